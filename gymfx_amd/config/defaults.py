@@ -1,0 +1,70 @@
+"""Canonical configuration defaults.
+
+Contract-compatible with the reference config surface (see
+/root/reference/app/config.py:1-47): same key names, same defaults, same
+precedence semantics (plugin-defaults < DEFAULT_VALUES < file < CLI-known
+< CLI-unknown).  Extra keys below the marker are MI355X-native additions
+(vectorized-env scale, PPO training, distributed) that the reference has no
+counterpart for.
+"""
+from __future__ import annotations
+
+DEFAULT_VALUES = {
+    # execution
+    "mode": "inference",  # training|optimization|inference
+    "driver_mode": "buy_hold",  # random|buy_hold|flat|replay
+    "steps": 500,
+
+    # plugin selection
+    "data_feed_plugin": "default_data_feed",
+    "broker_plugin": "default_broker",
+    "strategy_plugin": "default_strategy",
+    "preprocessor_plugin": "default_preprocessor",
+    "reward_plugin": "pnl_reward",
+    "metrics_plugin": "default_metrics",
+
+    # data + symbol
+    "input_data_file": "examples/data/eurusd.csv",
+    "date_column": "DATE_TIME",
+    "price_column": "CLOSE",
+    "instrument": "EUR_USD",
+    "timeframe": "M1",
+    "headers": True,
+    "max_rows": None,
+
+    # env and execution settings
+    "window_size": 32,
+    "initial_cash": 10000.0,
+    "position_size": 1.0,
+    "simulation_engine": "vectorized",   # vectorized (native) — reference had backtrader|nautilus
+    "execution_cost_profile": None,
+    "commission": 0.0,
+    "slippage": 0.0,
+
+    # optional replay actions
+    "replay_actions_file": None,
+
+    # config I/O
+    "remote_log": None,
+    "remote_load_config": None,
+    "remote_save_config": None,
+    "username": None,
+    "password": None,
+    "load_config": None,
+    "save_config": "./config_out.json",
+    "save_log": "./debug_out.json",
+    "results_file": "./results.json",
+    "quiet_mode": False,
+
+    # ------------------------------------------------------------------
+    # MI355X-native additions (no reference counterpart)
+    # ------------------------------------------------------------------
+    "n_envs": 1,                 # vectorized env count (SoA on device)
+    "device": "auto",            # auto|cpu|cuda
+    "seed": None,
+    "autoreset": False,          # auto-reset terminated envs (training)
+    "env_start_mode": "zero",    # zero|spread|random — per-env episode start offsets
+    "action_space_mode": "discrete",
+    "continuous_action_threshold": 0.33,
+    "min_equity": None,          # default: initial_cash * 0.01
+}

@@ -1,0 +1,299 @@
+"""Single-env Gymnasium-compatible wrapper over the vectorized engine.
+
+API parity with the reference GymFxEnv (/root/reference/app/env.py:93-328):
+same constructor signature (config + six plugins), same Dict observation
+contract, same info keys, same summary() shape.  The difference is under the
+hood: instead of a backtrader Cerebro on a worker thread synchronized with
+two events (app/bt_bridge.py:30-83), this wraps a VecFxEnv with N=1 whose
+step is a single vectorized transition (CPU torch here; the same state
+machine runs as one fused HIP kernel at N=4096+ on MI355X).
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, Optional
+
+import numpy as np
+import torch
+
+from .. import spaces
+from ..calendar import compute_fx_calendar_features, resolve_broker_metadata
+from .vec_env import VecFxEnv
+from .params import EnvParams
+
+
+def build_base_observation_space(config: Dict[str, Any], *, window_size: int) -> spaces.Dict:
+    """Observation-space contract of /root/reference/app/env.py:31-90."""
+    feature_columns = list(config.get("feature_columns") or [])
+    include_prices = bool(config.get("include_price_window", not feature_columns))
+    include_agent_state = bool(config.get("include_agent_state", True))
+    obs: Dict[str, Any] = {}
+    if feature_columns:
+        obs["features"] = spaces.Box(
+            low=-np.inf, high=np.inf, shape=(window_size, len(feature_columns)),
+            dtype=np.float32,
+        )
+    if include_prices:
+        obs["prices"] = spaces.Box(-np.inf, np.inf, shape=(window_size,), dtype=np.float32)
+        obs["returns"] = spaces.Box(-np.inf, np.inf, shape=(window_size,), dtype=np.float32)
+    if include_agent_state:
+        obs["position"] = spaces.Box(-1.0, 1.0, shape=(1,), dtype=np.float32)
+        obs["equity_norm"] = spaces.Box(-np.inf, np.inf, shape=(1,), dtype=np.float32)
+        obs["unrealized_pnl_norm"] = spaces.Box(-np.inf, np.inf, shape=(1,), dtype=np.float32)
+        obs["steps_remaining_norm"] = spaces.Box(0.0, 1.0, shape=(1,), dtype=np.float32)
+    if not obs:
+        raise ValueError("preprocessor observation contract emits no observation blocks")
+    return spaces.Dict(obs)
+
+
+_FC_KEYS = (
+    "bars_to_force_close",
+    "hours_to_force_close",
+    "is_force_close_zone",
+    "is_monday_entry_window",
+)
+
+_CAL_KEYS = (
+    "hours_to_fx_daily_break",
+    "bars_to_fx_daily_break",
+    "hours_to_friday_close",
+    "bars_to_friday_close",
+    "is_friday_risk_reduction_window",
+    "is_no_new_position_window",
+    "is_force_flat_window",
+    "is_broker_daily_break_near",
+    "broker_market_open",
+)
+
+
+class GymFxEnv(spaces.Env):
+    """Gymnasium-style FX trading env backed by the vectorized engine."""
+
+    metadata = {"render_modes": []}
+
+    def __init__(
+        self,
+        config: Dict[str, Any],
+        data_feed_plugin,
+        broker_plugin,
+        strategy_plugin,
+        preprocessor_plugin,
+        reward_plugin,
+        metrics_plugin,
+    ):
+        self.config = dict(config)
+        self.data_feed_plugin = data_feed_plugin
+        self.broker_plugin = broker_plugin
+        self.strategy_plugin = strategy_plugin
+        self.preprocessor_plugin = preprocessor_plugin
+        self.reward_plugin = reward_plugin
+        self.metrics_plugin = metrics_plugin
+
+        self.initial_cash = float(self.config.get("initial_cash", 10000.0))
+        self.position_size = float(self.config.get("position_size", 1.0))
+        self.window_size = int(self.config.get("window_size", 32))
+        self.price_column = self.config.get("price_column", "CLOSE")
+        self.min_equity = float(
+            self.config.get("min_equity") or self.initial_cash * 0.01
+        )
+
+        self.market_data = self.data_feed_plugin.load_data(self.config)
+        if self.market_data is None or len(self.market_data) < self.window_size + 2:
+            raise ValueError("input data is empty or too short for the configured window")
+        if not self.market_data.has_column(self.price_column):
+            raise ValueError(f"price_column '{self.price_column}' not found in data")
+        self.total_bars = len(self.market_data)
+
+        vec_config = dict(self.config)
+        vec_config["n_envs"] = 1
+        vec_config["autoreset"] = False
+        vec_config.setdefault("device", "cpu")
+        if hasattr(self.broker_plugin, "broker_params"):
+            bp = self.broker_plugin.broker_params(self.config)
+            vec_config["initial_cash"] = bp["initial_cash"]
+            vec_config["commission"] = bp["commission"]
+            vec_config["slippage"] = bp["slippage"]
+            vec_config["leverage"] = bp["leverage"]
+        self.vec = VecFxEnv(vec_config, self.market_data, use_native=False)
+        self.params: EnvParams = self.vec.params
+
+        # --- spaces -------------------------------------------------------
+        self.action_space_mode = str(self.config.get("action_space_mode", "discrete")).lower()
+        if self.action_space_mode == "continuous":
+            self.action_space = spaces.Box(low=-1.0, high=1.0, shape=(1,), dtype=np.float32)
+            self.continuous_action_threshold = float(
+                self.config.get("continuous_action_threshold", 0.33)
+            )
+        else:
+            self.action_space = spaces.Discrete(3)
+            self.continuous_action_threshold = None
+        self.observation_space = build_base_observation_space(
+            self.config, window_size=self.window_size
+        )
+        self.stage_b_force_close_obs = self.params.stage_b_force_close_obs
+        if self.stage_b_force_close_obs:
+            extra = {
+                "bars_to_force_close": spaces.Box(0.0, np.inf, shape=(1,), dtype=np.float32),
+                "hours_to_force_close": spaces.Box(0.0, np.inf, shape=(1,), dtype=np.float32),
+                "is_force_close_zone": spaces.Box(0.0, 1.0, shape=(1,), dtype=np.float32),
+                "is_monday_entry_window": spaces.Box(0.0, 1.0, shape=(1,), dtype=np.float32),
+            }
+            self.observation_space = spaces.Dict({**dict(self.observation_space.spaces), **extra})
+        self.oanda_fx_calendar_obs = self.params.oanda_fx_calendar_obs
+        if self.oanda_fx_calendar_obs:
+            extra = {k: spaces.Box(0.0, np.inf, shape=(1,), dtype=np.float32) for k in _CAL_KEYS}
+            extra["margin_closeout_percent"] = spaces.Box(0.0, np.inf, shape=(1,), dtype=np.float32)
+            extra["margin_available_norm"] = spaces.Box(0.0, np.inf, shape=(1,), dtype=np.float32)
+            self.observation_space = spaces.Dict({**dict(self.observation_space.spaces), **extra})
+
+        self._np_random = np.random.default_rng()
+        self._last_info_extras: Dict[str, Any] = {}
+        self._was_reset = False
+
+    # ------------------------------------------------------------------
+    def reset(self, *, seed: Optional[int] = None, options: Optional[Dict[str, Any]] = None):
+        if seed is not None:
+            self._np_random = np.random.default_rng(seed)
+        self.vec.reset(seed=seed)
+        self._was_reset = True
+        self._last_info_extras = {}
+        return self._make_observation(), self._make_info()
+
+    def step(self, action):
+        if not self._was_reset:
+            raise RuntimeError("Call reset() before step().")
+        raw = self._raw_action_value(action)
+        if self.action_space_mode == "continuous":
+            act_t = torch.tensor([raw], dtype=torch.float32)
+        else:
+            try:
+                act_t = torch.tensor([int(action)], dtype=torch.int64)
+            except (TypeError, ValueError):
+                act_t = torch.tensor([0], dtype=torch.int64)
+        out = self.vec.step(act_t)
+        reward = float(out["reward"][0].item())
+        base_reward = float(out["base_reward"][0].item())
+        penalty = float(out["force_close_reward_penalty"][0].item())
+        terminated = bool(out["terminated"][0].item())
+        bs = self.vec.bridge_state(0)
+        obs = self._make_observation()
+        info = self._make_info()
+        info.update(
+            reward=reward,
+            base_reward=base_reward,
+            force_close_reward_penalty=penalty,
+            pnl=bs["equity"] - bs["prev_equity"],
+            trade_cost=bs["last_trade_cost"],
+        )
+        return obs, reward, terminated, False, info
+
+    def close(self):
+        pass
+
+    def render(self):  # pragma: no cover
+        return None
+
+    # ------------------------------------------------------------------
+    def _raw_action_value(self, action) -> float:
+        try:
+            return float(np.asarray(action).reshape(-1)[0])
+        except Exception:
+            try:
+                return float(action)
+            except Exception:
+                return 0.0
+
+    def _make_observation(self) -> Dict[str, np.ndarray]:
+        bs = self.vec.bridge_state(0)
+        step_idx = max(0, min(bs["bar_index"], self.total_bars))
+        obs = self.preprocessor_plugin.make_observation(
+            data=self.market_data,
+            step=step_idx,
+            bridge_state={
+                "position": bs["position"],
+                "equity": bs["equity"],
+                "initial_cash": self.initial_cash,
+                "price": bs["price"],
+                "bar_index": bs["bar_index"],
+                "total_bars": self.total_bars,
+            },
+            config=self.config,
+        )
+        if self.stage_b_force_close_obs:
+            obs = dict(obs)
+            fc = self._force_close_features(step_idx)
+            for k in _FC_KEYS:
+                obs[k] = np.array([fc[k]], dtype=np.float32)
+        if self.oanda_fx_calendar_obs:
+            obs = dict(obs)
+            cal = self._oanda_calendar_features(step_idx)
+            for k in _CAL_KEYS:
+                obs[k] = np.array([cal[k]], dtype=np.float32)
+            obs["margin_closeout_percent"] = np.array([0.0], dtype=np.float32)
+            obs["margin_available_norm"] = np.array(
+                [bs["equity"] / (self.initial_cash or 1.0)], dtype=np.float32
+            )
+        return obs
+
+    def _force_close_features(self, step_idx: int) -> Dict[str, float]:
+        mt = self.vec.mt
+        if mt.force_close is None or mt.timestamps is None:
+            return {k: 0.0 for k in _FC_KEYS}
+        row = max(0, min(step_idx, self.total_bars - 1))
+        fc = mt.force_close[row]
+        return {
+            "bars_to_force_close": float(fc[0]),
+            "hours_to_force_close": float(fc[1]),
+            "is_force_close_zone": float(fc[2]),
+            "is_monday_entry_window": float(fc[3]),
+        }
+
+    def _oanda_calendar_features(self, step_idx: int) -> Dict[str, float]:
+        md = self.market_data
+        if md.timestamps is None:
+            ts = None
+        else:
+            row = max(0, min(step_idx, self.total_bars - 1))
+            ts = int(md.timestamps[row])
+        tf_h = float(md.timeframe_hours() or 1.0) or 1.0
+        return compute_fx_calendar_features(ts, timeframe_hours=tf_h)
+
+    def _make_info(self) -> Dict[str, Any]:
+        bs = self.vec.bridge_state(0)
+        info: Dict[str, Any] = {
+            "equity": bs["equity"],
+            "position": bs["position"],
+            "price": bs["price"],
+            "bar_index": bs["bar_index"],
+            "total_bars": self.total_bars,
+            "trades": bs["trade_count"],
+            "commission_paid": bs["commission_paid"],
+            "raw_action_value": None,
+            "coerced_action": None,
+            "action_diagnostics": self.vec.action_diagnostics(0),
+            "execution_diagnostics": self.vec.execution_diagnostics(0),
+        }
+        if self.stage_b_force_close_obs:
+            step_idx = max(0, min(bs["bar_index"], self.total_bars))
+            info.update(self._force_close_features(step_idx))
+        if self.oanda_fx_calendar_obs:
+            step_idx = max(0, min(bs["bar_index"], self.total_bars))
+            info.update(self._oanda_calendar_features(step_idx))
+            info["margin_closeout_percent"] = 0.0
+            info["margin_available_norm"] = bs["equity"] / (self.initial_cash or 1.0)
+            for k, v in resolve_broker_metadata(self.config).items():
+                if v is not None:
+                    info[k] = v
+        return info
+
+    def summary(self) -> Dict[str, Any]:
+        bs = self.vec.bridge_state(0)
+        summary = self.metrics_plugin.summarize(
+            initial_cash=self.initial_cash,
+            final_equity=bs["equity"],
+            analyzers=self.vec.analyzers(0),
+            config=self.config,
+        )
+        summary["action_diagnostics"] = self.vec.action_diagnostics(0)
+        summary["execution_diagnostics"] = self.vec.execution_diagnostics(0)
+        summary["event_context_diagnostics"] = dict(self._last_info_extras)
+        return summary

@@ -1,0 +1,224 @@
+"""VecFxEnv — N trading environments as SoA device tensors.
+
+The MI355X-native replacement for the reference's one-env-per-thread design
+(/root/reference/app/env.py:93 + app/bt_bridge.py): all N envs advance in
+lockstep; on GPU the whole step (action decode -> overlay -> broker fills ->
+brackets -> ATR -> strategy -> equity -> reward -> observation) is ONE fused
+HIP kernel (ops/csrc/env_step.hip); on CPU the same semantics run as
+vectorized torch ops (envs/reference_step.py — also the kernel's oracle).
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, Optional
+
+import numpy as np
+import torch
+
+from ..data.feed import MarketData
+from .market import MarketTensors, build_market_tensors
+from .params import EnvParams
+from .reference_step import build_obs_torch, coerce_actions, step_torch
+from .state import ACTION_COUNTERS, EXEC_COUNTERS, EnvState, alloc_state, reset_state_
+
+
+def resolve_device(spec: Any = "auto") -> torch.device:
+    if isinstance(spec, torch.device):
+        return spec
+    s = str(spec or "auto")
+    if s == "auto":
+        return torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+    return torch.device(s)
+
+
+class VecFxEnv:
+    """Vectorized FX trading environment (N envs, SoA state, device-resident)."""
+
+    def __init__(
+        self,
+        config: Dict[str, Any],
+        market_data: MarketData,
+        *,
+        device: Any = None,
+        use_native: Optional[bool] = None,
+    ):
+        self.config = dict(config)
+        self.market_data = market_data
+        self.device = resolve_device(device if device is not None else config.get("device", "auto"))
+        self.params = EnvParams.from_config(
+            self.config, timeframe_hours=market_data.timeframe_hours()
+        )
+        if market_data.n_rows < self.params.window_size + 2:
+            raise ValueError("input data is empty or too short for the configured window")
+        self.mt: MarketTensors = build_market_tensors(market_data, self.params, self.device)
+        self.st: EnvState = alloc_state(self.params, self.device)
+        self.total_bars = self.mt.T
+        self._rng = np.random.default_rng(self.params.seed)
+        self._native = None
+        if use_native is None:
+            use_native = self.device.type == "cuda"
+        if use_native:
+            # On a GPU the HIP extension is REQUIRED — never silently fall
+            # back to the eager torch path.
+            from ..ops import native  # noqa: PLC0415
+
+            self._native = native.require()
+        self._obs = torch.empty(
+            self.params.n_envs, self.params.obs_dim, dtype=torch.float32, device=self.device
+        )
+        self._assign_start_offsets()
+
+    # ------------------------------------------------------------------
+    @property
+    def n_envs(self) -> int:
+        return self.params.n_envs
+
+    @property
+    def obs_dim(self) -> int:
+        return self.params.obs_dim
+
+    def _assign_start_offsets(self) -> None:
+        N, T, W = self.params.n_envs, self.total_bars, self.params.window_size
+        mode = self.params.env_start_mode
+        max_off = max(0, T - W - 2)
+        if mode == "spread" and N > 1:
+            off = np.floor(np.linspace(0, max_off, N)).astype(np.int32)
+        elif mode == "random":
+            off = self._rng.integers(0, max_off + 1, size=N).astype(np.int32)
+        else:
+            off = np.zeros(N, dtype=np.int32)
+        self.st.start_offset = torch.from_numpy(off).to(self.device)
+
+    # ------------------------------------------------------------------
+    def reset(self, *, seed: Optional[int] = None) -> torch.Tensor:
+        if seed is not None:
+            self._rng = np.random.default_rng(seed)
+            if self.params.env_start_mode == "random":
+                self._assign_start_offsets()
+        mask = torch.ones(self.params.n_envs, dtype=torch.bool, device=self.device)
+        reset_state_(self.st, self.params, mask)
+        self._build_obs()
+        return self._obs
+
+    def _build_obs(self) -> None:
+        if self._native is not None:
+            self._native.build_obs(self.st, self.mt, self.params, self._obs)
+        else:
+            build_obs_torch(self.st, self.mt, self.params, out=self._obs)
+
+    def step(self, actions: torch.Tensor) -> Dict[str, torch.Tensor]:
+        """Advance all envs. Returns dict with obs/reward/terminated/info tensors."""
+        if not isinstance(actions, torch.Tensor):
+            actions = torch.as_tensor(actions, device=self.device)
+        actions = actions.to(self.device)
+        if self._native is not None:
+            info = self._native.env_step(self.st, self.mt, self.params, actions)
+        else:
+            info = step_torch(self.st, self.mt, self.params, actions)
+        terminated = info["terminated"]
+        if self.params.autoreset:
+            done = terminated.clone()
+            if bool(done.any()):
+                reset_state_(self.st, self.params, done)
+            info["terminated"] = done
+        self._build_obs()
+        info["obs"] = self._obs
+        return info
+
+    # ------------------------------------------------------------------
+    # Observability / summary
+    # ------------------------------------------------------------------
+    def bridge_state(self, i: int = 0) -> Dict[str, Any]:
+        """Reference-shaped per-env state view (BTBridge fields)."""
+        st = self.st
+        t = int(torch.clamp(st.cursor[i] - 1, min=0).item())
+        return {
+            "position": int(torch.sign(st.pos[i]).item()),
+            "equity": float(st.equity[i].item()),
+            "prev_equity": float(st.prev_equity[i].item()),
+            "initial_cash": self.params.initial_cash,
+            "price": float(self.mt.close[t].item()),
+            "bar_index": int(st.cursor[i].item()),
+            "total_bars": self.total_bars,
+            "trade_count": int(st.trade_count[i].item()),
+            "commission_paid": float(st.commission_paid[i].item()),
+            "last_trade_cost": float(st.last_trade_cost[i].item()),
+            "terminated": bool(st.terminated[i].item()),
+        }
+
+    def action_diagnostics(self, i: int = 0) -> Dict[str, Any]:
+        d = {
+            name: int(self.st.act_diag[i, k].item())
+            for k, name in enumerate(ACTION_COUNTERS)
+        }
+        d["raw_abs_sum"] = float(self.st.raw_abs_sum[i].item())
+        rmin = float(self.st.raw_min[i].item())
+        rmax = float(self.st.raw_max[i].item())
+        d["raw_min"] = None if rmin == float("inf") else rmin
+        d["raw_max"] = None if rmax == float("-inf") else rmax
+        d["continuous_action_threshold"] = (
+            self.params.continuous_action_threshold
+            if self.params.action_space_mode == "continuous"
+            else None
+        )
+        return d
+
+    def execution_diagnostics(self, i: int = 0) -> Dict[str, int]:
+        return {
+            name: int(self.st.exec_diag[i, k].item())
+            for k, name in enumerate(EXEC_COUNTERS)
+        }
+
+    def analyzers(self, i: int = 0) -> Dict[str, Any]:
+        """Analyzer-equivalents for the metrics plugin contract
+        (trades/sharpe/drawdown/sqn digest, bt_bridge.py:277-281)."""
+        st = self.st
+        n_tr = int(st.trade_count[i].item())
+        pnl_sum = float(st.trade_pnl_sum[i].item())
+        pnl_sumsq = float(st.trade_pnl_sumsq[i].item())
+        sqn = None
+        if n_tr >= 2:
+            mean = pnl_sum / n_tr
+            var = max(pnl_sumsq / n_tr - mean * mean, 0.0)
+            std = var ** 0.5
+            if std > 0:
+                sqn = (n_tr ** 0.5) * mean / std
+        n_ret = int(st.ret_count[i].item())
+        sharpe = None
+        if n_ret >= 2:
+            rs = float(st.ret_sum[i].item())
+            rss = float(st.ret_sumsq[i].item())
+            mean = rs / n_ret
+            var = max((rss - n_ret * mean * mean) / (n_ret - 1), 0.0)
+            std = var ** 0.5
+            if std > 0:
+                ann = float(self.config.get("annualization_factor", 252.0))
+                sharpe = mean / std * (ann ** 0.5)
+        return {
+            "trades": {
+                "total": {"total": n_tr},
+                "won": {"total": int(st.trade_won[i].item())},
+                "lost": {"total": int(st.trade_lost[i].item())},
+                "pnl": {"net": {"average": (pnl_sum / n_tr) if n_tr else None}},
+            },
+            "sharpe": {"sharperatio": sharpe},
+            "drawdown": {
+                "max": {
+                    "drawdown": float(st.max_dd_pct[i].item()),
+                    "moneydown": float(st.max_dd_money[i].item()),
+                }
+            },
+            "sqn": {"sqn": sqn},
+            "time_return": {},
+        }
+
+    def vec_summary(self) -> Dict[str, Any]:
+        """Whole-fleet aggregates (no reference counterpart)."""
+        st = self.st
+        return {
+            "n_envs": self.params.n_envs,
+            "mean_equity": float(st.equity.mean().item()),
+            "mean_episode_return": float(st.episode_return.mean().item()),
+            "mean_episode_steps": float(st.episode_step.float().mean().item()),
+            "terminated_envs": int(st.terminated.sum().item()),
+            "total_trades": int(st.trade_count.sum().item()),
+        }

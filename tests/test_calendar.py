@@ -1,0 +1,64 @@
+"""Calendar: scalar predicates (reference semantics app/oanda_calendar.py)
+and scalar-vs-vectorized-table equivalence across DST boundaries."""
+import datetime as dt
+
+import numpy as np
+import pytest
+
+from gymfx_amd.calendar import (
+    CALENDAR_FEATURE_KEYS,
+    broker_market_open,
+    compute_fx_calendar_features,
+    compute_fx_calendar_table,
+    is_force_flat_window,
+    is_no_new_position_window,
+    _to_ny,
+)
+
+NY = dt.timezone.utc  # inputs below are UTC epoch values
+
+
+def _epoch(y, m, d, hh, mm):
+    return int(dt.datetime(y, m, d, hh, mm, tzinfo=dt.timezone.utc).timestamp())
+
+
+def test_scalar_predicates():
+    # 2024-01-05 is a Friday. 19:30 UTC = 14:30 NY (EST, UTC-5).
+    ts = _to_ny(_epoch(2024, 1, 5, 19, 30))
+    assert ts.weekday() == 4
+    assert is_no_new_position_window(ts)
+    assert not is_force_flat_window(ts)
+    # 20:50 UTC = 15:50 NY -> force flat
+    ts2 = _to_ny(_epoch(2024, 1, 5, 20, 50))
+    assert is_force_flat_window(ts2)
+    # Saturday closed
+    assert not broker_market_open(_to_ny(_epoch(2024, 1, 6, 12, 0)))
+    # Sunday 22:30 UTC = 17:30 NY -> open
+    assert broker_market_open(_to_ny(_epoch(2024, 1, 7, 22, 30)))
+    # Monday daily break 21:59-22:05 UTC (EST): 22:00 UTC = 17:00 NY
+    assert not broker_market_open(_to_ny(_epoch(2024, 1, 8, 22, 0)))
+
+
+@pytest.mark.parametrize(
+    "start",
+    [
+        _epoch(2024, 1, 3, 0, 0),    # EST
+        _epoch(2024, 3, 9, 12, 0),   # spring-forward DST transition window
+        _epoch(2024, 11, 2, 12, 0),  # fall-back DST transition window
+        _epoch(2024, 7, 1, 0, 0),    # EDT
+    ],
+)
+def test_table_matches_scalar(start):
+    ts = start + np.arange(0, 96 * 3600, 1800, dtype=np.int64)  # 4 days, 30-min grid
+    table = compute_fx_calendar_table(ts, timeframe_hours=1.0)
+    for i in range(0, len(ts), 7):
+        scalar = compute_fx_calendar_features(int(ts[i]), timeframe_hours=1.0)
+        for j, key in enumerate(CALENDAR_FEATURE_KEYS):
+            assert table[i, j] == pytest.approx(scalar[key], abs=1e-4), (
+                f"{key} mismatch at ts={ts[i]}"
+            )
+
+
+def test_neutral_on_unparseable():
+    out = compute_fx_calendar_features("not-a-date", timeframe_hours=1.0)
+    assert all(v == 0.0 for v in out.values())

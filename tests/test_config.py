@@ -1,0 +1,58 @@
+"""Config precedence + typed coercion (reference semantics:
+app/config_merger.py:19-51, app/config_handler.py:11-24)."""
+import json
+
+from gymfx_amd.config import (
+    DEFAULT_VALUES,
+    compose_config,
+    convert_type,
+    merge_config,
+    process_unknown_args,
+)
+
+
+def test_convert_type():
+    assert convert_type("true") is True
+    assert convert_type("False") is False
+    assert convert_type("none") is None
+    assert convert_type("3") == 3
+    assert convert_type("3.5") == 3.5
+    assert convert_type("hello") == "hello"
+    assert convert_type(7) == 7
+
+
+def test_unknown_args_parsing():
+    out = process_unknown_args(["--alpha", "0.5", "--flag", "--name", "x"])
+    assert out == {"alpha": "0.5", "flag": True, "name": "x"}
+
+
+def test_merge_precedence():
+    defaults = {"a": 1, "b": 2, "c": 3}
+    plugin = {"a": 0, "z": 9}
+    file_cfg = {"b": 20}
+    cli = {"c": 30, "d": None}
+    unknown = {"e": "4.5"}
+    merged = merge_config(defaults, plugin, {}, file_cfg, cli, unknown)
+    assert merged["a"] == 1          # defaults beat plugin params
+    assert merged["b"] == 20         # file beats defaults
+    assert merged["c"] == 30         # cli beats file
+    assert "d" not in merged         # None cli args are skipped
+    assert merged["e"] == 4.5        # unknown args are type-coerced
+    assert merged["z"] == 9          # plugin-only keys survive
+
+
+def test_compose_config_saves_only_non_defaults(tmp_path):
+    cfg = dict(DEFAULT_VALUES)
+    cfg["steps"] = 123
+    cfg["custom_key"] = "abc"
+    out = compose_config(cfg)
+    assert out["steps"] == 123
+    assert out["custom_key"] == "abc"
+    assert "window_size" not in out  # unchanged default not saved
+
+    from gymfx_amd.config import save_config
+
+    path = tmp_path / "cfg.json"
+    save_config(cfg, str(path))
+    reloaded = json.loads(path.read_text())
+    assert reloaded == {k: v for k, v in out.items()}
