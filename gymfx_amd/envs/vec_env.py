@@ -54,6 +54,10 @@ class VecFxEnv:
         self.total_bars = self.mt.T
         self._rng = np.random.default_rng(self.params.seed)
         self._native = None
+        self._obs = torch.empty(
+            self.params.n_envs, self.params.obs_dim, dtype=torch.float32, device=self.device
+        )
+        self._assign_start_offsets()
         if use_native is None:
             use_native = self.device.type == "cuda"
         if use_native:
@@ -61,11 +65,10 @@ class VecFxEnv:
             # back to the eager torch path.
             from ..ops import native  # noqa: PLC0415
 
-            self._native = native.require()
-        self._obs = torch.empty(
-            self.params.n_envs, self.params.obs_dim, dtype=torch.float32, device=self.device
-        )
-        self._assign_start_offsets()
+            native.require()
+            from ..ops.wrappers import NativeEngine  # noqa: PLC0415
+
+            self._native = NativeEngine(self.st, self.mt, self.params)
 
     # ------------------------------------------------------------------
     @property
@@ -101,7 +104,7 @@ class VecFxEnv:
 
     def _build_obs(self) -> None:
         if self._native is not None:
-            self._native.build_obs(self.st, self.mt, self.params, self._obs)
+            self._native.build_obs(self._obs)
         else:
             build_obs_torch(self.st, self.mt, self.params, out=self._obs)
 
@@ -111,9 +114,11 @@ class VecFxEnv:
             actions = torch.as_tensor(actions, device=self.device)
         actions = actions.to(self.device)
         if self._native is not None:
-            info = self._native.env_step(self.st, self.mt, self.params, actions)
-        else:
-            info = step_torch(self.st, self.mt, self.params, actions)
+            # fused HIP path: step + autoreset + obs in two kernel launches
+            info = self._native.step(actions, self._obs)
+            info["obs"] = self._obs
+            return info
+        info = step_torch(self.st, self.mt, self.params, actions)
         terminated = info["terminated"]
         if self.params.autoreset:
             done = terminated.clone()

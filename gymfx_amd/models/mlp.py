@@ -1,0 +1,196 @@
+"""Actor-critic MLP(256,256) — bf16 MFMA compute, f32 master weights.
+
+North-star model (BASELINE.json config #2): obs -> tanh(256) -> tanh(256)
+-> fused head [n_actions logits | value].  Forward/backward run entirely on
+the hand-written MFMA GEMM kernels (ops/api.gemm / wgrad); the optimizer is
+the fused Adam kernel over one flat parameter buffer (single RCCL bucket in
+data-parallel runs).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from ..ops import api
+
+
+@dataclass
+class _ParamSlice:
+    name: str
+    shape: Tuple[int, ...]
+    sl: slice
+    is_weight: bool  # weights get a bf16 mirror; biases stay f32
+
+
+class ActorCriticMLP:
+    def __init__(
+        self,
+        obs_dim: int,
+        n_actions: int = 3,
+        hidden: int = 256,
+        *,
+        device: torch.device,
+        seed: int = 0,
+    ):
+        self.obs_dim = obs_dim
+        self.n_actions = n_actions
+        self.hidden = hidden
+        self.head_dim = n_actions + 1
+        self.device = device
+
+        dims = [
+            ("W1", (obs_dim, hidden), True),
+            ("b1", (hidden,), False),
+            ("W2", (hidden, hidden), True),
+            ("b2", (hidden,), False),
+            ("W3", (hidden, self.head_dim), True),
+            ("b3", (self.head_dim,), False),
+        ]
+        self.slices: Dict[str, _ParamSlice] = {}
+        off = 0
+        for name, shape, is_w in dims:
+            n = int(torch.tensor(shape).prod())
+            self.slices[name] = _ParamSlice(name, shape, slice(off, off + n), is_w)
+            off += n
+        self.n_params = off
+
+        g = torch.Generator().manual_seed(seed)
+        flat = torch.empty(off, dtype=torch.float32)
+        for name, shape, is_w in dims:
+            s = self.slices[name]
+            if is_w:
+                fan_in = shape[0]
+                bound = 1.0 / math.sqrt(fan_in)
+                flat[s.sl] = (torch.rand(s.sl.stop - s.sl.start, generator=g) * 2 - 1) * bound
+            else:
+                flat[s.sl] = 0.0
+        self.params = flat.to(device)
+        self.grads = torch.zeros_like(self.params)
+        self.m = torch.zeros_like(self.params)
+        self.v = torch.zeros_like(self.params)
+        # bf16 mirrors of the weight matrices (GEMM operands)
+        self.params_bf16 = self.params.to(torch.bfloat16)
+        self.adam_step = 0
+        # grad-clip workspace
+        self._clip_part = torch.zeros(256, dtype=torch.float32, device=device)
+        self._clip_scale = torch.ones(1, dtype=torch.float32, device=device)
+        # wgrad slab workspaces (allocated lazily per layer shape)
+        self._wg_ws: Dict[str, Tuple[torch.Tensor, Optional[torch.Tensor]]] = {}
+        self.wgrad_slabs = 64
+
+    # -- param views ----------------------------------------------------
+    def w(self, name: str) -> torch.Tensor:
+        s = self.slices[name]
+        return self.params_bf16[s.sl].view(*s.shape)
+
+    def f32(self, name: str) -> torch.Tensor:
+        s = self.slices[name]
+        return self.params[s.sl].view(*s.shape)
+
+    def grad(self, name: str) -> torch.Tensor:
+        s = self.slices[name]
+        return self.grads[s.sl].view(*s.shape)
+
+    def _wg_workspace(self, name: str, K: int, N: int, want_db: bool):
+        key = f"{name}:{K}x{N}"
+        if key not in self._wg_ws:
+            S = self.wgrad_slabs
+            dw = torch.empty(S, K, N, dtype=torch.float32, device=self.device)
+            db = torch.empty(S, N, dtype=torch.float32, device=self.device) if want_db else None
+            self._wg_ws[key] = (dw, db)
+        return self._wg_ws[key]
+
+    # -- forward ---------------------------------------------------------
+    def alloc_acts(self, M: int) -> Dict[str, torch.Tensor]:
+        dev = self.device
+        return {
+            "h1": torch.empty(M, self.hidden, dtype=torch.bfloat16, device=dev),
+            "h2": torch.empty(M, self.hidden, dtype=torch.bfloat16, device=dev),
+            "head": torch.empty(M, self.head_dim, dtype=torch.float32, device=dev),
+        }
+
+    def forward(self, obs_bf16: torch.Tensor, acts: Dict[str, torch.Tensor]) -> torch.Tensor:
+        """obs_bf16 [M, obs_dim] -> head f32 [M, A+1]; saves h1/h2 for bwd."""
+        api.gemm(obs_bf16, self.w("W1"), self.f32("b1"), acts["h1"], act=2)
+        api.gemm(acts["h1"], self.w("W2"), self.f32("b2"), acts["h2"], act=2)
+        api.gemm(acts["h2"], self.w("W3"), self.f32("b3"), acts["head"], act=0)
+        return acts["head"]
+
+    # -- backward (dhead [M, A+1] bf16 -> accumulate grads) --------------
+    def backward(
+        self,
+        obs_bf16: torch.Tensor,
+        acts: Dict[str, torch.Tensor],
+        dhead: torch.Tensor,
+        scratch: Dict[str, torch.Tensor],
+    ) -> None:
+        M = obs_bf16.shape[0]
+        dh2 = scratch["dh2"]
+        dh1 = scratch["dh1"]
+        # head layer
+        api.wgrad(
+            acts["h2"], dhead, self.grad("W3"), self.grad("b3"),
+            workspace=self._wg_workspace("W3", self.hidden, self.head_dim, True),
+            slabs=self.wgrad_slabs,
+        )
+        api.gemm(dhead, self.w("W3"), None, dh2, Yact=acts["h2"], trans_b=True,
+                 act=1, dact_tanh=True)
+        # layer 2
+        api.wgrad(
+            acts["h1"], dh2, self.grad("W2"), self.grad("b2"),
+            workspace=self._wg_workspace("W2", self.hidden, self.hidden, True),
+            slabs=self.wgrad_slabs,
+        )
+        api.gemm(dh2, self.w("W2"), None, dh1, Yact=acts["h1"], trans_b=True,
+                 act=1, dact_tanh=True)
+        # layer 1
+        api.wgrad(
+            obs_bf16, dh1, self.grad("W1"), self.grad("b1"),
+            workspace=self._wg_workspace("W1", self.obs_dim, self.hidden, True),
+            slabs=self.wgrad_slabs,
+        )
+
+    def alloc_scratch(self, M: int) -> Dict[str, torch.Tensor]:
+        dev = self.device
+        return {
+            "dh2": torch.empty(M, self.hidden, dtype=torch.bfloat16, device=dev),
+            "dh1": torch.empty(M, self.hidden, dtype=torch.bfloat16, device=dev),
+        }
+
+    # -- optimizer --------------------------------------------------------
+    def adam(self, lr: float, *, beta1=0.9, beta2=0.999, eps=1e-8,
+             max_grad_norm: float = 0.0) -> None:
+        self.adam_step += 1
+        gscale = None
+        if max_grad_norm and max_grad_norm > 0:
+            api.grad_clip_scale(self.grads, max_grad_norm, self._clip_part,
+                                self._clip_scale)
+            gscale = self._clip_scale
+        api.adam(self.params, self.grads, self.m, self.v, self.params_bf16,
+                 lr=lr, beta1=beta1, beta2=beta2, eps=eps, step=self.adam_step,
+                 gscale=gscale)
+
+    def zero_grad(self) -> None:
+        self.grads.zero_()
+
+    # -- checkpoint -------------------------------------------------------
+    def state_dict(self) -> Dict[str, torch.Tensor]:
+        return {
+            "params": self.params.detach().cpu(),
+            "m": self.m.detach().cpu(),
+            "v": self.v.detach().cpu(),
+            "adam_step": torch.tensor(self.adam_step),
+            "obs_dim": torch.tensor(self.obs_dim),
+            "n_actions": torch.tensor(self.n_actions),
+            "hidden": torch.tensor(self.hidden),
+        }
+
+    def load_state_dict(self, sd: Dict[str, torch.Tensor]) -> None:
+        self.params.copy_(sd["params"].to(self.device))
+        self.m.copy_(sd["m"].to(self.device))
+        self.v.copy_(sd["v"].to(self.device))
+        self.adam_step = int(sd["adam_step"])
+        self.params_bf16.copy_(self.params.to(torch.bfloat16))

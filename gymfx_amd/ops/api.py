@@ -1,0 +1,272 @@
+"""Op dispatch: HIP kernels on GPU, torch reference implementations on CPU.
+
+The torch implementations are the ORACLES — GPU tests assert the HIP kernel
+matches them.  On a CUDA/ROCm device the native extension is required
+(fail-loud, no silent eager fallback): see ops/native.py.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import native
+
+
+def _use_native(t: torch.Tensor) -> bool:
+    if t.is_cuda:
+        native.require()
+        return True
+    return False
+
+
+# ---------------------------------------------------------------------------
+# GEMM: C = act(A @ B (+bias));  A bf16 [M,K], B bf16 [K,N] (or [N,K] when
+# trans_b); act: 0 = f32 out, 1 = bf16 out, 2 = tanh bf16 out.
+# dact_tanh: multiply the product by (1 - Yact^2) (fused tanh backward).
+# ---------------------------------------------------------------------------
+
+def gemm(
+    A: torch.Tensor,
+    B: torch.Tensor,
+    bias: Optional[torch.Tensor],
+    C: torch.Tensor,
+    *,
+    Yact: Optional[torch.Tensor] = None,
+    trans_b: bool = False,
+    act: int = 1,
+    dact_tanh: bool = False,
+) -> torch.Tensor:
+    if _use_native(A):
+        native.require().gemm(A, B, bias, C, Yact, trans_b, act, dact_tanh)
+        return C
+    a = A.to(torch.float32)
+    b = B.to(torch.float32)
+    if trans_b:
+        b = b.t()
+    out = a @ b
+    if bias is not None:
+        out = out + bias
+    if act == 2:
+        out = torch.tanh(out)
+    if dact_tanh:
+        y = Yact.to(torch.float32)
+        out = out * (1.0 - y * y)
+    if act == 0:
+        C.copy_(out)
+    else:
+        C.copy_(out.to(torch.bfloat16))
+    return C
+
+
+def wgrad(
+    X: torch.Tensor,
+    dY: torch.Tensor,
+    dW: torch.Tensor,
+    db: Optional[torch.Tensor],
+    *,
+    workspace: Optional[Tuple[torch.Tensor, Optional[torch.Tensor]]] = None,
+    slabs: int = 64,
+) -> None:
+    """dW[K,N] = X^T @ dY ; db[N] = colsum(dY). Deterministic split-M."""
+    if _use_native(X):
+        K, N = dW.shape
+        if workspace is None:
+            dW_part = torch.empty(slabs, K, N, dtype=torch.float32, device=X.device)
+            db_part = (
+                torch.empty(slabs, N, dtype=torch.float32, device=X.device)
+                if db is not None
+                else None
+            )
+        else:
+            dW_part, db_part = workspace
+        native.require().wgrad(X, dY, dW_part, db_part, dW, db, slabs)
+        return
+    x = X.to(torch.float32)
+    dy = dY.to(torch.float32)
+    dW.copy_(x.t() @ dy)
+    if db is not None:
+        db.copy_(dy.sum(dim=0))
+
+
+def gae(
+    rewards: torch.Tensor,
+    values: torch.Tensor,
+    dones: torch.Tensor,
+    adv: torch.Tensor,
+    ret: torch.Tensor,
+    gamma: float,
+    lam: float,
+) -> None:
+    if _use_native(rewards):
+        native.require().gae(rewards, values, dones, adv, ret, gamma, lam)
+        return
+    T, N = rewards.shape
+    running = torch.zeros(N, dtype=torch.float32)
+    for t in range(T - 1, -1, -1):
+        nonterm = (~dones[t]).to(torch.float32)
+        delta = rewards[t] + gamma * values[t + 1] * nonterm - values[t]
+        running = delta + gamma * lam * nonterm * running
+        adv[t] = running
+        ret[t] = running + values[t]
+
+
+def adam(
+    p: torch.Tensor,
+    g: torch.Tensor,
+    m: torch.Tensor,
+    v: torch.Tensor,
+    p_bf16: Optional[torch.Tensor],
+    *,
+    lr: float,
+    beta1: float = 0.9,
+    beta2: float = 0.999,
+    eps: float = 1e-8,
+    step: int = 1,
+    gscale: Optional[torch.Tensor] = None,
+) -> None:
+    if _use_native(p):
+        native.require().adam(p, g, m, v, p_bf16, lr, beta1, beta2, eps, step, gscale)
+        return
+    s = float(gscale.item()) if gscale is not None else 1.0
+    geff = g * s
+    m.mul_(beta1).add_(geff, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(geff, geff, value=1 - beta2)
+    mhat = m / (1 - beta1 ** step)
+    vhat = v / (1 - beta2 ** step)
+    p.addcdiv_(mhat, vhat.sqrt() + eps, value=-lr)
+    if p_bf16 is not None:
+        p_bf16.copy_(p.to(torch.bfloat16))
+
+
+def grad_clip_scale(
+    g: torch.Tensor, max_norm: float, part: torch.Tensor, scale: torch.Tensor
+) -> None:
+    """scale = min(1, max_norm / ||g||) written into `scale` (device scalar)."""
+    if _use_native(g):
+        native.require().grad_clip(g, max_norm, part, scale)
+        return
+    norm = g.norm()
+    s = max_norm / norm if (max_norm > 0 and norm > max_norm) else torch.ones(())
+    scale.fill_(float(s))
+
+
+def sample_head(
+    head: torch.Tensor,
+    seed: int,
+    step: int,
+    actions: torch.Tensor,
+    logp: torch.Tensor,
+    value: Optional[torch.Tensor] = None,
+    entropy: Optional[torch.Tensor] = None,
+    greedy: bool = False,
+) -> None:
+    """head [M, A+1] f32 (logits | value) -> categorical sample + logp."""
+    if _use_native(head):
+        native.require().sample_head(head, seed, step, actions, logp, value, entropy, greedy)
+        return
+    M, W = head.shape
+    A = W - 1
+    logits = head[:, :A]
+    logz = torch.logsumexp(logits, dim=1, keepdim=True)
+    logpi = logits - logz
+    pi = logpi.exp()
+    if greedy:
+        a = logits.argmax(dim=1)
+    else:
+        # same counter-based RNG as the kernel (splitmix64)
+        u = _splitmix_uniform(seed, step, M)
+        cdf = pi.cumsum(dim=1)
+        a = (u.unsqueeze(1) >= cdf).sum(dim=1).clamp(max=A - 1)
+    actions.copy_(a)
+    logp.copy_(logpi.gather(1, a.unsqueeze(1)).squeeze(1))
+    if value is not None:
+        value.copy_(head[:, A])
+    if entropy is not None:
+        entropy.copy_(-(pi * logpi).sum(dim=1))
+
+
+def _splitmix_uniform(seed: int, step: int, M: int) -> torch.Tensor:
+    m = torch.arange(M, dtype=torch.int64)
+    x = (seed ^ (step * 0x51E1F5 + m * 0x9E37)) & 0xFFFFFFFFFFFFFFFF
+
+    def mix(x):
+        mask = (1 << 64) - 1
+        x = (x + 0x9E3779B97F4A7C15) & mask
+        x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & mask
+        x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & mask
+        return x ^ (x >> 31)
+
+    # python-int loop for exact uint64 semantics (CPU oracle only)
+    vals = [mix(int(v)) for v in x.tolist()]
+    u = torch.tensor([(v >> 11) * (1.0 / 9007199254740992.0) for v in vals],
+                     dtype=torch.float32)
+    return u.clamp(max=0.999999)
+
+
+def ppo_loss_bwd(
+    head: torch.Tensor,
+    actions: torch.Tensor,
+    old_logp: torch.Tensor,
+    adv: torch.Tensor,
+    ret: torch.Tensor,
+    dhead: torch.Tensor,
+    *,
+    clip_eps: float,
+    ent_coef: float,
+    vf_coef: float,
+    inv_count: float,
+    losses: Optional[torch.Tensor] = None,
+) -> None:
+    """Clipped-PPO backward: writes dhead (bf16) and accumulates logging
+    scalars [pi_loss, v_loss, entropy, approx_kl, clipfrac] into `losses`."""
+    if _use_native(head):
+        native.require().ppo_loss_bwd(
+            head, actions, old_logp, adv, ret, dhead, clip_eps, ent_coef,
+            vf_coef, inv_count, losses,
+        )
+        return
+    M, W = head.shape
+    A = W - 1
+    logits = head[:, :A]
+    logz = torch.logsumexp(logits, dim=1, keepdim=True)
+    logpi = logits - logz
+    pi = logpi.exp()
+    lp = logpi.gather(1, actions.unsqueeze(1)).squeeze(1)
+    ratio = (lp - old_logp).exp()
+    surr1 = ratio * adv
+    rclip = ratio.clamp(1 - clip_eps, 1 + clip_eps)
+    surr2 = rclip * adv
+    g_lp = torch.where(surr1 <= surr2, -adv * ratio, torch.zeros_like(ratio))
+    onehot = torch.nn.functional.one_hot(actions, A).to(torch.float32)
+    H = -(pi * logpi).sum(dim=1)
+    dlogits = g_lp.unsqueeze(1) * (onehot - pi)
+    dlogits = dlogits + ent_coef * pi * (logpi + H.unsqueeze(1))
+    v = head[:, A]
+    dv = vf_coef * (v - ret)
+    d = torch.cat([dlogits, dv.unsqueeze(1)], dim=1) * inv_count
+    dhead.copy_(d.to(torch.bfloat16))
+    if losses is not None:
+        losses[0] += float((-torch.minimum(surr1, surr2)).sum()) * inv_count
+        losses[1] += float((0.5 * (v - ret) ** 2).sum()) * inv_count
+        losses[2] += float(H.sum()) * inv_count
+        losses[3] += float((old_logp - lp).sum()) * inv_count
+        clipped = (ratio > 1 + clip_eps) | (ratio < 1 - clip_eps)
+        losses[4] += float(clipped.to(torch.float32).sum()) * inv_count
+
+
+def adv_normalize(adv: torch.Tensor, part: torch.Tensor) -> None:
+    if _use_native(adv):
+        native.require().adv_normalize(adv, part)
+        return
+    mean = adv.mean()
+    # population std (matches kernel): sqrt(E[x^2] - mean^2)
+    var = (adv * adv).mean() - mean * mean
+    adv.sub_(mean).div_(var.clamp(min=0).sqrt() + 1e-8)
+
+
+def f32_to_bf16(src: torch.Tensor, dst: torch.Tensor) -> None:
+    if _use_native(src):
+        native.require().f32_to_bf16(src, dst)
+        return
+    dst.copy_(src.to(torch.bfloat16))

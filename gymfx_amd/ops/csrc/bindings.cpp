@@ -1,0 +1,451 @@
+// pybind11 bindings: GymFxEngine holds raw device pointers + the param pack
+// so the per-step hot path does zero dict lookups / tensor re-validation.
+// Built in-tree via torch.utils.cpp_extension (hipcc, gfx950).
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <unordered_map>
+#include <string>
+#include <vector>
+
+#include "env_common.h"
+
+namespace gymfx {
+
+void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream);
+void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream);
+void launch_gemm(const void* A, const void* B, const float* bias, void* C,
+                 const void* Yact, int M, int N, int K, bool trans_b, int act,
+                 bool dact_tanh, bool add_bias, hipStream_t stream);
+void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
+                  float* dW, float* db, int M, int N, int K, int slabs,
+                  hipStream_t stream);
+void launch_gae(const float* rewards, const float* values, const bool* dones,
+                float* adv, float* ret, int T, int N, float gamma, float lam,
+                hipStream_t stream);
+void launch_adam(float* p, const float* g, float* m, float* v, void* p_bf16,
+                 int64_t n, float lr, float beta1, float beta2, float eps,
+                 float bc1, float bc2, const float* gscale, hipStream_t stream);
+void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
+                      float* scale, int nparts, hipStream_t stream);
+void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,
+                        uint64_t step, int64_t* actions, float* logp,
+                        float* value, float* entropy, int greedy,
+                        hipStream_t stream);
+void launch_ppo_loss_bwd(const float* head, const int64_t* actions,
+                         const float* old_logp, const float* adv,
+                         const float* ret, void* dhead, int M, int n_actions,
+                         float clip_eps, float ent_coef, float vf_coef,
+                         float inv_count, float* losses, hipStream_t stream);
+void launch_adv_normalize(float* adv, int64_t n, float* part, int nparts,
+                          hipStream_t stream);
+void launch_f32_to_bf16(const float* in, void* out, int64_t n,
+                        hipStream_t stream);
+
+namespace {
+
+using TensorMap = std::unordered_map<std::string, torch::Tensor>;
+
+template <typename T>
+T* ptr(TensorMap& m, const char* key, bool required = true) {
+  auto it = m.find(key);
+  if (it == m.end() || !it->second.defined()) {
+    TORCH_CHECK(!required, "GymFxEngine: missing tensor '", key, "'");
+    return nullptr;
+  }
+  return it->second.data_ptr<T>();
+}
+
+}  // namespace
+
+struct GymFxEngine {
+  EnvPtrs P{};
+  EnvParamsK K{};
+  TensorMap state, market, outputs;
+  bool continuous = false;
+
+  GymFxEngine(const py::dict& iparams, const py::dict& fparams,
+              const py::dict& state_d, const py::dict& market_d) {
+    for (auto item : state_d)
+      state[py::cast<std::string>(item.first)] = py::cast<torch::Tensor>(item.second);
+    for (auto item : market_d)
+      market[py::cast<std::string>(item.first)] = py::cast<torch::Tensor>(item.second);
+
+    auto gi = [&](const char* k) { return py::cast<int64_t>(iparams[k]); };
+    auto gf = [&](const char* k) { return py::cast<double>(fparams[k]); };
+
+    K.n_envs = (int)gi("n_envs");
+    K.T = (int)gi("T");
+    K.window = (int)gi("window");
+    K.n_features = (int)gi("n_features");
+    K.reward_id = (int)gi("reward_id");
+    K.strategy_id = (int)gi("strategy_id");
+    K.prep_id = (int)gi("prep_id");
+    K.scaling_mode = (int)gi("scaling_mode");
+    K.scale_window = (int)gi("scale_window");
+    K.sharpe_window = (int)gi("sharpe_window");
+    K.atr_period = (int)gi("atr_period");
+    K.size_mode = (int)gi("size_mode");
+    K.risk_mode = (int)gi("risk_mode");
+    K.flags = (int)gi("flags");
+    K.obs_dim = (int)gi("obs_dim");
+    K.off_features = (int)gi("off_features");
+    K.off_prices = (int)gi("off_prices");
+    K.off_returns = (int)gi("off_returns");
+    K.off_agent = (int)gi("off_agent");
+    K.off_fc = (int)gi("off_fc");
+    K.off_cal = (int)gi("off_cal");
+    continuous = (K.flags & F_CONTINUOUS) != 0;
+
+    K.initial_cash = gf("initial_cash");
+    K.position_size = gf("position_size");
+    K.commission = gf("commission");
+    K.slippage = gf("slippage");
+    K.leverage = gf("leverage");
+    K.min_equity = gf("min_equity");
+    K.cont_threshold = gf("cont_threshold");
+    K.reward_scale = gf("reward_scale");
+    K.annualization = gf("annualization");
+    K.penalty_lambda = gf("penalty_lambda");
+    K.sl_pips = gf("sl_pips");
+    K.tp_pips = gf("tp_pips");
+    K.pip_size = gf("pip_size");
+    K.k_sl = gf("k_sl");
+    K.k_tp = gf("k_tp");
+    K.rel_volume = gf("rel_volume");
+    K.min_order_volume = gf("min_order_volume");
+    K.max_order_volume = gf("max_order_volume");
+    K.min_sltp_frac = gf("min_sltp_frac");
+    K.max_sltp_frac = gf("max_sltp_frac");
+    K.baseline_rel_volume = gf("baseline_rel_volume");
+    K.max_risk_rel_volume = gf("max_risk_rel_volume");
+    K.sl_shrink_alpha = gf("sl_shrink_alpha");
+    K.tp_shrink_alpha = gf("tp_shrink_alpha");
+    K.min_k_sl = gf("min_k_sl");
+    K.min_rr = gf("min_rr");
+    K.mplf = gf("mplf");
+    K.fc_pen_coef = gf("fc_pen_coef");
+    K.fc_pen_window_hours = gf("fc_pen_window_hours");
+    K.feature_clip = gf("feature_clip");
+    K.overlay_threshold = gf("overlay_threshold");
+
+    // market pointers
+    P.open_px = ptr<float>(market, "open");
+    P.high_px = ptr<float>(market, "high");
+    P.low_px = ptr<float>(market, "low");
+    P.close_px = ptr<float>(market, "close");
+    P.price_px = ptr<float>(market, "price");
+    P.features = ptr<float>(market, "features", false);
+    P.p1 = ptr<double>(market, "feat_prefix1", false);
+    P.p2 = ptr<double>(market, "feat_prefix2", false);
+    P.binary_mask = ptr<bool>(market, "binary_mask", false);
+    P.ev_no_trade = ptr<float>(market, "ev_no_trade");
+    P.force_close = ptr<float>(market, "force_close", false);
+    P.calendar = ptr<float>(market, "calendar", false);
+    P.sess_entry = ptr<bool>(market, "sess_entry");
+    P.sess_close = ptr<bool>(market, "sess_close");
+
+    // state pointers
+    P.cursor = ptr<int>(state, "cursor");
+    P.started = ptr<bool>(state, "started");
+    P.terminated = ptr<bool>(state, "terminated");
+    P.pos = ptr<double>(state, "pos");
+    P.avg_entry = ptr<double>(state, "avg_entry");
+    P.cash = ptr<double>(state, "cash");
+    P.margin_used = ptr<double>(state, "margin_used");
+    P.equity = ptr<double>(state, "equity");
+    P.prev_equity = ptr<double>(state, "prev_equity");
+    P.peak_equity = ptr<double>(state, "peak_equity");
+    P.commission_paid = ptr<double>(state, "commission_paid");
+    P.last_trade_cost = ptr<double>(state, "last_trade_cost");
+    P.trade_count = ptr<int>(state, "trade_count");
+    P.pend_close = ptr<bool>(state, "pend_close");
+    P.pend_open_dir = ptr<int8_t>(state, "pend_open_dir");
+    P.pend_open_size = ptr<float>(state, "pend_open_size");
+    P.pend_sl = ptr<float>(state, "pend_sl");
+    P.pend_tp = ptr<float>(state, "pend_tp");
+    P.br_active = ptr<bool>(state, "br_active");
+    P.br_armed = ptr<bool>(state, "br_armed");
+    P.br_sl = ptr<float>(state, "br_sl");
+    P.br_tp = ptr<float>(state, "br_tp");
+    P.tr_ring = ptr<float>(state, "tr_ring");
+    P.tr_count = ptr<int>(state, "tr_count");
+    P.tr_sum = ptr<float>(state, "tr_sum");
+    P.prev_close_atr = ptr<float>(state, "prev_close_atr");
+    P.rew_ring = ptr<float>(state, "rew_ring");
+    P.rew_count = ptr<int>(state, "rew_count");
+    P.trade_won = ptr<int>(state, "trade_won");
+    P.trade_lost = ptr<int>(state, "trade_lost");
+    P.trade_pnl_sum = ptr<double>(state, "trade_pnl_sum");
+    P.trade_pnl_sumsq = ptr<double>(state, "trade_pnl_sumsq");
+    P.metric_peak = ptr<double>(state, "metric_peak");
+    P.max_dd_money = ptr<double>(state, "max_dd_money");
+    P.max_dd_pct = ptr<double>(state, "max_dd_pct");
+    P.ret_sum = ptr<double>(state, "ret_sum");
+    P.ret_sumsq = ptr<double>(state, "ret_sumsq");
+    P.ret_count = ptr<int>(state, "ret_count");
+    P.episode_step = ptr<int>(state, "episode_step");
+    P.episode_return = ptr<double>(state, "episode_return");
+    P.start_offset = ptr<int>(state, "start_offset");
+    P.exec_diag = ptr<int>(state, "exec_diag");
+    P.act_diag = ptr<int>(state, "act_diag");
+    P.raw_abs_sum = ptr<float>(state, "raw_abs_sum");
+    P.raw_min = ptr<float>(state, "raw_min");
+    P.raw_max = ptr<float>(state, "raw_max");
+
+    // persistent step outputs (allocated once, device)
+    auto dev = state["cursor"].device();
+    auto f32 = torch::TensorOptions().dtype(torch::kFloat32).device(dev);
+    outputs["reward"] = torch::zeros({K.n_envs}, f32);
+    outputs["base_reward"] = torch::zeros({K.n_envs}, f32);
+    outputs["penalty"] = torch::zeros({K.n_envs}, f32);
+    outputs["terminated"] = torch::zeros(
+        {K.n_envs}, torch::TensorOptions().dtype(torch::kBool).device(dev));
+    outputs["coerced"] = torch::zeros(
+        {K.n_envs}, torch::TensorOptions().dtype(torch::kInt64).device(dev));
+    P.reward_out = outputs["reward"].data_ptr<float>();
+    P.base_reward_out = outputs["base_reward"].data_ptr<float>();
+    P.penalty_out = outputs["penalty"].data_ptr<float>();
+    P.terminated_out = outputs["terminated"].data_ptr<bool>();
+    P.coerced_out = outputs["coerced"].data_ptr<int64_t>();
+  }
+
+  py::dict step(torch::Tensor actions, torch::Tensor obs_out) {
+    TORCH_CHECK(actions.is_cuda() == state["cursor"].is_cuda(),
+                "actions must live on the env device");
+    TORCH_CHECK(actions.numel() == K.n_envs, "actions numel != n_envs");
+    TORCH_CHECK(actions.is_contiguous(), "actions must be contiguous");
+    if (continuous) {
+      TORCH_CHECK(actions.scalar_type() == torch::kFloat32,
+                  "continuous mode wants float32 actions");
+    } else {
+      TORCH_CHECK(actions.scalar_type() == torch::kInt64,
+                  "discrete mode wants int64 actions");
+    }
+    P.actions = actions.data_ptr();
+    hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+    launch_env_step(P, K, stream);
+    build_obs(obs_out);
+    py::dict out;
+    out["reward"] = outputs["reward"];
+    out["base_reward"] = outputs["base_reward"];
+    out["force_close_reward_penalty"] = outputs["penalty"];
+    out["terminated"] = outputs["terminated"];
+    out["coerced_action"] = outputs["coerced"];
+    return out;
+  }
+
+  void build_obs(torch::Tensor obs_out) {
+    TORCH_CHECK(obs_out.is_contiguous() && obs_out.scalar_type() == torch::kFloat32,
+                "obs_out must be contiguous f32");
+    TORCH_CHECK(obs_out.numel() == (int64_t)K.n_envs * K.obs_dim, "obs_out shape");
+    P.obs_out = obs_out.data_ptr<float>();
+    hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+    launch_build_obs(P, K, stream);
+  }
+};
+
+}  // namespace gymfx
+
+// ---------------------------------------------------------------------------
+// PPO op wrappers (free functions)
+// ---------------------------------------------------------------------------
+namespace {
+
+hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_contiguous() && t.scalar_type() == torch::kBFloat16,
+              name, " must be contiguous bf16");
+}
+
+void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_contiguous() && t.scalar_type() == torch::kFloat32,
+              name, " must be contiguous f32");
+}
+
+void gemm_op(torch::Tensor A, torch::Tensor B, c10::optional<torch::Tensor> bias,
+             torch::Tensor C, c10::optional<torch::Tensor> Yact, bool trans_b,
+             int64_t act, bool dact_tanh) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  const int M = (int)A.size(0), K = (int)A.size(1);
+  const int N = trans_b ? (int)B.size(0) : (int)B.size(1);
+  TORCH_CHECK((trans_b ? B.size(1) : B.size(0)) == K, "gemm K mismatch");
+  TORCH_CHECK(C.size(0) == M && C.size(1) == N, "gemm C shape");
+  if (act == 0) check_f32(C, "C"); else check_bf16(C, "C");
+  const float* bias_p = nullptr;
+  if (bias.has_value()) {
+    check_f32(*bias, "bias");
+    TORCH_CHECK(bias->numel() == N, "bias numel");
+    bias_p = bias->data_ptr<float>();
+  }
+  const void* y_p = nullptr;
+  if (dact_tanh) {
+    TORCH_CHECK(Yact.has_value(), "dact_tanh needs Yact");
+    check_bf16(*Yact, "Yact");
+    y_p = Yact->data_ptr();
+  }
+  gymfx::launch_gemm(A.data_ptr(), B.data_ptr(), bias_p, C.data_ptr(), y_p, M,
+                     N, K, trans_b, (int)act, dact_tanh, bias_p != nullptr,
+                     cur_stream());
+}
+
+void wgrad_op(torch::Tensor X, torch::Tensor dY, torch::Tensor dW_part,
+              c10::optional<torch::Tensor> db_part, torch::Tensor dW,
+              c10::optional<torch::Tensor> db, int64_t slabs) {
+  check_bf16(X, "X");
+  check_bf16(dY, "dY");
+  check_f32(dW, "dW");
+  check_f32(dW_part, "dW_part");
+  const int M = (int)X.size(0), K = (int)X.size(1), N = (int)dY.size(1);
+  TORCH_CHECK(dY.size(0) == M, "wgrad M mismatch");
+  TORCH_CHECK(dW.size(0) == K && dW.size(1) == N, "dW shape");
+  TORCH_CHECK(dW_part.numel() >= slabs * (int64_t)K * N, "dW_part too small");
+  float* dbp = nullptr;
+  float* dbo = nullptr;
+  if (db_part.has_value() && db.has_value()) {
+    check_f32(*db_part, "db_part");
+    check_f32(*db, "db");
+    dbp = db_part->data_ptr<float>();
+    dbo = db->data_ptr<float>();
+  }
+  gymfx::launch_wgrad(X.data_ptr(), dY.data_ptr(), dW_part.data_ptr<float>(),
+                      dbp, dW.data_ptr<float>(), dbo, M, N, K, (int)slabs,
+                      cur_stream());
+}
+
+void gae_op(torch::Tensor rewards, torch::Tensor values, torch::Tensor dones,
+            torch::Tensor adv, torch::Tensor ret, double gamma, double lam) {
+  check_f32(rewards, "rewards");
+  check_f32(values, "values");
+  check_f32(adv, "adv");
+  check_f32(ret, "ret");
+  TORCH_CHECK(dones.scalar_type() == torch::kBool && dones.is_contiguous());
+  const int T = (int)rewards.size(0), N = (int)rewards.size(1);
+  TORCH_CHECK(values.size(0) == T + 1 && values.size(1) == N, "values shape");
+  gymfx::launch_gae(rewards.data_ptr<float>(), values.data_ptr<float>(),
+                    dones.data_ptr<bool>(), adv.data_ptr<float>(),
+                    ret.data_ptr<float>(), T, N, (float)gamma, (float)lam,
+                    cur_stream());
+}
+
+void adam_op(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
+             c10::optional<torch::Tensor> p_bf16, double lr, double beta1,
+             double beta2, double eps, int64_t step,
+             c10::optional<torch::Tensor> gscale) {
+  check_f32(p, "p");
+  check_f32(g, "g");
+  check_f32(m, "m");
+  check_f32(v, "v");
+  const int64_t n = p.numel();
+  TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n);
+  void* pb = nullptr;
+  if (p_bf16.has_value()) {
+    check_bf16(*p_bf16, "p_bf16");
+    pb = p_bf16->data_ptr();
+  }
+  const float* gs = nullptr;
+  if (gscale.has_value()) gs = gscale->data_ptr<float>();
+  const float bc1 = 1.f - powf((float)beta1, (float)step);
+  const float bc2 = 1.f - powf((float)beta2, (float)step);
+  gymfx::launch_adam(p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(), pb, n, (float)lr,
+                     (float)beta1, (float)beta2, (float)eps, bc1, bc2, gs,
+                     cur_stream());
+}
+
+void grad_clip_op(torch::Tensor g, double max_norm, torch::Tensor part,
+                  torch::Tensor scale) {
+  check_f32(g, "g");
+  check_f32(part, "part");
+  check_f32(scale, "scale");
+  gymfx::launch_grad_clip(g.data_ptr<float>(), g.numel(), (float)max_norm,
+                          part.data_ptr<float>(), scale.data_ptr<float>(),
+                          (int)part.numel(), cur_stream());
+}
+
+void sample_head_op(torch::Tensor head, int64_t seed, int64_t step,
+                    torch::Tensor actions, torch::Tensor logp,
+                    c10::optional<torch::Tensor> value,
+                    c10::optional<torch::Tensor> entropy, bool greedy) {
+  check_f32(head, "head");
+  const int M = (int)head.size(0);
+  const int n_actions = (int)head.size(1) - 1;
+  TORCH_CHECK(actions.scalar_type() == torch::kInt64);
+  gymfx::launch_sample_head(
+      head.data_ptr<float>(), M, n_actions, (uint64_t)seed, (uint64_t)step,
+      actions.data_ptr<int64_t>(), logp.data_ptr<float>(),
+      value.has_value() ? value->data_ptr<float>() : nullptr,
+      entropy.has_value() ? entropy->data_ptr<float>() : nullptr,
+      greedy ? 1 : 0, cur_stream());
+}
+
+void ppo_loss_bwd_op(torch::Tensor head, torch::Tensor actions,
+                     torch::Tensor old_logp, torch::Tensor adv,
+                     torch::Tensor ret, torch::Tensor dhead, double clip_eps,
+                     double ent_coef, double vf_coef, double inv_count,
+                     c10::optional<torch::Tensor> losses) {
+  check_f32(head, "head");
+  check_bf16(dhead, "dhead");
+  const int M = (int)head.size(0);
+  const int n_actions = (int)head.size(1) - 1;
+  gymfx::launch_ppo_loss_bwd(
+      head.data_ptr<float>(), actions.data_ptr<int64_t>(),
+      old_logp.data_ptr<float>(), adv.data_ptr<float>(), ret.data_ptr<float>(),
+      dhead.data_ptr(), M, n_actions, (float)clip_eps, (float)ent_coef,
+      (float)vf_coef, (float)inv_count,
+      losses.has_value() ? losses->data_ptr<float>() : nullptr, cur_stream());
+}
+
+void adv_normalize_op(torch::Tensor adv, torch::Tensor part) {
+  check_f32(adv, "adv");
+  check_f32(part, "part");
+  gymfx::launch_adv_normalize(adv.data_ptr<float>(), adv.numel(),
+                              part.data_ptr<float>(), (int)(part.numel() / 2),
+                              cur_stream());
+}
+
+void f32_to_bf16_op(torch::Tensor in, torch::Tensor out) {
+  check_f32(in, "in");
+  check_bf16(out, "out");
+  gymfx::launch_f32_to_bf16(in.data_ptr<float>(), out.data_ptr(), in.numel(),
+                            cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm", &gemm_op,
+        "C = act(A @ B + bias); trans_b reads B as [N,K]; act 0=f32 1=bf16 "
+        "2=tanh-bf16; dact_tanh multiplies by (1-Yact^2)",
+        py::arg("A"), py::arg("B"), py::arg("bias"), py::arg("C"),
+        py::arg("Yact") = py::none(), py::arg("trans_b") = false,
+        py::arg("act") = 1, py::arg("dact_tanh") = false);
+  m.def("wgrad", &wgrad_op, py::arg("X"), py::arg("dY"), py::arg("dW_part"),
+        py::arg("db_part"), py::arg("dW"), py::arg("db"), py::arg("slabs"));
+  m.def("gae", &gae_op);
+  m.def("adam", &adam_op, py::arg("p"), py::arg("g"), py::arg("m"),
+        py::arg("v"), py::arg("p_bf16"), py::arg("lr"), py::arg("beta1"),
+        py::arg("beta2"), py::arg("eps"), py::arg("step"),
+        py::arg("gscale") = py::none());
+  m.def("grad_clip", &grad_clip_op);
+  m.def("sample_head", &sample_head_op, py::arg("head"), py::arg("seed"),
+        py::arg("step"), py::arg("actions"), py::arg("logp"),
+        py::arg("value") = py::none(), py::arg("entropy") = py::none(),
+        py::arg("greedy") = false);
+  m.def("ppo_loss_bwd", &ppo_loss_bwd_op, py::arg("head"), py::arg("actions"),
+        py::arg("old_logp"), py::arg("adv"), py::arg("ret"), py::arg("dhead"),
+        py::arg("clip_eps"), py::arg("ent_coef"), py::arg("vf_coef"),
+        py::arg("inv_count"), py::arg("losses") = py::none());
+  m.def("adv_normalize", &adv_normalize_op);
+  m.def("f32_to_bf16", &f32_to_bf16_op);
+  py::class_<gymfx::GymFxEngine>(m, "GymFxEngine")
+      .def(py::init<const py::dict&, const py::dict&, const py::dict&, const py::dict&>())
+      .def("step", &gymfx::GymFxEngine::step)
+      .def("build_obs", &gymfx::GymFxEngine::build_obs);
+  m.attr("EXEC_COUNTER_N") = (int)gymfx::EXEC_COUNTER_N;
+  m.attr("ACT_COUNTER_N") = (int)gymfx::ACT_COUNTER_N;
+}

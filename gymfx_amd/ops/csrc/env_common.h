@@ -1,0 +1,136 @@
+// Shared structs for the fused env-step/obs kernels (gfx950 / CDNA4 only).
+// Keep enums in sync with gymfx_amd/envs/params.py.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define GFX_DEV __device__ __forceinline__
+
+namespace gymfx {
+
+// plugin dispatch enums (== envs/params.py)
+enum RewardId { REWARD_PNL = 0, REWARD_SHARPE = 1, REWARD_DD = 2 };
+enum StrategyId { STRAT_DIRECT = 0, STRAT_FIXED = 1, STRAT_ATR = 2 };
+enum PrepId { PREP_DEFAULT = 0, PREP_FEATURE_WINDOW = 1 };
+enum ScalingMode { SCALE_NONE = 0, SCALE_ROLLING = 1, SCALE_EXPANDING = 2 };
+enum SizeMode { SIZE_FX_UNITS = 0, SIZE_NOTIONAL = 1 };
+enum RiskMode { RISK_FIXED = 0, RISK_RELVOL = 1, RISK_MARGIN = 2 };
+
+// flag bits (== ops/wrappers.py FLAG_*)
+enum Flags {
+  F_CONTINUOUS = 1 << 0,
+  F_INCLUDE_PRICE = 1 << 1,
+  F_INCLUDE_AGENT = 1 << 2,
+  F_STAGEB_OBS = 1 << 3,
+  F_CALENDAR_OBS = 1 << 4,
+  F_OVERLAY = 1 << 5,
+  F_OVERLAY_BLOCK = 1 << 6,
+  F_OVERLAY_FF = 1 << 7,
+  F_SESSION_FILTER = 1 << 8,
+  F_HAS_MINFRAC = 1 << 9,
+  F_HAS_MAXFRAC = 1 << 10,
+  F_HAS_RELVOL = 1 << 11,
+  F_HAS_MPLF = 1 << 12,
+  F_STAGEB_PENALTY = 1 << 13,
+  F_AUTORESET = 1 << 14,
+};
+
+// execution-diagnostics counter indices (== envs/state.py EXEC_COUNTERS)
+enum ExecCounter {
+  E_ENTRY_ACTIONS_SEEN = 0,
+  E_ENTRY_ORDERS_SUBMITTED,
+  E_BLOCKED_SESSION_FILTER,
+  E_BLOCKED_ATR_WARMUP,
+  E_BLOCKED_NON_POSITIVE_ATR,
+  E_BLOCKED_NON_POSITIVE_SIZE,
+  E_BLOCKED_NON_POSITIVE_PRICE,
+  E_DEFAULT_ORDERS_SUBMITTED,
+  E_PLUGIN_APPLY_ERRORS,
+  E_EV_NO_TRADE_ACTIVE_STEPS,
+  E_EV_ACTION_OVERRIDES,
+  E_EV_BLOCKED_ENTRIES,
+  E_EV_FORCED_FLAT_ACTIONS,
+  E_EV_FORCED_FLAT_ORDERS,
+  E_SESSION_FORCE_CLOSES,
+  E_BRACKET_SL_FILLS,
+  E_BRACKET_TP_FILLS,
+  EXEC_COUNTER_N,
+};
+
+enum ActCounter {
+  A_STEPS = 0,
+  A_HOLD,
+  A_LONG,
+  A_SHORT,
+  A_NON_HOLD,
+  A_DEADBAND,
+  ACT_COUNTER_N,
+};
+
+struct EnvParamsK {
+  int n_envs, T, window, n_features;
+  int reward_id, strategy_id, prep_id;
+  int scaling_mode, scale_window, sharpe_window, atr_period;
+  int size_mode, risk_mode;
+  int flags;
+  int obs_dim;
+  int off_features, off_prices, off_returns, off_agent, off_fc, off_cal;
+
+  double initial_cash, position_size, commission, slippage, leverage, min_equity;
+  double cont_threshold, reward_scale, annualization, penalty_lambda;
+  double sl_pips, tp_pips, pip_size;
+  double k_sl, k_tp, rel_volume, min_order_volume, max_order_volume;
+  double min_sltp_frac, max_sltp_frac;
+  double baseline_rel_volume, max_risk_rel_volume;
+  double sl_shrink_alpha, tp_shrink_alpha, min_k_sl, min_rr, mplf;
+  double fc_pen_coef, fc_pen_window_hours;
+  double feature_clip, overlay_threshold;
+};
+
+struct EnvPtrs {
+  // market (device, read-only)
+  const float *open_px, *high_px, *low_px, *close_px, *price_px;
+  const float *features;      // [T, F]
+  const double *p1, *p2;      // [T+1, F] prefix sums
+  const bool *binary_mask;    // [F]
+  const float *ev_no_trade;   // [T]
+  const float *force_close;   // [T, 4] or null
+  const float *calendar;      // [T, 10] or null
+  const bool *sess_entry, *sess_close;  // [T]
+  // state (device, mutable)
+  int *cursor;
+  bool *started, *terminated;
+  double *pos, *avg_entry, *cash, *margin_used, *equity, *prev_equity, *peak_equity;
+  double *commission_paid, *last_trade_cost;
+  int *trade_count;
+  bool *pend_close;
+  int8_t *pend_open_dir;
+  float *pend_open_size, *pend_sl, *pend_tp;
+  bool *br_active, *br_armed;
+  float *br_sl, *br_tp;
+  float *tr_ring;            // [N, P]
+  int *tr_count;
+  float *tr_sum;
+  float *prev_close_atr;
+  float *rew_ring;           // [N, W_sharpe]
+  int *rew_count;
+  int *trade_won, *trade_lost;
+  double *trade_pnl_sum, *trade_pnl_sumsq;
+  double *metric_peak, *max_dd_money, *max_dd_pct, *ret_sum, *ret_sumsq;
+  int *ret_count;
+  int *episode_step;
+  double *episode_return;
+  int *start_offset;
+  int *exec_diag;            // [N, EXEC_COUNTER_N]
+  int *act_diag;             // [N, ACT_COUNTER_N]
+  float *raw_abs_sum, *raw_min, *raw_max;
+  // step inputs/outputs
+  const void *actions;       // int64 (discrete) or float32 (continuous)
+  float *reward_out, *base_reward_out, *penalty_out;
+  bool *terminated_out;
+  int64_t *coerced_out;
+  float *obs_out;            // [N, obs_dim]
+};
+
+}  // namespace gymfx

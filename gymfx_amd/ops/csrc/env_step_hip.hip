@@ -1,0 +1,564 @@
+#include "hip/hip_runtime.h"
+// Fused vectorized env step + observation build (gfx950).
+//
+// One lane per environment: the whole per-bar transition of the reference
+// engine (action decode -> event overlay -> pending market fills at open ->
+// bracket SL/TP children (worst-case ordering) -> ATR update -> strategy
+// decision -> equity publish -> reward -> metrics -> optional autoreset)
+// runs as straight-line scalar code per lane; N environments = N lanes.
+//
+// Semantics oracle: gymfx_amd/envs/reference_step.py (torch CPU) — GPU tests
+// assert elementwise equality.  Reference provenance for the behavior:
+// /root/reference/app/bt_bridge.py:136-248, app/env.py:279-440,
+// broker_plugins/default_broker.py, strategy_plugins/direct_{fixed,atr}_sltp.py,
+// reward_plugins/*.py.
+#include "env_common.h"
+
+namespace gymfx {
+
+GFX_DEV double buy_fill(double o, double slip) { return o * (1.0 + slip); }
+GFX_DEV double sell_fill(double o, double slip) { return o * (1.0 - slip); }
+
+// Close the full position at `fill`; mirrors reference_step._close_position.
+GFX_DEV void close_position(const EnvPtrs& P, const EnvParamsK& K, int n,
+                            double fill, int counter) {
+  double pos = P.pos[n];
+  double realized = pos * (fill - P.avg_entry[n]);
+  double comm = fabs(pos) * fill * K.commission;
+  P.cash[n] += P.margin_used[n] + realized - comm;
+  P.commission_paid[n] += comm;
+  P.last_trade_cost[n] += comm;
+  P.trade_count[n] += 1;
+  if (realized > 0) P.trade_won[n] += 1;
+  if (realized < 0) P.trade_lost[n] += 1;
+  double net = realized - comm;
+  P.trade_pnl_sum[n] += net;
+  P.trade_pnl_sumsq[n] += net * net;
+  P.pos[n] = 0.0;
+  P.avg_entry[n] = 0.0;
+  P.margin_used[n] = 0.0;
+  P.br_active[n] = false;
+  P.br_armed[n] = false;
+  if (counter >= 0) P.exec_diag[n * EXEC_COUNTER_N + counter] += 1;
+}
+
+GFX_DEV void reset_env(const EnvPtrs& P, const EnvParamsK& K, int n) {
+  const double ic = K.initial_cash;
+  P.cursor[n] = 1 + P.start_offset[n];
+  P.started[n] = false;
+  P.terminated[n] = false;
+  P.pos[n] = 0.0;
+  P.avg_entry[n] = 0.0;
+  P.cash[n] = ic;
+  P.margin_used[n] = 0.0;
+  P.equity[n] = ic;
+  P.prev_equity[n] = ic;
+  P.peak_equity[n] = 0.0;
+  P.commission_paid[n] = 0.0;
+  P.last_trade_cost[n] = 0.0;
+  P.trade_count[n] = 0;
+  P.pend_close[n] = false;
+  P.pend_open_dir[n] = 0;
+  P.pend_open_size[n] = 0.f;
+  P.pend_sl[n] = 0.f;
+  P.pend_tp[n] = 0.f;
+  P.br_active[n] = false;
+  P.br_armed[n] = false;
+  P.br_sl[n] = 0.f;
+  P.br_tp[n] = 0.f;
+  for (int i = 0; i < K.atr_period; ++i) P.tr_ring[(int64_t)n * K.atr_period + i] = 0.f;
+  P.tr_count[n] = 0;
+  P.tr_sum[n] = 0.f;
+  P.prev_close_atr[n] = __builtin_nanf("");
+  for (int i = 0; i < K.sharpe_window; ++i) P.rew_ring[(int64_t)n * K.sharpe_window + i] = 0.f;
+  P.rew_count[n] = 0;
+  P.trade_won[n] = 0;
+  P.trade_lost[n] = 0;
+  P.trade_pnl_sum[n] = 0.0;
+  P.trade_pnl_sumsq[n] = 0.0;
+  P.metric_peak[n] = ic;
+  P.max_dd_money[n] = 0.0;
+  P.max_dd_pct[n] = 0.0;
+  P.ret_sum[n] = 0.0;
+  P.ret_sumsq[n] = 0.0;
+  P.ret_count[n] = 0;
+  P.episode_step[n] = 0;
+  P.episode_return[n] = 0.0;
+  for (int i = 0; i < EXEC_COUNTER_N; ++i) P.exec_diag[n * EXEC_COUNTER_N + i] = 0;
+  for (int i = 0; i < ACT_COUNTER_N; ++i) P.act_diag[n * ACT_COUNTER_N + i] = 0;
+  P.raw_abs_sum[n] = 0.f;
+  P.raw_min[n] = INFINITY;
+  P.raw_max[n] = -INFINITY;
+}
+
+__global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
+  const int n = blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= K.n_envs) return;
+  const int T = K.T;
+  const double slip = K.slippage;
+  int* ediag = P.exec_diag + (int64_t)n * EXEC_COUNTER_N;
+  int* adiag = P.act_diag + (int64_t)n * ACT_COUNTER_N;
+
+  // ---- action decode (env.py:343-360) --------------------------------
+  float raw;
+  int64_t a;
+  if (K.flags & F_CONTINUOUS) {
+    raw = reinterpret_cast<const float*>(P.actions)[n];
+    const double thr = K.cont_threshold != 0.0 ? K.cont_threshold : 0.33;
+    a = (raw >= thr) ? 1 : (raw <= -thr) ? 2 : 0;
+  } else {
+    int64_t v = reinterpret_cast<const int64_t*>(P.actions)[n];
+    raw = (float)v;
+    a = (v >= 0 && v <= 2) ? v : 0;
+  }
+
+  const bool live = !P.terminated[n];
+
+  // ---- event-context overlay (env.py:394-440) ------------------------
+  // row = pre-advance bar_index (the reference reads the upcoming row).
+  if (K.flags & F_OVERLAY) {
+    int row_ov = min(P.cursor[n], T - 1);
+    bool active = P.ev_no_trade[row_ov] >= (float)K.overlay_threshold;
+    int psign = P.pos[n] > 0 ? 1 : (P.pos[n] < 0 ? -1 : 0);
+    if (live && active) {
+      ediag[E_EV_NO_TRADE_ACTIVE_STEPS] += 1;
+      bool forced = (K.flags & F_OVERLAY_FF) && psign != 0;
+      bool blocked = !forced && (K.flags & F_OVERLAY_BLOCK) && psign == 0 &&
+                     (a == 1 || a == 2);
+      if (forced) a = 3;
+      if (blocked) a = 0;
+      if (forced || blocked) ediag[E_EV_ACTION_OVERRIDES] += 1;
+      if (blocked) ediag[E_EV_BLOCKED_ENTRIES] += 1;
+      if (forced) ediag[E_EV_FORCED_FLAT_ACTIONS] += 1;
+    }
+  }
+
+  // ---- action diagnostics (env.py:744-761) ---------------------------
+  if (live) {
+    adiag[A_STEPS] += 1;
+    P.raw_abs_sum[n] += fabsf(raw);
+    P.raw_min[n] = fminf(P.raw_min[n], raw);
+    P.raw_max[n] = fmaxf(P.raw_max[n], raw);
+    if (a == 1) { adiag[A_LONG] += 1; adiag[A_NON_HOLD] += 1; }
+    else if (a == 2) { adiag[A_SHORT] += 1; adiag[A_NON_HOLD] += 1; }
+    else {
+      adiag[A_HOLD] += 1;
+      if (K.flags & F_CONTINUOUS) adiag[A_DEADBAND] += 1;
+    }
+  }
+
+  // ---- advance --------------------------------------------------------
+  const bool first = live && !P.started[n];
+  const bool adv = live && P.started[n];
+  bool exhausted = false;
+  if (adv) {
+    P.cursor[n] += 1;
+    if (P.cursor[n] > T) { exhausted = true; P.cursor[n] = T; P.terminated[n] = true; }
+  }
+  const bool valid = adv && !exhausted;
+  const int t = max(P.cursor[n] - 1, 0);
+  const double o_px = P.open_px[t];
+  const double h_px = P.high_px[t];
+  const double l_px = P.low_px[t];
+  const double c_px = P.close_px[t];
+
+  if (valid || first) P.last_trade_cost[n] = 0.0;
+
+  // ---- 1. pending market fills at open(t) ----------------------------
+  if (valid && P.pend_close[n] && P.pos[n] != 0.0) {
+    double fill = (P.pos[n] < 0) ? buy_fill(o_px, slip) : sell_fill(o_px, slip);
+    close_position(P, K, n, fill, -1);
+  }
+  if (valid && P.pend_open_dir[n] != 0 && P.pos[n] == 0.0) {
+    double dir = (double)P.pend_open_dir[n];
+    double fill = dir > 0 ? buy_fill(o_px, slip) : sell_fill(o_px, slip);
+    double size = (double)P.pend_open_size[n];
+    double notional = size * fill;
+    double comm = notional * K.commission;
+    double margin = notional / K.leverage;
+    P.cash[n] -= margin + comm;
+    P.margin_used[n] = margin;
+    P.commission_paid[n] += comm;
+    P.last_trade_cost[n] += comm;
+    P.pos[n] = dir * size;
+    P.avg_entry[n] = fill;
+    if (P.pend_sl[n] > 0.f || P.pend_tp[n] > 0.f) {
+      P.br_active[n] = true;
+      P.br_armed[n] = true;  // children active from NEXT bar
+      P.br_sl[n] = P.pend_sl[n];
+      P.br_tp[n] = P.pend_tp[n];
+    }
+  }
+  P.pend_close[n] = false;
+  P.pend_open_dir[n] = 0;
+  P.pend_open_size[n] = 0.f;
+  P.pend_sl[n] = 0.f;
+  P.pend_tp[n] = 0.f;
+
+  // ---- 2. bracket children, worst-case ordering ----------------------
+  if (valid && P.br_active[n] && !P.br_armed[n] && P.pos[n] != 0.0) {
+    const bool is_long = P.pos[n] > 0;
+    const double sl = P.br_sl[n], tp = P.br_tp[n];
+    double trig = 0.0;
+    int hit = 0;  // 0 none, 1 sl, 2 tp
+    if (is_long) {
+      if (o_px <= sl) { hit = 1; trig = o_px; }
+      else if (l_px <= sl) { hit = 1; trig = sl; }
+      else if (o_px >= tp) { hit = 2; trig = o_px; }
+      else if (h_px >= tp) { hit = 2; trig = tp; }
+    } else {
+      if (o_px >= sl) { hit = 1; trig = o_px; }
+      else if (h_px >= sl) { hit = 1; trig = sl; }
+      else if (o_px <= tp) { hit = 2; trig = o_px; }
+      else if (l_px <= tp) { hit = 2; trig = tp; }
+    }
+    if (hit) {
+      double fill = is_long ? sell_fill(trig, slip) : buy_fill(trig, slip);
+      close_position(P, K, n, fill, hit == 1 ? E_BRACKET_SL_FILLS : E_BRACKET_TP_FILLS);
+    }
+  }
+  if (valid) P.br_armed[n] = false;
+
+  const bool dec = valid || first;
+
+  // ---- 3. ATR true-range update (direct_atr_sltp.py:143-155) ---------
+  float atr = 0.f;
+  bool atr_ready = false;
+  if (K.strategy_id == STRAT_ATR && dec) {
+    const int Pn = K.atr_period;
+    float pc = P.prev_close_atr[n];
+    float tr = isnan(pc)
+                   ? (float)(h_px - l_px)
+                   : fmaxf((float)(h_px - l_px),
+                           fmaxf(fabsf((float)(h_px - pc)), fabsf((float)(l_px - pc))));
+    int cnt = P.tr_count[n];
+    int idx = cnt % Pn;
+    float old = P.tr_ring[(int64_t)n * Pn + idx];
+    P.tr_sum[n] += tr - (cnt >= Pn ? old : 0.f);
+    P.tr_ring[(int64_t)n * Pn + idx] = tr;
+    P.tr_count[n] = cnt + 1;
+    P.prev_close_atr[n] = (float)c_px;
+    int n_tr = min(P.tr_count[n], Pn);
+    atr = P.tr_sum[n] / (float)max(n_tr, 1);
+    atr_ready = P.tr_count[n] >= Pn;
+  }
+
+  // ---- 4. strategy decision on bar t ---------------------------------
+  if (dec && a == 3) {  // overlay force-flat bypasses the strategy
+    if (P.pos[n] != 0.0) {
+      P.pend_close[n] = true;
+      ediag[E_DEFAULT_ORDERS_SUBMITTED] += 1;
+      ediag[E_EV_FORCED_FLAT_ORDERS] += 1;
+    }
+  } else if (dec && K.strategy_id == STRAT_DIRECT) {
+    if (a == 1 || a == 2) {
+      ediag[E_ENTRY_ACTIONS_SEEN] += 1;
+      const double pos = P.pos[n];
+      int orders = 0;
+      if (a == 1) {
+        if (pos < 0) { P.pend_close[n] = true; orders += 1; }
+        if (pos <= 0) {
+          P.pend_open_dir[n] = 1;
+          P.pend_open_size[n] = (float)K.position_size;
+          orders += 1;
+        }
+      } else {
+        if (pos > 0) { P.pend_close[n] = true; orders += 1; }
+        if (pos >= 0) {
+          P.pend_open_dir[n] = -1;
+          P.pend_open_size[n] = (float)K.position_size;
+          orders += 1;
+        }
+      }
+      ediag[E_DEFAULT_ORDERS_SUBMITTED] += orders;
+    }
+  } else if (dec && K.strategy_id == STRAT_FIXED) {
+    if (a == 1 || a == 2) {
+      const double pos = P.pos[n];
+      const double sl_d = K.sl_pips * K.pip_size;
+      const double tp_d = K.tp_pips * K.pip_size;
+      if (a == 1) {
+        if (pos < 0) P.pend_close[n] = true;
+        if (pos <= 0) {
+          P.pend_open_dir[n] = 1;
+          P.pend_open_size[n] = (float)K.position_size;
+          P.pend_sl[n] = (float)(c_px - sl_d);
+          P.pend_tp[n] = (float)(c_px + tp_d);
+          ediag[E_ENTRY_ORDERS_SUBMITTED] += 1;
+        }
+      } else {
+        if (pos > 0) P.pend_close[n] = true;
+        if (pos >= 0) {
+          P.pend_open_dir[n] = -1;
+          P.pend_open_size[n] = (float)K.position_size;
+          P.pend_sl[n] = (float)(c_px + sl_d);
+          P.pend_tp[n] = (float)(c_px - tp_d);
+          ediag[E_ENTRY_ORDERS_SUBMITTED] += 1;
+        }
+      }
+    }
+  } else if (dec && K.strategy_id == STRAT_ATR) {
+    const bool sess_on = (K.flags & F_SESSION_FILTER) != 0;
+    const bool in_close_zone = sess_on ? P.sess_close[t] : false;
+    const bool in_entry_win = sess_on ? P.sess_entry[t] : true;
+    if (in_close_zone && P.pos[n] != 0.0) {
+      P.pend_close[n] = true;
+      ediag[E_SESSION_FORCE_CLOSES] += 1;
+    } else if (a == 1 || a == 2) {
+      ediag[E_ENTRY_ACTIONS_SEEN] += 1;
+      bool ok = true;
+      if (sess_on && !in_entry_win) { ediag[E_BLOCKED_SESSION_FILTER] += 1; ok = false; }
+      if (ok && !atr_ready) { ediag[E_BLOCKED_ATR_WARMUP] += 1; ok = false; }
+      if (ok && atr <= 0.f) { ediag[E_BLOCKED_NON_POSITIVE_ATR] += 1; ok = false; }
+      double size = K.position_size;
+      if (K.flags & F_HAS_RELVOL) {
+        double raw_sz = P.cash[n] * K.rel_volume * K.leverage;
+        if (K.size_mode == SIZE_NOTIONAL) raw_sz = c_px > 0 ? raw_sz / c_px : 0.0;
+        size = fmin(fmax(raw_sz, K.min_order_volume), K.max_order_volume);
+      }
+      if (ok && size <= 0.0) { ediag[E_BLOCKED_NON_POSITIVE_SIZE] += 1; ok = false; }
+      if (ok && c_px <= 0.0) { ediag[E_BLOCKED_NON_POSITIVE_PRICE] += 1; ok = false; }
+      if (ok) {
+        // risk-mode k shrink (direct_atr_sltp.py:263-289)
+        double k_sl_eff = K.k_sl, k_tp_eff = K.k_tp;
+        if (K.risk_mode != RISK_FIXED) {
+          double rel = fmax(0.0, (K.flags & F_HAS_RELVOL) ? K.rel_volume : 0.0);
+          if (rel > K.baseline_rel_volume) {
+            double prog = fmin(1.0, fmax(0.0, (rel - K.baseline_rel_volume) /
+                                                  (K.max_risk_rel_volume -
+                                                   K.baseline_rel_volume)));
+            k_sl_eff = fmax(K.min_k_sl, K.k_sl * (1.0 - K.sl_shrink_alpha * prog));
+            k_tp_eff = K.k_tp * (1.0 - K.tp_shrink_alpha * prog);
+          }
+          k_tp_eff = fmax(k_tp_eff, k_sl_eff * K.min_rr);
+        }
+        double sl_dist = k_sl_eff * (double)atr;
+        double tp_dist = k_tp_eff * (double)atr;
+        if (K.risk_mode == RISK_MARGIN && (K.flags & F_HAS_MPLF)) {
+          double rel = fmax(0.0, (K.flags & F_HAS_RELVOL) ? K.rel_volume : 0.0);
+          double mlf = fmax(0.0, K.mplf);
+          if (rel > 0.0 && mlf > 0.0)
+            sl_dist = fmin(sl_dist, c_px * mlf / (rel * K.leverage));
+        }
+        if (K.flags & F_HAS_MINFRAC) {
+          double floor_d = K.min_sltp_frac * c_px;
+          sl_dist = fmax(sl_dist, floor_d);
+          tp_dist = fmax(tp_dist, floor_d);
+        }
+        if (K.flags & F_HAS_MAXFRAC) {
+          double ceil_d = K.max_sltp_frac * c_px;
+          sl_dist = fmin(sl_dist, ceil_d);
+          tp_dist = fmin(tp_dist, ceil_d);
+        }
+        if (tp_dist >= c_px) tp_dist = c_px * 0.5;
+        const double pos = P.pos[n];
+        if (a == 1) {
+          if (pos < 0) P.pend_close[n] = true;
+          if (pos <= 0) {
+            P.pend_open_dir[n] = 1;
+            P.pend_open_size[n] = (float)size;
+            P.pend_sl[n] = (float)(c_px - sl_dist);
+            P.pend_tp[n] = (float)(c_px + tp_dist);
+            ediag[E_ENTRY_ORDERS_SUBMITTED] += 1;
+          }
+        } else {
+          if (pos > 0) P.pend_close[n] = true;
+          if (pos >= 0) {
+            P.pend_open_dir[n] = -1;
+            P.pend_open_size[n] = (float)size;
+            P.pend_sl[n] = (float)(c_px + sl_dist);
+            P.pend_tp[n] = (float)(c_px - tp_dist);
+            ediag[E_ENTRY_ORDERS_SUBMITTED] += 1;
+          }
+        }
+      }
+    }
+  }
+
+  if (first) P.started[n] = true;
+  if (dec) P.episode_step[n] += 1;
+
+  // ---- 5. publish (bt_bridge.py:239-248) ------------------------------
+  const bool pub = dec;
+  double reward = 0.0, base_reward = 0.0, penalty = 0.0;
+  if (pub) {
+    P.prev_equity[n] = P.equity[n];
+    double unreal = P.pos[n] * (c_px - P.avg_entry[n]);
+    P.equity[n] = P.cash[n] + P.margin_used[n] + unreal;
+    if (P.equity[n] <= K.min_equity) P.terminated[n] = true;
+    if (valid && P.cursor[n] >= T) P.terminated[n] = true;
+
+    // ---- 6. reward ---------------------------------------------------
+    const double ic = K.initial_cash != 0.0 ? K.initial_cash : 1.0;
+    const double r_step = (P.equity[n] - P.prev_equity[n]) / ic;
+    if (K.reward_id == REWARD_SHARPE) {
+      const int W = K.sharpe_window;
+      int cnt = P.rew_count[n];
+      P.rew_ring[(int64_t)n * W + (cnt % W)] = (float)r_step;
+      P.rew_count[n] = cnt + 1;
+      int m = min(P.rew_count[n], W);
+      if (m >= 2) {
+        float s1 = 0.f;
+        for (int i = 0; i < m; ++i) s1 += P.rew_ring[(int64_t)n * W + i];
+        float mean = s1 / (float)m;
+        float ss = 0.f;
+        for (int i = 0; i < m; ++i) {
+          float d = P.rew_ring[(int64_t)n * W + i] - mean;
+          ss += d * d;
+        }
+        float var = ss / (float)(m - 1);
+        float stdv = sqrtf(var);
+        if (stdv > 0.f) base_reward = (double)(mean / stdv) * sqrt(K.annualization);
+      }
+    } else if (K.reward_id == REWARD_DD) {
+      P.peak_equity[n] = fmax(P.peak_equity[n], fmax(P.equity[n], P.prev_equity[n]));
+      double dd = P.peak_equity[n] > 0 ? (P.peak_equity[n] - P.equity[n]) / ic : 0.0;
+      base_reward = r_step - K.penalty_lambda * dd;
+    } else {
+      base_reward = r_step * K.reward_scale;
+    }
+
+    if ((K.flags & F_STAGEB_PENALTY) && K.fc_pen_coef > 0 && P.force_close) {
+      int row_fc = min(P.cursor[n], T - 1);
+      float hours = P.force_close[row_fc * 4 + 1];
+      bool in_zone = P.force_close[row_fc * 4 + 2] > 0.f;
+      bool in_win = hours >= 0.f && hours <= fmaxf(0.f, (float)K.fc_pen_window_hours);
+      int psign = P.pos[n] > 0 ? 1 : (P.pos[n] < 0 ? -1 : 0);
+      if ((in_zone || in_win) && psign != 0) penalty = K.fc_pen_coef;
+    }
+    reward = base_reward - penalty;
+    P.episode_return[n] += reward;
+
+    // ---- 7. metrics ---------------------------------------------------
+    P.ret_sum[n] += r_step;
+    P.ret_sumsq[n] += r_step * r_step;
+    P.ret_count[n] += 1;
+    P.metric_peak[n] = fmax(P.metric_peak[n], P.equity[n]);
+    double dd_money = P.metric_peak[n] - P.equity[n];
+    P.max_dd_money[n] = fmax(P.max_dd_money[n], dd_money);
+    if (P.metric_peak[n] > 0)
+      P.max_dd_pct[n] = fmax(P.max_dd_pct[n], dd_money / P.metric_peak[n] * 100.0);
+  }
+
+  P.reward_out[n] = (float)reward;
+  P.base_reward_out[n] = (float)base_reward;
+  P.penalty_out[n] = (float)penalty;
+  P.terminated_out[n] = P.terminated[n];
+  P.coerced_out[n] = a;
+
+  // ---- 8. autoreset (training path; no reference counterpart) --------
+  if ((K.flags & F_AUTORESET) && P.terminated[n]) reset_env(P, K, n);
+}
+
+// ---------------------------------------------------------------------------
+// Observation build: elementwise over [N, obs_dim].
+// Semantics: envs/reference_step.py build_obs_torch (itself mirroring
+// default_preprocessor.py:34-77 / feature_window_preprocessor.py:99-191).
+// ---------------------------------------------------------------------------
+__global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K) {
+  const int64_t total = (int64_t)K.n_envs * K.obs_dim;
+  const int T = K.T, W = K.window, F = K.n_features;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int n = (int)(idx / K.obs_dim);
+    const int j = (int)(idx % K.obs_dim);
+    const int step = P.cursor[n];
+    float val = 0.f;
+
+    if (K.off_features >= 0 && j < K.off_features + W * F && j >= K.off_features) {
+      const int q = j - K.off_features;
+      const int w = q / F, f = q % F;
+      int row = step - W + w;
+      row = max(row, 0);
+      row = min(row, T - 1);
+      float x = P.features[(int64_t)row * F + f];
+      bool is_binary = P.binary_mask && P.binary_mask[f];
+      if (K.scaling_mode == SCALE_NONE || is_binary) {
+        val = x;
+      } else {
+        int hist_left = (K.scaling_mode == SCALE_ROLLING) ? max(step - K.scale_window, 0) : 0;
+        int m = step - hist_left;
+        if (m < 2) {
+          val = 0.f;
+        } else {
+          double s1 = P.p1[(int64_t)step * F + f] - P.p1[(int64_t)hist_left * F + f];
+          double s2 = P.p2[(int64_t)step * F + f] - P.p2[(int64_t)hist_left * F + f];
+          double mean = s1 / (double)m;
+          double var = s2 / (double)m - mean * mean;
+          var = fmax(var, 0.0);
+          double stdv = sqrt(var);
+          if (stdv < 1e-8) stdv = 1.0;
+          val = (float)(((double)x - mean) / stdv);
+        }
+      }
+      // NaN guard BEFORE the clamp (fminf/fmaxf silently drop NaNs on CDNA):
+      // torch maps nan->0, +/-inf->+/-clip (reference_step.build_obs_torch).
+      if (isnan(val)) val = 0.f;
+      if (K.feature_clip > 0) {
+        float c = (float)K.feature_clip;
+        if (isinf(val)) val = val > 0 ? c : -c;
+        val = fminf(fmaxf(val, -c), c);
+      } else if (isinf(val)) {
+        val = 0.f;
+      }
+    } else if (K.off_prices >= 0 && j >= K.off_prices && j < K.off_prices + W) {
+      const int w = j - K.off_prices;
+      int row = min(max(step - W + w, 0), T - 1);
+      val = P.price_px[row];
+    } else if (K.off_returns >= 0 && j >= K.off_returns && j < K.off_returns + W) {
+      const int w = j - K.off_returns;
+      int row = min(max(step - W + w, 0), T - 1);
+      int row_prev = min(max(step - W + w - 1, 0), T - 1);
+      if (w == 0) row_prev = row;
+      val = P.price_px[row] - P.price_px[row_prev];
+    } else if (K.off_agent >= 0 && j >= K.off_agent && j < K.off_agent + 4) {
+      const int q = j - K.off_agent;
+      const double ic = K.initial_cash != 0.0 ? K.initial_cash : 1.0;
+      if (q == 0) {
+        val = P.pos[n] > 0 ? 1.f : (P.pos[n] < 0 ? -1.f : 0.f);
+      } else if (q == 1) {
+        val = (float)((P.equity[n] - ic) / ic);
+      } else if (q == 2) {
+        int t_now = min(max(step - 1, 0), T - 1);
+        int row_last = min(max(step - 1, 0), T - 1);
+        float psign = P.pos[n] > 0 ? 1.f : (P.pos[n] < 0 ? -1.f : 0.f);
+        float unreal =
+            psign * (P.close_px[t_now] - P.price_px[row_last]) * (float)K.position_size;
+        val = unreal / (float)ic;
+      } else {
+        val = fmaxf((float)(T - step), 0.f) / (float)max(T, 1);
+      }
+    } else if (K.off_fc >= 0 && j >= K.off_fc && j < K.off_fc + 4) {
+      const int q = j - K.off_fc;
+      int row = min(step, T - 1);
+      val = P.force_close ? P.force_close[row * 4 + q] : 0.f;
+    } else if (K.off_cal >= 0 && j >= K.off_cal && j < K.off_cal + 11) {
+      const int q = j - K.off_cal;
+      int row = min(step, T - 1);
+      if (q < 9) val = P.calendar ? P.calendar[row * 10 + q] : 0.f;
+      else if (q == 9) val = 0.f;  // margin_closeout_percent
+      else {
+        const double ic = K.initial_cash != 0.0 ? K.initial_cash : 1.0;
+        val = (float)(P.equity[n] / ic);
+      }
+    }
+    P.obs_out[(int64_t)n * K.obs_dim + j] = val;
+  }
+}
+
+void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream) {
+  const int block = 256;
+  const int grid = (K.n_envs + block - 1) / block;
+  hipLaunchKernelGGL(env_step_kernel, dim3(grid), dim3(block), 0, stream, P, K);
+}
+
+void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream) {
+  const int block = 256;
+  int64_t total = (int64_t)K.n_envs * K.obs_dim;
+  int64_t blocks = (total + block - 1) / block;
+  int grid = (int)(blocks < 2048 ? blocks : 2048);
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(build_obs_kernel, dim3(grid), dim3(block), 0, stream, P, K);
+}
+
+}  // namespace gymfx

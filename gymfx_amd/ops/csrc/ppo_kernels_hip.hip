@@ -1,0 +1,643 @@
+#include "hip/hip_runtime.h"
+// PPO compute stack for gfx950 (CDNA4): hand-written MFMA bf16 GEMMs with
+// fused epilogues (bias+tanh fwd, tanh' dgrad), deterministic split-M wgrad,
+// GAE backward scan, fused Adam, categorical sampling with counter-based RNG,
+// and the PPO clipped-surrogate loss backward.
+//
+// The reference is agent-free by design (/root/reference/app/env.py:148-150);
+// this stack is the BASELINE.json north-star requirement (PPO MLP(256,256)
+// bf16 on MFMA, GAE scan, minibatch Adam).
+//
+// MFMA: v_mfma_f32_16x16x32_bf16 — per-wave 16x16 tile, K=32, fp32 accum.
+// Lane mapping (cdna_hip_programming.md §3):
+//   A: lane l holds A[row = l&15][k = (l>>4)*8 + i], i=0..7  (8 bf16)
+//   B: lane l holds B[k = (l>>4)*8 + i][col = l&15]
+//   C/D: lane l, reg r -> row = (l>>4)*4 + r, col = l&15
+// Verified on hardware by tests/test_gpu_gemm.py (identity x asymmetric-B
+// probes per guide rule G9).
+#include "env_common.h"
+
+#include <hip/hip_bf16.h>
+
+#include <algorithm>
+
+namespace gymfx {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+GFX_DEV float bf2f(__bf16 v) { return (float)v; }
+GFX_DEV __bf16 f2bf(float v) { return (__bf16)v; }
+
+// ---------------------------------------------------------------------------
+// GEMM: C[M,N] = act(A[M,K] @ B + bias)
+//   TRANS_B = false: B stored [K, N] (ldb = N)
+//   TRANS_B = true:  B stored [N, K] (ldb = K)  (used for dgrad: B = W^T)
+// Block: 256 threads = 4 waves in a 2x2 wave grid; block tile 64x64;
+// wave tile 32x32 = 2x2 MFMA fragments; BK = 32.
+// Epilogues: ACT 0=none(f32 out) 1=none(bf16) 2=tanh(bf16)
+//            DACT_TANH: multiply by (1 - Y^2) elementwise (dgrad fused tanh')
+// ---------------------------------------------------------------------------
+template <bool TRANS_B, int ACT, bool DACT_TANH, bool ADD_BIAS>
+__global__ __launch_bounds__(256) void gemm_kernel(
+    const __bf16* __restrict__ A, const __bf16* __restrict__ B,
+    const float* __restrict__ bias, void* __restrict__ C,
+    const __bf16* __restrict__ Yact,  // activation output (DACT_TANH)
+    int M, int N, int K) {
+  constexpr int BM = 64, BN = 64, BK = 32;
+  constexpr int LDA = BK + 8;  // bf16 elems; 16B-aligned rows, bank-spread
+  constexpr int LDBT = BK + 8;
+  __shared__ __bf16 As[BM][LDA];
+  __shared__ __bf16 Bs[BN][LDBT];  // stored TRANSPOSED: [n][k]
+
+  const int bm = blockIdx.x * BM;
+  const int bn = blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;         // 0..3
+  const int lane = tid & 63;
+  const int wr = wave >> 1;          // wave row 0..1
+  const int wc = wave & 1;           // wave col 0..1
+
+  f32x4 acc[2][2] = {};
+
+  const int row_a = lane & 15;
+  const int kseg = lane >> 4;  // 0..3 -> k-base = kseg*8
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A tile [64][32]: 256 threads x 8 elems = 2048 ----------
+    {
+      const int r = tid >> 2;             // 0..63
+      const int c8 = (tid & 3) * 8;       // 0,8,16,24
+      const int gr = bm + r;
+      bf16x8 v = {};
+      if (gr < M) {
+        const int gk = k0 + c8;
+        if (gk + 8 <= K) {
+          v = *reinterpret_cast<const bf16x8*>(&A[(int64_t)gr * K + gk]);
+        } else {
+          for (int i = 0; i < 8; ++i)
+            v[i] = (gk + i < K) ? A[(int64_t)gr * K + gk + i] : (__bf16)0.f;
+        }
+      }
+      *reinterpret_cast<bf16x8*>(&As[r][c8]) = v;
+    }
+    // ---- stage B tile transposed [64(n)][32(k)] -----------------------
+    {
+      const int nn = tid >> 2;            // 0..63 (col of C)
+      const int c8 = (tid & 3) * 8;
+      const int gn = bn + nn;
+      bf16x8 v = {};
+      if (gn < N) {
+        if (TRANS_B) {
+          // B stored [N, K]: row gn, cols k0+c8..+7 — contiguous
+          const int gk = k0 + c8;
+          if (gk + 8 <= K) {
+            v = *reinterpret_cast<const bf16x8*>(&B[(int64_t)gn * K + gk]);
+          } else {
+            for (int i = 0; i < 8; ++i)
+              v[i] = (gk + i < K) ? B[(int64_t)gn * K + gk + i] : (__bf16)0.f;
+          }
+        } else {
+          // B stored [K, N]: gather column gn (strided)
+          for (int i = 0; i < 8; ++i) {
+            const int gk = k0 + c8 + i;
+            v[i] = (gk < K) ? B[(int64_t)gk * N + gn] : (__bf16)0.f;
+          }
+        }
+      }
+      *reinterpret_cast<bf16x8*>(&Bs[nn][c8]) = v;
+    }
+    __syncthreads();
+
+    // ---- MFMA: 2x2 fragments per wave ---------------------------------
+    bf16x8 af[2], bf[2];
+    for (int mi = 0; mi < 2; ++mi)
+      af[mi] = *reinterpret_cast<const bf16x8*>(
+          &As[wr * 32 + mi * 16 + row_a][kseg * 8]);
+    for (int ni = 0; ni < 2; ++ni)
+      bf[ni] = *reinterpret_cast<const bf16x8*>(
+          &Bs[wc * 32 + ni * 16 + row_a][kseg * 8]);
+    for (int mi = 0; mi < 2; ++mi)
+      for (int ni = 0; ni < 2; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    __syncthreads();
+  }
+
+  // ---- epilogue -------------------------------------------------------
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+  for (int mi = 0; mi < 2; ++mi) {
+    for (int ni = 0; ni < 2; ++ni) {
+      const int gcol = bn + wc * 32 + ni * 16 + ccol;
+      if (gcol >= N) continue;
+      for (int r = 0; r < 4; ++r) {
+        const int grow = bm + wr * 32 + mi * 16 + crow_base + r;
+        if (grow >= M) continue;
+        float v = acc[mi][ni][r];
+        if (ADD_BIAS) v += bias[gcol];
+        if (ACT == 2) v = tanhf(v);
+        if (DACT_TANH) {
+          float y = bf2f(Yact[(int64_t)grow * N + gcol]);
+          v *= (1.f - y * y);
+        }
+        if (ACT == 0) {
+          reinterpret_cast<float*>(C)[(int64_t)grow * N + gcol] = v;
+        } else {
+          reinterpret_cast<__bf16*>(C)[(int64_t)grow * N + gcol] = f2bf(v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad partials: dW[K,N] = X^T @ dY, split over M into SLABS partial sums.
+// Each block: one (ktile, ntile, slab); loops its M-chunk with MFMA where
+// logical A = X^T [K, M] (i.e. A[k][m] = X[m][k]), logical B = dY [M, N].
+// Deterministic: fixed slab count + ordered tree reduce (no atomics).
+// Partials: f32 [S, K, N]; db partials f32 [S, N] (bias grad = colsum dY).
+// ---------------------------------------------------------------------------
+template <bool WANT_DB>
+__global__ __launch_bounds__(256) void wgrad_partial_kernel(
+    const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
+    float* __restrict__ dW_part, float* __restrict__ db_part,
+    int M, int N, int K, int slabs) {
+  constexpr int BM = 64, BN = 64, BK = 32;  // BM: K-dim tile, BN: N tile, BK: M chunk step
+  // naming: output tile is [BM of K] x [BN of N]; reduction dim is M.
+  __shared__ __bf16 Xs[BM][BK + 8];   // X^T tile: [k][m]
+  __shared__ __bf16 Ys[BN][BK + 8];   // dY^T tile: [n][m]
+
+  const int ktile = blockIdx.x;
+  const int ntile = blockIdx.y;
+  const int slab = blockIdx.z;
+  const int bk = ktile * BM;
+  const int bn = ntile * BN;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+
+  const int m_per_slab = (M + slabs - 1) / slabs;
+  const int m_begin = slab * m_per_slab;
+  const int m_end = min(M, m_begin + m_per_slab);
+
+  f32x4 acc[2][2] = {};
+  float db_acc = 0.f;  // per-thread partial for db (cols handled below)
+
+  for (int m0 = m_begin; m0 < m_end; m0 += BK) {
+    // stage X^T tile: Xs[k][m] = X[m0+m][bk+k]
+    {
+      const int k = tid >> 2;            // 0..63
+      const int m8 = (tid & 3) * 8;
+      bf16x8 v = {};
+      const int gk = bk + k;
+      if (gk < K) {
+        for (int i = 0; i < 8; ++i) {
+          const int gm = m0 + m8 + i;
+          v[i] = (gm < m_end) ? X[(int64_t)gm * K + gk] : (__bf16)0.f;
+        }
+      }
+      *reinterpret_cast<bf16x8*>(&Xs[k][m8]) = v;
+    }
+    // stage dY^T tile: Ys[n][m] = dY[m0+m][bn+n]
+    {
+      const int n = tid >> 2;
+      const int m8 = (tid & 3) * 8;
+      bf16x8 v = {};
+      const int gn = bn + n;
+      if (gn < N) {
+        for (int i = 0; i < 8; ++i) {
+          const int gm = m0 + m8 + i;
+          v[i] = (gm < m_end) ? dY[(int64_t)gm * N + gn] : (__bf16)0.f;
+        }
+      }
+      *reinterpret_cast<bf16x8*>(&Ys[n][m8]) = v;
+    }
+    __syncthreads();
+    if (WANT_DB && tid < BN) {
+      float s = 0.f;
+      for (int i = 0; i < BK && m0 + i < m_end; ++i) s += bf2f(Ys[tid][i]);
+      db_acc += s;
+    }
+
+    const int row_a = lane & 15;
+    const int kseg = lane >> 4;
+    bf16x8 af[2], bf_[2];
+    for (int mi = 0; mi < 2; ++mi)
+      af[mi] = *reinterpret_cast<const bf16x8*>(
+          &Xs[wr * 32 + mi * 16 + row_a][kseg * 8]);
+    for (int ni = 0; ni < 2; ++ni)
+      bf_[ni] = *reinterpret_cast<const bf16x8*>(
+          &Ys[wc * 32 + ni * 16 + row_a][kseg * 8]);
+    for (int mi = 0; mi < 2; ++mi)
+      for (int ni = 0; ni < 2; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
+    __syncthreads();
+  }
+
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+  float* out = dW_part + (int64_t)slab * K * N;
+  for (int mi = 0; mi < 2; ++mi)
+    for (int ni = 0; ni < 2; ++ni) {
+      const int gcol = bn + wc * 32 + ni * 16 + ccol;
+      if (gcol >= N) continue;
+      for (int r = 0; r < 4; ++r) {
+        const int grow = bk + wr * 32 + mi * 16 + crow_base + r;
+        if (grow >= K) continue;
+        out[(int64_t)grow * N + gcol] = acc[mi][ni][r];
+      }
+    }
+  if (WANT_DB && tid < BN && ktile == 0) {
+    const int gn = bn + tid;
+    if (gn < N) db_part[(int64_t)slab * N + gn] = db_acc;
+  }
+}
+
+__global__ void slab_reduce_kernel(const float* __restrict__ part,
+                                   float* __restrict__ out, int64_t elems,
+                                   int slabs) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < elems;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float s = 0.f;
+    for (int k = 0; k < slabs; ++k) s += part[(int64_t)k * elems + i];
+    out[i] = s;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// elementwise casts
+// ---------------------------------------------------------------------------
+__global__ void f32_to_bf16_kernel(const float* __restrict__ in,
+                                   __bf16* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = f2bf(in[i]);
+}
+
+// ---------------------------------------------------------------------------
+// GAE backward scan (per env): adv[t] = delta_t + gamma*lam*(1-done_t)*adv[t+1]
+// rewards/values/dones laid out [T, N] so lane reads coalesce across envs.
+// ---------------------------------------------------------------------------
+__global__ void gae_kernel(const float* __restrict__ rewards,
+                           const float* __restrict__ values,   // [T+1, N]
+                           const bool* __restrict__ dones,     // [T, N]
+                           float* __restrict__ adv, float* __restrict__ ret,
+                           int T, int N, float gamma, float lam) {
+  const int n = blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= N) return;
+  float running = 0.f;
+  for (int t = T - 1; t >= 0; --t) {
+    const float nonterm = dones[(int64_t)t * N + n] ? 0.f : 1.f;
+    const float delta = rewards[(int64_t)t * N + n] +
+                        gamma * values[(int64_t)(t + 1) * N + n] * nonterm -
+                        values[(int64_t)t * N + n];
+    running = delta + gamma * lam * nonterm * running;
+    adv[(int64_t)t * N + n] = running;
+    ret[(int64_t)t * N + n] = running + values[(int64_t)t * N + n];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused Adam (f32 master params; bf16 mirror refreshed in the same pass)
+// ---------------------------------------------------------------------------
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            __bf16* __restrict__ p_bf16, int64_t n, float lr,
+                            float beta1, float beta2, float eps, float bc1,
+                            float bc2, const float* __restrict__ gscale) {
+  const float s = gscale ? *gscale : 1.f;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const float gi = g[i] * s;
+    const float mi = beta1 * m[i] + (1.f - beta1) * gi;
+    const float vi = beta2 * v[i] + (1.f - beta2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    const float mhat = mi / bc1;
+    const float vhat = vi / bc2;
+    const float pi = p[i] - lr * mhat / (sqrtf(vhat) + eps);
+    p[i] = pi;
+    if (p_bf16) p_bf16[i] = f2bf(pi);
+  }
+}
+
+// grad-norm^2: deterministic two-pass (fixed grid of partials, ordered sum)
+__global__ void sumsq_partial_kernel(const float* __restrict__ g, int64_t n,
+                                     float* __restrict__ part) {
+  __shared__ float red[256];
+  float s = 0.f;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    s += g[i] * g[i];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) part[blockIdx.x] = red[0];
+}
+
+__global__ void clip_scale_kernel(const float* __restrict__ part, int nparts,
+                                  float max_norm, float* __restrict__ scale) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    float s = 0.f;
+    for (int i = 0; i < nparts; ++i) s += part[i];
+    const float norm = sqrtf(s);
+    *scale = (max_norm > 0.f && norm > max_norm) ? max_norm / norm : 1.f;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Categorical sampling + logp + entropy from the fused head output
+// head[M, A+1] f32: cols 0..A-1 logits, col A value.
+// Counter-based RNG (splitmix64): deterministic in (seed, step, env).
+// ---------------------------------------------------------------------------
+GFX_DEV uint64_t splitmix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+__global__ void sample_head_kernel(const float* __restrict__ head, int M,
+                                   int n_actions, uint64_t seed, uint64_t step,
+                                   int64_t* __restrict__ actions,
+                                   float* __restrict__ logp,
+                                   float* __restrict__ value,
+                                   float* __restrict__ entropy,
+                                   int greedy) {
+  const int m = blockIdx.x * blockDim.x + threadIdx.x;
+  if (m >= M) return;
+  const float* row = head + (int64_t)m * (n_actions + 1);
+  float mx = row[0];
+  for (int j = 1; j < n_actions; ++j) mx = fmaxf(mx, row[j]);
+  float z = 0.f;
+  for (int j = 0; j < n_actions; ++j) z += expf(row[j] - mx);
+  const float logz = logf(z) + mx;
+
+  int a = 0;
+  if (greedy) {
+    float best = row[0];
+    for (int j = 1; j < n_actions; ++j)
+      if (row[j] > best) { best = row[j]; a = j; }
+  } else {
+    const uint64_t r = splitmix64(seed ^ (step * 0x51E1F5ull + (uint64_t)m * 0x9E37ull));
+    float u = (float)((r >> 11) * (1.0 / 9007199254740992.0));  // [0,1)
+    u = fminf(u, 0.999999f);
+    float c = 0.f;
+    a = n_actions - 1;
+    for (int j = 0; j < n_actions; ++j) {
+      c += expf(row[j] - logz);
+      if (u < c) { a = j; break; }
+    }
+  }
+  actions[m] = a;
+  logp[m] = row[a] - logz;
+  if (value) value[m] = row[n_actions];
+  if (entropy) {
+    float h = 0.f;
+    for (int j = 0; j < n_actions; ++j) {
+      const float lp = row[j] - logz;
+      h -= expf(lp) * lp;
+    }
+    entropy[m] = h;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// PPO clipped-surrogate loss backward: head[M, A+1] f32 -> dhead[M, A+1] bf16
+// losses (logging): [pi_loss, v_loss, entropy, approx_kl, clipfrac] via
+// block-partial + atomicAdd (logging only; gradients are deterministic).
+// ---------------------------------------------------------------------------
+__global__ void ppo_loss_bwd_kernel(
+    const float* __restrict__ head, const int64_t* __restrict__ actions,
+    const float* __restrict__ old_logp, const float* __restrict__ adv,
+    const float* __restrict__ ret, __bf16* __restrict__ dhead, int M,
+    int n_actions, float clip_eps, float ent_coef, float vf_coef,
+    float inv_count, float* __restrict__ losses) {
+  const int m = blockIdx.x * blockDim.x + threadIdx.x;
+  float l_pi = 0.f, l_v = 0.f, l_ent = 0.f, l_kl = 0.f, l_clip = 0.f;
+  if (m < M) {
+    const float* row = head + (int64_t)m * (n_actions + 1);
+    __bf16* drow = dhead + (int64_t)m * (n_actions + 1);
+    float mx = row[0];
+    for (int j = 1; j < n_actions; ++j) mx = fmaxf(mx, row[j]);
+    float z = 0.f;
+    for (int j = 0; j < n_actions; ++j) z += expf(row[j] - mx);
+    const float logz = logf(z) + mx;
+    const int a = (int)actions[m];
+    const float lp = row[a] - logz;
+    const float ratio = expf(lp - old_logp[m]);
+    const float A = adv[m];
+    const bool clipped = (ratio > 1.f + clip_eps) || (ratio < 1.f - clip_eps);
+    const float surr1 = ratio * A;
+    const float rclip = fminf(fmaxf(ratio, 1.f - clip_eps), 1.f + clip_eps);
+    const float surr2 = rclip * A;
+    // d(-min(surr1,surr2))/dlogp = -A*ratio when surr1 <= surr2 (unclipped
+    // branch active), else 0
+    const float g_lp = (surr1 <= surr2) ? (-A * ratio) : 0.f;
+    // entropy bonus: loss += -ent_coef * H
+    float H = 0.f;
+    for (int j = 0; j < n_actions; ++j) {
+      const float lpj = row[j] - logz;
+      H -= expf(lpj) * lpj;
+    }
+    for (int j = 0; j < n_actions; ++j) {
+      const float pj = expf(row[j] - logz);
+      const float onehot = (j == a) ? 1.f : 0.f;
+      float d = g_lp * (onehot - pj);             // policy term
+      d += ent_coef * pj * ((row[j] - logz) + H);  // -ent_coef*dH/dlogit
+      drow[j] = f2bf(d * inv_count);
+    }
+    const float v = row[n_actions];
+    const float dv = vf_coef * (v - ret[m]);
+    drow[n_actions] = f2bf(dv * inv_count);
+
+    l_pi = -fminf(surr1, surr2);
+    l_v = 0.5f * (v - ret[m]) * (v - ret[m]);
+    l_ent = H;
+    l_kl = old_logp[m] - lp;
+    l_clip = clipped ? 1.f : 0.f;
+  }
+  // block reduction for logging scalars
+  __shared__ float red[256];
+  float vals[5] = {l_pi, l_v, l_ent, l_kl, l_clip};
+  for (int s = 0; s < 5; ++s) {
+    red[threadIdx.x] = vals[s];
+    __syncthreads();
+    for (int w = 128; w > 0; w >>= 1) {
+      if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0 && losses) atomicAdd(&losses[s], red[0] * inv_count);
+    __syncthreads();
+  }
+}
+
+// normalize advantages in-place: (a - mean) / (std + 1e-8), deterministic
+__global__ void adv_norm_stats_kernel(const float* __restrict__ adv, int64_t n,
+                                      float* __restrict__ part /*[blocks*2]*/) {
+  __shared__ float red[512];
+  float s = 0.f, ss = 0.f;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    s += adv[i];
+    ss += adv[i] * adv[i];
+  }
+  red[threadIdx.x] = s;
+  red[256 + threadIdx.x] = ss;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) {
+      red[threadIdx.x] += red[threadIdx.x + w];
+      red[256 + threadIdx.x] += red[256 + threadIdx.x + w];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    part[blockIdx.x] = red[0];
+    part[gridDim.x + blockIdx.x] = red[256];
+  }
+}
+
+__global__ void adv_norm_apply_kernel(float* __restrict__ adv, int64_t n,
+                                      const float* __restrict__ part,
+                                      int nparts) {
+  __shared__ float mean_s, inv_std_s;
+  if (threadIdx.x == 0) {
+    float s = 0.f, ss = 0.f;
+    for (int i = 0; i < nparts; ++i) { s += part[i]; ss += part[nparts + i]; }
+    const float mean = s / (float)n;
+    float var = ss / (float)n - mean * mean;
+    var = fmaxf(var, 0.f);
+    mean_s = mean;
+    inv_std_s = 1.f / (sqrtf(var) + 1e-8f);
+  }
+  __syncthreads();
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    adv[i] = (adv[i] - mean_s) * inv_std_s;
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
+
+void launch_gemm(const void* A, const void* B, const float* bias, void* C,
+                 const void* Yact, int M, int N, int K, bool trans_b, int act,
+                 bool dact_tanh, bool add_bias, hipStream_t stream) {
+  dim3 grid(ceil_div(M, 64), ceil_div(N, 64));
+  dim3 block(256);
+  const __bf16* a = reinterpret_cast<const __bf16*>(A);
+  const __bf16* b = reinterpret_cast<const __bf16*>(B);
+  const __bf16* y = reinterpret_cast<const __bf16*>(Yact);
+
+#define GEMM_CASE(TB, ACT, DT, AB)                                            \
+  hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB>), grid, block, 0, stream,  \
+                     a, b, bias, C, y, M, N, K)
+  if (!trans_b && !dact_tanh) {
+    if (act == 0) { if (add_bias) GEMM_CASE(false, 0, false, true); else GEMM_CASE(false, 0, false, false); }
+    else if (act == 1) { if (add_bias) GEMM_CASE(false, 1, false, true); else GEMM_CASE(false, 1, false, false); }
+    else { if (add_bias) GEMM_CASE(false, 2, false, true); else GEMM_CASE(false, 2, false, false); }
+  } else if (trans_b && !dact_tanh) {
+    if (act == 0) { if (add_bias) GEMM_CASE(true, 0, false, true); else GEMM_CASE(true, 0, false, false); }
+    else if (act == 1) { if (add_bias) GEMM_CASE(true, 1, false, true); else GEMM_CASE(true, 1, false, false); }
+    else { if (add_bias) GEMM_CASE(true, 2, false, true); else GEMM_CASE(true, 2, false, false); }
+  } else if (trans_b && dact_tanh) {
+    GEMM_CASE(true, 1, true, false);
+  } else {
+    GEMM_CASE(false, 1, true, false);
+  }
+#undef GEMM_CASE
+}
+
+void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
+                  float* dW, float* db, int M, int N, int K, int slabs,
+                  hipStream_t stream) {
+  dim3 grid(ceil_div(K, 64), ceil_div(N, 64), slabs);
+  if (db_part) {
+    hipLaunchKernelGGL((wgrad_partial_kernel<true>), grid, dim3(256), 0, stream,
+                       reinterpret_cast<const __bf16*>(X),
+                       reinterpret_cast<const __bf16*>(dY), dW_part, db_part, M,
+                       N, K, slabs);
+  } else {
+    hipLaunchKernelGGL((wgrad_partial_kernel<false>), grid, dim3(256), 0,
+                       stream, reinterpret_cast<const __bf16*>(X),
+                       reinterpret_cast<const __bf16*>(dY), dW_part, nullptr, M,
+                       N, K, slabs);
+  }
+  int64_t elems = (int64_t)K * N;
+  int blocks = (int)std::min<int64_t>((elems + 255) / 256, 1024);
+  hipLaunchKernelGGL(slab_reduce_kernel, dim3(blocks), dim3(256), 0, stream,
+                     dW_part, dW, elems, slabs);
+  if (db_part && db)
+    hipLaunchKernelGGL(slab_reduce_kernel, dim3(1), dim3(256), 0, stream,
+                       db_part, db, (int64_t)N, slabs);
+}
+
+void launch_gae(const float* rewards, const float* values, const bool* dones,
+                float* adv, float* ret, int T, int N, float gamma, float lam,
+                hipStream_t stream) {
+  hipLaunchKernelGGL(gae_kernel, dim3(ceil_div(N, 256)), dim3(256), 0, stream,
+                     rewards, values, dones, adv, ret, T, N, gamma, lam);
+}
+
+void launch_adam(float* p, const float* g, float* m, float* v, void* p_bf16,
+                 int64_t n, float lr, float beta1, float beta2, float eps,
+                 float bc1, float bc2, const float* gscale,
+                 hipStream_t stream) {
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, stream, p, g, m,
+                     v, reinterpret_cast<__bf16*>(p_bf16), n, lr, beta1, beta2,
+                     eps, bc1, bc2, gscale);
+}
+
+void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
+                      float* scale, int nparts, hipStream_t stream) {
+  hipLaunchKernelGGL(sumsq_partial_kernel, dim3(nparts), dim3(256), 0, stream,
+                     g, n, part);
+  hipLaunchKernelGGL(clip_scale_kernel, dim3(1), dim3(1), 0, stream, part,
+                     nparts, max_norm, scale);
+}
+
+void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,
+                        uint64_t step, int64_t* actions, float* logp,
+                        float* value, float* entropy, int greedy,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(sample_head_kernel, dim3(ceil_div(M, 256)), dim3(256), 0,
+                     stream, head, M, n_actions, seed, step, actions, logp,
+                     value, entropy, greedy);
+}
+
+void launch_ppo_loss_bwd(const float* head, const int64_t* actions,
+                         const float* old_logp, const float* adv,
+                         const float* ret, void* dhead, int M, int n_actions,
+                         float clip_eps, float ent_coef, float vf_coef,
+                         float inv_count, float* losses, hipStream_t stream) {
+  hipLaunchKernelGGL(ppo_loss_bwd_kernel, dim3(ceil_div(M, 256)), dim3(256), 0,
+                     stream, head, actions, old_logp, adv, ret,
+                     reinterpret_cast<__bf16*>(dhead), M, n_actions, clip_eps,
+                     ent_coef, vf_coef, inv_count, losses);
+}
+
+void launch_adv_normalize(float* adv, int64_t n, float* part, int nparts,
+                          hipStream_t stream) {
+  hipLaunchKernelGGL(adv_norm_stats_kernel, dim3(nparts), dim3(256), 0, stream,
+                     adv, n, part);
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(adv_norm_apply_kernel, dim3(blocks), dim3(256), 0, stream,
+                     adv, n, part, nparts);
+}
+
+void launch_f32_to_bf16(const float* in, void* out, int64_t n,
+                        hipStream_t stream) {
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(f32_to_bf16_kernel, dim3(blocks), dim3(256), 0, stream, in,
+                     reinterpret_cast<__bf16*>(out), n);
+}
+
+}  // namespace gymfx

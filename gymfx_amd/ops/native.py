@@ -34,7 +34,7 @@ def available() -> bool:
 
 
 def require():
-    """Return the native module wrapper; raise loudly when missing on GPU."""
+    """Return the native extension module; raise loudly when missing."""
     ext = load()
     if ext is None:
         raise RuntimeError(
@@ -43,9 +43,7 @@ def require():
             "(PYTORCH_ROCM_ARCH=gfx950). The eager torch path is only a CPU "
             "oracle and is not used on GPU."
         )
-    from . import wrappers  # noqa: PLC0415
-
-    return wrappers
+    return ext
 
 
 __all__ = ["load", "available", "require"]

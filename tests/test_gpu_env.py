@@ -1,0 +1,123 @@
+"""GPU: the fused HIP env_step/build_obs kernels vs the torch oracle
+(envs/reference_step.py), elementwise, across strategy/reward/preprocessor
+configurations and many steps."""
+import numpy as np
+import pytest
+import torch
+
+from gymfx_amd import build_vec_environment
+from gymfx_amd.data.feed import synthetic_ohlcv
+
+pytestmark = pytest.mark.gpu
+
+N = 64
+STEPS = 220
+
+
+def _market():
+    return synthetic_ohlcv(2000, seed=5, vol=4e-4, extra_feature_columns=3)
+
+
+BASE = {
+    "n_envs": N,
+    "window_size": 16,
+    "initial_cash": 10000.0,
+    "position_size": 1000.0,
+    "commission": 2e-5,
+    "slippage": 1e-5,
+    "env_start_mode": "spread",
+}
+
+CONFIGS = {
+    "default": {},
+    "fixed_sltp": {"strategy_plugin": "direct_fixed_sltp", "sl_pips": 10.0, "tp_pips": 20.0},
+    "atr": {
+        "strategy_plugin": "direct_atr_sltp",
+        "rel_volume": 0.1,
+        "leverage": 10.0,
+        "atr_period": 7,
+    },
+    "sharpe": {"reward_plugin": "sharpe_reward", "window": 16},
+    "dd": {"reward_plugin": "dd_penalized_reward", "penalty_lambda": 0.5},
+    "feature_window": {
+        "preprocessor_plugin": "feature_window_preprocessor",
+        "feature_columns": ["OPEN", "CLOSE", "FEAT_0", "FEAT_1"],
+        "feature_scaling": "rolling_zscore",
+        "feature_scaling_window": 64,
+    },
+    "autoreset": {"autoreset": True, "strategy_plugin": "direct_fixed_sltp",
+                  "sl_pips": 3.0, "tp_pips": 3.0},
+    "continuous": {"action_space_mode": "continuous"},
+}
+
+
+@pytest.mark.parametrize("name", sorted(CONFIGS))
+def test_kernel_matches_torch_oracle(name):
+    md = _market()
+    cfg = {**BASE, **CONFIGS[name]}
+    env_g = build_vec_environment({**cfg, "device": "cuda"}, md, use_native=True)
+    env_c = build_vec_environment({**cfg, "device": "cuda"}, md, use_native=False)
+    env_g.reset(seed=0)
+    env_c.reset(seed=0)
+    rng = np.random.default_rng(9)
+    for k in range(STEPS):
+        if cfg.get("action_space_mode") == "continuous":
+            a = torch.from_numpy(rng.uniform(-1, 1, N).astype(np.float32)).cuda()
+        else:
+            a = torch.from_numpy(rng.integers(0, 3, N)).cuda()
+        out_g = env_g.step(a)
+        out_c = env_c.step(a)
+        torch.cuda.synchronize()
+        for fld in ("equity", "cash", "pos", "avg_entry", "commission_paid"):
+            g = getattr(env_g.st, fld).cpu().numpy()
+            c = getattr(env_c.st, fld).cpu().numpy()
+            np.testing.assert_allclose(g, c, rtol=1e-9, atol=1e-9,
+                                       err_msg=f"{name} step {k} field {fld}")
+        np.testing.assert_array_equal(
+            env_g.st.cursor.cpu().numpy(), env_c.st.cursor.cpu().numpy()
+        )
+        np.testing.assert_array_equal(
+            env_g.st.terminated.cpu().numpy(), env_c.st.terminated.cpu().numpy()
+        )
+        np.testing.assert_allclose(
+            out_g["reward"].cpu().numpy(),
+            out_c["reward"].cpu().float().numpy(),
+            rtol=2e-5, atol=1e-7, err_msg=f"{name} step {k} reward",
+        )
+        np.testing.assert_allclose(
+            out_g["obs"].cpu().numpy(), out_c["obs"].cpu().numpy(),
+            rtol=3e-5, atol=3e-5, err_msg=f"{name} step {k} obs",
+        )
+        np.testing.assert_array_equal(
+            env_g.st.exec_diag.cpu().numpy(), env_c.st.exec_diag.cpu().numpy(),
+            err_msg=f"{name} step {k} exec_diag",
+        )
+
+
+def test_gpu_determinism_bitwise():
+    """Same run twice on GPU -> bit-identical equity trajectory."""
+    md = _market()
+    cfg = {**BASE, "strategy_plugin": "direct_fixed_sltp", "autoreset": True,
+           "device": "cuda"}
+
+    def run():
+        env = build_vec_environment(cfg, md, use_native=True)
+        env.reset(seed=0)
+        rng = np.random.default_rng(3)
+        tail = []
+        for _ in range(150):
+            a = torch.from_numpy(rng.integers(0, 3, N)).cuda()
+            env.step(a)
+        return env.st.equity.cpu().numpy().copy(), env.st.trade_count.cpu().numpy().copy()
+
+    e1, t1 = run()
+    e2, t2 = run()
+    np.testing.assert_array_equal(e1, e2)
+    np.testing.assert_array_equal(t1, t2)
+
+
+def test_native_required_on_gpu():
+    """GPU default path must use the HIP engine (no silent eager fallback)."""
+    md = _market()
+    env = build_vec_environment({**BASE, "device": "cuda"}, md)
+    assert env._native is not None
