@@ -15,8 +15,10 @@
 
 namespace gymfx {
 
-GFX_DEV double buy_fill(double o, double slip) { return o * (1.0 + slip); }
-GFX_DEV double sell_fill(double o, double slip) { return o * (1.0 - slip); }
+// Fill prices are computed in f32 to match the torch oracle exactly
+// (f32 market tensor * f32-cast scalar); the money ledger then runs in f64.
+GFX_DEV float buy_fill(float o, float slip) { return o * (1.0f + slip); }
+GFX_DEV float sell_fill(float o, float slip) { return o * (1.0f - slip); }
 
 // Close the full position at `fill`; mirrors reference_step._close_position.
 GFX_DEV void close_position(const EnvPtrs& P, const EnvParamsK& K, int n,
@@ -94,7 +96,7 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   const int n = blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= K.n_envs) return;
   const int T = K.T;
-  const double slip = K.slippage;
+  const float slip = (float)K.slippage;
   int* ediag = P.exec_diag + (int64_t)n * EXEC_COUNTER_N;
   int* adiag = P.act_diag + (int64_t)n * ACT_COUNTER_N;
 
@@ -156,21 +158,21 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   }
   const bool valid = adv && !exhausted;
   const int t = max(P.cursor[n] - 1, 0);
-  const double o_px = P.open_px[t];
-  const double h_px = P.high_px[t];
-  const double l_px = P.low_px[t];
-  const double c_px = P.close_px[t];
+  const float o_px = P.open_px[t];
+  const float h_px = P.high_px[t];
+  const float l_px = P.low_px[t];
+  const float c_px = P.close_px[t];
 
   if (valid || first) P.last_trade_cost[n] = 0.0;
 
   // ---- 1. pending market fills at open(t) ----------------------------
   if (valid && P.pend_close[n] && P.pos[n] != 0.0) {
-    double fill = (P.pos[n] < 0) ? buy_fill(o_px, slip) : sell_fill(o_px, slip);
-    close_position(P, K, n, fill, -1);
+    float fill = (P.pos[n] < 0) ? buy_fill(o_px, slip) : sell_fill(o_px, slip);
+    close_position(P, K, n, (double)fill, -1);
   }
   if (valid && P.pend_open_dir[n] != 0 && P.pos[n] == 0.0) {
     double dir = (double)P.pend_open_dir[n];
-    double fill = dir > 0 ? buy_fill(o_px, slip) : sell_fill(o_px, slip);
+    double fill = (double)(dir > 0 ? buy_fill(o_px, slip) : sell_fill(o_px, slip));
     double size = (double)P.pend_open_size[n];
     double notional = size * fill;
     double comm = notional * K.commission;
@@ -197,8 +199,8 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   // ---- 2. bracket children, worst-case ordering ----------------------
   if (valid && P.br_active[n] && !P.br_armed[n] && P.pos[n] != 0.0) {
     const bool is_long = P.pos[n] > 0;
-    const double sl = P.br_sl[n], tp = P.br_tp[n];
-    double trig = 0.0;
+    const float sl = P.br_sl[n], tp = P.br_tp[n];
+    float trig = 0.f;
     int hit = 0;  // 0 none, 1 sl, 2 tp
     if (is_long) {
       if (o_px <= sl) { hit = 1; trig = o_px; }
@@ -212,8 +214,9 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
       else if (l_px <= tp) { hit = 2; trig = tp; }
     }
     if (hit) {
-      double fill = is_long ? sell_fill(trig, slip) : buy_fill(trig, slip);
-      close_position(P, K, n, fill, hit == 1 ? E_BRACKET_SL_FILLS : E_BRACKET_TP_FILLS);
+      float fill = is_long ? sell_fill(trig, slip) : buy_fill(trig, slip);
+      close_position(P, K, n, (double)fill,
+                     hit == 1 ? E_BRACKET_SL_FILLS : E_BRACKET_TP_FILLS);
     }
   }
   if (valid) P.br_armed[n] = false;
@@ -227,16 +230,15 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
     const int Pn = K.atr_period;
     float pc = P.prev_close_atr[n];
     float tr = isnan(pc)
-                   ? (float)(h_px - l_px)
-                   : fmaxf((float)(h_px - l_px),
-                           fmaxf(fabsf((float)(h_px - pc)), fabsf((float)(l_px - pc))));
+                   ? (h_px - l_px)
+                   : fmaxf(h_px - l_px, fmaxf(fabsf(h_px - pc), fabsf(l_px - pc)));
     int cnt = P.tr_count[n];
     int idx = cnt % Pn;
     float old = P.tr_ring[(int64_t)n * Pn + idx];
     P.tr_sum[n] += tr - (cnt >= Pn ? old : 0.f);
     P.tr_ring[(int64_t)n * Pn + idx] = tr;
     P.tr_count[n] = cnt + 1;
-    P.prev_close_atr[n] = (float)c_px;
+    P.prev_close_atr[n] = c_px;
     int n_tr = min(P.tr_count[n], Pn);
     atr = P.tr_sum[n] / (float)max(n_tr, 1);
     atr_ready = P.tr_count[n] >= Pn;
@@ -274,15 +276,15 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   } else if (dec && K.strategy_id == STRAT_FIXED) {
     if (a == 1 || a == 2) {
       const double pos = P.pos[n];
-      const double sl_d = K.sl_pips * K.pip_size;
-      const double tp_d = K.tp_pips * K.pip_size;
+      const float sl_d = (float)(K.sl_pips * K.pip_size);
+      const float tp_d = (float)(K.tp_pips * K.pip_size);
       if (a == 1) {
         if (pos < 0) P.pend_close[n] = true;
         if (pos <= 0) {
           P.pend_open_dir[n] = 1;
           P.pend_open_size[n] = (float)K.position_size;
-          P.pend_sl[n] = (float)(c_px - sl_d);
-          P.pend_tp[n] = (float)(c_px + tp_d);
+          P.pend_sl[n] = c_px - sl_d;
+          P.pend_tp[n] = c_px + tp_d;
           ediag[E_ENTRY_ORDERS_SUBMITTED] += 1;
         }
       } else {
@@ -290,8 +292,8 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
         if (pos >= 0) {
           P.pend_open_dir[n] = -1;
           P.pend_open_size[n] = (float)K.position_size;
-          P.pend_sl[n] = (float)(c_px + sl_d);
-          P.pend_tp[n] = (float)(c_px - tp_d);
+          P.pend_sl[n] = c_px + sl_d;
+          P.pend_tp[n] = c_px - tp_d;
           ediag[E_ENTRY_ORDERS_SUBMITTED] += 1;
         }
       }
@@ -312,11 +314,11 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
       double size = K.position_size;
       if (K.flags & F_HAS_RELVOL) {
         double raw_sz = P.cash[n] * K.rel_volume * K.leverage;
-        if (K.size_mode == SIZE_NOTIONAL) raw_sz = c_px > 0 ? raw_sz / c_px : 0.0;
+        if (K.size_mode == SIZE_NOTIONAL) raw_sz = c_px > 0.f ? raw_sz / (double)c_px : 0.0;
         size = fmin(fmax(raw_sz, K.min_order_volume), K.max_order_volume);
       }
       if (ok && size <= 0.0) { ediag[E_BLOCKED_NON_POSITIVE_SIZE] += 1; ok = false; }
-      if (ok && c_px <= 0.0) { ediag[E_BLOCKED_NON_POSITIVE_PRICE] += 1; ok = false; }
+      if (ok && c_px <= 0.f) { ediag[E_BLOCKED_NON_POSITIVE_PRICE] += 1; ok = false; }
       if (ok) {
         // risk-mode k shrink (direct_atr_sltp.py:263-289)
         double k_sl_eff = K.k_sl, k_tp_eff = K.k_tp;
@@ -331,33 +333,34 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
           }
           k_tp_eff = fmax(k_tp_eff, k_sl_eff * K.min_rr);
         }
-        double sl_dist = k_sl_eff * (double)atr;
-        double tp_dist = k_tp_eff * (double)atr;
+        // f32 arithmetic with f32-cast scalars — op-for-op the torch oracle
+        float sl_dist = (float)k_sl_eff * atr;
+        float tp_dist = (float)k_tp_eff * atr;
         if (K.risk_mode == RISK_MARGIN && (K.flags & F_HAS_MPLF)) {
           double rel = fmax(0.0, (K.flags & F_HAS_RELVOL) ? K.rel_volume : 0.0);
           double mlf = fmax(0.0, K.mplf);
           if (rel > 0.0 && mlf > 0.0)
-            sl_dist = fmin(sl_dist, c_px * mlf / (rel * K.leverage));
+            sl_dist = fminf(sl_dist, c_px * (float)(mlf / (rel * K.leverage)));
         }
         if (K.flags & F_HAS_MINFRAC) {
-          double floor_d = K.min_sltp_frac * c_px;
-          sl_dist = fmax(sl_dist, floor_d);
-          tp_dist = fmax(tp_dist, floor_d);
+          float floor_d = (float)K.min_sltp_frac * c_px;
+          sl_dist = fmaxf(sl_dist, floor_d);
+          tp_dist = fmaxf(tp_dist, floor_d);
         }
         if (K.flags & F_HAS_MAXFRAC) {
-          double ceil_d = K.max_sltp_frac * c_px;
-          sl_dist = fmin(sl_dist, ceil_d);
-          tp_dist = fmin(tp_dist, ceil_d);
+          float ceil_d = (float)K.max_sltp_frac * c_px;
+          sl_dist = fminf(sl_dist, ceil_d);
+          tp_dist = fminf(tp_dist, ceil_d);
         }
-        if (tp_dist >= c_px) tp_dist = c_px * 0.5;
+        if (tp_dist >= c_px) tp_dist = c_px * 0.5f;
         const double pos = P.pos[n];
         if (a == 1) {
           if (pos < 0) P.pend_close[n] = true;
           if (pos <= 0) {
             P.pend_open_dir[n] = 1;
             P.pend_open_size[n] = (float)size;
-            P.pend_sl[n] = (float)(c_px - sl_dist);
-            P.pend_tp[n] = (float)(c_px + tp_dist);
+            P.pend_sl[n] = c_px - sl_dist;
+            P.pend_tp[n] = c_px + tp_dist;
             ediag[E_ENTRY_ORDERS_SUBMITTED] += 1;
           }
         } else {
@@ -365,8 +368,8 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
           if (pos >= 0) {
             P.pend_open_dir[n] = -1;
             P.pend_open_size[n] = (float)size;
-            P.pend_sl[n] = (float)(c_px + sl_dist);
-            P.pend_tp[n] = (float)(c_px - tp_dist);
+            P.pend_sl[n] = c_px + sl_dist;
+            P.pend_tp[n] = c_px - tp_dist;
             ediag[E_ENTRY_ORDERS_SUBMITTED] += 1;
           }
         }
@@ -382,7 +385,7 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   double reward = 0.0, base_reward = 0.0, penalty = 0.0;
   if (pub) {
     P.prev_equity[n] = P.equity[n];
-    double unreal = P.pos[n] * (c_px - P.avg_entry[n]);
+    double unreal = P.pos[n] * ((double)c_px - P.avg_entry[n]);
     P.equity[n] = P.cash[n] + P.margin_used[n] + unreal;
     if (P.equity[n] <= K.min_equity) P.terminated[n] = true;
     if (valid && P.cursor[n] >= T) P.terminated[n] = true;
