@@ -490,7 +490,9 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K) {
           var = fmax(var, 0.0);
           double stdv = sqrt(var);
           if (stdv < 1e-8) stdv = 1.0;
-          val = (float)(((double)x - mean) / stdv);
+          // f64 stats cast to f32 BEFORE the subtract/divide — the oracle's
+          // exact op order (reference_step.build_obs_torch).
+          val = (x - (float)mean) / (float)stdv;
         }
       }
       // NaN guard BEFORE the clamp (fminf/fmaxf silently drop NaNs on CDNA):

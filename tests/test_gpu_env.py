@@ -79,10 +79,12 @@ def test_kernel_matches_torch_oracle(name):
         np.testing.assert_array_equal(
             env_g.st.terminated.cpu().numpy(), env_c.st.terminated.cpu().numpy()
         )
+        # sharpe's ring mean/var differ by f32 summation ORDER between the
+        # kernel (sequential) and torch (pairwise) — tolerance, not equality.
         np.testing.assert_allclose(
             out_g["reward"].cpu().numpy(),
             out_c["reward"].cpu().float().numpy(),
-            rtol=2e-5, atol=1e-7, err_msg=f"{name} step {k} reward",
+            rtol=5e-4, atol=1e-6, err_msg=f"{name} step {k} reward",
         )
         np.testing.assert_allclose(
             out_g["obs"].cpu().numpy(), out_c["obs"].cpu().numpy(),
