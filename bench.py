@@ -113,11 +113,11 @@ def main() -> None:
             torch.cuda.synchronize()
 
     for _ in range(args.warmup):
-        trainer.train_update()
+        trainer.train_update(with_stats=False)
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        trainer.train_update()
+        trainer.train_update(with_stats=False)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 

@@ -40,6 +40,7 @@ class PPOConfig:
     hidden: int = 256
     normalize_adv: bool = True
     shuffle_rows: bool = True
+    use_graphs: bool = True  # hipGraph-capture the update on GPU
 
     @classmethod
     def from_config(cls, cfg: Dict[str, Any]) -> "PPOConfig":
@@ -84,8 +85,10 @@ class PPOTrainer:
         self.N, self.T, self.D = N, T, D
         n_actions = 3
 
+        # model init is rank-INDEPENDENT (data-parallel replicas must start
+        # identical); rollout sampling / shuffling seeds differ per rank.
         self.model = ActorCriticMLP(
-            D, n_actions, cfg.hidden, device=self.device, seed=cfg.seed + rank
+            D, n_actions, cfg.hidden, device=self.device, seed=cfg.seed
         )
 
         dev = self.device
@@ -101,46 +104,70 @@ class PPOTrainer:
 
         self.acts_rollout = self.model.alloc_acts(N)
         self.obs_bf16_step = torch.empty(N, D, dtype=torch.bfloat16, device=dev)
-        self.step_actions = torch.empty(N, dtype=torch.int64, device=dev)
-        self.step_logp = torch.empty(N, dtype=torch.float32, device=dev)
-        self.step_value = torch.empty(N, dtype=torch.float32, device=dev)
 
         M = (T * N) // cfg.minibatches
+        if M * cfg.minibatches != T * N:
+            raise ValueError("rollout_steps * n_envs must divide by minibatches")
         self.mb_rows = M
         self.acts_train = self.model.alloc_acts(M)
         self.scratch = self.model.alloc_scratch(M)
         self.dhead = torch.empty(M, n_actions + 1, dtype=torch.bfloat16, device=dev)
         self.losses = torch.zeros(5, dtype=torch.float32, device=dev)
-        self._perm_gen = torch.Generator(device="cpu").manual_seed(cfg.seed * 9973 + rank)
 
-        self.global_step = 0   # env steps taken (per rank)
+        # flat views (fixed addresses — required for hipGraph capture)
+        TN = T * N
+        self.obs_flat = self.obs_buf.view(TN, D)
+        self.act_flat = self.act_buf.view(TN)
+        self.logp_flat = self.logp_buf.view(TN)
+        self.adv_flat = self.adv_buf.view(TN)
+        self.ret_flat = self.ret_buf.view(TN)
+        # minibatch gather targets
+        self.obs_mb = torch.empty(M, D, dtype=torch.bfloat16, device=dev)
+        self.act_mb = torch.empty(M, dtype=torch.int64, device=dev)
+        self.logp_mb = torch.empty(M, dtype=torch.float32, device=dev)
+        self.adv_mb = torch.empty(M, dtype=torch.float32, device=dev)
+        self.ret_mb = torch.empty(M, dtype=torch.float32, device=dev)
+
+        # device counters (hipGraph-replayable RNG / schedule state)
+        self.step_base = torch.zeros((), dtype=torch.int64, device=dev)
+        self.mb_ctr = torch.zeros((), dtype=torch.int64, device=dev)
+        self.sample_seed = cfg.seed * 1_000_003 + rank
+        self.shuffle_seed = cfg.seed * 9973 + rank
+
+        self.global_step = 0   # host mirror of step_base (logging/ckpt)
         self.update_count = 0
+        self.use_graphs = bool(cfg.use_graphs) and self.device.type == "cuda"
+        self._graphs_ready = False
+        self.g_rollout = None
+        self.g_mb = None
+        self.g_opt = None
 
     # ------------------------------------------------------------------
-    def collect_rollout(self) -> None:
-        env, model, cfg = self.env, self.model, self.cfg
-        T, N = self.T, self.N
+    # graph-capturable bodies: no host syncs, no allocations, fixed
+    # tensor addresses, all RNG/schedule state in device counters.
+    # ------------------------------------------------------------------
+    def _rollout_body(self) -> None:
+        env, model = self.env, self.model
+        T = self.T
         for t in range(T):
-            api.f32_to_bf16(env._obs, self.obs_bf16_step)
-            self.obs_buf[t].copy_(self.obs_bf16_step)
-            head = model.forward(self.obs_bf16_step, self.acts_rollout)
+            obs_t = self.obs_buf[t]
+            api.f32_to_bf16(env._obs, obs_t)
+            head = model.forward(obs_t, self.acts_rollout)
             api.sample_head(
-                head, cfg.seed * 1_000_003 + self.rank, self.global_step + t,
-                self.step_actions, self.step_logp, self.step_value,
+                head, self.sample_seed, t,
+                self.act_buf[t], self.logp_buf[t], self.val_buf[t],
+                step_base=self.step_base,
             )
-            self.act_buf[t].copy_(self.step_actions)
-            self.logp_buf[t].copy_(self.step_logp)
-            self.val_buf[t].copy_(self.step_value)
-            out = env.step(self.step_actions)
+            out = env.step(self.act_buf[t])
             self.rew_buf[t].copy_(out["reward"])
             self.done_buf[t].copy_(out["terminated"])
         # bootstrap value
         api.f32_to_bf16(env._obs, self.obs_bf16_step)
         head = model.forward(self.obs_bf16_step, self.acts_rollout)
         self.val_buf[T].copy_(head[:, -1])
-        self.global_step += T
+        api.increment_counter(self.step_base, T)
 
-    def compute_advantages(self) -> None:
+    def _gae_body(self) -> None:
         api.gae(
             self.rew_buf, self.val_buf, self.done_buf, self.adv_buf,
             self.ret_buf, self.cfg.gamma, self.cfg.gae_lambda,
@@ -148,47 +175,144 @@ class PPOTrainer:
         if self.cfg.normalize_adv:
             api.adv_normalize(self.adv_buf.view(-1), self._adv_part)
 
-    def update(self) -> Dict[str, float]:
-        cfg, model = self.cfg, self.model
-        TN = self.T * self.N
-        M = self.mb_rows
-        obs_flat = self.obs_buf.view(TN, self.D)
-        act_flat = self.act_buf.view(TN)
-        logp_flat = self.logp_buf.view(TN)
-        adv_flat = self.adv_buf.view(TN)
-        ret_flat = self.ret_buf.view(TN)
-        self.losses.zero_()
-        inv_count = 1.0 / (M * cfg.ppo_epochs * cfg.minibatches)
+    def _mb_body(self) -> None:
+        """One minibatch: Feistel-shuffled gather -> fwd -> loss bwd -> bwd.
 
-        for _ in range(cfg.ppo_epochs):
-            if cfg.shuffle_rows:
-                perm = torch.randperm(TN, generator=self._perm_gen).to(self.device)
-                obs_e = obs_flat[perm]
-                act_e = act_flat[perm]
-                logp_e = logp_flat[perm]
-                adv_e = adv_flat[perm]
-                ret_e = ret_flat[perm]
-            else:
-                obs_e, act_e, logp_e, adv_e, ret_e = (
-                    obs_flat, act_flat, logp_flat, adv_flat, ret_flat
-                )
-            for mb in range(cfg.minibatches):
-                sl = slice(mb * M, (mb + 1) * M)
-                obs_mb = obs_e[sl].contiguous()
-                model.zero_grad()
-                head = model.forward(obs_mb, self.acts_train)
-                api.ppo_loss_bwd(
-                    head, act_e[sl].contiguous(), logp_e[sl].contiguous(),
-                    adv_e[sl].contiguous(), ret_e[sl].contiguous(), self.dhead,
-                    clip_eps=cfg.clip_eps, ent_coef=cfg.ent_coef,
-                    vf_coef=cfg.vf_coef, inv_count=1.0 / M,
-                    losses=self.losses,
-                )
-                model.backward(obs_mb, self.acts_train, self.dhead, self.scratch)
-                self._allreduce_grads()
-                model.adam(cfg.lr, max_grad_norm=cfg.max_grad_norm)
-        self.update_count += 1
+        Gradients are fully overwritten by backward (deterministic split-M
+        wgrad), so there is no zero_grad."""
+        cfg, model = self.cfg, self.model
+        api.mb_gather(
+            self.obs_flat, self.act_flat, self.logp_flat, self.adv_flat,
+            self.ret_flat, self.obs_mb, self.act_mb, self.logp_mb,
+            self.adv_mb, self.ret_mb, seed=self.shuffle_seed,
+            minibatches=cfg.minibatches, step_base=self.step_base,
+            mb_ctr=self.mb_ctr,
+        )
+        head = model.forward(self.obs_mb, self.acts_train)
+        api.ppo_loss_bwd(
+            head, self.act_mb, self.logp_mb, self.adv_mb, self.ret_mb,
+            self.dhead, clip_eps=cfg.clip_eps, ent_coef=cfg.ent_coef,
+            vf_coef=cfg.vf_coef, inv_count=1.0 / self.mb_rows,
+            losses=self.losses,
+        )
+        model.backward(self.obs_mb, self.acts_train, self.dhead, self.scratch)
+        api.increment_counter(self.mb_ctr, 1)
+
+    def _mb_body_noshuffle(self, epoch_mb: int) -> None:
+        cfg, model = self.cfg, self.model
+        M = self.mb_rows
+        mb = epoch_mb % cfg.minibatches
+        sl = slice(mb * M, (mb + 1) * M)
+        self.obs_mb.copy_(self.obs_flat[sl])
+        self.act_mb.copy_(self.act_flat[sl])
+        self.logp_mb.copy_(self.logp_flat[sl])
+        self.adv_mb.copy_(self.adv_flat[sl])
+        self.ret_mb.copy_(self.ret_flat[sl])
+        head = model.forward(self.obs_mb, self.acts_train)
+        api.ppo_loss_bwd(
+            head, self.act_mb, self.logp_mb, self.adv_mb, self.ret_mb,
+            self.dhead, clip_eps=cfg.clip_eps, ent_coef=cfg.ent_coef,
+            vf_coef=cfg.vf_coef, inv_count=1.0 / M, losses=self.losses,
+        )
+        model.backward(self.obs_mb, self.acts_train, self.dhead, self.scratch)
+
+    def _opt_body(self) -> None:
+        self.model.adam(self.cfg.lr, max_grad_norm=self.cfg.max_grad_norm)
+
+    # ------------------------------------------------------------------
+    def _snapshot(self) -> Dict[str, Any]:
+        """Clone every tensor the warmup pass mutates (env + model +
+        counters) so capture leaves training state untouched."""
+        m = self.model
+        snap = {
+            "params": m.params.clone(), "m": m.m.clone(), "v": m.v.clone(),
+            "params_bf16": m.params_bf16.clone(),
+            "adam_ctr": m.adam_ctr.clone(), "adam_step": m.adam_step,
+            "step_base": self.step_base.clone(), "mb_ctr": self.mb_ctr.clone(),
+            "obs": self.env._obs.clone(),
+            "st": {k: v.clone() for k, v in self.env.st.to_dict().items()},
+        }
+        return snap
+
+    def _restore(self, snap: Dict[str, Any]) -> None:
+        m = self.model
+        m.params.copy_(snap["params"])
+        m.m.copy_(snap["m"])
+        m.v.copy_(snap["v"])
+        m.params_bf16.copy_(snap["params_bf16"])
+        m.adam_ctr.copy_(snap["adam_ctr"])
+        m.adam_step = snap["adam_step"]
+        self.step_base.copy_(snap["step_base"])
+        self.mb_ctr.copy_(snap["mb_ctr"])
+        self.env._obs.copy_(snap["obs"])
+        st = self.env.st.to_dict()
+        for k, v in snap["st"].items():
+            st[k].copy_(v)
+
+    def _capture_graphs(self) -> None:
+        """Capture rollout+GAE, minibatch fwd/bwd, and optimizer as three
+        hipGraphs.  The gradient all-reduce (world_size > 1) stays eager
+        between the mb and opt replays.  Device counters make every replay
+        advance RNG / Adam step correctly.  The warmup pass initializes the
+        RCCL communicator and lazy workspaces; training state is snapshotted
+        and restored around it so graph mode is bit-equivalent to eager."""
+        snap = self._snapshot()
+        self._rollout_body()
+        self._gae_body()
+        self.mb_ctr.zero_()
+        self._mb_body()
+        self._allreduce_grads()
+        self._opt_body()
+        torch.cuda.synchronize()
+        self.g_rollout = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_rollout):
+            self._rollout_body()
+            self._gae_body()
+        self.g_mb = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_mb):
+            self._mb_body()
+        self.g_opt = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_opt):
+            self._opt_body()
+        torch.cuda.synchronize()
+        self._restore(snap)
+        torch.cuda.synchronize()
+        self._graphs_ready = True
+
+    # ------------------------------------------------------------------
+    def collect_rollout(self) -> None:
+        if self.use_graphs and self._graphs_ready:
+            self.g_rollout.replay()
+        else:
+            self._rollout_body()
+            self._gae_body()
+        self.global_step += self.T
+
+    def compute_advantages(self) -> None:
+        # folded into collect_rollout (one graph); kept for API compat
+        pass
+
+    def update(self, with_stats: bool = True) -> Dict[str, float]:
+        cfg = self.cfg
+        self.losses.zero_()
+        self.mb_ctr.zero_()
+        graphs = self.use_graphs and self._graphs_ready
         n_mb = cfg.ppo_epochs * cfg.minibatches
+        for i in range(n_mb):
+            if not cfg.shuffle_rows:
+                self._mb_body_noshuffle(i)
+            elif graphs:
+                self.g_mb.replay()
+            else:
+                self._mb_body()
+            self._allreduce_grads()
+            if graphs:
+                self.g_opt.replay()
+            else:
+                self._opt_body()
+        self.update_count += 1
+        if not with_stats:
+            return {}
         lv = (self.losses / n_mb).cpu()
         return {
             "pi_loss": float(lv[0]),
@@ -206,11 +330,12 @@ class PPOTrainer:
         dist.all_reduce(self.model.grads, op=dist.ReduceOp.SUM, group=self.pg)
         self.model.grads.mul_(1.0 / self.world_size)
 
-    def train_update(self) -> Dict[str, float]:
+    def train_update(self, with_stats: bool = True) -> Dict[str, float]:
         """One full PPO update (rollout + GAE + epochs)."""
+        if self.use_graphs and not self._graphs_ready:
+            self._capture_graphs()
         self.collect_rollout()
-        self.compute_advantages()
-        return self.update()
+        return self.update(with_stats)
 
 
 # ---------------------------------------------------------------------------

@@ -73,7 +73,9 @@ class ActorCriticMLP:
         self.v = torch.zeros_like(self.params)
         # bf16 mirrors of the weight matrices (GEMM operands)
         self.params_bf16 = self.params.to(torch.bfloat16)
-        self.adam_step = 0
+        self.adam_step = 0  # host mirror of adam_ctr (logging/checkpoint)
+        # device step counter: Adam bias correction inside captured hipGraphs
+        self.adam_ctr = torch.zeros((), dtype=torch.int32, device=device)
         # grad-clip workspace
         self._clip_part = torch.zeros(256, dtype=torch.float32, device=device)
         self._clip_scale = torch.ones(1, dtype=torch.float32, device=device)
@@ -169,9 +171,12 @@ class ActorCriticMLP:
             api.grad_clip_scale(self.grads, max_grad_norm, self._clip_part,
                                 self._clip_scale)
             gscale = self._clip_scale
+        # bias correction reads the DEVICE counter (graph-replayable); the
+        # host step is a fallback for paths that pass step_ctr=None.
         api.adam(self.params, self.grads, self.m, self.v, self.params_bf16,
                  lr=lr, beta1=beta1, beta2=beta2, eps=eps, step=self.adam_step,
-                 gscale=gscale)
+                 gscale=gscale, step_ctr=self.adam_ctr)
+        api.increment_counter(self.adam_ctr, 1)
 
     def zero_grad(self) -> None:
         self.grads.zero_()
@@ -193,4 +198,5 @@ class ActorCriticMLP:
         self.m.copy_(sd["m"].to(self.device))
         self.v.copy_(sd["v"].to(self.device))
         self.adam_step = int(sd["adam_step"])
+        self.adam_ctr.fill_(self.adam_step)
         self.params_bf16.copy_(self.params.to(torch.bfloat16))

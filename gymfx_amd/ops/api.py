@@ -124,10 +124,17 @@ def adam(
     eps: float = 1e-8,
     step: int = 1,
     gscale: Optional[torch.Tensor] = None,
+    step_ctr: Optional[torch.Tensor] = None,
 ) -> None:
+    """step_ctr (device int32 scalar): when given, bias correction uses
+    ``*step_ctr + 1`` instead of the host ``step`` (hipGraph-replayable; pair
+    with ``increment_counter(step_ctr, 1)`` after the call)."""
     if _use_native(p):
-        native.require().adam(p, g, m, v, p_bf16, lr, beta1, beta2, eps, step, gscale)
+        native.require().adam(p, g, m, v, p_bf16, lr, beta1, beta2, eps, step,
+                              gscale, step_ctr)
         return
+    if step_ctr is not None:
+        step = int(step_ctr.item()) + 1
     s = float(gscale.item()) if gscale is not None else 1.0
     geff = g * s
     m.mul_(beta1).add_(geff, alpha=1 - beta1)
@@ -160,11 +167,18 @@ def sample_head(
     value: Optional[torch.Tensor] = None,
     entropy: Optional[torch.Tensor] = None,
     greedy: bool = False,
+    step_base: Optional[torch.Tensor] = None,
 ) -> None:
-    """head [M, A+1] f32 (logits | value) -> categorical sample + logp."""
+    """head [M, A+1] f32 (logits | value) -> categorical sample + logp.
+
+    step_base (device u64/i64 scalar): added to ``step`` inside the kernel so
+    a captured hipGraph replays with fresh randomness."""
     if _use_native(head):
-        native.require().sample_head(head, seed, step, actions, logp, value, entropy, greedy)
+        native.require().sample_head(head, seed, step, actions, logp, value,
+                                     entropy, greedy, step_base)
         return
+    if step_base is not None:
+        step = step + int(step_base.item())
     M, W = head.shape
     A = W - 1
     logits = head[:, :A]
@@ -269,4 +283,104 @@ def f32_to_bf16(src: torch.Tensor, dst: torch.Tensor) -> None:
     if _use_native(src):
         native.require().f32_to_bf16(src, dst)
         return
-    dst.copy_(src.to(torch.bfloat16))
+    dst.view(-1).copy_(src.reshape(-1).to(torch.bfloat16))
+
+
+# ---------------------------------------------------------------------------
+# Feistel epoch permutation (stateless bijection) + fused minibatch gather.
+# ---------------------------------------------------------------------------
+
+_U64 = (1 << 64) - 1
+
+
+def _splitmix64_int(x: int) -> int:
+    x = (x + 0x9E3779B97F4A7C15) & _U64
+    x = ((x ^ (x >> 30)) * 0xBF58476D1CE4E5B9) & _U64
+    x = ((x ^ (x >> 27)) * 0x94D049BB133111EB) & _U64
+    return x ^ (x >> 31)
+
+
+def feistel_key(seed: int, step_base: int, epoch: int) -> int:
+    """Key schedule — must match mb_gather_kernel exactly."""
+    return _splitmix64_int(
+        (seed ^ (step_base * 0x9E3779B97F4A7C15) ^ (epoch << 32)) & _U64
+    )
+
+
+def feistel_perm(n: int, key: int) -> torch.Tensor:
+    """dst -> src index map: 4-round balanced Feistel over the smallest
+    even-bit power-of-two domain >= n, cycle-walked into [0, n).  CPU oracle
+    of feistel_perm_idx in ops/csrc/ppo_kernels.hip (vectorized numpy)."""
+    import numpy as np
+
+    bits = 2
+    while (1 << bits) < n:
+        bits += 2
+    half = bits // 2
+    mask = np.uint64((1 << half) - 1)
+    x = np.arange(n, dtype=np.uint64)
+    active = np.ones(n, dtype=bool)
+    while active.any():
+        xa = x[active]
+        a = xa & mask
+        b = xa >> np.uint64(half)
+        for r in range(4):
+            h = (np.uint64(key) ^ np.uint64(r << 48) ^ b).astype(np.uint64)
+            # splitmix64 vectorized (uint64 wraparound is numpy semantics)
+            with np.errstate(over="ignore"):
+                h = h + np.uint64(0x9E3779B97F4A7C15)
+                h = (h ^ (h >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+                h = (h ^ (h >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+                h = h ^ (h >> np.uint64(31))
+            f = h & mask
+            a, b = b, a ^ f
+        xa = (b << np.uint64(half)) | a
+        x[active] = xa
+        active = x >= n
+    return torch.from_numpy(x.astype(np.int64))
+
+
+def mb_gather(
+    obs_src: torch.Tensor,
+    act_src: torch.Tensor,
+    logp_src: torch.Tensor,
+    adv_src: torch.Tensor,
+    ret_src: torch.Tensor,
+    obs_mb: torch.Tensor,
+    act_mb: torch.Tensor,
+    logp_mb: torch.Tensor,
+    adv_mb: torch.Tensor,
+    ret_mb: torch.Tensor,
+    *,
+    seed: int,
+    minibatches: int,
+    step_base: torch.Tensor,
+    mb_ctr: torch.Tensor,
+) -> None:
+    """Gather minibatch ``mb_ctr % minibatches`` of epoch
+    ``mb_ctr // minibatches`` under the Feistel permutation keyed by
+    (seed, *step_base, epoch).  One fused kernel on GPU."""
+    if _use_native(obs_src):
+        native.require().mb_gather(
+            obs_src, act_src, logp_src, adv_src, ret_src, obs_mb, act_mb,
+            logp_mb, adv_mb, ret_mb, seed, minibatches, step_base, mb_ctr,
+        )
+        return
+    n = obs_src.shape[0]
+    M = obs_mb.shape[0]
+    ctr = int(mb_ctr.item())
+    epoch, mb = ctr // minibatches, ctr % minibatches
+    key = feistel_key(seed, int(step_base.item()), epoch)
+    src = feistel_perm(n, key)[mb * M:(mb + 1) * M]
+    obs_mb.copy_(obs_src[src])
+    act_mb.copy_(act_src[src])
+    logp_mb.copy_(logp_src[src])
+    adv_mb.copy_(adv_src[src])
+    ret_mb.copy_(ret_src[src])
+
+
+def increment_counter(ctr: torch.Tensor, delta: int) -> None:
+    if _use_native(ctr):
+        native.require().increment_counter(ctr, delta)
+        return
+    ctr.add_(delta)

@@ -27,13 +27,25 @@ void launch_gae(const float* rewards, const float* values, const bool* dones,
                 hipStream_t stream);
 void launch_adam(float* p, const float* g, float* m, float* v, void* p_bf16,
                  int64_t n, float lr, float beta1, float beta2, float eps,
-                 float bc1, float bc2, const float* gscale, hipStream_t stream);
+                 float bc1, float bc2, const float* gscale, const int* step_ctr,
+                 hipStream_t stream);
 void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
                       float* scale, int nparts, hipStream_t stream);
 void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,
                         uint64_t step, int64_t* actions, float* logp,
                         float* value, float* entropy, int greedy,
+                        const unsigned long long* step_base,
                         hipStream_t stream);
+void launch_increment_u64(unsigned long long* ctr, unsigned long long delta,
+                          hipStream_t stream);
+void launch_increment_i32(int* ctr, int delta, hipStream_t stream);
+void launch_mb_gather(const void* obs_src, const int64_t* act_src,
+                      const float* logp_src, const float* adv_src,
+                      const float* ret_src, void* obs_mb, int64_t* act_mb,
+                      float* logp_mb, float* adv_mb, float* ret_mb, int M,
+                      int D, uint32_t n, int half, uint64_t seed,
+                      int minibatches, const unsigned long long* step_base,
+                      const unsigned long long* mb_ctr, hipStream_t stream);
 void launch_ppo_loss_bwd(const float* head, const int64_t* actions,
                          const float* old_logp, const float* adv,
                          const float* ret, void* dhead, int M, int n_actions,
@@ -335,7 +347,8 @@ void gae_op(torch::Tensor rewards, torch::Tensor values, torch::Tensor dones,
 void adam_op(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
              c10::optional<torch::Tensor> p_bf16, double lr, double beta1,
              double beta2, double eps, int64_t step,
-             c10::optional<torch::Tensor> gscale) {
+             c10::optional<torch::Tensor> gscale,
+             c10::optional<torch::Tensor> step_ctr) {
   check_f32(p, "p");
   check_f32(g, "g");
   check_f32(m, "m");
@@ -349,11 +362,16 @@ void adam_op(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
   }
   const float* gs = nullptr;
   if (gscale.has_value()) gs = gscale->data_ptr<float>();
+  const int* sc = nullptr;
+  if (step_ctr.has_value()) {
+    TORCH_CHECK(step_ctr->scalar_type() == torch::kInt32, "step_ctr int32");
+    sc = step_ctr->data_ptr<int>();
+  }
   const float bc1 = 1.f - powf((float)beta1, (float)step);
   const float bc2 = 1.f - powf((float)beta2, (float)step);
   gymfx::launch_adam(p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(), pb, n, (float)lr,
-                     (float)beta1, (float)beta2, (float)eps, bc1, bc2, gs,
+                     (float)beta1, (float)beta2, (float)eps, bc1, bc2, gs, sc,
                      cur_stream());
 }
 
@@ -370,17 +388,74 @@ void grad_clip_op(torch::Tensor g, double max_norm, torch::Tensor part,
 void sample_head_op(torch::Tensor head, int64_t seed, int64_t step,
                     torch::Tensor actions, torch::Tensor logp,
                     c10::optional<torch::Tensor> value,
-                    c10::optional<torch::Tensor> entropy, bool greedy) {
+                    c10::optional<torch::Tensor> entropy, bool greedy,
+                    c10::optional<torch::Tensor> step_base) {
   check_f32(head, "head");
   const int M = (int)head.size(0);
   const int n_actions = (int)head.size(1) - 1;
   TORCH_CHECK(actions.scalar_type() == torch::kInt64);
+  const unsigned long long* sb = nullptr;
+  if (step_base.has_value()) {
+    TORCH_CHECK(step_base->scalar_type() == torch::kUInt64 ||
+                    step_base->scalar_type() == torch::kInt64,
+                "step_base must be a 64-bit counter tensor");
+    sb = reinterpret_cast<const unsigned long long*>(step_base->data_ptr());
+  }
   gymfx::launch_sample_head(
       head.data_ptr<float>(), M, n_actions, (uint64_t)seed, (uint64_t)step,
       actions.data_ptr<int64_t>(), logp.data_ptr<float>(),
       value.has_value() ? value->data_ptr<float>() : nullptr,
       entropy.has_value() ? entropy->data_ptr<float>() : nullptr,
-      greedy ? 1 : 0, cur_stream());
+      greedy ? 1 : 0, sb, cur_stream());
+}
+
+void increment_counter_op(torch::Tensor ctr, int64_t delta) {
+  TORCH_CHECK(ctr.numel() == 1, "counter must be scalar");
+  if (ctr.scalar_type() == torch::kInt32) {
+    gymfx::launch_increment_i32(ctr.data_ptr<int>(), (int)delta, cur_stream());
+  } else {
+    TORCH_CHECK(ctr.scalar_type() == torch::kUInt64 ||
+                    ctr.scalar_type() == torch::kInt64,
+                "counter must be i32/i64/u64");
+    gymfx::launch_increment_u64(
+        reinterpret_cast<unsigned long long*>(ctr.data_ptr()),
+        (unsigned long long)delta, cur_stream());
+  }
+}
+
+void mb_gather_op(torch::Tensor obs_src, torch::Tensor act_src,
+                  torch::Tensor logp_src, torch::Tensor adv_src,
+                  torch::Tensor ret_src, torch::Tensor obs_mb,
+                  torch::Tensor act_mb, torch::Tensor logp_mb,
+                  torch::Tensor adv_mb, torch::Tensor ret_mb, int64_t seed,
+                  int64_t minibatches, torch::Tensor step_base,
+                  torch::Tensor mb_ctr) {
+  check_bf16(obs_src, "obs_src");
+  check_bf16(obs_mb, "obs_mb");
+  const int64_t n = obs_src.size(0);
+  const int D = (int)obs_src.size(1);
+  const int M = (int)obs_mb.size(0);
+  TORCH_CHECK(obs_mb.size(1) == D, "obs_mb D mismatch");
+  TORCH_CHECK(act_src.numel() == n && logp_src.numel() == n &&
+                  adv_src.numel() == n && ret_src.numel() == n,
+              "src numel mismatch");
+  TORCH_CHECK(act_mb.numel() == M && logp_mb.numel() == M &&
+                  adv_mb.numel() == M && ret_mb.numel() == M,
+              "mb numel mismatch");
+  TORCH_CHECK((int64_t)M * minibatches == n, "M * minibatches != n");
+  // smallest even-bit domain 2^(2*half) >= n
+  int bits = 2;
+  while ((1ll << bits) < n) bits += 2;
+  gymfx::launch_mb_gather(
+      obs_src.data_ptr(), act_src.data_ptr<int64_t>(),
+      logp_src.data_ptr<float>(), adv_src.data_ptr<float>(),
+      ret_src.data_ptr<float>(), obs_mb.data_ptr(),
+      act_mb.data_ptr<int64_t>(), logp_mb.data_ptr<float>(),
+      adv_mb.data_ptr<float>(), ret_mb.data_ptr<float>(), M, D, (uint32_t)n,
+      bits / 2, (uint64_t)seed, (int)minibatches,
+      reinterpret_cast<const unsigned long long*>(step_base.data_ptr()),
+      reinterpret_cast<const unsigned long long*>(mb_ctr.data_ptr()),
+      cur_stream());
 }
 
 void ppo_loss_bwd_op(torch::Tensor head, torch::Tensor actions,
@@ -430,12 +505,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam", &adam_op, py::arg("p"), py::arg("g"), py::arg("m"),
         py::arg("v"), py::arg("p_bf16"), py::arg("lr"), py::arg("beta1"),
         py::arg("beta2"), py::arg("eps"), py::arg("step"),
-        py::arg("gscale") = py::none());
+        py::arg("gscale") = py::none(), py::arg("step_ctr") = py::none());
   m.def("grad_clip", &grad_clip_op);
   m.def("sample_head", &sample_head_op, py::arg("head"), py::arg("seed"),
         py::arg("step"), py::arg("actions"), py::arg("logp"),
         py::arg("value") = py::none(), py::arg("entropy") = py::none(),
-        py::arg("greedy") = false);
+        py::arg("greedy") = false, py::arg("step_base") = py::none());
+  m.def("increment_counter", &increment_counter_op, py::arg("ctr"),
+        py::arg("delta"));
+  m.def("mb_gather", &mb_gather_op, py::arg("obs_src"), py::arg("act_src"),
+        py::arg("logp_src"), py::arg("adv_src"), py::arg("ret_src"),
+        py::arg("obs_mb"), py::arg("act_mb"), py::arg("logp_mb"),
+        py::arg("adv_mb"), py::arg("ret_mb"), py::arg("seed"),
+        py::arg("minibatches"), py::arg("step_base"), py::arg("mb_ctr"));
   m.def("ppo_loss_bwd", &ppo_loss_bwd_op, py::arg("head"), py::arg("actions"),
         py::arg("old_logp"), py::arg("adv"), py::arg("ret"), py::arg("dhead"),
         py::arg("clip_eps"), py::arg("ent_coef"), py::arg("vf_coef"),

@@ -156,6 +156,10 @@ __global__ __launch_bounds__(256) void gemm_kernel(
 // logical A = X^T [K, M] (i.e. A[k][m] = X[m][k]), logical B = dY [M, N].
 // Deterministic: fixed slab count + ordered tree reduce (no atomics).
 // Partials: f32 [S, K, N]; db partials f32 [S, N] (bias grad = colsum dY).
+// Tiles are staged in NATURAL [m][k] / [m][n] layout with fully-coalesced
+// bf16x8 global loads; the transpose happens at the MFMA fragment reads from
+// LDS (scalar LDS reads are cheap; strided 2-byte GLOBAL reads were the
+// bottleneck of the first version of this kernel).
 // ---------------------------------------------------------------------------
 template <bool WANT_DB>
 __global__ __launch_bounds__(256) void wgrad_partial_kernel(
@@ -164,8 +168,8 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
     int M, int N, int K, int slabs) {
   constexpr int BM = 64, BN = 64, BK = 32;  // BM: K-dim tile, BN: N tile, BK: M chunk step
   // naming: output tile is [BM of K] x [BN of N]; reduction dim is M.
-  __shared__ __bf16 Xs[BM][BK + 8];   // X^T tile: [k][m]
-  __shared__ __bf16 Ys[BN][BK + 8];   // dY^T tile: [n][m]
+  __shared__ __bf16 Xs[BK][BM + 8];   // X tile, natural layout: [m][k]
+  __shared__ __bf16 Ys[BK][BN + 8];   // dY tile, natural layout: [m][n]
 
   const int ktile = blockIdx.x;
   const int ntile = blockIdx.y;
@@ -183,51 +187,59 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   f32x4 acc[2][2] = {};
   float db_acc = 0.f;  // per-thread partial for db (cols handled below)
 
+  // coalesced staging map: 256 threads x bf16x8 = 2048 elems = 32 rows x 64 cols
+  const int s_row = tid >> 3;        // 0..31  (m within chunk)
+  const int s_c8 = (tid & 7) * 8;    // 0,8,..,56 (k or n within tile)
+
   for (int m0 = m_begin; m0 < m_end; m0 += BK) {
-    // stage X^T tile: Xs[k][m] = X[m0+m][bk+k]
+    // stage X tile [32 m][64 k]: contiguous bf16x8 row segments
     {
-      const int k = tid >> 2;            // 0..63
-      const int m8 = (tid & 3) * 8;
       bf16x8 v = {};
-      const int gk = bk + k;
-      if (gk < K) {
-        for (int i = 0; i < 8; ++i) {
-          const int gm = m0 + m8 + i;
-          v[i] = (gm < m_end) ? X[(int64_t)gm * K + gk] : (__bf16)0.f;
+      const int gm = m0 + s_row;
+      if (gm < m_end) {
+        const int gk = bk + s_c8;
+        if (gk + 8 <= K) {
+          v = *reinterpret_cast<const bf16x8*>(&X[(int64_t)gm * K + gk]);
+        } else {
+          for (int i = 0; i < 8; ++i)
+            v[i] = (gk + i < K) ? X[(int64_t)gm * K + gk + i] : (__bf16)0.f;
         }
       }
-      *reinterpret_cast<bf16x8*>(&Xs[k][m8]) = v;
+      *reinterpret_cast<bf16x8*>(&Xs[s_row][s_c8]) = v;
     }
-    // stage dY^T tile: Ys[n][m] = dY[m0+m][bn+n]
+    // stage dY tile [32 m][64 n]
     {
-      const int n = tid >> 2;
-      const int m8 = (tid & 3) * 8;
       bf16x8 v = {};
-      const int gn = bn + n;
-      if (gn < N) {
-        for (int i = 0; i < 8; ++i) {
-          const int gm = m0 + m8 + i;
-          v[i] = (gm < m_end) ? dY[(int64_t)gm * N + gn] : (__bf16)0.f;
+      const int gm = m0 + s_row;
+      if (gm < m_end) {
+        const int gn = bn + s_c8;
+        if (gn + 8 <= N) {
+          v = *reinterpret_cast<const bf16x8*>(&dY[(int64_t)gm * N + gn]);
+        } else {
+          for (int i = 0; i < 8; ++i)
+            v[i] = (gn + i < N) ? dY[(int64_t)gm * N + gn + i] : (__bf16)0.f;
         }
       }
-      *reinterpret_cast<bf16x8*>(&Ys[n][m8]) = v;
+      *reinterpret_cast<bf16x8*>(&Ys[s_row][s_c8]) = v;
     }
     __syncthreads();
     if (WANT_DB && tid < BN) {
       float s = 0.f;
-      for (int i = 0; i < BK && m0 + i < m_end; ++i) s += bf2f(Ys[tid][i]);
+      for (int i = 0; i < BK && m0 + i < m_end; ++i) s += bf2f(Ys[i][tid]);
       db_acc += s;
     }
 
     const int row_a = lane & 15;
     const int kseg = lane >> 4;
+    // fragment reads do the transpose: logical A[kout][m] = X[m][kout],
+    // logical B[m][n] = dY[m][n]; reduction index m = kseg*8 + i.
     bf16x8 af[2], bf_[2];
     for (int mi = 0; mi < 2; ++mi)
-      af[mi] = *reinterpret_cast<const bf16x8*>(
-          &Xs[wr * 32 + mi * 16 + row_a][kseg * 8]);
+      for (int i = 0; i < 8; ++i)
+        af[mi][i] = Xs[kseg * 8 + i][wr * 32 + mi * 16 + row_a];
     for (int ni = 0; ni < 2; ++ni)
-      bf_[ni] = *reinterpret_cast<const bf16x8*>(
-          &Ys[wc * 32 + ni * 16 + row_a][kseg * 8]);
+      for (int i = 0; i < 8; ++i)
+        bf_[ni][i] = Ys[kseg * 8 + i][wc * 32 + ni * 16 + row_a];
     for (int mi = 0; mi < 2; ++mi)
       for (int ni = 0; ni < 2; ++ni)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -305,7 +317,15 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             __bf16* __restrict__ p_bf16, int64_t n, float lr,
                             float beta1, float beta2, float eps, float bc1,
-                            float bc2, const float* __restrict__ gscale) {
+                            float bc2, const float* __restrict__ gscale,
+                            const int* __restrict__ step_ctr) {
+  if (step_ctr) {
+    // device step counter (hipGraph-replayable): effective step = *ctr + 1;
+    // the companion increment kernel advances it after this launch.
+    const float st = (float)(*step_ctr + 1);
+    bc1 = 1.f - powf(beta1, st);
+    bc2 = 1.f - powf(beta2, st);
+  }
   const float s = gscale ? *gscale : 1.f;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -367,9 +387,11 @@ __global__ void sample_head_kernel(const float* __restrict__ head, int M,
                                    float* __restrict__ logp,
                                    float* __restrict__ value,
                                    float* __restrict__ entropy,
-                                   int greedy) {
+                                   int greedy,
+                                   const unsigned long long* __restrict__ step_base) {
   const int m = blockIdx.x * blockDim.x + threadIdx.x;
   if (m >= M) return;
+  if (step_base) step += *step_base;  // device counter: hipGraph-replayable RNG
   const float* row = head + (int64_t)m * (n_actions + 1);
   float mx = row[0];
   for (int j = 1; j < n_actions; ++j) mx = fmaxf(mx, row[j]);
@@ -522,9 +544,125 @@ __global__ void adv_norm_apply_kernel(float* __restrict__ adv, int64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// Device counters (hipGraph support): tiny kernels that advance the RNG /
+// optimizer-step counters inside a captured graph, so one captured update
+// replays with fresh randomness and correct Adam bias correction.
+// ---------------------------------------------------------------------------
+__global__ void increment_u64_kernel(unsigned long long* __restrict__ ctr,
+                                     unsigned long long delta) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *ctr += delta;
+}
+
+__global__ void increment_i32_kernel(int* __restrict__ ctr, int delta) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *ctr += delta;
+}
+
+// ---------------------------------------------------------------------------
+// Minibatch shuffle + gather, fused.
+//
+// The epoch permutation is a 4-round balanced Feistel network over the
+// smallest even-bit power-of-two domain >= n, cycle-walked back into [0, n):
+// a stateless bijection keyed by (seed, rollout counter, epoch).  This
+// replaces host randperm + five giant index-gathers (and is exactly
+// reproducible on the CPU oracle: ops/api.feistel_perm).
+// ---------------------------------------------------------------------------
+GFX_DEV uint32_t feistel_encrypt_once(uint32_t x, int half, uint32_t mask,
+                                      uint64_t key) {
+  uint32_t a = x & mask;        // low half
+  uint32_t b = x >> half;       // high half
+  for (int r = 0; r < 4; ++r) {
+    const uint32_t f =
+        (uint32_t)(splitmix64(key ^ ((uint64_t)r << 48) ^ (uint64_t)b)) & mask;
+    const uint32_t t = b;
+    b = a ^ f;
+    a = t;
+  }
+  return (b << half) | a;
+}
+
+GFX_DEV uint32_t feistel_perm_idx(uint32_t i, uint32_t n, int half,
+                                  uint64_t key) {
+  const uint32_t mask = (1u << half) - 1u;
+  uint32_t x = i;
+  do {
+    x = feistel_encrypt_once(x, half, mask, key);
+  } while (x >= n);  // cycle-walk: terminates (bijection over 2^(2*half))
+  return x;
+}
+
+// One wavefront copies one minibatch row. grid.x = ceil(M/4), block = 256.
+// mb_ctr (device) = epoch * minibatches + mb, advanced by increment after
+// the gather inside the captured graph; step_base keys the epoch permutation
+// to the rollout counter so every update reshuffles.
+__global__ __launch_bounds__(256) void mb_gather_kernel(
+    const __bf16* __restrict__ obs_src, const int64_t* __restrict__ act_src,
+    const float* __restrict__ logp_src, const float* __restrict__ adv_src,
+    const float* __restrict__ ret_src, __bf16* __restrict__ obs_mb,
+    int64_t* __restrict__ act_mb, float* __restrict__ logp_mb,
+    float* __restrict__ adv_mb, float* __restrict__ ret_mb, int M, int D,
+    uint32_t n, int half, uint64_t seed, int minibatches,
+    const unsigned long long* __restrict__ step_base,
+    const unsigned long long* __restrict__ mb_ctr) {
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (row >= M) return;
+  const unsigned long long ctr = *mb_ctr;
+  const uint32_t epoch = (uint32_t)(ctr / (unsigned)minibatches);
+  const uint32_t mb = (uint32_t)(ctr % (unsigned)minibatches);
+  const uint64_t key = splitmix64(seed ^ (*step_base * 0x9E3779B97F4A7C15ull) ^
+                                  ((uint64_t)epoch << 32));
+  const uint32_t dst = mb * (uint32_t)M + (uint32_t)row;
+  const uint32_t src = feistel_perm_idx(dst, n, half, key);
+
+  const __bf16* src_row = obs_src + (int64_t)src * D;
+  __bf16* dst_row = obs_mb + (int64_t)row * D;
+  if ((D & 3) == 0) {
+    // 8-byte chunks (bf16 x4)
+    const int chunks = D >> 2;
+    const uint64_t* s64 = reinterpret_cast<const uint64_t*>(src_row);
+    uint64_t* d64 = reinterpret_cast<uint64_t*>(dst_row);
+    for (int c = lane; c < chunks; c += 64) d64[c] = s64[c];
+  } else {
+    for (int c = lane; c < D; c += 64) dst_row[c] = src_row[c];
+  }
+  if (lane == 0) {
+    act_mb[row] = act_src[src];
+    logp_mb[row] = logp_src[src];
+    adv_mb[row] = adv_src[src];
+    ret_mb[row] = ret_src[src];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
 static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
+
+void launch_increment_u64(unsigned long long* ctr, unsigned long long delta,
+                          hipStream_t stream) {
+  hipLaunchKernelGGL(increment_u64_kernel, dim3(1), dim3(1), 0, stream, ctr,
+                     delta);
+}
+
+void launch_increment_i32(int* ctr, int delta, hipStream_t stream) {
+  hipLaunchKernelGGL(increment_i32_kernel, dim3(1), dim3(1), 0, stream, ctr,
+                     delta);
+}
+
+void launch_mb_gather(const void* obs_src, const int64_t* act_src,
+                      const float* logp_src, const float* adv_src,
+                      const float* ret_src, void* obs_mb, int64_t* act_mb,
+                      float* logp_mb, float* adv_mb, float* ret_mb, int M,
+                      int D, uint32_t n, int half, uint64_t seed,
+                      int minibatches, const unsigned long long* step_base,
+                      const unsigned long long* mb_ctr, hipStream_t stream) {
+  hipLaunchKernelGGL(mb_gather_kernel, dim3(ceil_div(M, 4)), dim3(256), 0,
+                     stream, reinterpret_cast<const __bf16*>(obs_src), act_src,
+                     logp_src, adv_src, ret_src,
+                     reinterpret_cast<__bf16*>(obs_mb), act_mb, logp_mb, adv_mb,
+                     ret_mb, M, D, n, half, seed, minibatches, step_base,
+                     mb_ctr);
+}
 
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
@@ -587,12 +725,12 @@ void launch_gae(const float* rewards, const float* values, const bool* dones,
 
 void launch_adam(float* p, const float* g, float* m, float* v, void* p_bf16,
                  int64_t n, float lr, float beta1, float beta2, float eps,
-                 float bc1, float bc2, const float* gscale,
+                 float bc1, float bc2, const float* gscale, const int* step_ctr,
                  hipStream_t stream) {
   int blocks = (int)std::min<int64_t>((n + 255) / 256, 2048);
   hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, stream, p, g, m,
                      v, reinterpret_cast<__bf16*>(p_bf16), n, lr, beta1, beta2,
-                     eps, bc1, bc2, gscale);
+                     eps, bc1, bc2, gscale, step_ctr);
 }
 
 void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
@@ -606,10 +744,11 @@ void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
 void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,
                         uint64_t step, int64_t* actions, float* logp,
                         float* value, float* entropy, int greedy,
+                        const unsigned long long* step_base,
                         hipStream_t stream) {
   hipLaunchKernelGGL(sample_head_kernel, dim3(ceil_div(M, 256)), dim3(256), 0,
                      stream, head, M, n_actions, seed, step, actions, logp,
-                     value, entropy, greedy);
+                     value, entropy, greedy, step_base);
 }
 
 void launch_ppo_loss_bwd(const float* head, const int64_t* actions,

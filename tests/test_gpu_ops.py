@@ -262,3 +262,79 @@ def test_model_fwd_bwd_matches_torch_oracle():
         scale = max(np.abs(cc).max(), 1e-6)
         np.testing.assert_allclose(gg / scale, cc / scale, atol=5e-2,
                                    err_msg=f"grad {name}")
+
+
+# ---------------------------------------------------------------------------
+# Feistel minibatch gather + hipGraph device counters
+# ---------------------------------------------------------------------------
+def test_mb_gather_matches_cpu_oracle():
+    from gymfx_amd.ops import api
+
+    ext = _ext()
+    n, mbs, D = 96 * 11, 8, 260  # non-pow2 n exercises cycle-walking
+    M = n // mbs
+    g = torch.Generator().manual_seed(3)
+    obs_cpu = torch.randn(n, D, generator=g).to(torch.bfloat16)
+    act_cpu = torch.randint(0, 3, (n,), generator=g)
+    logp_cpu = torch.randn(n, generator=g)
+    adv_cpu = torch.randn(n, generator=g)
+    ret_cpu = torch.randn(n, generator=g)
+    srcs = [t.cuda() for t in (obs_cpu, act_cpu, logp_cpu, adv_cpu, ret_cpu)]
+    outs_g = [torch.empty(M, D, dtype=torch.bfloat16).cuda(),
+              torch.empty(M, dtype=torch.int64).cuda(),
+              torch.empty(M).cuda(), torch.empty(M).cuda(),
+              torch.empty(M).cuda()]
+    outs_c = [torch.empty(M, D, dtype=torch.bfloat16),
+              torch.empty(M, dtype=torch.int64),
+              torch.empty(M), torch.empty(M), torch.empty(M)]
+    for ctr_val, sb_val in [(0, 0), (5, 128), (17, 999)]:
+        sb_g = torch.tensor(sb_val, dtype=torch.int64).cuda()
+        ctr_g = torch.tensor(ctr_val, dtype=torch.int64).cuda()
+        ext.mb_gather(*srcs, *outs_g, 42, mbs, sb_g, ctr_g)
+        api.mb_gather(obs_cpu, act_cpu, logp_cpu, adv_cpu, ret_cpu, *outs_c,
+                      seed=42, minibatches=mbs,
+                      step_base=torch.tensor(sb_val, dtype=torch.int64),
+                      mb_ctr=torch.tensor(ctr_val, dtype=torch.int64))
+        for gpu_t, cpu_t in zip(outs_g, outs_c):
+            assert torch.equal(gpu_t.cpu(), cpu_t)
+
+
+def test_sample_head_step_base_counter():
+    ext = _ext()
+    M = 512
+    g = torch.Generator().manual_seed(1)
+    head = torch.randn(M, 4, generator=g).cuda()
+    a1 = torch.empty(M, dtype=torch.int64).cuda()
+    a2 = torch.empty(M, dtype=torch.int64).cuda()
+    lp = torch.empty(M).cuda()
+    # step_base + step must equal host-summed step
+    sb = torch.tensor(100, dtype=torch.int64).cuda()
+    ext.sample_head(head, 7, 3, a1, lp, None, None, False, sb)
+    ext.sample_head(head, 7, 103, a2, lp, None, None, False, None)
+    assert torch.equal(a1, a2)
+    # increment advances the counter on-device
+    ext.increment_counter(sb, 28)
+    ext.sample_head(head, 7, 3, a1, lp, None, None, False, sb)
+    ext.sample_head(head, 7, 131, a2, lp, None, None, False, None)
+    assert torch.equal(a1, a2)
+
+
+def test_adam_device_step_ctr_matches_host_step():
+    ext = _ext()
+    n = 1000
+    g = torch.Generator().manual_seed(2)
+
+    def init():
+        p = torch.randn(n, generator=g).cuda()
+        return p, torch.randn(n, generator=g).cuda(), torch.zeros(n).cuda(), torch.zeros(n).cuda()
+
+    p1, g1, m1, v1 = init()
+    g.manual_seed(2)
+    p2, g2, m2, v2 = init()
+    ctr = torch.zeros((), dtype=torch.int32).cuda()
+    for step in range(1, 4):
+        ext.adam(p1, g1, m1, v1, None, 1e-3, 0.9, 0.999, 1e-8, step, None, None)
+        ext.adam(p2, g2, m2, v2, None, 1e-3, 0.9, 0.999, 1e-8, 999, None, ctr)
+        ext.increment_counter(ctr, 1)
+    assert torch.equal(p1, p2)
+    assert int(ctr.item()) == 3
