@@ -336,5 +336,6 @@ def test_adam_device_step_ctr_matches_host_step():
         ext.adam(p1, g1, m1, v1, None, 1e-3, 0.9, 0.999, 1e-8, step, None, None)
         ext.adam(p2, g2, m2, v2, None, 1e-3, 0.9, 0.999, 1e-8, 999, None, ctr)
         ext.increment_counter(ctr, 1)
-    assert torch.equal(p1, p2)
+    # host vs device powf for the bias correction differ in the last ulp
+    assert torch.allclose(p1, p2, rtol=1e-6, atol=1e-7)
     assert int(ctr.item()) == 3
