@@ -149,20 +149,22 @@ class PPOTrainer:
     def _rollout_body(self) -> None:
         env, model = self.env, self.model
         T = self.T
+        api.f32_to_bf16(env._obs, self.obs_buf[0])
         for t in range(T):
-            obs_t = self.obs_buf[t]
-            api.f32_to_bf16(env._obs, obs_t)
-            head = model.forward(obs_t, self.acts_rollout)
+            head = model.forward(self.obs_buf[t], self.acts_rollout)
             api.sample_head(
                 head, self.sample_seed, t,
                 self.act_buf[t], self.logp_buf[t], self.val_buf[t],
                 step_base=self.step_base,
             )
-            out = env.step(self.act_buf[t])
-            self.rew_buf[t].copy_(out["reward"])
-            self.done_buf[t].copy_(out["terminated"])
+            # fused env kernel writes reward/done into the rollout slabs and
+            # the NEXT observation (bf16) into obs_buf[t+1] directly
+            nxt = self.obs_buf[t + 1] if t + 1 < T else self.obs_bf16_step
+            env.step(
+                self.act_buf[t], reward_out=self.rew_buf[t],
+                terminated_out=self.done_buf[t], obs_bf16_out=nxt,
+            )
         # bootstrap value
-        api.f32_to_bf16(env._obs, self.obs_bf16_step)
         head = model.forward(self.obs_bf16_step, self.acts_rollout)
         self.val_buf[T].copy_(head[:, -1])
         api.increment_counter(self.step_base, T)

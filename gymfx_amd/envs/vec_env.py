@@ -108,14 +108,22 @@ class VecFxEnv:
         else:
             build_obs_torch(self.st, self.mt, self.params, out=self._obs)
 
-    def step(self, actions: torch.Tensor) -> Dict[str, torch.Tensor]:
-        """Advance all envs. Returns dict with obs/reward/terminated/info tensors."""
+    def step(self, actions: torch.Tensor, *,
+             reward_out: Optional[torch.Tensor] = None,
+             terminated_out: Optional[torch.Tensor] = None,
+             obs_bf16_out: Optional[torch.Tensor] = None) -> Dict[str, torch.Tensor]:
+        """Advance all envs. Returns dict with obs/reward/terminated tensors.
+
+        reward_out / terminated_out / obs_bf16_out: optional preallocated
+        device tensors the fused kernels write directly (rollout slabs — no
+        copy kernels; used by the PPO trainer's hipGraph bodies)."""
         if not isinstance(actions, torch.Tensor):
             actions = torch.as_tensor(actions, device=self.device)
         actions = actions.to(self.device)
         if self._native is not None:
             # fused HIP path: step + autoreset + obs in two kernel launches
-            info = self._native.step(actions, self._obs)
+            info = self._native.step(actions, self._obs, reward_out,
+                                     terminated_out, obs_bf16_out)
             info["obs"] = self._obs
             return info
         info = step_torch(self.st, self.mt, self.params, actions)
@@ -127,6 +135,12 @@ class VecFxEnv:
             info["terminated"] = done
         self._build_obs()
         info["obs"] = self._obs
+        if reward_out is not None:
+            reward_out.copy_(info["reward"])
+        if terminated_out is not None:
+            terminated_out.copy_(info["terminated"])
+        if obs_bf16_out is not None:
+            obs_bf16_out.view(-1).copy_(self._obs.reshape(-1).to(torch.bfloat16))
         return info
 
     # ------------------------------------------------------------------

@@ -224,7 +224,10 @@ struct GymFxEngine {
     P.coerced_out = outputs["coerced"].data_ptr<int64_t>();
   }
 
-  py::dict step(torch::Tensor actions, torch::Tensor obs_out) {
+  py::dict step(torch::Tensor actions, torch::Tensor obs_out,
+                c10::optional<torch::Tensor> reward_out,
+                c10::optional<torch::Tensor> terminated_out,
+                c10::optional<torch::Tensor> obs_bf16_out) {
     TORCH_CHECK(actions.is_cuda() == state["cursor"].is_cuda(),
                 "actions must live on the env device");
     TORCH_CHECK(actions.numel() == K.n_envs, "actions numel != n_envs");
@@ -237,23 +240,52 @@ struct GymFxEngine {
                   "discrete mode wants int64 actions");
     }
     P.actions = actions.data_ptr();
+    // optional external outputs (rollout slabs): the kernels write straight
+    // into the trainer's [T, N] buffers — no copy kernels afterwards.
+    torch::Tensor rew = outputs["reward"];
+    if (reward_out.has_value()) {
+      TORCH_CHECK(reward_out->is_contiguous() &&
+                      reward_out->scalar_type() == torch::kFloat32 &&
+                      reward_out->numel() == K.n_envs,
+                  "reward_out must be contiguous f32 [n_envs]");
+      rew = *reward_out;
+    }
+    torch::Tensor term = outputs["terminated"];
+    if (terminated_out.has_value()) {
+      TORCH_CHECK(terminated_out->is_contiguous() &&
+                      terminated_out->scalar_type() == torch::kBool &&
+                      terminated_out->numel() == K.n_envs,
+                  "terminated_out must be contiguous bool [n_envs]");
+      term = *terminated_out;
+    }
+    P.reward_out = rew.data_ptr<float>();
+    P.terminated_out = term.data_ptr<bool>();
     hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
     launch_env_step(P, K, stream);
-    build_obs(obs_out);
+    build_obs(obs_out, obs_bf16_out);
     py::dict out;
-    out["reward"] = outputs["reward"];
+    out["reward"] = rew;
     out["base_reward"] = outputs["base_reward"];
     out["force_close_reward_penalty"] = outputs["penalty"];
-    out["terminated"] = outputs["terminated"];
+    out["terminated"] = term;
     out["coerced_action"] = outputs["coerced"];
     return out;
   }
 
-  void build_obs(torch::Tensor obs_out) {
+  void build_obs(torch::Tensor obs_out,
+                 c10::optional<torch::Tensor> obs_bf16_out) {
     TORCH_CHECK(obs_out.is_contiguous() && obs_out.scalar_type() == torch::kFloat32,
                 "obs_out must be contiguous f32");
     TORCH_CHECK(obs_out.numel() == (int64_t)K.n_envs * K.obs_dim, "obs_out shape");
     P.obs_out = obs_out.data_ptr<float>();
+    P.obs_bf16_out = nullptr;
+    if (obs_bf16_out.has_value()) {
+      TORCH_CHECK(obs_bf16_out->is_contiguous() &&
+                      obs_bf16_out->scalar_type() == torch::kBFloat16 &&
+                      obs_bf16_out->numel() == (int64_t)K.n_envs * K.obs_dim,
+                  "obs_bf16_out must be contiguous bf16 [n_envs, obs_dim]");
+      P.obs_bf16_out = obs_bf16_out->data_ptr();
+    }
     hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
     launch_build_obs(P, K, stream);
   }
@@ -526,8 +558,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("f32_to_bf16", &f32_to_bf16_op);
   py::class_<gymfx::GymFxEngine>(m, "GymFxEngine")
       .def(py::init<const py::dict&, const py::dict&, const py::dict&, const py::dict&>())
-      .def("step", &gymfx::GymFxEngine::step)
-      .def("build_obs", &gymfx::GymFxEngine::build_obs);
+      .def("step", &gymfx::GymFxEngine::step, py::arg("actions"),
+           py::arg("obs_out"), py::arg("reward_out") = py::none(),
+           py::arg("terminated_out") = py::none(),
+           py::arg("obs_bf16_out") = py::none())
+      .def("build_obs", &gymfx::GymFxEngine::build_obs, py::arg("obs_out"),
+           py::arg("obs_bf16_out") = py::none());
   m.attr("EXEC_COUNTER_N") = (int)gymfx::EXEC_COUNTER_N;
   m.attr("ACT_COUNTER_N") = (int)gymfx::ACT_COUNTER_N;
 }

@@ -131,6 +131,7 @@ struct EnvPtrs {
   bool *terminated_out;
   int64_t *coerced_out;
   float *obs_out;            // [N, obs_dim]
+  void *obs_bf16_out;        // optional bf16 mirror (written by build_obs)
 };
 
 }  // namespace gymfx

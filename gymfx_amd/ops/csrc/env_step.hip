@@ -547,6 +547,9 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K) {
       }
     }
     P.obs_out[(int64_t)n * K.obs_dim + j] = val;
+    if (P.obs_bf16_out)  // bf16 mirror: feeds the policy GEMM directly
+      reinterpret_cast<__bf16*>(P.obs_bf16_out)[(int64_t)n * K.obs_dim + j] =
+          (__bf16)val;
   }
 }
 

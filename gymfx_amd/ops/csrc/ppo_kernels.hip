@@ -32,22 +32,28 @@ GFX_DEV __bf16 f2bf(float v) { return (__bf16)v; }
 // GEMM: C[M,N] = act(A[M,K] @ B + bias)
 //   TRANS_B = false: B stored [K, N] (ldb = N)
 //   TRANS_B = true:  B stored [N, K] (ldb = K)  (used for dgrad: B = W^T)
-// Block: 256 threads = 4 waves in a 2x2 wave grid; block tile 64x64;
-// wave tile 32x32 = 2x2 MFMA fragments; BK = 32.
+// Block: 256 threads = 4 waves in a 2x2 wave grid; BK = 32; double-buffered
+// LDS staging (global loads for tile k+1 issue in registers while MFMAs for
+// tile k run — one __syncthreads per K-step).
+// NFRAG = MFMA n-fragments per wave:
+//   NFRAG=2 -> block tile 64x64  (small/medium M: rollout forward at N=4096)
+//   NFRAG=8 -> block tile 64x256 (update phase, M=65536/N=256: the A operand
+//              is streamed from HBM exactly once instead of N/64 times)
 // Epilogues: ACT 0=none(f32 out) 1=none(bf16) 2=tanh(bf16)
 //            DACT_TANH: multiply by (1 - Y^2) elementwise (dgrad fused tanh')
 // ---------------------------------------------------------------------------
-template <bool TRANS_B, int ACT, bool DACT_TANH, bool ADD_BIAS>
+template <bool TRANS_B, int ACT, bool DACT_TANH, bool ADD_BIAS, int NFRAG>
 __global__ __launch_bounds__(256) void gemm_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     const float* __restrict__ bias, void* __restrict__ C,
     const __bf16* __restrict__ Yact,  // activation output (DACT_TANH)
     int M, int N, int K) {
-  constexpr int BM = 64, BN = 64, BK = 32;
+  constexpr int BM = 64, BK = 32;
+  constexpr int BN = 32 * NFRAG;
   constexpr int LDA = BK + 8;  // bf16 elems; 16B-aligned rows, bank-spread
-  constexpr int LDBT = BK + 8;
-  __shared__ __bf16 As[BM][LDA];
-  __shared__ __bf16 Bs[BN][LDBT];  // stored TRANSPOSED: [n][k]
+  constexpr int BVEC = NFRAG / 2;  // bf16x8 loads per thread for the B tile
+  __shared__ __bf16 As[2][BM][LDA];
+  __shared__ __bf16 Bs[2][BN][LDA];  // stored TRANSPOSED: [n][k]
 
   const int bm = blockIdx.x * BM;
   const int bn = blockIdx.y * BN;
@@ -57,69 +63,82 @@ __global__ __launch_bounds__(256) void gemm_kernel(
   const int wr = wave >> 1;          // wave row 0..1
   const int wc = wave & 1;           // wave col 0..1
 
-  f32x4 acc[2][2] = {};
+  f32x4 acc[2][NFRAG] = {};
 
   const int row_a = lane & 15;
   const int kseg = lane >> 4;  // 0..3 -> k-base = kseg*8
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // ---- stage A tile [64][32]: 256 threads x 8 elems = 2048 ----------
-    {
-      const int r = tid >> 2;             // 0..63
-      const int c8 = (tid & 3) * 8;       // 0,8,16,24
-      const int gr = bm + r;
-      bf16x8 v = {};
-      if (gr < M) {
-        const int gk = k0 + c8;
-        if (gk + 8 <= K) {
-          v = *reinterpret_cast<const bf16x8*>(&A[(int64_t)gr * K + gk]);
-        } else {
-          for (int i = 0; i < 8; ++i)
-            v[i] = (gk + i < K) ? A[(int64_t)gr * K + gk + i] : (__bf16)0.f;
-        }
+  // staging coordinates
+  const int a_r = tid >> 2;            // 0..63
+  const int a_c8 = (tid & 3) * 8;      // 0,8,16,24
+  const int b_r = tid >> 2;            // base n row (strided by 64 for NFRAG>2)
+  const int b_c8 = (tid & 3) * 8;
+
+  bf16x8 ra, rb[BVEC];
+
+  auto load_tile = [&](int k0) {
+    // A tile [64 m][32 k]
+    ra = bf16x8{};
+    const int gr = bm + a_r;
+    if (gr < M) {
+      const int gk = k0 + a_c8;
+      if (gk + 8 <= K) {
+        ra = *reinterpret_cast<const bf16x8*>(&A[(int64_t)gr * K + gk]);
+      } else {
+        for (int i = 0; i < 8; ++i)
+          ra[i] = (gk + i < K) ? A[(int64_t)gr * K + gk + i] : (__bf16)0.f;
       }
-      *reinterpret_cast<bf16x8*>(&As[r][c8]) = v;
     }
-    // ---- stage B tile transposed [64(n)][32(k)] -----------------------
-    {
-      const int nn = tid >> 2;            // 0..63 (col of C)
-      const int c8 = (tid & 3) * 8;
-      const int gn = bn + nn;
-      bf16x8 v = {};
+    // B tile transposed [BN n][32 k]
+    for (int j = 0; j < BVEC; ++j) {
+      rb[j] = bf16x8{};
+      const int gn = bn + b_r + j * 64;
       if (gn < N) {
         if (TRANS_B) {
-          // B stored [N, K]: row gn, cols k0+c8..+7 — contiguous
-          const int gk = k0 + c8;
+          const int gk = k0 + b_c8;
           if (gk + 8 <= K) {
-            v = *reinterpret_cast<const bf16x8*>(&B[(int64_t)gn * K + gk]);
+            rb[j] = *reinterpret_cast<const bf16x8*>(&B[(int64_t)gn * K + gk]);
           } else {
             for (int i = 0; i < 8; ++i)
-              v[i] = (gk + i < K) ? B[(int64_t)gn * K + gk + i] : (__bf16)0.f;
+              rb[j][i] = (gk + i < K) ? B[(int64_t)gn * K + gk + i] : (__bf16)0.f;
           }
         } else {
-          // B stored [K, N]: gather column gn (strided)
           for (int i = 0; i < 8; ++i) {
-            const int gk = k0 + c8 + i;
-            v[i] = (gk < K) ? B[(int64_t)gk * N + gn] : (__bf16)0.f;
+            const int gk = k0 + b_c8 + i;
+            rb[j][i] = (gk < K) ? B[(int64_t)gk * N + gn] : (__bf16)0.f;
           }
         }
       }
-      *reinterpret_cast<bf16x8*>(&Bs[nn][c8]) = v;
     }
-    __syncthreads();
+  };
+  auto store_tile = [&](int buf) {
+    *reinterpret_cast<bf16x8*>(&As[buf][a_r][a_c8]) = ra;
+    for (int j = 0; j < BVEC; ++j)
+      *reinterpret_cast<bf16x8*>(&Bs[buf][b_r + j * 64][b_c8]) = rb[j];
+  };
 
-    // ---- MFMA: 2x2 fragments per wave ---------------------------------
-    bf16x8 af[2], bf[2];
+  const int ktiles = (K + BK - 1) / BK;
+  load_tile(0);
+  store_tile(0);
+  __syncthreads();
+
+  for (int kt = 0; kt < ktiles; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < ktiles) load_tile((kt + 1) * BK);
+
+    bf16x8 af[2], bf[NFRAG];
     for (int mi = 0; mi < 2; ++mi)
       af[mi] = *reinterpret_cast<const bf16x8*>(
-          &As[wr * 32 + mi * 16 + row_a][kseg * 8]);
-    for (int ni = 0; ni < 2; ++ni)
+          &As[cur][wr * 32 + mi * 16 + row_a][kseg * 8]);
+    for (int ni = 0; ni < NFRAG; ++ni)
       bf[ni] = *reinterpret_cast<const bf16x8*>(
-          &Bs[wc * 32 + ni * 16 + row_a][kseg * 8]);
+          &Bs[cur][wc * (16 * NFRAG) + ni * 16 + row_a][kseg * 8]);
     for (int mi = 0; mi < 2; ++mi)
-      for (int ni = 0; ni < 2; ++ni)
+      for (int ni = 0; ni < NFRAG; ++ni)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+
+    if (kt + 1 < ktiles) store_tile(1 - cur);
     __syncthreads();
   }
 
@@ -127,8 +146,8 @@ __global__ __launch_bounds__(256) void gemm_kernel(
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
   for (int mi = 0; mi < 2; ++mi) {
-    for (int ni = 0; ni < 2; ++ni) {
-      const int gcol = bn + wc * 32 + ni * 16 + ccol;
+    for (int ni = 0; ni < NFRAG; ++ni) {
+      const int gcol = bn + wc * (16 * NFRAG) + ni * 16 + ccol;
       if (gcol >= N) continue;
       for (int r = 0; r < 4; ++r) {
         const int grow = bm + wr * 32 + mi * 16 + crow_base + r;
@@ -161,21 +180,24 @@ __global__ __launch_bounds__(256) void gemm_kernel(
 // LDS (scalar LDS reads are cheap; strided 2-byte GLOBAL reads were the
 // bottleneck of the first version of this kernel).
 // ---------------------------------------------------------------------------
-template <bool WANT_DB>
+template <bool WANT_DB, int FK, int FN>
 __global__ __launch_bounds__(256) void wgrad_partial_kernel(
     const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
     float* __restrict__ dW_part, float* __restrict__ db_part,
     int M, int N, int K, int slabs) {
-  constexpr int BM = 64, BN = 64, BK = 32;  // BM: K-dim tile, BN: N tile, BK: M chunk step
-  // naming: output tile is [BM of K] x [BN of N]; reduction dim is M.
-  __shared__ __bf16 Xs[BK][BM + 8];   // X tile, natural layout: [m][k]
-  __shared__ __bf16 Ys[BK][BN + 8];   // dY tile, natural layout: [m][n]
+  // FK/FN = MFMA fragments per wave along K/N: block tile (32*FK) x (32*FN)
+  // over the dW output; reduction dim is M, chunked BKm=32 rows at a time
+  // with double-buffered natural-layout LDS staging (coalesced bf16x8
+  // global loads; the transpose happens at the fragment reads).
+  constexpr int BKm = 32;
+  constexpr int TK = 32 * FK, TN = 32 * FN;
+  constexpr int LDX = TK + 8, LDY = TN + 8;
+  __shared__ __bf16 Xs[2][BKm][LDX];   // X tile, natural layout: [m][k]
+  __shared__ __bf16 Ys[2][BKm][LDY];   // dY tile, natural layout: [m][n]
 
-  const int ktile = blockIdx.x;
-  const int ntile = blockIdx.y;
+  const int bk = blockIdx.x * TK;
+  const int bn = blockIdx.y * TN;
   const int slab = blockIdx.z;
-  const int bk = ktile * BM;
-  const int bn = ntile * BN;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int wr = wave >> 1, wc = wave & 1;
@@ -184,83 +206,108 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   const int m_begin = slab * m_per_slab;
   const int m_end = min(M, m_begin + m_per_slab);
 
-  f32x4 acc[2][2] = {};
-  float db_acc = 0.f;  // per-thread partial for db (cols handled below)
+  f32x4 acc[FK][FN] = {};
+  float db_acc = 0.f;
 
-  // coalesced staging map: 256 threads x bf16x8 = 2048 elems = 32 rows x 64 cols
-  const int s_row = tid >> 3;        // 0..31  (m within chunk)
-  const int s_c8 = (tid & 7) * 8;    // 0,8,..,56 (k or n within tile)
-
-  for (int m0 = m_begin; m0 < m_end; m0 += BK) {
-    // stage X tile [32 m][64 k]: contiguous bf16x8 row segments
-    {
-      bf16x8 v = {};
-      const int gm = m0 + s_row;
+  bf16x8 rx[FK / 2], ry[FN / 2];
+  auto load_chunk = [&](int m0) {
+    // X chunk [32 m][TK k]: 32*TK elems = 256 threads * (FK/2) vec8
+    for (int j = 0; j < FK / 2; ++j) {
+      const int c = tid + j * 256;          // chunk id over [32][TK/8]
+      const int m = c / (4 * FK);
+      const int c8 = (c % (4 * FK)) * 8;
+      rx[j] = bf16x8{};
+      const int gm = m0 + m;
       if (gm < m_end) {
-        const int gk = bk + s_c8;
+        const int gk = bk + c8;
         if (gk + 8 <= K) {
-          v = *reinterpret_cast<const bf16x8*>(&X[(int64_t)gm * K + gk]);
+          rx[j] = *reinterpret_cast<const bf16x8*>(&X[(int64_t)gm * K + gk]);
         } else {
           for (int i = 0; i < 8; ++i)
-            v[i] = (gk + i < K) ? X[(int64_t)gm * K + gk + i] : (__bf16)0.f;
+            rx[j][i] = (gk + i < K) ? X[(int64_t)gm * K + gk + i] : (__bf16)0.f;
         }
       }
-      *reinterpret_cast<bf16x8*>(&Xs[s_row][s_c8]) = v;
     }
-    // stage dY tile [32 m][64 n]
-    {
-      bf16x8 v = {};
-      const int gm = m0 + s_row;
+    for (int j = 0; j < FN / 2; ++j) {
+      const int c = tid + j * 256;
+      const int m = c / (4 * FN);
+      const int c8 = (c % (4 * FN)) * 8;
+      ry[j] = bf16x8{};
+      const int gm = m0 + m;
       if (gm < m_end) {
-        const int gn = bn + s_c8;
+        const int gn = bn + c8;
         if (gn + 8 <= N) {
-          v = *reinterpret_cast<const bf16x8*>(&dY[(int64_t)gm * N + gn]);
+          ry[j] = *reinterpret_cast<const bf16x8*>(&dY[(int64_t)gm * N + gn]);
         } else {
           for (int i = 0; i < 8; ++i)
-            v[i] = (gn + i < N) ? dY[(int64_t)gm * N + gn + i] : (__bf16)0.f;
+            ry[j][i] = (gn + i < N) ? dY[(int64_t)gm * N + gn + i] : (__bf16)0.f;
         }
       }
-      *reinterpret_cast<bf16x8*>(&Ys[s_row][s_c8]) = v;
     }
+  };
+  auto store_chunk = [&](int buf) {
+    for (int j = 0; j < FK / 2; ++j) {
+      const int c = tid + j * 256;
+      *reinterpret_cast<bf16x8*>(&Xs[buf][c / (4 * FK)][(c % (4 * FK)) * 8]) = rx[j];
+    }
+    for (int j = 0; j < FN / 2; ++j) {
+      const int c = tid + j * 256;
+      *reinterpret_cast<bf16x8*>(&Ys[buf][c / (4 * FN)][(c % (4 * FN)) * 8]) = ry[j];
+    }
+  };
+
+  const int row_a = lane & 15;
+  const int kseg = lane >> 4;
+  const int nchunks = (m_end - m_begin + BKm - 1) / BKm;
+
+  if (nchunks > 0) {
+    load_chunk(m_begin);
+    store_chunk(0);
     __syncthreads();
-    if (WANT_DB && tid < BN) {
+  }
+  for (int ci = 0; ci < nchunks; ++ci) {
+    const int cur = ci & 1;
+    const int m0 = m_begin + ci * BKm;
+    if (ci + 1 < nchunks) load_chunk(m0 + BKm);
+
+    if (WANT_DB && tid < TN) {
       float s = 0.f;
-      for (int i = 0; i < BK && m0 + i < m_end; ++i) s += bf2f(Ys[i][tid]);
+      for (int i = 0; i < BKm && m0 + i < m_end; ++i)
+        s += bf2f(Ys[cur][i][tid]);
       db_acc += s;
     }
-
-    const int row_a = lane & 15;
-    const int kseg = lane >> 4;
     // fragment reads do the transpose: logical A[kout][m] = X[m][kout],
     // logical B[m][n] = dY[m][n]; reduction index m = kseg*8 + i.
-    bf16x8 af[2], bf_[2];
-    for (int mi = 0; mi < 2; ++mi)
+    bf16x8 af[FK], bf_[FN];
+    for (int fi = 0; fi < FK; ++fi)
       for (int i = 0; i < 8; ++i)
-        af[mi][i] = Xs[kseg * 8 + i][wr * 32 + mi * 16 + row_a];
-    for (int ni = 0; ni < 2; ++ni)
+        af[fi][i] = Xs[cur][kseg * 8 + i][wr * (16 * FK) + fi * 16 + row_a];
+    for (int ni = 0; ni < FN; ++ni)
       for (int i = 0; i < 8; ++i)
-        bf_[ni][i] = Ys[kseg * 8 + i][wc * 32 + ni * 16 + row_a];
-    for (int mi = 0; mi < 2; ++mi)
-      for (int ni = 0; ni < 2; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[mi], bf_[ni], acc[mi][ni], 0, 0, 0);
+        bf_[ni][i] = Ys[cur][kseg * 8 + i][wc * (16 * FN) + ni * 16 + row_a];
+    for (int fi = 0; fi < FK; ++fi)
+      for (int ni = 0; ni < FN; ++ni)
+        acc[fi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[fi], bf_[ni], acc[fi][ni], 0, 0, 0);
+
+    if (ci + 1 < nchunks) store_chunk(1 - cur);
     __syncthreads();
   }
 
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
   float* out = dW_part + (int64_t)slab * K * N;
-  for (int mi = 0; mi < 2; ++mi)
-    for (int ni = 0; ni < 2; ++ni) {
-      const int gcol = bn + wc * 32 + ni * 16 + ccol;
+  for (int fi = 0; fi < FK; ++fi)
+    for (int ni = 0; ni < FN; ++ni) {
+      const int gcol = bn + wc * (16 * FN) + ni * 16 + ccol;
       if (gcol >= N) continue;
       for (int r = 0; r < 4; ++r) {
-        const int grow = bk + wr * 32 + mi * 16 + crow_base + r;
+        const int grow = bk + wr * (16 * FK) + fi * 16 + crow_base + r;
         if (grow >= K) continue;
-        out[(int64_t)grow * N + gcol] = acc[mi][ni][r];
+        out[(int64_t)grow * N + gcol] = acc[fi][ni][r];
       }
     }
-  if (WANT_DB && tid < BN && ktile == 0) {
+  if (WANT_DB && tid < TN && blockIdx.x == 0) {
     const int gn = bn + tid;
     if (gn < N) db_part[(int64_t)slab * N + gn] = db_acc;
   }
@@ -269,8 +316,27 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
 __global__ void slab_reduce_kernel(const float* __restrict__ part,
                                    float* __restrict__ out, int64_t elems,
                                    int slabs) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < elems;
+  // float4 per thread + 4-way unrolled slab loop (keeps >=16 loads in
+  // flight; the scalar version was latency-bound on the slab chain).
+  const int64_t nvec = elems >> 2;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (int64_t)gridDim.x * blockDim.x) {
+    f32x4 s0 = {}, s1 = {}, s2 = {}, s3 = {};
+    int k = 0;
+    for (; k + 4 <= slabs; k += 4) {
+      s0 += *reinterpret_cast<const f32x4*>(&part[(int64_t)k * elems + i * 4]);
+      s1 += *reinterpret_cast<const f32x4*>(&part[(int64_t)(k + 1) * elems + i * 4]);
+      s2 += *reinterpret_cast<const f32x4*>(&part[(int64_t)(k + 2) * elems + i * 4]);
+      s3 += *reinterpret_cast<const f32x4*>(&part[(int64_t)(k + 3) * elems + i * 4]);
+    }
+    for (; k < slabs; ++k)
+      s0 += *reinterpret_cast<const f32x4*>(&part[(int64_t)k * elems + i * 4]);
+    // fixed association order: (s0 + s1) + (s2 + s3) — deterministic
+    *reinterpret_cast<f32x4*>(&out[i * 4]) = (s0 + s1) + (s2 + s3);
+  }
+  // scalar tail
+  for (int64_t i = (nvec << 2) + blockIdx.x * blockDim.x + threadIdx.x;
+       i < elems; i += (int64_t)gridDim.x * blockDim.x) {
     float s = 0.f;
     for (int k = 0; k < slabs; ++k) s += part[(int64_t)k * elems + i];
     out[i] = s;
@@ -361,10 +427,19 @@ __global__ void sumsq_partial_kernel(const float* __restrict__ g, int64_t n,
 
 __global__ void clip_scale_kernel(const float* __restrict__ part, int nparts,
                                   float max_norm, float* __restrict__ scale) {
-  if (threadIdx.x == 0 && blockIdx.x == 0) {
-    float s = 0.f;
-    for (int i = 0; i < nparts; ++i) s += part[i];
-    const float norm = sqrtf(s);
+  // one block of 256; tree reduce (deterministic, ~4x faster than the
+  // single-thread serial loop this replaced)
+  __shared__ float red[256];
+  float s = 0.f;
+  for (int i = threadIdx.x; i < nparts; i += 256) s += part[i];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    const float norm = sqrtf(red[0]);
     *scale = (max_norm > 0.f && norm > max_norm) ? max_norm / norm : 1.f;
   }
 }
@@ -667,48 +742,64 @@ void launch_mb_gather(const void* obs_src, const int64_t* act_src,
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
                  bool dact_tanh, bool add_bias, hipStream_t stream) {
-  dim3 grid(ceil_div(M, 64), ceil_div(N, 64));
-  dim3 block(256);
   const __bf16* a = reinterpret_cast<const __bf16*>(A);
   const __bf16* b = reinterpret_cast<const __bf16*>(B);
   const __bf16* y = reinterpret_cast<const __bf16*>(Yact);
+  // Wide-N (64x256) tiles when the tall-M operand would otherwise be
+  // re-streamed from HBM N/64 times AND the grid still fills the 256 CUs.
+  const bool wide = (N >= 192) && (M >= 16384);
+  dim3 grid(ceil_div(M, 64), ceil_div(N, wide ? 256 : 64));
+  dim3 block(256);
 
-#define GEMM_CASE(TB, ACT, DT, AB)                                            \
-  hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB>), grid, block, 0, stream,  \
-                     a, b, bias, C, y, M, N, K)
+#define GEMM_LAUNCH(TB, ACT, DT, AB)                                          \
+  do {                                                                        \
+    if (wide)                                                                 \
+      hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 8>), grid, block, 0,   \
+                         stream, a, b, bias, C, y, M, N, K);                  \
+    else                                                                      \
+      hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 2>), grid, block, 0,   \
+                         stream, a, b, bias, C, y, M, N, K);                  \
+  } while (0)
   if (!trans_b && !dact_tanh) {
-    if (act == 0) { if (add_bias) GEMM_CASE(false, 0, false, true); else GEMM_CASE(false, 0, false, false); }
-    else if (act == 1) { if (add_bias) GEMM_CASE(false, 1, false, true); else GEMM_CASE(false, 1, false, false); }
-    else { if (add_bias) GEMM_CASE(false, 2, false, true); else GEMM_CASE(false, 2, false, false); }
+    if (act == 0) { if (add_bias) GEMM_LAUNCH(false, 0, false, true); else GEMM_LAUNCH(false, 0, false, false); }
+    else if (act == 1) { if (add_bias) GEMM_LAUNCH(false, 1, false, true); else GEMM_LAUNCH(false, 1, false, false); }
+    else { if (add_bias) GEMM_LAUNCH(false, 2, false, true); else GEMM_LAUNCH(false, 2, false, false); }
   } else if (trans_b && !dact_tanh) {
-    if (act == 0) { if (add_bias) GEMM_CASE(true, 0, false, true); else GEMM_CASE(true, 0, false, false); }
-    else if (act == 1) { if (add_bias) GEMM_CASE(true, 1, false, true); else GEMM_CASE(true, 1, false, false); }
-    else { if (add_bias) GEMM_CASE(true, 2, false, true); else GEMM_CASE(true, 2, false, false); }
+    if (act == 0) { if (add_bias) GEMM_LAUNCH(true, 0, false, true); else GEMM_LAUNCH(true, 0, false, false); }
+    else if (act == 1) { if (add_bias) GEMM_LAUNCH(true, 1, false, true); else GEMM_LAUNCH(true, 1, false, false); }
+    else { if (add_bias) GEMM_LAUNCH(true, 2, false, true); else GEMM_LAUNCH(true, 2, false, false); }
   } else if (trans_b && dact_tanh) {
-    GEMM_CASE(true, 1, true, false);
+    GEMM_LAUNCH(true, 1, true, false);
   } else {
-    GEMM_CASE(false, 1, true, false);
+    GEMM_LAUNCH(false, 1, true, false);
   }
-#undef GEMM_CASE
+#undef GEMM_LAUNCH
 }
 
 void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                   float* dW, float* db, int M, int N, int K, int slabs,
                   hipStream_t stream) {
-  dim3 grid(ceil_div(K, 64), ceil_div(N, 64), slabs);
-  if (db_part) {
-    hipLaunchKernelGGL((wgrad_partial_kernel<true>), grid, dim3(256), 0, stream,
-                       reinterpret_cast<const __bf16*>(X),
-                       reinterpret_cast<const __bf16*>(dY), dW_part, db_part, M,
-                       N, K, slabs);
-  } else {
-    hipLaunchKernelGGL((wgrad_partial_kernel<false>), grid, dim3(256), 0,
-                       stream, reinterpret_cast<const __bf16*>(X),
-                       reinterpret_cast<const __bf16*>(dY), dW_part, nullptr, M,
-                       N, K, slabs);
-  }
+  const __bf16* x = reinterpret_cast<const __bf16*>(X);
+  const __bf16* dy = reinterpret_cast<const __bf16*>(dY);
+  // 128x128 output tiles when K and N are big enough (halves the HBM
+  // re-streaming of X and dY vs 64x64); 64x64 otherwise (e.g. the head
+  // layer, N = n_actions+1).
+  const bool big = (K >= 128) && (N >= 128);
+  dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);
+#define WGRAD_LAUNCH(DB)                                                       \
+  do {                                                                         \
+    if (big)                                                                   \
+      hipLaunchKernelGGL((wgrad_partial_kernel<DB, 4, 4>), grid, dim3(256), 0, \
+                         stream, x, dy, dW_part, db_part, M, N, K, slabs);     \
+    else                                                                       \
+      hipLaunchKernelGGL((wgrad_partial_kernel<DB, 2, 2>), grid, dim3(256), 0, \
+                         stream, x, dy, dW_part, db_part, M, N, K, slabs);     \
+  } while (0)
+  if (db_part) WGRAD_LAUNCH(true);
+  else WGRAD_LAUNCH(false);
+#undef WGRAD_LAUNCH
   int64_t elems = (int64_t)K * N;
-  int blocks = (int)std::min<int64_t>((elems + 255) / 256, 1024);
+  int blocks = (int)std::min<int64_t>((elems + 1023) / 1024, 1024);
   hipLaunchKernelGGL(slab_reduce_kernel, dim3(blocks), dim3(256), 0, stream,
                      dW_part, dW, elems, slabs);
   if (db_part && db)
@@ -737,7 +828,7 @@ void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
                       float* scale, int nparts, hipStream_t stream) {
   hipLaunchKernelGGL(sumsq_partial_kernel, dim3(nparts), dim3(256), 0, stream,
                      g, n, part);
-  hipLaunchKernelGGL(clip_scale_kernel, dim3(1), dim3(1), 0, stream, part,
+  hipLaunchKernelGGL(clip_scale_kernel, dim3(1), dim3(256), 0, stream, part,
                      nparts, max_norm, scale);
 }
 

@@ -164,12 +164,16 @@ class NativeEngine:
         self._params = params
         self._engine = ext.GymFxEngine(iparams, fparams, st.to_dict(), _market_dict(mt))
 
-    def step(self, actions: torch.Tensor, obs_out: torch.Tensor) -> Dict[str, torch.Tensor]:
+    def step(self, actions: torch.Tensor, obs_out: torch.Tensor,
+             reward_out: torch.Tensor = None, terminated_out: torch.Tensor = None,
+             obs_bf16_out: torch.Tensor = None) -> Dict[str, torch.Tensor]:
         if self._params.action_space_mode == "continuous":
             actions = actions.to(torch.float32).contiguous()
         else:
             actions = actions.to(torch.int64).contiguous()
-        return dict(self._engine.step(actions, obs_out))
+        return dict(self._engine.step(actions, obs_out, reward_out,
+                                      terminated_out, obs_bf16_out))
 
-    def build_obs(self, obs_out: torch.Tensor) -> None:
-        self._engine.build_obs(obs_out)
+    def build_obs(self, obs_out: torch.Tensor,
+                  obs_bf16_out: torch.Tensor = None) -> None:
+        self._engine.build_obs(obs_out, obs_bf16_out)
