@@ -96,13 +96,21 @@ class ActorCriticMLP:
         s = self.slices[name]
         return self.grads[s.sl].view(*s.shape)
 
+    def _wg_slabs(self, K: int, N: int) -> int:
+        # enough split-M slabs that tiles x slabs >= ~4 blocks per CU
+        tiles = ((K + 63) // 64) * ((N + 63) // 64)
+        s = self.wgrad_slabs
+        while tiles * s < 1024:
+            s *= 2
+        return s
+
     def _wg_workspace(self, name: str, K: int, N: int, want_db: bool):
         key = f"{name}:{K}x{N}"
         if key not in self._wg_ws:
-            S = self.wgrad_slabs
+            S = self._wg_slabs(K, N)
             dw = torch.empty(S, K, N, dtype=torch.float32, device=self.device)
             db = torch.empty(S, N, dtype=torch.float32, device=self.device) if want_db else None
-            self._wg_ws[key] = (dw, db)
+            self._wg_ws[key] = (dw, db, S)
         return self._wg_ws[key]
 
     # -- forward ---------------------------------------------------------
@@ -133,27 +141,21 @@ class ActorCriticMLP:
         dh2 = scratch["dh2"]
         dh1 = scratch["dh1"]
         # head layer
-        api.wgrad(
-            acts["h2"], dhead, self.grad("W3"), self.grad("b3"),
-            workspace=self._wg_workspace("W3", self.hidden, self.head_dim, True),
-            slabs=self.wgrad_slabs,
-        )
+        dw_p, db_p, slabs = self._wg_workspace("W3", self.hidden, self.head_dim, True)
+        api.wgrad(acts["h2"], dhead, self.grad("W3"), self.grad("b3"),
+                  workspace=(dw_p, db_p), slabs=slabs)
         api.gemm(dhead, self.w("W3"), None, dh2, Yact=acts["h2"], trans_b=True,
                  act=1, dact_tanh=True)
         # layer 2
-        api.wgrad(
-            acts["h1"], dh2, self.grad("W2"), self.grad("b2"),
-            workspace=self._wg_workspace("W2", self.hidden, self.hidden, True),
-            slabs=self.wgrad_slabs,
-        )
+        dw_p, db_p, slabs = self._wg_workspace("W2", self.hidden, self.hidden, True)
+        api.wgrad(acts["h1"], dh2, self.grad("W2"), self.grad("b2"),
+                  workspace=(dw_p, db_p), slabs=slabs)
         api.gemm(dh2, self.w("W2"), None, dh1, Yact=acts["h1"], trans_b=True,
                  act=1, dact_tanh=True)
         # layer 1
-        api.wgrad(
-            obs_bf16, dh1, self.grad("W1"), self.grad("b1"),
-            workspace=self._wg_workspace("W1", self.obs_dim, self.hidden, True),
-            slabs=self.wgrad_slabs,
-        )
+        dw_p, db_p, slabs = self._wg_workspace("W1", self.obs_dim, self.hidden, True)
+        api.wgrad(obs_bf16, dh1, self.grad("W1"), self.grad("b1"),
+                  workspace=(dw_p, db_p), slabs=slabs)
 
     def alloc_scratch(self, M: int) -> Dict[str, torch.Tensor]:
         dev = self.device

@@ -52,8 +52,14 @@ __global__ __launch_bounds__(256) void gemm_kernel(
   constexpr int BN = 32 * NFRAG;
   constexpr int LDA = BK + 8;  // bf16 elems; 16B-aligned rows, bank-spread
   constexpr int BVEC = NFRAG / 2;  // bf16x8 loads per thread for the B tile
+  // B tile layout follows the GLOBAL layout so staging loads are always
+  // contiguous bf16x8 (a strided 2-byte gather of B was 6x slower than the
+  // whole GEMM): TRANS_B stages [n][k], !TRANS_B stages [k][n]; the
+  // !TRANS_B fragment reads transpose out of LDS instead.
+  constexpr int BS_ROWS = TRANS_B ? BN : BK;
+  constexpr int BS_LD = (TRANS_B ? BK : BN) + 8;
   __shared__ __bf16 As[2][BM][LDA];
-  __shared__ __bf16 Bs[2][BN][LDA];  // stored TRANSPOSED: [n][k]
+  __shared__ __bf16 Bs[2][BS_ROWS][BS_LD];
 
   const int bm = blockIdx.x * BM;
   const int bn = blockIdx.y * BN;
@@ -89,12 +95,12 @@ __global__ __launch_bounds__(256) void gemm_kernel(
           ra[i] = (gk + i < K) ? A[(int64_t)gr * K + gk + i] : (__bf16)0.f;
       }
     }
-    // B tile transposed [BN n][32 k]
-    for (int j = 0; j < BVEC; ++j) {
-      rb[j] = bf16x8{};
-      const int gn = bn + b_r + j * 64;
-      if (gn < N) {
-        if (TRANS_B) {
+    if (TRANS_B) {
+      // stage [n][k]: contiguous along k in global [N, K]
+      for (int j = 0; j < BVEC; ++j) {
+        rb[j] = bf16x8{};
+        const int gn = bn + b_r + j * 64;
+        if (gn < N) {
           const int gk = k0 + b_c8;
           if (gk + 8 <= K) {
             rb[j] = *reinterpret_cast<const bf16x8*>(&B[(int64_t)gn * K + gk]);
@@ -102,10 +108,23 @@ __global__ __launch_bounds__(256) void gemm_kernel(
             for (int i = 0; i < 8; ++i)
               rb[j][i] = (gk + i < K) ? B[(int64_t)gn * K + gk + i] : (__bf16)0.f;
           }
-        } else {
-          for (int i = 0; i < 8; ++i) {
-            const int gk = k0 + b_c8 + i;
-            rb[j][i] = (gk < K) ? B[(int64_t)gk * N + gn] : (__bf16)0.f;
+        }
+      }
+    } else {
+      // stage [k][n]: contiguous along n in global [K, N]
+      for (int j = 0; j < BVEC; ++j) {
+        rb[j] = bf16x8{};
+        const int c = tid + j * 256;             // chunk over [BK][BN/8]
+        const int kk = c / (BN / 8);
+        const int n8 = (c % (BN / 8)) * 8;
+        const int gk = k0 + kk;
+        const int gn = bn + n8;
+        if (gk < K) {
+          if (gn + 8 <= N) {
+            rb[j] = *reinterpret_cast<const bf16x8*>(&B[(int64_t)gk * N + gn]);
+          } else {
+            for (int i = 0; i < 8; ++i)
+              rb[j][i] = (gn + i < N) ? B[(int64_t)gk * N + gn + i] : (__bf16)0.f;
           }
         }
       }
@@ -113,8 +132,16 @@ __global__ __launch_bounds__(256) void gemm_kernel(
   };
   auto store_tile = [&](int buf) {
     *reinterpret_cast<bf16x8*>(&As[buf][a_r][a_c8]) = ra;
-    for (int j = 0; j < BVEC; ++j)
-      *reinterpret_cast<bf16x8*>(&Bs[buf][b_r + j * 64][b_c8]) = rb[j];
+    if (TRANS_B) {
+      for (int j = 0; j < BVEC; ++j)
+        *reinterpret_cast<bf16x8*>(&Bs[buf][b_r + j * 64][b_c8]) = rb[j];
+    } else {
+      for (int j = 0; j < BVEC; ++j) {
+        const int c = tid + j * 256;
+        *reinterpret_cast<bf16x8*>(
+            &Bs[buf][c / (BN / 8)][(c % (BN / 8)) * 8]) = rb[j];
+      }
+    }
   };
 
   const int ktiles = (K + BK - 1) / BK;
@@ -130,9 +157,17 @@ __global__ __launch_bounds__(256) void gemm_kernel(
     for (int mi = 0; mi < 2; ++mi)
       af[mi] = *reinterpret_cast<const bf16x8*>(
           &As[cur][wr * 32 + mi * 16 + row_a][kseg * 8]);
-    for (int ni = 0; ni < NFRAG; ++ni)
-      bf[ni] = *reinterpret_cast<const bf16x8*>(
-          &Bs[cur][wc * (16 * NFRAG) + ni * 16 + row_a][kseg * 8]);
+    if (TRANS_B) {
+      for (int ni = 0; ni < NFRAG; ++ni)
+        bf[ni] = *reinterpret_cast<const bf16x8*>(
+            &Bs[cur][wc * (16 * NFRAG) + ni * 16 + row_a][kseg * 8]);
+    } else {
+      // transpose at the LDS read: Bs holds [k][n]
+      for (int ni = 0; ni < NFRAG; ++ni)
+        for (int i = 0; i < 8; ++i)
+          bf[ni][i] =
+              Bs[cur][kseg * 8 + i][wc * (16 * NFRAG) + ni * 16 + row_a];
+    }
     for (int mi = 0; mi < 2; ++mi)
       for (int ni = 0; ni < NFRAG; ++ni)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -781,23 +816,18 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                   hipStream_t stream) {
   const __bf16* x = reinterpret_cast<const __bf16*>(X);
   const __bf16* dy = reinterpret_cast<const __bf16*>(dY);
-  // 128x128 output tiles when K and N are big enough (halves the HBM
-  // re-streaming of X and dY vs 64x64); 64x64 otherwise (e.g. the head
-  // layer, N = n_actions+1).
-  const bool big = (K >= 128) && (N >= 128);
-  dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);
-#define WGRAD_LAUNCH(DB)                                                       \
-  do {                                                                         \
-    if (big)                                                                   \
-      hipLaunchKernelGGL((wgrad_partial_kernel<DB, 4, 4>), grid, dim3(256), 0, \
-                         stream, x, dy, dW_part, db_part, M, N, K, slabs);     \
-    else                                                                       \
-      hipLaunchKernelGGL((wgrad_partial_kernel<DB, 2, 2>), grid, dim3(256), 0, \
-                         stream, x, dy, dW_part, db_part, M, N, K, slabs);     \
-  } while (0)
-  if (db_part) WGRAD_LAUNCH(true);
-  else WGRAD_LAUNCH(false);
-#undef WGRAD_LAUNCH
+  // 64x64 output tiles: 128x128 halves the HBM re-streaming but drops the
+  // grid to ~1 block/CU at these layer shapes (measured slower).  The
+  // caller raises `slabs` instead when K/64 * N/64 is small so the grid
+  // still fills the 256 CUs.
+  dim3 grid(ceil_div(K, 64), ceil_div(N, 64), slabs);
+  if (db_part) {
+    hipLaunchKernelGGL((wgrad_partial_kernel<true, 2, 2>), grid, dim3(256), 0,
+                       stream, x, dy, dW_part, db_part, M, N, K, slabs);
+  } else {
+    hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), grid, dim3(256), 0,
+                       stream, x, dy, dW_part, db_part, M, N, K, slabs);
+  }
   int64_t elems = (int64_t)K * N;
   int blocks = (int)std::min<int64_t>((elems + 1023) / 1024, 1024);
   hipLaunchKernelGGL(slab_reduce_kernel, dim3(blocks), dim3(256), 0, stream,

@@ -22,7 +22,8 @@ def _rand_bf16(*shape, seed=0, scale=1.0):
 # GEMM
 # ---------------------------------------------------------------------------
 @pytest.mark.parametrize("M,N,K", [(64, 64, 64), (256, 256, 96), (1000, 68, 132),
-                                   (4096, 256, 324), (512, 4, 256)])
+                                   (4096, 256, 324), (512, 4, 256),
+                                   (16384, 256, 68), (20480, 224, 52)])
 def test_gemm_identity_probe_and_random(M, N, K):
     ext = _ext()
     # asymmetric-B identity probe (guide rule G9: transpose-detecting)
