@@ -73,6 +73,15 @@ class ActorCriticMLP:
         self.v = torch.zeros_like(self.params)
         # bf16 mirrors of the weight matrices (GEMM operands)
         self.params_bf16 = self.params.to(torch.bfloat16)
+        # transposed [N,K] mirrors of the weight matrices: forward GEMMs use
+        # the TRANS_B path (contiguous B staging); refreshed after every
+        # Adam step by the transpose kernel.
+        self._wt: Dict[str, torch.Tensor] = {}
+        for name in ("W1", "W2", "W3"):
+            s = self.slices[name]
+            K_, N_ = s.shape
+            self._wt[name] = torch.empty(N_, K_, dtype=torch.bfloat16, device=device)
+        self._refresh_wt()
         self.adam_step = 0  # host mirror of adam_ctr (logging/checkpoint)
         # device step counter: Adam bias correction inside captured hipGraphs
         self.adam_ctr = torch.zeros((), dtype=torch.int32, device=device)
@@ -87,6 +96,14 @@ class ActorCriticMLP:
     def w(self, name: str) -> torch.Tensor:
         s = self.slices[name]
         return self.params_bf16[s.sl].view(*s.shape)
+
+    def wt(self, name: str) -> torch.Tensor:
+        """Transposed [N,K] bf16 mirror (forward GEMM B operand)."""
+        return self._wt[name]
+
+    def _refresh_wt(self) -> None:
+        for name, t in self._wt.items():
+            api.transpose_bf16(self.w(name), t)
 
     def f32(self, name: str) -> torch.Tensor:
         s = self.slices[name]
@@ -123,10 +140,13 @@ class ActorCriticMLP:
         }
 
     def forward(self, obs_bf16: torch.Tensor, acts: Dict[str, torch.Tensor]) -> torch.Tensor:
-        """obs_bf16 [M, obs_dim] -> head f32 [M, A+1]; saves h1/h2 for bwd."""
-        api.gemm(obs_bf16, self.w("W1"), self.f32("b1"), acts["h1"], act=2)
-        api.gemm(acts["h1"], self.w("W2"), self.f32("b2"), acts["h2"], act=2)
-        api.gemm(acts["h2"], self.w("W3"), self.f32("b3"), acts["head"], act=0)
+        """obs_bf16 [M, obs_dim] -> head f32 [M, A+1]; saves h1/h2 for bwd.
+
+        B operands are the transposed mirrors (TRANS_B path): contiguous
+        vector staging on gfx950 regardless of tile width."""
+        api.gemm(obs_bf16, self.wt("W1"), self.f32("b1"), acts["h1"], act=2, trans_b=True)
+        api.gemm(acts["h1"], self.wt("W2"), self.f32("b2"), acts["h2"], act=2, trans_b=True)
+        api.gemm(acts["h2"], self.wt("W3"), self.f32("b3"), acts["head"], act=0, trans_b=True)
         return acts["head"]
 
     # -- backward (dhead [M, A+1] bf16 -> accumulate grads) --------------
@@ -179,6 +199,7 @@ class ActorCriticMLP:
                  lr=lr, beta1=beta1, beta2=beta2, eps=eps, step=self.adam_step,
                  gscale=gscale, step_ctr=self.adam_ctr)
         api.increment_counter(self.adam_ctr, 1)
+        self._refresh_wt()
 
     def zero_grad(self) -> None:
         self.grads.zero_()
@@ -202,3 +223,4 @@ class ActorCriticMLP:
         self.adam_step = int(sd["adam_step"])
         self.adam_ctr.fill_(self.adam_step)
         self.params_bf16.copy_(self.params.to(torch.bfloat16))
+        self._refresh_wt()

@@ -384,3 +384,12 @@ def increment_counter(ctr: torch.Tensor, delta: int) -> None:
         native.require().increment_counter(ctr, delta)
         return
     ctr.add_(delta)
+
+
+def transpose_bf16(src: torch.Tensor, dst: torch.Tensor) -> None:
+    """dst[N,K] = src[K,N]^T (bf16). Keeps the model's transposed weight
+    mirrors fresh so GEMM B operands always stage contiguously."""
+    if _use_native(src):
+        native.require().transpose_bf16(src, dst)
+        return
+    dst.copy_(src.t())

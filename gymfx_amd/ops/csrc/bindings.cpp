@@ -55,6 +55,8 @@ void launch_adv_normalize(float* adv, int64_t n, float* part, int nparts,
                           hipStream_t stream);
 void launch_f32_to_bf16(const float* in, void* out, int64_t n,
                         hipStream_t stream);
+void launch_transpose_bf16(const void* src, void* dst, int K, int N,
+                           hipStream_t stream);
 
 namespace {
 
@@ -522,6 +524,15 @@ void f32_to_bf16_op(torch::Tensor in, torch::Tensor out) {
                             cur_stream());
 }
 
+void transpose_bf16_op(torch::Tensor src, torch::Tensor dst) {
+  check_bf16(src, "src");
+  check_bf16(dst, "dst");
+  const int K = (int)src.size(0), N = (int)src.size(1);
+  TORCH_CHECK(dst.size(0) == N && dst.size(1) == K, "transpose shape");
+  gymfx::launch_transpose_bf16(src.data_ptr(), dst.data_ptr(), K, N,
+                               cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -556,6 +567,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("inv_count"), py::arg("losses") = py::none());
   m.def("adv_normalize", &adv_normalize_op);
   m.def("f32_to_bf16", &f32_to_bf16_op);
+  m.def("transpose_bf16", &transpose_bf16_op);
   py::class_<gymfx::GymFxEngine>(m, "GymFxEngine")
       .def(py::init<const py::dict&, const py::dict&, const py::dict&, const py::dict&>())
       .def("step", &gymfx::GymFxEngine::step, py::arg("actions"),

@@ -378,6 +378,73 @@ __global__ void slab_reduce_kernel(const float* __restrict__ part,
   }
 }
 
+// tiny-elems variant (e.g. the head bias, elems=4, slabs up to 256): one
+// block per element, threads parallel over slabs, deterministic tree reduce.
+__global__ void slab_reduce_small_kernel(const float* __restrict__ part,
+                                         float* __restrict__ out,
+                                         int64_t elems, int slabs) {
+  __shared__ float red[256];
+  const int64_t e = blockIdx.x;
+  if (e >= elems) return;
+  float s = 0.f;
+  for (int k = threadIdx.x; k < slabs; k += 256)
+    s += part[(int64_t)k * elems + e];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int w = 128; w > 0; w >>= 1) {
+    if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) out[e] = red[0];
+}
+
+// ---------------------------------------------------------------------------
+// bf16 transpose: dst[N,K] = src[K,N]^T.  Keeps per-weight transposed
+// mirrors fresh after each Adam step so every model GEMM can stage its B
+// operand with contiguous vector loads (TRANS_B path) — the strided
+// alternative spilled registers and ran 6x slower.  LDS 64x64 tile.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void transpose_bf16_kernel(
+    const __bf16* __restrict__ src, __bf16* __restrict__ dst, int K, int N) {
+  __shared__ __bf16 tile[64][64 + 4];
+  const int kb = blockIdx.x * 64;
+  const int nb = blockIdx.y * 64;
+  // load 64x64 tile of src (rows k, cols n), coalesced along n
+  for (int e = threadIdx.x; e < 64 * 8; e += 256) {
+    const int r = e >> 3;            // 0..63 (k)
+    const int c8 = (e & 7) * 8;      // 0..56 (n)
+    const int gk = kb + r;
+    const int gn = nb + c8;
+    bf16x8 v = {};
+    if (gk < K) {
+      if (gn + 8 <= N) {
+        v = *reinterpret_cast<const bf16x8*>(&src[(int64_t)gk * N + gn]);
+      } else {
+        for (int i = 0; i < 8; ++i)
+          v[i] = (gn + i < N) ? src[(int64_t)gk * N + gn + i] : (__bf16)0.f;
+      }
+    }
+    *reinterpret_cast<bf16x8*>(&tile[r][c8]) = v;
+  }
+  __syncthreads();
+  // write transposed, coalesced along k
+  for (int e = threadIdx.x; e < 64 * 8; e += 256) {
+    const int r = e >> 3;            // 0..63 (n)
+    const int c8 = (e & 7) * 8;      // 0..56 (k)
+    const int gn = nb + r;
+    const int gk = kb + c8;
+    if (gn >= N) continue;
+    if (gk + 8 <= K) {
+      bf16x8 v;
+      for (int i = 0; i < 8; ++i) v[i] = tile[c8 + i][r];
+      *reinterpret_cast<bf16x8*>(&dst[(int64_t)gn * K + gk]) = v;
+    } else {
+      for (int i = 0; i < 8 && gk + i < K; ++i)
+        dst[(int64_t)gn * K + gk + i] = tile[c8 + i][r];
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // elementwise casts
 // ---------------------------------------------------------------------------
@@ -829,11 +896,16 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                        stream, x, dy, dW_part, db_part, M, N, K, slabs);
   }
   int64_t elems = (int64_t)K * N;
-  int blocks = (int)std::min<int64_t>((elems + 1023) / 1024, 1024);
-  hipLaunchKernelGGL(slab_reduce_kernel, dim3(blocks), dim3(256), 0, stream,
-                     dW_part, dW, elems, slabs);
+  if (elems <= 4096) {
+    hipLaunchKernelGGL(slab_reduce_small_kernel, dim3((int)elems), dim3(256),
+                       0, stream, dW_part, dW, elems, slabs);
+  } else {
+    int blocks = (int)std::min<int64_t>((elems + 1023) / 1024, 1024);
+    hipLaunchKernelGGL(slab_reduce_kernel, dim3(blocks), dim3(256), 0, stream,
+                       dW_part, dW, elems, slabs);
+  }
   if (db_part && db)
-    hipLaunchKernelGGL(slab_reduce_kernel, dim3(1), dim3(256), 0, stream,
+    hipLaunchKernelGGL(slab_reduce_small_kernel, dim3(N), dim3(256), 0, stream,
                        db_part, db, (int64_t)N, slabs);
 }
 
@@ -890,6 +962,14 @@ void launch_adv_normalize(float* adv, int64_t n, float* part, int nparts,
   int blocks = (int)std::min<int64_t>((n + 255) / 256, 2048);
   hipLaunchKernelGGL(adv_norm_apply_kernel, dim3(blocks), dim3(256), 0, stream,
                      adv, n, part, nparts);
+}
+
+void launch_transpose_bf16(const void* src, void* dst, int K, int N,
+                           hipStream_t stream) {
+  dim3 grid(ceil_div(K, 64), ceil_div(N, 64));
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream,
+                     reinterpret_cast<const __bf16*>(src),
+                     reinterpret_cast<__bf16*>(dst), K, N);
 }
 
 void launch_f32_to_bf16(const float* in, void* out, int64_t n,
