@@ -242,6 +242,7 @@ class PPOTrainer:
         m.m.copy_(snap["m"])
         m.v.copy_(snap["v"])
         m.params_bf16.copy_(snap["params_bf16"])
+        m._refresh_wt()
         m.adam_ctr.copy_(snap["adam_ctr"])
         m.adam_step = snap["adam_step"]
         self.step_base.copy_(snap["step_base"])
