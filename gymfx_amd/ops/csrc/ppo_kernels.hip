@@ -34,8 +34,12 @@ GFX_DEV __bf16 f2bf(float v) { return (__bf16)v; }
 // ~2 ulp of fp32 — far below the bf16 output rounding.
 GFX_DEV float fast_tanh(float x) {
   x = fminf(fmaxf(x, -15.f), 15.f);
-  const float e = __expf(2.f * x);
-  return (e - 1.f) / (e + 1.f);
+  // raw v_exp_f32 + v_rcp_f32, both single instructions.  An IEEE f32
+  // divide here expands to a ~10-op guarded Newton sequence; unrolled 64x
+  // in the wide epilogue that blew register pressure enough to demote the
+  // MFMA accumulators to scratch for the whole kernel.
+  const float e = __builtin_amdgcn_exp2f(x * 2.885390081777927f);
+  return 1.f - 2.f * __builtin_amdgcn_rcpf(e + 1.f);
 }
 
 // ---------------------------------------------------------------------------
@@ -52,8 +56,12 @@ GFX_DEV float fast_tanh(float x) {
 // Epilogues: ACT 0=none(f32 out) 1=none(bf16) 2=tanh(bf16)
 //            DACT_TANH: multiply by (1 - Y^2) elementwise (dgrad fused tanh')
 // ---------------------------------------------------------------------------
+// min-waves floor 2: with the tanh epilogue the allocator otherwise chases
+// a higher occupancy target and demotes the 64 MFMA accumulators of the
+// NFRAG=8 variant to scratch — 1.2 GB of per-call scratch traffic, 4x
+// slower than just running 2 waves/SIMD with accs in AGPRs.
 template <bool TRANS_B, int ACT, bool DACT_TANH, bool ADD_BIAS, int NFRAG>
-__global__ __launch_bounds__(256) void gemm_kernel(
+__global__ __launch_bounds__(256, 2) void gemm_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     const float* __restrict__ bias, void* __restrict__ C,
     const __bf16* __restrict__ Yact,  // activation output (DACT_TANH)
