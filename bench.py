@@ -45,6 +45,8 @@ def main() -> None:
     ap.add_argument("--n-envs", type=int, default=N_ENVS)
     ap.add_argument("--rollout", type=int, default=ROLLOUT_T)
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--policy", type=str, default="mlp", choices=["mlp", "lstm"],
+                    help="actor-critic architecture (BASELINE config #2 / #4)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -101,6 +103,7 @@ def main() -> None:
         ppo_epochs=4,
         minibatches=8,
         seed=1000,
+        policy=args.policy,
     )
     trainer = PPOTrainer(env, ppo, rank=rank, world_size=world_size, process_group=pg)
 
@@ -150,7 +153,8 @@ def main() -> None:
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "PPO MLP(256,256) actor-critic bf16 MFMA",
+                "model": ("PPO LSTM(256) actor-critic bf16 MFMA" if args.policy == "lstm"
+                          else "PPO MLP(256,256) actor-critic bf16 MFMA"),
                 "global_batch": env_steps_per_update * world_size,
                 "seq_len": 32,
                 "parallelism": f"dp{world_size}",

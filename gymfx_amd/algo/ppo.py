@@ -20,6 +20,7 @@ from typing import Any, Dict, List, Optional
 import torch
 
 from ..envs.vec_env import VecFxEnv
+from ..models.lstm import ActorCriticLSTM
 from ..models.mlp import ActorCriticMLP
 from ..ops import api
 
@@ -41,6 +42,8 @@ class PPOConfig:
     normalize_adv: bool = True
     shuffle_rows: bool = True
     use_graphs: bool = True  # hipGraph-capture the update on GPU
+    policy: str = "mlp"      # "mlp" | "lstm" (BASELINE configs #2 / #4)
+    bptt_len: int = 16       # sequence-chunked BPTT length (lstm)
 
     @classmethod
     def from_config(cls, cfg: Dict[str, Any]) -> "PPOConfig":
@@ -59,6 +62,8 @@ class PPOConfig:
             "hidden": "hidden_size",
             "normalize_adv": "normalize_adv",
             "shuffle_rows": "shuffle_rows",
+            "policy": "policy_model",
+            "bptt_len": "bptt_len",
         }
         for attr, key in mapping.items():
             if cfg.get(key) is not None:
@@ -87,9 +92,19 @@ class PPOTrainer:
 
         # model init is rank-INDEPENDENT (data-parallel replicas must start
         # identical); rollout sampling / shuffling seeds differ per rank.
-        self.model = ActorCriticMLP(
-            D, n_actions, cfg.hidden, device=self.device, seed=cfg.seed
-        )
+        self.recurrent = cfg.policy == "lstm"
+        if self.recurrent:
+            if T % cfg.bptt_len != 0:
+                raise ValueError("rollout_steps must divide by bptt_len")
+            if not cfg.shuffle_rows:
+                raise ValueError("recurrent PPO requires shuffle_rows")
+            self.model = ActorCriticLSTM(
+                D, n_actions, cfg.hidden, device=self.device, seed=cfg.seed
+            )
+        else:
+            self.model = ActorCriticMLP(
+                D, n_actions, cfg.hidden, device=self.device, seed=cfg.seed
+            )
 
         dev = self.device
         self.obs_buf = torch.empty(T, N, D, dtype=torch.bfloat16, device=dev)
@@ -105,15 +120,6 @@ class PPOTrainer:
         self.acts_rollout = self.model.alloc_acts(N)
         self.obs_bf16_step = torch.empty(N, D, dtype=torch.bfloat16, device=dev)
 
-        M = (T * N) // cfg.minibatches
-        if M * cfg.minibatches != T * N:
-            raise ValueError("rollout_steps * n_envs must divide by minibatches")
-        self.mb_rows = M
-        self.acts_train = self.model.alloc_acts(M)
-        self.scratch = self.model.alloc_scratch(M)
-        self.dhead = torch.empty(M, n_actions + 1, dtype=torch.bfloat16, device=dev)
-        self.losses = torch.zeros(5, dtype=torch.float32, device=dev)
-
         # flat views (fixed addresses — required for hipGraph capture)
         TN = T * N
         self.obs_flat = self.obs_buf.view(TN, D)
@@ -121,12 +127,49 @@ class PPOTrainer:
         self.logp_flat = self.logp_buf.view(TN)
         self.adv_flat = self.adv_buf.view(TN)
         self.ret_flat = self.ret_buf.view(TN)
-        # minibatch gather targets
-        self.obs_mb = torch.empty(M, D, dtype=torch.bfloat16, device=dev)
-        self.act_mb = torch.empty(M, dtype=torch.int64, device=dev)
-        self.logp_mb = torch.empty(M, dtype=torch.float32, device=dev)
-        self.adv_mb = torch.empty(M, dtype=torch.float32, device=dev)
-        self.ret_mb = torch.empty(M, dtype=torch.float32, device=dev)
+        self.losses = torch.zeros(5, dtype=torch.float32, device=dev)
+
+        if self.recurrent:
+            H = cfg.hidden
+            L = cfg.bptt_len
+            self.n_chunks = T // L
+            n_seq = self.n_chunks * N
+            if n_seq % cfg.minibatches != 0:
+                raise ValueError("(T/bptt_len)*n_envs must divide by minibatches")
+            Mseq = n_seq // cfg.minibatches
+            self.mseq = Mseq
+            self.mb_rows = L * Mseq  # loss rows per minibatch
+            # persistent rollout recurrent state + chunk-boundary snapshots
+            self.rnn_state = self.model.alloc_state(N)
+            self.h0_buf = torch.zeros(self.n_chunks, N, H, dtype=torch.float32, device=dev)
+            self.c0_buf = torch.zeros(self.n_chunks, N, H, dtype=torch.float32, device=dev)
+            # sequence minibatch buffers (time-major)
+            self.obs_mb_seq = torch.empty(L, Mseq, D, dtype=torch.bfloat16, device=dev)
+            self.act_mb = torch.empty(L, Mseq, dtype=torch.int64, device=dev)
+            self.logp_mb = torch.empty(L, Mseq, dtype=torch.float32, device=dev)
+            self.adv_mb = torch.empty(L, Mseq, dtype=torch.float32, device=dev)
+            self.ret_mb = torch.empty(L, Mseq, dtype=torch.float32, device=dev)
+            self.done_mb = torch.empty(L, Mseq, dtype=torch.bool, device=dev)
+            self.h0_mb = torch.empty(Mseq, H, dtype=torch.bfloat16, device=dev)
+            self.c0_mb = torch.empty(Mseq, H, dtype=torch.float32, device=dev)
+            self.bptt = self.model.alloc_bptt(L, Mseq)
+            self.dhead = torch.empty(L * Mseq, n_actions + 1,
+                                     dtype=torch.bfloat16, device=dev)
+            self._boot_state = self.model.alloc_state(N)
+        else:
+            M = TN // cfg.minibatches
+            if M * cfg.minibatches != TN:
+                raise ValueError("rollout_steps * n_envs must divide by minibatches")
+            self.mb_rows = M
+            self.acts_train = self.model.alloc_acts(M)
+            self.scratch = self.model.alloc_scratch(M)
+            self.dhead = torch.empty(M, n_actions + 1, dtype=torch.bfloat16, device=dev)
+            # minibatch gather targets
+            self.obs_mb = torch.empty(M, D, dtype=torch.bfloat16, device=dev)
+            self.act_mb = torch.empty(M, dtype=torch.int64, device=dev)
+            self.logp_mb = torch.empty(M, dtype=torch.float32, device=dev)
+            self.adv_mb = torch.empty(M, dtype=torch.float32, device=dev)
+            self.ret_mb = torch.empty(M, dtype=torch.float32, device=dev)
 
         # device counters (hipGraph-replayable RNG / schedule state)
         self.step_base = torch.zeros((), dtype=torch.int64, device=dev)
@@ -149,9 +192,20 @@ class PPOTrainer:
     def _rollout_body(self) -> None:
         env, model = self.env, self.model
         T = self.T
+        rec = self.recurrent
+        L = self.cfg.bptt_len if rec else 0
         api.f32_to_bf16(env._obs, self.obs_buf[0])
         for t in range(T):
-            head = model.forward(self.obs_buf[t], self.acts_rollout)
+            if rec:
+                if t % L == 0:
+                    # chunk-boundary recurrent state for BPTT (f32 snapshot)
+                    ch = t // L
+                    self.h0_buf[ch].copy_(self.rnn_state["h"])
+                    self.c0_buf[ch].copy_(self.rnn_state["c"])
+                head = model.step_forward(self.obs_buf[t], self.rnn_state,
+                                          self.acts_rollout)
+            else:
+                head = model.forward(self.obs_buf[t], self.acts_rollout)
             api.sample_head(
                 head, self.sample_seed, t,
                 self.act_buf[t], self.logp_buf[t], self.val_buf[t],
@@ -164,8 +218,19 @@ class PPOTrainer:
                 self.act_buf[t], reward_out=self.rew_buf[t],
                 terminated_out=self.done_buf[t], obs_bf16_out=nxt,
             )
-        # bootstrap value
-        head = model.forward(self.obs_bf16_step, self.acts_rollout)
+            if rec:
+                # fresh episode -> zero recurrent state
+                api.mask_reset(self.rnn_state["h"], self.rnn_state["c"],
+                               self.done_buf[t])
+        # bootstrap value (recurrent: peek one cell step WITHOUT mutating the
+        # persistent state — use scratch state tensors)
+        if rec:
+            self._boot_state["h"].copy_(self.rnn_state["h"])
+            self._boot_state["c"].copy_(self.rnn_state["c"])
+            head = model.step_forward(self.obs_bf16_step, self._boot_state,
+                                      self.acts_rollout)
+        else:
+            head = model.forward(self.obs_bf16_step, self.acts_rollout)
         self.val_buf[T].copy_(head[:, -1])
         api.increment_counter(self.step_base, T)
 
@@ -183,6 +248,29 @@ class PPOTrainer:
         Gradients are fully overwritten by backward (deterministic split-M
         wgrad), so there is no zero_grad."""
         cfg, model = self.cfg, self.model
+        if self.recurrent:
+            api.mb_gather_seq(
+                self.obs_buf, self.act_buf, self.logp_buf, self.adv_buf,
+                self.ret_buf, self.done_buf, self.h0_buf, self.c0_buf,
+                self.obs_mb_seq, self.act_mb, self.logp_mb, self.adv_mb,
+                self.ret_mb, self.done_mb, self.h0_mb, self.c0_mb,
+                L=cfg.bptt_len, seed=self.shuffle_seed,
+                minibatches=cfg.minibatches, step_base=self.step_base,
+                mb_ctr=self.mb_ctr,
+            )
+            head = model.bptt_forward(self.obs_mb_seq, self.done_mb,
+                                      self.h0_mb, self.c0_mb, self.bptt)
+            api.ppo_loss_bwd(
+                head, self.act_mb.view(-1), self.logp_mb.view(-1),
+                self.adv_mb.view(-1), self.ret_mb.view(-1), self.dhead,
+                clip_eps=cfg.clip_eps, ent_coef=cfg.ent_coef,
+                vf_coef=cfg.vf_coef, inv_count=1.0 / self.mb_rows,
+                losses=self.losses,
+            )
+            model.bptt_backward(self.obs_mb_seq, self.done_mb, self.dhead,
+                                self.bptt)
+            api.increment_counter(self.mb_ctr, 1)
+            return
         api.mb_gather(
             self.obs_flat, self.act_flat, self.logp_flat, self.adv_flat,
             self.ret_flat, self.obs_mb, self.act_mb, self.logp_mb,
@@ -234,6 +322,9 @@ class PPOTrainer:
             "obs": self.env._obs.clone(),
             "st": {k: v.clone() for k, v in self.env.st.to_dict().items()},
         }
+        if self.recurrent:
+            snap["rnn_h"] = self.rnn_state["h"].clone()
+            snap["rnn_c"] = self.rnn_state["c"].clone()
         return snap
 
     def _restore(self, snap: Dict[str, Any]) -> None:
@@ -251,6 +342,9 @@ class PPOTrainer:
         st = self.env.st.to_dict()
         for k, v in snap["st"].items():
             st[k].copy_(v)
+        if self.recurrent:
+            self.rnn_state["h"].copy_(snap["rnn_h"])
+            self.rnn_state["c"].copy_(snap["rnn_c"])
 
     def _capture_graphs(self) -> None:
         """Capture rollout+GAE, minibatch fwd/bwd, and optimizer as three

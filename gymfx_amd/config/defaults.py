@@ -67,4 +67,22 @@ DEFAULT_VALUES = {
     "action_space_mode": "discrete",
     "continuous_action_threshold": 0.33,
     "min_equity": None,          # default: initial_cash * 0.01
+
+    # PPO training (mode=training; BASELINE configs #2-#4)
+    "policy_model": "mlp",       # mlp | lstm (recurrent PPO)
+    "hidden_size": 256,
+    "rollout_steps": 128,
+    "ppo_epochs": 4,
+    "minibatches": 8,
+    "bptt_len": 16,              # sequence-chunked BPTT length (lstm)
+    "learning_rate": 3e-4,
+    "gamma": 0.99,
+    "gae_lambda": 0.95,
+    "clip_eps": 0.2,
+    "ent_coef": 0.01,
+    "vf_coef": 0.5,
+    "max_grad_norm": 0.5,
+    "train_updates": 10,
+    "checkpoint_file": None,     # save/resume path (mode=training)
+    "resume": False,
 }

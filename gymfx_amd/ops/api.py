@@ -36,9 +36,11 @@ def gemm(
     trans_b: bool = False,
     act: int = 1,
     dact_tanh: bool = False,
+    accum: bool = False,
 ) -> torch.Tensor:
     if _use_native(A):
-        native.require().gemm(A, B, bias, C, Yact, trans_b, act, dact_tanh)
+        native.require().gemm(A, B, bias, C, Yact, trans_b, act, dact_tanh,
+                              accum)
         return C
     a = A.to(torch.float32)
     b = B.to(torch.float32)
@@ -53,10 +55,98 @@ def gemm(
         y = Yact.to(torch.float32)
         out = out * (1.0 - y * y)
     if act == 0:
-        C.copy_(out)
+        if accum:
+            C.add_(out)
+        else:
+            C.copy_(out)
     else:
         C.copy_(out.to(torch.bfloat16))
     return C
+
+
+def lstm_cell_fwd(
+    gates_pre: torch.Tensor,
+    c_prev: torch.Tensor,
+    c_new: torch.Tensor,
+    h_new: torch.Tensor,
+) -> None:
+    """gates_pre [M,4H] (i|f|g|o pre-activations) -> c_new f32, h_new bf16."""
+    if _use_native(gates_pre):
+        native.require().lstm_cell_fwd(gates_pre, c_prev, c_new, h_new)
+        return
+    M, H4 = gates_pre.shape
+    H = H4 // 4
+    i = torch.sigmoid(gates_pre[:, :H])
+    f = torch.sigmoid(gates_pre[:, H:2 * H])
+    g = torch.tanh(gates_pre[:, 2 * H:3 * H])
+    o = torch.sigmoid(gates_pre[:, 3 * H:])
+    c = f * c_prev + i * g
+    c_new.copy_(c)
+    h_new.copy_((o * torch.tanh(c)).to(torch.bfloat16))
+
+
+def lstm_cell_bwd(
+    gates_pre: torch.Tensor,
+    c_prev: torch.Tensor,
+    c_new: torch.Tensor,
+    dh_head: torch.Tensor,
+    dh_next: Optional[torch.Tensor],
+    dc_next: Optional[torch.Tensor],
+    done: Optional[torch.Tensor],
+    dgates: torch.Tensor,
+    dc_prev: torch.Tensor,
+) -> None:
+    """BPTT cell backward; dh_next/dc_next (grads arriving from step l+1)
+    are masked by `done` so nothing propagates across an episode reset."""
+    if _use_native(gates_pre):
+        native.require().lstm_cell_bwd(gates_pre, c_prev, c_new, dh_head,
+                                       dh_next, dc_next, done, dgates, dc_prev)
+        return
+    M, H4 = gates_pre.shape
+    H = H4 // 4
+    mask = None
+    if done is not None:
+        mask = (~done).to(torch.float32).unsqueeze(1)
+    i = torch.sigmoid(gates_pre[:, :H])
+    f = torch.sigmoid(gates_pre[:, H:2 * H])
+    g = torch.tanh(gates_pre[:, 2 * H:3 * H])
+    o = torch.sigmoid(gates_pre[:, 3 * H:])
+    tc = torch.tanh(c_new)
+    dh = dh_head.clone()
+    if dh_next is not None:
+        dh = dh + (mask * dh_next if mask is not None else dh_next)
+    dc = dh * o * (1 - tc * tc)
+    if dc_next is not None:
+        dc = dc + (mask * dc_next if mask is not None else dc_next)
+    d = torch.cat([
+        dc * g * i * (1 - i),
+        dc * c_prev * f * (1 - f),
+        dc * i * (1 - g * g),
+        dh * tc * o * (1 - o),
+    ], dim=1)
+    dgates.copy_(d.to(torch.bfloat16))
+    dc_prev.copy_(dc * f)
+
+
+def mask_reset(h: torch.Tensor, c: torch.Tensor, done: torch.Tensor) -> None:
+    """Zero the recurrent state of terminated envs (rollout autoreset)."""
+    if _use_native(c):
+        native.require().mask_reset(h, c, done)
+        return
+    h[done] = 0
+    c[done] = 0
+
+
+def masked_state(h_raw: torch.Tensor, c_raw: torch.Tensor,
+                 done: torch.Tensor, h_in: torch.Tensor,
+                 c_in: torch.Tensor) -> None:
+    """h_in/c_in = raw state zeroed where done (BPTT step input)."""
+    if _use_native(c_raw):
+        native.require().masked_state(h_raw, c_raw, done, h_in, c_in)
+        return
+    keep = (~done).unsqueeze(1)
+    h_in.copy_(torch.where(keep, h_raw, torch.zeros_like(h_raw)))
+    c_in.copy_(torch.where(keep, c_raw, torch.zeros_like(c_raw)))
 
 
 def wgrad(
@@ -377,6 +467,65 @@ def mb_gather(
     logp_mb.copy_(logp_src[src])
     adv_mb.copy_(adv_src[src])
     ret_mb.copy_(ret_src[src])
+
+
+def mb_gather_seq(
+    obs_src: torch.Tensor,      # [T, N, D] bf16
+    act_src: torch.Tensor,      # [T, N]
+    logp_src: torch.Tensor,
+    adv_src: torch.Tensor,
+    ret_src: torch.Tensor,
+    done_src: torch.Tensor,     # [T, N] bool
+    h0_src: torch.Tensor,       # [n_chunks, N, H] f32
+    c0_src: torch.Tensor,
+    obs_mb: torch.Tensor,       # [L, Mseq, D] bf16
+    act_mb: torch.Tensor,
+    logp_mb: torch.Tensor,
+    adv_mb: torch.Tensor,
+    ret_mb: torch.Tensor,
+    done_mb: torch.Tensor,      # [L, Mseq] bool
+    h0_mb: torch.Tensor,        # [Mseq, H] bf16
+    c0_mb: torch.Tensor,        # [Mseq, H] f32
+    *,
+    L: int,
+    seed: int,
+    minibatches: int,
+    step_base: torch.Tensor,
+    mb_ctr: torch.Tensor,
+) -> None:
+    """Recurrent-PPO sequence gather: minibatch of (chunk, env) sequences
+    under the Feistel permutation, time-major output + chunk-boundary
+    recurrent state."""
+    n_chunks, N, H = h0_src.shape
+    if _use_native(obs_src):
+        native.require().mb_gather_seq(
+            obs_src.view(-1, obs_src.shape[-1]), act_src.view(-1),
+            logp_src.view(-1), adv_src.view(-1), ret_src.view(-1),
+            done_src.view(-1),
+            h0_src, c0_src, obs_mb.view(-1, obs_mb.shape[-1]),
+            act_mb.view(-1), logp_mb.view(-1), adv_mb.view(-1),
+            ret_mb.view(-1), done_mb.view(-1), h0_mb, c0_mb, L, N, seed,
+            minibatches, step_base, mb_ctr,
+        )
+        return
+    Mseq = c0_mb.shape[0]
+    n_seq = n_chunks * N
+    ctr = int(mb_ctr.item())
+    epoch, mb = ctr // minibatches, ctr % minibatches
+    key = feistel_key(seed, int(step_base.item()), epoch)
+    src = feistel_perm(n_seq, key)[mb * Mseq:(mb + 1) * Mseq]
+    chunk = src // N
+    env = src % N
+    h0_mb.copy_(h0_src[chunk, env].to(torch.bfloat16))
+    c0_mb.copy_(c0_src[chunk, env])
+    for li in range(L):
+        t = chunk * L + li
+        obs_mb[li].copy_(obs_src[t, env])
+        act_mb[li].copy_(act_src[t, env])
+        logp_mb[li].copy_(logp_src[t, env])
+        adv_mb[li].copy_(adv_src[t, env])
+        ret_mb[li].copy_(ret_src[t, env])
+        done_mb[li].copy_(done_src[t, env])
 
 
 def increment_counter(ctr: torch.Tensor, delta: int) -> None:

@@ -18,7 +18,33 @@ void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream);
 void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream);
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
-                 bool dact_tanh, bool add_bias, hipStream_t stream);
+                 bool dact_tanh, bool add_bias, bool accum, hipStream_t stream);
+void launch_lstm_cell_fwd(const float* gates_pre, const float* c_prev,
+                          float* c_new, void* h_new, int64_t M, int H,
+                          hipStream_t stream);
+void launch_lstm_cell_bwd(const float* gates_pre, const float* c_prev,
+                          const float* c_new, const float* dh_head,
+                          const float* dh_next, const float* dc_next,
+                          const bool* done, void* dgates, float* dc_prev,
+                          int64_t M, int H, hipStream_t stream);
+void launch_mask_reset(void* h, float* c, const bool* done, int64_t M, int H,
+                       hipStream_t stream);
+void launch_masked_state(const void* h_raw, const float* c_raw,
+                         const bool* done, void* h_in, float* c_in, int64_t M,
+                         int H, hipStream_t stream);
+void launch_mb_gather_seq(const void* obs_src, const int64_t* act_src,
+                          const float* logp_src, const float* adv_src,
+                          const float* ret_src, const bool* done_src,
+                          const float* h0_src,
+                          const float* c0_src, void* obs_mb, int64_t* act_mb,
+                          float* logp_mb, float* adv_mb, float* ret_mb,
+                          bool* done_mb,
+                          void* h0_mb, float* c0_mb, int Mseq, int L, int D,
+                          int H, int N, uint32_t n_seq, int half,
+                          uint64_t seed, int minibatches,
+                          const unsigned long long* step_base,
+                          const unsigned long long* mb_ctr,
+                          hipStream_t stream);
 void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                   float* dW, float* db, int M, int N, int K, int slabs,
                   hipStream_t stream);
@@ -314,7 +340,9 @@ void check_f32(const torch::Tensor& t, const char* name) {
 
 void gemm_op(torch::Tensor A, torch::Tensor B, c10::optional<torch::Tensor> bias,
              torch::Tensor C, c10::optional<torch::Tensor> Yact, bool trans_b,
-             int64_t act, bool dact_tanh) {
+             int64_t act, bool dact_tanh, bool accum) {
+  TORCH_CHECK(!accum || (act == 0 && !dact_tanh && !bias.has_value()),
+              "accum only supported for plain f32-out gemm");
   check_bf16(A, "A");
   check_bf16(B, "B");
   const int M = (int)A.size(0), K = (int)A.size(1);
@@ -336,7 +364,111 @@ void gemm_op(torch::Tensor A, torch::Tensor B, c10::optional<torch::Tensor> bias
   }
   gymfx::launch_gemm(A.data_ptr(), B.data_ptr(), bias_p, C.data_ptr(), y_p, M,
                      N, K, trans_b, (int)act, dact_tanh, bias_p != nullptr,
-                     cur_stream());
+                     accum, cur_stream());
+}
+
+void lstm_cell_fwd_op(torch::Tensor gates_pre, torch::Tensor c_prev,
+                      torch::Tensor c_new, torch::Tensor h_new) {
+  check_f32(gates_pre, "gates_pre");
+  check_f32(c_prev, "c_prev");
+  check_f32(c_new, "c_new");
+  check_bf16(h_new, "h_new");
+  const int64_t M = c_prev.size(0);
+  const int H = (int)c_prev.size(1);
+  TORCH_CHECK(gates_pre.size(0) == M && gates_pre.size(1) == 4 * H,
+              "gates_pre shape");
+  gymfx::launch_lstm_cell_fwd(gates_pre.data_ptr<float>(),
+                              c_prev.data_ptr<float>(),
+                              c_new.data_ptr<float>(), h_new.data_ptr(), M, H,
+                              cur_stream());
+}
+
+void lstm_cell_bwd_op(torch::Tensor gates_pre, torch::Tensor c_prev,
+                      torch::Tensor c_new, torch::Tensor dh_head,
+                      c10::optional<torch::Tensor> dh_next,
+                      c10::optional<torch::Tensor> dc_next,
+                      c10::optional<torch::Tensor> done,
+                      torch::Tensor dgates, torch::Tensor dc_prev) {
+  check_f32(gates_pre, "gates_pre");
+  check_f32(c_prev, "c_prev");
+  check_f32(c_new, "c_new");
+  check_f32(dh_head, "dh_head");
+  check_bf16(dgates, "dgates");
+  check_f32(dc_prev, "dc_prev");
+  const int64_t M = c_prev.size(0);
+  const int H = (int)c_prev.size(1);
+  const float* dhn = dh_next.has_value() ? dh_next->data_ptr<float>() : nullptr;
+  const float* dcn = dc_next.has_value() ? dc_next->data_ptr<float>() : nullptr;
+  const bool* dn = done.has_value() ? done->data_ptr<bool>() : nullptr;
+  gymfx::launch_lstm_cell_bwd(
+      gates_pre.data_ptr<float>(), c_prev.data_ptr<float>(),
+      c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
+      dgates.data_ptr(), dc_prev.data_ptr<float>(), M, H, cur_stream());
+}
+
+void mask_reset_op(torch::Tensor h, torch::Tensor c, torch::Tensor done) {
+  check_bf16(h, "h");
+  check_f32(c, "c");
+  const int64_t M = c.size(0);
+  const int H = (int)c.size(1);
+  gymfx::launch_mask_reset(h.data_ptr(), c.data_ptr<float>(),
+                           done.data_ptr<bool>(), M, H, cur_stream());
+}
+
+void masked_state_op(torch::Tensor h_raw, torch::Tensor c_raw,
+                     torch::Tensor done, torch::Tensor h_in,
+                     torch::Tensor c_in) {
+  check_bf16(h_raw, "h_raw");
+  check_f32(c_raw, "c_raw");
+  check_bf16(h_in, "h_in");
+  check_f32(c_in, "c_in");
+  const int64_t M = c_raw.size(0);
+  const int H = (int)c_raw.size(1);
+  gymfx::launch_masked_state(h_raw.data_ptr(), c_raw.data_ptr<float>(),
+                             done.data_ptr<bool>(), h_in.data_ptr(),
+                             c_in.data_ptr<float>(), M, H, cur_stream());
+}
+
+void mb_gather_seq_op(torch::Tensor obs_src, torch::Tensor act_src,
+                      torch::Tensor logp_src, torch::Tensor adv_src,
+                      torch::Tensor ret_src, torch::Tensor done_src,
+                      torch::Tensor h0_src,
+                      torch::Tensor c0_src, torch::Tensor obs_mb,
+                      torch::Tensor act_mb, torch::Tensor logp_mb,
+                      torch::Tensor adv_mb, torch::Tensor ret_mb,
+                      torch::Tensor done_mb,
+                      torch::Tensor h0_mb, torch::Tensor c0_mb, int64_t L,
+                      int64_t N, int64_t seed, int64_t minibatches,
+                      torch::Tensor step_base, torch::Tensor mb_ctr) {
+  check_bf16(obs_src, "obs_src");
+  check_bf16(obs_mb, "obs_mb");
+  check_bf16(h0_mb, "h0_mb");
+  check_f32(h0_src, "h0_src");
+  check_f32(c0_src, "c0_src");
+  check_f32(c0_mb, "c0_mb");
+  const int D = (int)obs_src.size(-1);
+  const int H = (int)c0_mb.size(-1);
+  const int Mseq = (int)c0_mb.size(0);
+  const int64_t n_chunks = h0_src.numel() / ((int64_t)N * H);
+  const int64_t n_seq = n_chunks * N;
+  TORCH_CHECK((int64_t)Mseq * minibatches == n_seq,
+              "Mseq * minibatches != n_seq");
+  int bits = 2;
+  while ((1ll << bits) < n_seq) bits += 2;
+  gymfx::launch_mb_gather_seq(
+      obs_src.data_ptr(), act_src.data_ptr<int64_t>(),
+      logp_src.data_ptr<float>(), adv_src.data_ptr<float>(),
+      ret_src.data_ptr<float>(), done_src.data_ptr<bool>(),
+      h0_src.data_ptr<float>(),
+      c0_src.data_ptr<float>(), obs_mb.data_ptr(),
+      act_mb.data_ptr<int64_t>(), logp_mb.data_ptr<float>(),
+      adv_mb.data_ptr<float>(), ret_mb.data_ptr<float>(),
+      done_mb.data_ptr<bool>(), h0_mb.data_ptr(),
+      c0_mb.data_ptr<float>(), Mseq, (int)L, D, H, (int)N, (uint32_t)n_seq,
+      bits / 2, (uint64_t)seed, (int)minibatches,
+      reinterpret_cast<const unsigned long long*>(step_base.data_ptr()),
+      reinterpret_cast<const unsigned long long*>(mb_ctr.data_ptr()),
+      cur_stream());
 }
 
 void wgrad_op(torch::Tensor X, torch::Tensor dY, torch::Tensor dW_part,
@@ -538,10 +670,30 @@ void transpose_bf16_op(torch::Tensor src, torch::Tensor dst) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm", &gemm_op,
         "C = act(A @ B + bias); trans_b reads B as [N,K]; act 0=f32 1=bf16 "
-        "2=tanh-bf16; dact_tanh multiplies by (1-Yact^2)",
+        "2=tanh-bf16; dact_tanh multiplies by (1-Yact^2); accum does C += ",
         py::arg("A"), py::arg("B"), py::arg("bias"), py::arg("C"),
         py::arg("Yact") = py::none(), py::arg("trans_b") = false,
-        py::arg("act") = 1, py::arg("dact_tanh") = false);
+        py::arg("act") = 1, py::arg("dact_tanh") = false,
+        py::arg("accum") = false);
+  m.def("lstm_cell_fwd", &lstm_cell_fwd_op, py::arg("gates_pre"),
+        py::arg("c_prev"), py::arg("c_new"), py::arg("h_new"));
+  m.def("lstm_cell_bwd", &lstm_cell_bwd_op, py::arg("gates_pre"),
+        py::arg("c_prev"), py::arg("c_new"), py::arg("dh_head"),
+        py::arg("dh_next"), py::arg("dc_next"), py::arg("done"),
+        py::arg("dgates"), py::arg("dc_prev"));
+  m.def("mask_reset", &mask_reset_op, py::arg("h"), py::arg("c"),
+        py::arg("done"));
+  m.def("masked_state", &masked_state_op, py::arg("h_raw"), py::arg("c_raw"),
+        py::arg("done"), py::arg("h_in"), py::arg("c_in"));
+  m.def("mb_gather_seq", &mb_gather_seq_op, py::arg("obs_src"),
+        py::arg("act_src"), py::arg("logp_src"), py::arg("adv_src"),
+        py::arg("ret_src"), py::arg("done_src"), py::arg("h0_src"),
+        py::arg("c0_src"),
+        py::arg("obs_mb"), py::arg("act_mb"), py::arg("logp_mb"),
+        py::arg("adv_mb"), py::arg("ret_mb"), py::arg("done_mb"),
+        py::arg("h0_mb"),
+        py::arg("c0_mb"), py::arg("L"), py::arg("N"), py::arg("seed"),
+        py::arg("minibatches"), py::arg("step_base"), py::arg("mb_ctr"));
   m.def("wgrad", &wgrad_op, py::arg("X"), py::arg("dY"), py::arg("dW_part"),
         py::arg("db_part"), py::arg("dW"), py::arg("db"), py::arg("slabs"));
   m.def("gae", &gae_op);

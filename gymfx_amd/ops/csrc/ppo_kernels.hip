@@ -60,7 +60,8 @@ GFX_DEV float fast_tanh(float x) {
 // a higher occupancy target and demotes the 64 MFMA accumulators of the
 // NFRAG=8 variant to scratch — 1.2 GB of per-call scratch traffic, 4x
 // slower than just running 2 waves/SIMD with accs in AGPRs.
-template <bool TRANS_B, int ACT, bool DACT_TANH, bool ADD_BIAS, int NFRAG>
+template <bool TRANS_B, int ACT, bool DACT_TANH, bool ADD_BIAS, int NFRAG,
+          bool ACCUM = false>
 __global__ __launch_bounds__(256, 2) void gemm_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     const float* __restrict__ bias, void* __restrict__ C,
@@ -213,7 +214,8 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
           v *= (1.f - y * y);
         }
         if (ACT == 0) {
-          reinterpret_cast<float*>(C)[(int64_t)grow * N + gcol] = v;
+          float* cp = reinterpret_cast<float*>(C) + (int64_t)grow * N + gcol;
+          *cp = ACCUM ? (*cp + v) : v;
         } else {
           reinterpret_cast<__bf16*>(C)[(int64_t)grow * N + gcol] = f2bf(v);
         }
@@ -739,6 +741,115 @@ __global__ void adv_norm_apply_kernel(float* __restrict__ adv, int64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// LSTM cell, elementwise over [M, H] (the GEMM part — x@Wx + h@Wh + b — runs
+// on the MFMA GEMM above; these kernels do the gate math).
+// Gate layout in gates_pre [M, 4H]: [i | f | g | o].
+//   c_new = sig(f)*c_prev + sig(i)*tanh(g);  h_new = sig(o)*tanh(c_new)
+// fwd keeps gates_pre and c_prev/c_new around so bwd can recompute the
+// activations (cheaper than storing five activation planes).
+// ---------------------------------------------------------------------------
+GFX_DEV float fast_sigmoid(float x) {
+  x = fminf(fmaxf(x, -30.f), 30.f);
+  return __builtin_amdgcn_rcpf(1.f + __builtin_amdgcn_exp2f(-x * 1.4426950408889634f));
+}
+
+__global__ void lstm_cell_fwd_kernel(
+    const float* __restrict__ gates_pre,  // [M, 4H]
+    const float* __restrict__ c_prev,     // [M, H]
+    float* __restrict__ c_new,            // [M, H]
+    __bf16* __restrict__ h_new,           // [M, H] (bf16: feeds next GEMM)
+    int64_t M, int H) {
+  const int64_t total = M * H;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t m = idx / H;
+    const int h = (int)(idx % H);
+    const float* gp = gates_pre + m * 4 * H;
+    const float i = fast_sigmoid(gp[h]);
+    const float f = fast_sigmoid(gp[H + h]);
+    const float g = fast_tanh(gp[2 * H + h]);
+    const float o = fast_sigmoid(gp[3 * H + h]);
+    const float c = f * c_prev[idx] + i * g;
+    c_new[idx] = c;
+    h_new[idx] = f2bf(o * fast_tanh(c));
+  }
+}
+
+// dh_head: gradient from the policy/value head (w.r.t. the RAW h output of
+// this step).  dh_next: gradient arriving from step l+1's recurrent GEMM
+// (w.r.t. the MASKED h input of step l+1) — masked by `done` (the episode
+// boundary AFTER this step) along with dc_next, so no gradient crosses a
+// reset.
+__global__ void lstm_cell_bwd_kernel(
+    const float* __restrict__ gates_pre,  // [M, 4H] (saved fwd)
+    const float* __restrict__ c_prev,     // [M, H] (masked input c)
+    const float* __restrict__ c_new,      // [M, H] (raw output c)
+    const float* __restrict__ dh_head,    // [M, H]
+    const float* __restrict__ dh_next,    // [M, H] or null
+    const float* __restrict__ dc_next,    // [M, H] or null (last step)
+    const bool* __restrict__ done,        // [M] or null
+    __bf16* __restrict__ dgates,          // [M, 4H] out (bf16: feeds wgrad)
+    float* __restrict__ dc_prev,          // [M, H] out
+    int64_t M, int H) {
+  const int64_t total = M * H;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t m = idx / H;
+    const int h = (int)(idx % H);
+    const float mask = (done && done[m]) ? 0.f : 1.f;
+    const float* gp = gates_pre + m * 4 * H;
+    const float i = fast_sigmoid(gp[h]);
+    const float f = fast_sigmoid(gp[H + h]);
+    const float g = fast_tanh(gp[2 * H + h]);
+    const float o = fast_sigmoid(gp[3 * H + h]);
+    const float c = c_new[idx];
+    const float tc = fast_tanh(c);
+    float dhv = dh_head[idx];
+    if (dh_next) dhv += mask * dh_next[idx];
+    float dc = dhv * o * (1.f - tc * tc);
+    if (dc_next) dc += mask * dc_next[idx];
+    __bf16* dg = dgates + m * 4 * H;
+    dg[h] = f2bf(dc * g * i * (1.f - i));
+    dg[H + h] = f2bf(dc * c_prev[idx] * f * (1.f - f));
+    dg[2 * H + h] = f2bf(dc * i * (1.f - g * g));
+    dg[3 * H + h] = f2bf(dhv * tc * o * (1.f - o));
+    dc_prev[idx] = dc * f;
+  }
+}
+
+// zero the recurrent state of terminated envs (rollout path: applied right
+// after the fused env step, so the next policy step starts fresh)
+__global__ void mask_reset_kernel(__bf16* __restrict__ h,
+                                  float* __restrict__ c,
+                                  const bool* __restrict__ done, int64_t M,
+                                  int H) {
+  const int64_t total = M * H;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    if (done[idx / H]) {
+      h[idx] = (__bf16)0.f;
+      c[idx] = 0.f;
+    }
+  }
+}
+
+// masked copy: out = in * (1 - done)  (BPTT fwd: the h/c INPUT of step l+1)
+__global__ void masked_state_kernel(const __bf16* __restrict__ h_raw,
+                                    const float* __restrict__ c_raw,
+                                    const bool* __restrict__ done,
+                                    __bf16* __restrict__ h_in,
+                                    float* __restrict__ c_in, int64_t M,
+                                    int H) {
+  const int64_t total = M * H;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    const bool d = done[idx / H];
+    h_in[idx] = d ? (__bf16)0.f : h_raw[idx];
+    c_in[idx] = d ? 0.f : c_raw[idx];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Device counters (hipGraph support): tiny kernels that advance the RNG /
 // optimizer-step counters inside a captured graph, so one captured update
 // replays with fresh randomness and correct Adam bias correction.
@@ -829,9 +940,145 @@ __global__ __launch_bounds__(256) void mb_gather_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Sequence minibatch gather for recurrent PPO (chunked BPTT): sequences are
+// (chunk, env) pairs; sequence s of this minibatch maps through the Feistel
+// permutation to a source sequence, and all L of its timesteps are gathered
+// so the minibatch tensors are [L, Mseq, ...] (time-major for the BPTT
+// loop).  h0/c0 are the saved chunk-boundary recurrent states.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void mb_gather_seq_kernel(
+    const __bf16* __restrict__ obs_src,   // [T, N, D] flat
+    const int64_t* __restrict__ act_src,  // [T, N]
+    const float* __restrict__ logp_src, const float* __restrict__ adv_src,
+    const float* __restrict__ ret_src,
+    const bool* __restrict__ done_src,    // [T, N]
+    const float* __restrict__ h0_src,     // [n_chunks, N, H] f32
+    const float* __restrict__ c0_src,     // [n_chunks, N, H]
+    __bf16* __restrict__ obs_mb,          // [L, Mseq, D]
+    int64_t* __restrict__ act_mb,         // [L, Mseq]
+    float* __restrict__ logp_mb, float* __restrict__ adv_mb,
+    float* __restrict__ ret_mb,
+    bool* __restrict__ done_mb,           // [L, Mseq]
+    __bf16* __restrict__ h0_mb,           // [Mseq, H] bf16 (GEMM operand)
+    float* __restrict__ c0_mb,            // [Mseq, H]
+    int Mseq, int L, int D, int H, int N, uint32_t n_seq, int half,
+    uint64_t seed, int minibatches,
+    const unsigned long long* __restrict__ step_base,
+    const unsigned long long* __restrict__ mb_ctr) {
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);  // seq within mb
+  const int lane = threadIdx.x & 63;
+  if (row >= Mseq) return;
+  const unsigned long long ctr = *mb_ctr;
+  const uint32_t epoch = (uint32_t)(ctr / (unsigned)minibatches);
+  const uint32_t mb = (uint32_t)(ctr % (unsigned)minibatches);
+  const uint64_t key = splitmix64(seed ^ (*step_base * 0x9E3779B97F4A7C15ull) ^
+                                  ((uint64_t)epoch << 32));
+  const uint32_t dst = mb * (uint32_t)Mseq + (uint32_t)row;
+  const uint32_t src = feistel_perm_idx(dst, n_seq, half, key);
+  const int chunk = (int)(src / (uint32_t)N);
+  const int env = (int)(src % (uint32_t)N);
+  // recurrent initial state
+  {
+    const float* hs = h0_src + ((int64_t)chunk * N + env) * H;
+    const float* cs = c0_src + ((int64_t)chunk * N + env) * H;
+    __bf16* hd = h0_mb + (int64_t)row * H;
+    float* cd = c0_mb + (int64_t)row * H;
+    for (int j = lane; j < H; j += 64) {
+      hd[j] = f2bf(hs[j]);
+      cd[j] = cs[j];
+    }
+  }
+  for (int l = 0; l < L; ++l) {
+    const int64_t t = (int64_t)chunk * L + l;
+    const __bf16* srow = obs_src + (t * N + env) * D;
+    __bf16* drow = obs_mb + ((int64_t)l * Mseq + row) * D;
+    if ((D & 3) == 0) {
+      const int chunks8 = D >> 2;
+      const uint64_t* s64 = reinterpret_cast<const uint64_t*>(srow);
+      uint64_t* d64 = reinterpret_cast<uint64_t*>(drow);
+      for (int c = lane; c < chunks8; c += 64) d64[c] = s64[c];
+    } else {
+      for (int c = lane; c < D; c += 64) drow[c] = srow[c];
+    }
+    if (lane == 0) {
+      const int64_t si = t * N + env;
+      const int64_t di = (int64_t)l * Mseq + row;
+      act_mb[di] = act_src[si];
+      logp_mb[di] = logp_src[si];
+      adv_mb[di] = adv_src[si];
+      ret_mb[di] = ret_src[si];
+      done_mb[di] = done_src[si];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
 static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
+
+void launch_lstm_cell_fwd(const float* gates_pre, const float* c_prev,
+                          float* c_new, void* h_new, int64_t M, int H,
+                          hipStream_t stream) {
+  int64_t total = M * H;
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(lstm_cell_fwd_kernel, dim3(blocks), dim3(256), 0, stream,
+                     gates_pre, c_prev, c_new,
+                     reinterpret_cast<__bf16*>(h_new), M, H);
+}
+
+void launch_lstm_cell_bwd(const float* gates_pre, const float* c_prev,
+                          const float* c_new, const float* dh_head,
+                          const float* dh_next, const float* dc_next,
+                          const bool* done, void* dgates, float* dc_prev,
+                          int64_t M, int H, hipStream_t stream) {
+  int64_t total = M * H;
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(lstm_cell_bwd_kernel, dim3(blocks), dim3(256), 0, stream,
+                     gates_pre, c_prev, c_new, dh_head, dh_next, dc_next, done,
+                     reinterpret_cast<__bf16*>(dgates), dc_prev, M, H);
+}
+
+void launch_mask_reset(void* h, float* c, const bool* done, int64_t M, int H,
+                       hipStream_t stream) {
+  int64_t total = M * H;
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(mask_reset_kernel, dim3(blocks), dim3(256), 0, stream,
+                     reinterpret_cast<__bf16*>(h), c, done, M, H);
+}
+
+void launch_masked_state(const void* h_raw, const float* c_raw,
+                         const bool* done, void* h_in, float* c_in, int64_t M,
+                         int H, hipStream_t stream) {
+  int64_t total = M * H;
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(masked_state_kernel, dim3(blocks), dim3(256), 0, stream,
+                     reinterpret_cast<const __bf16*>(h_raw), c_raw, done,
+                     reinterpret_cast<__bf16*>(h_in), c_in, M, H);
+}
+
+void launch_mb_gather_seq(const void* obs_src, const int64_t* act_src,
+                          const float* logp_src, const float* adv_src,
+                          const float* ret_src, const bool* done_src,
+                          const float* h0_src,
+                          const float* c0_src, void* obs_mb, int64_t* act_mb,
+                          float* logp_mb, float* adv_mb, float* ret_mb,
+                          bool* done_mb,
+                          void* h0_mb, float* c0_mb, int Mseq, int L, int D,
+                          int H, int N, uint32_t n_seq, int half,
+                          uint64_t seed, int minibatches,
+                          const unsigned long long* step_base,
+                          const unsigned long long* mb_ctr,
+                          hipStream_t stream) {
+  hipLaunchKernelGGL(mb_gather_seq_kernel, dim3(ceil_div(Mseq, 4)), dim3(256),
+                     0, stream, reinterpret_cast<const __bf16*>(obs_src),
+                     act_src, logp_src, adv_src, ret_src, done_src, h0_src,
+                     c0_src, reinterpret_cast<__bf16*>(obs_mb), act_mb,
+                     logp_mb, adv_mb, ret_mb, done_mb,
+                     reinterpret_cast<__bf16*>(h0_mb), c0_mb,
+                     Mseq, L, D, H, N, n_seq, half, seed, minibatches,
+                     step_base, mb_ctr);
+}
 
 void launch_increment_u64(unsigned long long* ctr, unsigned long long delta,
                           hipStream_t stream) {
@@ -861,7 +1108,8 @@ void launch_mb_gather(const void* obs_src, const int64_t* act_src,
 
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
-                 bool dact_tanh, bool add_bias, hipStream_t stream) {
+                 bool dact_tanh, bool add_bias, bool accum,
+                 hipStream_t stream) {
   const __bf16* a = reinterpret_cast<const __bf16*>(A);
   const __bf16* b = reinterpret_cast<const __bf16*>(B);
   const __bf16* y = reinterpret_cast<const __bf16*>(Yact);
@@ -880,6 +1128,16 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
       hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 2>), grid, block, 0,   \
                          stream, a, b, bias, C, y, M, N, K);                  \
   } while (0)
+  if (accum) {
+    // C += A@B (f32 out, no bias/activation): the LSTM recurrent GEMM
+    if (trans_b)
+      if (wide) hipLaunchKernelGGL((gemm_kernel<true, 0, false, false, 8, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
+      else hipLaunchKernelGGL((gemm_kernel<true, 0, false, false, 2, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
+    else
+      if (wide) hipLaunchKernelGGL((gemm_kernel<false, 0, false, false, 8, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
+      else hipLaunchKernelGGL((gemm_kernel<false, 0, false, false, 2, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
+    return;
+  }
   if (!trans_b && !dact_tanh) {
     if (act == 0) { if (add_bias) GEMM_LAUNCH(false, 0, false, true); else GEMM_LAUNCH(false, 0, false, false); }
     else if (act == 1) { if (add_bias) GEMM_LAUNCH(false, 1, false, true); else GEMM_LAUNCH(false, 1, false, false); }

@@ -340,3 +340,127 @@ def test_adam_device_step_ctr_matches_host_step():
     # host vs device powf for the bias correction differ in the last ulp
     assert torch.allclose(p1, p2, rtol=1e-6, atol=1e-7)
     assert int(ctr.item()) == 3
+
+
+# ---------------------------------------------------------------------------
+# LSTM kernels (recurrent PPO, BASELINE config #4)
+# ---------------------------------------------------------------------------
+def test_lstm_cell_kernels_match_cpu_oracle():
+    from gymfx_amd.ops import api
+
+    ext = _ext()
+    M, H = 4096, 256
+    g = torch.Generator().manual_seed(5)
+    gates = torch.randn(M, 4 * H, generator=g)
+    c_prev = torch.randn(M, H, generator=g)
+    dh_head = torch.randn(M, H, generator=g)
+    dh_next = torch.randn(M, H, generator=g)
+    dc_next = torch.randn(M, H, generator=g)
+    done = torch.rand(M, generator=g) < 0.1
+    # fwd
+    c_new_c = torch.empty(M, H)
+    h_new_c = torch.empty(M, H, dtype=torch.bfloat16)
+    api.lstm_cell_fwd(gates, c_prev, c_new_c, h_new_c)
+    c_new_g = torch.empty(M, H).cuda()
+    h_new_g = torch.empty(M, H, dtype=torch.bfloat16).cuda()
+    ext.lstm_cell_fwd(gates.cuda(), c_prev.cuda(), c_new_g, h_new_g)
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(c_new_g.cpu().numpy(), c_new_c.numpy(),
+                               rtol=1e-4, atol=1e-5)
+    np.testing.assert_allclose(h_new_g.float().cpu().numpy(),
+                               h_new_c.float().numpy(), rtol=1e-2, atol=1e-2)
+    # bwd (with masking)
+    dg_c = torch.empty(M, 4 * H, dtype=torch.bfloat16)
+    dcp_c = torch.empty(M, H)
+    api.lstm_cell_bwd(gates, c_prev, c_new_c, dh_head, dh_next, dc_next, done,
+                      dg_c, dcp_c)
+    dg_g = torch.empty(M, 4 * H, dtype=torch.bfloat16).cuda()
+    dcp_g = torch.empty(M, H).cuda()
+    ext.lstm_cell_bwd(gates.cuda(), c_prev.cuda(), c_new_g, dh_head.cuda(),
+                      dh_next.cuda(), dc_next.cuda(), done.cuda(), dg_g, dcp_g)
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(dg_g.float().cpu().numpy(),
+                               dg_c.float().numpy(), rtol=2e-2, atol=2e-2)
+    np.testing.assert_allclose(dcp_g.cpu().numpy(), dcp_c.numpy(),
+                               rtol=1e-3, atol=1e-4)
+
+
+def test_mask_ops_match_cpu():
+    from gymfx_amd.ops import api
+
+    ext = _ext()
+    M, H = 1024, 64
+    g = torch.Generator().manual_seed(6)
+    h = torch.randn(M, H, generator=g).to(torch.bfloat16)
+    c = torch.randn(M, H, generator=g)
+    done = torch.rand(M, generator=g) < 0.3
+    hg, cg = h.clone().cuda(), c.clone().cuda()
+    ext.mask_reset(hg, cg, done.cuda())
+    hc, cc = h.clone(), c.clone()
+    api.mask_reset(hc, cc, done)
+    torch.cuda.synchronize()
+    assert torch.equal(hg.cpu(), hc) and torch.equal(cg.cpu(), cc)
+    h_in_g = torch.empty_like(hg)
+    c_in_g = torch.empty_like(cg)
+    ext.masked_state(hg, cg, done.cuda(), h_in_g, c_in_g)
+    h_in_c = torch.empty_like(hc)
+    c_in_c = torch.empty_like(cc)
+    api.masked_state(hc, cc, done, h_in_c, c_in_c)
+    torch.cuda.synchronize()
+    assert torch.equal(h_in_g.cpu(), h_in_c) and torch.equal(c_in_g.cpu(), c_in_c)
+
+
+def test_mb_gather_seq_matches_cpu_oracle():
+    from gymfx_amd.ops import api
+
+    T, N, D, H, L, mbs = 16, 24, 20, 8, 4, 3
+    n_chunks = T // L
+    Mseq = n_chunks * N // mbs
+    g = torch.Generator().manual_seed(7)
+    obs = torch.randn(T, N, D, generator=g).to(torch.bfloat16)
+    act = torch.randint(0, 3, (T, N), generator=g)
+    logp = torch.randn(T, N, generator=g)
+    adv = torch.randn(T, N, generator=g)
+    ret = torch.randn(T, N, generator=g)
+    done = torch.rand(T, N, generator=g) < 0.1
+    h0 = torch.randn(n_chunks, N, H, generator=g)
+    c0 = torch.randn(n_chunks, N, H, generator=g)
+
+    def outs(dev):
+        kw = {"device": dev}
+        return [torch.empty(L, Mseq, D, dtype=torch.bfloat16, **kw),
+                torch.empty(L, Mseq, dtype=torch.int64, **kw),
+                torch.empty(L, Mseq, **kw), torch.empty(L, Mseq, **kw),
+                torch.empty(L, Mseq, **kw),
+                torch.empty(L, Mseq, dtype=torch.bool, **kw),
+                torch.empty(Mseq, H, dtype=torch.bfloat16, **kw),
+                torch.empty(Mseq, H, **kw)]
+
+    for ctr, sb in [(0, 0), (4, 16)]:
+        oc = outs("cpu")
+        api.mb_gather_seq(obs, act, logp, adv, ret, done, h0, c0, *oc,
+                          L=L, seed=3, minibatches=mbs,
+                          step_base=torch.tensor(sb), mb_ctr=torch.tensor(ctr))
+        og = outs("cuda")
+        api.mb_gather_seq(obs.cuda(), act.cuda(), logp.cuda(), adv.cuda(),
+                          ret.cuda(), done.cuda(), h0.cuda(), c0.cuda(), *og,
+                          L=L, seed=3, minibatches=mbs,
+                          step_base=torch.tensor(sb).cuda(),
+                          mb_ctr=torch.tensor(ctr).cuda())
+        torch.cuda.synchronize()
+        for a, b in zip(oc, og):
+            assert torch.equal(a, b.cpu())
+
+
+def test_gemm_accum():
+    ext = _ext()
+    M, N, K = 512, 256, 64
+    A = _rand_bf16(M, K, seed=30)
+    B = _rand_bf16(N, K, seed=31)  # trans_b layout
+    C = torch.randn(M, N, device="cuda")
+    C0 = C.clone()
+    ext.gemm(A, B, None, C, None, True, 0, False, True)
+    torch.cuda.synchronize()
+    ref = C0 + A.float() @ B.float().t()
+    np.testing.assert_allclose(C.cpu().numpy(), ref.cpu().numpy(), rtol=2e-2,
+                               atol=2e-2)

@@ -56,3 +56,41 @@ def test_graph_mode_deterministic_across_trainers():
         t2.train_update(with_stats=False)
     torch.cuda.synchronize()
     assert torch.equal(t1.model.params, t2.model.params)
+
+
+def _make_rec_trainer(use_graphs):
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    md = synthetic_ohlcv(2000, seed=5, vol=4e-4, extra_feature_columns=3)
+    cfg = {
+        "n_envs": 256,
+        "device": "cuda",
+        "window_size": 16,
+        "preprocessor_plugin": "feature_window_preprocessor",
+        "feature_columns": ["OPEN", "HIGH", "LOW", "CLOSE",
+                            "FEAT_0", "FEAT_1", "FEAT_2"],
+        "env_start_mode": "spread",
+        "autoreset": True,
+        "position_size": 1000.0,
+        "seed": 4,
+    }
+    env = build_vec_environment(cfg, md)
+    env.reset(seed=4)
+    pc = PPOConfig(rollout_steps=16, minibatches=4, ppo_epochs=2, seed=4,
+                   policy="lstm", bptt_len=4, use_graphs=use_graphs)
+    return PPOTrainer(env, pc)
+
+
+def test_recurrent_graph_replay_equals_eager():
+    tg = _make_rec_trainer(use_graphs=True)
+    te = _make_rec_trainer(use_graphs=False)
+    for _ in range(3):
+        sg = tg.train_update()
+        se = te.train_update()
+    torch.cuda.synchronize()
+    assert tg._graphs_ready
+    assert torch.equal(tg.model.params, te.model.params)
+    for k in sg:
+        assert sg[k] == pytest.approx(se[k], rel=1e-5, abs=1e-7)
