@@ -182,8 +182,11 @@ class PPOTrainer:
         self.use_graphs = bool(cfg.use_graphs) and self.device.type == "cuda"
         self._graphs_ready = False
         self.g_rollout = None
-        self.g_mb = None
+        self.g_gather = None
+        self.g_fwd_bwd = None
         self.g_opt = None
+        from ..parallel.ddp import GradAllReducer
+        self._reducer = GradAllReducer(world_size, process_group)
 
     # ------------------------------------------------------------------
     # graph-capturable bodies: no host syncs, no allocations, fixed
@@ -242,12 +245,11 @@ class PPOTrainer:
         if self.cfg.normalize_adv:
             api.adv_normalize(self.adv_buf.view(-1), self._adv_part)
 
-    def _mb_body(self) -> None:
-        """One minibatch: Feistel-shuffled gather -> fwd -> loss bwd -> bwd.
-
-        Gradients are fully overwritten by backward (deterministic split-M
-        wgrad), so there is no zero_grad."""
-        cfg, model = self.cfg, self.model
+    def _gather_body(self) -> None:
+        """Feistel-shuffled minibatch gather (parameter-independent: in
+        data-parallel runs this overlaps the previous minibatch's async
+        gradient all-reduce)."""
+        cfg = self.cfg
         if self.recurrent:
             api.mb_gather_seq(
                 self.obs_buf, self.act_buf, self.logp_buf, self.adv_buf,
@@ -258,6 +260,22 @@ class PPOTrainer:
                 minibatches=cfg.minibatches, step_base=self.step_base,
                 mb_ctr=self.mb_ctr,
             )
+        else:
+            api.mb_gather(
+                self.obs_flat, self.act_flat, self.logp_flat, self.adv_flat,
+                self.ret_flat, self.obs_mb, self.act_mb, self.logp_mb,
+                self.adv_mb, self.ret_mb, seed=self.shuffle_seed,
+                minibatches=cfg.minibatches, step_base=self.step_base,
+                mb_ctr=self.mb_ctr,
+            )
+        api.increment_counter(self.mb_ctr, 1)
+
+    def _fwd_bwd_body(self) -> None:
+        """Forward + PPO loss backward + model backward on the gathered
+        minibatch.  Gradients are fully overwritten by backward
+        (deterministic split-M wgrad), so there is no zero_grad."""
+        cfg, model = self.cfg, self.model
+        if self.recurrent:
             head = model.bptt_forward(self.obs_mb_seq, self.done_mb,
                                       self.h0_mb, self.c0_mb, self.bptt)
             api.ppo_loss_bwd(
@@ -269,15 +287,7 @@ class PPOTrainer:
             )
             model.bptt_backward(self.obs_mb_seq, self.done_mb, self.dhead,
                                 self.bptt)
-            api.increment_counter(self.mb_ctr, 1)
             return
-        api.mb_gather(
-            self.obs_flat, self.act_flat, self.logp_flat, self.adv_flat,
-            self.ret_flat, self.obs_mb, self.act_mb, self.logp_mb,
-            self.adv_mb, self.ret_mb, seed=self.shuffle_seed,
-            minibatches=cfg.minibatches, step_base=self.step_base,
-            mb_ctr=self.mb_ctr,
-        )
         head = model.forward(self.obs_mb, self.acts_train)
         api.ppo_loss_bwd(
             head, self.act_mb, self.logp_mb, self.adv_mb, self.ret_mb,
@@ -286,7 +296,10 @@ class PPOTrainer:
             losses=self.losses,
         )
         model.backward(self.obs_mb, self.acts_train, self.dhead, self.scratch)
-        api.increment_counter(self.mb_ctr, 1)
+
+    def _mb_body(self) -> None:
+        self._gather_body()
+        self._fwd_bwd_body()
 
     def _mb_body_noshuffle(self, epoch_mb: int) -> None:
         cfg, model = self.cfg, self.model
@@ -365,9 +378,12 @@ class PPOTrainer:
         with torch.cuda.graph(self.g_rollout):
             self._rollout_body()
             self._gae_body()
-        self.g_mb = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_mb):
-            self._mb_body()
+        self.g_gather = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_gather):
+            self._gather_body()
+        self.g_fwd_bwd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_fwd_bwd):
+            self._fwd_bwd_body()
         self.g_opt = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_opt):
             self._opt_body()
@@ -395,21 +411,39 @@ class PPOTrainer:
         self.mb_ctr.zero_()
         graphs = self.use_graphs and self._graphs_ready
         n_mb = cfg.ppo_epochs * cfg.minibatches
-        for i in range(n_mb):
-            if not cfg.shuffle_rows:
+        if not cfg.shuffle_rows:
+            for i in range(n_mb):
                 self._mb_body_noshuffle(i)
-            elif graphs:
-                self.g_mb.replay()
+                self._allreduce_grads()
+                self._opt_body()
+            self.update_count += 1
+            return self._stats(n_mb) if with_stats else {}
+        # pipelined loop: gather(i+1) overlaps the async all-reduce of
+        # minibatch i's gradients (parallel/ddp.py)
+        if graphs:
+            self.g_gather.replay()
+        else:
+            self._gather_body()
+        for i in range(n_mb):
+            if graphs:
+                self.g_fwd_bwd.replay()
             else:
-                self._mb_body()
-            self._allreduce_grads()
+                self._fwd_bwd_body()
+            self._reducer.start(self.model.grads)
+            if i + 1 < n_mb:
+                if graphs:
+                    self.g_gather.replay()
+                else:
+                    self._gather_body()
+            self._reducer.finish()
             if graphs:
                 self.g_opt.replay()
             else:
                 self._opt_body()
         self.update_count += 1
-        if not with_stats:
-            return {}
+        return self._stats(n_mb) if with_stats else {}
+
+    def _stats(self, n_mb: int) -> Dict[str, float]:
         lv = (self.losses / n_mb).cpu()
         return {
             "pi_loss": float(lv[0]),
@@ -449,6 +483,13 @@ def train_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
     env = build_vec_environment(vec_cfg)
     env.reset(seed=cfg.seed)
     trainer = PPOTrainer(env, cfg)
+    ckpt_path = config.get("checkpoint_file")
+    resumed = False
+    if ckpt_path and config.get("resume"):
+        from ..utils.checkpoint import load_checkpoint
+
+        load_checkpoint(trainer, ckpt_path)
+        resumed = True
     updates = int(config.get("train_updates", 10))
     t0 = time.perf_counter()
     history: List[Dict[str, float]] = []
@@ -458,10 +499,19 @@ def train_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
         if not config.get("quiet_mode") and (u % max(1, updates // 10) == 0):
             print(f"update {u}: {stats}")
     wall = time.perf_counter() - t0
+    if ckpt_path:
+        from ..utils.checkpoint import save_checkpoint
+
+        save_checkpoint(trainer, ckpt_path,
+                        extra={"config": {k: v for k, v in config.items()
+                                          if isinstance(v, (int, float, str, bool, type(None)))}})
     steps = trainer.global_step * env.n_envs
     summary = {
         "mode": "training",
         "updates": updates,
+        "resumed": resumed,
+        "checkpoint_file": ckpt_path,
+        "policy_model": cfg.policy,
         "env_steps": steps,
         "wall_seconds": wall,
         "env_steps_per_sec": steps / wall if wall > 0 else 0.0,
