@@ -47,6 +47,8 @@ def main() -> None:
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--policy", type=str, default="mlp", choices=["mlp", "lstm"],
                     help="actor-critic architecture (BASELINE config #2 / #4)")
+    ap.add_argument("--pairs", type=int, default=1,
+                    help="instruments in the market tensor (BASELINE config #5)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -74,6 +76,7 @@ def main() -> None:
         "data_feed_plugin": "synthetic_data_feed",
         "synthetic_rows": SYN_ROWS,
         "synthetic_seed": 1234,
+        "synthetic_pairs": args.pairs,
         "synthetic_extra_features": 3,
         "synthetic_bar_minutes": 1,
         "instrument": "EUR_USD",
@@ -163,6 +166,7 @@ def main() -> None:
                 "obs_dim": env.obs_dim,
                 "preprocessor": "feature_window_preprocessor",
                 "reward": "pnl_reward",
+                "pairs": args.pairs,
             },
         }))
 

@@ -241,3 +241,49 @@ def uptrend_ohlcv(n_rows: int = 500, *, lo: float = 1.10, hi: float = 1.20) -> M
         timestamps=t0 + np.arange(n_rows, dtype=np.int64) * 60,
         meta={"source": "uptrend"},
     )
+
+
+def concat_markets(mds: "list[MarketData]") -> MarketData:
+    """Concatenate several instruments' series along the bar axis into one
+    MarketData (BASELINE config #5: multi-pair market tensor).
+
+    Per-instrument block boundaries, names and pip sizes land in
+    ``meta["instrument_blocks"]``; the vectorized env assigns each env an
+    instrument block and bounds its episode/observation windows to it
+    (per-env lo_bar/end_bar), so no window or scaling statistic ever crosses
+    an instrument boundary.
+    """
+    if not mds:
+        raise ValueError("concat_markets needs at least one MarketData")
+    if len(mds) == 1:
+        md = mds[0]
+        md.meta.setdefault("instrument_blocks", [{
+            "instrument": md.instrument, "lo": 0, "end": md.n_rows,
+            "pip_size": md.meta.get("pip_size", 0.0001),
+        }])
+        return md
+    keys = set(mds[0].columns)
+    for m in mds[1:]:
+        if set(m.columns) != keys:
+            raise ValueError("all instruments must share the same columns")
+    blocks = []
+    off = 0
+    for m in mds:
+        blocks.append({
+            "instrument": m.instrument, "lo": off, "end": off + m.n_rows,
+            "pip_size": m.meta.get("pip_size", 0.0001),
+        })
+        off += m.n_rows
+    cols = {k: np.concatenate([m.columns[k] for m in mds]) for k in keys}
+    ts = None
+    if all(m.timestamps is not None for m in mds):
+        ts = np.concatenate([m.timestamps for m in mds])
+    ds = None
+    if all(m.date_strings is not None for m in mds):
+        ds = np.concatenate([m.date_strings for m in mds])
+    return MarketData(
+        columns=cols, timestamps=ts, date_strings=ds,
+        instrument="+".join(m.instrument for m in mds),
+        timeframe=mds[0].timeframe,
+        meta={"instrument_blocks": blocks},
+    )

@@ -122,7 +122,9 @@ def step_torch(
     live = ~st.terminated
     # ---- event-context overlay (env.py:394-440): row = bar_index (pre-
     # advance), position = last published --------------------------------
-    row_ov = torch.clamp(st.cursor, max=T - 1).to(torch.int64)
+    eb_i = torch.where(st.end_bar > 0, st.end_bar,
+                       torch.full_like(st.end_bar, T))
+    row_ov = torch.minimum(st.cursor, eb_i - 1).to(torch.int64)
     pos_sign = torch.sign(st.pos).to(torch.int64)
     blocked_entry = torch.zeros(N, dtype=torch.bool, device=device)
     forced_flat = torch.zeros(N, dtype=torch.bool, device=device)
@@ -161,12 +163,12 @@ def step_torch(
     first = live & ~st.started
     adv = live & st.started
     st.cursor = torch.where(adv, st.cursor + 1, st.cursor)
-    exhausted = adv & (st.cursor > T)
-    st.cursor = torch.clamp(st.cursor, max=T)
+    exhausted = adv & (st.cursor > eb_i)
+    st.cursor = torch.minimum(st.cursor, eb_i)
     valid = adv & ~exhausted
     st.terminated = st.terminated | exhausted
 
-    t = torch.clamp(st.cursor - 1, min=0).to(torch.int64)  # current bar row
+    t = torch.maximum(st.cursor - 1, st.lo_bar).to(torch.int64)  # current bar row
     o_px = mt.open[t]
     h_px = mt.high[t]
     l_px = mt.low[t]
@@ -299,8 +301,8 @@ def step_torch(
         st.exec_diag[:, _E["default_orders_submitted"]] += n_orders
 
     elif params.strategy_id == STRATEGY_FIXED_SLTP:
-        sl_d = params.sl_pips * params.pip_size
-        tp_d = params.tp_pips * params.pip_size
+        sl_d = params.sl_pips * st.pip_env
+        tp_d = params.tp_pips * st.pip_env
         flip_l = want_long & (st.pos < 0)
         open_l = want_long & (st.pos <= 0)
         flip_s = want_short & (st.pos > 0)
@@ -428,7 +430,7 @@ def step_torch(
     new_eq = st.cash + st.margin_used + unreal
     st.equity = torch.where(pub, new_eq, st.equity)
     busted = pub & (st.equity <= params.min_equity)
-    data_done = valid & (st.cursor >= T)
+    data_done = valid & (st.cursor >= eb_i)
     st.terminated = st.terminated | busted | data_done
 
     # ---- 6. reward ------------------------------------------------------
@@ -545,9 +547,12 @@ def build_obs_torch(
         out = torch.empty(N, params.obs_dim, dtype=torch.float32, device=device)
 
     step = st.cursor.to(torch.int64)  # bar_index
+    lo = st.lo_bar.to(torch.int64)
+    eb = torch.where(st.end_bar > 0, st.end_bar,
+                     torch.full_like(st.end_bar, T)).to(torch.int64)
     w_idx = torch.arange(W, device=device, dtype=torch.int64).unsqueeze(0)
-    rows = torch.clamp(step.unsqueeze(1) - W + w_idx, min=0)
-    rows = torch.clamp(rows, max=T - 1)
+    rows = torch.maximum(step.unsqueeze(1) - W + w_idx, lo.unsqueeze(1))
+    rows = torch.minimum(rows, (eb - 1).unsqueeze(1))
 
     off = 0
     slices = params.obs_slices()
@@ -560,9 +565,9 @@ def build_obs_torch(
             scaled = feat_win.clone()
         else:
             if params.feature_scaling == "rolling_zscore":
-                hist_left = torch.clamp(step - S, min=0)
+                hist_left = torch.maximum(step - S, lo)
             else:  # expanding
-                hist_left = torch.zeros_like(step)
+                hist_left = lo.clone()
             m = (step - hist_left).to(torch.float64)
             s1 = mt.feat_prefix1[step] - mt.feat_prefix1[hist_left]  # [N, F] f64
             s2 = mt.feat_prefix2[step] - mt.feat_prefix2[hist_left]
@@ -594,11 +599,12 @@ def build_obs_torch(
         out[:, slices["returns"]] = returns
         window_last = prices[:, -1]
     else:
-        window_last = mt.price[torch.clamp(step - 1, min=0, max=T - 1)]
+        window_last = mt.price[
+            torch.minimum(torch.maximum(step - 1, lo), eb - 1)]
 
     if params.include_agent_state:
         ic = params.initial_cash or 1.0
-        t_now = torch.clamp(step - 1, min=0, max=T - 1)
+        t_now = torch.minimum(torch.maximum(step - 1, lo), eb - 1)
         price_now = mt.close[t_now]
         pos_sign = torch.sign(st.pos).to(torch.float32)
         unreal = pos_sign * (price_now - window_last) * params.position_size
@@ -607,18 +613,19 @@ def build_obs_torch(
                 pos_sign,
                 ((st.equity - ic) / ic).to(torch.float32),
                 unreal / ic,
-                torch.clamp((T - step).to(torch.float32), min=0.0) / max(1, T),
+                torch.clamp((eb - step).to(torch.float32), min=0.0)
+                / torch.clamp((eb - lo).to(torch.float32), min=1.0),
             ],
             dim=1,
         )
         out[:, slices["agent_state"]] = agent
 
     if params.stage_b_force_close_obs and mt.force_close is not None:
-        row_fc = torch.clamp(step, max=T - 1)
+        row_fc = torch.minimum(step, eb - 1)
         out[:, slices["force_close"]] = mt.force_close[row_fc]
 
     if params.oanda_fx_calendar_obs and mt.calendar is not None:
-        row_cal = torch.clamp(step, max=T - 1)
+        row_cal = torch.minimum(step, eb - 1)
         cal = mt.calendar[row_cal][:, :9]  # 9 calendar keys in obs
         ic = params.initial_cash or 1.0
         margin_closeout = torch.zeros(N, 1, dtype=torch.float32, device=device)

@@ -98,6 +98,11 @@ class EnvState:
     episode_step: torch.Tensor    # i32 [N] steps since reset
     episode_return: torch.Tensor  # f32 [N]
     start_offset: torch.Tensor    # i32 [N] — reset cursor base
+    # multi-instrument bounds (BASELINE config #5); single-pair: lo=0, end=T
+    lo_bar: torch.Tensor          # i32 [N] — first bar of this env's block
+    end_bar: torch.Tensor         # i32 [N] — one past last bar of the block
+    inst_id: torch.Tensor         # i32 [N] — instrument index (diagnostics)
+    pip_env: torch.Tensor         # f32 [N] — per-env pip size
     # diagnostics
     exec_diag: torch.Tensor       # i32 [N, len(EXEC_COUNTERS)]
     act_diag: torch.Tensor        # i32 [N, len(ACTION_COUNTERS)]
@@ -161,6 +166,10 @@ def alloc_state(params: EnvParams, device: torch.device) -> EnvState:
         episode_step=torch.zeros(N, **i32),
         episode_return=torch.zeros(N, **f64),
         start_offset=torch.zeros(N, **i32),
+        lo_bar=torch.zeros(N, **i32),
+        end_bar=torch.zeros(N, **i32),
+        inst_id=torch.zeros(N, **i32),
+        pip_env=torch.full((N,), params.pip_size, **f32),
         exec_diag=torch.zeros(N, len(EXEC_COUNTERS), **i32),
         act_diag=torch.zeros(N, len(ACTION_COUNTERS), **i32),
         raw_abs_sum=torch.zeros(N, **f32),

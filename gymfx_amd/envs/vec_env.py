@@ -14,7 +14,7 @@ from typing import Any, Dict, Optional
 import numpy as np
 import torch
 
-from ..data.feed import MarketData
+from ..data.feed import MarketData, concat_markets
 from .market import MarketTensors, build_market_tensors
 from .params import EnvParams
 from .reference_step import build_obs_torch, coerce_actions, step_torch
@@ -36,18 +36,27 @@ class VecFxEnv:
     def __init__(
         self,
         config: Dict[str, Any],
-        market_data: MarketData,
+        market_data,
         *,
         device: Any = None,
         use_native: Optional[bool] = None,
     ):
         self.config = dict(config)
+        if isinstance(market_data, (list, tuple)):
+            market_data = concat_markets(list(market_data))
         self.market_data = market_data
         self.device = resolve_device(device if device is not None else config.get("device", "auto"))
         self.params = EnvParams.from_config(
             self.config, timeframe_hours=market_data.timeframe_hours()
         )
-        if market_data.n_rows < self.params.window_size + 2:
+        # per-instrument blocks (multi-pair: BASELINE config #5)
+        self.instrument_blocks = market_data.meta.get(
+            "instrument_blocks",
+            [{"instrument": market_data.instrument, "lo": 0,
+              "end": market_data.n_rows, "pip_size": self.params.pip_size}],
+        )
+        min_rows = min(b["end"] - b["lo"] for b in self.instrument_blocks)
+        if min_rows < self.params.window_size + 2:
             raise ValueError("input data is empty or too short for the configured window")
         self.mt: MarketTensors = build_market_tensors(market_data, self.params, self.device)
         self.st: EnvState = alloc_state(self.params, self.device)
@@ -57,6 +66,7 @@ class VecFxEnv:
         self._obs = torch.empty(
             self.params.n_envs, self.params.obs_dim, dtype=torch.float32, device=self.device
         )
+        self._assign_instrument_blocks()
         self._assign_start_offsets()
         if use_native is None:
             use_native = self.device.type == "cuda"
@@ -79,16 +89,41 @@ class VecFxEnv:
     def obs_dim(self) -> int:
         return self.params.obs_dim
 
+    def _assign_instrument_blocks(self) -> None:
+        """Round-robin envs over instrument blocks; set per-env episode
+        bounds (lo_bar/end_bar) and pip size."""
+        N = self.params.n_envs
+        B = len(self.instrument_blocks)
+        inst = np.arange(N, dtype=np.int32) % B
+        lo = np.array([self.instrument_blocks[i]["lo"] for i in inst], dtype=np.int32)
+        end = np.array([self.instrument_blocks[i]["end"] for i in inst], dtype=np.int32)
+        pip = np.array([self.instrument_blocks[i].get("pip_size", self.params.pip_size)
+                        for i in inst], dtype=np.float32)
+        self.st.inst_id.copy_(torch.from_numpy(inst).to(self.device))
+        self.st.lo_bar.copy_(torch.from_numpy(lo).to(self.device))
+        self.st.end_bar.copy_(torch.from_numpy(end).to(self.device))
+        self.st.pip_env.copy_(torch.from_numpy(pip).to(self.device))
+
     def _assign_start_offsets(self) -> None:
-        N, T, W = self.params.n_envs, self.total_bars, self.params.window_size
+        N, W = self.params.n_envs, self.params.window_size
         mode = self.params.env_start_mode
-        max_off = max(0, T - W - 2)
+        lo = self.st.lo_bar.cpu().numpy().astype(np.int64)
+        end = self.st.end_bar.cpu().numpy().astype(np.int64)
+        max_off = np.maximum(end - W - 2, lo)
         if mode == "spread" and N > 1:
-            off = np.floor(np.linspace(0, max_off, N)).astype(np.int32)
+            # spread each env within its own instrument block
+            frac = np.zeros(N)
+            B = len(self.instrument_blocks)
+            for b in range(B):
+                idx = np.where((np.arange(N) % B) == b)[0]
+                if len(idx) > 1:
+                    frac[idx] = np.linspace(0.0, 1.0, len(idx))
+            off = np.floor(lo + frac * (max_off - lo)).astype(np.int32)
         elif mode == "random":
-            off = self._rng.integers(0, max_off + 1, size=N).astype(np.int32)
+            off = (lo + self._rng.integers(0, 1 << 30, size=N)
+                   % np.maximum(max_off - lo + 1, 1)).astype(np.int32)
         else:
-            off = np.zeros(N, dtype=np.int32)
+            off = lo.astype(np.int32)
         self.st.start_offset = torch.from_numpy(off).to(self.device)
 
     # ------------------------------------------------------------------

@@ -229,6 +229,10 @@ struct GymFxEngine {
     P.episode_step = ptr<int>(state, "episode_step");
     P.episode_return = ptr<double>(state, "episode_return");
     P.start_offset = ptr<int>(state, "start_offset");
+    P.lo_bar = ptr<int>(state, "lo_bar");
+    P.end_bar = ptr<int>(state, "end_bar");
+    P.inst_id = ptr<int>(state, "inst_id");
+    P.pip_env = ptr<float>(state, "pip_env");
     P.exec_diag = ptr<int>(state, "exec_diag");
     P.act_diag = ptr<int>(state, "act_diag");
     P.raw_abs_sum = ptr<float>(state, "raw_abs_sum");

@@ -122,6 +122,10 @@ struct EnvPtrs {
   int *episode_step;
   double *episode_return;
   int *start_offset;
+  const int *lo_bar;         // [N] first bar of env's instrument block
+  const int *end_bar;        // [N] one past last bar
+  const int *inst_id;        // [N]
+  const float *pip_env;      // [N] per-env pip size
   int *exec_diag;            // [N, EXEC_COUNTER_N]
   int *act_diag;             // [N, ACT_COUNTER_N]
   float *raw_abs_sum, *raw_min, *raw_max;

@@ -96,6 +96,9 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   const int n = blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= K.n_envs) return;
   const int T = K.T;
+  // per-env instrument block bounds (multi-pair market; single-pair: 0..T)
+  const int LO = P.lo_bar ? P.lo_bar[n] : 0;
+  const int EB = (P.end_bar && P.end_bar[n] > 0) ? P.end_bar[n] : T;
   const float slip = (float)K.slippage;
   int* ediag = P.exec_diag + (int64_t)n * EXEC_COUNTER_N;
   int* adiag = P.act_diag + (int64_t)n * ACT_COUNTER_N;
@@ -118,7 +121,7 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   // ---- event-context overlay (env.py:394-440) ------------------------
   // row = pre-advance bar_index (the reference reads the upcoming row).
   if (K.flags & F_OVERLAY) {
-    int row_ov = min(P.cursor[n], T - 1);
+    int row_ov = min(P.cursor[n], EB - 1);
     bool active = P.ev_no_trade[row_ov] >= (float)K.overlay_threshold;
     int psign = P.pos[n] > 0 ? 1 : (P.pos[n] < 0 ? -1 : 0);
     if (live && active) {
@@ -154,10 +157,10 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   bool exhausted = false;
   if (adv) {
     P.cursor[n] += 1;
-    if (P.cursor[n] > T) { exhausted = true; P.cursor[n] = T; P.terminated[n] = true; }
+    if (P.cursor[n] > EB) { exhausted = true; P.cursor[n] = EB; P.terminated[n] = true; }
   }
   const bool valid = adv && !exhausted;
-  const int t = max(P.cursor[n] - 1, 0);
+  const int t = max(P.cursor[n] - 1, LO);
   const float o_px = P.open_px[t];
   const float h_px = P.high_px[t];
   const float l_px = P.low_px[t];
@@ -276,8 +279,9 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
   } else if (dec && K.strategy_id == STRAT_FIXED) {
     if (a == 1 || a == 2) {
       const double pos = P.pos[n];
-      const float sl_d = (float)(K.sl_pips * K.pip_size);
-      const float tp_d = (float)(K.tp_pips * K.pip_size);
+      const float pip = P.pip_env ? P.pip_env[n] : (float)K.pip_size;
+      const float sl_d = (float)K.sl_pips * pip;
+      const float tp_d = (float)K.tp_pips * pip;
       if (a == 1) {
         if (pos < 0) P.pend_close[n] = true;
         if (pos <= 0) {
@@ -388,7 +392,7 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
     double unreal = P.pos[n] * ((double)c_px - P.avg_entry[n]);
     P.equity[n] = P.cash[n] + P.margin_used[n] + unreal;
     if (P.equity[n] <= K.min_equity) P.terminated[n] = true;
-    if (valid && P.cursor[n] >= T) P.terminated[n] = true;
+    if (valid && P.cursor[n] >= EB) P.terminated[n] = true;
 
     // ---- 6. reward ---------------------------------------------------
     const double ic = K.initial_cash != 0.0 ? K.initial_cash : 1.0;
@@ -421,7 +425,7 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
     }
 
     if ((K.flags & F_STAGEB_PENALTY) && K.fc_pen_coef > 0 && P.force_close) {
-      int row_fc = min(P.cursor[n], T - 1);
+      int row_fc = min(P.cursor[n], EB - 1);
       float hours = P.force_close[row_fc * 4 + 1];
       bool in_zone = P.force_close[row_fc * 4 + 2] > 0.f;
       bool in_win = hours >= 0.f && hours <= fmaxf(0.f, (float)K.fc_pen_window_hours);
@@ -465,20 +469,23 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K) {
     const int n = (int)(idx / K.obs_dim);
     const int j = (int)(idx % K.obs_dim);
     const int step = P.cursor[n];
+    const int LO = P.lo_bar ? P.lo_bar[n] : 0;
+    const int EB = (P.end_bar && P.end_bar[n] > 0) ? P.end_bar[n] : T;
     float val = 0.f;
 
     if (K.off_features >= 0 && j < K.off_features + W * F && j >= K.off_features) {
       const int q = j - K.off_features;
       const int w = q / F, f = q % F;
       int row = step - W + w;
-      row = max(row, 0);
-      row = min(row, T - 1);
+      row = max(row, LO);
+      row = min(row, EB - 1);
       float x = P.features[(int64_t)row * F + f];
       bool is_binary = P.binary_mask && P.binary_mask[f];
       if (K.scaling_mode == SCALE_NONE || is_binary) {
         val = x;
       } else {
-        int hist_left = (K.scaling_mode == SCALE_ROLLING) ? max(step - K.scale_window, 0) : 0;
+        int hist_left = (K.scaling_mode == SCALE_ROLLING)
+                            ? max(step - K.scale_window, LO) : LO;
         int m = step - hist_left;
         if (m < 2) {
           val = 0.f;
@@ -507,12 +514,12 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K) {
       }
     } else if (K.off_prices >= 0 && j >= K.off_prices && j < K.off_prices + W) {
       const int w = j - K.off_prices;
-      int row = min(max(step - W + w, 0), T - 1);
+      int row = min(max(step - W + w, LO), EB - 1);
       val = P.price_px[row];
     } else if (K.off_returns >= 0 && j >= K.off_returns && j < K.off_returns + W) {
       const int w = j - K.off_returns;
-      int row = min(max(step - W + w, 0), T - 1);
-      int row_prev = min(max(step - W + w - 1, 0), T - 1);
+      int row = min(max(step - W + w, LO), EB - 1);
+      int row_prev = min(max(step - W + w - 1, LO), EB - 1);
       if (w == 0) row_prev = row;
       val = P.price_px[row] - P.price_px[row_prev];
     } else if (K.off_agent >= 0 && j >= K.off_agent && j < K.off_agent + 4) {
@@ -523,22 +530,22 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K) {
       } else if (q == 1) {
         val = (float)((P.equity[n] - ic) / ic);
       } else if (q == 2) {
-        int t_now = min(max(step - 1, 0), T - 1);
-        int row_last = min(max(step - 1, 0), T - 1);
+        int t_now = min(max(step - 1, LO), EB - 1);
+        int row_last = min(max(step - 1, LO), EB - 1);
         float psign = P.pos[n] > 0 ? 1.f : (P.pos[n] < 0 ? -1.f : 0.f);
         float unreal =
             psign * (P.close_px[t_now] - P.price_px[row_last]) * (float)K.position_size;
         val = unreal / (float)ic;
       } else {
-        val = fmaxf((float)(T - step), 0.f) / (float)max(T, 1);
+        val = fmaxf((float)(EB - step), 0.f) / (float)max(EB - LO, 1);
       }
     } else if (K.off_fc >= 0 && j >= K.off_fc && j < K.off_fc + 4) {
       const int q = j - K.off_fc;
-      int row = min(step, T - 1);
+      int row = min(step, EB - 1);
       val = P.force_close ? P.force_close[row * 4 + q] : 0.f;
     } else if (K.off_cal >= 0 && j >= K.off_cal && j < K.off_cal + 11) {
       const int q = j - K.off_cal;
-      int row = min(step, T - 1);
+      int row = min(step, EB - 1);
       if (q < 9) val = P.calendar ? P.calendar[row * 10 + q] : 0.f;
       else if (q == 9) val = 0.f;  // margin_closeout_percent
       else {

@@ -54,16 +54,25 @@ class SyntheticDataFeed(PluginBase):
 
     def load_data(self, config: Dict[str, Any]) -> MarketData:
         p = self._resolve(config)
-        return synthetic_ohlcv(
-            int(p["synthetic_rows"]),
-            seed=int(p["synthetic_seed"]),
-            start_price=float(p["synthetic_start_price"]),
-            vol=float(p["synthetic_vol"]),
-            drift=float(p["synthetic_drift"]),
-            bar_minutes=int(p["synthetic_bar_minutes"]),
-            instrument=str(config.get("instrument", "EUR_USD")),
-            extra_feature_columns=int(p["synthetic_extra_features"]),
-        )
+        pairs = int(config.get("synthetic_pairs", 1) or 1)
+        mds = []
+        for i in range(pairs):
+            name = (str(config.get("instrument", "EUR_USD")) if i == 0
+                    else f"PAIR_{i}")
+            mds.append(synthetic_ohlcv(
+                int(p["synthetic_rows"]),
+                seed=int(p["synthetic_seed"]) + 7919 * i,
+                start_price=float(p["synthetic_start_price"]) * (1.0 + 0.1 * i),
+                vol=float(p["synthetic_vol"]),
+                drift=float(p["synthetic_drift"]),
+                bar_minutes=int(p["synthetic_bar_minutes"]),
+                instrument=name,
+                extra_feature_columns=int(p["synthetic_extra_features"]),
+            ))
+        if len(mds) == 1:
+            return mds[0]
+        from ..data.feed import concat_markets
+        return concat_markets(mds)
 
     def build_market(self, data: MarketData, config: Dict[str, Any]) -> MarketData:
         return data

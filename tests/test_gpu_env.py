@@ -14,8 +14,15 @@ N = 64
 STEPS = 220
 
 
-def _market():
-    return synthetic_ohlcv(2000, seed=5, vol=4e-4, extra_feature_columns=3)
+def _market(pairs=1):
+    if pairs == 1:
+        return synthetic_ohlcv(2000, seed=5, vol=4e-4, extra_feature_columns=3)
+    from gymfx_amd.data.feed import concat_markets
+    return concat_markets([
+        synthetic_ohlcv(700 + 100 * i, seed=5 + i, vol=4e-4,
+                        extra_feature_columns=3, instrument=f"PAIR_{i}")
+        for i in range(pairs)
+    ])
 
 
 BASE = {
@@ -48,13 +55,16 @@ CONFIGS = {
     "autoreset": {"autoreset": True, "strategy_plugin": "direct_fixed_sltp",
                   "sl_pips": 3.0, "tp_pips": 3.0},
     "continuous": {"action_space_mode": "continuous"},
+    "multipair": {"_pairs": 3, "autoreset": True,
+                  "strategy_plugin": "direct_fixed_sltp",
+                  "sl_pips": 6.0, "tp_pips": 9.0},
 }
 
 
 @pytest.mark.parametrize("name", sorted(CONFIGS))
 def test_kernel_matches_torch_oracle(name):
-    md = _market()
     cfg = {**BASE, **CONFIGS[name]}
+    md = _market(cfg.pop("_pairs", 1))
     env_g = build_vec_environment({**cfg, "device": "cuda"}, md, use_native=True)
     env_c = build_vec_environment({**cfg, "device": "cuda"}, md, use_native=False)
     env_g.reset(seed=0)
