@@ -264,22 +264,26 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   f32x4 acc[FK][FN] = {};
   float db_acc = 0.f;
 
-  bf16x8 rx[FK / 2], ry[FN / 2];
-  auto load_chunk = [&](int m0) {
+  // register prefetch ring, PF chunks deep: with only one chunk in flight
+  // the waves parked on vmcnt 68% of their cycles (SQ_WAIT_ANY) — the
+  // per-CU bytes in flight were far below the latency-bandwidth product.
+  constexpr int PF = 4;
+  bf16x8 rx[PF][FK / 2], ry[PF][FN / 2];
+  auto load_chunk = [&](int m0, bf16x8 (&rxs)[FK / 2], bf16x8 (&rys)[FN / 2]) {
     // X chunk [32 m][TK k]: 32*TK elems = 256 threads * (FK/2) vec8
     for (int j = 0; j < FK / 2; ++j) {
       const int c = tid + j * 256;          // chunk id over [32][TK/8]
       const int m = c / (4 * FK);
       const int c8 = (c % (4 * FK)) * 8;
-      rx[j] = bf16x8{};
+      rxs[j] = bf16x8{};
       const int gm = m0 + m;
       if (gm < m_end) {
         const int gk = bk + c8;
         if (gk + 8 <= K) {
-          rx[j] = *reinterpret_cast<const bf16x8*>(&X[(int64_t)gm * K + gk]);
+          rxs[j] = *reinterpret_cast<const bf16x8*>(&X[(int64_t)gm * K + gk]);
         } else {
           for (int i = 0; i < 8; ++i)
-            rx[j][i] = (gk + i < K) ? X[(int64_t)gm * K + gk + i] : (__bf16)0.f;
+            rxs[j][i] = (gk + i < K) ? X[(int64_t)gm * K + gk + i] : (__bf16)0.f;
         }
       }
     }
@@ -287,15 +291,15 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
       const int c = tid + j * 256;
       const int m = c / (4 * FN);
       const int c8 = (c % (4 * FN)) * 8;
-      ry[j] = bf16x8{};
+      rys[j] = bf16x8{};
       const int gm = m0 + m;
       if (gm < m_end) {
         const int gn = bn + c8;
         if (gn + 8 <= N) {
-          ry[j] = *reinterpret_cast<const bf16x8*>(&dY[(int64_t)gm * N + gn]);
+          rys[j] = *reinterpret_cast<const bf16x8*>(&dY[(int64_t)gm * N + gn]);
         } else {
           for (int i = 0; i < 8; ++i)
-            ry[j][i] = (gn + i < N) ? dY[(int64_t)gm * N + gn + i] : (__bf16)0.f;
+            rys[j][i] = (gn + i < N) ? dY[(int64_t)gm * N + gn + i] : (__bf16)0.f;
         }
       }
     }
@@ -306,16 +310,17 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   // lost, SQ_LDS_BANK_CONFLICT).  g ^ ((m>>4)&1)*4 separates them; the
   // staging writes keep their minimum cycle count.
   auto swz = [](int m, int g) { return g ^ (((m >> 4) & 1) << 2); };
-  auto store_chunk = [&](int buf) {
+  auto store_chunk = [&](int buf, bf16x8 (&rxs)[FK / 2],
+                         bf16x8 (&rys)[FN / 2]) {
     for (int j = 0; j < FK / 2; ++j) {
       const int c = tid + j * 256;
       const int m = c / (4 * FK), g = c % (4 * FK);
-      *reinterpret_cast<bf16x8*>(&Xs[buf][m][swz(m, g) * 8]) = rx[j];
+      *reinterpret_cast<bf16x8*>(&Xs[buf][m][swz(m, g) * 8]) = rxs[j];
     }
     for (int j = 0; j < FN / 2; ++j) {
       const int c = tid + j * 256;
       const int m = c / (4 * FN), g = c % (4 * FN);
-      *reinterpret_cast<bf16x8*>(&Ys[buf][m][swz(m, g) * 8]) = ry[j];
+      *reinterpret_cast<bf16x8*>(&Ys[buf][m][swz(m, g) * 8]) = rys[j];
     }
   };
 
@@ -324,14 +329,17 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   const int nchunks = (m_end - m_begin + BKm - 1) / BKm;
 
   if (nchunks > 0) {
-    load_chunk(m_begin);
-    store_chunk(0);
+    for (int d = 0; d < PF && d < nchunks; ++d)
+      load_chunk(m_begin + d * BKm, rx[d % PF], ry[d % PF]);
+    store_chunk(0, rx[0], ry[0]);
     __syncthreads();
   }
   for (int ci = 0; ci < nchunks; ++ci) {
     const int cur = ci & 1;
     const int m0 = m_begin + ci * BKm;
-    if (ci + 1 < nchunks) load_chunk(m0 + BKm);
+    const int pld = ci + PF;
+    if (pld < nchunks)
+      load_chunk(m_begin + pld * BKm, rx[pld % PF], ry[pld % PF]);
 
     if (WANT_DB && tid < TN) {
       float s = 0.f;
@@ -361,7 +369,8 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
         acc[fi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[fi], bf_[ni], acc[fi][ni], 0, 0, 0);
 
-    if (ci + 1 < nchunks) store_chunk(1 - cur);
+    if (ci + 1 < nchunks)
+      store_chunk(1 - cur, rx[(ci + 1) % PF], ry[(ci + 1) % PF]);
     __syncthreads();
   }
 
