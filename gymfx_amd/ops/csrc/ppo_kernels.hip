@@ -329,50 +329,62 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   const int nchunks = (m_end - m_begin + BKm - 1) / BKm;
 
   if (nchunks > 0) {
-    for (int d = 0; d < PF && d < nchunks; ++d)
-      load_chunk(m_begin + d * BKm, rx[d % PF], ry[d % PF]);
+#pragma unroll
+    for (int d = 0; d < PF; ++d)
+      if (d < nchunks)
+        load_chunk(m_begin + d * BKm, rx[d], ry[d]);
     store_chunk(0, rx[0], ry[0]);
     __syncthreads();
   }
-  for (int ci = 0; ci < nchunks; ++ci) {
-    const int cur = ci & 1;
-    const int m0 = m_begin + ci * BKm;
-    const int pld = ci + PF;
-    if (pld < nchunks)
-      load_chunk(m_begin + pld * BKm, rx[pld % PF], ry[pld % PF]);
-
-    if (WANT_DB && tid < TN) {
-      float s = 0.f;
-      for (int i = 0; i < BKm && m0 + i < m_end; ++i) {
-        const int col = tid;
-        s += bf2f(Ys[cur][i][swz(i, col >> 3) * 8 + (col & 7)]);
-      }
-      db_acc += s;
-    }
-    // fragment reads do the transpose: logical A[kout][m] = X[m][kout],
-    // logical B[m][n] = dY[m][n]; reduction index m = kseg*8 + i.
-    bf16x8 af[FK], bf_[FN];
-    for (int fi = 0; fi < FK; ++fi)
-      for (int i = 0; i < 8; ++i) {
-        const int m = kseg * 8 + i;
-        const int col = wr * (16 * FK) + fi * 16 + row_a;
-        af[fi][i] = Xs[cur][m][swz(m, col >> 3) * 8 + (col & 7)];
-      }
-    for (int ni = 0; ni < FN; ++ni)
-      for (int i = 0; i < 8; ++i) {
-        const int m = kseg * 8 + i;
-        const int col = wc * (16 * FN) + ni * 16 + row_a;
-        bf_[ni][i] = Ys[cur][m][swz(m, col >> 3) * 8 + (col & 7)];
-      }
-    for (int fi = 0; fi < FK; ++fi)
-      for (int ni = 0; ni < FN; ++ni)
-        acc[fi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[fi], bf_[ni], acc[fi][ni], 0, 0, 0);
-
-    if (ci + 1 < nchunks)
-      store_chunk(1 - cur, rx[(ci + 1) % PF], ry[(ci + 1) % PF]);
-    __syncthreads();
+  // manually software-pipelined in blocks of PF so every register-ring
+  // slot index is a literal constant: runtime-indexed per-lane arrays
+  // lower to v_movrel chains (measured 13x slower).
+#define WG_STEP(S)                                                            \
+  {                                                                           \
+    const int c_ = cb + (S);                                                  \
+    if (c_ < nchunks) {                                                       \
+      const int cur = c_ & 1;                                                 \
+      const int m0 = m_begin + c_ * BKm;                                      \
+      const int pld = c_ + PF;                                                \
+      if (pld < nchunks)                                                      \
+        load_chunk(m_begin + pld * BKm, rx[(S)], ry[(S)]);                    \
+      if (WANT_DB && tid < TN) {                                              \
+        float s = 0.f;                                                        \
+        for (int i = 0; i < BKm && m0 + i < m_end; ++i) {                     \
+          const int col = tid;                                                \
+          s += bf2f(Ys[cur][i][swz(i, col >> 3) * 8 + (col & 7)]);            \
+        }                                                                     \
+        db_acc += s;                                                          \
+      }                                                                       \
+      bf16x8 af[FK], bf_[FN];                                                 \
+      for (int fi = 0; fi < FK; ++fi)                                         \
+        for (int i = 0; i < 8; ++i) {                                         \
+          const int m = kseg * 8 + i;                                         \
+          const int col = wr * (16 * FK) + fi * 16 + row_a;                   \
+          af[fi][i] = Xs[cur][m][swz(m, col >> 3) * 8 + (col & 7)];           \
+        }                                                                     \
+      for (int ni = 0; ni < FN; ++ni)                                         \
+        for (int i = 0; i < 8; ++i) {                                         \
+          const int m = kseg * 8 + i;                                         \
+          const int col = wc * (16 * FN) + ni * 16 + row_a;                   \
+          bf_[ni][i] = Ys[cur][m][swz(m, col >> 3) * 8 + (col & 7)];          \
+        }                                                                     \
+      for (int fi = 0; fi < FK; ++fi)                                         \
+        for (int ni = 0; ni < FN; ++ni)                                       \
+          acc[fi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(              \
+              af[fi], bf_[ni], acc[fi][ni], 0, 0, 0);                         \
+      if (c_ + 1 < nchunks)                                                   \
+        store_chunk(1 - cur, rx[((S) + 1) % PF], ry[((S) + 1) % PF]);         \
+      __syncthreads();                                                        \
+    }                                                                         \
   }
+  for (int cb = 0; cb < nchunks; cb += PF) {
+    WG_STEP(0)
+    WG_STEP(1)
+    WG_STEP(2)
+    WG_STEP(3)
+  }
+#undef WG_STEP
 
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
