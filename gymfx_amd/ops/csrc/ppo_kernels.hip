@@ -264,26 +264,22 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   f32x4 acc[FK][FN] = {};
   float db_acc = 0.f;
 
-  // register prefetch ring, PF chunks deep: with only one chunk in flight
-  // the waves parked on vmcnt 68% of their cycles (SQ_WAIT_ANY) — the
-  // per-CU bytes in flight were far below the latency-bandwidth product.
-  constexpr int PF = 4;
-  bf16x8 rx[PF][FK / 2], ry[PF][FN / 2];
-  auto load_chunk = [&](int m0, bf16x8 (&rxs)[FK / 2], bf16x8 (&rys)[FN / 2]) {
+  bf16x8 rx[FK / 2], ry[FN / 2];
+  auto load_chunk = [&](int m0) {
     // X chunk [32 m][TK k]: 32*TK elems = 256 threads * (FK/2) vec8
     for (int j = 0; j < FK / 2; ++j) {
       const int c = tid + j * 256;          // chunk id over [32][TK/8]
       const int m = c / (4 * FK);
       const int c8 = (c % (4 * FK)) * 8;
-      rxs[j] = bf16x8{};
+      rx[j] = bf16x8{};
       const int gm = m0 + m;
       if (gm < m_end) {
         const int gk = bk + c8;
         if (gk + 8 <= K) {
-          rxs[j] = *reinterpret_cast<const bf16x8*>(&X[(int64_t)gm * K + gk]);
+          rx[j] = *reinterpret_cast<const bf16x8*>(&X[(int64_t)gm * K + gk]);
         } else {
           for (int i = 0; i < 8; ++i)
-            rxs[j][i] = (gk + i < K) ? X[(int64_t)gm * K + gk + i] : (__bf16)0.f;
+            rx[j][i] = (gk + i < K) ? X[(int64_t)gm * K + gk + i] : (__bf16)0.f;
         }
       }
     }
@@ -291,36 +287,27 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
       const int c = tid + j * 256;
       const int m = c / (4 * FN);
       const int c8 = (c % (4 * FN)) * 8;
-      rys[j] = bf16x8{};
+      ry[j] = bf16x8{};
       const int gm = m0 + m;
       if (gm < m_end) {
         const int gn = bn + c8;
         if (gn + 8 <= N) {
-          rys[j] = *reinterpret_cast<const bf16x8*>(&dY[(int64_t)gm * N + gn]);
+          ry[j] = *reinterpret_cast<const bf16x8*>(&dY[(int64_t)gm * N + gn]);
         } else {
           for (int i = 0; i < 8; ++i)
-            rys[j][i] = (gn + i < N) ? dY[(int64_t)gm * N + gn + i] : (__bf16)0.f;
+            ry[j][i] = (gn + i < N) ? dY[(int64_t)gm * N + gn + i] : (__bf16)0.f;
         }
       }
     }
   };
-  // XOR swizzle of 16B granules by row bit 4: with any 16-byte-aligned row
-  // stride, rows 16 apart alias to the same LDS banks, which made the
-  // transposed fragment reads 2-way conflicted (measured 40% of LDS cycles
-  // lost, SQ_LDS_BANK_CONFLICT).  g ^ ((m>>4)&1)*4 separates them; the
-  // staging writes keep their minimum cycle count.
-  auto swz = [](int m, int g) { return g ^ (((m >> 4) & 1) << 2); };
-  auto store_chunk = [&](int buf, bf16x8 (&rxs)[FK / 2],
-                         bf16x8 (&rys)[FN / 2]) {
+  auto store_chunk = [&](int buf) {
     for (int j = 0; j < FK / 2; ++j) {
       const int c = tid + j * 256;
-      const int m = c / (4 * FK), g = c % (4 * FK);
-      *reinterpret_cast<bf16x8*>(&Xs[buf][m][swz(m, g) * 8]) = rxs[j];
+      *reinterpret_cast<bf16x8*>(&Xs[buf][c / (4 * FK)][(c % (4 * FK)) * 8]) = rx[j];
     }
     for (int j = 0; j < FN / 2; ++j) {
       const int c = tid + j * 256;
-      const int m = c / (4 * FN), g = c % (4 * FN);
-      *reinterpret_cast<bf16x8*>(&Ys[buf][m][swz(m, g) * 8]) = rys[j];
+      *reinterpret_cast<bf16x8*>(&Ys[buf][c / (4 * FN)][(c % (4 * FN)) * 8]) = ry[j];
     }
   };
 
@@ -329,62 +316,38 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   const int nchunks = (m_end - m_begin + BKm - 1) / BKm;
 
   if (nchunks > 0) {
-#pragma unroll
-    for (int d = 0; d < PF; ++d)
-      if (d < nchunks)
-        load_chunk(m_begin + d * BKm, rx[d], ry[d]);
-    store_chunk(0, rx[0], ry[0]);
+    load_chunk(m_begin);
+    store_chunk(0);
     __syncthreads();
   }
-  // manually software-pipelined in blocks of PF so every register-ring
-  // slot index is a literal constant: runtime-indexed per-lane arrays
-  // lower to v_movrel chains (measured 13x slower).
-#define WG_STEP(S)                                                            \
-  {                                                                           \
-    const int c_ = cb + (S);                                                  \
-    if (c_ < nchunks) {                                                       \
-      const int cur = c_ & 1;                                                 \
-      const int m0 = m_begin + c_ * BKm;                                      \
-      const int pld = c_ + PF;                                                \
-      if (pld < nchunks)                                                      \
-        load_chunk(m_begin + pld * BKm, rx[(S)], ry[(S)]);                    \
-      if (WANT_DB && tid < TN) {                                              \
-        float s = 0.f;                                                        \
-        for (int i = 0; i < BKm && m0 + i < m_end; ++i) {                     \
-          const int col = tid;                                                \
-          s += bf2f(Ys[cur][i][swz(i, col >> 3) * 8 + (col & 7)]);            \
-        }                                                                     \
-        db_acc += s;                                                          \
-      }                                                                       \
-      bf16x8 af[FK], bf_[FN];                                                 \
-      for (int fi = 0; fi < FK; ++fi)                                         \
-        for (int i = 0; i < 8; ++i) {                                         \
-          const int m = kseg * 8 + i;                                         \
-          const int col = wr * (16 * FK) + fi * 16 + row_a;                   \
-          af[fi][i] = Xs[cur][m][swz(m, col >> 3) * 8 + (col & 7)];           \
-        }                                                                     \
-      for (int ni = 0; ni < FN; ++ni)                                         \
-        for (int i = 0; i < 8; ++i) {                                         \
-          const int m = kseg * 8 + i;                                         \
-          const int col = wc * (16 * FN) + ni * 16 + row_a;                   \
-          bf_[ni][i] = Ys[cur][m][swz(m, col >> 3) * 8 + (col & 7)];          \
-        }                                                                     \
-      for (int fi = 0; fi < FK; ++fi)                                         \
-        for (int ni = 0; ni < FN; ++ni)                                       \
-          acc[fi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(              \
-              af[fi], bf_[ni], acc[fi][ni], 0, 0, 0);                         \
-      if (c_ + 1 < nchunks)                                                   \
-        store_chunk(1 - cur, rx[((S) + 1) % PF], ry[((S) + 1) % PF]);         \
-      __syncthreads();                                                        \
-    }                                                                         \
+  for (int ci = 0; ci < nchunks; ++ci) {
+    const int cur = ci & 1;
+    const int m0 = m_begin + ci * BKm;
+    if (ci + 1 < nchunks) load_chunk(m0 + BKm);
+
+    if (WANT_DB && tid < TN) {
+      float s = 0.f;
+      for (int i = 0; i < BKm && m0 + i < m_end; ++i)
+        s += bf2f(Ys[cur][i][tid]);
+      db_acc += s;
+    }
+    // fragment reads do the transpose: logical A[kout][m] = X[m][kout],
+    // logical B[m][n] = dY[m][n]; reduction index m = kseg*8 + i.
+    bf16x8 af[FK], bf_[FN];
+    for (int fi = 0; fi < FK; ++fi)
+      for (int i = 0; i < 8; ++i)
+        af[fi][i] = Xs[cur][kseg * 8 + i][wr * (16 * FK) + fi * 16 + row_a];
+    for (int ni = 0; ni < FN; ++ni)
+      for (int i = 0; i < 8; ++i)
+        bf_[ni][i] = Ys[cur][kseg * 8 + i][wc * (16 * FN) + ni * 16 + row_a];
+    for (int fi = 0; fi < FK; ++fi)
+      for (int ni = 0; ni < FN; ++ni)
+        acc[fi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[fi], bf_[ni], acc[fi][ni], 0, 0, 0);
+
+    if (ci + 1 < nchunks) store_chunk(1 - cur);
+    __syncthreads();
   }
-  for (int cb = 0; cb < nchunks; cb += PF) {
-    WG_STEP(0)
-    WG_STEP(1)
-    WG_STEP(2)
-    WG_STEP(3)
-  }
-#undef WG_STEP
 
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
