@@ -117,7 +117,7 @@ class ActorCriticLSTM:
     def _wg_slabs(self, K: int, N: int) -> int:
         tiles = ((K + 63) // 64) * ((N + 63) // 64)
         s = self.wgrad_slabs
-        while tiles * s < 1024:
+        while tiles * s < 2048:
             s *= 2
         return s
 

@@ -117,7 +117,7 @@ class ActorCriticMLP:
         # enough split-M slabs that tiles x slabs >= ~4 blocks per CU
         tiles = ((K + 63) // 64) * ((N + 63) // 64)
         s = self.wgrad_slabs
-        while tiles * s < 1024:
+        while tiles * s < 2048:
             s *= 2
         return s
 
