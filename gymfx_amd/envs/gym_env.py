@@ -148,6 +148,13 @@ class GymFxEnv(spaces.Env):
         self._np_random = np.random.default_rng()
         self._last_info_extras: Dict[str, Any] = {}
         self._was_reset = False
+        # bracket/force-close audit trail (parity:
+        # /root/reference/strategy_plugins/direct_atr_sltp.py:40-50 —
+        # GYMFX_BRACKET_AUDIT=<path> enables append-only JSONL records)
+        import os as _os
+        self._audit_path = _os.environ.get("GYMFX_BRACKET_AUDIT") or \
+            config.get("bracket_audit_file")
+        self._prev_audit_state: Dict[str, Any] = {}
 
     # ------------------------------------------------------------------
     def reset(self, *, seed: Optional[int] = None, options: Optional[Dict[str, Any]] = None):
@@ -169,7 +176,18 @@ class GymFxEnv(spaces.Env):
                 act_t = torch.tensor([int(action)], dtype=torch.int64)
             except (TypeError, ValueError):
                 act_t = torch.tensor([0], dtype=torch.int64)
+        if self._audit_path:
+            st = self.vec.st
+            self._prev_audit_state = {
+                "br_active": bool(st.br_active[0].item()),
+                "pos": float(st.pos[0].item()),
+                "sl_fills": int(st.exec_diag[0, 15].item()),
+                "tp_fills": int(st.exec_diag[0, 16].item()),
+                "session_fc": int(st.exec_diag[0, 14].item()),
+            }
         out = self.vec.step(act_t)
+        if self._audit_path:
+            self._emit_audit()
         reward = float(out["reward"][0].item())
         base_reward = float(out["base_reward"][0].item())
         penalty = float(out["force_close_reward_penalty"][0].item())
@@ -185,6 +203,39 @@ class GymFxEnv(spaces.Env):
             trade_cost=bs["last_trade_cost"],
         )
         return obs, reward, terminated, False, info
+
+    def _emit_audit(self) -> None:
+        """Append JSONL audit records for bracket arms/fills and session
+        force-closes, derived from per-step state deltas."""
+        import json as _json
+
+        st = self.vec.st
+        prev = self._prev_audit_state
+        recs = []
+        bar = int(st.cursor[0].item())
+        now_active = bool(st.br_active[0].item())
+        if now_active and not prev.get("br_active"):
+            recs.append({
+                "event": "bracket_armed", "bar_index": bar,
+                "sl": float(st.br_sl[0].item()),
+                "tp": float(st.br_tp[0].item()),
+                "position": float(st.pos[0].item()),
+            })
+        sl_d = int(st.exec_diag[0, 15].item()) - prev.get("sl_fills", 0)
+        tp_d = int(st.exec_diag[0, 16].item()) - prev.get("tp_fills", 0)
+        fc_d = int(st.exec_diag[0, 14].item()) - prev.get("session_fc", 0)
+        if sl_d > 0:
+            recs.append({"event": "bracket_sl_fill", "bar_index": bar,
+                         "equity": float(st.equity[0].item())})
+        if tp_d > 0:
+            recs.append({"event": "bracket_tp_fill", "bar_index": bar,
+                         "equity": float(st.equity[0].item())})
+        if fc_d > 0:
+            recs.append({"event": "session_force_close", "bar_index": bar})
+        if recs:
+            with open(self._audit_path, "a", encoding="utf-8") as fh:
+                for r in recs:
+                    fh.write(_json.dumps(r) + "\n")
 
     def close(self):
         pass
