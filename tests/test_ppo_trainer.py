@@ -148,3 +148,29 @@ def test_ddp_gloo_ws2_ranks_stay_in_sync():
     for _ in range(2):
         solo.train_update()
     assert results[0] != solo.model.params.numpy().tobytes()
+
+
+def test_ppo_learns_to_go_long_on_uptrend():
+    """End-to-end training efficacy: on a strongly trending market the
+    policy must shift toward long and the mean step reward must rise
+    (behavioral upgrade of the reference's buy_hold>0 smoke invariant,
+    tools/smoke_test.py:108-155)."""
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    md = synthetic_ohlcv(3000, seed=8, vol=1e-4, drift=4e-4)
+    cfg = {"n_envs": 64, "device": "cpu", "window_size": 8,
+           "env_start_mode": "spread", "autoreset": True,
+           "position_size": 1000.0, "seed": 5}
+    env = build_vec_environment(cfg, md)
+    env.reset(seed=5)
+    pc = PPOConfig(rollout_steps=32, minibatches=4, ppo_epochs=4, seed=5,
+                   hidden=32, lr=3e-3, ent_coef=0.003)
+    tr = PPOTrainer(env, pc)
+    tr.train_update()
+    first = float(tr.rew_buf.mean())
+    for _ in range(19):
+        tr.train_update()
+    last = float(tr.rew_buf.mean())
+    long_frac = float((tr.act_buf == 1).float().mean())
+    assert last > max(first * 5, 2e-5), (first, last)
+    assert long_frac > 0.5, long_frac
