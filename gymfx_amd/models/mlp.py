@@ -210,7 +210,10 @@ class ActorCriticMLP:
             "params": self.params.detach().cpu(),
             "m": self.m.detach().cpu(),
             "v": self.v.detach().cpu(),
-            "adam_step": torch.tensor(self.adam_step),
+            # the DEVICE counter is the source of truth: under hipGraph
+            # replay the host mirror self.adam_step only advanced during
+            # capture (checkpoint bug found by the GPU resume test)
+            "adam_step": torch.tensor(int(self.adam_ctr.item())),
             "obs_dim": torch.tensor(self.obs_dim),
             "n_actions": torch.tensor(self.n_actions),
             "hidden": torch.tensor(self.hidden),
