@@ -44,6 +44,7 @@ class PPOConfig:
     use_graphs: bool = True  # hipGraph-capture the update on GPU
     policy: str = "mlp"      # "mlp" | "lstm" (BASELINE configs #2 / #4)
     bptt_len: int = 16       # sequence-chunked BPTT length (lstm)
+    rollout_streams: int = 2  # split rollout across HIP streams (GPU; 1=off)
 
     @classmethod
     def from_config(cls, cfg: Dict[str, Any]) -> "PPOConfig":
@@ -64,6 +65,7 @@ class PPOConfig:
             "shuffle_rows": "shuffle_rows",
             "policy": "policy_model",
             "bptt_len": "bptt_len",
+            "rollout_streams": "rollout_streams",
         }
         for attr, key in mapping.items():
             if cfg.get(key) is not None:
@@ -180,6 +182,15 @@ class PPOTrainer:
         self.global_step = 0   # host mirror of step_base (logging/ckpt)
         self.update_count = 0
         self.use_graphs = bool(cfg.use_graphs) and self.device.type == "cuda"
+        # split rollout: the per-step kernel chain is latency-bound at
+        # N=4096 (halving N barely changes kernel times), so running two
+        # env halves on two HIP streams overlaps the chains.
+        self._split = (self.device.type == "cuda" and cfg.rollout_streams > 1
+                       and N % 2 == 0)
+        if self._split:
+            self._s2 = torch.cuda.Stream()
+            self.acts_half = [self.model.alloc_acts(N // 2),
+                              self.model.alloc_acts(N // 2)]
         self._graphs_ready = False
         self.g_rollout = None
         self.g_gather = None
@@ -192,27 +203,32 @@ class PPOTrainer:
     # graph-capturable bodies: no host syncs, no allocations, fixed
     # tensor addresses, all RNG/schedule state in device counters.
     # ------------------------------------------------------------------
-    def _rollout_body(self) -> None:
+    def _rollout_half(self, lo: int, hi: int, acts) -> None:
+        """The T-step policy/env chain for env rows [lo, hi)."""
         env, model = self.env, self.model
         T = self.T
         rec = self.recurrent
         L = self.cfg.bptt_len if rec else 0
-        api.f32_to_bf16(env._obs, self.obs_buf[0])
+        state = None
+        if rec:
+            state = {"h": self.rnn_state["h"][lo:hi],
+                     "c": self.rnn_state["c"][lo:hi]}
         for t in range(T):
+            obs_t = self.obs_buf[t][lo:hi]
             if rec:
                 if t % L == 0:
                     # chunk-boundary recurrent state for BPTT (f32 snapshot)
                     ch = t // L
-                    self.h0_buf[ch].copy_(self.rnn_state["h"])
-                    self.c0_buf[ch].copy_(self.rnn_state["c"])
-                head = model.step_forward(self.obs_buf[t], self.rnn_state,
-                                          self.acts_rollout)
+                    self.h0_buf[ch][lo:hi].copy_(state["h"])
+                    self.c0_buf[ch][lo:hi].copy_(state["c"])
+                head = model.step_forward(obs_t, state, acts)
             else:
-                head = model.forward(self.obs_buf[t], self.acts_rollout)
+                head = model.forward(obs_t, acts)
             api.sample_head(
                 head, self.sample_seed, t,
-                self.act_buf[t], self.logp_buf[t], self.val_buf[t],
-                step_base=self.step_base,
+                self.act_buf[t][lo:hi], self.logp_buf[t][lo:hi],
+                self.val_buf[t][lo:hi], step_base=self.step_base,
+                row_offset=lo,
             )
             # fused env kernel writes reward/done into the rollout slabs and
             # the NEXT observation (bf16) into obs_buf[t+1] directly
@@ -220,14 +236,33 @@ class PPOTrainer:
             env.step(
                 self.act_buf[t], reward_out=self.rew_buf[t],
                 terminated_out=self.done_buf[t], obs_bf16_out=nxt,
+                env_lo=lo, env_hi=hi,
             )
             if rec:
                 # fresh episode -> zero recurrent state
-                api.mask_reset(self.rnn_state["h"], self.rnn_state["c"],
-                               self.done_buf[t])
+                api.mask_reset(state["h"], state["c"],
+                               self.done_buf[t][lo:hi])
+
+    def _rollout_body(self) -> None:
+        env, model = self.env, self.model
+        T = self.T
+        api.f32_to_bf16(env._obs, self.obs_buf[0])
+        if self._split:
+            half = self.N // 2
+            ev_fork = torch.cuda.Event()
+            ev_join = torch.cuda.Event()
+            ev_fork.record()
+            with torch.cuda.stream(self._s2):
+                self._s2.wait_event(ev_fork)
+                self._rollout_half(half, self.N, self.acts_half[1])
+                ev_join.record()
+            self._rollout_half(0, half, self.acts_half[0])
+            torch.cuda.current_stream().wait_event(ev_join)
+        else:
+            self._rollout_half(0, self.N, self.acts_rollout)
         # bootstrap value (recurrent: peek one cell step WITHOUT mutating the
         # persistent state — use scratch state tensors)
-        if rec:
+        if self.recurrent:
             self._boot_state["h"].copy_(self.rnn_state["h"])
             self._boot_state["c"].copy_(self.rnn_state["c"])
             head = model.step_forward(self.obs_bf16_step, self._boot_state,

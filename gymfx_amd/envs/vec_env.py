@@ -146,7 +146,8 @@ class VecFxEnv:
     def step(self, actions: torch.Tensor, *,
              reward_out: Optional[torch.Tensor] = None,
              terminated_out: Optional[torch.Tensor] = None,
-             obs_bf16_out: Optional[torch.Tensor] = None) -> Dict[str, torch.Tensor]:
+             obs_bf16_out: Optional[torch.Tensor] = None,
+             env_lo: int = 0, env_hi: int = 0) -> Dict[str, torch.Tensor]:
         """Advance all envs. Returns dict with obs/reward/terminated tensors.
 
         reward_out / terminated_out / obs_bf16_out: optional preallocated
@@ -158,9 +159,12 @@ class VecFxEnv:
         if self._native is not None:
             # fused HIP path: step + autoreset + obs in two kernel launches
             info = self._native.step(actions, self._obs, reward_out,
-                                     terminated_out, obs_bf16_out)
+                                     terminated_out, obs_bf16_out,
+                                     env_lo, env_hi)
             info["obs"] = self._obs
             return info
+        if env_lo != 0 or env_hi not in (0, self.params.n_envs):
+            raise ValueError("env range stepping requires the native engine")
         info = step_torch(self.st, self.mt, self.params, actions)
         terminated = info["terminated"]
         if self.params.autoreset:

@@ -258,14 +258,17 @@ def sample_head(
     entropy: Optional[torch.Tensor] = None,
     greedy: bool = False,
     step_base: Optional[torch.Tensor] = None,
+    row_offset: int = 0,
 ) -> None:
     """head [M, A+1] f32 (logits | value) -> categorical sample + logp.
 
     step_base (device u64/i64 scalar): added to ``step`` inside the kernel so
-    a captured hipGraph replays with fresh randomness."""
+    a captured hipGraph replays with fresh randomness.  row_offset: global
+    row index of row 0 (split launches draw the same RNG stream as one
+    full-width launch)."""
     if _use_native(head):
         native.require().sample_head(head, seed, step, actions, logp, value,
-                                     entropy, greedy, step_base)
+                                     entropy, greedy, step_base, row_offset)
         return
     if step_base is not None:
         step = step + int(step_base.item())
@@ -279,7 +282,7 @@ def sample_head(
         a = logits.argmax(dim=1)
     else:
         # same counter-based RNG as the kernel (splitmix64)
-        u = _splitmix_uniform(seed, step, M)
+        u = _splitmix_uniform(seed, step, M, row_offset)
         cdf = pi.cumsum(dim=1)
         a = (u.unsqueeze(1) >= cdf).sum(dim=1).clamp(max=A - 1)
     actions.copy_(a)
@@ -290,8 +293,9 @@ def sample_head(
         entropy.copy_(-(pi * logpi).sum(dim=1))
 
 
-def _splitmix_uniform(seed: int, step: int, M: int) -> torch.Tensor:
-    m = torch.arange(M, dtype=torch.int64)
+def _splitmix_uniform(seed: int, step: int, M: int,
+                      row_offset: int = 0) -> torch.Tensor:
+    m = torch.arange(row_offset, row_offset + M, dtype=torch.int64)
     x = (seed ^ (step * 0x51E1F5 + m * 0x9E37)) & 0xFFFFFFFFFFFFFFFF
 
     def mix(x):

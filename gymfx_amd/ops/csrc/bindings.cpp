@@ -14,8 +14,10 @@
 
 namespace gymfx {
 
-void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream);
-void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream);
+void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
+                     int env_cnt, hipStream_t stream);
+void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
+                      int env_cnt, hipStream_t stream);
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
                  bool dact_tanh, bool add_bias, bool accum, hipStream_t stream);
@@ -60,7 +62,7 @@ void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
 void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,
                         uint64_t step, int64_t* actions, float* logp,
                         float* value, float* entropy, int greedy,
-                        const unsigned long long* step_base,
+                        const unsigned long long* step_base, int row_offset,
                         hipStream_t stream);
 void launch_increment_u64(unsigned long long* ctr, unsigned long long delta,
                           hipStream_t stream);
@@ -259,10 +261,14 @@ struct GymFxEngine {
   py::dict step(torch::Tensor actions, torch::Tensor obs_out,
                 c10::optional<torch::Tensor> reward_out,
                 c10::optional<torch::Tensor> terminated_out,
-                c10::optional<torch::Tensor> obs_bf16_out) {
+                c10::optional<torch::Tensor> obs_bf16_out,
+                int64_t env_lo, int64_t env_hi) {
     TORCH_CHECK(actions.is_cuda() == state["cursor"].is_cuda(),
                 "actions must live on the env device");
     TORCH_CHECK(actions.numel() == K.n_envs, "actions numel != n_envs");
+    if (env_hi <= 0) env_hi = K.n_envs;
+    TORCH_CHECK(0 <= env_lo && env_lo < env_hi && env_hi <= K.n_envs,
+                "bad env range");
     TORCH_CHECK(actions.is_contiguous(), "actions must be contiguous");
     if (continuous) {
       TORCH_CHECK(actions.scalar_type() == torch::kFloat32,
@@ -293,8 +299,8 @@ struct GymFxEngine {
     P.reward_out = rew.data_ptr<float>();
     P.terminated_out = term.data_ptr<bool>();
     hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
-    launch_env_step(P, K, stream);
-    build_obs(obs_out, obs_bf16_out);
+    launch_env_step(P, K, (int)env_lo, (int)(env_hi - env_lo), stream);
+    build_obs(obs_out, obs_bf16_out, env_lo, env_hi);
     py::dict out;
     out["reward"] = rew;
     out["base_reward"] = outputs["base_reward"];
@@ -305,7 +311,9 @@ struct GymFxEngine {
   }
 
   void build_obs(torch::Tensor obs_out,
-                 c10::optional<torch::Tensor> obs_bf16_out) {
+                 c10::optional<torch::Tensor> obs_bf16_out,
+                 int64_t env_lo = 0, int64_t env_hi = 0) {
+    if (env_hi <= 0) env_hi = K.n_envs;
     TORCH_CHECK(obs_out.is_contiguous() && obs_out.scalar_type() == torch::kFloat32,
                 "obs_out must be contiguous f32");
     TORCH_CHECK(obs_out.numel() == (int64_t)K.n_envs * K.obs_dim, "obs_out shape");
@@ -319,7 +327,7 @@ struct GymFxEngine {
       P.obs_bf16_out = obs_bf16_out->data_ptr();
     }
     hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
-    launch_build_obs(P, K, stream);
+    launch_build_obs(P, K, (int)env_lo, (int)(env_hi - env_lo), stream);
   }
 };
 
@@ -559,7 +567,8 @@ void sample_head_op(torch::Tensor head, int64_t seed, int64_t step,
                     torch::Tensor actions, torch::Tensor logp,
                     c10::optional<torch::Tensor> value,
                     c10::optional<torch::Tensor> entropy, bool greedy,
-                    c10::optional<torch::Tensor> step_base) {
+                    c10::optional<torch::Tensor> step_base,
+                    int64_t row_offset) {
   check_f32(head, "head");
   const int M = (int)head.size(0);
   const int n_actions = (int)head.size(1) - 1;
@@ -576,7 +585,7 @@ void sample_head_op(torch::Tensor head, int64_t seed, int64_t step,
       actions.data_ptr<int64_t>(), logp.data_ptr<float>(),
       value.has_value() ? value->data_ptr<float>() : nullptr,
       entropy.has_value() ? entropy->data_ptr<float>() : nullptr,
-      greedy ? 1 : 0, sb, cur_stream());
+      greedy ? 1 : 0, sb, (int)row_offset, cur_stream());
 }
 
 void increment_counter_op(torch::Tensor ctr, int64_t delta) {
@@ -709,7 +718,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_head", &sample_head_op, py::arg("head"), py::arg("seed"),
         py::arg("step"), py::arg("actions"), py::arg("logp"),
         py::arg("value") = py::none(), py::arg("entropy") = py::none(),
-        py::arg("greedy") = false, py::arg("step_base") = py::none());
+        py::arg("greedy") = false, py::arg("step_base") = py::none(),
+        py::arg("row_offset") = 0);
   m.def("increment_counter", &increment_counter_op, py::arg("ctr"),
         py::arg("delta"));
   m.def("mb_gather", &mb_gather_op, py::arg("obs_src"), py::arg("act_src"),
@@ -729,9 +739,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("step", &gymfx::GymFxEngine::step, py::arg("actions"),
            py::arg("obs_out"), py::arg("reward_out") = py::none(),
            py::arg("terminated_out") = py::none(),
-           py::arg("obs_bf16_out") = py::none())
+           py::arg("obs_bf16_out") = py::none(), py::arg("env_lo") = 0,
+           py::arg("env_hi") = 0)
       .def("build_obs", &gymfx::GymFxEngine::build_obs, py::arg("obs_out"),
-           py::arg("obs_bf16_out") = py::none());
+           py::arg("obs_bf16_out") = py::none(), py::arg("env_lo") = 0,
+           py::arg("env_hi") = 0);
   m.attr("EXEC_COUNTER_N") = (int)gymfx::EXEC_COUNTER_N;
   m.attr("ACT_COUNTER_N") = (int)gymfx::ACT_COUNTER_N;
 }

@@ -92,9 +92,10 @@ GFX_DEV void reset_env(const EnvPtrs& P, const EnvParamsK& K, int n) {
   P.raw_max[n] = -INFINITY;
 }
 
-__global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
-  const int n = blockIdx.x * blockDim.x + threadIdx.x;
-  if (n >= K.n_envs) return;
+__global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
+                                const int env_lo, const int env_cnt) {
+  const int n = env_lo + blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= env_lo + env_cnt) return;
   const int T = K.T;
   // per-env instrument block bounds (multi-pair market; single-pair: 0..T)
   const int LO = P.lo_bar ? P.lo_bar[n] : 0;
@@ -461,12 +462,13 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K) {
 // Semantics: envs/reference_step.py build_obs_torch (itself mirroring
 // default_preprocessor.py:34-77 / feature_window_preprocessor.py:99-191).
 // ---------------------------------------------------------------------------
-__global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K) {
-  const int64_t total = (int64_t)K.n_envs * K.obs_dim;
+__global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K,
+                                 const int env_lo, const int env_cnt) {
+  const int64_t total = (int64_t)env_cnt * K.obs_dim;
   const int T = K.T, W = K.window, F = K.n_features;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (int64_t)gridDim.x * blockDim.x) {
-    const int n = (int)(idx / K.obs_dim);
+    const int n = env_lo + (int)(idx / K.obs_dim);
     const int j = (int)(idx % K.obs_dim);
     const int step = P.cursor[n];
     const int LO = P.lo_bar ? P.lo_bar[n] : 0;
@@ -560,19 +562,23 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K) {
   }
 }
 
-void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream) {
+void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
+                     int env_cnt, hipStream_t stream) {
   const int block = 256;
-  const int grid = (K.n_envs + block - 1) / block;
-  hipLaunchKernelGGL(env_step_kernel, dim3(grid), dim3(block), 0, stream, P, K);
+  const int grid = (env_cnt + block - 1) / block;
+  hipLaunchKernelGGL(env_step_kernel, dim3(grid), dim3(block), 0, stream, P, K,
+                     env_lo, env_cnt);
 }
 
-void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, hipStream_t stream) {
+void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
+                      int env_cnt, hipStream_t stream) {
   const int block = 256;
-  int64_t total = (int64_t)K.n_envs * K.obs_dim;
+  int64_t total = (int64_t)env_cnt * K.obs_dim;
   int64_t blocks = (total + block - 1) / block;
   int grid = (int)(blocks < 2048 ? blocks : 2048);
   if (grid < 1) grid = 1;
-  hipLaunchKernelGGL(build_obs_kernel, dim3(grid), dim3(block), 0, stream, P, K);
+  hipLaunchKernelGGL(build_obs_kernel, dim3(grid), dim3(block), 0, stream, P, K,
+                     env_lo, env_cnt);
 }
 
 }  // namespace gymfx

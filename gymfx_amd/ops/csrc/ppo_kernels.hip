@@ -585,10 +585,12 @@ __global__ void sample_head_kernel(const float* __restrict__ head, int M,
                                    float* __restrict__ value,
                                    float* __restrict__ entropy,
                                    int greedy,
-                                   const unsigned long long* __restrict__ step_base) {
+                                   const unsigned long long* __restrict__ step_base,
+                                   int row_offset) {
   const int m = blockIdx.x * blockDim.x + threadIdx.x;
   if (m >= M) return;
   if (step_base) step += *step_base;  // device counter: hipGraph-replayable RNG
+  const int64_t mg = m + row_offset;  // global row: split launches keep RNG
   const float* row = head + (int64_t)m * (n_actions + 1);
   float mx = row[0];
   for (int j = 1; j < n_actions; ++j) mx = fmaxf(mx, row[j]);
@@ -602,7 +604,7 @@ __global__ void sample_head_kernel(const float* __restrict__ head, int M,
     for (int j = 1; j < n_actions; ++j)
       if (row[j] > best) { best = row[j]; a = j; }
   } else {
-    const uint64_t r = splitmix64(seed ^ (step * 0x51E1F5ull + (uint64_t)m * 0x9E37ull));
+    const uint64_t r = splitmix64(seed ^ (step * 0x51E1F5ull + (uint64_t)mg * 0x9E37ull));
     float u = (float)((r >> 11) * (1.0 / 9007199254740992.0));  // [0,1)
     u = fminf(u, 0.999999f);
     float c = 0.f;
@@ -1213,11 +1215,11 @@ void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
 void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,
                         uint64_t step, int64_t* actions, float* logp,
                         float* value, float* entropy, int greedy,
-                        const unsigned long long* step_base,
+                        const unsigned long long* step_base, int row_offset,
                         hipStream_t stream) {
   hipLaunchKernelGGL(sample_head_kernel, dim3(ceil_div(M, 256)), dim3(256), 0,
                      stream, head, M, n_actions, seed, step, actions, logp,
-                     value, entropy, greedy, step_base);
+                     value, entropy, greedy, step_base, row_offset);
 }
 
 void launch_ppo_loss_bwd(const float* head, const int64_t* actions,

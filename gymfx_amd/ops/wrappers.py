@@ -166,13 +166,15 @@ class NativeEngine:
 
     def step(self, actions: torch.Tensor, obs_out: torch.Tensor,
              reward_out: torch.Tensor = None, terminated_out: torch.Tensor = None,
-             obs_bf16_out: torch.Tensor = None) -> Dict[str, torch.Tensor]:
+             obs_bf16_out: torch.Tensor = None, env_lo: int = 0,
+             env_hi: int = 0) -> Dict[str, torch.Tensor]:
         if self._params.action_space_mode == "continuous":
             actions = actions.to(torch.float32).contiguous()
         else:
             actions = actions.to(torch.int64).contiguous()
         return dict(self._engine.step(actions, obs_out, reward_out,
-                                      terminated_out, obs_bf16_out))
+                                      terminated_out, obs_bf16_out,
+                                      env_lo, env_hi))
 
     def build_obs(self, obs_out: torch.Tensor,
                   obs_bf16_out: torch.Tensor = None) -> None:

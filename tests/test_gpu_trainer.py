@@ -114,3 +114,36 @@ def test_checkpoint_resume_bit_identical_on_gpu():
     torch.cuda.synchronize()
     assert torch.equal(t_full.model.params, t_res.model.params)
     assert torch.equal(t_full.env.st.equity, t_res.env.st.equity)
+
+
+def test_split_rollout_equals_single_stream():
+    """Two-stream split rollout must be bit-identical to one stream (row
+    blocks are tile-aligned; the sampler draws by global row index)."""
+    from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    def make(streams):
+        md = synthetic_ohlcv(2000, seed=5, vol=4e-4, extra_feature_columns=3)
+        cfg = {"n_envs": 256, "device": "cuda", "window_size": 16,
+               "preprocessor_plugin": "feature_window_preprocessor",
+               "feature_columns": ["OPEN", "HIGH", "LOW", "CLOSE",
+                                   "FEAT_0", "FEAT_1", "FEAT_2"],
+               "env_start_mode": "spread", "autoreset": True,
+               "position_size": 1000.0, "seed": 9}
+        env = build_vec_environment(cfg, md)
+        env.reset(seed=9)
+        pc = PPOConfig(rollout_steps=16, minibatches=4, ppo_epochs=2, seed=9,
+                       rollout_streams=streams)
+        return PPOTrainer(env, pc)
+
+    t1 = make(1)
+    t2 = make(2)
+    for _ in range(2):
+        t1.train_update(with_stats=False)
+        t2.train_update(with_stats=False)
+    torch.cuda.synchronize()
+    assert t2._split
+    assert torch.equal(t1.act_buf, t2.act_buf)
+    assert torch.equal(t1.model.params, t2.model.params)
+    assert torch.equal(t1.env.st.equity, t2.env.st.equity)
