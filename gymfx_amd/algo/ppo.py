@@ -44,7 +44,10 @@ class PPOConfig:
     use_graphs: bool = True  # hipGraph-capture the update on GPU
     policy: str = "mlp"      # "mlp" | "lstm" (BASELINE configs #2 / #4)
     bptt_len: int = 16       # sequence-chunked BPTT length (lstm)
-    rollout_streams: int = 2  # split rollout across HIP streams (GPU; 1=off)
+    rollout_streams: int = 1  # >1: split rollout across HIP streams (GPU).
+                             # Measured neutral-to-slightly-negative at
+                             # N=4096 (per-kernel latency does not shrink
+                             # with width) — kept for bigger fleets.
 
     @classmethod
     def from_config(cls, cfg: Dict[str, Any]) -> "PPOConfig":

@@ -564,7 +564,9 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K,
 
 void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
                      int env_cnt, hipStream_t stream) {
-  const int block = 256;
+  // 64-thread blocks: the kernel is a per-env dependent-load latency chain;
+  // smaller blocks spread the waves over 4x more CUs.
+  const int block = 64;
   const int grid = (env_cnt + block - 1) / block;
   hipLaunchKernelGGL(env_step_kernel, dim3(grid), dim3(block), 0, stream, P, K,
                      env_lo, env_cnt);
@@ -575,7 +577,7 @@ void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
   const int block = 256;
   int64_t total = (int64_t)env_cnt * K.obs_dim;
   int64_t blocks = (total + block - 1) / block;
-  int grid = (int)(blocks < 2048 ? blocks : 2048);
+  int grid = (int)(blocks < 16384 ? blocks : 16384);
   if (grid < 1) grid = 1;
   hipLaunchKernelGGL(build_obs_kernel, dim3(grid), dim3(block), 0, stream, P, K,
                      env_lo, env_cnt);
