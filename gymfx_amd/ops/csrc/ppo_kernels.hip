@@ -1161,17 +1161,27 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                   hipStream_t stream) {
   const __bf16* x = reinterpret_cast<const __bf16*>(X);
   const __bf16* dy = reinterpret_cast<const __bf16*>(dY);
-  // 64x64 output tiles: 128x128 halves the HBM re-streaming but drops the
-  // grid to ~1 block/CU at these layer shapes (measured slower).  The
-  // caller raises `slabs` instead when K/64 * N/64 is small so the grid
-  // still fills the 256 CUs.
-  dim3 grid(ceil_div(K, 64), ceil_div(N, 64), slabs);
+  // Tile choice: 128x128 halves the HBM re-streaming of X/dY (which hits
+  // the roofline for the LSTM's K~260/N=1024 shapes) but needs the grid to
+  // still fill the 256 CUs; otherwise 64x64 (the MLP's 260x256 shapes
+  // measured slower at 128x128 because the grid collapsed to ~1 block/CU).
+  const bool big = K >= 128 && N >= 128 &&
+                   (int64_t)ceil_div(K, 128) * ceil_div(N, 128) * slabs >= 1024;
+  dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);
   if (db_part) {
-    hipLaunchKernelGGL((wgrad_partial_kernel<true, 2, 2>), grid, dim3(256), 0,
-                       stream, x, dy, dW_part, db_part, M, N, K, slabs);
+    if (big)
+      hipLaunchKernelGGL((wgrad_partial_kernel<true, 4, 4>), grid, dim3(256),
+                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs);
+    else
+      hipLaunchKernelGGL((wgrad_partial_kernel<true, 2, 2>), grid, dim3(256),
+                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs);
   } else {
-    hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), grid, dim3(256), 0,
-                       stream, x, dy, dW_part, db_part, M, N, K, slabs);
+    if (big)
+      hipLaunchKernelGGL((wgrad_partial_kernel<false, 4, 4>), grid, dim3(256),
+                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs);
+    else
+      hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), grid, dim3(256),
+                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs);
   }
   int64_t elems = (int64_t)K * N;
   if (elems <= 4096) {
