@@ -77,17 +77,8 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
   // !TRANS_B fragment reads transpose out of LDS instead.
   constexpr int BS_ROWS = TRANS_B ? BN : BK;
   constexpr int BS_LD = (TRANS_B ? BK : BN) + 8;
-  // one shared block: the K-loop staging buffers alias an output-staging
-  // image used by the wide epilogue (per-half-tile, f32 padded rows)
-  constexpr int AS_BYTES = 2 * BM * LDA * 2;
-  constexpr int BS_BYTES = 2 * BS_ROWS * BS_LD * 2;
-  constexpr int EP_LD = BN + 4;              // f32 elems; row pad kills
-  constexpr int EP_BYTES = 32 * EP_LD * 4;   // 4-way write conflicts
-  constexpr int SMEM_BYTES =
-      (AS_BYTES + BS_BYTES) > EP_BYTES ? (AS_BYTES + BS_BYTES) : EP_BYTES;
-  __shared__ __align__(16) char smem[SMEM_BYTES];
-  auto As = reinterpret_cast<__bf16(*)[BM][LDA]>(smem);
-  auto Bs = reinterpret_cast<__bf16(*)[BS_ROWS][BS_LD]>(smem + AS_BYTES);
+  __shared__ __bf16 As[2][BM][LDA];
+  __shared__ __bf16 Bs[2][BS_ROWS][BS_LD];
 
   const int bm = blockIdx.x * BM;
   const int bn = blockIdx.y * BN;
@@ -208,50 +199,6 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
   // ---- epilogue -------------------------------------------------------
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
-  if constexpr (NFRAG == 8 && !ACCUM) {
-    // Stage the output tile through LDS (reusing the K-loop buffers) so the
-    // global stores are full coalesced lines: the fragment layout's native
-    // stores are 16-column (64 B f32 / 32 B bf16) chunks — measured write-
-    // limited on the [65536, 1024] f32 gate projections.
-    float* ep = reinterpret_cast<float*>(smem);
-    for (int half = 0; half < 2; ++half) {   // output rows [half*32, +32)
-      __syncthreads();
-      if (wr == half) {
-        for (int mi = 0; mi < 2; ++mi)
-          for (int ni = 0; ni < NFRAG; ++ni) {
-            const int lcol = wc * (16 * NFRAG) + ni * 16 + ccol;
-            const int gcol = bn + lcol;
-            for (int r = 0; r < 4; ++r) {
-              const int lrow = mi * 16 + crow_base + r;
-              const int grow = bm + half * 32 + lrow;
-              float v = acc[mi][ni][r];
-              if (ADD_BIAS) v += bias[gcol < N ? gcol : 0];
-              if (ACT == 2) v = fast_tanh(v);
-              if (DACT_TANH && grow < M && gcol < N) {
-                float y = bf2f(Yact[(int64_t)grow * N + gcol]);
-                v *= (1.f - y * y);
-              }
-              ep[lrow * EP_LD + lcol] = v;
-            }
-          }
-      }
-      __syncthreads();
-      for (int idx = tid; idx < 32 * BN; idx += 256) {
-        const int lrow = idx / BN;
-        const int lcol = idx % BN;
-        const int grow = bm + half * 32 + lrow;
-        const int gcol = bn + lcol;
-        if (grow >= M || gcol >= N) continue;
-        const float v = ep[lrow * EP_LD + lcol];
-        if (ACT == 0) {
-          reinterpret_cast<float*>(C)[(int64_t)grow * N + gcol] = v;
-        } else {
-          reinterpret_cast<__bf16*>(C)[(int64_t)grow * N + gcol] = f2bf(v);
-        }
-      }
-    }
-    return;
-  }
   for (int mi = 0; mi < 2; ++mi) {
     for (int ni = 0; ni < NFRAG; ++ni) {
       const int gcol = bn + wc * (16 * NFRAG) + ni * 16 + ccol;
