@@ -142,6 +142,7 @@ class ActorCriticLSTM:
         dev, H = self.device, self.hidden
         return {
             "gates": torch.empty(M, 4 * H, dtype=torch.float32, device=dev),
+            "gates_h": torch.empty(M, 4 * H, dtype=torch.bfloat16, device=dev),
             "head": torch.empty(M, self.head_dim, dtype=torch.float32, device=dev),
         }
 
@@ -152,8 +153,8 @@ class ActorCriticLSTM:
         head [M, A+1] f32."""
         gates = acts["gates"]
         api.gemm(obs_bf16, self.wt("Wx"), self.f32("b"), gates, act=0, trans_b=True)
-        api.gemm(state["h"], self.wt("Wh"), None, gates, act=0, trans_b=True, accum=True)
-        api.lstm_cell_fwd(gates, state["c"], state["c"], state["h"])
+        api.gemm(state["h"], self.wt("Wh"), None, acts["gates_h"], act=1, trans_b=True)
+        api.lstm_cell_fwd(gates, acts["gates_h"], state["c"], state["c"], state["h"])
         api.gemm(state["h"], self.wt("Wy"), self.f32("by"), acts["head"], act=0, trans_b=True)
         return acts["head"]
 
@@ -169,6 +170,7 @@ class ActorCriticLSTM:
             "h_raw": torch.empty(L, M, H, dtype=torch.bfloat16, device=dev),
             "c_raw": torch.empty(L, M, H, dtype=torch.float32, device=dev),
             "gates": torch.empty(L, M, 4 * H, dtype=torch.float32, device=dev),
+            "gates_h": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
             "head": torch.empty(L * M, self.head_dim, dtype=torch.float32, device=dev),
             "dgates": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
             "dh_all": torch.empty(L, M, H, dtype=torch.float32, device=dev),
@@ -192,9 +194,11 @@ class ActorCriticLSTM:
         buf["c_in"][0].copy_(c0)
         for l in range(L):
             g = buf["gates"][l]
-            api.gemm(buf["h_in"][l], self.wt("Wh"), None, g, act=0,
-                     trans_b=True, accum=True)
-            api.lstm_cell_fwd(g, buf["c_in"][l], buf["c_raw"][l], buf["h_raw"][l])
+            gh = buf["gates_h"][l]
+            api.gemm(buf["h_in"][l], self.wt("Wh"), None, gh, act=1,
+                     trans_b=True)
+            api.lstm_cell_fwd(g, gh, buf["c_in"][l], buf["c_raw"][l],
+                              buf["h_raw"][l])
             # masked state feeds step l+1 (zero across episode resets)
             api.masked_state(buf["h_raw"][l], buf["c_raw"][l], done_seq[l],
                              buf["h_in"][l + 1], buf["c_in"][l + 1])
@@ -224,7 +228,8 @@ class ActorCriticLSTM:
                 api.gemm(buf["dgates"][l + 1], self.w("Wh"), None, dh_next,
                          act=0, trans_b=True)
             dc_out = dc_bufs[l & 1]
-            api.lstm_cell_bwd(buf["gates"][l], buf["c_in"][l], buf["c_raw"][l],
+            api.lstm_cell_bwd(buf["gates"][l], buf["gates_h"][l],
+                              buf["c_in"][l], buf["c_raw"][l],
                               buf["dh_all"][l], dh_next, dc_next,
                               done_seq[l], buf["dgates"][l], dc_out)
             dc_next = dc_out

@@ -66,20 +66,26 @@ def gemm(
 
 def lstm_cell_fwd(
     gates_pre: torch.Tensor,
+    gates_h: "Optional[torch.Tensor]",
     c_prev: torch.Tensor,
     c_new: torch.Tensor,
     h_new: torch.Tensor,
 ) -> None:
-    """gates_pre [M,4H] (i|f|g|o pre-activations) -> c_new f32, h_new bf16."""
+    """gate pre-activations = gates_pre (f32, x@Wx + b) [+ gates_h (bf16,
+    h@Wh kept separate so the recurrent GEMM writes 2-byte outputs instead
+    of read-modify-writing the f32 buffer)] -> c_new f32, h_new bf16."""
     if _use_native(gates_pre):
-        native.require().lstm_cell_fwd(gates_pre, c_prev, c_new, h_new)
+        native.require().lstm_cell_fwd(gates_pre, gates_h, c_prev, c_new, h_new)
         return
-    M, H4 = gates_pre.shape
+    gp = gates_pre
+    if gates_h is not None:
+        gp = gp + gates_h.to(torch.float32)
+    M, H4 = gp.shape
     H = H4 // 4
-    i = torch.sigmoid(gates_pre[:, :H])
-    f = torch.sigmoid(gates_pre[:, H:2 * H])
-    g = torch.tanh(gates_pre[:, 2 * H:3 * H])
-    o = torch.sigmoid(gates_pre[:, 3 * H:])
+    i = torch.sigmoid(gp[:, :H])
+    f = torch.sigmoid(gp[:, H:2 * H])
+    g = torch.tanh(gp[:, 2 * H:3 * H])
+    o = torch.sigmoid(gp[:, 3 * H:])
     c = f * c_prev + i * g
     c_new.copy_(c)
     h_new.copy_((o * torch.tanh(c)).to(torch.bfloat16))
@@ -87,6 +93,7 @@ def lstm_cell_fwd(
 
 def lstm_cell_bwd(
     gates_pre: torch.Tensor,
+    gates_h: "Optional[torch.Tensor]",
     c_prev: torch.Tensor,
     c_new: torch.Tensor,
     dh_head: torch.Tensor,
@@ -99,18 +106,22 @@ def lstm_cell_bwd(
     """BPTT cell backward; dh_next/dc_next (grads arriving from step l+1)
     are masked by `done` so nothing propagates across an episode reset."""
     if _use_native(gates_pre):
-        native.require().lstm_cell_bwd(gates_pre, c_prev, c_new, dh_head,
-                                       dh_next, dc_next, done, dgates, dc_prev)
+        native.require().lstm_cell_bwd(gates_pre, gates_h, c_prev, c_new,
+                                       dh_head, dh_next, dc_next, done,
+                                       dgates, dc_prev)
         return
-    M, H4 = gates_pre.shape
+    gp = gates_pre
+    if gates_h is not None:
+        gp = gp + gates_h.to(torch.float32)
+    M, H4 = gp.shape
     H = H4 // 4
     mask = None
     if done is not None:
         mask = (~done).to(torch.float32).unsqueeze(1)
-    i = torch.sigmoid(gates_pre[:, :H])
-    f = torch.sigmoid(gates_pre[:, H:2 * H])
-    g = torch.tanh(gates_pre[:, 2 * H:3 * H])
-    o = torch.sigmoid(gates_pre[:, 3 * H:])
+    i = torch.sigmoid(gp[:, :H])
+    f = torch.sigmoid(gp[:, H:2 * H])
+    g = torch.tanh(gp[:, 2 * H:3 * H])
+    o = torch.sigmoid(gp[:, 3 * H:])
     tc = torch.tanh(c_new)
     dh = dh_head.clone()
     if dh_next is not None:

@@ -21,10 +21,11 @@ void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
                  bool dact_tanh, bool add_bias, bool accum, hipStream_t stream);
-void launch_lstm_cell_fwd(const float* gates_pre, const float* c_prev,
-                          float* c_new, void* h_new, int64_t M, int H,
-                          hipStream_t stream);
-void launch_lstm_cell_bwd(const float* gates_pre, const float* c_prev,
+void launch_lstm_cell_fwd(const float* gates_pre, const void* gates_h,
+                          const float* c_prev, float* c_new, void* h_new,
+                          int64_t M, int H, hipStream_t stream);
+void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
+                          const float* c_prev,
                           const float* c_new, const float* dh_head,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
@@ -379,7 +380,9 @@ void gemm_op(torch::Tensor A, torch::Tensor B, c10::optional<torch::Tensor> bias
                      accum, cur_stream());
 }
 
-void lstm_cell_fwd_op(torch::Tensor gates_pre, torch::Tensor c_prev,
+void lstm_cell_fwd_op(torch::Tensor gates_pre,
+                      c10::optional<torch::Tensor> gates_h,
+                      torch::Tensor c_prev,
                       torch::Tensor c_new, torch::Tensor h_new) {
   check_f32(gates_pre, "gates_pre");
   check_f32(c_prev, "c_prev");
@@ -389,13 +392,20 @@ void lstm_cell_fwd_op(torch::Tensor gates_pre, torch::Tensor c_prev,
   const int H = (int)c_prev.size(1);
   TORCH_CHECK(gates_pre.size(0) == M && gates_pre.size(1) == 4 * H,
               "gates_pre shape");
-  gymfx::launch_lstm_cell_fwd(gates_pre.data_ptr<float>(),
+  const void* gh = nullptr;
+  if (gates_h.has_value()) {
+    check_bf16(*gates_h, "gates_h");
+    gh = gates_h->data_ptr();
+  }
+  gymfx::launch_lstm_cell_fwd(gates_pre.data_ptr<float>(), gh,
                               c_prev.data_ptr<float>(),
                               c_new.data_ptr<float>(), h_new.data_ptr(), M, H,
                               cur_stream());
 }
 
-void lstm_cell_bwd_op(torch::Tensor gates_pre, torch::Tensor c_prev,
+void lstm_cell_bwd_op(torch::Tensor gates_pre,
+                      c10::optional<torch::Tensor> gates_h,
+                      torch::Tensor c_prev,
                       torch::Tensor c_new, torch::Tensor dh_head,
                       c10::optional<torch::Tensor> dh_next,
                       c10::optional<torch::Tensor> dc_next,
@@ -412,8 +422,13 @@ void lstm_cell_bwd_op(torch::Tensor gates_pre, torch::Tensor c_prev,
   const float* dhn = dh_next.has_value() ? dh_next->data_ptr<float>() : nullptr;
   const float* dcn = dc_next.has_value() ? dc_next->data_ptr<float>() : nullptr;
   const bool* dn = done.has_value() ? done->data_ptr<bool>() : nullptr;
+  const void* gh = nullptr;
+  if (gates_h.has_value()) {
+    check_bf16(*gates_h, "gates_h");
+    gh = gates_h->data_ptr();
+  }
   gymfx::launch_lstm_cell_bwd(
-      gates_pre.data_ptr<float>(), c_prev.data_ptr<float>(),
+      gates_pre.data_ptr<float>(), gh, c_prev.data_ptr<float>(),
       c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
       dgates.data_ptr(), dc_prev.data_ptr<float>(), M, H, cur_stream());
 }
@@ -689,8 +704,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("act") = 1, py::arg("dact_tanh") = false,
         py::arg("accum") = false);
   m.def("lstm_cell_fwd", &lstm_cell_fwd_op, py::arg("gates_pre"),
-        py::arg("c_prev"), py::arg("c_new"), py::arg("h_new"));
+        py::arg("gates_h"), py::arg("c_prev"), py::arg("c_new"),
+        py::arg("h_new"));
   m.def("lstm_cell_bwd", &lstm_cell_bwd_op, py::arg("gates_pre"),
+        py::arg("gates_h"),
         py::arg("c_prev"), py::arg("c_new"), py::arg("dh_head"),
         py::arg("dh_next"), py::arg("dc_next"), py::arg("done"),
         py::arg("dgates"), py::arg("dc_prev"));

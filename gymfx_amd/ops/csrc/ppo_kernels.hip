@@ -755,8 +755,19 @@ GFX_DEV float fast_sigmoid(float x) {
   return __builtin_amdgcn_rcpf(1.f + __builtin_amdgcn_exp2f(-x * 1.4426950408889634f));
 }
 
+// gates_h (optional, bf16): the recurrent projection h@Wh kept as a
+// SEPARATE bf16 tensor and summed here — the fused alternative (f32
+// accumulate-GEMM into gates_pre) read-modify-writes a 4x bigger buffer
+// and was the dominant BPTT kernel.
+GFX_DEV float gate_pre(const float* gx, const __bf16* gh, int64_t off) {
+  float v = gx[off];
+  if (gh) v += bf2f(gh[off]);
+  return v;
+}
+
 __global__ void lstm_cell_fwd_kernel(
-    const float* __restrict__ gates_pre,  // [M, 4H]
+    const float* __restrict__ gates_pre,  // [M, 4H] (x-projection + bias)
+    const __bf16* __restrict__ gates_h,   // [M, 4H] or null (h-projection)
     const float* __restrict__ c_prev,     // [M, H]
     float* __restrict__ c_new,            // [M, H]
     __bf16* __restrict__ h_new,           // [M, H] (bf16: feeds next GEMM)
@@ -766,11 +777,13 @@ __global__ void lstm_cell_fwd_kernel(
        idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
     const int64_t m = idx / H;
     const int h = (int)(idx % H);
-    const float* gp = gates_pre + m * 4 * H;
-    const float i = fast_sigmoid(gp[h]);
-    const float f = fast_sigmoid(gp[H + h]);
-    const float g = fast_tanh(gp[2 * H + h]);
-    const float o = fast_sigmoid(gp[3 * H + h]);
+    const int64_t base = m * 4 * H;
+    const __bf16* gh = gates_h ? gates_h + base : nullptr;
+    const float* gx = gates_pre + base;
+    const float i = fast_sigmoid(gate_pre(gx, gh, h));
+    const float f = fast_sigmoid(gate_pre(gx, gh, H + h));
+    const float g = fast_tanh(gate_pre(gx, gh, 2 * H + h));
+    const float o = fast_sigmoid(gate_pre(gx, gh, 3 * H + h));
     const float c = f * c_prev[idx] + i * g;
     c_new[idx] = c;
     h_new[idx] = f2bf(o * fast_tanh(c));
@@ -783,7 +796,8 @@ __global__ void lstm_cell_fwd_kernel(
 // boundary AFTER this step) along with dc_next, so no gradient crosses a
 // reset.
 __global__ void lstm_cell_bwd_kernel(
-    const float* __restrict__ gates_pre,  // [M, 4H] (saved fwd)
+    const float* __restrict__ gates_pre,  // [M, 4H] (saved fwd, x part)
+    const __bf16* __restrict__ gates_h,   // [M, 4H] or null (saved h part)
     const float* __restrict__ c_prev,     // [M, H] (masked input c)
     const float* __restrict__ c_new,      // [M, H] (raw output c)
     const float* __restrict__ dh_head,    // [M, H]
@@ -799,11 +813,13 @@ __global__ void lstm_cell_bwd_kernel(
     const int64_t m = idx / H;
     const int h = (int)(idx % H);
     const float mask = (done && done[m]) ? 0.f : 1.f;
-    const float* gp = gates_pre + m * 4 * H;
-    const float i = fast_sigmoid(gp[h]);
-    const float f = fast_sigmoid(gp[H + h]);
-    const float g = fast_tanh(gp[2 * H + h]);
-    const float o = fast_sigmoid(gp[3 * H + h]);
+    const int64_t base = m * 4 * H;
+    const float* gx = gates_pre + base;
+    const __bf16* gh = gates_h ? gates_h + base : nullptr;
+    const float i = fast_sigmoid(gate_pre(gx, gh, h));
+    const float f = fast_sigmoid(gate_pre(gx, gh, H + h));
+    const float g = fast_tanh(gate_pre(gx, gh, 2 * H + h));
+    const float o = fast_sigmoid(gate_pre(gx, gh, 3 * H + h));
     const float c = c_new[idx];
     const float tc = fast_tanh(c);
     float dhv = dh_head[idx];
@@ -1019,17 +1035,18 @@ __global__ __launch_bounds__(256) void mb_gather_seq_kernel(
 // ---------------------------------------------------------------------------
 static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
-void launch_lstm_cell_fwd(const float* gates_pre, const float* c_prev,
-                          float* c_new, void* h_new, int64_t M, int H,
-                          hipStream_t stream) {
+void launch_lstm_cell_fwd(const float* gates_pre, const void* gates_h,
+                          const float* c_prev, float* c_new, void* h_new,
+                          int64_t M, int H, hipStream_t stream) {
   int64_t total = M * H;
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_fwd_kernel, dim3(blocks), dim3(256), 0, stream,
-                     gates_pre, c_prev, c_new,
-                     reinterpret_cast<__bf16*>(h_new), M, H);
+                     gates_pre, reinterpret_cast<const __bf16*>(gates_h),
+                     c_prev, c_new, reinterpret_cast<__bf16*>(h_new), M, H);
 }
 
-void launch_lstm_cell_bwd(const float* gates_pre, const float* c_prev,
+void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
+                          const float* c_prev,
                           const float* c_new, const float* dh_head,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
@@ -1037,7 +1054,8 @@ void launch_lstm_cell_bwd(const float* gates_pre, const float* c_prev,
   int64_t total = M * H;
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_bwd_kernel, dim3(blocks), dim3(256), 0, stream,
-                     gates_pre, c_prev, c_new, dh_head, dh_next, dc_next, done,
+                     gates_pre, reinterpret_cast<const __bf16*>(gates_h),
+                     c_prev, c_new, dh_head, dh_next, dc_next, done,
                      reinterpret_cast<__bf16*>(dgates), dc_prev, M, H);
 }
 

@@ -360,10 +360,12 @@ def test_lstm_cell_kernels_match_cpu_oracle():
     # fwd
     c_new_c = torch.empty(M, H)
     h_new_c = torch.empty(M, H, dtype=torch.bfloat16)
-    api.lstm_cell_fwd(gates, c_prev, c_new_c, h_new_c)
+    gates_h = (torch.randn(M, 4 * H, generator=g) * 0.1).to(torch.bfloat16)
+    api.lstm_cell_fwd(gates, gates_h, c_prev, c_new_c, h_new_c)
     c_new_g = torch.empty(M, H).cuda()
     h_new_g = torch.empty(M, H, dtype=torch.bfloat16).cuda()
-    ext.lstm_cell_fwd(gates.cuda(), c_prev.cuda(), c_new_g, h_new_g)
+    ext.lstm_cell_fwd(gates.cuda(), gates_h.cuda(), c_prev.cuda(), c_new_g,
+                      h_new_g)
     torch.cuda.synchronize()
     np.testing.assert_allclose(c_new_g.cpu().numpy(), c_new_c.numpy(),
                                rtol=1e-4, atol=1e-5)
@@ -372,11 +374,12 @@ def test_lstm_cell_kernels_match_cpu_oracle():
     # bwd (with masking)
     dg_c = torch.empty(M, 4 * H, dtype=torch.bfloat16)
     dcp_c = torch.empty(M, H)
-    api.lstm_cell_bwd(gates, c_prev, c_new_c, dh_head, dh_next, dc_next, done,
-                      dg_c, dcp_c)
+    api.lstm_cell_bwd(gates, gates_h, c_prev, c_new_c, dh_head, dh_next,
+                      dc_next, done, dg_c, dcp_c)
     dg_g = torch.empty(M, 4 * H, dtype=torch.bfloat16).cuda()
     dcp_g = torch.empty(M, H).cuda()
-    ext.lstm_cell_bwd(gates.cuda(), c_prev.cuda(), c_new_g, dh_head.cuda(),
+    ext.lstm_cell_bwd(gates.cuda(), gates_h.cuda(), c_prev.cuda(), c_new_g,
+                      dh_head.cuda(),
                       dh_next.cuda(), dc_next.cuda(), done.cuda(), dg_g, dcp_g)
     torch.cuda.synchronize()
     np.testing.assert_allclose(dg_g.float().cpu().numpy(),

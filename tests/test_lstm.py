@@ -19,7 +19,7 @@ def test_lstm_cell_fwd_matches_torch():
     c_prev = torch.randn(M, H, generator=g)
     c_new = torch.empty(M, H)
     h_new = torch.empty(M, H, dtype=torch.bfloat16)
-    api.lstm_cell_fwd(gates, c_prev, c_new, h_new)
+    api.lstm_cell_fwd(gates, None, c_prev, c_new, h_new)
     i, f, gg, o = gates.chunk(4, dim=1)
     c_ref = torch.sigmoid(f) * c_prev + torch.sigmoid(i) * torch.tanh(gg)
     h_ref = torch.sigmoid(o) * torch.tanh(c_ref)
@@ -45,8 +45,8 @@ def test_lstm_cell_bwd_matches_autograd():
 
     dgates = torch.empty(M, 4 * H, dtype=torch.bfloat16)
     dc_prev = torch.empty(M, H)
-    api.lstm_cell_bwd(gates.detach(), c_prev.detach(), c_new.detach(), dh,
-                      None, dc_next, None, dgates, dc_prev)
+    api.lstm_cell_bwd(gates.detach(), None, c_prev.detach(), c_new.detach(),
+                      dh, None, dc_next, None, dgates, dc_prev)
     assert torch.allclose(dgates.float(), dgates_ref, atol=2e-2, rtol=2e-2)
     assert torch.allclose(dc_prev, dc_prev_ref, atol=1e-5, rtol=1e-4)
 
@@ -64,12 +64,12 @@ def test_lstm_cell_bwd_done_masks_recurrent_grads():
     done[::2] = True
     dgates_m = torch.empty(M, 4 * H, dtype=torch.bfloat16)
     dc_prev_m = torch.empty(M, H)
-    api.lstm_cell_bwd(gates, c_prev, c_new, dh_head, dh_next, dc_next, done,
-                      dgates_m, dc_prev_m)
+    api.lstm_cell_bwd(gates, None, c_prev, c_new, dh_head, dh_next, dc_next,
+                      done, dgates_m, dc_prev_m)
     # for done rows the result must equal the no-next-grad case
     dgates_0 = torch.empty(M, 4 * H, dtype=torch.bfloat16)
     dc_prev_0 = torch.empty(M, H)
-    api.lstm_cell_bwd(gates, c_prev, c_new, dh_head, None, None, None,
+    api.lstm_cell_bwd(gates, None, c_prev, c_new, dh_head, None, None, None,
                       dgates_0, dc_prev_0)
     assert torch.equal(dgates_m[::2], dgates_0[::2])
     assert torch.equal(dc_prev_m[::2], dc_prev_0[::2])
@@ -106,7 +106,9 @@ def test_bptt_matches_autograd_replica():
     c = c0.clone()
     heads = []
     for l in range(L):
-        gates = obs[l].float() @ Wx_c + h @ Wh_c + b
+        gh = (h @ Wh_c)
+        gh = gh.to(torch.bfloat16).float() + (gh - gh.detach())  # bf16 fwd
+        gates = obs[l].float() @ Wx_c + gh + b
         i, f, gg, o = gates.chunk(4, dim=1)
         c = torch.sigmoid(f) * c + torch.sigmoid(i) * torch.tanh(gg)
         h_raw = torch.sigmoid(o) * torch.tanh(c)
