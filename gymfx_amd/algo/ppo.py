@@ -44,6 +44,7 @@ class PPOConfig:
     use_graphs: bool = True  # hipGraph-capture the update on GPU
     policy: str = "mlp"      # "mlp" | "lstm" (BASELINE configs #2 / #4)
     bptt_len: int = 16       # sequence-chunked BPTT length (lstm)
+    fused_rollout: bool = True  # single-kernel MLP policy step on GPU
     rollout_streams: int = 1  # >1: split rollout across HIP streams (GPU).
                              # Measured neutral-to-slightly-negative at
                              # N=4096 (per-kernel latency does not shrink
@@ -69,6 +70,7 @@ class PPOConfig:
             "policy": "policy_model",
             "bptt_len": "bptt_len",
             "rollout_streams": "rollout_streams",
+            "fused_rollout": "fused_rollout",
         }
         for attr, key in mapping.items():
             if cfg.get(key) is not None:
@@ -190,6 +192,8 @@ class PPOTrainer:
         # env halves on two HIP streams overlaps the chains.
         self._split = (self.device.type == "cuda" and cfg.rollout_streams > 1
                        and N % 2 == 0)
+        self._fused = (self.device.type == "cuda" and cfg.fused_rollout
+                       and not self.recurrent and cfg.hidden == 256)
         if self._split:
             self._s2 = torch.cuda.Stream()
             self.acts_half = [self.model.alloc_acts(N // 2),
@@ -225,14 +229,22 @@ class PPOTrainer:
                     self.h0_buf[ch][lo:hi].copy_(state["h"])
                     self.c0_buf[ch][lo:hi].copy_(state["c"])
                 head = model.step_forward(obs_t, state, acts)
+            elif self._fused:
+                model.fused_step(
+                    obs_t, self.act_buf[t][lo:hi], self.logp_buf[t][lo:hi],
+                    self.val_buf[t][lo:hi], seed=self.sample_seed, step=t,
+                    step_base=self.step_base, row_offset=lo,
+                )
+                head = None
             else:
                 head = model.forward(obs_t, acts)
-            api.sample_head(
-                head, self.sample_seed, t,
-                self.act_buf[t][lo:hi], self.logp_buf[t][lo:hi],
-                self.val_buf[t][lo:hi], step_base=self.step_base,
-                row_offset=lo,
-            )
+            if head is not None:
+                api.sample_head(
+                    head, self.sample_seed, t,
+                    self.act_buf[t][lo:hi], self.logp_buf[t][lo:hi],
+                    self.val_buf[t][lo:hi], step_base=self.step_base,
+                    row_offset=lo,
+                )
             # fused env kernel writes reward/done into the rollout slabs and
             # the NEXT observation (bf16) into obs_buf[t+1] directly
             nxt = self.obs_buf[t + 1] if t + 1 < T else self.obs_bf16_step

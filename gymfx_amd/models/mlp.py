@@ -139,6 +139,18 @@ class ActorCriticMLP:
             "head": torch.empty(M, self.head_dim, dtype=torch.float32, device=dev),
         }
 
+    def fused_step(self, obs_bf16: torch.Tensor, actions: torch.Tensor,
+                   logp: torch.Tensor, value: torch.Tensor, *, seed: int,
+                   step: int, step_base: torch.Tensor, row_offset: int = 0,
+                   greedy: bool = False) -> None:
+        """One fused policy step on GPU (forward + categorical sample in a
+        single kernel; bitwise identical to forward()+sample_head)."""
+        from ..ops import native
+        native.require().mlp_policy_rollout(
+            obs_bf16, self.wt("W1"), self.f32("b1"), self.wt("W2"),
+            self.f32("b2"), self.wt("W3"), self.f32("b3"), actions, logp,
+            value, seed, step, step_base, row_offset, greedy)
+
     def forward(self, obs_bf16: torch.Tensor, acts: Dict[str, torch.Tensor]) -> torch.Tensor:
         """obs_bf16 [M, obs_dim] -> head f32 [M, A+1]; saves h1/h2 for bwd.
 

@@ -67,6 +67,14 @@ void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,
                         hipStream_t stream);
 void launch_increment_u64(unsigned long long* ctr, unsigned long long delta,
                           hipStream_t stream);
+void launch_mlp_policy_rollout(const void* obs, const void* W1t,
+                               const float* b1, const void* W2t,
+                               const float* b2, const void* W3t,
+                               const float* b3, int64_t* actions, float* logp,
+                               float* value, int N, int D, int n_actions,
+                               uint64_t seed, uint64_t step,
+                               const unsigned long long* step_base,
+                               int row_offset, int greedy, hipStream_t stream);
 void launch_increment_i32(int* ctr, int delta, hipStream_t stream);
 void launch_mb_gather(const void* obs_src, const int64_t* act_src,
                       const float* logp_src, const float* adv_src,
@@ -603,6 +611,37 @@ void sample_head_op(torch::Tensor head, int64_t seed, int64_t step,
       greedy ? 1 : 0, sb, (int)row_offset, cur_stream());
 }
 
+void mlp_policy_rollout_op(torch::Tensor obs, torch::Tensor W1t,
+                           torch::Tensor b1, torch::Tensor W2t,
+                           torch::Tensor b2, torch::Tensor W3t,
+                           torch::Tensor b3, torch::Tensor actions,
+                           torch::Tensor logp, torch::Tensor value,
+                           int64_t seed, int64_t step,
+                           c10::optional<torch::Tensor> step_base,
+                           int64_t row_offset, bool greedy) {
+  check_bf16(obs, "obs");
+  check_bf16(W1t, "W1t");
+  check_bf16(W2t, "W2t");
+  check_bf16(W3t, "W3t");
+  const int N = (int)obs.size(0);
+  const int D = (int)obs.size(1);
+  const int H = (int)W1t.size(0);
+  const int head_dim = (int)W3t.size(0);
+  TORCH_CHECK(H == 256 && W2t.size(0) == 256 && W2t.size(1) == 256,
+              "fused rollout requires hidden == 256");
+  TORCH_CHECK(W1t.size(1) == D, "W1t shape");
+  TORCH_CHECK(head_dim <= 16 && W3t.size(1) == 256, "W3t shape");
+  const unsigned long long* sb = nullptr;
+  if (step_base.has_value())
+    sb = reinterpret_cast<const unsigned long long*>(step_base->data_ptr());
+  gymfx::launch_mlp_policy_rollout(
+      obs.data_ptr(), W1t.data_ptr(), b1.data_ptr<float>(), W2t.data_ptr(),
+      b2.data_ptr<float>(), W3t.data_ptr(), b3.data_ptr<float>(),
+      actions.data_ptr<int64_t>(), logp.data_ptr<float>(),
+      value.data_ptr<float>(), N, D, head_dim - 1, (uint64_t)seed,
+      (uint64_t)step, sb, (int)row_offset, greedy, cur_stream());
+}
+
 void increment_counter_op(torch::Tensor ctr, int64_t delta) {
   TORCH_CHECK(ctr.numel() == 1, "counter must be scalar");
   if (ctr.scalar_type() == torch::kInt32) {
@@ -739,6 +778,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("row_offset") = 0);
   m.def("increment_counter", &increment_counter_op, py::arg("ctr"),
         py::arg("delta"));
+  m.def("mlp_policy_rollout", &mlp_policy_rollout_op, py::arg("obs"),
+        py::arg("W1t"), py::arg("b1"), py::arg("W2t"), py::arg("b2"),
+        py::arg("W3t"), py::arg("b3"), py::arg("actions"), py::arg("logp"),
+        py::arg("value"), py::arg("seed"), py::arg("step"),
+        py::arg("step_base") = py::none(), py::arg("row_offset") = 0,
+        py::arg("greedy") = false);
   m.def("mb_gather", &mb_gather_op, py::arg("obs_src"), py::arg("act_src"),
         py::arg("logp_src"), py::arg("adv_src"), py::arg("ret_src"),
         py::arg("obs_mb"), py::arg("act_mb"), py::arg("logp_mb"),

@@ -32,6 +32,16 @@ GFX_DEV __bf16 f2bf(float v) { return (__bf16)v; }
 // CALL on gfx950: inside the GEMM epilogue it forced every live MFMA
 // accumulator to scratch (272 B/thread of spill, 6x slowdown).  Accuracy is
 // ~2 ulp of fp32 — far below the bf16 output rounding.
+// single-instruction exp/ln (libm expf/logf are CALLS on gfx950 — see the
+// fast_tanh note).  Used by BOTH sampler implementations so the fused
+// rollout kernel and sample_head_kernel agree bitwise.
+GFX_DEV float fast_exp(float x) {
+  return __builtin_amdgcn_exp2f(x * 1.4426950408889634f);
+}
+GFX_DEV float fast_log(float x) {
+  return __builtin_amdgcn_logf(x) * 0.6931471805599453f;
+}
+
 GFX_DEV float fast_tanh(float x) {
   x = fminf(fmaxf(x, -15.f), 15.f);
   // raw v_exp_f32 + v_rcp_f32, both single instructions.  An IEEE f32
@@ -595,8 +605,8 @@ __global__ void sample_head_kernel(const float* __restrict__ head, int M,
   float mx = row[0];
   for (int j = 1; j < n_actions; ++j) mx = fmaxf(mx, row[j]);
   float z = 0.f;
-  for (int j = 0; j < n_actions; ++j) z += expf(row[j] - mx);
-  const float logz = logf(z) + mx;
+  for (int j = 0; j < n_actions; ++j) z += fast_exp(row[j] - mx);
+  const float logz = fast_log(z) + mx;
 
   int a = 0;
   if (greedy) {
@@ -610,7 +620,7 @@ __global__ void sample_head_kernel(const float* __restrict__ head, int M,
     float c = 0.f;
     a = n_actions - 1;
     for (int j = 0; j < n_actions; ++j) {
-      c += expf(row[j] - logz);
+      c += fast_exp(row[j] - logz);
       if (u < c) { a = j; break; }
     }
   }
@@ -621,7 +631,7 @@ __global__ void sample_head_kernel(const float* __restrict__ head, int M,
     float h = 0.f;
     for (int j = 0; j < n_actions; ++j) {
       const float lp = row[j] - logz;
-      h -= expf(lp) * lp;
+      h -= fast_exp(lp) * lp;
     }
     entropy[m] = h;
   }
@@ -868,6 +878,210 @@ __global__ void masked_state_kernel(const __bf16* __restrict__ h_raw,
 }
 
 // ---------------------------------------------------------------------------
+// Fused MLP policy rollout step: obs -> tanh(obs@W1+b1) -> tanh(.@W2+b2)
+// -> head(.@W3+b3) -> categorical sample, ONE kernel.
+//
+// The unfused rollout chain (3 GEMMs + sampler) is execution-floor bound
+// (~8-10us per kernel at N=4096 regardless of bytes); fusing removes three
+// kernel floors per env step (x128 steps per update).  The math replicates
+// the GEMM kernel exactly (same BK=32 chunking, same MFMA sequence, bias
+// added in f32, tanh via fast_tanh, h1/h2 rounded to bf16 between layers),
+// so outputs are BITWISE identical to the unfused path — asserted by
+// tests/test_gpu_trainer.py::test_fused_rollout_equals_unfused.
+//
+// Shapes: H = 256 (4 waves each own 32 rows x 128 cols), head_dim <= 16,
+// D (obs_dim) arbitrary.  Block = 256 threads = 64 env rows; weights are
+// the transposed mirrors (W1t [256, D], W2t [256, 256], W3t [A+1, 256]).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 2) void mlp_policy_rollout_kernel(
+    const __bf16* __restrict__ obs,   // [N, D]
+    const __bf16* __restrict__ W1t,   // [256, D]
+    const float* __restrict__ b1,     // [256]
+    const __bf16* __restrict__ W2t,   // [256, 256]
+    const float* __restrict__ b2,     // [256]
+    const __bf16* __restrict__ W3t,   // [head_dim, 256]
+    const float* __restrict__ b3,     // [head_dim]
+    int64_t* __restrict__ actions, float* __restrict__ logp,
+    float* __restrict__ value, int N, int D, int n_actions, uint64_t seed,
+    uint64_t step, const unsigned long long* __restrict__ step_base,
+    int row_offset, int greedy) {
+  constexpr int BM = 64, BK = 32, H = 256;
+  constexpr int LDT = BK + 8;     // staged-operand row (bf16)
+  constexpr int LDH = H + 8;      // h1/h2 image row (bf16)
+  __shared__ __bf16 As[2][BM][LDT];
+  __shared__ __bf16 Bs[2][H][LDT];       // weight tile [n][k]
+  __shared__ __bf16 Himg[2][BM][LDH];    // h1 / h2 images
+  __shared__ float Head[BM][20];         // head values (row pad: 20 f32)
+  const int bm = blockIdx.x * BM;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int row_a = lane & 15;
+  const int kseg = lane >> 4;
+  const int a_r = tid >> 2, a_c8 = (tid & 3) * 8;
+  const int b_r = tid >> 2, b_c8 = (tid & 3) * 8;
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+  f32x4 acc[2][8];
+  bf16x8 ra, rb[4];
+
+  // Each FUSED_DENSE(...) computes Yimg = tanh(X @ Wt^T + bias) with the
+  // exact GEMM-kernel math (same BK chunking / MFMA order / f32 bias /
+  // fast_tanh / bf16 rounding) so results are bitwise identical to the
+  // unfused path.  FROM_GLOBAL selects the A source (global obs vs the
+  // previous layer's LDS image).
+#define FUSED_LOAD_A_GLOBAL(k0, K_)                                           \
+  {                                                                           \
+    ra = bf16x8{};                                                            \
+    const int gr = bm + a_r;                                                  \
+    if (gr < N) {                                                             \
+      const int gk = (k0) + a_c8;                                             \
+      if (gk + 8 <= (K_)) {                                                   \
+        ra = *reinterpret_cast<const bf16x8*>(&obs[(int64_t)gr * D + gk]);    \
+      } else {                                                                \
+        for (int i = 0; i < 8; ++i)                                           \
+          ra[i] = (gk + i < (K_)) ? obs[(int64_t)gr * D + gk + i]             \
+                                  : (__bf16)0.f;                              \
+      }                                                                       \
+    }                                                                         \
+  }
+#define FUSED_LOAD_B(Wt, k0, K_)                                              \
+  for (int j = 0; j < 4; ++j) {                                               \
+    rb[j] = bf16x8{};                                                         \
+    const int gn = b_r + j * 64;                                              \
+    const int gk = (k0) + b_c8;                                               \
+    if (gk + 8 <= (K_)) {                                                     \
+      rb[j] = *reinterpret_cast<const bf16x8*>(&(Wt)[(int64_t)gn * (K_) + gk]); \
+    } else {                                                                  \
+      for (int i = 0; i < 8; ++i)                                             \
+        rb[j][i] = (gk + i < (K_)) ? (Wt)[(int64_t)gn * (K_) + gk + i]        \
+                                   : (__bf16)0.f;                             \
+    }                                                                         \
+  }
+#define FUSED_DENSE(FROM_GLOBAL, XIMG, Wt, bias, K_, YIMG)                    \
+  {                                                                           \
+    for (int mi = 0; mi < 2; ++mi)                                            \
+      for (int ni = 0; ni < 8; ++ni) acc[mi][ni] = f32x4{};                   \
+    const int ktiles = ((K_) + BK - 1) / BK;                                  \
+    if (FROM_GLOBAL) FUSED_LOAD_A_GLOBAL(0, K_);                              \
+    FUSED_LOAD_B(Wt, 0, K_);                                                  \
+    if (FROM_GLOBAL)                                                          \
+      *reinterpret_cast<bf16x8*>(&As[0][a_r][a_c8]) = ra;                     \
+    else                                                                      \
+      *reinterpret_cast<bf16x8*>(&As[0][a_r][a_c8]) =                         \
+          *reinterpret_cast<const bf16x8*>(&(XIMG)[a_r][a_c8]);               \
+    for (int j = 0; j < 4; ++j)                                               \
+      *reinterpret_cast<bf16x8*>(&Bs[0][b_r + j * 64][b_c8]) = rb[j];         \
+    __syncthreads();                                                          \
+    for (int kt = 0; kt < ktiles; ++kt) {                                     \
+      const int cur = kt & 1;                                                 \
+      const int knext = (kt + 1) * BK;                                        \
+      if (kt + 1 < ktiles) {                                                  \
+        if (FROM_GLOBAL) FUSED_LOAD_A_GLOBAL(knext, K_);                      \
+        FUSED_LOAD_B(Wt, knext, K_);                                          \
+      }                                                                       \
+      bf16x8 af[2], bf[8];                                                    \
+      for (int mi = 0; mi < 2; ++mi)                                          \
+        af[mi] = *reinterpret_cast<const bf16x8*>(                            \
+            &As[cur][wr * 32 + mi * 16 + row_a][kseg * 8]);                   \
+      for (int ni = 0; ni < 8; ++ni)                                          \
+        bf[ni] = *reinterpret_cast<const bf16x8*>(                            \
+            &Bs[cur][wc * 128 + ni * 16 + row_a][kseg * 8]);                  \
+      for (int mi = 0; mi < 2; ++mi)                                          \
+        for (int ni = 0; ni < 8; ++ni)                                        \
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(              \
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);                          \
+      if (kt + 1 < ktiles) {                                                  \
+        if (FROM_GLOBAL)                                                      \
+          *reinterpret_cast<bf16x8*>(&As[1 - cur][a_r][a_c8]) = ra;           \
+        else                                                                  \
+          *reinterpret_cast<bf16x8*>(&As[1 - cur][a_r][a_c8]) =               \
+              *reinterpret_cast<const bf16x8*>(&(XIMG)[a_r][knext + a_c8]);   \
+        for (int j = 0; j < 4; ++j)                                           \
+          *reinterpret_cast<bf16x8*>(&Bs[1 - cur][b_r + j * 64][b_c8]) =      \
+              rb[j];                                                          \
+      }                                                                       \
+      __syncthreads();                                                        \
+    }                                                                         \
+    for (int mi = 0; mi < 2; ++mi)                                            \
+      for (int ni = 0; ni < 8; ++ni) {                                        \
+        const int lcol = wc * 128 + ni * 16 + ccol;                           \
+        for (int r = 0; r < 4; ++r) {                                         \
+          const int lrow = wr * 32 + mi * 16 + crow_base + r;                 \
+          float v = acc[mi][ni][r] + (bias)[lcol];                            \
+          (YIMG)[lrow][lcol] = f2bf(fast_tanh(v));                            \
+        }                                                                     \
+      }                                                                       \
+    __syncthreads();                                                          \
+  }
+
+  FUSED_DENSE(true, Himg[0], W1t, b1, D, Himg[0]);
+  FUSED_DENSE(false, Himg[0], W2t, b2, H, Himg[1]);
+#undef FUSED_DENSE
+#undef FUSED_LOAD_B
+#undef FUSED_LOAD_A_GLOBAL
+
+  // ---- head: [64, head_dim] = h2 @ W3 + b3 via one MFMA column tile ----
+  {
+    // wave w handles rows w*16..w*16+15 (16x16 MFMA, cols 0..15)
+    f32x4 hacc = {};
+    const int hrow = wave * 16 + row_a;
+    for (int kt = 0; kt < H / BK; ++kt) {
+      bf16x8 af, bf;
+      af = *reinterpret_cast<const bf16x8*>(
+          &Himg[1][hrow][kt * BK + kseg * 8]);
+      for (int i = 0; i < 8; ++i) {
+        const int k = kt * BK + kseg * 8 + i;
+        bf[i] = (row_a <= n_actions) ? W3t[(int64_t)row_a * H + k] : (__bf16)0.f;
+      }
+      hacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, hacc, 0, 0, 0);
+    }
+    for (int r = 0; r < 4; ++r) {
+      const int lrow = wave * 16 + crow_base + r;
+      if (ccol <= n_actions)
+        Head[lrow][ccol] = hacc[r] + b3[ccol];
+    }
+  }
+  __syncthreads();
+
+  // ---- categorical sample (== sample_head_kernel math) ----------------
+  if (tid < BM) {
+    const int gr = bm + tid;
+    if (gr < N) {
+      const float* row = Head[tid];
+      float mx = row[0];
+      for (int j = 1; j < n_actions; ++j) mx = fmaxf(mx, row[j]);
+      float z = 0.f;
+      for (int j = 0; j < n_actions; ++j) z += fast_exp(row[j] - mx);
+      const float logz = fast_log(z) + mx;
+      uint64_t st = step;
+      if (step_base) st += *step_base;
+      int a = 0;
+      if (greedy) {
+        float best = row[0];
+        for (int j = 1; j < n_actions; ++j)
+          if (row[j] > best) { best = row[j]; a = j; }
+      } else {
+        const int64_t mg = gr + row_offset;
+        const uint64_t r =
+            splitmix64(seed ^ (st * 0x51E1F5ull + (uint64_t)mg * 0x9E37ull));
+        float u = (float)((r >> 11) * (1.0 / 9007199254740992.0));
+        u = fminf(u, 0.999999f);
+        float c = 0.f;
+        a = n_actions - 1;
+        for (int j = 0; j < n_actions; ++j) {
+          c += fast_exp(row[j] - logz);
+          if (u < c) { a = j; break; }
+        }
+      }
+      actions[gr] = a;
+      logp[gr] = row[a] - logz;
+      if (value) value[gr] = row[n_actions];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Device counters (hipGraph support): tiny kernels that advance the RNG /
 // optimizer-step counters inside a captured graph, so one captured update
 // replays with fresh randomness and correct Adam bias correction.
@@ -1098,6 +1312,25 @@ void launch_mb_gather_seq(const void* obs_src, const int64_t* act_src,
                      reinterpret_cast<__bf16*>(h0_mb), c0_mb,
                      Mseq, L, D, H, N, n_seq, half, seed, minibatches,
                      step_base, mb_ctr);
+}
+
+void launch_mlp_policy_rollout(const void* obs, const void* W1t,
+                               const float* b1, const void* W2t,
+                               const float* b2, const void* W3t,
+                               const float* b3, int64_t* actions, float* logp,
+                               float* value, int N, int D, int n_actions,
+                               uint64_t seed, uint64_t step,
+                               const unsigned long long* step_base,
+                               int row_offset, int greedy,
+                               hipStream_t stream) {
+  hipLaunchKernelGGL(mlp_policy_rollout_kernel, dim3(ceil_div(N, 64)),
+                     dim3(256), 0, stream,
+                     reinterpret_cast<const __bf16*>(obs),
+                     reinterpret_cast<const __bf16*>(W1t), b1,
+                     reinterpret_cast<const __bf16*>(W2t), b2,
+                     reinterpret_cast<const __bf16*>(W3t), b3, actions, logp,
+                     value, N, D, n_actions, seed, step, step_base, row_offset,
+                     greedy);
 }
 
 void launch_increment_u64(unsigned long long* ctr, unsigned long long delta,

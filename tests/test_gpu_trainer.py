@@ -147,3 +147,38 @@ def test_split_rollout_equals_single_stream():
     assert torch.equal(t1.act_buf, t2.act_buf)
     assert torch.equal(t1.model.params, t2.model.params)
     assert torch.equal(t1.env.st.equity, t2.env.st.equity)
+
+
+def test_fused_rollout_equals_unfused():
+    """The fused MLP policy kernel (2 MFMA layers + MFMA head + sampler in
+    one launch) must be BITWISE identical to the unfused GEMM+sampler
+    chain (same BK chunking, MFMA sequence, bias/tanh/rounding, RNG)."""
+    from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    def make(fused):
+        md = synthetic_ohlcv(2000, seed=5, vol=4e-4, extra_feature_columns=3)
+        cfg = {"n_envs": 256, "device": "cuda", "window_size": 16,
+               "preprocessor_plugin": "feature_window_preprocessor",
+               "feature_columns": ["OPEN", "HIGH", "LOW", "CLOSE",
+                                   "FEAT_0", "FEAT_1", "FEAT_2"],
+               "env_start_mode": "spread", "autoreset": True,
+               "position_size": 1000.0, "seed": 13}
+        env = build_vec_environment(cfg, md)
+        env.reset(seed=13)
+        pc = PPOConfig(rollout_steps=16, minibatches=4, ppo_epochs=2, seed=13,
+                       fused_rollout=fused)
+        return PPOTrainer(env, pc)
+
+    tf = make(True)
+    tu = make(False)
+    for _ in range(2):
+        tf.train_update(with_stats=False)
+        tu.train_update(with_stats=False)
+    torch.cuda.synchronize()
+    assert tf._fused and not tu._fused
+    assert torch.equal(tf.act_buf, tu.act_buf)
+    assert torch.equal(tf.logp_buf, tu.logp_buf)
+    assert torch.equal(tf.val_buf, tu.val_buf)
+    assert torch.equal(tf.model.params, tu.model.params)
