@@ -44,7 +44,10 @@ class PPOConfig:
     use_graphs: bool = True  # hipGraph-capture the update on GPU
     policy: str = "mlp"      # "mlp" | "lstm" (BASELINE configs #2 / #4)
     bptt_len: int = 16       # sequence-chunked BPTT length (lstm)
-    fused_rollout: bool = True  # single-kernel MLP policy step on GPU
+    fused_rollout: bool = False  # single-kernel MLP policy step (bitwise-
+                                 # identical option; measured SLOWER: the
+                                 # mega-kernel spills ~2.5k SGPRs and runs
+                                 # at occupancy 1 — kept for the record)
     rollout_streams: int = 1  # >1: split rollout across HIP streams (GPU).
                              # Measured neutral-to-slightly-negative at
                              # N=4096 (per-kernel latency does not shrink
