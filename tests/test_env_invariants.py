@@ -125,3 +125,18 @@ def test_continuous_action_mode(data_files):
     assert a.tolist() == [1, 2, 0]
     d = env.action_diagnostics(2)
     assert d["continuous_deadband_actions"] == 1
+
+
+def test_gym_api_compliance():
+    """In-house env-checker (the reference runs gymnasium's check_env,
+    tools/check_gym_compliance.py; gymnasium is absent in this image)."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    r = subprocess.run(
+        [sys.executable, str(Path(__file__).resolve().parents[1] / "tools" /
+                             "check_gym_compliance.py")],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert r.returncode == 0, r.stdout + r.stderr
