@@ -148,6 +148,8 @@ class GymFxEnv(spaces.Env):
         self._np_random = np.random.default_rng()
         self._last_info_extras: Dict[str, Any] = {}
         self._was_reset = False
+        self._last_raw_action_value = None
+        self._last_coerced_action = None
         # bracket/force-close audit trail (parity:
         # /root/reference/strategy_plugins/direct_atr_sltp.py:40-50 —
         # GYMFX_BRACKET_AUDIT=<path> enables append-only JSONL records)
@@ -188,6 +190,8 @@ class GymFxEnv(spaces.Env):
         out = self.vec.step(act_t)
         if self._audit_path:
             self._emit_audit()
+        self._last_raw_action_value = raw
+        self._last_coerced_action = int(out["coerced_action"][0].item())
         reward = float(out["reward"][0].item())
         base_reward = float(out["base_reward"][0].item())
         penalty = float(out["force_close_reward_penalty"][0].item())
@@ -318,8 +322,8 @@ class GymFxEnv(spaces.Env):
             "total_bars": self.total_bars,
             "trades": bs["trade_count"],
             "commission_paid": bs["commission_paid"],
-            "raw_action_value": None,
-            "coerced_action": None,
+            "raw_action_value": self._last_raw_action_value,
+            "coerced_action": self._last_coerced_action,
             "action_diagnostics": self.vec.action_diagnostics(0),
             "execution_diagnostics": self.vec.execution_diagnostics(0),
         }
