@@ -14,7 +14,7 @@ latency-bound and a single fused bucket is optimal on this topology.
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional
 
 import torch

@@ -925,101 +925,87 @@ __global__ __launch_bounds__(256, 2) void mlp_policy_rollout_kernel(
   f32x4 acc[2][8];
   bf16x8 ra, rb[4];
 
-  // Each FUSED_DENSE(...) computes Yimg = tanh(X @ Wt^T + bias) with the
-  // exact GEMM-kernel math (same BK chunking / MFMA order / f32 bias /
-  // fast_tanh / bf16 rounding) so results are bitwise identical to the
-  // unfused path.  FROM_GLOBAL selects the A source (global obs vs the
-  // previous layer's LDS image).
-#define FUSED_LOAD_A_GLOBAL(k0, K_)                                           \
-  {                                                                           \
-    ra = bf16x8{};                                                            \
-    const int gr = bm + a_r;                                                  \
-    if (gr < N) {                                                             \
-      const int gk = (k0) + a_c8;                                             \
-      if (gk + 8 <= (K_)) {                                                   \
-        ra = *reinterpret_cast<const bf16x8*>(&obs[(int64_t)gr * D + gk]);    \
-      } else {                                                                \
-        for (int i = 0; i < 8; ++i)                                           \
-          ra[i] = (gk + i < (K_)) ? obs[(int64_t)gr * D + gk + i]             \
-                                  : (__bf16)0.f;                              \
-      }                                                                       \
-    }                                                                         \
+  // ONE shared dense-layer loop (layer selected at runtime): the first
+  // version macro-duplicated the whole GEMM body per layer and the
+  // allocator thrashed (~2.5k SGPR spills).  Math is bitwise identical to
+  // the unfused GEMM kernel (same BK chunking, MFMA order, f32 bias,
+  // fast_tanh, bf16 rounding).
+  for (int layer = 0; layer < 2; ++layer) {
+    const bool from_global = layer == 0;
+    const __bf16* Wt = from_global ? W1t : W2t;
+    const float* bias = from_global ? b1 : b2;
+    const int K = from_global ? D : H;
+    const __bf16(*Ximg)[LDH] = Himg[0];
+    __bf16(*Yimg)[LDH] = Himg[layer];
+    const int ktiles = (K + BK - 1) / BK;
+    for (int mi = 0; mi < 2; ++mi)
+      for (int ni = 0; ni < 8; ++ni) acc[mi][ni] = f32x4{};
+    for (int kt = -1; kt + 1 < ktiles + 1; ++kt) {
+      // stage chunk kt+1 (kt==-1 is the prologue), compute chunk kt
+      const int knext = (kt + 1) * BK;
+      if (kt + 1 < ktiles) {
+        if (from_global) {
+          ra = bf16x8{};
+          const int gr = bm + a_r;
+          if (gr < N) {
+            const int gk = knext + a_c8;
+            if (gk + 8 <= K) {
+              ra = *reinterpret_cast<const bf16x8*>(&obs[(int64_t)gr * D + gk]);
+            } else {
+              for (int i = 0; i < 8; ++i)
+                ra[i] = (gk + i < K) ? obs[(int64_t)gr * D + gk + i]
+                                     : (__bf16)0.f;
+            }
+          }
+        } else {
+          ra = *reinterpret_cast<const bf16x8*>(&Ximg[a_r][knext + a_c8]);
+        }
+        for (int j = 0; j < 4; ++j) {
+          const int gn = b_r + j * 64;
+          const int gk = knext + b_c8;
+          if (gk + 8 <= K) {
+            rb[j] = *reinterpret_cast<const bf16x8*>(&Wt[(int64_t)gn * K + gk]);
+          } else {
+            rb[j] = bf16x8{};
+            for (int i = 0; i < 8; ++i)
+              rb[j][i] = (gk + i < K) ? Wt[(int64_t)gn * K + gk + i]
+                                      : (__bf16)0.f;
+          }
+        }
+      }
+      if (kt >= 0) {
+        const int cur = kt & 1;
+        bf16x8 af[2], bf[8];
+        for (int mi = 0; mi < 2; ++mi)
+          af[mi] = *reinterpret_cast<const bf16x8*>(
+              &As[cur][wr * 32 + mi * 16 + row_a][kseg * 8]);
+        for (int ni = 0; ni < 8; ++ni)
+          bf[ni] = *reinterpret_cast<const bf16x8*>(
+              &Bs[cur][wc * 128 + ni * 16 + row_a][kseg * 8]);
+        for (int mi = 0; mi < 2; ++mi)
+          for (int ni = 0; ni < 8; ++ni)
+            acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      }
+      if (kt + 1 < ktiles) {
+        const int nxt = (kt + 1) & 1;
+        *reinterpret_cast<bf16x8*>(&As[nxt][a_r][a_c8]) = ra;
+        for (int j = 0; j < 4; ++j)
+          *reinterpret_cast<bf16x8*>(&Bs[nxt][b_r + j * 64][b_c8]) = rb[j];
+      }
+      __syncthreads();
+    }
+    for (int mi = 0; mi < 2; ++mi)
+      for (int ni = 0; ni < 8; ++ni) {
+        const int lcol = wc * 128 + ni * 16 + ccol;
+        for (int r = 0; r < 4; ++r) {
+          const int lrow = wr * 32 + mi * 16 + crow_base + r;
+          float v = acc[mi][ni][r] + bias[lcol];
+          Yimg[lrow][lcol] = f2bf(fast_tanh(v));
+        }
+      }
+    __syncthreads();
   }
-#define FUSED_LOAD_B(Wt, k0, K_)                                              \
-  for (int j = 0; j < 4; ++j) {                                               \
-    rb[j] = bf16x8{};                                                         \
-    const int gn = b_r + j * 64;                                              \
-    const int gk = (k0) + b_c8;                                               \
-    if (gk + 8 <= (K_)) {                                                     \
-      rb[j] = *reinterpret_cast<const bf16x8*>(&(Wt)[(int64_t)gn * (K_) + gk]); \
-    } else {                                                                  \
-      for (int i = 0; i < 8; ++i)                                             \
-        rb[j][i] = (gk + i < (K_)) ? (Wt)[(int64_t)gn * (K_) + gk + i]        \
-                                   : (__bf16)0.f;                             \
-    }                                                                         \
-  }
-#define FUSED_DENSE(FROM_GLOBAL, XIMG, Wt, bias, K_, YIMG)                    \
-  {                                                                           \
-    for (int mi = 0; mi < 2; ++mi)                                            \
-      for (int ni = 0; ni < 8; ++ni) acc[mi][ni] = f32x4{};                   \
-    const int ktiles = ((K_) + BK - 1) / BK;                                  \
-    if (FROM_GLOBAL) FUSED_LOAD_A_GLOBAL(0, K_);                              \
-    FUSED_LOAD_B(Wt, 0, K_);                                                  \
-    if (FROM_GLOBAL)                                                          \
-      *reinterpret_cast<bf16x8*>(&As[0][a_r][a_c8]) = ra;                     \
-    else                                                                      \
-      *reinterpret_cast<bf16x8*>(&As[0][a_r][a_c8]) =                         \
-          *reinterpret_cast<const bf16x8*>(&(XIMG)[a_r][a_c8]);               \
-    for (int j = 0; j < 4; ++j)                                               \
-      *reinterpret_cast<bf16x8*>(&Bs[0][b_r + j * 64][b_c8]) = rb[j];         \
-    __syncthreads();                                                          \
-    for (int kt = 0; kt < ktiles; ++kt) {                                     \
-      const int cur = kt & 1;                                                 \
-      const int knext = (kt + 1) * BK;                                        \
-      if (kt + 1 < ktiles) {                                                  \
-        if (FROM_GLOBAL) FUSED_LOAD_A_GLOBAL(knext, K_);                      \
-        FUSED_LOAD_B(Wt, knext, K_);                                          \
-      }                                                                       \
-      bf16x8 af[2], bf[8];                                                    \
-      for (int mi = 0; mi < 2; ++mi)                                          \
-        af[mi] = *reinterpret_cast<const bf16x8*>(                            \
-            &As[cur][wr * 32 + mi * 16 + row_a][kseg * 8]);                   \
-      for (int ni = 0; ni < 8; ++ni)                                          \
-        bf[ni] = *reinterpret_cast<const bf16x8*>(                            \
-            &Bs[cur][wc * 128 + ni * 16 + row_a][kseg * 8]);                  \
-      for (int mi = 0; mi < 2; ++mi)                                          \
-        for (int ni = 0; ni < 8; ++ni)                                        \
-          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(              \
-              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);                          \
-      if (kt + 1 < ktiles) {                                                  \
-        if (FROM_GLOBAL)                                                      \
-          *reinterpret_cast<bf16x8*>(&As[1 - cur][a_r][a_c8]) = ra;           \
-        else                                                                  \
-          *reinterpret_cast<bf16x8*>(&As[1 - cur][a_r][a_c8]) =               \
-              *reinterpret_cast<const bf16x8*>(&(XIMG)[a_r][knext + a_c8]);   \
-        for (int j = 0; j < 4; ++j)                                           \
-          *reinterpret_cast<bf16x8*>(&Bs[1 - cur][b_r + j * 64][b_c8]) =      \
-              rb[j];                                                          \
-      }                                                                       \
-      __syncthreads();                                                        \
-    }                                                                         \
-    for (int mi = 0; mi < 2; ++mi)                                            \
-      for (int ni = 0; ni < 8; ++ni) {                                        \
-        const int lcol = wc * 128 + ni * 16 + ccol;                           \
-        for (int r = 0; r < 4; ++r) {                                         \
-          const int lrow = wr * 32 + mi * 16 + crow_base + r;                 \
-          float v = acc[mi][ni][r] + (bias)[lcol];                            \
-          (YIMG)[lrow][lcol] = f2bf(fast_tanh(v));                            \
-        }                                                                     \
-      }                                                                       \
-    __syncthreads();                                                          \
-  }
-
-  FUSED_DENSE(true, Himg[0], W1t, b1, D, Himg[0]);
-  FUSED_DENSE(false, Himg[0], W2t, b2, H, Himg[1]);
-#undef FUSED_DENSE
-#undef FUSED_LOAD_B
-#undef FUSED_LOAD_A_GLOBAL
 
   // ---- head: [64, head_dim] = h2 @ W3 + b3 via one MFMA column tile ----
   {
