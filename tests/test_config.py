@@ -88,3 +88,54 @@ def test_optimization_mode_random_search(tmp_path):
     assert out["trials"] == 3
     assert set(out["best_params"]) == {"atr_period", "k_sl", "k_tp"}
     assert len(out["top5"]) == 3
+
+
+def test_remote_config_http_roundtrip():
+    """Remote config/log over HTTP with basic auth + degrade-on-failure
+    (parity: /root/reference/app/config_handler.py:30-73)."""
+    import http.server
+    import json as _json
+    import threading
+    from urllib.parse import parse_qs
+
+    from gymfx_amd.config.handler import (remote_load_config, remote_log,
+                                          remote_save_config)
+
+    received = {}
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            body = self.rfile.read(int(self.headers["Content-Length"]))
+            received.update({k: v[0] for k, v in
+                             parse_qs(body.decode()).items()})
+            received["auth"] = self.headers.get("Authorization", "")
+            self.send_response(200)
+            self.end_headers()
+
+        def do_GET(self):
+            payload = _json.dumps({"window_size": 12}).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.end_headers()
+            self.wfile.write(payload)
+
+        def log_message(self, *a):
+            pass
+
+    srv = http.server.HTTPServer(("127.0.0.1", 0), Handler)
+    port = srv.server_address[1]
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        url = f"http://127.0.0.1:{port}/cfg"
+        assert remote_save_config({"window_size": 9}, url, "u", "p")
+        assert _json.loads(received["json_config"]) == {"window_size": 9}
+        assert received["auth"].startswith("Basic ")
+        assert remote_load_config(url) == {"window_size": 12}
+        assert remote_log({"window_size": 9}, {"result": 1}, url, "u", "p")
+        assert _json.loads(received["json_result"]) == {"result": 1}
+        # degrade: unreachable endpoint returns False/None, never raises
+        assert remote_save_config({}, "http://127.0.0.1:9/none", "u", "p") is False
+        assert remote_load_config("http://127.0.0.1:9/none") is None
+    finally:
+        srv.shutdown()
