@@ -50,9 +50,11 @@ class GradAllReducer:
     between the two overlaps the collective.
     """
 
-    def __init__(self, world_size: int, process_group=None):
+    def __init__(self, world_size: int, process_group=None,
+                 timeout_s: float = 300.0):
         self.world_size = world_size
         self.pg = process_group
+        self.timeout_s = timeout_s
         self._work = None
 
     def start(self, grads: torch.Tensor) -> None:
@@ -65,9 +67,18 @@ class GradAllReducer:
         self._grads = grads
 
     def finish(self) -> None:
+        """Wait with a watchdog: a dead peer rank surfaces as a clear error
+        instead of an indefinite hang (failure-detection stance of
+        SURVEY.md §5.3 — fail loudly at the collective boundary)."""
         if self.world_size <= 1 or self._work is None:
             return
-        self._work.wait()
+        from datetime import timedelta
+
+        ok = self._work.wait(timedelta(seconds=self.timeout_s))
+        if ok is False:  # gloo returns False on timeout
+            raise RuntimeError(
+                f"gradient all-reduce timed out after {self.timeout_s}s "
+                "(peer rank dead or desynchronized)")
         self._grads.mul_(1.0 / self.world_size)
         self._work = None
 
