@@ -133,3 +133,49 @@ def test_native_required_on_gpu():
     md = _market()
     env = build_vec_environment({**BASE, "device": "cuda"}, md)
     assert env._native is not None
+
+
+def test_replay_reconciles_hip_engine_vs_scalar_ledger():
+    """The fused HIP engine's ledger vs the independent pure-Python scalar
+    oracle (replay.py; bakeoff.py:228-303 idiom) — on device."""
+    import numpy as np
+
+    from gymfx_amd.replay import ReplayAdapter
+
+    md = synthetic_ohlcv(400, seed=9, vol=5e-4)
+    rng = np.random.default_rng(3)
+    actions = rng.integers(0, 3, size=250).tolist()
+    cfg = {"window_size": 8, "initial_cash": 10000.0, "position_size": 1000.0,
+           "commission": 2e-5, "slippage": 5e-6, "device": "cuda", "seed": 0,
+           "strategy_plugin": "direct_fixed_sltp", "sl_pips": 8.0,
+           "tp_pips": 16.0}
+    res = ReplayAdapter().run(cfg, md, actions)
+    assert res["reconciled"], res["reconciliation"]
+    kinds = {e["type"] for e in res["events"]}
+    assert "order_filled" in kinds
+
+
+def test_update_speed_regression_guard():
+    """Catastrophic-regression guard: one full PPO update (4096 envs,
+    T=128) must stay well under 3x the measured ~23 ms."""
+    import time
+
+    from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+
+    md = synthetic_ohlcv(65536, seed=5, vol=4e-4, extra_feature_columns=3)
+    cfg = {**BASE, "n_envs": 4096, "device": "cuda", "autoreset": True,
+           "preprocessor_plugin": "feature_window_preprocessor",
+           "feature_columns": ["OPEN", "HIGH", "LOW", "CLOSE",
+                               "FEAT_0", "FEAT_1", "FEAT_2"]}
+    env = build_vec_environment(cfg, md)
+    env.reset(seed=0)
+    tr = PPOTrainer(env, PPOConfig(seed=0))
+    for _ in range(3):
+        tr.train_update(with_stats=False)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(5):
+        tr.train_update(with_stats=False)
+    torch.cuda.synchronize()
+    ms = (time.perf_counter() - t0) / 5 * 1000
+    assert ms < 80.0, f"PPO update took {ms:.1f} ms (expected ~23 ms)"
