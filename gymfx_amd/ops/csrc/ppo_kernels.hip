@@ -245,11 +245,118 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
 // LDS (scalar LDS reads are cheap; strided 2-byte GLOBAL reads were the
 // bottleneck of the first version of this kernel).
 // ---------------------------------------------------------------------------
+// ---------------------------------------------------------------------------
+// glds wgrad, 64x64 output tile, BKm=64 m-chunks: stages both operand
+// chunks with __builtin_amdgcn_global_load_lds (16B per lane, LDS image
+// linear in lane order — guide §5: 64²-tile step-3 structure), two LDS
+// buffers, one __syncthreads per chunk (the barrier's vmcnt(0) drains the
+// in-flight glds of the NEXT chunk — the simple 2-barrier recipe).
+// Requires full, aligned tiles: bk+64<=K, bn+64<=N and a full 64-row
+// m-chunk; the launcher falls back to the register-staged kernel for edge
+// tiles (K%64 tails, the head layer's N=4, partial slabs).
+// ---------------------------------------------------------------------------
+template <bool WANT_DB>
+__global__ __launch_bounds__(256) void wgrad_glds_kernel(
+    const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
+    float* __restrict__ dW_part, float* __restrict__ db_part,
+    int M, int N, int K, int slabs) {
+  constexpr int BKm = 64, TK = 64, TN = 64;
+  // UNPADDED linear images (glds cannot scatter past a row pad)
+  __shared__ __bf16 Xs[2][BKm][TK];
+  __shared__ __bf16 Ys[2][BKm][TN];
+
+  const int bk = blockIdx.x * TK;
+  const int bn = blockIdx.y * TN;
+  const int slab = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+
+  const int m_per_slab = (M + slabs - 1) / slabs;
+  const int m_begin = slab * m_per_slab;
+  const int m_end = min(M, m_begin + m_per_slab);
+  const int nchunks = (m_end - m_begin) / BKm;  // full chunks only (caller
+                                                // guarantees divisibility)
+  f32x4 acc[2][2] = {};
+  float db_acc = 0.f;
+
+  // one glds instruction stages 64 lanes x 16B = 1KB; a 64x64 bf16 chunk is
+  // 8KB = 8 instructions = 2 per wave.  element e = lane*8 within the 512-
+  // element span; row = e/64, col = e%64.
+  const int g_span = (wave * 2) * 512;          // element base of this
+                                                // wave's first span
+  auto glds_chunk = [&](int buf, int m0) {
+    for (int j = 0; j < 2; ++j) {
+      const int e0 = g_span + j * 512;
+      const int row = (e0 + lane * 8) / 64;
+      const int col = (e0 + lane * 8) % 64;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)(
+              &X[(int64_t)(m0 + row) * K + bk + col]),
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              &Xs[buf][0][0] + e0),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)(
+              &dY[(int64_t)(m0 + row) * N + bn + col]),
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              &Ys[buf][0][0] + e0),
+          16, 0, 0);
+    }
+  };
+
+  const int row_a = lane & 15;
+  const int kseg = lane >> 4;
+
+  if (nchunks > 0) glds_chunk(0, m_begin);
+  __syncthreads();  // vmcnt(0) inside the barrier drains the glds
+  for (int ci = 0; ci < nchunks; ++ci) {
+    const int cur = ci & 1;
+    if (ci + 1 < nchunks) glds_chunk(1 - cur, m_begin + (ci + 1) * BKm);
+
+    if (WANT_DB && tid < TN) {
+      float s = 0.f;
+      for (int i = 0; i < BKm; ++i) s += bf2f(Ys[cur][i][tid]);
+      db_acc += s;
+    }
+    // two 32-row reduction batches per 64-row chunk
+    for (int half = 0; half < 2; ++half) {
+      const int mb = half * 32;
+      bf16x8 af[2], bf_[2];
+      for (int fi = 0; fi < 2; ++fi)
+        for (int i = 0; i < 8; ++i)
+          af[fi][i] = Xs[cur][mb + kseg * 8 + i][wr * 32 + fi * 16 + row_a];
+      for (int ni = 0; ni < 2; ++ni)
+        for (int i = 0; i < 8; ++i)
+          bf_[ni][i] = Ys[cur][mb + kseg * 8 + i][wc * 32 + ni * 16 + row_a];
+      for (int fi = 0; fi < 2; ++fi)
+        for (int ni = 0; ni < 2; ++ni)
+          acc[fi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[fi], bf_[ni], acc[fi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+  float* out = dW_part + (int64_t)slab * K * N;
+  for (int fi = 0; fi < 2; ++fi)
+    for (int ni = 0; ni < 2; ++ni) {
+      const int gcol = bn + wc * 32 + ni * 16 + ccol;
+      for (int r = 0; r < 4; ++r) {
+        const int grow = bk + wr * 32 + fi * 16 + crow_base + r;
+        out[(int64_t)grow * N + gcol] = acc[fi][ni][r];
+      }
+    }
+  if (WANT_DB && tid < TN && blockIdx.x == 0)
+    db_part[(int64_t)slab * N + bn + tid] = db_acc;
+}
+
 template <bool WANT_DB, int FK, int FN>
 __global__ __launch_bounds__(256) void wgrad_partial_kernel(
     const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
     float* __restrict__ dW_part, float* __restrict__ db_part,
-    int M, int N, int K, int slabs) {
+    int M, int N, int K, int slabs, int kt0) {
   // FK/FN = MFMA fragments per wave along K/N: block tile (32*FK) x (32*FN)
   // over the dW output; reduction dim is M, chunked BKm=32 rows at a time
   // with double-buffered natural-layout LDS staging (coalesced bf16x8
@@ -260,7 +367,7 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   __shared__ __bf16 Xs[2][BKm][LDX];   // X tile, natural layout: [m][k]
   __shared__ __bf16 Ys[2][BKm][LDY];   // dY tile, natural layout: [m][n]
 
-  const int bk = blockIdx.x * TK;
+  const int bk = (kt0 + blockIdx.x) * TK;
   const int bn = blockIdx.y * TN;
   const int slab = blockIdx.z;
   const int tid = threadIdx.x;
@@ -372,7 +479,7 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
         out[(int64_t)grow * N + gcol] = acc[fi][ni][r];
       }
     }
-  if (WANT_DB && tid < TN && blockIdx.x == 0) {
+  if (WANT_DB && tid < TN && kt0 == 0 && blockIdx.x == 0) {
     const int gn = bn + tid;
     if (gn < N) db_part[(int64_t)slab * N + gn] = db_acc;
   }
@@ -1404,21 +1511,48 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
   // measured slower at 128x128 because the grid collapsed to ~1 block/CU).
   const bool big = K >= 128 && N >= 128 &&
                    (int64_t)ceil_div(K, 128) * ceil_div(N, 128) * slabs >= 1024;
-  dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);
-  if (db_part) {
-    if (big)
-      hipLaunchKernelGGL((wgrad_partial_kernel<true, 4, 4>), grid, dim3(256),
-                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs);
+  const int m_per_slab = (M + slabs - 1) / slabs;
+  // glds path: full 64-aligned tiles only; the K tail (K % 64) runs the
+  // register-staged kernel with a k-tile offset.
+  const bool glds_ok = !big && N % 64 == 0 && K >= 64 &&
+                       m_per_slab % 64 == 0 && M % slabs == 0;
+  if (glds_ok) {
+    const int kfull = K / 64;
+    dim3 g0(kfull, N / 64, slabs);
+    if (db_part)
+      hipLaunchKernelGGL((wgrad_glds_kernel<true>), g0, dim3(256), 0, stream,
+                         x, dy, dW_part, db_part, M, N, K, slabs);
     else
-      hipLaunchKernelGGL((wgrad_partial_kernel<true, 2, 2>), grid, dim3(256),
-                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs);
+      hipLaunchKernelGGL((wgrad_glds_kernel<false>), g0, dim3(256), 0, stream,
+                         x, dy, dW_part, db_part, M, N, K, slabs);
+    if (K % 64) {
+      dim3 g1(1, N / 64, slabs);
+      if (db_part)
+        hipLaunchKernelGGL((wgrad_partial_kernel<true, 2, 2>), g1, dim3(256),
+                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs,
+                           kfull);
+      else
+        hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), g1, dim3(256),
+                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs,
+                           kfull);
+    }
   } else {
-    if (big)
-      hipLaunchKernelGGL((wgrad_partial_kernel<false, 4, 4>), grid, dim3(256),
-                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs);
-    else
-      hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), grid, dim3(256),
-                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs);
+    dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);
+    if (db_part) {
+      if (big)
+        hipLaunchKernelGGL((wgrad_partial_kernel<true, 4, 4>), grid, dim3(256),
+                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs, 0);
+      else
+        hipLaunchKernelGGL((wgrad_partial_kernel<true, 2, 2>), grid, dim3(256),
+                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs, 0);
+    } else {
+      if (big)
+        hipLaunchKernelGGL((wgrad_partial_kernel<false, 4, 4>), grid, dim3(256),
+                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs, 0);
+      else
+        hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), grid, dim3(256),
+                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs, 0);
+    }
   }
   int64_t elems = (int64_t)K * N;
   if (elems <= 4096) {
