@@ -255,12 +255,13 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
 // m-chunk; the launcher falls back to the register-staged kernel for edge
 // tiles (K%64 tails, the head layer's N=4, partial slabs).
 // ---------------------------------------------------------------------------
-template <bool WANT_DB>
+template <bool WANT_DB, int FK, int FN>
 __global__ __launch_bounds__(256) void wgrad_glds_kernel(
     const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
     float* __restrict__ dW_part, float* __restrict__ db_part,
     int M, int N, int K, int slabs) {
-  constexpr int BKm = 64, TK = 64, TN = 64;
+  constexpr int BKm = 64;
+  constexpr int TK = 32 * FK, TN = 32 * FN;
   // UNPADDED linear images (glds cannot scatter past a row pad)
   __shared__ __bf16 Xs[2][BKm][TK];
   __shared__ __bf16 Ys[2][BKm][TN];
@@ -277,25 +278,27 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
   const int m_end = min(M, m_begin + m_per_slab);
   const int nchunks = (m_end - m_begin) / BKm;  // full chunks only (caller
                                                 // guarantees divisibility)
-  f32x4 acc[2][2] = {};
+  f32x4 acc[FK][FN] = {};
   float db_acc = 0.f;
 
-  // one glds instruction stages 64 lanes x 16B = 1KB; a 64x64 bf16 chunk is
-  // 8KB = 8 instructions = 2 per wave.  element e = lane*8 within the 512-
-  // element span; row = e/64, col = e%64.
-  const int g_span = (wave * 2) * 512;          // element base of this
-                                                // wave's first span
+  // one glds instruction stages 64 lanes x 16B = 1KB = 512 bf16; a 64xTW
+  // chunk is 64*TW*2 B = TW/8 instructions = TW/32 per wave.
   auto glds_chunk = [&](int buf, int m0) {
-    for (int j = 0; j < 2; ++j) {
-      const int e0 = g_span + j * 512;
-      const int row = (e0 + lane * 8) / 64;
-      const int col = (e0 + lane * 8) % 64;
+    for (int j = 0; j < FK; ++j) {
+      const int e0 = (wave * FK + j) * 512;
+      const int row = (e0 + lane * 8) / TK;
+      const int col = (e0 + lane * 8) % TK;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)(uintptr_t)(
               &X[(int64_t)(m0 + row) * K + bk + col]),
           (__attribute__((address_space(3))) void*)(uintptr_t)(
               &Xs[buf][0][0] + e0),
           16, 0, 0);
+    }
+    for (int j = 0; j < FN; ++j) {
+      const int e0 = (wave * FN + j) * 512;
+      const int row = (e0 + lane * 8) / TN;
+      const int col = (e0 + lane * 8) % TN;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)(uintptr_t)(
               &dY[(int64_t)(m0 + row) * N + bn + col]),
@@ -322,15 +325,17 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
     // two 32-row reduction batches per 64-row chunk
     for (int half = 0; half < 2; ++half) {
       const int mb = half * 32;
-      bf16x8 af[2], bf_[2];
-      for (int fi = 0; fi < 2; ++fi)
+      bf16x8 af[FK], bf_[FN];
+      for (int fi = 0; fi < FK; ++fi)
         for (int i = 0; i < 8; ++i)
-          af[fi][i] = Xs[cur][mb + kseg * 8 + i][wr * 32 + fi * 16 + row_a];
-      for (int ni = 0; ni < 2; ++ni)
+          af[fi][i] =
+              Xs[cur][mb + kseg * 8 + i][wr * (16 * FK) + fi * 16 + row_a];
+      for (int ni = 0; ni < FN; ++ni)
         for (int i = 0; i < 8; ++i)
-          bf_[ni][i] = Ys[cur][mb + kseg * 8 + i][wc * 32 + ni * 16 + row_a];
-      for (int fi = 0; fi < 2; ++fi)
-        for (int ni = 0; ni < 2; ++ni)
+          bf_[ni][i] =
+              Ys[cur][mb + kseg * 8 + i][wc * (16 * FN) + ni * 16 + row_a];
+      for (int fi = 0; fi < FK; ++fi)
+        for (int ni = 0; ni < FN; ++ni)
           acc[fi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[fi], bf_[ni], acc[fi][ni], 0, 0, 0);
     }
@@ -340,11 +345,11 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
   float* out = dW_part + (int64_t)slab * K * N;
-  for (int fi = 0; fi < 2; ++fi)
-    for (int ni = 0; ni < 2; ++ni) {
-      const int gcol = bn + wc * 32 + ni * 16 + ccol;
+  for (int fi = 0; fi < FK; ++fi)
+    for (int ni = 0; ni < FN; ++ni) {
+      const int gcol = bn + wc * (16 * FN) + ni * 16 + ccol;
       for (int r = 0; r < 4; ++r) {
-        const int grow = bk + wr * 32 + fi * 16 + crow_base + r;
+        const int grow = bk + wr * (16 * FK) + fi * 16 + crow_base + r;
         out[(int64_t)grow * N + gcol] = acc[fi][ni][r];
       }
     }
@@ -363,7 +368,7 @@ __global__ __launch_bounds__(1024) void wgrad_ktail_kernel(
   const int slab = blockIdx.x;
   const int ktail = K - k0;
   const int pairs = ktail * N;
-  const int tid = threadIdx.x;
+  const int tid = blockIdx.y * 1024 + threadIdx.x;
   const int m_per_slab = (M + slabs - 1) / slabs;
   const int m_begin = slab * m_per_slab;
   const int m_end = min(M, m_begin + m_per_slab);
@@ -1547,36 +1552,35 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
   const bool big = K >= 128 && N >= 128 &&
                    (int64_t)ceil_div(K, 128) * ceil_div(N, 128) * slabs >= 1024;
   const int m_per_slab = (M + slabs - 1) / slabs;
-  // glds path: full 64-aligned tiles only; the K tail (K % 64) runs the
-  // register-staged kernel with a k-tile offset.
-  const bool glds_ok = !big && N % 64 == 0 && K >= 64 &&
-                       m_per_slab % 64 == 0 && M % slabs == 0;
+  // glds path (both tile configs): full tiles only; the K tail runs the
+  // skinny per-slab streaming kernel.
+  const int TW = big ? 128 : 64;
+  const bool glds_ok = N % TW == 0 && K >= TW && m_per_slab % 64 == 0 &&
+                       M % slabs == 0;
   if (glds_ok) {
-    const int kfull = K / 64;
-    dim3 g0(kfull, N / 64, slabs);
-    if (db_part)
-      hipLaunchKernelGGL((wgrad_glds_kernel<true>), g0, dim3(256), 0, stream,
-                         x, dy, dW_part, db_part, M, N, K, slabs);
-    else
-      hipLaunchKernelGGL((wgrad_glds_kernel<false>), g0, dim3(256), 0, stream,
-                         x, dy, dW_part, db_part, M, N, K, slabs);
-    if (K % 64) {
-      const int ktail = K - kfull * 64;
-      if (ktail * N <= 1024) {
-        hipLaunchKernelGGL(wgrad_ktail_kernel, dim3(slabs), dim3(ktail * N),
-                           0, stream, x, dy, dW_part, M, N, K, slabs,
-                           kfull * 64);
-      } else {
-        dim3 g1(1, N / 64, slabs);
-        if (db_part)
-          hipLaunchKernelGGL((wgrad_partial_kernel<true, 2, 2>), g1,
-                             dim3(256), 0, stream, x, dy, dW_part, db_part, M,
-                             N, K, slabs, kfull);
-        else
-          hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), g1,
-                             dim3(256), 0, stream, x, dy, dW_part, db_part, M,
-                             N, K, slabs, kfull);
-      }
+    const int kfull = K / TW;
+    dim3 g0(kfull, N / TW, slabs);
+    if (big) {
+      if (db_part)
+        hipLaunchKernelGGL((wgrad_glds_kernel<true, 4, 4>), g0, dim3(256), 0,
+                           stream, x, dy, dW_part, db_part, M, N, K, slabs);
+      else
+        hipLaunchKernelGGL((wgrad_glds_kernel<false, 4, 4>), g0, dim3(256), 0,
+                           stream, x, dy, dW_part, db_part, M, N, K, slabs);
+    } else {
+      if (db_part)
+        hipLaunchKernelGGL((wgrad_glds_kernel<true, 2, 2>), g0, dim3(256), 0,
+                           stream, x, dy, dW_part, db_part, M, N, K, slabs);
+      else
+        hipLaunchKernelGGL((wgrad_glds_kernel<false, 2, 2>), g0, dim3(256), 0,
+                           stream, x, dy, dW_part, db_part, M, N, K, slabs);
+    }
+    if (K % TW) {
+      const int pairs = (K - kfull * TW) * N;
+      dim3 g1(slabs, ceil_div(pairs, 1024));
+      hipLaunchKernelGGL(wgrad_ktail_kernel, g1,
+                         dim3(pairs < 1024 ? pairs : 1024), 0, stream, x, dy,
+                         dW_part, M, N, K, slabs, kfull * TW);
     }
   } else {
     dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);

@@ -75,9 +75,10 @@ def test_gemm_trans_b_with_dact():
                                rtol=3e-2, atol=3e-2)
 
 
-def test_wgrad_matches_and_deterministic():
+@pytest.mark.parametrize("M,K,N", [(8192, 324, 256), (8192, 260, 1024),
+                                   (4096, 256, 256), (2048, 68, 64)])
+def test_wgrad_matches_and_deterministic(M, K, N):
     ext = _ext()
-    M, K, N = 8192, 324, 256
     X = _rand_bf16(M, K, seed=8, scale=0.3)
     dY = _rand_bf16(M, N, seed=9, scale=0.05)
     slabs = 64
