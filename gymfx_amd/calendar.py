@@ -286,3 +286,66 @@ def compute_fx_calendar_table(
     out[:, 8] = open_
     out[:, 9] = ntw
     return out
+
+
+# ---------------------------------------------------------------------------
+# FX rollover (financing) schedule
+# ---------------------------------------------------------------------------
+
+_CCY_LOCATION = {
+    "EUR": "EA19", "USD": "USA", "JPY": "JPN", "GBP": "GBR", "AUD": "AUS",
+    "NZD": "NZL", "CAD": "CAN", "CHF": "CHE",
+}
+
+
+def rollover_rate_lookup(rate_data, location: str, month: str) -> float:
+    """rate_data: iterable of {LOCATION, TIME(YYYY-MM), Value(percent p.a.)}
+    — the reference's monthly central-bank rate schema
+    (simulation_engines/bakeoff.py:104-113).  Falls back to the latest
+    month at or before the requested one."""
+    best = None
+    best_time = ""
+    for row in rate_data:
+        if str(row["LOCATION"]) != location:
+            continue
+        t = str(row["TIME"])
+        if t <= month and t >= best_time:
+            best, best_time = float(row["Value"]), t
+    if best is None:
+        raise ValueError(f"no rollover rate for {location} <= {month}")
+    return best
+
+
+def compute_rollover_schedule(timestamps, instrument: str, rate_data,
+                              rollover_hour_utc: int = 22):
+    """Per-bar financing multiplier [T] (numpy f32): at the first bar at or
+    after each day's rollover time, ``units * (base - quote) / 100 / 365``
+    where units is 3 on Wednesday (weekend rollover) else 1; 0 elsewhere.
+    Financing applied by the env step as ``cash += pos * close * sched[t]``
+    (FXRolloverInterestModule semantics,
+    /root/reference/simulation_engines/nautilus_adapter.py:363-368)."""
+    import numpy as np
+
+    base_ccy, _, quote_ccy = instrument.partition("_")
+    if not quote_ccy and "/" in instrument:
+        base_ccy, _, quote_ccy = instrument.partition("/")
+    base_loc = _CCY_LOCATION.get(base_ccy[:3].upper())
+    quote_loc = _CCY_LOCATION.get(quote_ccy[:3].upper())
+    if base_loc is None or quote_loc is None:
+        raise ValueError(f"unknown currencies in instrument '{instrument}'")
+    ts = np.asarray(timestamps, dtype=np.int64)
+    out = np.zeros(len(ts), dtype=np.float32)
+    day = ts // 86400
+    secs = ts % 86400
+    boundary = rollover_hour_utc * 3600
+    seen_days = set()
+    for i in range(len(ts)):
+        if secs[i] >= boundary and int(day[i]) not in seen_days:
+            seen_days.add(int(day[i]))
+            t = _dt.datetime.fromtimestamp(int(ts[i]), tz=_dt.timezone.utc)
+            month = f"{t.year:04d}-{t.month:02d}"
+            diff = (rollover_rate_lookup(rate_data, base_loc, month)
+                    - rollover_rate_lookup(rate_data, quote_loc, month))
+            units = 3.0 if t.weekday() == 2 else 1.0  # Wed: weekend rollover
+            out[i] = units * diff / 100.0 / 365.0
+    return out

@@ -67,6 +67,9 @@ DEFAULT_VALUES = {
     "action_space_mode": "discrete",
     "continuous_action_threshold": 0.33,
     "min_equity": None,          # default: initial_cash * 0.01
+    "financing_enabled": False,  # FX rollover interest at 22:00 UTC
+    "rollover_rate_data": None,  # monthly central-bank rates (LOCATION/TIME/Value)
+    "rollover_hour_utc": 22,
 
     # PPO training (mode=training; BASELINE configs #2-#4)
     "policy_model": "mlp",       # mlp | lstm (recurrent PPO)

@@ -43,6 +43,8 @@ class MarketTensors:
     # ATR-session filter masks [T] bool (entry window / close zone)
     sess_entry: Optional[torch.Tensor] = None
     sess_close: Optional[torch.Tensor] = None
+    # financing: per-bar rollover multiplier (cash += pos*close*roll_rate[t])
+    roll_rate: Optional[torch.Tensor] = None
     timestamps: Optional[torch.Tensor] = None      # [T] i64 epoch seconds
 
 

@@ -135,6 +135,8 @@ class EnvParams:
     force_close_exposure_penalty_window_hours: float = 4.0
     # oanda calendar obs
     oanda_fx_calendar_obs: bool = False
+    financing_enabled: bool = False
+    rollover_hour_utc: int = 22
     timeframe_hours: float = 0.0
     # event-context overlay
     event_context_execution_overlay: bool = False
@@ -248,6 +250,8 @@ class EnvParams:
         p.force_close_dow = _i(config, "force_close_dow", 4)
         p.force_close_hour = _i(config, "force_close_hour", 20)
 
+        p.financing_enabled = _b(config, "financing_enabled", False)
+        p.rollover_hour_utc = int(config.get("rollover_hour_utc", 22) or 22)
         p.stage_b_force_close_obs = _b(config, "stage_b_force_close_obs", False)
         p.force_close_window_hours = _i(config, "force_close_window_hours", 4)
         p.monday_entry_window_hours = _i(config, "monday_entry_window_hours", 4)

@@ -423,6 +423,12 @@ def step_torch(
     st.started = st.started | first
     st.episode_step = st.episode_step + dec.to(torch.int32)
 
+    # ---- 4b. financing: FX rollover interest at the scheduled bars ------
+    if params.financing_enabled and mt.roll_rate is not None:
+        fin = valid & (st.pos != 0)
+        interest = st.pos * c_px.to(torch.float64) * mt.roll_rate[t].to(torch.float64)
+        st.cash = torch.where(fin, st.cash + interest, st.cash)
+
     # ---- 5. publish (bt_bridge.py:239-248) ------------------------------
     pub = dec
     st.prev_equity = torch.where(pub, st.equity, st.prev_equity)

@@ -59,6 +59,8 @@ class VecFxEnv:
         if min_rows < self.params.window_size + 2:
             raise ValueError("input data is empty or too short for the configured window")
         self.mt: MarketTensors = build_market_tensors(market_data, self.params, self.device)
+        if self.params.financing_enabled:
+            self._build_rollover_schedule()
         self.st: EnvState = alloc_state(self.params, self.device)
         self.total_bars = self.mt.T
         self._rng = np.random.default_rng(self.params.seed)
@@ -88,6 +90,25 @@ class VecFxEnv:
     @property
     def obs_dim(self) -> int:
         return self.params.obs_dim
+
+    def _build_rollover_schedule(self) -> None:
+        """Per-bar financing multipliers, per instrument block
+        (calendar.compute_rollover_schedule; FXRolloverInterestModule
+        parity, nautilus_adapter.py:363-368)."""
+        from ..calendar import compute_rollover_schedule
+
+        md = self.market_data
+        if md.timestamps is None:
+            raise ValueError("financing_enabled requires bar timestamps")
+        rate_data = self.config.get("rollover_rate_data")
+        if not rate_data:
+            raise ValueError("financing_enabled requires rollover_rate_data")
+        sched = np.zeros(md.n_rows, dtype=np.float32)
+        for b in self.instrument_blocks:
+            sched[b["lo"]:b["end"]] = compute_rollover_schedule(
+                md.timestamps[b["lo"]:b["end"]], b["instrument"], rate_data,
+                self.params.rollover_hour_utc)
+        self.mt.roll_rate = torch.from_numpy(sched).to(self.device)
 
     def _assign_instrument_blocks(self) -> None:
         """Round-robin envs over instrument blocks; set per-env episode

@@ -197,6 +197,7 @@ struct GymFxEngine {
     P.calendar = ptr<float>(market, "calendar", false);
     P.sess_entry = ptr<bool>(market, "sess_entry");
     P.sess_close = ptr<bool>(market, "sess_close");
+    P.roll_rate = ptr<float>(market, "roll_rate", false);
 
     // state pointers
     P.cursor = ptr<int>(state, "cursor");

@@ -34,6 +34,7 @@ enum Flags {
   F_HAS_MPLF = 1 << 12,
   F_STAGEB_PENALTY = 1 << 13,
   F_AUTORESET = 1 << 14,
+  F_FINANCING = 1 << 15,
 };
 
 // execution-diagnostics counter indices (== envs/state.py EXEC_COUNTERS)
@@ -98,6 +99,7 @@ struct EnvPtrs {
   const float *force_close;   // [T, 4] or null
   const float *calendar;      // [T, 10] or null
   const bool *sess_entry, *sess_close;  // [T]
+  const float *roll_rate;    // [T] financing multiplier or null
   // state (device, mutable)
   int *cursor;
   bool *started, *terminated;

@@ -384,6 +384,10 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
 
   if (first) P.started[n] = true;
   if (dec) P.episode_step[n] += 1;
+  // ---- 4b. financing: FX rollover interest at the scheduled bars ------
+  if ((K.flags & F_FINANCING) && P.roll_rate && valid && P.pos[n] != 0.0)
+    P.cash[n] += P.pos[n] * (double)c_px * (double)P.roll_rate[t];
+
 
   // ---- 5. publish (bt_bridge.py:239-248) ------------------------------
   const bool pub = dec;

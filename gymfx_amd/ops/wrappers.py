@@ -28,6 +28,7 @@ F_HAS_RELVOL = 1 << 11
 F_HAS_MPLF = 1 << 12
 F_STAGEB_PENALTY = 1 << 13
 F_AUTORESET = 1 << 14
+F_FINANCING = 1 << 15
 
 _SCALING = {"none": 0, "rolling_zscore": 1, "expanding_zscore": 2}
 
@@ -64,6 +65,8 @@ def _flags(p: EnvParams) -> int:
         f |= F_STAGEB_PENALTY
     if p.autoreset:
         f |= F_AUTORESET
+    if p.financing_enabled:
+        f |= F_FINANCING
     return f
 
 
@@ -146,7 +149,7 @@ def _market_dict(mt: MarketTensors) -> Dict[str, torch.Tensor]:
         "sess_close": mt.sess_close,
     }
     for k in ("features", "feat_prefix1", "feat_prefix2", "binary_mask",
-              "force_close", "calendar"):
+              "force_close", "calendar", "roll_rate"):
         v = getattr(mt, k)
         if v is not None:
             d[k] = v

@@ -58,6 +58,10 @@ CONFIGS = {
     "multipair": {"_pairs": 3, "autoreset": True,
                   "strategy_plugin": "direct_fixed_sltp",
                   "sl_pips": 6.0, "tp_pips": 9.0},
+    "financing": {"financing_enabled": True,
+                  "rollover_rate_data": [
+                      {"LOCATION": "EA19", "TIME": "2024-01", "Value": 5.0},
+                      {"LOCATION": "USA", "TIME": "2024-01", "Value": 4.0}]},
 }
 
 
