@@ -70,6 +70,7 @@ DEFAULT_VALUES = {
     "financing_enabled": False,  # FX rollover interest at 22:00 UTC
     "rollover_rate_data": None,  # monthly central-bank rates (LOCATION/TIME/Value)
     "rollover_hour_utc": 22,
+    "enforce_margin_preflight": False,  # deny fills lacking free margin
 
     # PPO training (mode=training; BASELINE configs #2-#4)
     "policy_model": "mlp",       # mlp | lstm (recurrent PPO)

@@ -137,6 +137,7 @@ class EnvParams:
     oanda_fx_calendar_obs: bool = False
     financing_enabled: bool = False
     rollover_hour_utc: int = 22
+    enforce_margin_preflight: bool = False
     timeframe_hours: float = 0.0
     # event-context overlay
     event_context_execution_overlay: bool = False
@@ -251,6 +252,7 @@ class EnvParams:
         p.force_close_hour = _i(config, "force_close_hour", 20)
 
         p.financing_enabled = _b(config, "financing_enabled", False)
+        p.enforce_margin_preflight = _b(config, "enforce_margin_preflight", False)
         p.rollover_hour_utc = int(config.get("rollover_hour_utc", 22) or 22)
         p.stage_b_force_close_obs = _b(config, "stage_b_force_close_obs", False)
         p.force_close_window_hours = _i(config, "force_close_window_hours", 4)
@@ -299,6 +301,9 @@ class EnvParams:
             f = prof.as_floats()
             p.commission = f["commission_rate_per_side"]
             p.slippage = f["quote_adverse_rate_per_side"]
+            p.financing_enabled = p.financing_enabled or prof.financing_enabled
+            p.enforce_margin_preflight = (p.enforce_margin_preflight
+                                          or prof.enforce_margin_preflight)
 
         p.finalize()
         return p

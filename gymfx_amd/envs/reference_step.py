@@ -193,6 +193,12 @@ def step_torch(
         notional = size * fill64
         comm = notional * params.commission
         margin = notional / params.leverage
+        if params.enforce_margin_preflight:
+            # margin preflight denial (nautilus_gym.py:128-171): drop the
+            # order, count it, keep the episode alive
+            denied = open_m & (st.cash < margin + comm)
+            st.exec_diag[:, _E["margin_preflight_denied"]] += denied.to(torch.int32)
+            open_m = open_m & ~denied
         st.cash = torch.where(open_m, st.cash - margin - comm, st.cash)
         st.margin_used = torch.where(open_m, margin, st.margin_used)
         st.commission_paid = torch.where(open_m, st.commission_paid + comm, st.commission_paid)

@@ -33,6 +33,7 @@ EXEC_COUNTERS = (
     "session_force_closes",
     "bracket_sl_fills",
     "bracket_tp_fills",
+    "margin_preflight_denied",
 )
 
 ACTION_COUNTERS = (

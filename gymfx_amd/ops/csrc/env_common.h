@@ -35,6 +35,7 @@ enum Flags {
   F_STAGEB_PENALTY = 1 << 13,
   F_AUTORESET = 1 << 14,
   F_FINANCING = 1 << 15,
+  F_PREFLIGHT = 1 << 16,
 };
 
 // execution-diagnostics counter indices (== envs/state.py EXEC_COUNTERS)
@@ -56,6 +57,7 @@ enum ExecCounter {
   E_SESSION_FORCE_CLOSES,
   E_BRACKET_SL_FILLS,
   E_BRACKET_TP_FILLS,
+  E_MARGIN_PREFLIGHT_DENIED,
   EXEC_COUNTER_N,
 };
 

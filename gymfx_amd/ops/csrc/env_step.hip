@@ -181,6 +181,12 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
     double notional = size * fill;
     double comm = notional * K.commission;
     double margin = notional / K.leverage;
+    if ((K.flags & F_PREFLIGHT) && P.cash[n] < margin + comm) {
+      // margin preflight denial (nautilus_gym.py:128-171 semantics):
+      // the order is dropped, diagnostics incremented, episode continues
+      ediag[E_MARGIN_PREFLIGHT_DENIED] += 1;
+      goto preflight_denied;
+    }
     P.cash[n] -= margin + comm;
     P.margin_used[n] = margin;
     P.commission_paid[n] += comm;
@@ -193,6 +199,7 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
       P.br_sl[n] = P.pend_sl[n];
       P.br_tp[n] = P.pend_tp[n];
     }
+  preflight_denied:;
   }
   P.pend_close[n] = false;
   P.pend_open_dir[n] = 0;

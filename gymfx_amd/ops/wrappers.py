@@ -29,6 +29,7 @@ F_HAS_MPLF = 1 << 12
 F_STAGEB_PENALTY = 1 << 13
 F_AUTORESET = 1 << 14
 F_FINANCING = 1 << 15
+F_PREFLIGHT = 1 << 16
 
 _SCALING = {"none": 0, "rolling_zscore": 1, "expanding_zscore": 2}
 
@@ -67,6 +68,8 @@ def _flags(p: EnvParams) -> int:
         f |= F_AUTORESET
     if p.financing_enabled:
         f |= F_FINANCING
+    if p.enforce_margin_preflight:
+        f |= F_PREFLIGHT
     return f
 
 

@@ -146,3 +146,32 @@ def test_data_exhaustion_terminates():
             break
     assert terminated
     assert env.bridge_state(0)["bar_index"] == 6
+
+
+def test_margin_preflight_denies_oversized_order():
+    """enforce_margin_preflight: an order whose margin+commission exceeds
+    free cash is denied with a diagnostics count, the episode continues
+    (nautilus_gym.py:128-171 / bakeoff.py:166-176 margin-rejection fixture)."""
+    import torch
+
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    md = synthetic_ohlcv(100, seed=4, start_price=1.1)
+    base = {"n_envs": 1, "device": "cpu", "window_size": 4,
+            "initial_cash": 100.0, "position_size": 1_000_000.0,
+            "leverage": 1.0, "env_start_mode": "zero", "seed": 0}
+    # preflight ON: denied
+    env = build_vec_environment({**base, "enforce_margin_preflight": True}, md)
+    env.reset(seed=0)
+    env.step(torch.tensor([1]))
+    env.step(torch.tensor([0]))   # fill bar
+    assert float(env.st.pos[0]) == 0.0
+    assert env.execution_diagnostics(0)["margin_preflight_denied"] == 1
+    assert not bool(env.st.terminated[0])
+    # preflight OFF: fills (cash goes deeply negative -> bust path)
+    env2 = build_vec_environment(base, md)
+    env2.reset(seed=0)
+    env2.step(torch.tensor([1]))
+    env2.step(torch.tensor([0]))
+    assert float(env2.st.pos[0]) != 0.0

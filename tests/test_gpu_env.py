@@ -58,6 +58,8 @@ CONFIGS = {
     "multipair": {"_pairs": 3, "autoreset": True,
                   "strategy_plugin": "direct_fixed_sltp",
                   "sl_pips": 6.0, "tp_pips": 9.0},
+    "preflight": {"enforce_margin_preflight": True, "leverage": 1.0,
+                  "position_size": 50000.0, "initial_cash": 10000.0},
     "financing": {"financing_enabled": True,
                   "rollover_rate_data": [
                       {"LOCATION": "EA19", "TIME": "2024-01", "Value": 5.0},
