@@ -87,6 +87,13 @@ def main(argv=None) -> None:
         from .algo.optimize import optimize_from_config  # noqa: PLC0415
 
         summary = optimize_from_config(config)
+    elif (config.get("mode") == "inference" and config.get("checkpoint_file")
+          and int(config.get("n_envs", 1) or 1) > 1):
+        # vectorized policy evaluation (checkpoint -> greedy rollout);
+        # without a checkpoint the classic driver loop below runs instead
+        from .algo.evaluate import evaluate_from_config  # noqa: PLC0415
+
+        summary = evaluate_from_config(config)
     else:
         summary = run_env(config)
 

@@ -59,3 +59,41 @@ def test_checkpoint_rejects_mismatch(tmp_path):
     other = _make("lstm")
     with pytest.raises(ValueError, match="policy"):
         load_checkpoint(other, path)
+
+
+def test_train_then_evaluate_roundtrip(tmp_path):
+    """mode=training with a checkpoint, then vectorized greedy evaluation
+    from that checkpoint (the serving-side counterpart of training)."""
+    import json
+
+    from gymfx_amd.main import main as cli_main
+
+    ckpt = str(tmp_path / "m.pt")
+    train_res = tmp_path / "train.json"
+    eval_res = tmp_path / "eval.json"
+    common = [
+        "--quiet_mode", "true",
+        "--data_feed_plugin", "synthetic_data_feed",
+        "--synthetic_rows", "2000",
+        "--n_envs", "16",
+        "--window_size", "8",
+        "--position_size", "1000.0",
+        "--rollout_steps", "16",
+        "--minibatches", "2",
+        "--ppo_epochs", "1",
+        "--hidden_size", "16",
+        "--seed", "0",
+        "--checkpoint_file", ckpt,
+        "--save_config", "",
+    ]
+    cli_main(["--mode", "training", "--train_updates", "2",
+              "--results_file", str(train_res), *common])
+    out = json.loads(train_res.read_text())
+    assert out["updates"] == 2 and out["checkpoint_file"] == ckpt
+    cli_main(["--mode", "inference", "--eval_steps", "64",
+              "--results_file", str(eval_res), *common])
+    ev = json.loads(eval_res.read_text())
+    assert ev["mode"] == "inference"
+    assert ev["eval_steps"] == 64 and ev["n_envs"] == 16
+    assert ev["mean_final_equity"] > 0
+    assert ev["total_trades"] >= 0
