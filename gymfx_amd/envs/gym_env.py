@@ -178,6 +178,8 @@ class GymFxEnv(spaces.Env):
                 act_t = torch.tensor([int(action)], dtype=torch.int64)
             except (TypeError, ValueError):
                 act_t = torch.tensor([0], dtype=torch.int64)
+        if self.vec.params.event_context_execution_overlay:
+            self._last_info_extras = self._event_context_info(act_t)
         if self._audit_path:
             st = self.vec.st
             self._prev_audit_state = {
@@ -312,6 +314,42 @@ class GymFxEnv(spaces.Env):
         tf_h = float(md.timeframe_hours() or 1.0) or 1.0
         return compute_fx_calendar_features(ts, timeframe_hours=tf_h)
 
+    def _event_context_info(self, act_t) -> Dict[str, Any]:
+        """Per-step event-context info block (env.py:383-440 field set);
+        the authoritative action rewrite happens inside the fused kernel —
+        this mirrors the decision for observability."""
+        p = self.vec.params
+        mt = self.vec.mt
+        st = self.vec.st
+        row = int(min(int(st.cursor[0].item()), self.total_bars - 1))
+        no_trade = float(mt.ev_no_trade[row].item()) if mt.ev_no_trade is not None else 0.0
+        spread_m = float(mt.ev_spread_mult[row].item()) if mt.ev_spread_mult is not None else 1.0
+        slip_m = float(mt.ev_slip_mult[row].item()) if mt.ev_slip_mult is not None else 1.0
+        active = no_trade >= p.event_context_no_trade_threshold
+        position = int(torch.sign(st.pos[0]).item())
+        before = int(act_t.reshape(-1)[0].item()) if act_t.dtype == torch.int64 else 0
+        after = before
+        blocked = forced = False
+        if active:
+            if p.event_context_force_flat and position != 0:
+                after, forced = 3, True
+            elif (p.event_context_block_new_entries and position == 0
+                  and before in (1, 2)):
+                after, blocked = 0, True
+        return {
+            "event_context_no_trade_value": no_trade,
+            "event_context_no_trade_active": 1.0 if active else 0.0,
+            "event_context_spread_stress_multiplier": spread_m,
+            "event_context_slippage_stress_multiplier": slip_m,
+            "event_context_execution_overlay": True,
+            "event_context_action_before_overlay": before,
+            "event_context_action_after_overlay": after,
+            "event_context_action_overridden": bool(after != before),
+            "event_context_blocked_entry": blocked,
+            "event_context_forced_flat": forced,
+            "event_context_position_before_overlay": position,
+        }
+
     def _make_info(self) -> Dict[str, Any]:
         bs = self.vec.bridge_state(0)
         info: Dict[str, Any] = {
@@ -327,6 +365,7 @@ class GymFxEnv(spaces.Env):
             "action_diagnostics": self.vec.action_diagnostics(0),
             "execution_diagnostics": self.vec.execution_diagnostics(0),
         }
+        info.update(self._last_info_extras)
         if self.stage_b_force_close_obs:
             step_idx = max(0, min(bs["bar_index"], self.total_bars))
             info.update(self._force_close_features(step_idx))

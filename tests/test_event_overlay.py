@@ -76,3 +76,59 @@ def test_overlay_inactive_without_flag():
     out = env.step(torch.tensor([1]))
     assert int(out["coerced_action"][0]) == 1
     assert env.execution_diagnostics(0)["event_context_blocked_entries"] == 0
+
+
+def test_gym_wrapper_event_context_info_fields():
+    """Per-step event_context_* info block (env.py:383-440 field parity)."""
+    import numpy as np
+    import torch
+
+    from gymfx_amd import build_environment
+    from gymfx_amd.config import DEFAULT_VALUES
+    from gymfx_amd.data.feed import synthetic_ohlcv
+    from gymfx_amd.plugins import load_plugin
+
+    md = synthetic_ohlcv(200, seed=3, extra_feature_columns=1)
+    md.columns["EV_NO_TRADE"] = np.ones(md.n_rows, dtype=np.float64)
+    cfg = dict(DEFAULT_VALUES)
+    cfg.update({
+        "window_size": 8, "device": "cpu", "seed": 3,
+        "data_feed_plugin": "synthetic_data_feed",
+        "event_context_execution_overlay": True,
+        "event_context_block_new_entries": True,
+        "event_context_no_trade_column": "EV_NO_TRADE",
+        "event_context_no_trade_threshold": 0.5,
+    })
+    plugins = {}
+    for group, key in [("data_feed.plugins", "data_feed_plugin"),
+                       ("broker.plugins", "broker_plugin"),
+                       ("strategy.plugins", "strategy_plugin"),
+                       ("preprocessor.plugins", "preprocessor_plugin"),
+                       ("reward.plugins", "reward_plugin"),
+                       ("metrics.plugins", "metrics_plugin")]:
+        klass, _ = load_plugin(group, cfg[key])
+        plugins[key] = klass(cfg)
+
+    class MdFeed:
+        plugin_params = {}
+        def set_params(self, **kw): pass
+        def load_data(self, config): return md
+        def build_market(self, data, config): return data
+
+    env = build_environment(
+        config=cfg, data_feed_plugin=MdFeed(),
+        broker_plugin=plugins["broker_plugin"],
+        strategy_plugin=plugins["strategy_plugin"],
+        preprocessor_plugin=plugins["preprocessor_plugin"],
+        reward_plugin=plugins["reward_plugin"],
+        metrics_plugin=plugins["metrics_plugin"])
+    env.reset(seed=3)
+    obs, r, term, trunc, info = env.step(1)  # entry attempt under no-trade
+    assert info["event_context_no_trade_active"] == 1.0
+    assert info["event_context_action_before_overlay"] == 1
+    assert info["event_context_action_after_overlay"] == 0
+    assert info["event_context_blocked_entry"] is True
+    assert info["event_context_spread_stress_multiplier"] == 1.0
+    # the kernel really blocked it: no position after fill bar
+    env.step(0)
+    assert info["event_context_position_before_overlay"] == 0
