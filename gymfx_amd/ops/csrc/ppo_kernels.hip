@@ -79,16 +79,19 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
     int M, int N, int K) {
   constexpr int BM = 64, BK = 32;
   constexpr int BN = 32 * NFRAG;
-  constexpr int LDA = BK + 8;  // bf16 elems; 16B-aligned rows, bank-spread
+  // TRANS_B images are UNPADDED linear (required by the glds staging; the
+  // b128 fragment-read bank multiplicity is already at its minimum for both
+  // strides).  !TRANS_B keeps the +8 pad with register staging.
+  constexpr int LDA = TRANS_B ? BK : (BK + 8);
   constexpr int BVEC = NFRAG / 2;  // bf16x8 loads per thread for the B tile
   // B tile layout follows the GLOBAL layout so staging loads are always
   // contiguous bf16x8 (a strided 2-byte gather of B was 6x slower than the
   // whole GEMM): TRANS_B stages [n][k], !TRANS_B stages [k][n]; the
   // !TRANS_B fragment reads transpose out of LDS instead.
   constexpr int BS_ROWS = TRANS_B ? BN : BK;
-  constexpr int BS_LD = (TRANS_B ? BK : BN) + 8;
-  __shared__ __bf16 As[2][BM][LDA];
-  __shared__ __bf16 Bs[2][BS_ROWS][BS_LD];
+  constexpr int BS_LD = TRANS_B ? BK : (BN + 8);
+  __shared__ __align__(16) __bf16 As[2][BM][LDA];
+  __shared__ __align__(16) __bf16 Bs[2][BS_ROWS][BS_LD];
 
   const int bm = blockIdx.x * BM;
   const int bn = blockIdx.y * BN;
@@ -174,13 +177,52 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
   };
 
   const int ktiles = (K + BK - 1) / BK;
-  load_tile(0);
-  store_tile(0);
+  // glds staging (TRANS_B, in-bounds blocks, full k-chunks): 16B per lane
+  // into the linear images — no staging registers, loads still in flight
+  // across the MFMA section and drained by the barrier's vmcnt(0).
+  const bool can_glds = TRANS_B && (bm + BM <= M) && (bn + BN <= N);
+  auto glds_tile = [&](int buf, int k0) {
+    {  // A tile [BM][BK]: BM*BK*2 B = BM*BK/512 wave-instr
+      const int e0 = wave * 512;  // BM=64: 1 instr/wave
+      const int row = (e0 + lane * 8) / BK;
+      const int col = (e0 + lane * 8) % BK;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)(
+              &A[(int64_t)(bm + row) * K + k0 + col]),
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              &As[buf][0][0] + e0),
+          16, 0, 0);
+    }
+    for (int j = 0; j < BVEC; ++j) {  // B tile [BN][BK]
+      const int e0 = (wave * BVEC + j) * 512;
+      const int row = (e0 + lane * 8) / BK;
+      const int col = (e0 + lane * 8) % BK;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)(
+              &B[(int64_t)(bn + row) * K + k0 + col]),
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              &Bs[buf][0][0] + e0),
+          16, 0, 0);
+    }
+  };
+  auto stage = [&](int buf, int k0) {
+    if (can_glds && k0 + BK <= K) {
+      glds_tile(buf, k0);
+    } else {
+      load_tile(k0);
+      store_tile(buf);
+    }
+  };
+
+  stage(0, 0);
   __syncthreads();
 
   for (int kt = 0; kt < ktiles; ++kt) {
     const int cur = kt & 1;
-    if (kt + 1 < ktiles) load_tile((kt + 1) * BK);
+    const bool nxt_glds =
+        kt + 1 < ktiles && can_glds && (kt + 1) * BK + BK <= K;
+    if (nxt_glds) glds_tile(1 - cur, (kt + 1) * BK);
+    else if (kt + 1 < ktiles) load_tile((kt + 1) * BK);
 
     bf16x8 af[2], bf[NFRAG];
     for (int mi = 0; mi < 2; ++mi)
@@ -202,7 +244,7 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
 
-    if (kt + 1 < ktiles) store_tile(1 - cur);
+    if (kt + 1 < ktiles && !nxt_glds) store_tile(1 - cur);
     __syncthreads();
   }
 
