@@ -72,3 +72,29 @@ def test_replay_no_future_leakage():
     r2 = ReplayAdapter().run(_cfg(), md, acts)
     assert r1["event_hash"] == r2["event_hash"]
     assert r1["engine"] == r2["engine"]
+
+
+@pytest.mark.parametrize("trial", range(8))
+def test_replay_reconciles_under_random_configs(trial):
+    """Property sweep: random market/cost/strategy configs must all
+    reconcile engine vs the independent ledger (robustness currency of
+    the reference's bakeoff)."""
+    rng = np.random.default_rng(100 + trial)
+    md = synthetic_ohlcv(int(rng.integers(150, 400)),
+                         seed=int(rng.integers(0, 1000)),
+                         vol=float(rng.uniform(1e-4, 1e-3)),
+                         drift=float(rng.uniform(-2e-4, 2e-4)))
+    cfg = _cfg(
+        commission=float(rng.choice([0.0, 1e-5, 5e-5])),
+        slippage=float(rng.choice([0.0, 5e-6, 2e-5])),
+        position_size=float(rng.choice([100.0, 1000.0, 5000.0])),
+        leverage=float(rng.choice([1.0, 10.0, 50.0])),
+    )
+    if rng.random() < 0.5:
+        cfg.update(strategy_plugin="direct_fixed_sltp",
+                   sl_pips=float(rng.uniform(2, 15)),
+                   tp_pips=float(rng.uniform(2, 25)))
+    acts = rng.integers(0, 3, size=int(rng.integers(80, 250))).tolist()
+    res = ReplayAdapter().run(cfg, md, acts)
+    assert res["reconciled"], (trial, res["reconciliation"],
+                               res["engine"], res["oracle"])
