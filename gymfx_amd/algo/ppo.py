@@ -544,13 +544,33 @@ def train_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
         load_checkpoint(trainer, ckpt_path)
         resumed = True
     updates = int(config.get("train_updates", 10))
+    trace_path = config.get("trace_file")
+    tracer = writer = None
+    if trace_path:
+        from ..utils.trace import PhaseTimer, TraceWriter
+
+        tracer = PhaseTimer(env.device)
+        writer = TraceWriter(str(trace_path))
     t0 = time.perf_counter()
     history: List[Dict[str, float]] = []
     for u in range(updates):
-        stats = trainer.train_update()
+        if tracer is None:
+            stats = trainer.train_update()
+        else:
+            if trainer.use_graphs and not trainer._graphs_ready:
+                trainer._capture_graphs()
+            with tracer.phase("rollout"):
+                trainer.collect_rollout()
+            with tracer.phase("update"):
+                stats = trainer.update()
+            phases = tracer.drain()
+            writer.write({"update": u, "global_step": trainer.global_step,
+                          "phases_ms": phases, **stats})
         history.append(stats)
         if not config.get("quiet_mode") and (u % max(1, updates // 10) == 0):
             print(f"update {u}: {stats}")
+    if writer is not None:
+        writer.close()
     wall = time.perf_counter() - t0
     if ckpt_path:
         from ..utils.checkpoint import save_checkpoint

@@ -89,4 +89,5 @@ DEFAULT_VALUES = {
     "train_updates": 10,
     "checkpoint_file": None,     # save/resume path (mode=training)
     "resume": False,
+    "trace_file": None,          # per-update phase-timing JSONL (HIP events)
 }

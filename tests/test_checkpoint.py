@@ -97,3 +97,30 @@ def test_train_then_evaluate_roundtrip(tmp_path):
     assert ev["eval_steps"] == 64 and ev["n_envs"] == 16
     assert ev["mean_final_equity"] > 0
     assert ev["total_trades"] >= 0
+
+
+def test_trace_file_phase_timings(tmp_path):
+    """trace_file emits one JSONL record per update with rollout/update
+    phase wall-clock (SURVEY §5.1 observability upgrade)."""
+    import json
+
+    from gymfx_amd.main import main as cli_main
+
+    trace = tmp_path / "trace.jsonl"
+    cli_main([
+        "--mode", "training", "--quiet_mode", "true",
+        "--data_feed_plugin", "synthetic_data_feed",
+        "--synthetic_rows", "1200",
+        "--n_envs", "16", "--window_size", "8",
+        "--rollout_steps", "16", "--minibatches", "2", "--ppo_epochs", "1",
+        "--hidden_size", "16", "--seed", "0", "--train_updates", "3",
+        "--trace_file", str(trace),
+        "--results_file", str(tmp_path / "r.json"), "--save_config", "",
+    ])
+    lines = [json.loads(l) for l in trace.read_text().splitlines()]
+    assert len(lines) == 3
+    for i, rec in enumerate(lines):
+        assert rec["update"] == i
+        assert rec["phases_ms"]["rollout"] > 0
+        assert rec["phases_ms"]["update"] > 0
+        assert "pi_loss" in rec
