@@ -52,6 +52,9 @@ class PPOConfig:
                              # Measured neutral-to-slightly-negative at
                              # N=4096 (per-kernel latency does not shrink
                              # with width) — kept for bigger fleets.
+    fuse_sample: bool = True  # sample the action inside the env-step kernel
+                              # (native engine): one fewer launch per rollout
+                              # step; bitwise == the sample_head kernel.
 
     @classmethod
     def from_config(cls, cfg: Dict[str, Any]) -> "PPOConfig":
@@ -74,6 +77,7 @@ class PPOConfig:
             "bptt_len": "bptt_len",
             "rollout_streams": "rollout_streams",
             "fused_rollout": "fused_rollout",
+            "fuse_sample": "fuse_sample",
         }
         for attr, key in mapping.items():
             if cfg.get(key) is not None:
@@ -197,6 +201,9 @@ class PPOTrainer:
                        and N % 2 == 0)
         self._fused = (self.device.type == "cuda" and cfg.fused_rollout
                        and not self.recurrent and cfg.hidden == 256)
+        self._fuse_sample = (self.device.type == "cuda" and cfg.fuse_sample
+                             and getattr(env, "_native", None) is not None
+                             and env.params.action_space_mode != "continuous")
         if self._split:
             self._s2 = torch.cuda.Stream()
             self.acts_half = [self.model.alloc_acts(N // 2),
@@ -241,7 +248,8 @@ class PPOTrainer:
                 head = None
             else:
                 head = model.forward(obs_t, acts)
-            if head is not None:
+            fuse = self._fuse_sample and head is not None
+            if head is not None and not fuse:
                 api.sample_head(
                     head, self.sample_seed, t,
                     self.act_buf[t][lo:hi], self.logp_buf[t][lo:hi],
@@ -249,12 +257,18 @@ class PPOTrainer:
                     row_offset=lo,
                 )
             # fused env kernel writes reward/done into the rollout slabs and
-            # the NEXT observation (bf16) into obs_buf[t+1] directly
+            # the NEXT observation (bf16) into obs_buf[t+1] directly; with
+            # fuse_sample it also samples the action from `head` itself.
             nxt = self.obs_buf[t + 1] if t + 1 < T else self.obs_bf16_step
             env.step(
                 self.act_buf[t], reward_out=self.rew_buf[t],
                 terminated_out=self.done_buf[t], obs_bf16_out=nxt,
                 env_lo=lo, env_hi=hi,
+                head=head if fuse else None,
+                logp_out=self.logp_buf[t] if fuse else None,
+                value_out=self.val_buf[t] if fuse else None,
+                step_base=self.step_base if fuse else None,
+                sample_seed=self.sample_seed, sample_step=t,
             )
             if rec:
                 # fresh episode -> zero recurrent state

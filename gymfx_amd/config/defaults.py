@@ -87,6 +87,7 @@ DEFAULT_VALUES = {
     "vf_coef": 0.5,
     "max_grad_norm": 0.5,
     "train_updates": 10,
+    "fuse_sample": True,         # sample inside the env-step kernel (GPU)
     "checkpoint_file": None,     # save/resume path (mode=training)
     "resume": False,
     "trace_file": None,          # per-update phase-timing JSONL (HIP events)

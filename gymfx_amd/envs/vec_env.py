@@ -168,12 +168,23 @@ class VecFxEnv:
              reward_out: Optional[torch.Tensor] = None,
              terminated_out: Optional[torch.Tensor] = None,
              obs_bf16_out: Optional[torch.Tensor] = None,
-             env_lo: int = 0, env_hi: int = 0) -> Dict[str, torch.Tensor]:
+             env_lo: int = 0, env_hi: int = 0,
+             head: Optional[torch.Tensor] = None,
+             logp_out: Optional[torch.Tensor] = None,
+             value_out: Optional[torch.Tensor] = None,
+             step_base: Optional[torch.Tensor] = None,
+             sample_seed: int = 0,
+             sample_step: int = 0) -> Dict[str, torch.Tensor]:
         """Advance all envs. Returns dict with obs/reward/terminated tensors.
 
         reward_out / terminated_out / obs_bf16_out: optional preallocated
         device tensors the fused kernels write directly (rollout slabs — no
-        copy kernels; used by the PPO trainer's hipGraph bodies)."""
+        copy kernels; used by the PPO trainer's hipGraph bodies).
+
+        head / logp_out / value_out / step_base / sample_seed / sample_step:
+        fused policy sampling (native engine only) — the step kernel samples
+        the action from `head` itself, writing into `actions`, saving one
+        kernel launch per rollout step."""
         if not isinstance(actions, torch.Tensor):
             actions = torch.as_tensor(actions, device=self.device)
         actions = actions.to(self.device)
@@ -181,9 +192,13 @@ class VecFxEnv:
             # fused HIP path: step + autoreset + obs in two kernel launches
             info = self._native.step(actions, self._obs, reward_out,
                                      terminated_out, obs_bf16_out,
-                                     env_lo, env_hi)
+                                     env_lo, env_hi, head, logp_out,
+                                     value_out, step_base, sample_seed,
+                                     sample_step)
             info["obs"] = self._obs
             return info
+        if head is not None:
+            raise ValueError("fused sampling requires the native engine")
         if env_lo != 0 or env_hi not in (0, self.params.n_envs):
             raise ValueError("env range stepping requires the native engine")
         info = step_torch(self.st, self.mt, self.params, actions)

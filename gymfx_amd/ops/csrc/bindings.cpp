@@ -272,7 +272,12 @@ struct GymFxEngine {
                 c10::optional<torch::Tensor> reward_out,
                 c10::optional<torch::Tensor> terminated_out,
                 c10::optional<torch::Tensor> obs_bf16_out,
-                int64_t env_lo, int64_t env_hi) {
+                int64_t env_lo, int64_t env_hi,
+                c10::optional<torch::Tensor> head,
+                c10::optional<torch::Tensor> logp_out,
+                c10::optional<torch::Tensor> value_out,
+                c10::optional<torch::Tensor> step_base,
+                int64_t sample_seed, int64_t sample_step) {
     TORCH_CHECK(actions.is_cuda() == state["cursor"].is_cuda(),
                 "actions must live on the env device");
     TORCH_CHECK(actions.numel() == K.n_envs, "actions numel != n_envs");
@@ -288,6 +293,45 @@ struct GymFxEngine {
                   "discrete mode wants int64 actions");
     }
     P.actions = actions.data_ptr();
+    // optional fused sampling: the step kernel samples from `head` and
+    // writes actions/logp/value itself (one fewer launch per step).
+    P.head = nullptr;
+    P.actions_out = nullptr;
+    P.logp_out = nullptr;
+    P.value_out = nullptr;
+    P.step_base = nullptr;
+    if (head.has_value()) {
+      TORCH_CHECK(!continuous, "fused sampling requires discrete actions");
+      TORCH_CHECK(head->is_contiguous() &&
+                      head->scalar_type() == torch::kFloat32 &&
+                      head->dim() == 2 &&
+                      head->size(0) >= (env_hi <= 0 ? K.n_envs : env_hi) - env_lo,
+                  "head must be contiguous f32 [env_cnt, A+1]");
+      TORCH_CHECK(logp_out.has_value() && value_out.has_value(),
+                  "fused sampling needs logp_out and value_out");
+      TORCH_CHECK(logp_out->is_contiguous() &&
+                      logp_out->scalar_type() == torch::kFloat32 &&
+                      logp_out->numel() == K.n_envs,
+                  "logp_out must be contiguous f32 [n_envs]");
+      TORCH_CHECK(value_out->is_contiguous() &&
+                      value_out->scalar_type() == torch::kFloat32 &&
+                      value_out->numel() == K.n_envs,
+                  "value_out must be contiguous f32 [n_envs]");
+      P.head = head->data_ptr<float>();
+      P.actions_out = actions.data_ptr<int64_t>();
+      P.logp_out = logp_out->data_ptr<float>();
+      P.value_out = value_out->data_ptr<float>();
+      K.sample_nact = (int)head->size(1) - 1;
+      K.sample_seed = (unsigned long long)sample_seed;
+      K.sample_step = (long long)sample_step;
+      if (step_base.has_value()) {
+        TORCH_CHECK(step_base->scalar_type() == torch::kUInt64 ||
+                        step_base->scalar_type() == torch::kInt64,
+                    "step_base must be a 64-bit counter tensor");
+        P.step_base = reinterpret_cast<const unsigned long long*>(
+            step_base->data_ptr());
+      }
+    }
     // optional external outputs (rollout slabs): the kernels write straight
     // into the trainer's [T, N] buffers — no copy kernels afterwards.
     torch::Tensor rew = outputs["reward"];
@@ -803,7 +847,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("obs_out"), py::arg("reward_out") = py::none(),
            py::arg("terminated_out") = py::none(),
            py::arg("obs_bf16_out") = py::none(), py::arg("env_lo") = 0,
-           py::arg("env_hi") = 0)
+           py::arg("env_hi") = 0, py::arg("head") = py::none(),
+           py::arg("logp_out") = py::none(), py::arg("value_out") = py::none(),
+           py::arg("step_base") = py::none(), py::arg("sample_seed") = 0,
+           py::arg("sample_step") = 0)
       .def("build_obs", &gymfx::GymFxEngine::build_obs, py::arg("obs_out"),
            py::arg("obs_bf16_out") = py::none(), py::arg("env_lo") = 0,
            py::arg("env_hi") = 0);

@@ -71,6 +71,50 @@ enum ActCounter {
   ACT_COUNTER_N,
 };
 
+// ---------------------------------------------------------------------------
+// Shared categorical-sampler math (used by sample_head_kernel AND the fused
+// env_step sampling path so both are bitwise-identical by construction).
+// libm expf/logf are CALLS on gfx950; these are single v_exp_f32/v_log_f32.
+// hipcc-only: bindings.cpp (plain host c++) includes this header for the
+// structs and must not see the amdgcn builtins.
+// ---------------------------------------------------------------------------
+#ifdef __HIPCC__
+GFX_DEV float fast_exp(float x) {
+  return __builtin_amdgcn_exp2f(x * 1.4426950408889634f);
+}
+GFX_DEV float fast_log(float x) {
+  return __builtin_amdgcn_logf(x) * 0.6931471805599453f;
+}
+// Counter-based RNG: deterministic in (seed, step, env).
+GFX_DEV uint64_t splitmix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+GFX_DEV float head_logz(const float* row, int n_actions) {
+  float mx = row[0];
+  for (int j = 1; j < n_actions; ++j) mx = fmaxf(mx, row[j]);
+  float z = 0.f;
+  for (int j = 0; j < n_actions; ++j) z += fast_exp(row[j] - mx);
+  return fast_log(z) + mx;
+}
+GFX_DEV int sample_categorical(const float* row, int n_actions, float logz,
+                               uint64_t seed, uint64_t step, int64_t mg) {
+  const uint64_t r =
+      splitmix64(seed ^ (step * 0x51E1F5ull + (uint64_t)mg * 0x9E37ull));
+  float u = (float)((r >> 11) * (1.0 / 9007199254740992.0));  // [0,1)
+  u = fminf(u, 0.999999f);
+  float c = 0.f;
+  int a = n_actions - 1;
+  for (int j = 0; j < n_actions; ++j) {
+    c += fast_exp(row[j] - logz);
+    if (u < c) { a = j; break; }
+  }
+  return a;
+}
+#endif  // __HIPCC__
+
 struct EnvParamsK {
   int n_envs, T, window, n_features;
   int reward_id, strategy_id, prep_id;
@@ -89,6 +133,10 @@ struct EnvParamsK {
   double sl_shrink_alpha, tp_shrink_alpha, min_k_sl, min_rr, mplf;
   double fc_pen_coef, fc_pen_window_hours;
   double feature_clip, overlay_threshold;
+  // fused policy sampling (set per step() call when head != null)
+  unsigned long long sample_seed;
+  long long sample_step;
+  int sample_nact;
 };
 
 struct EnvPtrs {
@@ -133,6 +181,15 @@ struct EnvPtrs {
   int *exec_diag;            // [N, EXEC_COUNTER_N]
   int *act_diag;             // [N, ACT_COUNTER_N]
   float *raw_abs_sum, *raw_min, *raw_max;
+  // fused policy sampling (optional; saves one kernel launch per rollout
+  // step): when `head` is non-null the step kernel samples the action
+  // itself — identical math to sample_head_kernel — and writes it to
+  // actions_out before decoding it.
+  const float *head;         // [env_cnt, sample_nact+1] f32 (rows local to env_lo)
+  int64_t *actions_out;      // sampled action (same buffer `actions` reads)
+  float *logp_out;           // [N] f32
+  float *value_out;          // [N] f32
+  const unsigned long long *step_base;  // device counter (graph-replayable)
   // step inputs/outputs
   const void *actions;       // int64 (discrete) or float32 (continuous)
   float *reward_out, *base_reward_out, *penalty_out;

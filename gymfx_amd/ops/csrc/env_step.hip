@@ -104,6 +104,22 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
   int* ediag = P.exec_diag + (int64_t)n * EXEC_COUNTER_N;
   int* adiag = P.act_diag + (int64_t)n * ACT_COUNTER_N;
 
+  // ---- fused policy sampling (optional) ------------------------------
+  // One thread already owns env row n; sampling here (identical math to
+  // sample_head_kernel, shared helpers in env_common.h) removes one kernel
+  // launch per rollout step from the latency-bound per-step chain.
+  if (P.head) {
+    const int A = K.sample_nact;
+    const float* row = P.head + (int64_t)(n - env_lo) * (A + 1);
+    const float logz = head_logz(row, A);
+    uint64_t stp = (uint64_t)K.sample_step;
+    if (P.step_base) stp += *P.step_base;
+    const int sa = sample_categorical(row, A, logz, K.sample_seed, stp, n);
+    P.actions_out[n] = sa;
+    P.logp_out[n] = row[sa] - logz;
+    P.value_out[n] = row[A];
+  }
+
   // ---- action decode (env.py:343-360) --------------------------------
   float raw;
   int64_t a;

@@ -32,16 +32,8 @@ GFX_DEV __bf16 f2bf(float v) { return (__bf16)v; }
 // CALL on gfx950: inside the GEMM epilogue it forced every live MFMA
 // accumulator to scratch (272 B/thread of spill, 6x slowdown).  Accuracy is
 // ~2 ulp of fp32 — far below the bf16 output rounding.
-// single-instruction exp/ln (libm expf/logf are CALLS on gfx950 — see the
-// fast_tanh note).  Used by BOTH sampler implementations so the fused
-// rollout kernel and sample_head_kernel agree bitwise.
-GFX_DEV float fast_exp(float x) {
-  return __builtin_amdgcn_exp2f(x * 1.4426950408889634f);
-}
-GFX_DEV float fast_log(float x) {
-  return __builtin_amdgcn_logf(x) * 0.6931471805599453f;
-}
-
+// fast_exp/fast_log/splitmix64 live in env_common.h (shared with the fused
+// env_step sampling path so every sampler implementation agrees bitwise).
 GFX_DEV float fast_tanh(float x) {
   x = fminf(fmaxf(x, -15.f), 15.f);
   // raw v_exp_f32 + v_rcp_f32, both single instructions.  An IEEE f32
@@ -770,13 +762,6 @@ __global__ void clip_scale_kernel(const float* __restrict__ part, int nparts,
 // head[M, A+1] f32: cols 0..A-1 logits, col A value.
 // Counter-based RNG (splitmix64): deterministic in (seed, step, env).
 // ---------------------------------------------------------------------------
-GFX_DEV uint64_t splitmix64(uint64_t x) {
-  x += 0x9E3779B97F4A7C15ull;
-  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
-  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
-  return x ^ (x >> 31);
-}
-
 __global__ void sample_head_kernel(const float* __restrict__ head, int M,
                                    int n_actions, uint64_t seed, uint64_t step,
                                    int64_t* __restrict__ actions,
@@ -791,11 +776,7 @@ __global__ void sample_head_kernel(const float* __restrict__ head, int M,
   if (step_base) step += *step_base;  // device counter: hipGraph-replayable RNG
   const int64_t mg = m + row_offset;  // global row: split launches keep RNG
   const float* row = head + (int64_t)m * (n_actions + 1);
-  float mx = row[0];
-  for (int j = 1; j < n_actions; ++j) mx = fmaxf(mx, row[j]);
-  float z = 0.f;
-  for (int j = 0; j < n_actions; ++j) z += fast_exp(row[j] - mx);
-  const float logz = fast_log(z) + mx;
+  const float logz = head_logz(row, n_actions);
 
   int a = 0;
   if (greedy) {
@@ -803,15 +784,7 @@ __global__ void sample_head_kernel(const float* __restrict__ head, int M,
     for (int j = 1; j < n_actions; ++j)
       if (row[j] > best) { best = row[j]; a = j; }
   } else {
-    const uint64_t r = splitmix64(seed ^ (step * 0x51E1F5ull + (uint64_t)mg * 0x9E37ull));
-    float u = (float)((r >> 11) * (1.0 / 9007199254740992.0));  // [0,1)
-    u = fminf(u, 0.999999f);
-    float c = 0.f;
-    a = n_actions - 1;
-    for (int j = 0; j < n_actions; ++j) {
-      c += fast_exp(row[j] - logz);
-      if (u < c) { a = j; break; }
-    }
+    a = sample_categorical(row, n_actions, logz, seed, step, mg);
   }
   actions[m] = a;
   logp[m] = row[a] - logz;

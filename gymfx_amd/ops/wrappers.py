@@ -173,14 +173,19 @@ class NativeEngine:
     def step(self, actions: torch.Tensor, obs_out: torch.Tensor,
              reward_out: torch.Tensor = None, terminated_out: torch.Tensor = None,
              obs_bf16_out: torch.Tensor = None, env_lo: int = 0,
-             env_hi: int = 0) -> Dict[str, torch.Tensor]:
+             env_hi: int = 0, head: torch.Tensor = None,
+             logp_out: torch.Tensor = None, value_out: torch.Tensor = None,
+             step_base: torch.Tensor = None, sample_seed: int = 0,
+             sample_step: int = 0) -> Dict[str, torch.Tensor]:
         if self._params.action_space_mode == "continuous":
             actions = actions.to(torch.float32).contiguous()
         else:
             actions = actions.to(torch.int64).contiguous()
         return dict(self._engine.step(actions, obs_out, reward_out,
                                       terminated_out, obs_bf16_out,
-                                      env_lo, env_hi))
+                                      env_lo, env_hi, head, logp_out,
+                                      value_out, step_base, sample_seed,
+                                      sample_step))
 
     def build_obs(self, obs_out: torch.Tensor,
                   obs_bf16_out: torch.Tensor = None) -> None:

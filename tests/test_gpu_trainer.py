@@ -182,3 +182,38 @@ def test_fused_rollout_equals_unfused():
     assert torch.equal(tf.logp_buf, tu.logp_buf)
     assert torch.equal(tf.val_buf, tu.val_buf)
     assert torch.equal(tf.model.params, tu.model.params)
+
+
+@pytest.mark.parametrize("policy", ["mlp", "lstm"])
+def test_fuse_sample_equals_sample_head(policy):
+    """In-env-kernel sampling (fuse_sample) must be BITWISE identical to the
+    standalone sample_head kernel chain: same shared helpers (env_common.h),
+    same RNG keying (seed, step_base+t, global env row)."""
+    from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    def make(fuse):
+        md = synthetic_ohlcv(2000, seed=7, vol=4e-4)
+        cfg = {"n_envs": 256, "device": "cuda", "window_size": 16,
+               "env_start_mode": "spread", "autoreset": True,
+               "position_size": 1000.0, "seed": 21}
+        env = build_vec_environment(cfg, md)
+        env.reset(seed=21)
+        pc = PPOConfig(rollout_steps=16, minibatches=4, ppo_epochs=2, seed=21,
+                       policy=policy, bptt_len=4, hidden=64, fuse_sample=fuse)
+        return PPOTrainer(env, pc)
+
+    tf = make(True)
+    tu = make(False)
+    for _ in range(2):
+        tf.train_update(with_stats=False)
+        tu.train_update(with_stats=False)
+    torch.cuda.synchronize()
+    assert tf._fuse_sample and not tu._fuse_sample
+    assert torch.equal(tf.act_buf, tu.act_buf)
+    assert torch.equal(tf.logp_buf, tu.logp_buf)
+    assert torch.equal(tf.val_buf, tu.val_buf)
+    assert torch.equal(tf.rew_buf, tu.rew_buf)
+    assert torch.equal(tf.model.params, tu.model.params)
+    assert torch.equal(tf.env.st.equity, tu.env.st.equity)
