@@ -174,7 +174,8 @@ class VecFxEnv:
              value_out: Optional[torch.Tensor] = None,
              step_base: Optional[torch.Tensor] = None,
              sample_seed: int = 0,
-             sample_step: int = 0) -> Dict[str, torch.Tensor]:
+             sample_step: int = 0,
+             fuse_obs: bool = False) -> Dict[str, torch.Tensor]:
         """Advance all envs. Returns dict with obs/reward/terminated tensors.
 
         reward_out / terminated_out / obs_bf16_out: optional preallocated
@@ -194,7 +195,7 @@ class VecFxEnv:
                                      terminated_out, obs_bf16_out,
                                      env_lo, env_hi, head, logp_out,
                                      value_out, step_base, sample_seed,
-                                     sample_step)
+                                     sample_step, fuse_obs)
             info["obs"] = self._obs
             return info
         if head is not None:

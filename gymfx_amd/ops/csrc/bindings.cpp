@@ -18,6 +18,8 @@ void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
                      int env_cnt, hipStream_t stream);
 void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
                       int env_cnt, hipStream_t stream);
+void launch_env_step_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
+                         int env_cnt, hipStream_t stream);
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
                  bool dact_tanh, bool add_bias, bool accum, hipStream_t stream);
@@ -277,7 +279,7 @@ struct GymFxEngine {
                 c10::optional<torch::Tensor> logp_out,
                 c10::optional<torch::Tensor> value_out,
                 c10::optional<torch::Tensor> step_base,
-                int64_t sample_seed, int64_t sample_step) {
+                int64_t sample_seed, int64_t sample_step, bool fuse_obs) {
     TORCH_CHECK(actions.is_cuda() == state["cursor"].is_cuda(),
                 "actions must live on the env device");
     TORCH_CHECK(actions.numel() == K.n_envs, "actions numel != n_envs");
@@ -353,8 +355,14 @@ struct GymFxEngine {
     P.reward_out = rew.data_ptr<float>();
     P.terminated_out = term.data_ptr<bool>();
     hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
-    launch_env_step(P, K, (int)env_lo, (int)(env_hi - env_lo), stream);
-    build_obs(obs_out, obs_bf16_out, env_lo, env_hi);
+    if (fuse_obs) {
+      // single fused launch: lane-0 step chain + wave-wide obs build
+      set_obs_ptrs(obs_out, obs_bf16_out);
+      launch_env_step_obs(P, K, (int)env_lo, (int)(env_hi - env_lo), stream);
+    } else {
+      launch_env_step(P, K, (int)env_lo, (int)(env_hi - env_lo), stream);
+      build_obs(obs_out, obs_bf16_out, env_lo, env_hi);
+    }
     py::dict out;
     out["reward"] = rew;
     out["base_reward"] = outputs["base_reward"];
@@ -364,10 +372,8 @@ struct GymFxEngine {
     return out;
   }
 
-  void build_obs(torch::Tensor obs_out,
-                 c10::optional<torch::Tensor> obs_bf16_out,
-                 int64_t env_lo = 0, int64_t env_hi = 0) {
-    if (env_hi <= 0) env_hi = K.n_envs;
+  void set_obs_ptrs(torch::Tensor& obs_out,
+                    c10::optional<torch::Tensor>& obs_bf16_out) {
     TORCH_CHECK(obs_out.is_contiguous() && obs_out.scalar_type() == torch::kFloat32,
                 "obs_out must be contiguous f32");
     TORCH_CHECK(obs_out.numel() == (int64_t)K.n_envs * K.obs_dim, "obs_out shape");
@@ -380,6 +386,13 @@ struct GymFxEngine {
                   "obs_bf16_out must be contiguous bf16 [n_envs, obs_dim]");
       P.obs_bf16_out = obs_bf16_out->data_ptr();
     }
+  }
+
+  void build_obs(torch::Tensor obs_out,
+                 c10::optional<torch::Tensor> obs_bf16_out,
+                 int64_t env_lo = 0, int64_t env_hi = 0) {
+    if (env_hi <= 0) env_hi = K.n_envs;
+    set_obs_ptrs(obs_out, obs_bf16_out);
     hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
     launch_build_obs(P, K, (int)env_lo, (int)(env_hi - env_lo), stream);
   }
@@ -850,7 +863,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("env_hi") = 0, py::arg("head") = py::none(),
            py::arg("logp_out") = py::none(), py::arg("value_out") = py::none(),
            py::arg("step_base") = py::none(), py::arg("sample_seed") = 0,
-           py::arg("sample_step") = 0)
+           py::arg("sample_step") = 0, py::arg("fuse_obs") = false)
       .def("build_obs", &gymfx::GymFxEngine::build_obs, py::arg("obs_out"),
            py::arg("obs_bf16_out") = py::none(), py::arg("env_lo") = 0,
            py::arg("env_hi") = 0);

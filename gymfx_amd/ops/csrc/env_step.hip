@@ -92,10 +92,8 @@ GFX_DEV void reset_env(const EnvPtrs& P, const EnvParamsK& K, int n) {
   P.raw_max[n] = -INFINITY;
 }
 
-__global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
-                                const int env_lo, const int env_cnt) {
-  const int n = env_lo + blockIdx.x * blockDim.x + threadIdx.x;
-  if (n >= env_lo + env_cnt) return;
+GFX_DEV void env_step_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
+                          const int env_lo) {
   const int T = K.T;
   // per-env instrument block bounds (multi-pair market; single-pair: 0..T)
   const int LO = P.lo_bar ? P.lo_bar[n] : 0;
@@ -484,19 +482,22 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
   if ((K.flags & F_AUTORESET) && P.terminated[n]) reset_env(P, K, n);
 }
 
+__global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
+                                const int env_lo, const int env_cnt) {
+  const int n = env_lo + blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= env_lo + env_cnt) return;
+  env_step_one(P, K, n, env_lo);
+}
+
 // ---------------------------------------------------------------------------
 // Observation build: elementwise over [N, obs_dim].
 // Semantics: envs/reference_step.py build_obs_torch (itself mirroring
 // default_preprocessor.py:34-77 / feature_window_preprocessor.py:99-191).
 // ---------------------------------------------------------------------------
-__global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K,
-                                 const int env_lo, const int env_cnt) {
-  const int64_t total = (int64_t)env_cnt * K.obs_dim;
+GFX_DEV void build_obs_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
+                           const int j) {
   const int T = K.T, W = K.window, F = K.n_features;
-  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (int64_t)gridDim.x * blockDim.x) {
-    const int n = env_lo + (int)(idx / K.obs_dim);
-    const int j = (int)(idx % K.obs_dim);
+  {
     const int step = P.cursor[n];
     const int LO = P.lo_bar ? P.lo_bar[n] : 0;
     const int EB = (P.end_bar && P.end_bar[n] > 0) ? P.end_bar[n] : T;
@@ -589,6 +590,39 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K,
   }
 }
 
+__global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K,
+                                 const int env_lo, const int env_cnt) {
+  const int64_t total = (int64_t)env_cnt * K.obs_dim;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    const int n = env_lo + (int)(idx / K.obs_dim);
+    const int j = (int)(idx % K.obs_dim);
+    build_obs_one(P, K, n, j);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused step + observation: one env per 64-lane wave.  Lane 0 runs the
+// serial per-env step chain, then the whole wave builds that env's
+// observation row.  MEASURED SLOWER than the two-launch split at N=4096
+// (22.55 vs 22.0 ms/update; a block-per-env variant was worse still at
+// 24.4): the obs build is work-bound, and no intra-kernel arrangement
+// matches the standalone kernel's full-width parallelism — the saved
+// launch (~9 us) does not cover the lost width.  Kept behind fuse_obs
+// (default off) with a bitwise-equality GPU test, same as fused_rollout.
+// ---------------------------------------------------------------------------
+__global__ void env_step_obs_kernel(const EnvPtrs P, const EnvParamsK K,
+                                    const int env_lo, const int env_cnt) {
+  const int wpb = blockDim.x >> 6;  // waves per block
+  const int n = env_lo + blockIdx.x * wpb + ((int)threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  const bool valid = n < env_lo + env_cnt;
+  if (valid && lane == 0) env_step_one(P, K, n, env_lo);
+  __syncthreads();  // step-phase global writes visible before the obs reads
+  if (valid)
+    for (int j = lane; j < K.obs_dim; j += 64) build_obs_one(P, K, n, j);
+}
+
 void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
                      int env_cnt, hipStream_t stream) {
   // 64-thread blocks: the kernel is a per-env dependent-load latency chain;
@@ -608,6 +642,15 @@ void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
   if (grid < 1) grid = 1;
   hipLaunchKernelGGL(build_obs_kernel, dim3(grid), dim3(block), 0, stream, P, K,
                      env_lo, env_cnt);
+}
+
+void launch_env_step_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
+                         int env_cnt, hipStream_t stream) {
+  const int block = 256;  // 4 envs (waves) per block
+  const int wpb = block / 64;
+  const int grid = (env_cnt + wpb - 1) / wpb;
+  hipLaunchKernelGGL(env_step_obs_kernel, dim3(grid), dim3(block), 0, stream,
+                     P, K, env_lo, env_cnt);
 }
 
 }  // namespace gymfx

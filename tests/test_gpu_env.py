@@ -185,3 +185,29 @@ def test_update_speed_regression_guard():
     torch.cuda.synchronize()
     ms = (time.perf_counter() - t0) / 5 * 1000
     assert ms < 80.0, f"PPO update took {ms:.1f} ms (expected ~23 ms)"
+
+
+def test_fused_obs_equals_two_launch():
+    """env_step_obs_kernel (one env per wave: lane-0 step + wave obs build)
+    must be bitwise identical to the separate env_step + build_obs launches."""
+    md = synthetic_ohlcv(3000, seed=11, vol=4e-4)
+    def make():
+        cfg = {"n_envs": 193, "device": "cuda", "window_size": 16,
+               "env_start_mode": "spread", "autoreset": True,
+               "position_size": 1000.0, "seed": 3,
+               "strategy_plugin": "direct_atr_sltp"}
+        env = build_vec_environment(cfg, md)
+        env.reset(seed=3)
+        return env
+    e1, e2 = make(), make()
+    g = torch.Generator().manual_seed(44)
+    for _ in range(60):
+        acts = torch.randint(0, 3, (193,), generator=g).cuda()
+        o1 = e1.step(acts, fuse_obs=True)
+        o2 = e2.step(acts, fuse_obs=False)
+        assert torch.equal(o1["obs"], o2["obs"])
+        assert torch.equal(o1["reward"], o2["reward"])
+        assert torch.equal(o1["terminated"], o2["terminated"])
+    assert torch.equal(e1.st.equity, e2.st.equity)
+    assert torch.equal(e1.st.cursor, e2.st.cursor)
+    assert torch.equal(e1.st.trade_count, e2.st.trade_count)
