@@ -55,6 +55,15 @@ class PPOConfig:
     fuse_sample: bool = True  # sample the action inside the env-step kernel
                               # (native engine): one fewer launch per rollout
                               # step; bitwise == the sample_head kernel.
+    overlap_gather: bool = False  # run minibatch gather on a side HIP
+                                  # stream overlapped with the previous
+                                  # minibatch's fwd/bwd (ping-pong slots).
+                                  # Measured SLOWER at N=4096 (23.1 vs 22.0
+                                  # ms/update): gather and the GEMMs are
+                                  # both HBM-bound, so concurrency only
+                                  # splits bandwidth and adds event-sync
+                                  # overhead.  Kept with a bit-equality
+                                  # GPU test, like fused_rollout.
 
     @classmethod
     def from_config(cls, cfg: Dict[str, Any]) -> "PPOConfig":
@@ -78,6 +87,7 @@ class PPOConfig:
             "rollout_streams": "rollout_streams",
             "fused_rollout": "fused_rollout",
             "fuse_sample": "fuse_sample",
+            "overlap_gather": "overlap_gather",
         }
         for attr, key in mapping.items():
             if cfg.get(key) is not None:
@@ -170,6 +180,7 @@ class PPOTrainer:
             self.dhead = torch.empty(L * Mseq, n_actions + 1,
                                      dtype=torch.bfloat16, device=dev)
             self._boot_state = self.model.alloc_state(N)
+            self._overlap = False  # BPTT buffers are large; single slot
         else:
             M = TN // cfg.minibatches
             if M * cfg.minibatches != TN:
@@ -178,12 +189,27 @@ class PPOTrainer:
             self.acts_train = self.model.alloc_acts(M)
             self.scratch = self.model.alloc_scratch(M)
             self.dhead = torch.empty(M, n_actions + 1, dtype=torch.bfloat16, device=dev)
-            # minibatch gather targets
-            self.obs_mb = torch.empty(M, D, dtype=torch.bfloat16, device=dev)
-            self.act_mb = torch.empty(M, dtype=torch.int64, device=dev)
-            self.logp_mb = torch.empty(M, dtype=torch.float32, device=dev)
-            self.adv_mb = torch.empty(M, dtype=torch.float32, device=dev)
-            self.ret_mb = torch.empty(M, dtype=torch.float32, device=dev)
+            # minibatch gather targets — 2 ping-pong slots when the gather
+            # overlaps the previous minibatch's fwd/bwd on a side stream
+            self._overlap = (self.device.type == "cuda" and cfg.overlap_gather
+                             and cfg.shuffle_rows)
+            ns = 2 if self._overlap else 1
+            self.obs_mb_s = [torch.empty(M, D, dtype=torch.bfloat16, device=dev)
+                             for _ in range(ns)]
+            self.act_mb_s = [torch.empty(M, dtype=torch.int64, device=dev)
+                             for _ in range(ns)]
+            self.logp_mb_s = [torch.empty(M, dtype=torch.float32, device=dev)
+                              for _ in range(ns)]
+            self.adv_mb_s = [torch.empty(M, dtype=torch.float32, device=dev)
+                             for _ in range(ns)]
+            self.ret_mb_s = [torch.empty(M, dtype=torch.float32, device=dev)
+                             for _ in range(ns)]
+            # slot-0 aliases (noshuffle path, tests)
+            self.obs_mb = self.obs_mb_s[0]
+            self.act_mb = self.act_mb_s[0]
+            self.logp_mb = self.logp_mb_s[0]
+            self.adv_mb = self.adv_mb_s[0]
+            self.ret_mb = self.ret_mb_s[0]
 
         # device counters (hipGraph-replayable RNG / schedule state)
         self.step_base = torch.zeros((), dtype=torch.int64, device=dev)
@@ -208,6 +234,8 @@ class PPOTrainer:
             self._s2 = torch.cuda.Stream()
             self.acts_half = [self.model.alloc_acts(N // 2),
                               self.model.alloc_acts(N // 2)]
+        if self._overlap:
+            self._gs = torch.cuda.Stream()
         self._graphs_ready = False
         self.g_rollout = None
         self.g_gather = None
@@ -312,10 +340,10 @@ class PPOTrainer:
         if self.cfg.normalize_adv:
             api.adv_normalize(self.adv_buf.view(-1), self._adv_part)
 
-    def _gather_body(self) -> None:
-        """Feistel-shuffled minibatch gather (parameter-independent: in
-        data-parallel runs this overlaps the previous minibatch's async
-        gradient all-reduce)."""
+    def _gather_body(self, slot: int = 0) -> None:
+        """Feistel-shuffled minibatch gather (parameter-independent: it
+        overlaps the previous minibatch's fwd/bwd on a side stream and, in
+        data-parallel runs, the async gradient all-reduce)."""
         cfg = self.cfg
         if self.recurrent:
             api.mb_gather_seq(
@@ -330,14 +358,14 @@ class PPOTrainer:
         else:
             api.mb_gather(
                 self.obs_flat, self.act_flat, self.logp_flat, self.adv_flat,
-                self.ret_flat, self.obs_mb, self.act_mb, self.logp_mb,
-                self.adv_mb, self.ret_mb, seed=self.shuffle_seed,
-                minibatches=cfg.minibatches, step_base=self.step_base,
-                mb_ctr=self.mb_ctr,
+                self.ret_flat, self.obs_mb_s[slot], self.act_mb_s[slot],
+                self.logp_mb_s[slot], self.adv_mb_s[slot], self.ret_mb_s[slot],
+                seed=self.shuffle_seed, minibatches=cfg.minibatches,
+                step_base=self.step_base, mb_ctr=self.mb_ctr,
             )
         api.increment_counter(self.mb_ctr, 1)
 
-    def _fwd_bwd_body(self) -> None:
+    def _fwd_bwd_body(self, slot: int = 0) -> None:
         """Forward + PPO loss backward + model backward on the gathered
         minibatch.  Gradients are fully overwritten by backward
         (deterministic split-M wgrad), so there is no zero_grad."""
@@ -355,18 +383,20 @@ class PPOTrainer:
             model.bptt_backward(self.obs_mb_seq, self.done_mb, self.dhead,
                                 self.bptt)
             return
-        head = model.forward(self.obs_mb, self.acts_train)
+        head = model.forward(self.obs_mb_s[slot], self.acts_train)
         api.ppo_loss_bwd(
-            head, self.act_mb, self.logp_mb, self.adv_mb, self.ret_mb,
+            head, self.act_mb_s[slot], self.logp_mb_s[slot],
+            self.adv_mb_s[slot], self.ret_mb_s[slot],
             self.dhead, clip_eps=cfg.clip_eps, ent_coef=cfg.ent_coef,
             vf_coef=cfg.vf_coef, inv_count=1.0 / self.mb_rows,
             losses=self.losses,
         )
-        model.backward(self.obs_mb, self.acts_train, self.dhead, self.scratch)
+        model.backward(self.obs_mb_s[slot], self.acts_train, self.dhead,
+                       self.scratch)
 
-    def _mb_body(self) -> None:
-        self._gather_body()
-        self._fwd_bwd_body()
+    def _mb_body(self, slot: int = 0) -> None:
+        self._gather_body(slot)
+        self._fwd_bwd_body(slot)
 
     def _mb_body_noshuffle(self, epoch_mb: int) -> None:
         cfg, model = self.cfg, self.model
@@ -437,7 +467,9 @@ class PPOTrainer:
         self._rollout_body()
         self._gae_body()
         self.mb_ctr.zero_()
-        self._mb_body()
+        nslots = 2 if self._overlap else 1
+        for s in range(nslots):
+            self._mb_body(s)
         self._allreduce_grads()
         self._opt_body()
         torch.cuda.synchronize()
@@ -445,12 +477,17 @@ class PPOTrainer:
         with torch.cuda.graph(self.g_rollout):
             self._rollout_body()
             self._gae_body()
-        self.g_gather = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_gather):
-            self._gather_body()
-        self.g_fwd_bwd = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_fwd_bwd):
-            self._fwd_bwd_body()
+        self.g_gather = []
+        self.g_fwd_bwd = []
+        for s in range(nslots):
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._gather_body(s)
+            self.g_gather.append(g)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._fwd_bwd_body(s)
+            self.g_fwd_bwd.append(g)
         self.g_opt = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_opt):
             self._opt_body()
@@ -485,28 +522,63 @@ class PPOTrainer:
                 self._opt_body()
             self.update_count += 1
             return self._stats(n_mb) if with_stats else {}
-        # pipelined loop: gather(i+1) overlaps the async all-reduce of
-        # minibatch i's gradients (parallel/ddp.py)
-        if graphs:
-            self.g_gather.replay()
-        else:
-            self._gather_body()
-        for i in range(n_mb):
+        def gather(slot: int) -> None:
             if graphs:
-                self.g_fwd_bwd.replay()
+                self.g_gather[slot].replay()
             else:
-                self._fwd_bwd_body()
-            self._reducer.start(self.model.grads)
-            if i + 1 < n_mb:
-                if graphs:
-                    self.g_gather.replay()
-                else:
-                    self._gather_body()
-            self._reducer.finish()
+                self._gather_body(slot)
+
+        def fwd_bwd(slot: int) -> None:
+            if graphs:
+                self.g_fwd_bwd[slot].replay()
+            else:
+                self._fwd_bwd_body(slot)
+
+        def opt() -> None:
             if graphs:
                 self.g_opt.replay()
             else:
                 self._opt_body()
+
+        if not self._overlap:
+            # pipelined loop: gather(i+1) overlaps the async all-reduce of
+            # minibatch i's gradients (parallel/ddp.py)
+            gather(0)
+            for i in range(n_mb):
+                fwd_bwd(0)
+                self._reducer.start(self.model.grads)
+                if i + 1 < n_mb:
+                    gather(0)
+                self._reducer.finish()
+                opt()
+            self.update_count += 1
+            return self._stats(n_mb) if with_stats else {}
+        # two-stream pipeline (ping-pong minibatch slots): gather(i+1) is
+        # parameter-independent, so it runs on a side stream concurrently
+        # with fwd/bwd(i) and the async gradient all-reduce.  Buffer slot
+        # i&1 alternates; gather(i+1) into slot s is safe once fwd_bwd(i-1)
+        # (the last reader of s) has been enqueued-before on the main
+        # stream (ev_fork), and fwd_bwd(i+1) waits on its gather (ev_g).
+        s2 = self._gs
+        main = torch.cuda.current_stream()
+        ev_fork = torch.cuda.Event()
+        ev_g = [torch.cuda.Event(), torch.cuda.Event()]
+        gather(0)
+        for i in range(n_mb):
+            slot = i & 1
+            nxt = slot ^ 1
+            if i + 1 < n_mb:
+                ev_fork.record()
+                with torch.cuda.stream(s2):
+                    s2.wait_event(ev_fork)
+                    gather(nxt)
+                    ev_g[nxt].record()
+            fwd_bwd(slot)
+            self._reducer.start(self.model.grads)
+            self._reducer.finish()
+            opt()
+            if i + 1 < n_mb:
+                main.wait_event(ev_g[nxt])
         self.update_count += 1
         return self._stats(n_mb) if with_stats else {}
 

@@ -88,6 +88,7 @@ DEFAULT_VALUES = {
     "max_grad_norm": 0.5,
     "train_updates": 10,
     "fuse_sample": True,         # sample inside the env-step kernel (GPU)
+    "overlap_gather": False,     # side-stream mb gather (measured slower)
     "checkpoint_file": None,     # save/resume path (mode=training)
     "resume": False,
     "trace_file": None,          # per-update phase-timing JSONL (HIP events)

@@ -217,3 +217,36 @@ def test_fuse_sample_equals_sample_head(policy):
     assert torch.equal(tf.rew_buf, tu.rew_buf)
     assert torch.equal(tf.model.params, tu.model.params)
     assert torch.equal(tf.env.st.equity, tu.env.st.equity)
+
+
+def test_overlap_gather_equals_serial():
+    """Side-stream minibatch gather (ping-pong slots) must be bitwise
+    identical to the serial single-slot pipeline — same kernels, same
+    Feistel sequence, only the stream placement differs."""
+    from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    def make(overlap):
+        md = synthetic_ohlcv(2000, seed=9, vol=4e-4)
+        cfg = {"n_envs": 256, "device": "cuda", "window_size": 16,
+               "env_start_mode": "spread", "autoreset": True,
+               "position_size": 1000.0, "seed": 17}
+        env = build_vec_environment(cfg, md)
+        env.reset(seed=17)
+        pc = PPOConfig(rollout_steps=16, minibatches=4, ppo_epochs=3, seed=17,
+                       overlap_gather=overlap)
+        return PPOTrainer(env, pc)
+
+    to = make(True)
+    ts = make(False)
+    for _ in range(3):
+        to.train_update(with_stats=False)
+        ts.train_update(with_stats=False)
+    torch.cuda.synchronize()
+    assert to._overlap and not ts._overlap
+    assert torch.equal(to.model.params, ts.model.params)
+    assert torch.equal(to.model.m, ts.model.m)
+    # losses are atomicAdd logging sums: order (not value) may differ
+    assert torch.allclose(to.losses, ts.losses, rtol=1e-5, atol=1e-6)
+    assert torch.equal(to.env.st.equity, ts.env.st.equity)
