@@ -1514,16 +1514,28 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
   const __bf16* a = reinterpret_cast<const __bf16*>(A);
   const __bf16* b = reinterpret_cast<const __bf16*>(B);
   const __bf16* y = reinterpret_cast<const __bf16*>(Yact);
-  // Wide-N (64x256) tiles when the tall-M operand would otherwise be
-  // re-streamed from HBM N/64 times AND the grid still fills the 256 CUs.
-  const bool wide = (N >= 192) && (M >= 16384);
-  dim3 grid(ceil_div(M, 64), ceil_div(N, wide ? 256 : 64));
+  // Tile width: 64x64 (NFRAG=2) tiles run at occupancy 8 and beat the
+  // 64x256 (NFRAG=8, occupancy 3) wide tiles even at tall update shapes
+  // (M=65536: 20.3 vs 22.0 ms/update whole-trainer) — the extra occupancy
+  // hides HBM latency better than wide tiles save A-operand re-streaming
+  // (L2 absorbs the re-reads).  GYMFX_GEMM_WIDE=1 forces the wide tile,
+  // =4 the 64x128 middle tile (tuning knobs).
+  static const int wide_env = [] {
+    const char* e = getenv("GYMFX_GEMM_WIDE");
+    return e ? atoi(e) : -1;
+  }();
+  const bool mid = (N >= 96) && wide_env == 4;
+  const bool wide = (N >= 192) && !mid && wide_env == 1;
+  dim3 grid(ceil_div(M, 64), ceil_div(N, wide ? 256 : (mid ? 128 : 64)));
   dim3 block(256);
 
 #define GEMM_LAUNCH(TB, ACT, DT, AB)                                          \
   do {                                                                        \
     if (wide)                                                                 \
       hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 8>), grid, block, 0,   \
+                         stream, a, b, bias, C, y, M, N, K);                  \
+    else if (mid)                                                             \
+      hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 4>), grid, block, 0,   \
                          stream, a, b, bias, C, y, M, N, K);                  \
     else                                                                      \
       hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 2>), grid, block, 0,   \
@@ -1533,9 +1545,11 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
     // C += A@B (f32 out, no bias/activation): the LSTM recurrent GEMM
     if (trans_b)
       if (wide) hipLaunchKernelGGL((gemm_kernel<true, 0, false, false, 8, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
+      else if (mid) hipLaunchKernelGGL((gemm_kernel<true, 0, false, false, 4, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
       else hipLaunchKernelGGL((gemm_kernel<true, 0, false, false, 2, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
     else
       if (wide) hipLaunchKernelGGL((gemm_kernel<false, 0, false, false, 8, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
+      else if (mid) hipLaunchKernelGGL((gemm_kernel<false, 0, false, false, 4, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
       else hipLaunchKernelGGL((gemm_kernel<false, 0, false, false, 2, true>), grid, block, 0, stream, a, b, bias, C, y, M, N, K);
     return;
   }
