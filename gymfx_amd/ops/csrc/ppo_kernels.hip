@@ -1605,11 +1605,16 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                            stream, x, dy, dW_part, db_part, M, N, K, slabs);
     }
     if (K % TW) {
-      const int pairs = (K - kfull * TW) * N;
-      dim3 g1(slabs, ceil_div(pairs, 1024));
-      hipLaunchKernelGGL(wgrad_ktail_kernel, g1,
-                         dim3(pairs < 1024 ? pairs : 1024), 0, stream, x, dy,
-                         dW_part, M, N, K, slabs, kfull * TW);
+      // K tail (obs_dim=260 -> 4 leftover dW rows): one MFMA tile-column
+      // of the register-staged kernel at tile offset kfull*TW/64.  The
+      // scalar wgrad_ktail_kernel (one (k,n) pair per thread, dependent
+      // scalar loads over m) measured 48.7 us — as much as the whole main
+      // wgrad; the LDS-staged MFMA column streams dY once and masks the
+      // dead K columns.
+      dim3 g1(1, ceil_div(N, 64), slabs);
+      hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), g1, dim3(256),
+                         0, stream, x, dy, dW_part, nullptr, M, N, K, slabs,
+                         kfull * TW / 64);
     }
   } else {
     dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);
