@@ -1574,12 +1574,21 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                   hipStream_t stream) {
   const __bf16* x = reinterpret_cast<const __bf16*>(X);
   const __bf16* dy = reinterpret_cast<const __bf16*>(dY);
-  // Tile choice: 128x128 halves the HBM re-streaming of X/dY (which hits
-  // the roofline for the LSTM's K~260/N=1024 shapes) but needs the grid to
-  // still fill the 256 CUs; otherwise 64x64 (the MLP's 260x256 shapes
-  // measured slower at 128x128 because the grid collapsed to ~1 block/CU).
+  // Tile choice: 128x128 halves the HBM re-streaming of X/dY — wgrad runs
+  // at ~5.7 TB/s with 64x64 tiles, i.e. bandwidth-bound, so the bigger tile
+  // wins whenever the grid still fills the chip (MLP 260x256 shapes:
+  // 19.3 -> 18.5 ms/update whole-trainer).  384+ workgroups = 1.5 waves/SIMD
+  // at the kernel's occupancy 2 keeps every CU fed.
+  // GYMFX_WGRAD_BIG=0/1 overrides (tuning knob).
+  static const int big_env = [] {
+    const char* e = getenv("GYMFX_WGRAD_BIG");
+    return e ? atoi(e) : -1;
+  }();
   const bool big = K >= 128 && N >= 128 &&
-                   (int64_t)ceil_div(K, 128) * ceil_div(N, 128) * slabs >= 1024;
+                   (big_env >= 0
+                        ? big_env != 0
+                        : (int64_t)ceil_div(K, 128) * ceil_div(N, 128) * slabs >=
+                              384);
   const int m_per_slab = (M + slabs - 1) / slabs;
   // glds path (both tile configs): full tiles only; the K tail runs the
   // skinny per-slab streaming kernel.
@@ -1611,7 +1620,8 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
       // scalar loads over m) measured 48.7 us — as much as the whole main
       // wgrad; the LDS-staged MFMA column streams dY once and masks the
       // dead K columns.
-      dim3 g1(1, ceil_div(N, 64), slabs);
+      // tail width < TW but can span up to two 64-tiles when TW=128
+      dim3 g1(ceil_div(K - kfull * TW, 64), ceil_div(N, 64), slabs);
       hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), g1, dim3(256),
                          0, stream, x, dy, dW_part, nullptr, M, N, K, slabs,
                          kfull * TW / 64);
