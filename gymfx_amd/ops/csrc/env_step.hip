@@ -494,8 +494,46 @@ __global__ void env_step_kernel(const EnvPtrs P, const EnvParamsK K,
 // Semantics: envs/reference_step.py build_obs_torch (itself mirroring
 // default_preprocessor.py:34-77 / feature_window_preprocessor.py:99-191).
 // ---------------------------------------------------------------------------
+// Per-(env, feature) z-score stats: mean/std depend only on (step, f), NOT
+// on the window row w — computing them once per feature instead of once per
+// element drops the p1/p2 prefix-sum traffic by the window factor (32x).
+// Encoding in (sm, sd): sd > 0 -> z-score with mean sm / std sd;
+// sd == 0 -> raw passthrough (binary or scaling off); sd < 0 -> emit 0
+// (fewer than 2 history rows).  Cast points match build_obs_torch exactly.
+GFX_DEV void obs_feature_stats(const EnvPtrs& P, const EnvParamsK& K,
+                               const int step, const int LO, const int f,
+                               float* sm, float* sd) {
+  const int F = K.n_features;
+  const bool is_binary = P.binary_mask && P.binary_mask[f];
+  if (K.scaling_mode == SCALE_NONE || is_binary) {
+    *sm = 0.f;
+    *sd = 0.f;
+    return;
+  }
+  int hist_left = (K.scaling_mode == SCALE_ROLLING)
+                      ? max(step - K.scale_window, LO) : LO;
+  int m = step - hist_left;
+  if (m < 2) {
+    *sm = 0.f;
+    *sd = -1.f;
+    return;
+  }
+  double s1 = P.p1[(int64_t)step * F + f] - P.p1[(int64_t)hist_left * F + f];
+  double s2 = P.p2[(int64_t)step * F + f] - P.p2[(int64_t)hist_left * F + f];
+  double mean = s1 / (double)m;
+  double var = s2 / (double)m - mean * mean;
+  var = fmax(var, 0.0);
+  double stdv = sqrt(var);
+  if (stdv < 1e-8) stdv = 1.0;
+  // f64 stats cast to f32 BEFORE the subtract/divide — the oracle's
+  // exact op order (reference_step.build_obs_torch).
+  *sm = (float)mean;
+  *sd = (float)stdv;
+}
+
 GFX_DEV void build_obs_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
-                           const int j) {
+                           const int j, const float* stat_m = nullptr,
+                           const float* stat_d = nullptr) {
   const int T = K.T, W = K.window, F = K.n_features;
   {
     const int step = P.cursor[n];
@@ -510,27 +548,19 @@ GFX_DEV void build_obs_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
       row = max(row, LO);
       row = min(row, EB - 1);
       float x = P.features[(int64_t)row * F + f];
-      bool is_binary = P.binary_mask && P.binary_mask[f];
-      if (K.scaling_mode == SCALE_NONE || is_binary) {
-        val = x;
+      float sm, sd;
+      if (stat_m) {
+        sm = stat_m[f];
+        sd = stat_d[f];
       } else {
-        int hist_left = (K.scaling_mode == SCALE_ROLLING)
-                            ? max(step - K.scale_window, LO) : LO;
-        int m = step - hist_left;
-        if (m < 2) {
-          val = 0.f;
-        } else {
-          double s1 = P.p1[(int64_t)step * F + f] - P.p1[(int64_t)hist_left * F + f];
-          double s2 = P.p2[(int64_t)step * F + f] - P.p2[(int64_t)hist_left * F + f];
-          double mean = s1 / (double)m;
-          double var = s2 / (double)m - mean * mean;
-          var = fmax(var, 0.0);
-          double stdv = sqrt(var);
-          if (stdv < 1e-8) stdv = 1.0;
-          // f64 stats cast to f32 BEFORE the subtract/divide — the oracle's
-          // exact op order (reference_step.build_obs_torch).
-          val = (x - (float)mean) / (float)stdv;
-        }
+        obs_feature_stats(P, K, step, LO, f, &sm, &sd);
+      }
+      if (sd == 0.f) {
+        val = x;
+      } else if (sd < 0.f) {
+        val = 0.f;
+      } else {
+        val = (x - sm) / sd;
       }
       // NaN guard BEFORE the clamp (fminf/fmaxf silently drop NaNs on CDNA):
       // torch maps nan->0, +/-inf->+/-clip (reference_step.build_obs_torch).
@@ -601,6 +631,27 @@ __global__ void build_obs_kernel(const EnvPtrs P, const EnvParamsK K,
   }
 }
 
+// One env per block: the per-feature z-score stats are computed once into
+// LDS and shared by all window rows (the elementwise kernel re-derived them
+// per element — W x more p1/p2 prefix-sum loads, the kernel's dominant HBM
+// traffic at 14 us/step).
+constexpr int OBS_MAX_F = 64;
+__global__ void build_obs_env_kernel(const EnvPtrs P, const EnvParamsK K,
+                                     const int env_lo, const int env_cnt) {
+  __shared__ float sm[OBS_MAX_F], sd[OBS_MAX_F];
+  const int n = env_lo + blockIdx.x;
+  if (n >= env_lo + env_cnt) return;  // uniform per block
+  const int T = K.T, F = K.n_features;
+  const int step = P.cursor[n];
+  const int LO = P.lo_bar ? P.lo_bar[n] : 0;
+  (void)T;
+  for (int f = threadIdx.x; f < F; f += blockDim.x)
+    obs_feature_stats(P, K, step, LO, f, &sm[f], &sd[f]);
+  __syncthreads();
+  for (int j = threadIdx.x; j < K.obs_dim; j += blockDim.x)
+    build_obs_one(P, K, n, j, sm, sd);
+}
+
 // ---------------------------------------------------------------------------
 // Fused step + observation: one env per 64-lane wave.  Lane 0 runs the
 // serial per-env step chain, then the whole wave builds that env's
@@ -636,6 +687,14 @@ void launch_env_step(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
 void launch_build_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
                       int env_cnt, hipStream_t stream) {
   const int block = 256;
+  // per-env LDS-stats kernel whenever the z-score path is active and the
+  // feature count fits the LDS stats arrays
+  if (K.off_features >= 0 && K.scaling_mode != SCALE_NONE &&
+      K.n_features <= OBS_MAX_F) {
+    hipLaunchKernelGGL(build_obs_env_kernel, dim3(env_cnt), dim3(block), 0,
+                       stream, P, K, env_lo, env_cnt);
+    return;
+  }
   int64_t total = (int64_t)env_cnt * K.obs_dim;
   int64_t blocks = (total + block - 1) / block;
   int grid = (int)(blocks < 16384 ? blocks : 16384);
