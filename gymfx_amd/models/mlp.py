@@ -115,6 +115,11 @@ class ActorCriticMLP:
 
     def _wg_slabs(self, K: int, N: int) -> int:
         # enough split-M slabs that tiles x slabs >= ~4 blocks per CU
+        # (GYMFX_WGRAD_SLABS overrides — tuning knob)
+        import os
+        ov = os.environ.get("GYMFX_WGRAD_SLABS")
+        if ov:
+            return int(ov)
         tiles = ((K + 63) // 64) * ((N + 63) // 64)
         s = self.wgrad_slabs
         while tiles * s < 2048:
