@@ -82,6 +82,7 @@ class PPOConfig:
             "hidden": "hidden_size",
             "normalize_adv": "normalize_adv",
             "shuffle_rows": "shuffle_rows",
+            "use_graphs": "use_graphs",
             "policy": "policy_model",
             "bptt_len": "bptt_len",
             "rollout_streams": "rollout_streams",
@@ -89,10 +90,15 @@ class PPOConfig:
             "fuse_sample": "fuse_sample",
             "overlap_gather": "overlap_gather",
         }
+        from ..config.merger import convert_type
+
         for attr, key in mapping.items():
             if cfg.get(key) is not None:
                 cur = getattr(out, attr)
-                setattr(out, attr, type(cur)(cfg[key]))
+                # bool("false") is True — route strings through the config
+                # layer's typed coercion first
+                val = convert_type(cfg[key]) if isinstance(cur, bool) else cfg[key]
+                setattr(out, attr, type(cur)(val))
         if cfg.get("seed") is not None:
             out.seed = int(cfg["seed"])
         return out

@@ -87,6 +87,7 @@ DEFAULT_VALUES = {
     "vf_coef": 0.5,
     "max_grad_norm": 0.5,
     "train_updates": 10,
+    "use_graphs": True,          # hipGraph-capture the training loop (GPU)
     "fuse_sample": True,         # sample inside the env-step kernel (GPU)
     "overlap_gather": False,     # side-stream mb gather (measured slower)
     "checkpoint_file": None,     # save/resume path (mode=training)

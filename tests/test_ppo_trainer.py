@@ -174,3 +174,13 @@ def test_ppo_learns_to_go_long_on_uptrend():
     long_frac = float((tr.act_buf == 1).float().mean())
     assert last > max(first * 5, 2e-5), (first, last)
     assert long_frac > 0.5, long_frac
+
+
+def test_ppo_config_bool_string_coercion():
+    """String 'false' from a raw config dict must not coerce to True
+    (bool('false') is True; from_config routes through convert_type)."""
+    from gymfx_amd.algo.ppo import PPOConfig
+
+    assert PPOConfig.from_config({"fuse_sample": "false"}).fuse_sample is False
+    assert PPOConfig.from_config({"use_graphs": "false"}).use_graphs is False
+    assert PPOConfig.from_config({"shuffle_rows": "true"}).shuffle_rows is True
