@@ -7,6 +7,15 @@ REPO_ROOT = Path(__file__).resolve().parents[1]
 if str(REPO_ROOT) not in sys.path:
     sys.path.insert(0, str(REPO_ROOT))
 
+# In small shared containers torch's default all-core OMP pool thrashes:
+# measured 6x SLOWER than single-thread on the suite's small CPU GEMMs.
+try:
+    import torch
+
+    torch.set_num_threads(1)
+except Exception:
+    pass
+
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: needs an AMD GPU (MI355X) to run")
