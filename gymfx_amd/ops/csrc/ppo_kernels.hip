@@ -391,41 +391,6 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
     db_part[(int64_t)slab * N + bn + tid] = db_acc;
 }
 
-// K-tail companion for the glds path: dW rows [k0, K) x all N for shapes
-// where K % 64 != 0 (e.g. obs_dim 260).  The 64x64-tile kernel would
-// re-stream dY once per n-tile just for these few columns; here one block
-// per slab owns every (k, n) pair (block = 1024 threads, one pair each for
-// ktail*N <= 1024) and streams X-tail + dY exactly once.
-__global__ __launch_bounds__(1024) void wgrad_ktail_kernel(
-    const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
-    float* __restrict__ dW_part, int M, int N, int K, int slabs, int k0) {
-  const int slab = blockIdx.x;
-  const int ktail = K - k0;
-  const int pairs = ktail * N;
-  const int tid = blockIdx.y * 1024 + threadIdx.x;
-  const int m_per_slab = (M + slabs - 1) / slabs;
-  const int m_begin = slab * m_per_slab;
-  const int m_end = min(M, m_begin + m_per_slab);
-  if (tid >= pairs) return;
-  const int k = tid / N;
-  const int n = tid % N;
-  const __bf16* xp = X + k0 + k;
-  const __bf16* yp = dY + n;
-  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-  int m = m_begin;
-  for (; m + 4 <= m_end; m += 4) {
-    a0 += bf2f(xp[(int64_t)m * K]) * bf2f(yp[(int64_t)m * N]);
-    a1 += bf2f(xp[(int64_t)(m + 1) * K]) * bf2f(yp[(int64_t)(m + 1) * N]);
-    a2 += bf2f(xp[(int64_t)(m + 2) * K]) * bf2f(yp[(int64_t)(m + 2) * N]);
-    a3 += bf2f(xp[(int64_t)(m + 3) * K]) * bf2f(yp[(int64_t)(m + 3) * N]);
-  }
-  for (; m < m_end; ++m)
-    a0 += bf2f(xp[(int64_t)m * K]) * bf2f(yp[(int64_t)m * N]);
-  // fixed association: ((a0+a1)+(a2+a3)) + tail-in-a0 — deterministic
-  dW_part[(int64_t)slab * K * N + (int64_t)(k0 + k) * N + n] =
-      (a0 + a1) + (a2 + a3);
-}
-
 template <bool WANT_DB, int FK, int FN>
 __global__ __launch_bounds__(256) void wgrad_partial_kernel(
     const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
