@@ -10,7 +10,7 @@ on an MI355X.
 """
 from __future__ import annotations
 
-import math
+
 from typing import Any, Dict, List, Sequence, Tuple
 
 import numpy as np

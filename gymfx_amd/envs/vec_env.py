@@ -14,10 +14,10 @@ from typing import Any, Dict, Optional
 import numpy as np
 import torch
 
-from ..data.feed import MarketData, concat_markets
+from ..data.feed import concat_markets
 from .market import MarketTensors, build_market_tensors
 from .params import EnvParams
-from .reference_step import build_obs_torch, coerce_actions, step_torch
+from .reference_step import build_obs_torch, step_torch
 from .state import ACTION_COUNTERS, EXEC_COUNTERS, EnvState, alloc_state, reset_state_
 
 

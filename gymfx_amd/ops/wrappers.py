@@ -8,7 +8,7 @@ from typing import Dict
 import torch
 
 from ..envs.market import MarketTensors
-from ..envs.params import EnvParams, PREP_FEATURE_WINDOW
+from ..envs.params import EnvParams
 from ..envs.state import EnvState
 from . import native
 

@@ -18,7 +18,7 @@ xGMI work — container hostnames may not resolve).
 from __future__ import annotations
 
 import os
-from typing import Any, Dict, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 
