@@ -197,11 +197,11 @@ class ActorCriticLSTM:
             gh = buf["gates_h"][l]
             api.gemm(buf["h_in"][l], self.wt("Wh"), None, gh, act=1,
                      trans_b=True)
+            # the cell also emits the reset-masked state feeding step l+1
+            # (fused masked_state: one fewer launch per sequential step)
             api.lstm_cell_fwd(g, gh, buf["c_in"][l], buf["c_raw"][l],
-                              buf["h_raw"][l])
-            # masked state feeds step l+1 (zero across episode resets)
-            api.masked_state(buf["h_raw"][l], buf["c_raw"][l], done_seq[l],
-                             buf["h_in"][l + 1], buf["c_in"][l + 1])
+                              buf["h_raw"][l], done_seq[l],
+                              buf["h_in"][l + 1], buf["c_in"][l + 1])
         api.gemm(buf["h_raw"].view(L * M, H), self.wt("Wy"), self.f32("by"),
                  buf["head"], act=0, trans_b=True)
         return buf["head"]

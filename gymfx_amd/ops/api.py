@@ -70,12 +70,18 @@ def lstm_cell_fwd(
     c_prev: torch.Tensor,
     c_new: torch.Tensor,
     h_new: torch.Tensor,
+    done: "Optional[torch.Tensor]" = None,
+    h_masked: "Optional[torch.Tensor]" = None,
+    c_masked: "Optional[torch.Tensor]" = None,
 ) -> None:
     """gate pre-activations = gates_pre (f32, x@Wx + b) [+ gates_h (bf16,
     h@Wh kept separate so the recurrent GEMM writes 2-byte outputs instead
-    of read-modify-writing the f32 buffer)] -> c_new f32, h_new bf16."""
+    of read-modify-writing the f32 buffer)] -> c_new f32, h_new bf16.
+    With done/h_masked/c_masked the kernel also emits the reset-masked
+    state that feeds the next BPTT step (fuses the masked_state launch)."""
     if _use_native(gates_pre):
-        native.require().lstm_cell_fwd(gates_pre, gates_h, c_prev, c_new, h_new)
+        native.require().lstm_cell_fwd(gates_pre, gates_h, c_prev, c_new,
+                                       h_new, done, h_masked, c_masked)
         return
     gp = gates_pre
     if gates_h is not None:
@@ -89,6 +95,10 @@ def lstm_cell_fwd(
     c = f * c_prev + i * g
     c_new.copy_(c)
     h_new.copy_((o * torch.tanh(c)).to(torch.bfloat16))
+    if done is not None:
+        keep = (~done).unsqueeze(1)
+        h_masked.copy_(torch.where(keep, h_new, torch.zeros_like(h_new)))
+        c_masked.copy_(torch.where(keep, c_new, torch.zeros_like(c_new)))
 
 
 def lstm_cell_bwd(

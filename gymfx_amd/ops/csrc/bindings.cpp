@@ -25,6 +25,7 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  bool dact_tanh, bool add_bias, bool accum, hipStream_t stream);
 void launch_lstm_cell_fwd(const float* gates_pre, const void* gates_h,
                           const float* c_prev, float* c_new, void* h_new,
+                          const bool* done, void* h_masked, float* c_masked,
                           int64_t M, int H, hipStream_t stream);
 void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
                           const float* c_prev,
@@ -449,7 +450,10 @@ void gemm_op(torch::Tensor A, torch::Tensor B, c10::optional<torch::Tensor> bias
 void lstm_cell_fwd_op(torch::Tensor gates_pre,
                       c10::optional<torch::Tensor> gates_h,
                       torch::Tensor c_prev,
-                      torch::Tensor c_new, torch::Tensor h_new) {
+                      torch::Tensor c_new, torch::Tensor h_new,
+                      c10::optional<torch::Tensor> done,
+                      c10::optional<torch::Tensor> h_masked,
+                      c10::optional<torch::Tensor> c_masked) {
   check_f32(gates_pre, "gates_pre");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
@@ -463,10 +467,24 @@ void lstm_cell_fwd_op(torch::Tensor gates_pre,
     check_bf16(*gates_h, "gates_h");
     gh = gates_h->data_ptr();
   }
+  const bool* done_p = nullptr;
+  void* hm_p = nullptr;
+  float* cm_p = nullptr;
+  if (done.has_value()) {
+    TORCH_CHECK(h_masked.has_value() && c_masked.has_value(),
+                "done needs h_masked and c_masked outputs");
+    TORCH_CHECK(done->scalar_type() == torch::kBool && done->numel() == M,
+                "done must be bool [M]");
+    check_bf16(*h_masked, "h_masked");
+    check_f32(*c_masked, "c_masked");
+    done_p = done->data_ptr<bool>();
+    hm_p = h_masked->data_ptr();
+    cm_p = c_masked->data_ptr<float>();
+  }
   gymfx::launch_lstm_cell_fwd(gates_pre.data_ptr<float>(), gh,
                               c_prev.data_ptr<float>(),
-                              c_new.data_ptr<float>(), h_new.data_ptr(), M, H,
-                              cur_stream());
+                              c_new.data_ptr<float>(), h_new.data_ptr(),
+                              done_p, hm_p, cm_p, M, H, cur_stream());
 }
 
 void lstm_cell_bwd_op(torch::Tensor gates_pre,
@@ -802,7 +820,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("accum") = false);
   m.def("lstm_cell_fwd", &lstm_cell_fwd_op, py::arg("gates_pre"),
         py::arg("gates_h"), py::arg("c_prev"), py::arg("c_new"),
-        py::arg("h_new"));
+        py::arg("h_new"), py::arg("done") = py::none(),
+        py::arg("h_masked") = py::none(), py::arg("c_masked") = py::none());
   m.def("lstm_cell_bwd", &lstm_cell_bwd_op, py::arg("gates_pre"),
         py::arg("gates_h"),
         py::arg("c_prev"), py::arg("c_new"), py::arg("dh_head"),

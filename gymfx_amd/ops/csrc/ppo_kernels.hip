@@ -908,7 +908,10 @@ __global__ void lstm_cell_fwd_kernel(
     const float* __restrict__ c_prev,     // [M, H]
     float* __restrict__ c_new,            // [M, H]
     __bf16* __restrict__ h_new,           // [M, H] (bf16: feeds next GEMM)
-    int64_t M, int H) {
+    const bool* __restrict__ done,        // [M] or null: also emit the
+    __bf16* __restrict__ h_masked,        //   reset-masked state that feeds
+    float* __restrict__ c_masked,         //   the NEXT step (one fewer
+    int64_t M, int H) {                   //   launch than masked_state)
   const int64_t total = M * H;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
@@ -923,7 +926,13 @@ __global__ void lstm_cell_fwd_kernel(
     const float o = fast_sigmoid(gate_pre(gx, gh, 3 * H + h));
     const float c = f * c_prev[idx] + i * g;
     c_new[idx] = c;
-    h_new[idx] = f2bf(o * fast_tanh(c));
+    const __bf16 hb = f2bf(o * fast_tanh(c));
+    h_new[idx] = hb;
+    if (done) {
+      const bool d = done[m];
+      h_masked[idx] = d ? (__bf16)0.f : hb;
+      c_masked[idx] = d ? 0.f : c;
+    }
   }
 }
 
@@ -1364,12 +1373,14 @@ static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
 void launch_lstm_cell_fwd(const float* gates_pre, const void* gates_h,
                           const float* c_prev, float* c_new, void* h_new,
+                          const bool* done, void* h_masked, float* c_masked,
                           int64_t M, int H, hipStream_t stream) {
   int64_t total = M * H;
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_fwd_kernel, dim3(blocks), dim3(256), 0, stream,
                      gates_pre, reinterpret_cast<const __bf16*>(gates_h),
-                     c_prev, c_new, reinterpret_cast<__bf16*>(h_new), M, H);
+                     c_prev, c_new, reinterpret_cast<__bf16*>(h_new), done,
+                     reinterpret_cast<__bf16*>(h_masked), c_masked, M, H);
 }
 
 void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
