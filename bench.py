@@ -47,6 +47,10 @@ def main() -> None:
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--policy", type=str, default="mlp", choices=["mlp", "lstm"],
                     help="actor-critic architecture (BASELINE config #2 / #4)")
+    ap.add_argument("--reward", type=str, default=None,
+                    help="reward plugin (default: BASELINE config for the run shape)")
+    ap.add_argument("--strategy", type=str, default=None,
+                    help="strategy plugin (default: BASELINE config for the run shape)")
     ap.add_argument("--pairs", type=int, default=1,
                     help="instruments in the market tensor (BASELINE config #5)")
     args = ap.parse_args()
@@ -82,11 +86,20 @@ def main() -> None:
         "instrument": "EUR_USD",
         "timeframe": "M1",
         "preprocessor_plugin": "feature_window_preprocessor",
+        # BASELINE.json named configs: #4 (LSTM) dd_penalized +
+        # direct_atr_sltp.  The MLP scale runs use pnl_reward at EVERY N so
+        # the driver's weak-scaling efficiency compares identical work
+        # (config #3's sharpe_reward variant is measured separately:
+        # profiles/bench_mlp_sharpe.json, ~8% env-kernel cost).
+        # Overridable via --reward/--strategy.
+        "reward_plugin": args.reward or (
+            "dd_penalized_reward" if args.policy == "lstm" else "pnl_reward"),
+        "strategy_plugin": args.strategy or (
+            "direct_atr_sltp" if args.policy == "lstm" else "default_strategy"),
         "feature_columns": FEATURES,
         "feature_scaling": "rolling_zscore",
         "feature_scaling_window": 256,
         "window_size": 32,
-        "reward_plugin": "pnl_reward",
         "n_envs": args.n_envs,
         "device": str(device),
         "autoreset": True,
@@ -165,7 +178,8 @@ def main() -> None:
                 "rollout_steps": args.rollout,
                 "obs_dim": env.obs_dim,
                 "preprocessor": "feature_window_preprocessor",
-                "reward": "pnl_reward",
+                "reward": cfg["reward_plugin"],
+                "strategy": cfg["strategy_plugin"],
                 "pairs": args.pairs,
             },
         }))
