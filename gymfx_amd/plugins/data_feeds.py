@@ -55,20 +55,33 @@ class SyntheticDataFeed(PluginBase):
     def load_data(self, config: Dict[str, Any]) -> MarketData:
         p = self._resolve(config)
         pairs = int(config.get("synthetic_pairs", 1) or 1)
+        # BASELINE config #5 names EURUSD+GBPUSD+USDJPY: realistic start
+        # prices and per-instrument pip sizes (JPY quotes pip = 0.01)
+        majors = [
+            (str(config.get("instrument", "EUR_USD")), None, 0.0001),
+            ("GBP_USD", 1.27, 0.0001),
+            ("USD_JPY", 150.0, 0.01),
+        ]
         mds = []
         for i in range(pairs):
-            name = (str(config.get("instrument", "EUR_USD")) if i == 0
-                    else f"PAIR_{i}")
-            mds.append(synthetic_ohlcv(
+            if i < len(majors):
+                name, px, pip = majors[i]
+            else:
+                name, px, pip = f"PAIR_{i}", None, 0.0001
+            if px is None:
+                px = float(p["synthetic_start_price"]) * (1.0 + 0.1 * i)
+            md = synthetic_ohlcv(
                 int(p["synthetic_rows"]),
                 seed=int(p["synthetic_seed"]) + 7919 * i,
-                start_price=float(p["synthetic_start_price"]) * (1.0 + 0.1 * i),
+                start_price=px,
                 vol=float(p["synthetic_vol"]),
                 drift=float(p["synthetic_drift"]),
                 bar_minutes=int(p["synthetic_bar_minutes"]),
                 instrument=name,
                 extra_feature_columns=int(p["synthetic_extra_features"]),
-            ))
+            )
+            md.meta["pip_size"] = pip
+            mds.append(md)
         if len(mds) == 1:
             return mds[0]
         from ..data.feed import concat_markets
