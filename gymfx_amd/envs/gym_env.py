@@ -17,6 +17,7 @@ import torch
 
 from .. import spaces
 from ..calendar import compute_fx_calendar_features, resolve_broker_metadata
+from .state import EXEC_COUNTERS
 from .vec_env import VecFxEnv
 from .params import EnvParams
 
@@ -157,6 +158,9 @@ class GymFxEnv(spaces.Env):
         self._audit_path = _os.environ.get("GYMFX_BRACKET_AUDIT") or \
             config.get("bracket_audit_file")
         self._prev_audit_state: Dict[str, Any] = {}
+        self._ec_sl = EXEC_COUNTERS.index("bracket_sl_fills")
+        self._ec_tp = EXEC_COUNTERS.index("bracket_tp_fills")
+        self._ec_fc = EXEC_COUNTERS.index("session_force_closes")
 
     # ------------------------------------------------------------------
     def reset(self, *, seed: Optional[int] = None, options: Optional[Dict[str, Any]] = None):
@@ -185,9 +189,9 @@ class GymFxEnv(spaces.Env):
             self._prev_audit_state = {
                 "br_active": bool(st.br_active[0].item()),
                 "pos": float(st.pos[0].item()),
-                "sl_fills": int(st.exec_diag[0, 15].item()),
-                "tp_fills": int(st.exec_diag[0, 16].item()),
-                "session_fc": int(st.exec_diag[0, 14].item()),
+                "sl_fills": int(st.exec_diag[0, self._ec_sl].item()),
+                "tp_fills": int(st.exec_diag[0, self._ec_tp].item()),
+                "session_fc": int(st.exec_diag[0, self._ec_fc].item()),
             }
         out = self.vec.step(act_t)
         if self._audit_path:
@@ -227,9 +231,9 @@ class GymFxEnv(spaces.Env):
                 "tp": float(st.br_tp[0].item()),
                 "position": float(st.pos[0].item()),
             })
-        sl_d = int(st.exec_diag[0, 15].item()) - prev.get("sl_fills", 0)
-        tp_d = int(st.exec_diag[0, 16].item()) - prev.get("tp_fills", 0)
-        fc_d = int(st.exec_diag[0, 14].item()) - prev.get("session_fc", 0)
+        sl_d = int(st.exec_diag[0, self._ec_sl].item()) - prev.get("sl_fills", 0)
+        tp_d = int(st.exec_diag[0, self._ec_tp].item()) - prev.get("tp_fills", 0)
+        fc_d = int(st.exec_diag[0, self._ec_fc].item()) - prev.get("session_fc", 0)
         if sl_d > 0:
             recs.append({"event": "bracket_sl_fill", "bar_index": bar,
                          "equity": float(st.equity[0].item())})
