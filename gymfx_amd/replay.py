@@ -149,7 +149,7 @@ class ScalarLedger:
             want = 1 if action == 1 else -1
             if self.pos * want < 0:
                 self.pend_close = True
-            if self.pos * want <= 0 and not (self.pos != 0 and self.pos * want > 0):
+            if self.pos * want <= 0:  # flat or opposite (flip = close+open)
                 self.pend_dir = want
                 if self.use_brackets:
                     sl_d = self.sl_pips * self.pip_size
