@@ -250,3 +250,29 @@ def test_overlap_gather_equals_serial():
     # losses are atomicAdd logging sums: order (not value) may differ
     assert torch.allclose(to.losses, ts.losses, rtol=1e-5, atol=1e-6)
     assert torch.equal(to.env.st.equity, ts.env.st.equity)
+
+
+def test_trace_file_on_gpu(tmp_path):
+    """trace_file phase timing must work when the phases are hipGraph
+    replays (HIP events bracket graph launches; lazy drain)."""
+    import json
+
+    from gymfx_amd.algo.ppo import train_from_config
+    from gymfx_amd.config import DEFAULT_VALUES
+
+    trace = tmp_path / "t.jsonl"
+    cfg = {**DEFAULT_VALUES,
+           "data_feed_plugin": "synthetic_data_feed",
+           "synthetic_rows": 3000, "n_envs": 256, "window_size": 8,
+           "env_start_mode": "spread", "autoreset": True,
+           "position_size": 1000.0, "device": "cuda", "seed": 3,
+           "rollout_steps": 16, "minibatches": 2, "ppo_epochs": 1,
+           "hidden_size": 64, "train_updates": 3, "quiet_mode": True,
+           "trace_file": str(trace)}
+    out = train_from_config(cfg)
+    assert out["updates"] == 3
+    lines = [json.loads(l) for l in trace.read_text().splitlines()]
+    assert len(lines) == 3
+    for rec in lines:
+        assert rec["phases_ms"]["rollout"] > 0
+        assert rec["phases_ms"]["update"] > 0
