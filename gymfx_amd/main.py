@@ -74,6 +74,22 @@ def main(argv=None) -> None:
     config = DEFAULT_VALUES.copy()
     file_config = load_config(args.load_config) if args.load_config else {}
     unknown_args_dict = process_unknown_args(unknown_args)
+    # remote-loaded config sits at the file tier (below CLI): an upgrade
+    # over the reference, which defines remote config I/O
+    # (app/config_handler.py:30-73) but never wires it into the runner.
+    remote_url = (cli_args.get("remote_load_config")
+                  or unknown_args_dict.get("remote_load_config")
+                  or file_config.get("remote_load_config"))
+    if remote_url:
+        from .config import remote_load_config  # noqa: PLC0415
+
+        remote_cfg = remote_load_config(
+            remote_url,
+            cli_args.get("username") or file_config.get("username"),
+            cli_args.get("password") or file_config.get("password"),
+        )
+        if remote_cfg:
+            file_config = {**remote_cfg, **file_config}
     config = merge_config(config, {}, {}, file_config, cli_args, unknown_args_dict)
 
     if config.get("mode") not in {"training", "optimization", "inference"}:
@@ -104,6 +120,25 @@ def main(argv=None) -> None:
 
     if config.get("save_config"):
         save_config(config, config["save_config"])
+    # debug log only when a path was explicitly given (the default
+    # "./debug_out.json" would otherwise litter every run's cwd)
+    explicit_save_log = (cli_args.get("save_log")
+                         or unknown_args_dict.get("save_log")
+                         or file_config.get("save_log"))
+    if explicit_save_log:
+        from .config import save_debug_info  # noqa: PLC0415
+
+        save_debug_info(summary, str(explicit_save_log))
+    if config.get("remote_save_config"):
+        from .config import remote_save_config  # noqa: PLC0415
+
+        remote_save_config(config, config["remote_save_config"],
+                           config.get("username"), config.get("password"))
+    if config.get("remote_log"):
+        from .config import remote_log  # noqa: PLC0415
+
+        remote_log(config, summary, config["remote_log"],
+                   config.get("username"), config.get("password"))
 
     if not config.get("quiet_mode", False):
         print(json.dumps(summary, indent=2, default=str))

@@ -139,3 +139,63 @@ def test_remote_config_http_roundtrip():
         assert remote_load_config("http://127.0.0.1:9/none") is None
     finally:
         srv.shutdown()
+
+
+def test_main_wires_remote_config_and_log(tmp_path):
+    """mode=inference CLI run with remote_load_config / remote_log /
+    save_log: the runner merges the remote config at the file tier and
+    posts the summary at the end (an upgrade over the reference, which
+    defines but never wires app/config_handler.py:30-73)."""
+    import http.server
+    import json as _json
+    import threading
+    from urllib.parse import parse_qs
+
+    from gymfx_amd.main import main as cli_main
+
+    received = {}
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        def do_POST(self):
+            body = self.rfile.read(int(self.headers["Content-Length"]))
+            received.update({k: v[0] for k, v in
+                             parse_qs(body.decode()).items()})
+            self.send_response(200)
+            self.end_headers()
+
+        def do_GET(self):
+            payload = _json.dumps({"steps": 7}).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.end_headers()
+            self.wfile.write(payload)
+
+        def log_message(self, *a):
+            pass
+
+    srv = http.server.HTTPServer(("127.0.0.1", 0), Handler)
+    port = srv.server_address[1]
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        url = f"http://127.0.0.1:{port}/x"
+        log_path = tmp_path / "dbg.json"
+        cli_main([
+            "--mode", "inference", "--quiet_mode", "true",
+            "--data_feed_plugin", "synthetic_data_feed",
+            "--synthetic_rows", "300", "--window_size", "8",
+            "--results_file", str(tmp_path / "r.json"),
+            "--save_config", "",
+            "--remote_load_config", url,
+            "--remote_log", url,
+            "--save_log", str(log_path),
+            "--username", "u", "--password", "p",
+        ])
+        out = _json.loads((tmp_path / "r.json").read_text())
+        assert out  # run completed, summary written
+        # remote-loaded config (steps=7) reached the merged config: the
+        # posted json_config carries the non-default value
+        assert _json.loads(received["json_config"])["steps"] == 7
+        assert _json.loads(received["json_result"])  # summary was posted
+        assert _json.loads(log_path.read_text())     # save_log written
+    finally:
+        srv.shutdown()
