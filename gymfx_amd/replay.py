@@ -142,7 +142,12 @@ class ScalarLedger:
                 fill = sell(trig) if is_long else buy(trig)
                 self._close(fill, bar, kind)
         self.br_armed = False
-        # 3. strategy decision (direct / fixed-bracket)
+        # 3. strategy decision (direct / fixed-bracket).  Out-of-range
+        # actions coerce to hold, exactly like the engine's decode
+        # (env_step.hip: a = (v >= 0 && v <= 2) ? v : 0); action 3
+        # (force-flat) exists only INTERNALLY via the event overlay.
+        if action < 0 or action > 2:
+            action = 0
         if action in (1, 2):
             self.events.append({"type": "target_requested", "bar": bar,
                                 "action": action})
@@ -156,8 +161,6 @@ class ScalarLedger:
                     tp_d = self.tp_pips * self.pip_size
                     self.pend_sl = c - sl_d if want > 0 else c + sl_d
                     self.pend_tp = c + tp_d if want > 0 else c - tp_d
-        elif action == 3 and self.pos != 0:
-            self.pend_close = True
 
     def equity(self, close: float) -> float:
         return self.cash + self.margin + self.pos * (close - self.avg_entry)

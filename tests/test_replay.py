@@ -98,3 +98,28 @@ def test_replay_reconciles_under_random_configs(trial):
     res = ReplayAdapter().run(cfg, md, acts)
     assert res["reconciled"], (trial, res["reconciliation"],
                                res["engine"], res["oracle"])
+
+
+def test_replay_reconciles_hypothesis_fuzz():
+    """Hypothesis-driven action-sequence fuzz: the engine and the
+    independent ledger must reconcile for ARBITRARY action strings
+    (incl. pathological runs of flips / force-closes), not just uniform
+    random ones."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    md = synthetic_ohlcv(300, seed=77, vol=6e-4)
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.lists(st.integers(min_value=0, max_value=3),
+                    min_size=10, max_size=120),
+           st.booleans())
+    def check(actions, brackets):
+        cfg = _cfg()
+        if brackets:
+            cfg.update(strategy_plugin="direct_fixed_sltp",
+                       sl_pips=8.0, tp_pips=12.0)
+        res = ReplayAdapter().run(cfg, md, actions)
+        assert res["reconciled"], (actions[:20], res["reconciliation"])
+
+    check()
