@@ -199,3 +199,44 @@ def test_main_wires_remote_config_and_log(tmp_path):
         assert _json.loads(log_path.read_text())     # save_log written
     finally:
         srv.shutdown()
+
+
+def test_merge_precedence_property():
+    """5-tier precedence as a property (parity:
+    app/config_merger.py:37-51): for ANY key sets, later tiers win and
+    unknown-arg values pass through typed coercion."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from gymfx_amd.config import merge_config
+
+    keys = st.sampled_from(["a", "b", "c", "d"])
+    vals = st.one_of(st.integers(-5, 5), st.text(max_size=3),
+                     st.booleans(), st.none())
+    tier = st.dictionaries(keys, vals, max_size=4)
+
+    @settings(max_examples=60, deadline=None)
+    @given(tier, tier, tier, tier, tier)
+    def check(p1, p2, filec, cli, unknown):
+        defaults = {"a": 0, "b": 0}
+        merged = merge_config(defaults, p1, p2, filec, cli, unknown)
+        from gymfx_amd.config import convert_type
+        for k in set().union(defaults, p1, p2, filec, cli, unknown):
+            if k in unknown:
+                assert merged[k] == convert_type(unknown[k])
+            elif k in cli and cli[k] is not None:
+                assert merged[k] == cli[k]
+            elif k in filec:
+                assert merged[k] == filec[k]
+            elif k in defaults:
+                assert merged[k] == defaults[k]
+            elif k in p2:
+                assert merged[k] == p2[k]
+            elif k in p1:
+                assert merged[k] == p1[k]
+            else:
+                # only a None CLI value mentioned it: argparse "not
+                # provided" — the key must be absent entirely
+                assert k not in merged
+
+    check()
