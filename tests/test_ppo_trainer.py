@@ -184,3 +184,22 @@ def test_ppo_config_bool_string_coercion():
     assert PPOConfig.from_config({"fuse_sample": "false"}).fuse_sample is False
     assert PPOConfig.from_config({"use_graphs": "false"}).use_graphs is False
     assert PPOConfig.from_config({"shuffle_rows": "true"}).shuffle_rows is True
+
+
+def test_feistel_bijection_property():
+    """Hypothesis property: the cycle-walked Feistel permutation is a
+    bijection on [0, n) for ARBITRARY n (odd, prime, tiny, non-power-of-2)
+    and key — the correctness backbone of the minibatch shuffle."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=50, deadline=None)
+    @given(st.integers(min_value=2, max_value=5000),
+           st.integers(min_value=0, max_value=2**31),
+           st.integers(min_value=0, max_value=64),
+           st.integers(min_value=0, max_value=7))
+    def check(n, seed, step, epoch):
+        perm = api.feistel_perm(n, api.feistel_key(seed, step, epoch))
+        assert torch.equal(torch.sort(perm).values, torch.arange(n))
+
+    check()
