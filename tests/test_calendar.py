@@ -62,3 +62,33 @@ def test_table_matches_scalar(start):
 def test_neutral_on_unparseable():
     out = compute_fx_calendar_features("not-a-date", timeframe_hours=1.0)
     assert all(v == 0.0 for v in out.values())
+
+
+def test_calendar_feature_ranges_property():
+    """Hypothesis property over arbitrary epoch timestamps: every feature
+    is finite and in range (binary flags in {0,1}, countdowns >= 0, the
+    force-flat window implies the no-new-position window)."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from gymfx_amd.calendar import CALENDAR_FEATURE_KEYS, compute_fx_calendar_features
+
+    flags = {"is_friday_risk_reduction_window", "is_no_new_position_window",
+             "is_force_flat_window", "is_broker_daily_break_near",
+             "broker_market_open", "is_no_trade_window"}
+
+    @settings(max_examples=120, deadline=None)
+    @given(st.integers(min_value=0, max_value=2_500_000_000))
+    def check(ts):
+        f = compute_fx_calendar_features(ts, timeframe_hours=1.0)
+        for k in CALENDAR_FEATURE_KEYS:
+            v = f[k]
+            assert v == v and abs(v) != float("inf"), (k, v)
+            if k in flags:
+                assert v in (0.0, 1.0), (k, v)
+            else:
+                assert v >= 0.0, (k, v)
+        if f["is_force_flat_window"]:
+            assert f["is_no_new_position_window"] == 1.0
+
+    check()
