@@ -203,3 +203,25 @@ def test_feistel_bijection_property():
         assert torch.equal(torch.sort(perm).values, torch.arange(n))
 
     check()
+
+
+def test_ddp_gloo_ws3_odd_world_size():
+    """Odd world size: the 1/world_size gradient averaging and the
+    rank-keyed rollout seeds must keep 3 replicas bit-identical too
+    (guards the scale run at any N)."""
+    import multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_ddp_worker, args=(r, 3, 29737, q))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(3):
+        rank, params = q.get(timeout=300)
+        results[rank] = params
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert results[0] == results[1] == results[2]
