@@ -63,10 +63,16 @@ def load_trainer_state_dict(trainer, sd: Dict[str, Any]) -> None:
 
 
 def save_checkpoint(trainer, path: str, *, extra: Dict[str, Any] = None) -> str:
+    """Atomic save: write to a sibling temp file, then rename — a crash
+    mid-save never corrupts an existing resume file."""
+    import os
+
     sd = trainer_state_dict(trainer)
     if extra:
         sd["extra"] = extra
-    torch.save(sd, path)
+    tmp = f"{path}.tmp.{os.getpid()}"
+    torch.save(sd, tmp)
+    os.replace(tmp, path)
     return path
 
 

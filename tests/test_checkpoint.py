@@ -124,3 +124,32 @@ def test_trace_file_phase_timings(tmp_path):
         assert rec["phases_ms"]["rollout"] > 0
         assert rec["phases_ms"]["update"] > 0
         assert "pi_loss" in rec
+
+
+def test_checkpoint_save_is_atomic(tmp_path, monkeypatch):
+    """A crash during torch.save must leave any pre-existing checkpoint
+    intact (save goes to a temp sibling, then os.replace)."""
+    import torch as _torch
+
+    from gymfx_amd.utils import checkpoint as ck
+
+    path = str(tmp_path / "c.pt")
+    t = _make("mlp")
+    save_checkpoint(t, path)
+    good = open(path, "rb").read()
+
+    real_save = _torch.save
+
+    def boom(obj, f, *a, **kw):
+        # write partial garbage then die, like an OOM/kill mid-serialize
+        with open(f, "wb") as fh:
+            fh.write(b"partial")
+        raise RuntimeError("crash mid-save")
+
+    monkeypatch.setattr(_torch, "save", boom)
+    with pytest.raises(RuntimeError):
+        ck.save_checkpoint(t, path)
+    assert open(path, "rb").read() == good  # original untouched
+    monkeypatch.setattr(_torch, "save", real_save)
+    t2 = _make("mlp")
+    load_checkpoint(t2, path)  # still loadable
