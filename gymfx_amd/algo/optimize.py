@@ -90,12 +90,31 @@ def optimize_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
         if not config.get("quiet_mode"):
             print(f"trial {t}: rap={res['rap']:.6f} {trial}")
     results.sort(key=lambda r: r["rap"], reverse=True)
+    # optional refinement: re-sample around the incumbent best inside a
+    # narrowed range (+-25% of each parameter's original span)
+    refine = int(config.get("optimization_refine_trials", 0) or 0)
+    for t in range(refine):
+        center = results[0]["params"]
+        narrowed: List[Tuple[str, float, float, str]] = []
+        for name, lo, hi, kind in schema:
+            span = (float(hi) - float(lo)) * 0.25
+            c = float(center[name])
+            narrowed.append((name, max(float(lo), c - span),
+                             min(float(hi), c + span), kind))
+        trial = sample_params(narrowed, rng)
+        res = _score_trial(config, trial)
+        res["trial"] = trials + t
+        res["refined"] = True
+        results.append(res)
+        results.sort(key=lambda r: r["rap"], reverse=True)
+        if not config.get("quiet_mode"):
+            print(f"refine {t}: rap={res['rap']:.6f} {trial}")
     best = results[0]
     return {
         "mode": "optimization",
         "strategy_plugin": strat_name,
         "schema": [list(s) for s in schema],
-        "trials": trials,
+        "trials": trials + refine,
         "best_params": best["params"],
         "best": best,
         "top5": results[:5],

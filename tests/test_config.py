@@ -240,3 +240,29 @@ def test_merge_precedence_property():
                 assert k not in merged
 
     check()
+
+
+def test_optimization_refinement_narrows_around_best():
+    """optimization_refine_trials re-samples inside ±25% of the incumbent
+    best's parameters and can only improve (or keep) the best rap."""
+    from gymfx_amd.algo.optimize import optimize_from_config
+    from gymfx_amd.config import DEFAULT_VALUES
+
+    cfg = {**DEFAULT_VALUES,
+           "data_feed_plugin": "synthetic_data_feed",
+           "synthetic_rows": 400, "n_envs": 8, "window_size": 8,
+           "device": "cpu", "seed": 0, "quiet_mode": True,
+           "strategy_plugin": "direct_atr_sltp",
+           "optimization_trials": 3, "optimization_steps": 32,
+           "optimization_refine_trials": 3}
+    out = optimize_from_config(cfg)
+    assert out["trials"] == 6
+    refined = [r for r in out["top5"] if r.get("refined")]
+    base_only = optimize_from_config({**cfg, "optimization_refine_trials": 0})
+    assert out["best"]["rap"] >= base_only["best"]["rap"] - 1e-12
+    # refined trials stay within the narrowed box around some incumbent
+    schema = {s[0]: (s[1], s[2]) for s in out["schema"]}
+    for r in refined:
+        for k, v in r["params"].items():
+            lo, hi = schema[k]
+            assert lo - 1e-9 <= float(v) <= hi + 1e-9
