@@ -93,4 +93,9 @@ DEFAULT_VALUES = {
     "checkpoint_file": None,     # save/resume path (mode=training)
     "resume": False,
     "trace_file": None,          # per-update phase-timing JSONL (HIP events)
+
+    # mode=optimization (built-in random search over hparam_schema)
+    "optimization_trials": 16,
+    "optimization_steps": 256,         # env steps scored per trial
+    "optimization_refine_trials": 0,   # extra trials around the incumbent
 }
