@@ -52,9 +52,11 @@ GFX_DEV float fast_tanh(float x) {
 // LDS staging (global loads for tile k+1 issue in registers while MFMAs for
 // tile k run — one __syncthreads per K-step).
 // NFRAG = MFMA n-fragments per wave:
-//   NFRAG=2 -> block tile 64x64  (small/medium M: rollout forward at N=4096)
-//   NFRAG=8 -> block tile 64x256 (update phase, M=65536/N=256: the A operand
-//              is streamed from HBM exactly once instead of N/64 times)
+//   NFRAG=2 -> block tile 64x64  (the DEFAULT at every shape: occupancy 8
+//              hides memory latency better than wider tiles save A-operand
+//              re-streaming — L2 absorbs the re-reads; see launch_gemm)
+//   NFRAG=4/8 -> 64x128 / 64x256 tiles, reachable via GYMFX_GEMM_WIDE
+//              (measured slower: occupancy 2-3, 67% SQ_WAIT)
 // Epilogues: ACT 0=none(f32 out) 1=none(bf16) 2=tanh(bf16)
 //            DACT_TANH: multiply by (1 - Y^2) elementwise (dgrad fused tanh')
 // ---------------------------------------------------------------------------
