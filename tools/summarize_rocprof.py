@@ -14,7 +14,13 @@ def main():
     ap.add_argument("--out", default="-")
     ap.add_argument("--top", type=int, default=25)
     args = ap.parse_args()
-    paths = sorted(glob.glob(args.db_glob))
+    import os
+    pattern = args.db_glob
+    if os.path.isdir(pattern):  # accept a rocprofv3 -d output directory
+        paths = sorted(glob.glob(os.path.join(pattern, "**", "*.db"),
+                                 recursive=True))
+    else:
+        paths = sorted(glob.glob(pattern))
     if not paths:
         print(f"no db matches {args.db_glob}", file=sys.stderr)
         sys.exit(1)
