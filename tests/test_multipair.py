@@ -98,3 +98,36 @@ def test_multipair_trainer_runs():
     tr = PPOTrainer(env, pc)
     stats = tr.train_update()
     assert np.isfinite(list(stats.values())).all()
+
+
+def test_concat_markets_block_meta_invariants():
+    """instrument_blocks tile [0, total) contiguously with the right names,
+    pip sizes, and column concatenation."""
+    import numpy as np
+
+    mds = [synthetic_ohlcv(100 + 30 * i, seed=i, instrument=f"I{i}")
+           for i in range(4)]
+    for i, m in enumerate(mds):
+        m.meta["pip_size"] = 0.0001 * (i + 1)
+    cat = concat_markets(mds)
+    blocks = cat.meta["instrument_blocks"]
+    assert len(blocks) == 4
+    off = 0
+    for i, b in enumerate(blocks):
+        assert b["lo"] == off
+        assert b["end"] == off + (100 + 30 * i)
+        assert b["instrument"] == f"I{i}"
+        assert b["pip_size"] == 0.0001 * (i + 1)
+        off = b["end"]
+    assert off == cat.n_rows == sum(100 + 30 * i for i in range(4))
+    # column data concatenated in order
+    np.testing.assert_array_equal(
+        cat.columns["CLOSE"][blocks[2]["lo"]:blocks[2]["end"]],
+        mds[2].columns["CLOSE"])
+
+
+def test_concat_markets_rejects_mismatched_columns():
+    a = synthetic_ohlcv(50, seed=1)
+    b = synthetic_ohlcv(50, seed=2, extra_feature_columns=2)
+    with pytest.raises(ValueError, match="same columns"):
+        concat_markets([a, b])
