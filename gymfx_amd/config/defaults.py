@@ -11,7 +11,7 @@ from __future__ import annotations
 
 DEFAULT_VALUES = {
     # execution
-    "mode": "inference",  # training|optimization|inference
+    "mode": "inference",  # training|optimization|inference|serve
     "driver_mode": "buy_hold",  # random|buy_hold|flat|replay
     "steps": 500,
 
@@ -93,6 +93,10 @@ DEFAULT_VALUES = {
     "checkpoint_file": None,     # save/resume path (mode=training)
     "resume": False,
     "trace_file": None,          # per-update phase-timing JSONL (HIP events)
+
+    # mode=serve (policy serving over HTTP; see serve.py)
+    "serve_host": "127.0.0.1",
+    "serve_port": 8400,
 
     # mode=optimization (built-in random search over hparam_schema)
     "optimization_trials": 16,

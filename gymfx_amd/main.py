@@ -92,8 +92,10 @@ def main(argv=None) -> None:
             file_config = {**remote_cfg, **file_config}
     config = merge_config(config, {}, {}, file_config, cli_args, unknown_args_dict)
 
-    if config.get("mode") not in {"training", "optimization", "inference"}:
-        raise ValueError("mode must be one of training|optimization|inference")
+    if config.get("mode") not in {"training", "optimization", "inference",
+                                   "serve"}:
+        raise ValueError(
+            "mode must be one of training|optimization|inference|serve")
 
     if config.get("mode") == "training":
         from .algo.ppo import train_from_config  # noqa: PLC0415
@@ -103,6 +105,10 @@ def main(argv=None) -> None:
         from .algo.optimize import optimize_from_config  # noqa: PLC0415
 
         summary = optimize_from_config(config)
+    elif config.get("mode") == "serve":
+        from .serve import serve_from_config  # noqa: PLC0415
+
+        summary = serve_from_config(config)
     elif (config.get("mode") == "inference" and config.get("checkpoint_file")
           and int(config.get("n_envs", 1) or 1) > 1):
         # vectorized policy evaluation (checkpoint -> greedy rollout);

@@ -1,0 +1,155 @@
+"""Policy serving: checkpoint -> batched actions over HTTP (mode=serve).
+
+The reference has no serving story (its inference mode drives one env with
+a scripted strategy, app/main.py:57-66).  This is the deployment-side
+counterpart of mode=training for the MI355X stack: load a checkpoint, hold
+the policy resident on the device, and answer batched observation requests
+with greedy (or sampled) actions through the same MFMA forward kernels the
+trainer uses — one process per GPU scales service capacity exactly like
+training.
+
+Endpoints
+---------
+GET  /health            -> {status, policy, obs_dim, n_actions, device}
+POST /act               -> {"obs": [[f32 x obs_dim] x B], "greedy": bool,
+                            "session": str|null}
+                        -> {actions, logp, value}
+POST /session/reset     -> {"session": str} zero a recurrent session state
+Recurrent policies keep per-session (h, c) server-side, keyed by the
+``session`` field; stateless MLP ignores it.
+
+NOTE: no ``from __future__ import annotations`` here — PEP 563 string
+annotations cannot resolve the closure-local pydantic models FastAPI needs
+to see as real classes (they would silently become query params).
+"""
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+import torch
+
+from .ops import api
+
+
+class PolicyServer:
+    """Device-resident policy with batched act(); framework-agnostic core
+    (the FastAPI app below is a thin shell so this stays testable)."""
+
+    def __init__(self, config: Dict[str, Any]):
+        from . import build_vec_environment
+        from .algo.ppo import PPOConfig, PPOTrainer
+        from .utils.checkpoint import load_checkpoint
+
+        cfg = dict(config)
+        cfg.setdefault("autoreset", False)
+        cfg.setdefault("env_start_mode", "zero")
+        # a tiny template env provides obs_dim + device + model shape
+        env = build_vec_environment(cfg)
+        env.reset(seed=int(cfg.get("seed") or 0))
+        pc = PPOConfig.from_config(cfg)
+        trainer = PPOTrainer(env, pc)
+        ckpt = cfg.get("checkpoint_file")
+        if ckpt:
+            load_checkpoint(trainer, ckpt)
+        self.policy = pc.policy
+        self.model = trainer.model
+        self.recurrent = trainer.recurrent
+        self.device = env.device
+        self.obs_dim = env.obs_dim
+        self.n_actions = 3
+        self.checkpoint = ckpt
+        self.sample_seed = int(cfg.get("seed") or 0) * 7919 + 17
+        self._step = 0
+        self._sessions: Dict[str, Dict[str, torch.Tensor]] = {}
+
+    # -- recurrent session state ---------------------------------------
+    def _session_state(self, session: str, batch: int) -> Dict[str, torch.Tensor]:
+        st = self._sessions.get(session)
+        if st is None or st["h"].shape[0] != batch:
+            st = self.model.alloc_state(batch)
+            self._sessions[session] = st
+        return st
+
+    def reset_session(self, session: str) -> bool:
+        return self._sessions.pop(session, None) is not None
+
+    # -- inference ------------------------------------------------------
+    def act(self, obs: np.ndarray, *, greedy: bool = True,
+            session: Optional[str] = None) -> Dict[str, List[float]]:
+        if obs.ndim != 2 or obs.shape[1] != self.obs_dim:
+            raise ValueError(
+                f"obs must be [batch, {self.obs_dim}], got {list(obs.shape)}")
+        B = obs.shape[0]
+        obs_bf16 = torch.from_numpy(np.ascontiguousarray(obs, dtype=np.float32)) \
+            .to(self.device).to(torch.bfloat16)
+        acts_buf = self.model.alloc_acts(B)
+        if self.recurrent:
+            state = self._session_state(session or "default", B)
+            head = self.model.step_forward(obs_bf16, state, acts_buf)
+        else:
+            head = self.model.forward(obs_bf16, acts_buf)
+        actions = torch.empty(B, dtype=torch.int64, device=self.device)
+        logp = torch.empty(B, dtype=torch.float32, device=self.device)
+        value = torch.empty(B, dtype=torch.float32, device=self.device)
+        api.sample_head(head, self.sample_seed, self._step, actions, logp,
+                        value, greedy=greedy)
+        self._step += 1
+        return {
+            "actions": actions.cpu().tolist(),
+            "logp": [float(x) for x in logp.cpu().tolist()],
+            "value": [float(x) for x in value.cpu().tolist()],
+        }
+
+
+def create_app(config: Dict[str, Any]):
+    """FastAPI app around a PolicyServer (import-light: fastapi only
+    needed when serving is actually used)."""
+    from fastapi import FastAPI, HTTPException
+    from pydantic import BaseModel
+
+    server = PolicyServer(config)
+    app = FastAPI(title="gymfx-amd policy server")
+    app.state.server = server
+
+    class ActRequest(BaseModel):
+        obs: List[List[float]]
+        greedy: bool = True
+        session: Optional[str] = None
+
+    class SessionRequest(BaseModel):
+        session: str
+
+    @app.get("/health")
+    def health():
+        return {
+            "status": "ok",
+            "policy": server.policy,
+            "obs_dim": server.obs_dim,
+            "n_actions": server.n_actions,
+            "device": str(server.device),
+            "checkpoint": server.checkpoint,
+        }
+
+    @app.post("/act")
+    def act(req: ActRequest):
+        try:
+            arr = np.asarray(req.obs, dtype=np.float32)
+            return server.act(arr, greedy=req.greedy, session=req.session)
+        except ValueError as exc:
+            raise HTTPException(status_code=422, detail=str(exc))
+
+    @app.post("/session/reset")
+    def session_reset(req: SessionRequest):
+        return {"reset": server.reset_session(req.session)}
+
+    return app
+
+
+def serve_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
+    """mode=serve entry: run uvicorn until interrupted."""
+    import uvicorn
+
+    host = str(config.get("serve_host", "127.0.0.1"))
+    port = int(config.get("serve_port", 8400))
+    app = create_app(config)
+    uvicorn.run(app, host=host, port=port, log_level="warning")
+    return {"mode": "serve", "host": host, "port": port}

@@ -1,0 +1,83 @@
+"""Policy serving (mode=serve): checkpoint -> batched actions over HTTP.
+
+The reference has no serving; this is the deployment-side counterpart of
+mode=training (serve.py).  Tested in-process with the starlette TestClient
+(no network)."""
+import numpy as np
+import pytest
+
+from gymfx_amd import build_vec_environment
+from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+from gymfx_amd.config import DEFAULT_VALUES
+from gymfx_amd.data.feed import synthetic_ohlcv
+from gymfx_amd.utils.checkpoint import save_checkpoint
+
+
+def _serve_cfg(tmp_path, policy="mlp"):
+    """Train 1 update, checkpoint, return a serve config for it."""
+    cfg = {**DEFAULT_VALUES,
+           "data_feed_plugin": "synthetic_data_feed",
+           "synthetic_rows": 1200, "n_envs": 16, "window_size": 8,
+           "device": "cpu", "seed": 4, "env_start_mode": "spread",
+           "autoreset": True, "position_size": 1000.0,
+           "policy_model": policy, "hidden_size": 16, "bptt_len": 4}
+    env = build_vec_environment(cfg)
+    env.reset(seed=4)
+    pc = PPOConfig.from_config({**cfg, "rollout_steps": 8, "minibatches": 2,
+                                "ppo_epochs": 1})
+    pc.rollout_steps, pc.minibatches, pc.ppo_epochs = 8, 2, 1
+    tr = PPOTrainer(env, pc)
+    tr.train_update()
+    ckpt = str(tmp_path / f"{policy}.pt")
+    save_checkpoint(tr, ckpt)
+    return {**cfg, "checkpoint_file": ckpt,
+            "rollout_steps": 8, "minibatches": 2, "ppo_epochs": 1}
+
+
+@pytest.mark.parametrize("policy", ["mlp", "lstm"])
+def test_serve_health_and_act(tmp_path, policy):
+    from fastapi.testclient import TestClient
+
+    from gymfx_amd.serve import create_app
+
+    cfg = _serve_cfg(tmp_path, policy)
+    app = create_app(cfg)
+    client = TestClient(app)
+
+    h = client.get("/health").json()
+    assert h["status"] == "ok" and h["policy"] == policy
+    D = h["obs_dim"]
+
+    rng = np.random.default_rng(0)
+    obs = rng.normal(size=(5, D)).astype(np.float32).tolist()
+    r = client.post("/act", json={"obs": obs, "greedy": True}).json()
+    assert len(r["actions"]) == 5
+    assert all(0 <= a <= 2 for a in r["actions"])
+    assert len(r["logp"]) == 5 and len(r["value"]) == 5
+    # greedy is deterministic for the same obs (stateless mlp only)
+    if policy == "mlp":
+        r2 = client.post("/act", json={"obs": obs, "greedy": True}).json()
+        assert r2["actions"] == r["actions"]
+    # wrong obs width -> 422, not a crash
+    bad = client.post("/act", json={"obs": [[1.0, 2.0]]})
+    assert bad.status_code == 422
+
+
+def test_serve_recurrent_sessions(tmp_path):
+    from fastapi.testclient import TestClient
+
+    from gymfx_amd.serve import create_app
+
+    app = create_app(_serve_cfg(tmp_path, "lstm"))
+    client = TestClient(app)
+    D = client.get("/health").json()["obs_dim"]
+    rng = np.random.default_rng(1)
+    obs = rng.normal(size=(3, D)).astype(np.float32).tolist()
+    # two sessions evolve independent recurrent state
+    a1 = client.post("/act", json={"obs": obs, "session": "s1"}).json()
+    for _ in range(3):  # advance s2's state so it diverges
+        client.post("/act", json={"obs": obs, "session": "s2"})
+    # resetting s1 then acting reproduces the first response (zero state)
+    assert client.post("/session/reset", json={"session": "s1"}).json()["reset"]
+    a1b = client.post("/act", json={"obs": obs, "session": "s1"}).json()
+    assert a1b["value"] == a1["value"]
