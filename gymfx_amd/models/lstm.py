@@ -115,6 +115,11 @@ class ActorCriticLSTM:
             api.transpose_bf16(self.w(name), t)
 
     def _wg_slabs(self, K: int, N: int) -> int:
+        # (GYMFX_WGRAD_SLABS overrides — tuning knob, same as models/mlp.py)
+        import os
+        ov = os.environ.get("GYMFX_WGRAD_SLABS")
+        if ov:
+            return int(ov)
         tiles = ((K + 63) // 64) * ((N + 63) // 64)
         s = self.wgrad_slabs
         while tiles * s < 2048:
