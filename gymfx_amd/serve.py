@@ -22,6 +22,7 @@ NOTE: no ``from __future__ import annotations`` here — PEP 563 string
 annotations cannot resolve the closure-local pydantic models FastAPI needs
 to see as real classes (they would silently become query params).
 """
+import threading
 from typing import Any, Dict, List, Optional
 
 import numpy as np
@@ -60,6 +61,11 @@ class PolicyServer:
         self.sample_seed = int(cfg.get("seed") or 0) * 7919 + 17
         self._step = 0
         self._sessions: Dict[str, Dict[str, torch.Tensor]] = {}
+        # FastAPI serves sync endpoints from a threadpool: one lock keeps
+        # the RNG step counter and per-session recurrent state consistent
+        # (GPU throughput comes from batching within a request, not from
+        # concurrent kernel submission).
+        self._lock = threading.Lock()
 
     # -- recurrent session state ---------------------------------------
     def _session_state(self, session: str, batch: int) -> Dict[str, torch.Tensor]:
@@ -70,7 +76,8 @@ class PolicyServer:
         return st
 
     def reset_session(self, session: str) -> bool:
-        return self._sessions.pop(session, None) is not None
+        with self._lock:
+            return self._sessions.pop(session, None) is not None
 
     # -- inference ------------------------------------------------------
     def act(self, obs: np.ndarray, *, greedy: bool = True,
@@ -78,6 +85,11 @@ class PolicyServer:
         if obs.ndim != 2 or obs.shape[1] != self.obs_dim:
             raise ValueError(
                 f"obs must be [batch, {self.obs_dim}], got {list(obs.shape)}")
+        with self._lock:
+            return self._act_locked(obs, greedy=greedy, session=session)
+
+    def _act_locked(self, obs: np.ndarray, *, greedy: bool,
+                    session: Optional[str]) -> Dict[str, List[float]]:
         B = obs.shape[0]
         obs_bf16 = torch.from_numpy(np.ascontiguousarray(obs, dtype=np.float32)) \
             .to(self.device).to(torch.bfloat16)

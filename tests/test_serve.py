@@ -81,3 +81,29 @@ def test_serve_recurrent_sessions(tmp_path):
     assert client.post("/session/reset", json={"session": "s1"}).json()["reset"]
     a1b = client.post("/act", json={"obs": obs, "session": "s1"}).json()
     assert a1b["value"] == a1["value"]
+
+
+def test_serve_concurrent_requests_consistent(tmp_path):
+    """Threaded clients hammering /act must produce valid, complete
+    responses (the server serializes RNG/session state internally)."""
+    from concurrent.futures import ThreadPoolExecutor
+
+    from fastapi.testclient import TestClient
+
+    from gymfx_amd.serve import create_app
+
+    app = create_app(_serve_cfg(tmp_path, "lstm"))
+    client = TestClient(app)
+    D = client.get("/health").json()["obs_dim"]
+    obs = np.zeros((4, D), dtype=np.float32).tolist()
+
+    def hit(i):
+        r = client.post("/act", json={"obs": obs, "session": f"s{i % 3}"})
+        assert r.status_code == 200
+        body = r.json()
+        assert len(body["actions"]) == 4
+        return body
+
+    with ThreadPoolExecutor(8) as ex:
+        results = list(ex.map(hit, range(40)))
+    assert len(results) == 40
