@@ -63,7 +63,13 @@ def _i(config: Dict[str, Any], key: str, default: int) -> int:
 
 def _b(config: Dict[str, Any], key: str, default: bool) -> bool:
     v = config.get(key, default)
-    return default if v is None else bool(v)
+    if v is None:
+        return default
+    if isinstance(v, str):  # bool("false") is True — route through the
+        from ..config.merger import convert_type  # config layer's coercion
+
+        v = convert_type(v)
+    return bool(v)
 
 
 @dataclass
