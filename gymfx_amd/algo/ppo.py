@@ -636,6 +636,9 @@ def train_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
         load_checkpoint(trainer, ckpt_path)
         resumed = True
     updates = int(config.get("train_updates", 10))
+    # periodic save (atomic temp+rename): a crash mid-run resumes from the
+    # last interval instead of zero.  0/None disables.
+    ckpt_every = int(config.get("checkpoint_interval") or 0)
     trace_path = config.get("trace_file")
     tracer = writer = None
     if trace_path:
@@ -659,6 +662,10 @@ def train_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
             writer.write({"update": u, "global_step": trainer.global_step,
                           "phases_ms": phases, **stats})
         history.append(stats)
+        if ckpt_path and ckpt_every and (u + 1) % ckpt_every == 0 and u + 1 < updates:
+            from ..utils.checkpoint import save_checkpoint
+
+            save_checkpoint(trainer, ckpt_path)
         if not config.get("quiet_mode") and (u % max(1, updates // 10) == 0):
             print(f"update {u}: {stats}")
     if writer is not None:

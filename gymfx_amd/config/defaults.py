@@ -91,6 +91,7 @@ DEFAULT_VALUES = {
     "fuse_sample": True,         # sample inside the env-step kernel (GPU)
     "overlap_gather": False,     # side-stream mb gather (measured slower)
     "checkpoint_file": None,     # save/resume path (mode=training)
+    "checkpoint_interval": 0,    # save every N updates (0 = final only)
     "resume": False,
     "trace_file": None,          # per-update phase-timing JSONL (HIP events)
 

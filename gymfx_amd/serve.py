@@ -23,6 +23,7 @@ annotations cannot resolve the closure-local pydantic models FastAPI needs
 to see as real classes (they would silently become query params).
 """
 import threading
+from collections import OrderedDict
 from typing import Any, Dict, List, Optional
 
 import numpy as np
@@ -31,36 +32,60 @@ import torch
 from .ops import api
 
 
+class SessionBatchMismatch(ValueError):
+    """A recurrent session was continued with a different batch size —
+    silently reallocating would corrupt LSTM continuity, so refuse."""
+
+
 class PolicyServer:
     """Device-resident policy with batched act(); framework-agnostic core
     (the FastAPI app below is a thin shell so this stays testable)."""
 
     def __init__(self, config: Dict[str, Any]):
         from . import build_vec_environment
-        from .algo.ppo import PPOConfig, PPOTrainer
-        from .utils.checkpoint import load_checkpoint
+        from .algo.ppo import PPOConfig
 
         cfg = dict(config)
         cfg.setdefault("autoreset", False)
         cfg.setdefault("env_start_mode", "zero")
-        # a tiny template env provides obs_dim + device + model shape
+        # a tiny template env provides obs_dim + device; the model shape
+        # comes from the checkpoint (inference-only load: the training-time
+        # n_envs does NOT have to match this template env)
         env = build_vec_environment(cfg)
         env.reset(seed=int(cfg.get("seed") or 0))
         pc = PPOConfig.from_config(cfg)
-        trainer = PPOTrainer(env, pc)
         ckpt = cfg.get("checkpoint_file")
         if ckpt:
-            load_checkpoint(trainer, ckpt)
-        self.policy = pc.policy
-        self.model = trainer.model
-        self.recurrent = trainer.recurrent
+            from .utils.checkpoint import load_model_for_inference
+
+            model, meta = load_model_for_inference(ckpt, env.device)
+            if meta["obs_dim"] != env.obs_dim:
+                raise ValueError(
+                    f"checkpoint obs_dim {meta['obs_dim']} != serving env "
+                    f"obs_dim {env.obs_dim} (feature/window config mismatch)")
+            self.policy = meta["policy"]
+        else:
+            if pc.policy == "lstm":
+                from .models.lstm import ActorCriticLSTM
+
+                model = ActorCriticLSTM(env.obs_dim, 3, pc.hidden,
+                                        device=env.device, seed=pc.seed)
+            else:
+                from .models.mlp import ActorCriticMLP
+
+                model = ActorCriticMLP(env.obs_dim, 3, pc.hidden,
+                                       device=env.device, seed=pc.seed)
+            self.policy = pc.policy
+        self.model = model
+        self.recurrent = self.policy == "lstm"
         self.device = env.device
         self.obs_dim = env.obs_dim
         self.n_actions = 3
         self.checkpoint = ckpt
         self.sample_seed = int(cfg.get("seed") or 0) * 7919 + 17
+        self.max_sessions = int(cfg.get("serve_max_sessions") or 1024)
         self._step = 0
-        self._sessions: Dict[str, Dict[str, torch.Tensor]] = {}
+        self._sessions: "OrderedDict[str, Dict[str, torch.Tensor]]" = OrderedDict()
         # FastAPI serves sync endpoints from a threadpool: one lock keeps
         # the RNG step counter and per-session recurrent state consistent
         # (GPU throughput comes from batching within a request, not from
@@ -70,9 +95,20 @@ class PolicyServer:
     # -- recurrent session state ---------------------------------------
     def _session_state(self, session: str, batch: int) -> Dict[str, torch.Tensor]:
         st = self._sessions.get(session)
-        if st is None or st["h"].shape[0] != batch:
-            st = self.model.alloc_state(batch)
-            self._sessions[session] = st
+        if st is not None:
+            if st["h"].shape[0] != batch:
+                raise SessionBatchMismatch(
+                    f"session {session!r} holds state for batch "
+                    f"{st['h'].shape[0]}, request has batch {batch}; reset "
+                    "the session or keep the batch size constant")
+            self._sessions.move_to_end(session)  # LRU touch
+            return st
+        st = self.model.alloc_state(batch)
+        self._sessions[session] = st
+        # bounded session map: arbitrary client-supplied keys must not grow
+        # device memory without limit — evict least-recently-used
+        while len(self._sessions) > self.max_sessions:
+            self._sessions.popitem(last=False)
         return st
 
     def reset_session(self, session: str) -> bool:
@@ -146,6 +182,8 @@ def create_app(config: Dict[str, Any]):
         try:
             arr = np.asarray(req.obs, dtype=np.float32)
             return server.act(arr, greedy=req.greedy, session=req.session)
+        except SessionBatchMismatch as exc:
+            raise HTTPException(status_code=409, detail=str(exc))
         except ValueError as exc:
             raise HTTPException(status_code=422, detail=str(exc))
 

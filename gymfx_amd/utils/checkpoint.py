@@ -77,6 +77,47 @@ def save_checkpoint(trainer, path: str, *, extra: Dict[str, Any] = None) -> str:
 
 
 def load_checkpoint(trainer, path: str) -> Dict[str, Any]:
-    sd = torch.load(path, map_location="cpu", weights_only=False)
+    # weights_only: the payload is tensors / scalars / plain dicts only, so
+    # the restricted unpickler suffices — an untrusted checkpoint file can
+    # not execute code on load.
+    sd = torch.load(path, map_location="cpu", weights_only=True)
     load_trainer_state_dict(trainer, sd)
     return sd.get("extra", {})
+
+
+def load_model_for_inference(path: str, device) -> tuple:
+    """Weights-only load for serving / evaluation.
+
+    Unlike :func:`load_trainer_state_dict` this does NOT require the
+    checkpoint's ``n_envs`` to match anything — a policy trained with
+    n_envs=4096 serves from a 16-env template just fine.  Only the model
+    shape (policy kind, obs_dim, hidden) is taken from the checkpoint.
+
+    Returns ``(model, meta)`` where ``meta`` has ``policy``, ``obs_dim``,
+    ``hidden``, ``n_envs`` (the training-time value, informational) and
+    ``update_count``.
+    """
+    sd = torch.load(path, map_location="cpu", weights_only=True)
+    if sd.get("schema") != SCHEMA:
+        raise ValueError(f"unknown checkpoint schema: {sd.get('schema')!r}")
+    policy = sd["policy"]
+    obs_dim = int(sd["obs_dim"])
+    msd = sd["model"]
+    hidden = int(msd["hidden"].item()) if "hidden" in msd else 256
+    if policy == "lstm":
+        from ..models.lstm import ActorCriticLSTM
+
+        model = ActorCriticLSTM(obs_dim, 3, hidden, device=device, seed=0)
+    else:
+        from ..models.mlp import ActorCriticMLP
+
+        model = ActorCriticMLP(obs_dim, 3, hidden, device=device, seed=0)
+    model.load_state_dict(msd)
+    meta = {
+        "policy": policy,
+        "obs_dim": obs_dim,
+        "hidden": hidden,
+        "n_envs": int(sd.get("n_envs", 0)),
+        "update_count": int(sd.get("update_count", 0)),
+    }
+    return model, meta

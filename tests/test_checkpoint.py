@@ -153,3 +153,60 @@ def test_checkpoint_save_is_atomic(tmp_path, monkeypatch):
     monkeypatch.setattr(_torch, "save", real_save)
     t2 = _make("mlp")
     load_checkpoint(t2, path)  # still loadable
+
+
+def test_load_model_for_inference_ignores_n_envs(tmp_path):
+    """Inference load takes only the model shape from the checkpoint —
+    training-time n_envs is informational (serving from a
+    4096-env checkpoint with a 16-env template must work)."""
+    from gymfx_amd.utils.checkpoint import load_model_for_inference
+
+    path = str(tmp_path / "ckpt.pt")
+    tr = _make("mlp")
+    tr.train_update()
+    save_checkpoint(tr, path)
+    model, meta = load_model_for_inference(path, torch.device("cpu"))
+    assert meta["policy"] == "mlp" and meta["n_envs"] == 16
+    assert meta["obs_dim"] == tr.env.obs_dim and meta["hidden"] == 16
+    assert torch.equal(model.params, tr.model.params)
+
+
+def test_checkpoint_loads_with_weights_only(tmp_path):
+    """The checkpoint payload must stay within torch.load(weights_only=True)
+    territory — no pickled code objects in the file."""
+    path = str(tmp_path / "ckpt.pt")
+    tr = _make("lstm")
+    tr.train_update()
+    save_checkpoint(tr, path, extra={"config": {"seed": 31, "note": "x"}})
+    sd = torch.load(path, map_location="cpu", weights_only=True)
+    assert sd["schema"] == "gymfx.ckpt.v1"
+
+
+def test_periodic_checkpoint_interval(tmp_path):
+    """checkpoint_interval=N saves inside the update loop, so a crash after
+    update k>=N resumes from the last interval, not from zero."""
+    from gymfx_amd.algo.ppo import train_from_config
+    from gymfx_amd.config import DEFAULT_VALUES
+
+    path = str(tmp_path / "ckpt.pt")
+    cfg = {**DEFAULT_VALUES,
+           "data_feed_plugin": "synthetic_data_feed", "synthetic_rows": 600,
+           "n_envs": 8, "window_size": 8, "device": "cpu", "seed": 3,
+           "rollout_steps": 8, "minibatches": 2, "ppo_epochs": 1,
+           "hidden_size": 16, "train_updates": 5, "checkpoint_interval": 2,
+           "checkpoint_file": path, "quiet_mode": True}
+    saves = []
+    import gymfx_amd.utils.checkpoint as ck
+    orig = ck.save_checkpoint
+
+    def spy(trainer, p, **kw):
+        saves.append(trainer.update_count)
+        return orig(trainer, p, **kw)
+
+    ck.save_checkpoint = spy
+    try:
+        train_from_config(cfg)
+    finally:
+        ck.save_checkpoint = orig
+    # interval saves after updates 2 and 4, final save after 5
+    assert saves == [2, 4, 5]

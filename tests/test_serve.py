@@ -107,3 +107,60 @@ def test_serve_concurrent_requests_consistent(tmp_path):
     with ThreadPoolExecutor(8) as ex:
         results = list(ex.map(hit, range(40)))
     assert len(results) == 40
+
+
+def test_serve_loads_checkpoint_from_different_n_envs(tmp_path):
+    """Serving only needs model weights: a policy trained at one n_envs
+    must load into a serve config with a different (smaller) template env
+    (inference-only load path, utils/checkpoint.load_model_for_inference)."""
+    from fastapi.testclient import TestClient
+
+    from gymfx_amd.serve import create_app
+
+    cfg = _serve_cfg(tmp_path, "mlp")   # trained with n_envs=16
+    cfg = {**cfg, "n_envs": 4}          # serve template is 4 envs
+    app = create_app(cfg)
+    client = TestClient(app)
+    h = client.get("/health").json()
+    obs = np.zeros((2, h["obs_dim"]), dtype=np.float32).tolist()
+    r = client.post("/act", json={"obs": obs})
+    assert r.status_code == 200 and len(r.json()["actions"]) == 2
+
+
+def test_serve_session_batch_mismatch_409(tmp_path):
+    """Continuing a recurrent session with a different batch size must be
+    refused (409), never silently reset; after /session/reset the new batch
+    size is accepted."""
+    from fastapi.testclient import TestClient
+
+    from gymfx_amd.serve import create_app
+
+    app = create_app(_serve_cfg(tmp_path, "lstm"))
+    client = TestClient(app)
+    D = client.get("/health").json()["obs_dim"]
+    obs3 = np.zeros((3, D), dtype=np.float32).tolist()
+    obs2 = np.zeros((2, D), dtype=np.float32).tolist()
+    assert client.post("/act", json={"obs": obs3, "session": "x"}).status_code == 200
+    r = client.post("/act", json={"obs": obs2, "session": "x"})
+    assert r.status_code == 409
+    client.post("/session/reset", json={"session": "x"})
+    assert client.post("/act", json={"obs": obs2, "session": "x"}).status_code == 200
+
+
+def test_serve_session_map_bounded_lru(tmp_path):
+    """Arbitrary client session keys must not grow state unboundedly:
+    the map is LRU-bounded by serve_max_sessions."""
+    from gymfx_amd.serve import PolicyServer
+
+    cfg = {**_serve_cfg(tmp_path, "lstm"), "serve_max_sessions": 4}
+    srv = PolicyServer(cfg)
+    obs = np.zeros((1, srv.obs_dim), dtype=np.float32)
+    for i in range(10):
+        srv.act(obs, session=f"k{i}")
+    assert len(srv._sessions) == 4
+    # most-recent keys survive
+    assert set(srv._sessions) == {"k6", "k7", "k8", "k9"}
+    # touching k6 then adding a new key evicts k7, not k6
+    srv.act(obs, session="k6")
+    srv.act(obs, session="new")
+    assert "k6" in srv._sessions and "k7" not in srv._sessions
