@@ -68,7 +68,10 @@ def main() -> None:
         torch.cuda.set_device(device)
 
     pg = None
-    if world_size > 1:
+    # init the process group whenever a launcher provided rendezvous env
+    # (even at world_size 1): the driver's torchrun N=1 run then exercises
+    # RCCL communicator init on hardware, not just the ws>1 path.
+    if world_size > 1 or ("MASTER_ADDR" in os.environ and "MASTER_PORT" in os.environ):
         import torch.distributed as dist
 
         backend = "nccl" if device.type == "cuda" else "gloo"
@@ -133,12 +136,34 @@ def main() -> None:
 
     for _ in range(args.warmup):
         trainer.train_update(with_stats=False)
+    if trainer.use_graphs and not trainer._graphs_ready:
+        trainer._capture_graphs()  # warmup=0 safety: keep capture untimed
+    # per-phase attribution: CUDA events bracket the
+    # rollout and update phases of every timed step — no host syncs inside
+    # the timed region; elapsed_time is read after the closing barrier.
+    use_ev = device.type == "cuda"
+    if use_ev:
+        evs = [tuple(torch.cuda.Event(enable_timing=True) for _ in range(3))
+               for _ in range(args.steps)]
     barrier_sync()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        trainer.train_update(with_stats=False)
+    for i in range(args.steps):
+        if use_ev:
+            e0, e1, e2 = evs[i]
+            e0.record()
+            trainer.collect_rollout()
+            e1.record()
+            trainer.update(with_stats=False)
+            e2.record()
+        else:
+            trainer.collect_rollout()
+            trainer.update(with_stats=False)
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    rollout_ms = update_ms = None
+    if use_ev:
+        rollout_ms = sum(e0.elapsed_time(e1) for e0, e1, _ in evs) / args.steps
+        update_ms = sum(e1.elapsed_time(e2) for _, e1, e2 in evs) / args.steps
 
     # MAX over ranks (driver contract)
     if world_size > 1:
@@ -164,6 +189,7 @@ def main() -> None:
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
+            "phases_ms": {"rollout": rollout_ms, "update": update_ms},
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "bf16",
@@ -184,7 +210,7 @@ def main() -> None:
             },
         }))
 
-    if world_size > 1:
+    if pg is not None:
         import torch.distributed as dist
 
         dist.destroy_process_group()
