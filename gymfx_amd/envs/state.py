@@ -82,6 +82,8 @@ class EnvState:
     # sharpe reward ring
     rew_ring: torch.Tensor        # f32 [N, sharpe_window]
     rew_count: torch.Tensor       # i32 [N]
+    rew_s1: torch.Tensor          # f64 [N] sharpe running sum (O(1)/step)
+    rew_s2: torch.Tensor          # f64 [N] sharpe running sum of squares
     # per-trade stats (metrics rollup)
     trade_won: torch.Tensor       # i32 [N]
     trade_lost: torch.Tensor      # i32 [N]
@@ -154,6 +156,8 @@ def alloc_state(params: EnvParams, device: torch.device) -> EnvState:
         prev_close_atr=torch.full((N,), float("nan"), **f32),
         rew_ring=torch.zeros(N, max(params.sharpe_window, 2), **f32),
         rew_count=torch.zeros(N, **i32),
+        rew_s1=torch.zeros(N, **f64),
+        rew_s2=torch.zeros(N, **f64),
         trade_won=torch.zeros(N, **i32),
         trade_lost=torch.zeros(N, **i32),
         trade_pnl_sum=torch.zeros(N, **f64),
@@ -211,6 +215,8 @@ def reset_state_(st: EnvState, params: EnvParams, mask: torch.Tensor) -> None:
     st.prev_close_atr[mask] = float("nan")
     st.rew_ring[mask] = 0.0
     st.rew_count[mask] = 0
+    st.rew_s1[mask] = 0.0
+    st.rew_s2[mask] = 0.0
     st.trade_won[mask] = 0
     st.trade_lost[mask] = 0
     st.trade_pnl_sum[mask] = 0.0

@@ -231,6 +231,8 @@ struct GymFxEngine {
     P.prev_close_atr = ptr<float>(state, "prev_close_atr");
     P.rew_ring = ptr<float>(state, "rew_ring");
     P.rew_count = ptr<int>(state, "rew_count");
+    P.rew_s1 = ptr<double>(state, "rew_s1");
+    P.rew_s2 = ptr<double>(state, "rew_s2");
     P.trade_won = ptr<int>(state, "trade_won");
     P.trade_lost = ptr<int>(state, "trade_lost");
     P.trade_pnl_sum = ptr<double>(state, "trade_pnl_sum");

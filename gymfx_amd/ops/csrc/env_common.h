@@ -167,6 +167,7 @@ struct EnvPtrs {
   float *prev_close_atr;
   float *rew_ring;           // [N, W_sharpe]
   int *rew_count;
+  double *rew_s1, *rew_s2;   // O(1) sharpe running sum / sum-of-squares
   int *trade_won, *trade_lost;
   double *trade_pnl_sum, *trade_pnl_sumsq;
   double *metric_peak, *max_dd_money, *max_dd_pct, *ret_sum, *ret_sumsq;

@@ -73,6 +73,8 @@ GFX_DEV void reset_env(const EnvPtrs& P, const EnvParamsK& K, int n) {
   P.prev_close_atr[n] = __builtin_nanf("");
   for (int i = 0; i < K.sharpe_window; ++i) P.rew_ring[(int64_t)n * K.sharpe_window + i] = 0.f;
   P.rew_count[n] = 0;
+  P.rew_s1[n] = 0.0;
+  P.rew_s2[n] = 0.0;
   P.trade_won[n] = 0;
   P.trade_lost[n] = 0;
   P.trade_pnl_sum[n] = 0.0;
@@ -424,23 +426,28 @@ GFX_DEV void env_step_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
     const double ic = K.initial_cash != 0.0 ? K.initial_cash : 1.0;
     const double r_step = (P.equity[n] - P.prev_equity[n]) / ic;
     if (K.reward_id == REWARD_SHARPE) {
+      // O(1)/step rolling Sharpe: the ring holds the
+      // window values only so the evicted term can be subtracted from the
+      // f64 running sums; the f32 rescans of round 1 cost ~8% of the env
+      // kernel at W=64 (profiles/bench_mlp_sharpe.json).  f64 accumulators
+      // make the cancellation in s2 - m*mean^2 benign at FX return scale.
       const int W = K.sharpe_window;
-      int cnt = P.rew_count[n];
-      P.rew_ring[(int64_t)n * W + (cnt % W)] = (float)r_step;
+      const int cnt = P.rew_count[n];
+      const int slot = cnt % W;
+      const float evicted = P.rew_ring[(int64_t)n * W + slot];
+      const float nv = (float)r_step;
+      P.rew_ring[(int64_t)n * W + slot] = nv;
       P.rew_count[n] = cnt + 1;
-      int m = min(P.rew_count[n], W);
+      const double s1 = P.rew_s1[n] + ((double)nv - (double)evicted);
+      const double s2 = P.rew_s2[n] + ((double)nv * nv - (double)evicted * evicted);
+      P.rew_s1[n] = s1;
+      P.rew_s2[n] = s2;
+      const int m = min(cnt + 1, W);
       if (m >= 2) {
-        float s1 = 0.f;
-        for (int i = 0; i < m; ++i) s1 += P.rew_ring[(int64_t)n * W + i];
-        float mean = s1 / (float)m;
-        float ss = 0.f;
-        for (int i = 0; i < m; ++i) {
-          float d = P.rew_ring[(int64_t)n * W + i] - mean;
-          ss += d * d;
-        }
-        float var = ss / (float)(m - 1);
-        float stdv = sqrtf(var);
-        if (stdv > 0.f) base_reward = (double)(mean / stdv) * sqrt(K.annualization);
+        const double mean = s1 / (double)m;
+        const double var = fmax((s2 - (double)m * mean * mean) / (double)(m - 1), 0.0);
+        const double stdv = sqrt(var);
+        if (stdv > 0.0) base_reward = (mean / stdv) * sqrt(K.annualization);
       }
     } else if (K.reward_id == REWARD_DD) {
       P.peak_equity[n] = fmax(P.peak_equity[n], fmax(P.equity[n], P.prev_equity[n]));

@@ -9,6 +9,9 @@ any other failure is a real bug.
 import os
 import sys
 
+# torchrun children get sys.path[0]=tests/ — put the repo root back
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 
@@ -41,9 +44,22 @@ def main() -> int:
         print(f"rank {rank}: WS2_PROBE_OK")
         return 0
     except Exception as exc:  # noqa: BLE001
+        import traceback
+
         msg = str(exc)
-        if "Duplicate GPU" in msg or "invalid usage" in msg.lower():
-            print(f"rank {rank}: RCCL refuses shared device: {msg}")
+        tb = traceback.format_exc()
+        # torchrun swallows child tracebacks: persist them for the caller
+        try:
+            os.makedirs("gpurun_out", exist_ok=True)
+            with open(f"gpurun_out/ws2_probe_rank{rank}.log", "w") as f:
+                f.write(tb)
+        except OSError:
+            pass
+        print(tb, flush=True)
+        low = msg.lower()
+        if ("duplicate gpu" in low or "invalid usage" in low
+                or "invalid argument" in low or "unhandled system error" in low):
+            print(f"rank {rank}: RCCL refuses shared device: {msg}", flush=True)
             return 77
         raise
 
