@@ -1,50 +1,57 @@
 """Config precedence merge + typed coercion of unknown CLI args.
 
-Same precedence contract as the reference merge
-(/root/reference/app/config_merger.py:37-51): plugin defaults are the lowest
-tier, then DEFAULT_VALUES, then the config file, then known CLI args (only
-when not None), then unknown ``--key value`` args with string->typed coercion
-(/root/reference/app/config_merger.py:19-35).
+Behavioral contract (matches the reference's merge semantics,
+/root/reference/app/config_merger.py:37-51, re-expressed here as a tier
+fold): plugin defaults are the lowest tier, then DEFAULT_VALUES, then the
+config file, then known CLI args (only when not None), then unknown
+``--key value`` args with string->typed coercion.
 """
 from __future__ import annotations
 
 from typing import Any, Dict, List, Optional
 
+_BOOL_WORDS = {"true": True, "false": False}
+_NONE_WORDS = frozenset({"none", "null"})
 
-def process_unknown_args(unknown_args: List[str]) -> Dict[str, Any]:
-    parsed: Dict[str, Any] = {}
-    i = 0
-    while i < len(unknown_args):
-        key = unknown_args[i]
-        if not key.startswith("--"):
-            i += 1
-            continue
-        if i + 1 < len(unknown_args) and not unknown_args[i + 1].startswith("--"):
-            parsed[key.lstrip("-")] = unknown_args[i + 1]
-            i += 2
-        else:
-            parsed[key.lstrip("-")] = True
-            i += 1
-    return parsed
+
+def process_unknown_args(tokens: List[str]) -> Dict[str, Any]:
+    """Pair up ``--key value`` tokens from argparse's unknown-args list.
+
+    A ``--flag`` immediately followed by another ``--flag`` (or by the end
+    of the list) is a valueless switch and maps to True; stray positional
+    tokens with no preceding flag are dropped.
+    """
+    overrides: Dict[str, Any] = {}
+    pending: Optional[str] = None
+    for tok in tokens:
+        if tok.startswith("--"):
+            if pending is not None:
+                overrides[pending] = True
+            pending = tok.lstrip("-")
+        elif pending is not None:
+            overrides[pending] = tok
+            pending = None
+    if pending is not None:
+        overrides[pending] = True
+    return overrides
 
 
 def convert_type(value: Any) -> Any:
-    if isinstance(value, bool):
-        return value
+    """Best-effort typed coercion of a CLI string: bool words, none words,
+    int, float, else the string unchanged.  Non-strings pass through."""
     if not isinstance(value, str):
         return value
-    lowered = value.strip().lower()
-    if lowered in {"true", "false"}:
-        return lowered == "true"
-    if lowered in {"none", "null"}:
+    word = value.strip().lower()
+    if word in _BOOL_WORDS:
+        return _BOOL_WORDS[word]
+    if word in _NONE_WORDS:
         return None
-    try:
-        return int(value)
-    except ValueError:
+    for parse in (int, float):
         try:
-            return float(value)
+            return parse(value)
         except ValueError:
-            return value
+            continue
+    return value
 
 
 def merge_config(
@@ -55,14 +62,21 @@ def merge_config(
     cli_args: Optional[Dict[str, Any]],
     unknown_args: Optional[Dict[str, Any]],
 ) -> Dict[str, Any]:
+    """Fold the precedence tiers lowest-to-highest into one dict.
+
+    Known CLI args participate only when set (not None); unknown CLI args
+    win over everything and are coerced from their raw string form.
+    """
+    tiers = (
+        plugin_params1,
+        plugin_params2,
+        defaults,
+        file_config,
+        {k: v for k, v in (cli_args or {}).items() if v is not None},
+        {k: convert_type(v) for k, v in (unknown_args or {}).items()},
+    )
     merged: Dict[str, Any] = {}
-    merged.update(plugin_params1 or {})
-    merged.update(plugin_params2 or {})
-    merged.update(defaults or {})
-    merged.update(file_config or {})
-    for key, value in (cli_args or {}).items():
-        if value is not None:
-            merged[key] = value
-    for key, value in (unknown_args or {}).items():
-        merged[key] = convert_type(value)
+    for tier in tiers:
+        if tier:
+            merged.update(tier)
     return merged
