@@ -225,3 +225,50 @@ def test_ddp_gloo_ws3_odd_world_size():
         p.join(timeout=60)
         assert p.exitcode == 0
     assert results[0] == results[1] == results[2]
+
+
+# ---------------------------------------------------------------------------
+# failure path: a dead peer rank must surface as a timely error, not a hang
+# (failure-detection stance of SURVEY.md §5.3)
+# ---------------------------------------------------------------------------
+
+def _ddp_dying_worker(rank, world_size, port, out_q):
+    import time as _time
+
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    if rank == 1:
+        # die mid-job WITHOUT destroying the pg — simulates a crashed rank
+        os._exit(17)
+    t = _make_trainer(rank=rank, world_size=world_size, pg=dist.group.WORLD)
+    t._reducer.timeout_s = 5.0
+    t0 = _time.perf_counter()
+    try:
+        t.train_update()
+        out_q.put((rank, "no-error", _time.perf_counter() - t0))
+    except RuntimeError as exc:
+        out_q.put((rank, f"raised: {exc}", _time.perf_counter() - t0))
+
+
+def test_ddp_dead_rank_surfaces_watchdog_error():
+    """Kill rank 1 before its first collective: rank 0's GradAllReducer
+    watchdog must raise within its timeout instead of hanging forever."""
+    import multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_ddp_dying_worker, args=(r, 2, 29739, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    rank, outcome, elapsed = q.get(timeout=180)
+    for p in procs:
+        p.join(timeout=60)
+        p.kill()  # belt-and-braces; join above should have reaped both
+    assert rank == 0
+    assert outcome.startswith("raised:"), outcome
+    # raised promptly (5 s watchdog + slack), not after some giant default
+    assert elapsed < 60.0, elapsed
