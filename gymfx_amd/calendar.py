@@ -303,6 +303,8 @@ def rollover_rate_lookup(rate_data, location: str, month: str) -> float:
     — the reference's monthly central-bank rate schema
     (simulation_engines/bakeoff.py:104-113).  Falls back to the latest
     month at or before the requested one."""
+    if hasattr(rate_data, "to_dict"):  # pandas DataFrame (reference schema)
+        rate_data = rate_data.to_dict("records")
     best = None
     best_time = ""
     for row in rate_data:
