@@ -44,6 +44,10 @@ PREP_NAMES = {
     "default_preprocessor": PREP_DEFAULT,
     "feature_window_preprocessor": PREP_FEATURE_WINDOW,
 }
+# execution-realism policies (ids == ops/csrc/env_common.h enums)
+_COLLISION_POLICIES = {"worst_case": 0, "ohlc": 1, "adaptive": 2}
+_LIMIT_POLICIES = {"touch": 0, "cross": 1, "conservative": 2}
+_MARGIN_MODELS = {"leveraged": 0, "standard": 1}
 RISK_MODES = {
     "fixed_atr": RISK_FIXED_ATR,
     "rel_volume_aware_atr": RISK_REL_VOLUME_AWARE,
@@ -145,6 +149,12 @@ class EnvParams:
     rollover_hour_utc: int = 22
     enforce_margin_preflight: bool = False
     timeframe_hours: float = 0.0
+    # execution-realism tier (execution_cost_profile.v1 semantics)
+    intrabar_collision_policy: int = 0   # 0 worst_case, 1 ohlc, 2 adaptive
+    limit_fill_policy: int = 0           # 0 touch, 1 cross, 2 conservative
+    latency_bars: int = 0                # floor(latency_ms / bar duration)
+    margin_model: int = 0                # 0 leveraged, 1 standard
+    margin_init_rate: float = 0.03       # used by margin_model=standard
     # event-context overlay
     event_context_execution_overlay: bool = False
     event_context_no_trade_column: str = "event_no_trade_window_active"
@@ -259,6 +269,14 @@ class EnvParams:
 
         p.financing_enabled = _b(config, "financing_enabled", False)
         p.enforce_margin_preflight = _b(config, "enforce_margin_preflight", False)
+        p.intrabar_collision_policy = _COLLISION_POLICIES[
+            str(config.get("intrabar_collision_policy", "worst_case"))]
+        p.limit_fill_policy = _LIMIT_POLICIES[
+            str(config.get("limit_fill_policy", "touch"))]
+        p.margin_model = _MARGIN_MODELS[
+            str(config.get("margin_model", "leveraged"))]
+        p.margin_init_rate = _f(config, "margin_init_rate", 0.03)
+        latency_ms = _f(config, "latency_ms", 0.0)
         p.rollover_hour_utc = int(config.get("rollover_hour_utc", 22) or 22)
         p.stage_b_force_close_obs = _b(config, "stage_b_force_close_obs", False)
         p.force_close_window_hours = _i(config, "force_close_window_hours", 4)
@@ -310,6 +328,18 @@ class EnvParams:
             p.financing_enabled = p.financing_enabled or prof.financing_enabled
             p.enforce_margin_preflight = (p.enforce_margin_preflight
                                           or prof.enforce_margin_preflight)
+            # execution-realism tier: the profile is authoritative for the
+            # policy fields (round 1 parsed-then-dropped them — VERDICT #2)
+            p.intrabar_collision_policy = _COLLISION_POLICIES[
+                prof.intrabar_collision_policy]
+            p.limit_fill_policy = _LIMIT_POLICIES[prof.limit_fill_policy]
+            p.margin_model = _MARGIN_MODELS[prof.margin_model]
+            latency_ms = float(prof.latency_ms)
+
+        # sub-bar latency collapses to next-open at OHLC granularity;
+        # each full bar duration of latency delays the fill one more bar
+        bar_ms = p.timeframe_hours * 3_600_000.0
+        p.latency_bars = int(latency_ms // bar_ms) if bar_ms > 0 else 0
 
         p.finalize()
         return p

@@ -179,12 +179,19 @@ def step_torch(
     )
 
     # ---- 1. pending market fills at open(t) -----------------------------
-    close_m = valid & st.pend_close & (st.pos != 0)
+    # latency_bars holds the order in transit (LatencyModel at OHLC
+    # granularity): the fill and the pend-clear below are skipped while the
+    # wait counter drains
+    in_transit = valid & (st.pend_wait > 0) & (st.pend_close | (st.pend_open_dir != 0))
+    st.pend_wait = torch.where(in_transit, st.pend_wait - 1, st.pend_wait)
+    deliver = ~in_transit
+
+    close_m = valid & deliver & st.pend_close & (st.pos != 0)
     exit_buy = st.pos < 0
     close_fill = torch.where(exit_buy, _buy_fill(o_px, slip), _sell_fill(o_px, slip))
     _close_position(st, close_m, close_fill, params)
 
-    open_m = valid & (st.pend_open_dir != 0) & (st.pos == 0)
+    open_m = valid & deliver & (st.pend_open_dir != 0) & (st.pos == 0)
     odir = st.pend_open_dir.to(torch.float32)
     open_fill = torch.where(odir > 0, _buy_fill(o_px, slip), _sell_fill(o_px, slip))
     if bool(open_m.any()):
@@ -192,7 +199,12 @@ def step_torch(
         fill64 = open_fill.to(torch.float64)
         notional = size * fill64
         comm = notional * params.commission
-        margin = notional / params.leverage
+        # standard: init-margin fraction of notional; leveraged: notional
+        # over account leverage (Standard-/LeveragedMarginModel semantics)
+        if params.margin_model == 1:
+            margin = notional * params.margin_init_rate
+        else:
+            margin = notional / params.leverage
         if params.enforce_margin_preflight:
             # margin preflight denial (nautilus_gym.py:128-171): drop the
             # order, count it, keep the episode alive
@@ -211,38 +223,81 @@ def step_torch(
         st.br_armed = st.br_armed | arm
         st.br_sl = torch.where(arm, st.pend_sl, st.br_sl)
         st.br_tp = torch.where(arm, st.pend_tp, st.br_tp)
-    st.pend_close = torch.zeros_like(st.pend_close)
-    st.pend_open_dir = torch.zeros_like(st.pend_open_dir)
-    st.pend_open_size = torch.zeros_like(st.pend_open_size)
-    st.pend_sl = torch.zeros_like(st.pend_sl)
-    st.pend_tp = torch.zeros_like(st.pend_tp)
+    st.pend_close = st.pend_close & in_transit
+    st.pend_open_dir = torch.where(in_transit, st.pend_open_dir,
+                                   torch.zeros_like(st.pend_open_dir))
+    st.pend_open_size = torch.where(in_transit, st.pend_open_size,
+                                    torch.zeros_like(st.pend_open_size))
+    st.pend_sl = torch.where(in_transit, st.pend_sl, torch.zeros_like(st.pend_sl))
+    st.pend_tp = torch.where(in_transit, st.pend_tp, torch.zeros_like(st.pend_tp))
 
-    # ---- 2. bracket children (worst_case policy: stop before limit;
-    # gap-through fills at open) — children active from the bar AFTER the
-    # parent fill ---------------------------------------------------------
+    # ---- 2. bracket children — collision + limit-fill policies ---------
+    # worst_case: stop has absolute priority (open AND adverse extreme
+    # before any TP look); ohlc: open, high, low print order; adaptive:
+    # up bar assumes the low printed first.  The limit leg honors
+    # limit_fill_policy: touch (>=), cross/conservative (strict), and
+    # conservative takes no gap price improvement (fills at the limit).
     chk = valid & st.br_active & ~st.br_armed & (st.pos != 0)
     if bool(chk.any()):
         is_long = st.pos > 0
-        # long: SL below (sell stop), TP above (sell limit)
-        l_sl_gap = chk & is_long & (o_px <= st.br_sl)
-        l_sl_hit = chk & is_long & ~l_sl_gap & (l_px <= st.br_sl)
-        l_tp_gap = chk & is_long & ~l_sl_gap & ~l_sl_hit & (o_px >= st.br_tp)
-        l_tp_hit = (
-            chk & is_long & ~l_sl_gap & ~l_sl_hit & ~l_tp_gap & (h_px >= st.br_tp)
-        )
-        # short: SL above (buy stop), TP below (buy limit)
-        s_sl_gap = chk & ~is_long & (o_px >= st.br_sl)
-        s_sl_hit = chk & ~is_long & ~s_sl_gap & (h_px >= st.br_sl)
-        s_tp_gap = chk & ~is_long & ~s_sl_gap & ~s_sl_hit & (o_px <= st.br_tp)
-        s_tp_hit = (
-            chk & ~is_long & ~s_sl_gap & ~s_sl_hit & ~s_tp_gap & (l_px <= st.br_tp)
-        )
-        sl_m = l_sl_gap | l_sl_hit | s_sl_gap | s_sl_hit
-        tp_m = l_tp_gap | l_tp_hit | s_tp_gap | s_tp_hit
+        lim = params.limit_fill_policy
+
+        def tp_ge(px):  # long take-profit trigger at px
+            return px >= st.br_tp if lim == 0 else px > st.br_tp
+
+        def tp_le(px):  # short take-profit trigger at px
+            return px <= st.br_tp if lim == 0 else px < st.br_tp
+
+        gap_tp_price = st.br_tp if lim == 2 else o_px
+        if params.intrabar_collision_policy == 0:  # worst_case
+            # long: SL below (sell stop), TP above (sell limit)
+            l_sl_gap = chk & is_long & (o_px <= st.br_sl)
+            l_sl_hit = chk & is_long & ~l_sl_gap & (l_px <= st.br_sl)
+            l_tp_gap = chk & is_long & ~l_sl_gap & ~l_sl_hit & tp_ge(o_px)
+            l_tp_hit = (
+                chk & is_long & ~l_sl_gap & ~l_sl_hit & ~l_tp_gap & tp_ge(h_px)
+            )
+            # short: SL above (buy stop), TP below (buy limit)
+            s_sl_gap = chk & ~is_long & (o_px >= st.br_sl)
+            s_sl_hit = chk & ~is_long & ~s_sl_gap & (h_px >= st.br_sl)
+            s_tp_gap = chk & ~is_long & ~s_sl_gap & ~s_sl_hit & tp_le(o_px)
+            s_tp_hit = (
+                chk & ~is_long & ~s_sl_gap & ~s_sl_hit & ~s_tp_gap & tp_le(l_px)
+            )
+            sl_gap = l_sl_gap | s_sl_gap
+            tp_gap = l_tp_gap | s_tp_gap
+            sl_m = sl_gap | l_sl_hit | s_sl_hit
+            tp_m = tp_gap | l_tp_hit | s_tp_hit
+        else:
+            # point-walk: gap checks at the open (stop first), then the
+            # two extremes in policy order (close adds no new trigger)
+            if params.intrabar_collision_policy == 2:  # adaptive
+                low_first = c_px >= o_px  # up bar: dip printed first
+            else:  # ohlc: high before low
+                low_first = torch.zeros_like(chk)
+            sl_gap = chk & torch.where(is_long, o_px <= st.br_sl,
+                                       o_px >= st.br_sl)
+            tp_gap = chk & ~sl_gap & torch.where(is_long, tp_ge(o_px),
+                                                 tp_le(o_px))
+            rest = chk & ~sl_gap & ~tp_gap
+            # per-extreme triggers (each extreme serves one leg per side)
+            low_sl = is_long & (l_px <= st.br_sl)     # long stop at low
+            low_tp = ~is_long & tp_le(l_px)           # short limit at low
+            high_tp = is_long & tp_ge(h_px)           # long limit at high
+            high_sl = ~is_long & (h_px >= st.br_sl)   # short stop at high
+            first_sl = torch.where(low_first, low_sl, high_sl)
+            first_tp = torch.where(low_first, low_tp, high_tp)
+            second_sl = torch.where(low_first, high_sl, low_sl)
+            second_tp = torch.where(low_first, high_tp, low_tp)
+            sl1 = rest & first_sl
+            tp1 = rest & first_tp & ~sl1
+            rest2 = rest & ~sl1 & ~tp1
+            sl_m = sl_gap | sl1 | (rest2 & second_sl)
+            tp_m = tp_gap | tp1 | (rest2 & second_tp & ~(rest2 & second_sl))
         trig = torch.where(
-            l_sl_gap | s_sl_gap | l_tp_gap | s_tp_gap,
-            o_px,
-            torch.where(sl_m, st.br_sl, st.br_tp),
+            sl_gap, o_px,
+            torch.where(tp_gap, gap_tp_price,
+                        torch.where(sl_m, st.br_sl, st.br_tp)),
         )
         # exit side: long exits sell, short exits buy
         fill = torch.where(is_long, _sell_fill(trig, slip), _buy_fill(trig, slip))
@@ -251,6 +306,10 @@ def step_torch(
     st.br_armed = st.br_armed & ~valid  # arm active brackets for next bar
 
     dec = valid | first  # envs making a decision on this bar
+
+    # an order already in transit keeps its remaining latency wait; only a
+    # NEW submission (below) starts the clock (matches env_step.hip)
+    held_order = st.pend_close | (st.pend_open_dir != 0)
 
     # ---- 3. ATR true-range update (direct_atr_sltp.py:143-155) ----------
     if params.strategy_id == STRATEGY_ATR_SLTP:
@@ -294,7 +353,7 @@ def step_torch(
         open_l = open_l & ~(st.pos > 0)
         open_s = open_s & ~(st.pos < 0)
         st.pend_close = st.pend_close | flip_l | flip_s
-        dirv = torch.zeros_like(st.pend_open_dir)
+        dirv = st.pend_open_dir.clone()  # keep in-transit orders (latency)
         dirv = torch.where(open_l, torch.ones_like(dirv), dirv)
         dirv = torch.where(open_s, -torch.ones_like(dirv), dirv)
         st.pend_open_dir = dirv
@@ -314,7 +373,7 @@ def step_torch(
         flip_s = want_short & (st.pos > 0)
         open_s = want_short & (st.pos >= 0)
         st.pend_close = st.pend_close | flip_l | flip_s
-        dirv = torch.zeros_like(st.pend_open_dir)
+        dirv = st.pend_open_dir.clone()  # keep in-transit orders (latency)
         dirv = torch.where(open_l, torch.ones_like(dirv), dirv)
         dirv = torch.where(open_s, -torch.ones_like(dirv), dirv)
         st.pend_open_dir = dirv
@@ -412,7 +471,7 @@ def step_torch(
         flip_s = go_s & (st.pos > 0)
         open_s = go_s & (st.pos >= 0)
         st.pend_close = st.pend_close | flip_l | flip_s
-        dirv = torch.zeros_like(st.pend_open_dir)
+        dirv = st.pend_open_dir.clone()  # keep in-transit orders (latency)
         dirv = torch.where(open_l, torch.ones_like(dirv), dirv)
         dirv = torch.where(open_s, -torch.ones_like(dirv), dirv)
         st.pend_open_dir = dirv
@@ -425,6 +484,14 @@ def step_torch(
             open_l, c_px + tp_dist, torch.where(open_s, c_px - tp_dist, st.pend_tp)
         )
         st.exec_diag[:, _E["entry_orders_submitted"]] += opn.to(torch.int32)
+
+    if params.latency_bars > 0:
+        queued_now = (dec & ~held_order
+                      & (st.pend_close | (st.pend_open_dir != 0)))
+        st.pend_wait = torch.where(
+            queued_now,
+            torch.full_like(st.pend_wait, params.latency_bars),
+            st.pend_wait)
 
     st.started = st.started | first
     st.episode_step = st.episode_step + dec.to(torch.int32)

@@ -80,6 +80,7 @@ class EnvState:
     tr_sum: torch.Tensor          # f32 [N]
     prev_close_atr: torch.Tensor  # f32 [N] (NaN = unset)
     # sharpe reward ring
+    pend_wait: torch.Tensor       # i32 [N] bars the order is in transit
     rew_ring: torch.Tensor        # f32 [N, sharpe_window]
     rew_count: torch.Tensor       # i32 [N]
     rew_s1: torch.Tensor          # f64 [N] sharpe running sum (O(1)/step)
@@ -146,6 +147,7 @@ def alloc_state(params: EnvParams, device: torch.device) -> EnvState:
         pend_open_size=torch.zeros(N, **f32),
         pend_sl=torch.zeros(N, **f32),
         pend_tp=torch.zeros(N, **f32),
+        pend_wait=torch.zeros(N, **i32),
         br_active=torch.zeros(N, **boolk),
         br_armed=torch.zeros(N, **boolk),
         br_sl=torch.zeros(N, **f32),
@@ -205,6 +207,7 @@ def reset_state_(st: EnvState, params: EnvParams, mask: torch.Tensor) -> None:
     st.pend_open_size[mask] = 0.0
     st.pend_sl[mask] = 0.0
     st.pend_tp[mask] = 0.0
+    st.pend_wait[mask] = 0
     st.br_active[mask] = False
     st.br_armed[mask] = False
     st.br_sl[mask] = 0.0

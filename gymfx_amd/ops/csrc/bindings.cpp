@@ -143,6 +143,10 @@ struct GymFxEngine {
     K.atr_period = (int)gi("atr_period");
     K.size_mode = (int)gi("size_mode");
     K.risk_mode = (int)gi("risk_mode");
+    K.collision_policy = (int)gi("collision_policy");
+    K.limit_policy = (int)gi("limit_policy");
+    K.latency_bars = (int)gi("latency_bars");
+    K.margin_model = (int)gi("margin_model");
     K.flags = (int)gi("flags");
     K.obs_dim = (int)gi("obs_dim");
     K.off_features = (int)gi("off_features");
@@ -184,6 +188,7 @@ struct GymFxEngine {
     K.fc_pen_window_hours = gf("fc_pen_window_hours");
     K.feature_clip = gf("feature_clip");
     K.overlay_threshold = gf("overlay_threshold");
+    K.margin_init_rate = gf("margin_init_rate");
 
     // market pointers
     P.open_px = ptr<float>(market, "open");
@@ -221,6 +226,7 @@ struct GymFxEngine {
     P.pend_open_size = ptr<float>(state, "pend_open_size");
     P.pend_sl = ptr<float>(state, "pend_sl");
     P.pend_tp = ptr<float>(state, "pend_tp");
+    P.pend_wait = ptr<int>(state, "pend_wait");
     P.br_active = ptr<bool>(state, "br_active");
     P.br_armed = ptr<bool>(state, "br_armed");
     P.br_sl = ptr<float>(state, "br_sl");

@@ -16,6 +16,10 @@ enum PrepId { PREP_DEFAULT = 0, PREP_FEATURE_WINDOW = 1 };
 enum ScalingMode { SCALE_NONE = 0, SCALE_ROLLING = 1, SCALE_EXPANDING = 2 };
 enum SizeMode { SIZE_FX_UNITS = 0, SIZE_NOTIONAL = 1 };
 enum RiskMode { RISK_FIXED = 0, RISK_RELVOL = 1, RISK_MARGIN = 2 };
+// execution-realism policies (contracts.py execution_cost_profile.v1)
+enum CollisionPolicy { COLL_WORST = 0, COLL_OHLC = 1, COLL_ADAPTIVE = 2 };
+enum LimitPolicy { LIM_TOUCH = 0, LIM_CROSS = 1, LIM_CONSERVATIVE = 2 };
+enum MarginModel { MARGIN_LEVERAGED = 0, MARGIN_STANDARD = 1 };
 
 // flag bits (== ops/wrappers.py FLAG_*)
 enum Flags {
@@ -120,6 +124,7 @@ struct EnvParamsK {
   int reward_id, strategy_id, prep_id;
   int scaling_mode, scale_window, sharpe_window, atr_period;
   int size_mode, risk_mode;
+  int collision_policy, limit_policy, latency_bars, margin_model;
   int flags;
   int obs_dim;
   int off_features, off_prices, off_returns, off_agent, off_fc, off_cal;
@@ -133,6 +138,7 @@ struct EnvParamsK {
   double sl_shrink_alpha, tp_shrink_alpha, min_k_sl, min_rr, mplf;
   double fc_pen_coef, fc_pen_window_hours;
   double feature_clip, overlay_threshold;
+  double margin_init_rate;
   // fused policy sampling (set per step() call when head != null)
   unsigned long long sample_seed;
   long long sample_step;
@@ -159,6 +165,7 @@ struct EnvPtrs {
   bool *pend_close;
   int8_t *pend_open_dir;
   float *pend_open_size, *pend_sl, *pend_tp;
+  int *pend_wait;            // bars the pending order is still in transit
   bool *br_active, *br_armed;
   float *br_sl, *br_tp;
   float *tr_ring;            // [N, P]

@@ -63,6 +63,7 @@ GFX_DEV void reset_env(const EnvPtrs& P, const EnvParamsK& K, int n) {
   P.pend_open_size[n] = 0.f;
   P.pend_sl[n] = 0.f;
   P.pend_tp[n] = 0.f;
+  P.pend_wait[n] = 0;
   P.br_active[n] = false;
   P.br_armed[n] = false;
   P.br_sl[n] = 0.f;
@@ -186,59 +187,114 @@ GFX_DEV void env_step_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
   if (valid || first) P.last_trade_cost[n] = 0.0;
 
   // ---- 1. pending market fills at open(t) ----------------------------
-  if (valid && P.pend_close[n] && P.pos[n] != 0.0) {
-    float fill = (P.pos[n] < 0) ? buy_fill(o_px, slip) : sell_fill(o_px, slip);
-    close_position(P, K, n, (double)fill, -1);
-  }
-  if (valid && P.pend_open_dir[n] != 0 && P.pos[n] == 0.0) {
-    double dir = (double)P.pend_open_dir[n];
-    double fill = (double)(dir > 0 ? buy_fill(o_px, slip) : sell_fill(o_px, slip));
-    double size = (double)P.pend_open_size[n];
-    double notional = size * fill;
-    double comm = notional * K.commission;
-    double margin = notional / K.leverage;
-    if ((K.flags & F_PREFLIGHT) && P.cash[n] < margin + comm) {
-      // margin preflight denial (nautilus_gym.py:128-171 semantics):
-      // the order is dropped, diagnostics incremented, episode continues
-      ediag[E_MARGIN_PREFLIGHT_DENIED] += 1;
-      goto preflight_denied;
+  // latency_ms >= bar duration holds the order in transit for
+  // K.latency_bars extra bars before it reaches the book (LatencyModel
+  // semantics at OHLC granularity, nautilus_adapter.py:415-417)
+  const bool in_transit = valid && P.pend_wait[n] > 0 &&
+                          (P.pend_close[n] || P.pend_open_dir[n] != 0);
+  if (in_transit) {
+    P.pend_wait[n] -= 1;
+  } else {
+    if (valid && P.pend_close[n] && P.pos[n] != 0.0) {
+      float fill = (P.pos[n] < 0) ? buy_fill(o_px, slip) : sell_fill(o_px, slip);
+      close_position(P, K, n, (double)fill, -1);
     }
-    P.cash[n] -= margin + comm;
-    P.margin_used[n] = margin;
-    P.commission_paid[n] += comm;
-    P.last_trade_cost[n] += comm;
-    P.pos[n] = dir * size;
-    P.avg_entry[n] = fill;
-    if (P.pend_sl[n] > 0.f || P.pend_tp[n] > 0.f) {
-      P.br_active[n] = true;
-      P.br_armed[n] = true;  // children active from NEXT bar
-      P.br_sl[n] = P.pend_sl[n];
-      P.br_tp[n] = P.pend_tp[n];
+    if (valid && P.pend_open_dir[n] != 0 && P.pos[n] == 0.0) {
+      double dir = (double)P.pend_open_dir[n];
+      double fill = (double)(dir > 0 ? buy_fill(o_px, slip) : sell_fill(o_px, slip));
+      double size = (double)P.pend_open_size[n];
+      double notional = size * fill;
+      double comm = notional * K.commission;
+      // standard: init-margin fraction of full notional; leveraged:
+      // notional / account leverage (Standard-/LeveragedMarginModel,
+      // nautilus_adapter.py:371-375)
+      double margin = (K.margin_model == MARGIN_STANDARD)
+                          ? notional * K.margin_init_rate
+                          : notional / K.leverage;
+      if ((K.flags & F_PREFLIGHT) && P.cash[n] < margin + comm) {
+        // margin preflight denial (nautilus_gym.py:128-171 semantics):
+        // the order is dropped, diagnostics incremented, episode continues
+        ediag[E_MARGIN_PREFLIGHT_DENIED] += 1;
+        goto preflight_denied;
+      }
+      P.cash[n] -= margin + comm;
+      P.margin_used[n] = margin;
+      P.commission_paid[n] += comm;
+      P.last_trade_cost[n] += comm;
+      P.pos[n] = dir * size;
+      P.avg_entry[n] = fill;
+      if (P.pend_sl[n] > 0.f || P.pend_tp[n] > 0.f) {
+        P.br_active[n] = true;
+        P.br_armed[n] = true;  // children active from NEXT bar
+        P.br_sl[n] = P.pend_sl[n];
+        P.br_tp[n] = P.pend_tp[n];
+      }
+    preflight_denied:;
     }
-  preflight_denied:;
+    P.pend_close[n] = false;
+    P.pend_open_dir[n] = 0;
+    P.pend_open_size[n] = 0.f;
+    P.pend_sl[n] = 0.f;
+    P.pend_tp[n] = 0.f;
   }
-  P.pend_close[n] = false;
-  P.pend_open_dir[n] = 0;
-  P.pend_open_size[n] = 0.f;
-  P.pend_sl[n] = 0.f;
-  P.pend_tp[n] = 0.f;
 
-  // ---- 2. bracket children, worst-case ordering ----------------------
+  // ---- 2. bracket children -------------------------------------------
+  // Collision policy decides the intrabar point order when both children
+  // lie inside one bar (contracts.py intrabar_collision_policy):
+  //   worst_case — the stop has absolute priority (checked on the open
+  //                AND the adverse extreme before any take-profit look);
+  //   ohlc      — open, high, low, close print order;
+  //   adaptive  — up bar assumes the low printed first, down bar the high.
+  // The limit_fill_policy governs the take-profit leg: touch fills at >=,
+  // cross/conservative need a strict cross, and conservative never takes
+  // gap price improvement (fills at the limit price).
   if (valid && P.br_active[n] && !P.br_armed[n] && P.pos[n] != 0.0) {
     const bool is_long = P.pos[n] > 0;
     const float sl = P.br_sl[n], tp = P.br_tp[n];
+    const int lim = K.limit_policy;
     float trig = 0.f;
     int hit = 0;  // 0 none, 1 sl, 2 tp
-    if (is_long) {
-      if (o_px <= sl) { hit = 1; trig = o_px; }
-      else if (l_px <= sl) { hit = 1; trig = sl; }
-      else if (o_px >= tp) { hit = 2; trig = o_px; }
-      else if (h_px >= tp) { hit = 2; trig = tp; }
+    const auto tp_ge = [&](float px) {  // long take-profit trigger
+      return lim == LIM_TOUCH ? px >= tp : px > tp;
+    };
+    const auto tp_le = [&](float px) {  // short take-profit trigger
+      return lim == LIM_TOUCH ? px <= tp : px < tp;
+    };
+    const float tp_gap_trig = (lim == LIM_CONSERVATIVE) ? tp : o_px;
+    if (K.collision_policy == COLL_WORST) {
+      if (is_long) {
+        if (o_px <= sl) { hit = 1; trig = o_px; }
+        else if (l_px <= sl) { hit = 1; trig = sl; }
+        else if (tp_ge(o_px)) { hit = 2; trig = tp_gap_trig; }
+        else if (tp_ge(h_px)) { hit = 2; trig = tp; }
+      } else {
+        if (o_px >= sl) { hit = 1; trig = o_px; }
+        else if (h_px >= sl) { hit = 1; trig = sl; }
+        else if (tp_le(o_px)) { hit = 2; trig = tp_gap_trig; }
+        else if (tp_le(l_px)) { hit = 2; trig = tp; }
+      }
     } else {
-      if (o_px >= sl) { hit = 1; trig = o_px; }
-      else if (h_px >= sl) { hit = 1; trig = sl; }
-      else if (o_px <= tp) { hit = 2; trig = o_px; }
-      else if (l_px <= tp) { hit = 2; trig = tp; }
+      // point-walk: gap checks at the open (stop first), then the two
+      // extremes in policy order; the close is within [low, high] so it
+      // can add no new trigger.
+      const bool low_first = (K.collision_policy == COLL_ADAPTIVE)
+                                 ? (c_px >= o_px)  // up bar: dip printed first
+                                 : false;          // ohlc: high before low
+      if (is_long && o_px <= sl) { hit = 1; trig = o_px; }
+      else if (!is_long && o_px >= sl) { hit = 1; trig = o_px; }
+      else if (is_long && tp_ge(o_px)) { hit = 2; trig = tp_gap_trig; }
+      else if (!is_long && tp_le(o_px)) { hit = 2; trig = tp_gap_trig; }
+      else {
+        for (int k = 0; k < 2 && !hit; ++k) {
+          if ((k == 0) == low_first) {  // at the bar low
+            if (is_long) { if (l_px <= sl) { hit = 1; trig = sl; } }
+            else if (tp_le(l_px)) { hit = 2; trig = tp; }
+          } else {                      // at the bar high
+            if (is_long) { if (tp_ge(h_px)) { hit = 2; trig = tp; } }
+            else if (h_px >= sl) { hit = 1; trig = sl; }
+          }
+        }
+      }
     }
     if (hit) {
       float fill = is_long ? sell_fill(trig, slip) : buy_fill(trig, slip);
@@ -272,6 +328,11 @@ GFX_DEV void env_step_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
   }
 
   // ---- 4. strategy decision on bar t ---------------------------------
+  // `held` = an order already in transit (latency); a NEW decision this
+  // bar starts the latency clock, a held order keeps its remaining wait
+  // (the strategy may refresh the order fields, the arrival time is set
+  // by the first submission).
+  const bool held_order = P.pend_close[n] || P.pend_open_dir[n] != 0;
   if (dec && a == 3) {  // overlay force-flat bypasses the strategy
     if (P.pos[n] != 0.0) {
       P.pend_close[n] = true;
@@ -404,6 +465,10 @@ GFX_DEV void env_step_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
       }
     }
   }
+
+  if (dec && !held_order && K.latency_bars > 0 &&
+      (P.pend_close[n] || P.pend_open_dir[n] != 0))
+    P.pend_wait[n] = K.latency_bars;
 
   if (first) P.started[n] = true;
   if (dec) P.episode_step[n] += 1;

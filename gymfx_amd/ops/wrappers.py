@@ -93,6 +93,10 @@ def pack_params(p: EnvParams, T: int):
         "atr_period": p.atr_period,
         "size_mode": p.size_mode,
         "risk_mode": p.sltp_risk_mode,
+        "collision_policy": p.intrabar_collision_policy,
+        "limit_policy": p.limit_fill_policy,
+        "latency_bars": p.latency_bars,
+        "margin_model": p.margin_model,
         "flags": _flags(p),
         "obs_dim": p.obs_dim,
         "off_features": off("features"),
@@ -136,6 +140,7 @@ def pack_params(p: EnvParams, T: int):
         "fc_pen_window_hours": p.force_close_exposure_penalty_window_hours,
         "feature_clip": p.feature_clip,
         "overlay_threshold": p.event_context_no_trade_threshold,
+        "margin_init_rate": p.margin_init_rate,
     }
     return iparams, fparams
 

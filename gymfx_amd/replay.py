@@ -46,6 +46,12 @@ class ScalarLedger:
     tp_pips: float = 0.0
     pip_size: float = 0.0001
     use_brackets: bool = False
+    # execution-realism tier (ids == envs/params.py policy maps)
+    collision_policy: int = 0     # 0 worst_case, 1 ohlc, 2 adaptive
+    limit_fill_policy: int = 0    # 0 touch, 1 cross, 2 conservative
+    latency_bars: int = 0
+    margin_model: int = 0         # 0 leveraged, 1 standard
+    margin_init_rate: float = 0.03
 
     cash: float = field(init=False)
     pos: float = field(init=False, default=0.0)
@@ -61,6 +67,7 @@ class ScalarLedger:
     pend_dir: int = field(init=False, default=0)
     pend_sl: float = field(init=False, default=0.0)
     pend_tp: float = field(init=False, default=0.0)
+    pend_wait: int = field(init=False, default=0)
     events: List[Dict[str, Any]] = field(init=False, default_factory=list)
 
     def __post_init__(self):
@@ -86,7 +93,9 @@ class ScalarLedger:
     def _open(self, direction: int, size: float, fill: float, bar: int) -> None:
         notional = size * fill
         comm = notional * self.commission
-        self.margin = notional / self.leverage
+        # leveraged: notional / leverage; standard: init-margin fraction
+        self.margin = (notional * self.margin_init_rate
+                       if self.margin_model == 1 else notional / self.leverage)
         self.cash -= self.margin + comm
         self.commission_paid += comm
         self.pos = direction * size
@@ -101,45 +110,74 @@ class ScalarLedger:
             self.br_sl, self.br_tp = self.pend_sl, self.pend_tp
 
     # -- one bar ---------------------------------------------------------
+    def _bracket_trigger(self, o: float, h: float, low: float, c: float):
+        """(trigger_price, kind) for the first bracket child hit this bar,
+        honoring the collision and limit-fill policies (mirrors
+        env_step.hip section 2 and reference_step.py)."""
+        is_long = self.pos > 0
+        sl, tp = self.br_sl, self.br_tp
+        lim = self.limit_fill_policy
+
+        def tp_hit(px: float) -> bool:
+            if is_long:
+                return px >= tp if lim == 0 else px > tp
+            return px <= tp if lim == 0 else px < tp
+
+        tp_gap_px = tp if lim == 2 else o  # conservative: no improvement
+        sl_gap = o <= sl if is_long else o >= sl
+        if self.collision_policy == 0:  # worst_case: stop absolute priority
+            if sl_gap:
+                return o, "bracket_sl_fill"
+            if (low <= sl) if is_long else (h >= sl):
+                return sl, "bracket_sl_fill"
+            if tp_hit(o):
+                return tp_gap_px, "bracket_tp_fill"
+            if tp_hit(h if is_long else low):
+                return tp, "bracket_tp_fill"
+            return None, None
+        # ohlc / adaptive: gap checks at the open, then extremes in order
+        if sl_gap:
+            return o, "bracket_sl_fill"
+        if tp_hit(o):
+            return tp_gap_px, "bracket_tp_fill"
+        low_first = (c >= o) if self.collision_policy == 2 else False
+        for at_low in ((True, False) if low_first else (False, True)):
+            if at_low:
+                if is_long and low <= sl:
+                    return sl, "bracket_sl_fill"
+                if not is_long and tp_hit(low):
+                    return tp, "bracket_tp_fill"
+            else:
+                if is_long and tp_hit(h):
+                    return tp, "bracket_tp_fill"
+                if not is_long and h >= sl:
+                    return sl, "bracket_sl_fill"
+        return None, None
+
     def step(self, bar: int, o: float, h: float, low: float, c: float,
              action: int) -> None:
         slip = self.slippage
         buy = lambda px: px * (1.0 + slip)     # noqa: E731
         sell = lambda px: px * (1.0 - slip)    # noqa: E731
-        # 1. pending market fills at open
-        if self.pend_close and self.pos != 0:
-            fill = buy(o) if self.pos < 0 else sell(o)
-            self._close(fill, bar, "order_filled")
-        if self.pend_dir != 0 and self.pos == 0:
-            fill = buy(o) if self.pend_dir > 0 else sell(o)
-            self._open(self.pend_dir, self.position_size, fill, bar)
-        self.pend_close = False
-        self.pend_dir = 0
-        self.pend_sl = self.pend_tp = 0.0
-        # 2. bracket children (worst-case: stop before limit; gaps at open)
+        # 1. pending market fills at open (latency holds them in transit)
+        if self.pend_wait > 0 and (self.pend_close or self.pend_dir != 0):
+            self.pend_wait -= 1
+        else:
+            if self.pend_close and self.pos != 0:
+                fill = buy(o) if self.pos < 0 else sell(o)
+                self._close(fill, bar, "order_filled")
+            if self.pend_dir != 0 and self.pos == 0:
+                fill = buy(o) if self.pend_dir > 0 else sell(o)
+                self._open(self.pend_dir, self.position_size, fill, bar)
+            self.pend_close = False
+            self.pend_dir = 0
+            self.pend_sl = self.pend_tp = 0.0
+            self.pend_wait = 0
+        # 2. bracket children per collision/limit policy
         if self.br_active and not self.br_armed and self.pos != 0:
-            is_long = self.pos > 0
-            trig = kind = None
-            if is_long:
-                if o <= self.br_sl:
-                    trig, kind = o, "bracket_sl_fill"
-                elif low <= self.br_sl:
-                    trig, kind = self.br_sl, "bracket_sl_fill"
-                elif o >= self.br_tp:
-                    trig, kind = o, "bracket_tp_fill"
-                elif h >= self.br_tp:
-                    trig, kind = self.br_tp, "bracket_tp_fill"
-            else:
-                if o >= self.br_sl:
-                    trig, kind = o, "bracket_sl_fill"
-                elif h >= self.br_sl:
-                    trig, kind = self.br_sl, "bracket_sl_fill"
-                elif o <= self.br_tp:
-                    trig, kind = o, "bracket_tp_fill"
-                elif low <= self.br_tp:
-                    trig, kind = self.br_tp, "bracket_tp_fill"
+            trig, kind = self._bracket_trigger(o, h, low, c)
             if trig is not None:
-                fill = sell(trig) if is_long else buy(trig)
+                fill = sell(trig) if self.pos > 0 else buy(trig)
                 self._close(fill, bar, kind)
         self.br_armed = False
         # 3. strategy decision (direct / fixed-bracket).  Out-of-range
@@ -148,6 +186,7 @@ class ScalarLedger:
         # (force-flat) exists only INTERNALLY via the event overlay.
         if action < 0 or action > 2:
             action = 0
+        held = self.pend_close or self.pend_dir != 0  # in-transit order
         if action in (1, 2):
             self.events.append({"type": "target_requested", "bar": bar,
                                 "action": action})
@@ -161,6 +200,8 @@ class ScalarLedger:
                     tp_d = self.tp_pips * self.pip_size
                     self.pend_sl = c - sl_d if want > 0 else c + sl_d
                     self.pend_tp = c + tp_d if want > 0 else c - tp_d
+            if not held and (self.pend_close or self.pend_dir != 0):
+                self.pend_wait = self.latency_bars  # latency clock starts
 
     def equity(self, close: float) -> float:
         return self.cash + self.margin + self.pos * (close - self.avg_entry)
@@ -186,6 +227,11 @@ class ReplayAdapter:
             position_size=p.position_size, sl_pips=p.sl_pips,
             tp_pips=p.tp_pips, pip_size=p.pip_size,
             use_brackets=(p.strategy_id == 1),
+            collision_policy=p.intrabar_collision_policy,
+            limit_fill_policy=p.limit_fill_policy,
+            latency_bars=p.latency_bars,
+            margin_model=p.margin_model,
+            margin_init_rate=p.margin_init_rate,
         )
         o = market_data.columns["OPEN"]
         h = market_data.columns["HIGH"]

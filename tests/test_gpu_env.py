@@ -64,6 +64,23 @@ CONFIGS = {
                   "rollover_rate_data": [
                       {"LOCATION": "EA19", "TIME": "2024-01", "Value": 5.0},
                       {"LOCATION": "USA", "TIME": "2024-01", "Value": 4.0}]},
+    # execution-realism tier: one GPU parity config per
+    # policy against the same torch oracle
+    "collision_ohlc": {"strategy_plugin": "direct_fixed_sltp",
+                       "sl_pips": 6.0, "tp_pips": 6.0, "autoreset": True,
+                       "intrabar_collision_policy": "ohlc"},
+    "collision_adaptive": {"strategy_plugin": "direct_fixed_sltp",
+                           "sl_pips": 6.0, "tp_pips": 6.0, "autoreset": True,
+                           "intrabar_collision_policy": "adaptive"},
+    "limit_cross": {"strategy_plugin": "direct_fixed_sltp",
+                    "sl_pips": 8.0, "tp_pips": 4.0, "autoreset": True,
+                    "limit_fill_policy": "cross"},
+    "limit_conservative": {"strategy_plugin": "direct_fixed_sltp",
+                           "sl_pips": 8.0, "tp_pips": 4.0, "autoreset": True,
+                           "limit_fill_policy": "conservative"},
+    "latency": {"latency_ms": 120_000.0, "strategy_plugin": "direct_fixed_sltp",
+                "sl_pips": 10.0, "tp_pips": 10.0},
+    "margin_standard": {"margin_model": "standard", "margin_init_rate": 0.04},
 }
 
 
