@@ -91,6 +91,16 @@ except ImportError:
             if seed is not None:
                 self.seed(seed)
 
+        def seed(self, seed=None):
+            # gymnasium semantics: seeding a Dict seeds every subspace
+            # with a derived seed, so sample() is deterministic
+            rng = np.random.default_rng(seed)
+            seeds = [seed]
+            for sub in self.spaces.values():
+                sub_seed = int(rng.integers(0, 2**31 - 1))
+                seeds.extend(sub.seed(sub_seed))
+            return seeds
+
         def sample(self):
             return {k: s.sample() for k, s in self.spaces.items()}
 

@@ -1,47 +1,27 @@
 #!/usr/bin/env python3
 """Gym API compliance check (parity:
 /root/reference/tools/check_gym_compliance.py:49-56 — the reference runs
-gymnasium.utils.env_checker; gymnasium is not in this image, so this is the
-equivalent in-house contract check against gymfx_amd.spaces)."""
+gymnasium.utils.env_checker; gymnasium is not installable here, so this
+runs the PORTED checker assertions in gymfx_amd/gym_check.py, which defer
+to the real gymnasium automatically when it is importable)."""
 import json
 import sys
 from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
 
-import numpy as np
-
 from gymfx_amd import build_environment
 from gymfx_amd.config import DEFAULT_VALUES
+from gymfx_amd.gym_check import check_env as _check_env
 from gymfx_amd.plugins import load_plugin
 
 
 def check_env(env) -> list:
-    errors = []
-    obs, info = env.reset(seed=3)
-    if not isinstance(info, dict):
-        errors.append("reset info is not a dict")
-    if not env.observation_space.contains(obs):
-        errors.append("reset obs not in observation_space")
-    obs2, _ = env.reset(seed=3)
-    for k in obs:
-        if not np.array_equal(obs[k], obs2[k]):
-            errors.append(f"seeded reset not deterministic for key {k}")
-            break
-    for _ in range(20):
-        a = env.action_space.sample()
-        obs, reward, terminated, truncated, info = env.step(a)
-        if not isinstance(reward, float):
-            errors.append(f"reward type {type(reward)} != float")
-        if not isinstance(terminated, bool) or not isinstance(truncated, bool):
-            errors.append("terminated/truncated not bool")
-        if not env.observation_space.contains(obs):
-            errors.append("step obs not in observation_space")
-        if not isinstance(info, dict):
-            errors.append("step info not dict")
-        if errors or terminated:
-            break
-    return errors
+    try:
+        _check_env(env)
+        return []
+    except AssertionError as exc:
+        return [str(exc)]
 
 
 def main():
