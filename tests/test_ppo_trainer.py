@@ -272,3 +272,47 @@ def test_ddp_dead_rank_surfaces_watchdog_error():
     assert outcome.startswith("raised:"), outcome
     # raised promptly (5 s watchdog + slack), not after some giant default
     assert elapsed < 60.0, elapsed
+
+
+# ---------------------------------------------------------------------------
+# optimization mode with policy-driven trials
+# ---------------------------------------------------------------------------
+
+def test_optimize_policy_driver_uses_checkpoint(tmp_path):
+    """driver_mode=policy scores trials with a trained policy's greedy
+    actions instead of scripted random/buy-hold rollouts."""
+    from gymfx_amd.algo.optimize import optimize_from_config
+    from gymfx_amd.utils.checkpoint import save_checkpoint
+
+    t = _make_trainer()
+    t.train_update()
+    ckpt = str(tmp_path / "p.pt")
+    save_checkpoint(t, ckpt)
+    cfg = {
+        "n_envs": 16, "device": "cpu", "window_size": 8,
+        "data_feed_plugin": "synthetic_data_feed", "synthetic_rows": 400,
+        "strategy_plugin": "direct_atr_sltp", "seed": 5,
+        "optimization_trials": 2, "optimization_steps": 24,
+        "driver_mode": "policy", "checkpoint_file": ckpt,
+        "position_size": 1000.0, "quiet_mode": True,
+    }
+    out = optimize_from_config(cfg)
+    assert out["trials"] == 2 and "best_params" in out
+
+
+def test_optimize_train_driver_trains_per_trial():
+    """driver_mode=train runs a short PPO fit per trial and scores the
+    trained policy — optimization can now tune reward/strategy hparams the
+    PPO path cares about."""
+    from gymfx_amd.algo.optimize import optimize_from_config
+
+    cfg = {
+        "n_envs": 16, "device": "cpu", "window_size": 8,
+        "data_feed_plugin": "synthetic_data_feed", "synthetic_rows": 500,
+        "strategy_plugin": "direct_atr_sltp", "seed": 6,
+        "optimization_trials": 2, "optimization_steps": 16,
+        "driver_mode": "train", "optimization_train_updates": 1,
+        "hidden_size": 16, "position_size": 1000.0, "quiet_mode": True,
+    }
+    out = optimize_from_config(cfg)
+    assert out["trials"] == 2 and "best_params" in out
