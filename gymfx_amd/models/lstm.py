@@ -74,9 +74,16 @@ class ActorCriticLSTM:
                 flat[s.sl] = (torch.rand(s.sl.stop - s.sl.start, generator=g) * 2 - 1) * bound
             else:
                 flat[s.sl] = 0.0
-        # forget-gate bias init to 1 (standard LSTM practice)
-        b = flat[self.slices["b"].sl].view(4 * H)
-        b[H:2 * H] = 1.0
+        # Gate layout is INTERLEAVED along the 4H axis: column 4*k + g holds
+        # gate g of hidden unit k (g: 0=i 1=f 2=g~ 3=o).  A 16-column GEMM
+        # output tile then contains 4 COMPLETE hidden units, which is what
+        # lets the recurrent GEMM compute the LSTM cell in its epilogue
+        # (ops/csrc/ppo_kernels.hip lstm_gemm_cell_fwd) and makes every
+        # cell-kernel gate access a contiguous 4-wide vector load instead of
+        # four reads H apart.  Init is layout-invariant (iid uniform) except
+        # the forget-gate bias:
+        b = flat[self.slices["b"].sl].view(H, 4)
+        b[:, 1] = 1.0  # forget gate
         self.params = flat.to(device)
         self.grads = torch.zeros_like(self.params)
         self.m = torch.zeros_like(self.params)

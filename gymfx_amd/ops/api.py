@@ -88,10 +88,12 @@ def lstm_cell_fwd(
         gp = gp + gates_h.to(torch.float32)
     M, H4 = gp.shape
     H = H4 // 4
-    i = torch.sigmoid(gp[:, :H])
-    f = torch.sigmoid(gp[:, H:2 * H])
-    g = torch.tanh(gp[:, 2 * H:3 * H])
-    o = torch.sigmoid(gp[:, 3 * H:])
+    # interleaved gate layout: column 4*k + {0=i,1=f,2=g,3=o} (models/lstm.py)
+    gv = gp.view(M, H, 4)
+    i = torch.sigmoid(gv[..., 0])
+    f = torch.sigmoid(gv[..., 1])
+    g = torch.tanh(gv[..., 2])
+    o = torch.sigmoid(gv[..., 3])
     c = f * c_prev + i * g
     c_new.copy_(c)
     h_new.copy_((o * torch.tanh(c)).to(torch.bfloat16))
@@ -128,10 +130,12 @@ def lstm_cell_bwd(
     mask = None
     if done is not None:
         mask = (~done).to(torch.float32).unsqueeze(1)
-    i = torch.sigmoid(gp[:, :H])
-    f = torch.sigmoid(gp[:, H:2 * H])
-    g = torch.tanh(gp[:, 2 * H:3 * H])
-    o = torch.sigmoid(gp[:, 3 * H:])
+    # interleaved gate layout: column 4*k + {0=i,1=f,2=g,3=o} (models/lstm.py)
+    gv = gp.view(M, H, 4)
+    i = torch.sigmoid(gv[..., 0])
+    f = torch.sigmoid(gv[..., 1])
+    g = torch.tanh(gv[..., 2])
+    o = torch.sigmoid(gv[..., 3])
     tc = torch.tanh(c_new)
     dh = dh_head.clone()
     if dh_next is not None:
@@ -139,12 +143,12 @@ def lstm_cell_bwd(
     dc = dh * o * (1 - tc * tc)
     if dc_next is not None:
         dc = dc + (mask * dc_next if mask is not None else dc_next)
-    d = torch.cat([
+    d = torch.stack([
         dc * g * i * (1 - i),
         dc * c_prev * f * (1 - f),
         dc * i * (1 - g * g),
         dh * tc * o * (1 - o),
-    ], dim=1)
+    ], dim=2).view(M, 4 * H)
     dgates.copy_(d.to(torch.bfloat16))
     dc_prev.copy_(dc * f)
 

@@ -20,7 +20,8 @@ def test_lstm_cell_fwd_matches_torch():
     c_new = torch.empty(M, H)
     h_new = torch.empty(M, H, dtype=torch.bfloat16)
     api.lstm_cell_fwd(gates, None, c_prev, c_new, h_new)
-    i, f, gg, o = gates.chunk(4, dim=1)
+    # interleaved gate layout: column 4*k + {0=i,1=f,2=g,3=o}
+    i, f, gg, o = gates.view(gates.shape[0], -1, 4).unbind(2)
     c_ref = torch.sigmoid(f) * c_prev + torch.sigmoid(i) * torch.tanh(gg)
     h_ref = torch.sigmoid(o) * torch.tanh(c_ref)
     assert torch.allclose(c_new, c_ref, atol=1e-6)
@@ -32,7 +33,8 @@ def test_lstm_cell_bwd_matches_autograd():
     g = torch.Generator().manual_seed(1)
     gates = torch.randn(M, 4 * H, generator=g, requires_grad=True)
     c_prev = torch.randn(M, H, generator=g, requires_grad=True)
-    i, f, gg, o = gates.chunk(4, dim=1)
+    # interleaved gate layout: column 4*k + {0=i,1=f,2=g,3=o}
+    i, f, gg, o = gates.view(gates.shape[0], -1, 4).unbind(2)
     c_new = torch.sigmoid(f) * c_prev + torch.sigmoid(i) * torch.tanh(gg)
     h_new = torch.sigmoid(o) * torch.tanh(c_new)
     dh = torch.randn(M, H, generator=g)
@@ -109,7 +111,8 @@ def test_bptt_matches_autograd_replica():
         gh = (h @ Wh_c)
         gh = gh.to(torch.bfloat16).float() + (gh - gh.detach())  # bf16 fwd
         gates = obs[l].float() @ Wx_c + gh + b
-        i, f, gg, o = gates.chunk(4, dim=1)
+        # interleaved gate layout: column 4*k + {0=i,1=f,2=g,3=o}
+        i, f, gg, o = gates.view(gates.shape[0], -1, 4).unbind(2)
         c = torch.sigmoid(f) * c + torch.sigmoid(i) * torch.tanh(gg)
         h_raw = torch.sigmoid(o) * torch.tanh(c)
         heads.append(h_raw @ Wy_c + by)
