@@ -331,8 +331,11 @@ class PPOTrainer:
         if self.recurrent:
             self._boot_state["h"].copy_(self.rnn_state["h"])
             self._boot_state["c"].copy_(self.rnn_state["c"])
+            # fused=False: the fused step SWAPS the state's h tensor with
+            # acts['h_tmp'] — a once-per-rollout swap would break the fixed
+            # pointer cycle hipGraph replay requires
             head = model.step_forward(self.obs_bf16_step, self._boot_state,
-                                      self.acts_rollout)
+                                      self.acts_rollout, fused=False)
         else:
             head = model.forward(self.obs_bf16_step, self.acts_rollout)
         self.val_buf[T].copy_(head[:, -1])

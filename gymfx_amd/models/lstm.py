@@ -156,17 +156,33 @@ class ActorCriticLSTM:
             "gates": torch.empty(M, 4 * H, dtype=torch.float32, device=dev),
             "gates_h": torch.empty(M, 4 * H, dtype=torch.bfloat16, device=dev),
             "head": torch.empty(M, self.head_dim, dtype=torch.float32, device=dev),
+            # fused-step h output (the fused GEMM+cell kernel must not write
+            # h in place: other workgroups still read it as the A operand)
+            "h_tmp": torch.empty(M, H, dtype=torch.bfloat16, device=dev),
         }
 
     # -- single step (rollout) -------------------------------------------
     def step_forward(self, obs_bf16: torch.Tensor, state: Dict[str, torch.Tensor],
-                     acts: Dict[str, torch.Tensor]) -> torch.Tensor:
-        """One policy step: updates state['h']/state['c'] IN PLACE, returns
-        head [M, A+1] f32."""
+                     acts: Dict[str, torch.Tensor], *,
+                     fused: bool = True) -> torch.Tensor:
+        """One policy step: updates state['h']/state['c'] IN PLACE (the h
+        TENSOR IDENTITY rotates with acts['h_tmp'] on the fused path —
+        callers must read state['h'] through the dict), returns head
+        [M, A+1] f32."""
         gates = acts["gates"]
         api.gemm(obs_bf16, self.wt("Wx"), self.f32("b"), gates, act=0, trans_b=True)
-        api.gemm(state["h"], self.wt("Wh"), None, acts["gates_h"], act=1, trans_b=True)
-        api.lstm_cell_fwd(gates, acts["gates_h"], state["c"], state["c"], state["h"])
+        if fused and "h_tmp" in acts:
+            # recurrent GEMM + cell in one kernel; rollout never reads
+            # gates_h back, so its global write is skipped on the GPU path
+            api.lstm_step_fused(state["h"], self.wt("Wh"), gates,
+                                acts["gates_h"], state["c"], state["c"],
+                                acts["h_tmp"], keep_gates_h=False)
+            state["h"], acts["h_tmp"] = acts["h_tmp"], state["h"]
+        else:
+            api.gemm(state["h"], self.wt("Wh"), None, acts["gates_h"], act=1,
+                     trans_b=True)
+            api.lstm_cell_fwd(gates, acts["gates_h"], state["c"], state["c"],
+                              state["h"])
         api.gemm(state["h"], self.wt("Wy"), self.f32("by"), acts["head"], act=0, trans_b=True)
         return acts["head"]
 
@@ -186,7 +202,10 @@ class ActorCriticLSTM:
             "head": torch.empty(L * M, self.head_dim, dtype=torch.float32, device=dev),
             "dgates": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
             "dh_all": torch.empty(L, M, H, dtype=torch.float32, device=dev),
-            "dh_next": torch.empty(M, H, dtype=torch.float32, device=dev),
+            # ping-pong recurrent-grad buffers (the fused bwd kernel writes
+            # dh for step l-1 while reading this step's dh_next)
+            "dh_na": torch.empty(M, H, dtype=torch.float32, device=dev),
+            "dh_nb": torch.empty(M, H, dtype=torch.float32, device=dev),
             "dc_a": torch.empty(M, H, dtype=torch.float32, device=dev),
             "dc_b": torch.empty(M, H, dtype=torch.float32, device=dev),
         }
@@ -205,15 +224,13 @@ class ActorCriticLSTM:
         buf["h_in"][0].copy_(h0)
         buf["c_in"][0].copy_(c0)
         for l in range(L):
-            g = buf["gates"][l]
-            gh = buf["gates_h"][l]
-            api.gemm(buf["h_in"][l], self.wt("Wh"), None, gh, act=1,
-                     trans_b=True)
-            # the cell also emits the reset-masked state feeding step l+1
-            # (fused masked_state: one fewer launch per sequential step)
-            api.lstm_cell_fwd(g, gh, buf["c_in"][l], buf["c_raw"][l],
-                              buf["h_raw"][l], done_seq[l],
-                              buf["h_in"][l + 1], buf["c_in"][l + 1])
+            # one fused kernel per sequential step: recurrent GEMM + cell +
+            # reset-masked next-step state; gates_h is kept (bwd recompute)
+            api.lstm_step_fused(buf["h_in"][l], self.wt("Wh"),
+                                buf["gates"][l], buf["gates_h"][l],
+                                buf["c_in"][l], buf["c_raw"][l],
+                                buf["h_raw"][l], done_seq[l],
+                                buf["h_in"][l + 1], buf["c_in"][l + 1])
         api.gemm(buf["h_raw"].view(L * M, H), self.wt("Wy"), self.f32("by"),
                  buf["head"], act=0, trans_b=True)
         return buf["head"]
@@ -230,21 +247,22 @@ class ActorCriticLSTM:
                   workspace=(dw_p, db_p), slabs=slabs)
         dh_flat = buf["dh_all"].view(L * M, H)
         api.gemm(dhead, self.w("Wy"), None, dh_flat, act=0, trans_b=True)
-        # backward through time
+        # backward through time: ONE fused kernel per step computes dgates
+        # (cell bwd) AND the recurrent dgrad dh for step l-1
         dc_next: Optional[torch.Tensor] = None
+        dh_next: Optional[torch.Tensor] = None
         dc_bufs = (buf["dc_a"], buf["dc_b"])
+        dh_bufs = (buf["dh_na"], buf["dh_nb"])
         for l in range(L - 1, -1, -1):
-            dh_next = None
-            if l < L - 1:
-                dh_next = buf["dh_next"]
-                api.gemm(buf["dgates"][l + 1], self.w("Wh"), None, dh_next,
-                         act=0, trans_b=True)
             dc_out = dc_bufs[l & 1]
-            api.lstm_cell_bwd(buf["gates"][l], buf["gates_h"][l],
+            dh_out = dh_bufs[l & 1] if l > 0 else None
+            api.lstm_bwd_step(buf["gates"][l], buf["gates_h"][l],
                               buf["c_in"][l], buf["c_raw"][l],
                               buf["dh_all"][l], dh_next, dc_next,
-                              done_seq[l], buf["dgates"][l], dc_out)
+                              done_seq[l], self.w("Wh"), buf["dgates"][l],
+                              dc_out, dh_out)
             dc_next = dc_out
+            dh_next = dh_out
         # weight grads over the whole sequence batch
         dgates_flat = buf["dgates"].view(L * M, 4 * H)
         dw_p, db_p, slabs = self._wg_workspace("Wx", D, 4 * H, True)

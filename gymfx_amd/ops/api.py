@@ -8,9 +8,15 @@ from __future__ import annotations
 
 from typing import Optional, Tuple
 
+import os
+
 import torch
 
 from . import native
+
+# A/B knob for the fused LSTM step kernels (GYMFX_LSTM_FUSED=0 forces the
+# unfused gemm+cell pairs — bitwise-identical, for measurement)
+_LSTM_FUSED = os.environ.get("GYMFX_LSTM_FUSED", "1") != "0"
 
 
 def _use_native(t: torch.Tensor) -> bool:
@@ -151,6 +157,66 @@ def lstm_cell_bwd(
     ], dim=2).view(M, 4 * H)
     dgates.copy_(d.to(torch.bfloat16))
     dc_prev.copy_(dc * f)
+
+
+def lstm_step_fused(
+    h_in: torch.Tensor,
+    wh_t: torch.Tensor,
+    gates_pre: torch.Tensor,
+    gates_h_buf: torch.Tensor,
+    c_prev: torch.Tensor,
+    c_new: torch.Tensor,
+    h_new: torch.Tensor,
+    done: "Optional[torch.Tensor]" = None,
+    h_masked: "Optional[torch.Tensor]" = None,
+    c_masked: "Optional[torch.Tensor]" = None,
+    keep_gates_h: bool = True,
+) -> None:
+    """One recurrent step: gates_h = h_in @ Wh^T then the LSTM cell, fused
+    into a single kernel on GPU (cell in the GEMM epilogue — removes one
+    launch from the sequential BPTT chain).  ``keep_gates_h=False`` (rollout)
+    skips the gates_h global write entirely.  Falls back to the unfused
+    gemm + lstm_cell_fwd pair (bitwise-identical result) on CPU or for
+    shapes without a fused kernel."""
+    if _LSTM_FUSED and _use_native(gates_pre):
+        ok = native.require().lstm_gemm_cell_fwd(
+            h_in, wh_t, gates_pre, gates_h_buf if keep_gates_h else None,
+            c_prev, c_new, h_new, done, h_masked, c_masked)
+        if ok:
+            return
+    gemm(h_in, wh_t, None, gates_h_buf, act=1, trans_b=True)
+    lstm_cell_fwd(gates_pre, gates_h_buf, c_prev, c_new, h_new, done,
+                  h_masked, c_masked)
+
+
+def lstm_bwd_step(
+    gates_pre: torch.Tensor,
+    gates_h: "Optional[torch.Tensor]",
+    c_prev: torch.Tensor,
+    c_new: torch.Tensor,
+    dh_head: torch.Tensor,
+    dh_next: "Optional[torch.Tensor]",
+    dc_next: "Optional[torch.Tensor]",
+    done: "Optional[torch.Tensor]",
+    wh: torch.Tensor,
+    dgates: torch.Tensor,
+    dc_prev: torch.Tensor,
+    dh_prev: "Optional[torch.Tensor]" = None,
+) -> None:
+    """One BPTT backward step: cell backward (dgates, dc_prev) plus the
+    recurrent dgrad ``dh_prev = dgates @ Wh^T`` for step l-1, fused into a
+    single kernel on GPU.  ``wh`` is the [H, 4H] weight (trans_b layout).
+    Falls back to lstm_cell_bwd + gemm (bitwise-identical)."""
+    if _LSTM_FUSED and _use_native(gates_pre):
+        ok = native.require().lstm_bwd_fused(
+            gates_pre, gates_h, c_prev, c_new, dh_head, dh_next, dc_next,
+            done, wh, dgates, dc_prev, dh_prev)
+        if ok:
+            return
+    lstm_cell_bwd(gates_pre, gates_h, c_prev, c_new, dh_head, dh_next,
+                  dc_next, done, dgates, dc_prev)
+    if dh_prev is not None:
+        gemm(dgates, wh, None, dh_prev, act=0, trans_b=True)
 
 
 def mask_reset(h: torch.Tensor, c: torch.Tensor, done: torch.Tensor) -> None:

@@ -33,6 +33,18 @@ void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
                           int64_t M, int H, hipStream_t stream);
+bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
+                               const float* gates_pre, void* gates_h,
+                               const float* c_prev, float* c_new, void* h_new,
+                               const bool* done, void* h_masked,
+                               float* c_masked, int M, int N, int K,
+                               hipStream_t stream);
+bool launch_lstm_bwd_fused(const float* gates_pre, const void* gates_h,
+                           const float* c_prev, const float* c_new,
+                           const float* dh_head, const float* dh_next,
+                           const float* dc_next, const bool* done,
+                           const void* B, void* dgates, float* dc_prev,
+                           float* dh_prev, int M, int H, hipStream_t stream);
 void launch_mask_reset(void* h, float* c, const bool* done, int64_t M, int H,
                        hipStream_t stream);
 void launch_masked_state(const void* h_raw, const float* c_raw,
@@ -525,6 +537,83 @@ void lstm_cell_bwd_op(torch::Tensor gates_pre,
       dgates.data_ptr(), dc_prev.data_ptr<float>(), M, H, cur_stream());
 }
 
+bool lstm_gemm_cell_fwd_op(torch::Tensor A, torch::Tensor B,
+                           torch::Tensor gates_pre,
+                           c10::optional<torch::Tensor> gates_h,
+                           torch::Tensor c_prev, torch::Tensor c_new,
+                           torch::Tensor h_new,
+                           c10::optional<torch::Tensor> done,
+                           c10::optional<torch::Tensor> h_masked,
+                           c10::optional<torch::Tensor> c_masked) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  check_f32(gates_pre, "gates_pre");
+  check_f32(c_prev, "c_prev");
+  check_f32(c_new, "c_new");
+  check_bf16(h_new, "h_new");
+  const int M = (int)A.size(0);
+  const int K = (int)A.size(1);
+  const int N = (int)B.size(0);
+  TORCH_CHECK(B.size(1) == K, "B must be Wh^T [4H, H]");
+  TORCH_CHECK(N == 4 * K, "N must equal 4*H");
+  TORCH_CHECK(gates_pre.size(0) == M && gates_pre.size(1) == N,
+              "gates_pre shape");
+  void* gh_p = nullptr;
+  if (gates_h.has_value()) {
+    check_bf16(*gates_h, "gates_h");
+    gh_p = gates_h->data_ptr();
+  }
+  const bool* done_p = nullptr;
+  void* hm_p = nullptr;
+  float* cm_p = nullptr;
+  if (done.has_value()) {
+    TORCH_CHECK(h_masked.has_value() && c_masked.has_value(),
+                "done needs h_masked and c_masked outputs");
+    done_p = done->data_ptr<bool>();
+    hm_p = h_masked->data_ptr();
+    cm_p = c_masked->data_ptr<float>();
+  }
+  return gymfx::launch_lstm_gemm_cell_fwd(
+      A.data_ptr(), B.data_ptr(), gates_pre.data_ptr<float>(), gh_p,
+      c_prev.data_ptr<float>(), c_new.data_ptr<float>(), h_new.data_ptr(),
+      done_p, hm_p, cm_p, M, N, K, cur_stream());
+}
+
+bool lstm_bwd_fused_op(torch::Tensor gates_pre,
+                       c10::optional<torch::Tensor> gates_h,
+                       torch::Tensor c_prev, torch::Tensor c_new,
+                       torch::Tensor dh_head,
+                       c10::optional<torch::Tensor> dh_next,
+                       c10::optional<torch::Tensor> dc_next,
+                       c10::optional<torch::Tensor> done, torch::Tensor B,
+                       torch::Tensor dgates, torch::Tensor dc_prev,
+                       c10::optional<torch::Tensor> dh_prev) {
+  check_f32(gates_pre, "gates_pre");
+  check_f32(c_prev, "c_prev");
+  check_f32(c_new, "c_new");
+  check_f32(dh_head, "dh_head");
+  check_bf16(B, "B");
+  check_bf16(dgates, "dgates");
+  check_f32(dc_prev, "dc_prev");
+  const int M = (int)c_prev.size(0);
+  const int H = (int)c_prev.size(1);
+  TORCH_CHECK(B.size(0) == H && B.size(1) == 4 * H, "B must be Wh [H, 4H]");
+  const void* gh = nullptr;
+  if (gates_h.has_value()) {
+    check_bf16(*gates_h, "gates_h");
+    gh = gates_h->data_ptr();
+  }
+  const float* dhn = dh_next.has_value() ? dh_next->data_ptr<float>() : nullptr;
+  const float* dcn = dc_next.has_value() ? dc_next->data_ptr<float>() : nullptr;
+  const bool* dn = done.has_value() ? done->data_ptr<bool>() : nullptr;
+  float* dhp = dh_prev.has_value() ? dh_prev->data_ptr<float>() : nullptr;
+  return gymfx::launch_lstm_bwd_fused(
+      gates_pre.data_ptr<float>(), gh, c_prev.data_ptr<float>(),
+      c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
+      B.data_ptr(), dgates.data_ptr(), dc_prev.data_ptr<float>(), dhp, M, H,
+      cur_stream());
+}
+
 void mask_reset_op(torch::Tensor h, torch::Tensor c, torch::Tensor done) {
   check_bf16(h, "h");
   check_f32(c, "c");
@@ -830,6 +919,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("gates_h"), py::arg("c_prev"), py::arg("c_new"),
         py::arg("h_new"), py::arg("done") = py::none(),
         py::arg("h_masked") = py::none(), py::arg("c_masked") = py::none());
+  m.def("lstm_gemm_cell_fwd", &lstm_gemm_cell_fwd_op,
+        "fused h@Wh^T recurrent GEMM + LSTM cell epilogue; returns False "
+        "when the shape has no fused kernel (caller falls back)",
+        py::arg("A"), py::arg("B"), py::arg("gates_pre"), py::arg("gates_h"),
+        py::arg("c_prev"), py::arg("c_new"), py::arg("h_new"),
+        py::arg("done") = py::none(), py::arg("h_masked") = py::none(),
+        py::arg("c_masked") = py::none());
+  m.def("lstm_bwd_fused", &lstm_bwd_fused_op,
+        "fused LSTM cell backward + dgates @ Wh recurrent dgrad; returns "
+        "False when the shape has no fused kernel (caller falls back)",
+        py::arg("gates_pre"), py::arg("gates_h"), py::arg("c_prev"),
+        py::arg("c_new"), py::arg("dh_head"), py::arg("dh_next"),
+        py::arg("dc_next"), py::arg("done"), py::arg("B"), py::arg("dgates"),
+        py::arg("dc_prev"), py::arg("dh_prev") = py::none());
   m.def("lstm_cell_bwd", &lstm_cell_bwd_op, py::arg("gates_pre"),
         py::arg("gates_h"),
         py::arg("c_prev"), py::arg("c_new"), py::arg("dh_head"),

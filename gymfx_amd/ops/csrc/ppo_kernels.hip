@@ -894,13 +894,25 @@ GFX_DEV float fast_sigmoid(float x) {
   return __builtin_amdgcn_rcpf(1.f + __builtin_amdgcn_exp2f(-x * 1.4426950408889634f));
 }
 
+// Gate layout is INTERLEAVED along the 4H axis (models/lstm.py): column
+// 4*k + g holds gate g of hidden unit k (g: 0=i 1=f 2=g~ 3=o), so every
+// per-unit gate access is ONE contiguous vector load (f32x4 / bf16x4)
+// instead of four reads H apart.
+//
 // gates_h (optional, bf16): the recurrent projection h@Wh kept as a
 // SEPARATE bf16 tensor and summed here — the fused alternative (f32
 // accumulate-GEMM into gates_pre) read-modify-writes a 4x bigger buffer
 // and was the dominant BPTT kernel.
-GFX_DEV float gate_pre(const float* gx, const __bf16* gh, int64_t off) {
-  float v = gx[off];
-  if (gh) v += bf2f(gh[off]);
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+
+// 4-wide interleaved gate pre-activations of hidden unit k for row m
+GFX_DEV f32x4 gate_pre4(const float* __restrict__ gates_pre,
+                        const __bf16* __restrict__ gates_h, int64_t idx4) {
+  f32x4 v = *reinterpret_cast<const f32x4*>(&gates_pre[idx4]);
+  if (gates_h) {
+    const bf16x4 gh = *reinterpret_cast<const bf16x4*>(&gates_h[idx4]);
+    for (int j = 0; j < 4; ++j) v[j] += bf2f(gh[j]);
+  }
   return v;
 }
 
@@ -918,14 +930,11 @@ __global__ void lstm_cell_fwd_kernel(
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
     const int64_t m = idx / H;
-    const int h = (int)(idx % H);
-    const int64_t base = m * 4 * H;
-    const __bf16* gh = gates_h ? gates_h + base : nullptr;
-    const float* gx = gates_pre + base;
-    const float i = fast_sigmoid(gate_pre(gx, gh, h));
-    const float f = fast_sigmoid(gate_pre(gx, gh, H + h));
-    const float g = fast_tanh(gate_pre(gx, gh, 2 * H + h));
-    const float o = fast_sigmoid(gate_pre(gx, gh, 3 * H + h));
+    const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
+    const float i = fast_sigmoid(gp[0]);
+    const float f = fast_sigmoid(gp[1]);
+    const float g = fast_tanh(gp[2]);
+    const float o = fast_sigmoid(gp[3]);
     const float c = f * c_prev[idx] + i * g;
     c_new[idx] = c;
     const __bf16 hb = f2bf(o * fast_tanh(c));
@@ -959,27 +968,320 @@ __global__ void lstm_cell_bwd_kernel(
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
     const int64_t m = idx / H;
-    const int h = (int)(idx % H);
     const float mask = (done && done[m]) ? 0.f : 1.f;
-    const int64_t base = m * 4 * H;
-    const float* gx = gates_pre + base;
-    const __bf16* gh = gates_h ? gates_h + base : nullptr;
-    const float i = fast_sigmoid(gate_pre(gx, gh, h));
-    const float f = fast_sigmoid(gate_pre(gx, gh, H + h));
-    const float g = fast_tanh(gate_pre(gx, gh, 2 * H + h));
-    const float o = fast_sigmoid(gate_pre(gx, gh, 3 * H + h));
+    const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
+    const float i = fast_sigmoid(gp[0]);
+    const float f = fast_sigmoid(gp[1]);
+    const float g = fast_tanh(gp[2]);
+    const float o = fast_sigmoid(gp[3]);
     const float c = c_new[idx];
     const float tc = fast_tanh(c);
     float dhv = dh_head[idx];
     if (dh_next) dhv += mask * dh_next[idx];
     float dc = dhv * o * (1.f - tc * tc);
     if (dc_next) dc += mask * dc_next[idx];
-    __bf16* dg = dgates + m * 4 * H;
-    dg[h] = f2bf(dc * g * i * (1.f - i));
-    dg[H + h] = f2bf(dc * c_prev[idx] * f * (1.f - f));
-    dg[2 * H + h] = f2bf(dc * i * (1.f - g * g));
-    dg[3 * H + h] = f2bf(dhv * tc * o * (1.f - o));
+    bf16x4 dg;
+    dg[0] = f2bf(dc * g * i * (1.f - i));
+    dg[1] = f2bf(dc * c_prev[idx] * f * (1.f - f));
+    dg[2] = f2bf(dc * i * (1.f - g * g));
+    dg[3] = f2bf(dhv * tc * o * (1.f - o));
+    *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
     dc_prev[idx] = dc * f;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// FUSED recurrent step: gates_h = h_in @ Wh^T computed by the standard MFMA
+// pipeline (TRANS_B, NFRAG=2, 64x64 tile — same staging/double-buffering as
+// gemm_kernel), with the LSTM CELL as the epilogue.  The interleaved gate
+// layout makes each 64-column output tile hold 16 COMPLETE hidden units, so
+// the cell needs no cross-tile exchange: accumulators go to an LDS tile,
+// one barrier, then each thread does the gate math for 4 (row, unit) pairs.
+// Removes the separate cell launch from the latency-bound sequential chain
+// (profiles/lstm_kernel_stats_r2.txt: cell_fwd 11.3 us x 512/update).
+//
+// Bitwise parity with the unfused pair: the GEMM result is rounded to bf16
+// BEFORE the gate sum (matching the bf16 gates_h tensor of the unfused
+// path) and the MFMA K-chunk order is identical.
+// ---------------------------------------------------------------------------
+template <bool WRITE_GH>
+__global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
+    const __bf16* __restrict__ A,         // h_in [M, K=H]
+    const __bf16* __restrict__ B,         // Wh^T [N=4H, K=H]
+    const float* __restrict__ gates_pre,  // [M, 4H] (x-proj + bias)
+    __bf16* __restrict__ gates_h,         // [M, 4H] out (if WRITE_GH)
+    const float* __restrict__ c_prev,     // [M, H]
+    float* __restrict__ c_new,            // [M, H]
+    __bf16* __restrict__ h_new,           // [M, H]
+    const bool* __restrict__ done,        // [M] or null
+    __bf16* __restrict__ h_masked,        // [M, H] (if done)
+    float* __restrict__ c_masked,         // [M, H] (if done)
+    int M, int N, int K) {
+  constexpr int BM = 64, BK = 32, BN = 64;
+  constexpr int LDA = BK;  // TRANS_B images are unpadded linear (glds)
+  __shared__ __align__(16) __bf16 As[2][BM][LDA];
+  __shared__ __align__(16) __bf16 Bs[2][BN][BK];
+  __shared__ float Cs[BM][BN + 1];  // accum tile for the cell phase
+
+  const int bm = blockIdx.x * BM;
+  const int bn = blockIdx.y * BN;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+
+  f32x4 acc[2][2] = {};
+  const int row_a = lane & 15;
+  const int kseg = lane >> 4;
+  const int a_r = tid >> 2;
+  const int a_c8 = (tid & 3) * 8;
+
+  bf16x8 ra, rb;
+  auto load_tile = [&](int k0) {
+    ra = bf16x8{};
+    const int gr = bm + a_r;
+    if (gr < M) {
+      const int gk = k0 + a_c8;
+      if (gk + 8 <= K) {
+        ra = *reinterpret_cast<const bf16x8*>(&A[(int64_t)gr * K + gk]);
+      } else {
+        for (int i = 0; i < 8; ++i)
+          ra[i] = (gk + i < K) ? A[(int64_t)gr * K + gk + i] : (__bf16)0.f;
+      }
+    }
+    rb = bf16x8{};
+    const int gn = bn + a_r;  // BN == BM: same staging coordinates
+    if (gn < N) {
+      const int gk = k0 + a_c8;
+      if (gk + 8 <= K) {
+        rb = *reinterpret_cast<const bf16x8*>(&B[(int64_t)gn * K + gk]);
+      } else {
+        for (int i = 0; i < 8; ++i)
+          rb[i] = (gk + i < K) ? B[(int64_t)gn * K + gk + i] : (__bf16)0.f;
+      }
+    }
+  };
+  auto store_tile = [&](int buf) {
+    *reinterpret_cast<bf16x8*>(&As[buf][a_r][a_c8]) = ra;
+    *reinterpret_cast<bf16x8*>(&Bs[buf][a_r][a_c8]) = rb;
+  };
+  const bool can_glds = (bm + BM <= M) && (bn + BN <= N);
+  auto glds_tile = [&](int buf, int k0) {
+    {
+      const int e0 = wave * 512;
+      const int row = (e0 + lane * 8) / BK;
+      const int col = (e0 + lane * 8) % BK;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)(
+              &A[(int64_t)(bm + row) * K + k0 + col]),
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              &As[buf][0][0] + e0),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)(
+              &B[(int64_t)(bn + row) * K + k0 + col]),
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              &Bs[buf][0][0] + e0),
+          16, 0, 0);
+    }
+  };
+  auto stage = [&](int buf, int k0) {
+    if (can_glds && k0 + BK <= K) {
+      glds_tile(buf, k0);
+    } else {
+      load_tile(k0);
+      store_tile(buf);
+    }
+  };
+
+  const int ktiles = (K + BK - 1) / BK;
+  stage(0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < ktiles; ++kt) {
+    const int cur = kt & 1;
+    const bool nxt_glds = kt + 1 < ktiles && can_glds && (kt + 1) * BK + BK <= K;
+    if (nxt_glds) glds_tile(1 - cur, (kt + 1) * BK);
+    else if (kt + 1 < ktiles) load_tile((kt + 1) * BK);
+
+    bf16x8 af[2], bf[2];
+    for (int mi = 0; mi < 2; ++mi)
+      af[mi] = *reinterpret_cast<const bf16x8*>(
+          &As[cur][wr * 32 + mi * 16 + row_a][kseg * 8]);
+    for (int ni = 0; ni < 2; ++ni)
+      bf[ni] = *reinterpret_cast<const bf16x8*>(
+          &Bs[cur][wc * 32 + ni * 16 + row_a][kseg * 8]);
+    for (int mi = 0; mi < 2; ++mi)
+      for (int ni = 0; ni < 2; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+
+    if (kt + 1 < ktiles && !nxt_glds) store_tile(1 - cur);
+    __syncthreads();
+  }
+
+  // accumulators -> LDS tile (bf16-round to match the unfused gates_h)
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+  for (int mi = 0; mi < 2; ++mi)
+    for (int ni = 0; ni < 2; ++ni)
+      for (int r = 0; r < 4; ++r)
+        Cs[wr * 32 + mi * 16 + crow_base + r][wc * 32 + ni * 16 + ccol] =
+            bf2f(f2bf(acc[mi][ni][r]));
+  __syncthreads();
+
+  // ---- cell phase: 16 units x 64 rows; thread t -> unit t%16, rows
+  // (t/16)*4 .. +3 ------------------------------------------------------
+  const int ul = tid & 15;            // unit within the tile
+  const int r0 = (tid >> 4) * 4;      // first local row
+  const int H = N / 4;
+  const int ug = bn / 4 + ul;         // global hidden unit
+  for (int r = 0; r < 4; ++r) {
+    const int grow = bm + r0 + r;
+    if (grow >= M) continue;
+    const int64_t idx = (int64_t)grow * H + ug;
+    f32x4 gh;
+    for (int j = 0; j < 4; ++j) gh[j] = Cs[r0 + r][ul * 4 + j];
+    if (WRITE_GH) {
+      bf16x4 ghb;
+      for (int j = 0; j < 4; ++j) ghb[j] = f2bf(gh[j]);
+      *reinterpret_cast<bf16x4*>(&gates_h[(int64_t)grow * N + (int64_t)(bn + ul * 4)]) = ghb;
+    }
+    const f32x4 gx = *reinterpret_cast<const f32x4*>(
+        &gates_pre[(int64_t)grow * N + (int64_t)(bn + ul * 4)]);
+    const float i = fast_sigmoid(gx[0] + gh[0]);
+    const float f = fast_sigmoid(gx[1] + gh[1]);
+    const float g = fast_tanh(gx[2] + gh[2]);
+    const float o = fast_sigmoid(gx[3] + gh[3]);
+    const float c = f * c_prev[idx] + i * g;
+    c_new[idx] = c;
+    const __bf16 hb = f2bf(o * fast_tanh(c));
+    h_new[idx] = hb;
+    if (done) {
+      const bool d = done[grow];
+      h_masked[idx] = d ? (__bf16)0.f : hb;
+      c_masked[idx] = d ? 0.f : c;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// FUSED BPTT backward step: the cell backward (dgates from saved gates /
+// cell state / incoming grads) and the recurrent dgrad GEMM
+// dh_prev[M,H] = dgates[M,4H] @ Wh^T merged into ONE kernel — the two were
+// 15.3 us + 14.2 us per sequential step (profiles/lstm_kernel_stats_r2.txt).
+// One workgroup owns a 16-row slab across the FULL 4H gate width: phase 1
+// computes dgates elementwise into LDS (and to global, for the wgrads),
+// phase 2 runs the MFMA K-loop with A already resident in LDS.
+// NT = H/64 = 16-column output tiles per wave (H in {64,128,192,256}).
+// Bitwise parity: dgates are bf16-rounded before the MFMA (matching the
+// unfused global tensor) and the K-chunk order equals gemm_kernel's.
+// ---------------------------------------------------------------------------
+template <int NT>
+__global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
+    const float* __restrict__ gates_pre,  // [M, 4H]
+    const __bf16* __restrict__ gates_h,   // [M, 4H] or null
+    const float* __restrict__ c_prev,     // [M, H]
+    const float* __restrict__ c_new,      // [M, H]
+    const float* __restrict__ dh_head,    // [M, H]
+    const float* __restrict__ dh_next,    // [M, H] or null
+    const float* __restrict__ dc_next,    // [M, H] or null
+    const bool* __restrict__ done,        // [M] or null
+    const __bf16* __restrict__ B,         // Wh [N=H, K=4H] (trans_b layout)
+    __bf16* __restrict__ dgates,          // [M, 4H] out
+    float* __restrict__ dc_prev,          // [M, H] out
+    float* __restrict__ dh_prev,          // [M, H] out, or null (step 0)
+    int M, int H) {
+  constexpr int BM = 16, BK = 32;
+  const int N = H;
+  const int K = 4 * H;
+  // LDS: As [BM][4H] bf16 (dgates slab), Bs [2][N][BK] bf16
+  extern __shared__ __align__(16) __bf16 lds[];
+  __bf16* As = lds;                       // BM * K
+  __bf16* Bs = lds + BM * K;              // 2 * N * BK
+
+  const int bm = blockIdx.x * BM;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+
+  // ---- phase 1: cell backward into LDS + global ----------------------
+  for (int p = tid; p < BM * H; p += 256) {
+    const int rl = p / H;
+    const int u = p % H;
+    const int grow = bm + rl;
+    bf16x4 dg = {};
+    float dcp = 0.f;
+    if (grow < M) {
+      const int64_t idx = (int64_t)grow * H + u;
+      const float mask = (done && done[grow]) ? 0.f : 1.f;
+      const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
+      const float i = fast_sigmoid(gp[0]);
+      const float f = fast_sigmoid(gp[1]);
+      const float g = fast_tanh(gp[2]);
+      const float o = fast_sigmoid(gp[3]);
+      const float c = c_new[idx];
+      const float tc = fast_tanh(c);
+      float dhv = dh_head[idx];
+      if (dh_next) dhv += mask * dh_next[idx];
+      float dc = dhv * o * (1.f - tc * tc);
+      if (dc_next) dc += mask * dc_next[idx];
+      dg[0] = f2bf(dc * g * i * (1.f - i));
+      dg[1] = f2bf(dc * c_prev[idx] * f * (1.f - f));
+      dg[2] = f2bf(dc * i * (1.f - g * g));
+      dg[3] = f2bf(dhv * tc * o * (1.f - o));
+      dcp = dc * f;
+      *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
+      dc_prev[idx] = dcp;
+    }
+    *reinterpret_cast<bf16x4*>(&As[rl * K + u * 4]) = dg;
+  }
+  if (dh_prev == nullptr) return;  // step 0: no recurrent grad consumer
+
+  // ---- phase 2: dh_prev = dgates @ Wh^T (A in LDS) -------------------
+  auto stage_b = [&](int buf, int k0) {
+    // B rows [0, N), k chunk [k0, k0+32): one row per thread (N <= 256)
+    for (int r = tid; r < N; r += 256) {
+      for (int j = 0; j < 4; ++j) {
+        *reinterpret_cast<bf16x8*>(&Bs[(buf * N + r) * BK + j * 8]) =
+            *reinterpret_cast<const bf16x8*>(&B[(int64_t)r * K + k0 + j * 8]);
+      }
+    }
+  };
+  f32x4 acc[NT] = {};
+  const int row_a = lane & 15;
+  const int kseg = lane >> 4;
+  const int ktiles = K / BK;  // 4H % 32 == 0 for H % 8 == 0
+  stage_b(0, 0);
+  __syncthreads();
+  for (int kt = 0; kt < ktiles; ++kt) {
+    const int cur = kt & 1;
+    bf16x8 af = *reinterpret_cast<const bf16x8*>(
+        &As[row_a * K + kt * BK + kseg * 8]);
+    bf16x8 bf[NT];
+    for (int ni = 0; ni < NT; ++ni) {
+      const int col = wave * (NT * 16) + ni * 16 + row_a;
+      bf[ni] = *reinterpret_cast<const bf16x8*>(
+          &Bs[(cur * N + col) * BK + kseg * 8]);
+    }
+    for (int ni = 0; ni < NT; ++ni)
+      acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf[ni], acc[ni],
+                                                        0, 0, 0);
+    if (kt + 1 < ktiles) {
+      __syncthreads();
+      stage_b(1 - cur, (kt + 1) * BK);
+      __syncthreads();
+    }
+  }
+  const int crow = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+  for (int ni = 0; ni < NT; ++ni) {
+    const int gcol = wave * (NT * 16) + ni * 16 + ccol;
+    for (int r = 0; r < 4; ++r) {
+      const int grow = bm + crow + r;
+      if (grow < M && gcol < N)
+        dh_prev[(int64_t)grow * N + gcol] = acc[ni][r];
+    }
   }
 }
 
@@ -1397,6 +1699,64 @@ void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
                      gates_pre, reinterpret_cast<const __bf16*>(gates_h),
                      c_prev, c_new, dh_head, dh_next, dc_next, done,
                      reinterpret_cast<__bf16*>(dgates), dc_prev, M, H);
+}
+
+bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
+                               const float* gates_pre, void* gates_h,
+                               const float* c_prev, float* c_new, void* h_new,
+                               const bool* done, void* h_masked,
+                               float* c_masked, int M, int N, int K,
+                               hipStream_t stream) {
+  if (N % 64 != 0) return false;  // 64-col tiles must hold whole units
+  dim3 grid(ceil_div(M, 64), N / 64);
+  if (gates_h) {
+    hipLaunchKernelGGL((lstm_gemm_cell_fwd_kernel<true>), grid, dim3(256), 0,
+                       stream, reinterpret_cast<const __bf16*>(A),
+                       reinterpret_cast<const __bf16*>(B), gates_pre,
+                       reinterpret_cast<__bf16*>(gates_h), c_prev, c_new,
+                       reinterpret_cast<__bf16*>(h_new), done,
+                       reinterpret_cast<__bf16*>(h_masked), c_masked, M, N, K);
+  } else {
+    hipLaunchKernelGGL((lstm_gemm_cell_fwd_kernel<false>), grid, dim3(256), 0,
+                       stream, reinterpret_cast<const __bf16*>(A),
+                       reinterpret_cast<const __bf16*>(B), gates_pre, nullptr,
+                       c_prev, c_new, reinterpret_cast<__bf16*>(h_new), done,
+                       reinterpret_cast<__bf16*>(h_masked), c_masked, M, N, K);
+  }
+  return true;
+}
+
+bool launch_lstm_bwd_fused(const float* gates_pre, const void* gates_h,
+                           const float* c_prev, const float* c_new,
+                           const float* dh_head, const float* dh_next,
+                           const float* dc_next, const bool* done,
+                           const void* B, void* dgates, float* dc_prev,
+                           float* dh_prev, int M, int H,
+                           hipStream_t stream) {
+  if (H % 64 != 0 || H > 256) return false;
+  dim3 grid(ceil_div(M, 16));
+  const size_t lds = (size_t)16 * 4 * H * sizeof(__bf16) +
+                     (size_t)2 * H * 32 * sizeof(__bf16);
+  const __bf16* Bw = reinterpret_cast<const __bf16*>(B);
+  __bf16* dg = reinterpret_cast<__bf16*>(dgates);
+  const __bf16* gh = reinterpret_cast<const __bf16*>(gates_h);
+  switch (H / 64) {
+#define GYMFX_LSTM_BWD_CASE(NT)                                               \
+  case NT:                                                                    \
+    hipLaunchKernelGGL((lstm_bwd_fused_kernel<NT>), grid, dim3(256), lds,     \
+                       stream, gates_pre, gh, c_prev, c_new, dh_head,         \
+                       dh_next, dc_next, done, Bw, dg, dc_prev, dh_prev, M,   \
+                       H);                                                    \
+    break;
+    GYMFX_LSTM_BWD_CASE(1)
+    GYMFX_LSTM_BWD_CASE(2)
+    GYMFX_LSTM_BWD_CASE(3)
+    GYMFX_LSTM_BWD_CASE(4)
+#undef GYMFX_LSTM_BWD_CASE
+    default:
+      return false;
+  }
+  return true;
 }
 
 void launch_mask_reset(void* h, float* c, const bool* done, int64_t M, int H,
