@@ -1239,23 +1239,36 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
   if (dh_prev == nullptr) return;  // step 0: no recurrent grad consumer
 
   // ---- phase 2: dh_prev = dgates @ Wh^T (A in LDS) -------------------
-  auto stage_b = [&](int buf, int k0) {
-    // B rows [0, N), k chunk [k0, k0+32): one row per thread (N <= 256)
-    for (int r = tid; r < N; r += 256) {
-      for (int j = 0; j < 4; ++j) {
-        *reinterpret_cast<bf16x8*>(&Bs[(buf * N + r) * BK + j * 8]) =
-            *reinterpret_cast<const bf16x8*>(&B[(int64_t)r * K + k0 + j * 8]);
-      }
+  // B staged via global_load_lds double-buffering, exactly the gemm_kernel
+  // recipe: the glds for chunk k+1 is in flight across chunk k's MFMAs and
+  // the barrier's vmcnt(0) drains it — ONE barrier per chunk.  (The first
+  // version staged through registers with two barriers per chunk and
+  // serialized ~32 L2 round-trips; measured SLOWER than the unfused pair.)
+  auto glds_b = [&](int buf, int k0) {
+    // B chunk [N][BK] = N*BK elements, 8 bf16 (16 B) per lane-instr
+    const int total8 = N * BK / 8;          // bf16x8 transfers
+    const int per_wave = total8 / 4;        // lanes*instrs per wave
+    for (int j = 0; j < per_wave / 64; ++j) {
+      const int e0 = (wave * (per_wave / 64) + j) * 512;
+      const int row = (e0 + lane * 8) / BK;
+      const int col = (e0 + lane * 8) % BK;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(uintptr_t)(
+              &B[(int64_t)row * K + k0 + col]),
+          (__attribute__((address_space(3))) void*)(uintptr_t)(
+              &Bs[(int64_t)buf * N * BK] + e0),
+          16, 0, 0);
     }
   };
   f32x4 acc[NT] = {};
   const int row_a = lane & 15;
   const int kseg = lane >> 4;
   const int ktiles = K / BK;  // 4H % 32 == 0 for H % 8 == 0
-  stage_b(0, 0);
+  glds_b(0, 0);
   __syncthreads();
   for (int kt = 0; kt < ktiles; ++kt) {
     const int cur = kt & 1;
+    if (kt + 1 < ktiles) glds_b(1 - cur, (kt + 1) * BK);
     bf16x8 af = *reinterpret_cast<const bf16x8*>(
         &As[row_a * K + kt * BK + kseg * 8]);
     bf16x8 bf[NT];
@@ -1267,11 +1280,7 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
     for (int ni = 0; ni < NT; ++ni)
       acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf[ni], acc[ni],
                                                         0, 0, 0);
-    if (kt + 1 < ktiles) {
-      __syncthreads();
-      stage_b(1 - cur, (kt + 1) * BK);
-      __syncthreads();
-    }
+    __syncthreads();
   }
   const int crow = (lane >> 4) * 4;
   const int ccol = lane & 15;
