@@ -17,6 +17,11 @@ from . import native
 # A/B knob for the fused LSTM step kernels (GYMFX_LSTM_FUSED=0 forces the
 # unfused gemm+cell pairs — bitwise-identical, for measurement)
 _LSTM_FUSED = os.environ.get("GYMFX_LSTM_FUSED", "1") != "0"
+# The fused BACKWARD step measured SLOWER than the unfused pair at the
+# flagship shape (row-slab reads 4x the Wh traffic of the 64x64-tiled
+# dgrad GEMM and runs 1 workgroup/CU): 54.3 vs 42.9 ms/update — kept
+# behind an opt-in knob with its bitwise tests (profiles/PERF_NOTES.md).
+_LSTM_BWD_FUSED = os.environ.get("GYMFX_LSTM_BWD_FUSED", "0") == "1"
 
 
 def _use_native(t: torch.Tensor) -> bool:
@@ -207,7 +212,7 @@ def lstm_bwd_step(
     recurrent dgrad ``dh_prev = dgates @ Wh^T`` for step l-1, fused into a
     single kernel on GPU.  ``wh`` is the [H, 4H] weight (trans_b layout).
     Falls back to lstm_cell_bwd + gemm (bitwise-identical)."""
-    if _LSTM_FUSED and _use_native(gates_pre):
+    if _LSTM_BWD_FUSED and _use_native(gates_pre):
         ok = native.require().lstm_bwd_fused(
             gates_pre, gates_h, c_prev, c_new, dh_head, dh_next, dc_next,
             done, wh, dgates, dc_prev, dh_prev)

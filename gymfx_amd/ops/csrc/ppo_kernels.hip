@@ -64,20 +64,24 @@ GFX_DEV float fast_tanh(float x) {
 // a higher occupancy target and demotes the 64 MFMA accumulators of the
 // NFRAG=8 variant to scratch — 1.2 GB of per-call scratch traffic, 4x
 // slower than just running 2 waves/SIMD with accs in AGPRs.
+// BKT: K-chunk depth of the LDS staging pipeline.  32 is the default; 64
+// halves the barrier count for long-K tall shapes (the dgrad/fwd GEMMs of
+// the update phase are latency-limited at 32 — ROADMAP lever 2), reachable
+// via GYMFX_GEMM_BK64.  The MFMA k order is unchanged (two sequential
+// 16x16x32 steps per 64-chunk), so results stay bitwise identical.
 template <bool TRANS_B, int ACT, bool DACT_TANH, bool ADD_BIAS, int NFRAG,
-          bool ACCUM = false>
+          bool ACCUM = false, int BKT = 32>
 __global__ __launch_bounds__(256, 2) void gemm_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     const float* __restrict__ bias, void* __restrict__ C,
     const __bf16* __restrict__ Yact,  // activation output (DACT_TANH)
     int M, int N, int K) {
-  constexpr int BM = 64, BK = 32;
+  constexpr int BM = 64, BK = BKT;
   constexpr int BN = 32 * NFRAG;
   // TRANS_B images are UNPADDED linear (required by the glds staging; the
   // b128 fragment-read bank multiplicity is already at its minimum for both
   // strides).  !TRANS_B keeps the +8 pad with register staging.
   constexpr int LDA = TRANS_B ? BK : (BK + 8);
-  constexpr int BVEC = NFRAG / 2;  // bf16x8 loads per thread for the B tile
   // B tile layout follows the GLOBAL layout so staging loads are always
   // contiguous bf16x8 (a strided 2-byte gather of B was 6x slower than the
   // whole GEMM): TRANS_B stages [n][k], !TRANS_B stages [k][n]; the
@@ -100,34 +104,37 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
   const int row_a = lane & 15;
   const int kseg = lane >> 4;  // 0..3 -> k-base = kseg*8
 
-  // staging coordinates
-  const int a_r = tid >> 2;            // 0..63
-  const int a_c8 = (tid & 3) * 8;      // 0,8,16,24
-  const int b_r = tid >> 2;            // base n row (strided by 64 for NFRAG>2)
-  const int b_c8 = (tid & 3) * 8;
+  // staging: element-offset addressed so any BKT works; for BKT=32 the
+  // (row, col) decomposition reduces to the classic tid>>2 / (tid&3)*8
+  constexpr int ALPT = (BM * BK) / 2048;  // bf16x8 loads per thread, A tile
+  constexpr int BLPT = (BN * BK) / 2048;  // bf16x8 loads per thread, B tile
 
-  bf16x8 ra, rb[BVEC];
+  bf16x8 ra[ALPT], rb[BLPT];
 
   auto load_tile = [&](int k0) {
-    // A tile [64 m][32 k]
-    ra = bf16x8{};
-    const int gr = bm + a_r;
-    if (gr < M) {
-      const int gk = k0 + a_c8;
-      if (gk + 8 <= K) {
-        ra = *reinterpret_cast<const bf16x8*>(&A[(int64_t)gr * K + gk]);
-      } else {
-        for (int i = 0; i < 8; ++i)
-          ra[i] = (gk + i < K) ? A[(int64_t)gr * K + gk + i] : (__bf16)0.f;
+    // A tile [BM m][BK k]
+    for (int j = 0; j < ALPT; ++j) {
+      ra[j] = bf16x8{};
+      const int e = (tid + j * 256) * 8;
+      const int gr = bm + e / BK;
+      if (gr < M) {
+        const int gk = k0 + e % BK;
+        if (gk + 8 <= K) {
+          ra[j] = *reinterpret_cast<const bf16x8*>(&A[(int64_t)gr * K + gk]);
+        } else {
+          for (int i = 0; i < 8; ++i)
+            ra[j][i] = (gk + i < K) ? A[(int64_t)gr * K + gk + i] : (__bf16)0.f;
+        }
       }
     }
     if (TRANS_B) {
       // stage [n][k]: contiguous along k in global [N, K]
-      for (int j = 0; j < BVEC; ++j) {
+      for (int j = 0; j < BLPT; ++j) {
         rb[j] = bf16x8{};
-        const int gn = bn + b_r + j * 64;
+        const int e = (tid + j * 256) * 8;
+        const int gn = bn + e / BK;
         if (gn < N) {
-          const int gk = k0 + b_c8;
+          const int gk = k0 + e % BK;
           if (gk + 8 <= K) {
             rb[j] = *reinterpret_cast<const bf16x8*>(&B[(int64_t)gn * K + gk]);
           } else {
@@ -138,7 +145,7 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
       }
     } else {
       // stage [k][n]: contiguous along n in global [K, N]
-      for (int j = 0; j < BVEC; ++j) {
+      for (int j = 0; j < BLPT; ++j) {
         rb[j] = bf16x8{};
         const int c = tid + j * 256;             // chunk over [BK][BN/8]
         const int kk = c / (BN / 8);
@@ -157,12 +164,17 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
     }
   };
   auto store_tile = [&](int buf) {
-    *reinterpret_cast<bf16x8*>(&As[buf][a_r][a_c8]) = ra;
+    for (int j = 0; j < ALPT; ++j) {
+      const int e = (tid + j * 256) * 8;
+      *reinterpret_cast<bf16x8*>(&As[buf][e / BK][e % BK]) = ra[j];
+    }
     if (TRANS_B) {
-      for (int j = 0; j < BVEC; ++j)
-        *reinterpret_cast<bf16x8*>(&Bs[buf][b_r + j * 64][b_c8]) = rb[j];
+      for (int j = 0; j < BLPT; ++j) {
+        const int e = (tid + j * 256) * 8;
+        *reinterpret_cast<bf16x8*>(&Bs[buf][e / BK][e % BK]) = rb[j];
+      }
     } else {
-      for (int j = 0; j < BVEC; ++j) {
+      for (int j = 0; j < BLPT; ++j) {
         const int c = tid + j * 256;
         *reinterpret_cast<bf16x8*>(
             &Bs[buf][c / (BN / 8)][(c % (BN / 8)) * 8]) = rb[j];
@@ -176,8 +188,8 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
   // across the MFMA section and drained by the barrier's vmcnt(0).
   const bool can_glds = TRANS_B && (bm + BM <= M) && (bn + BN <= N);
   auto glds_tile = [&](int buf, int k0) {
-    {  // A tile [BM][BK]: BM*BK*2 B = BM*BK/512 wave-instr
-      const int e0 = wave * 512;  // BM=64: 1 instr/wave
+    for (int j = 0; j < ALPT; ++j) {  // A tile [BM][BK]
+      const int e0 = (wave * ALPT + j) * 512;
       const int row = (e0 + lane * 8) / BK;
       const int col = (e0 + lane * 8) % BK;
       __builtin_amdgcn_global_load_lds(
@@ -187,8 +199,8 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
               &As[buf][0][0] + e0),
           16, 0, 0);
     }
-    for (int j = 0; j < BVEC; ++j) {  // B tile [BN][BK]
-      const int e0 = (wave * BVEC + j) * 512;
+    for (int j = 0; j < BLPT; ++j) {  // B tile [BN][BK]
+      const int e0 = (wave * BLPT + j) * 512;
       const int row = (e0 + lane * 8) / BK;
       const int col = (e0 + lane * 8) % BK;
       __builtin_amdgcn_global_load_lds(
@@ -218,25 +230,27 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
     if (nxt_glds) glds_tile(1 - cur, (kt + 1) * BK);
     else if (kt + 1 < ktiles) load_tile((kt + 1) * BK);
 
-    bf16x8 af[2], bf[NFRAG];
-    for (int mi = 0; mi < 2; ++mi)
-      af[mi] = *reinterpret_cast<const bf16x8*>(
-          &As[cur][wr * 32 + mi * 16 + row_a][kseg * 8]);
-    if (TRANS_B) {
-      for (int ni = 0; ni < NFRAG; ++ni)
-        bf[ni] = *reinterpret_cast<const bf16x8*>(
-            &Bs[cur][wc * (16 * NFRAG) + ni * 16 + row_a][kseg * 8]);
-    } else {
-      // transpose at the LDS read: Bs holds [k][n]
-      for (int ni = 0; ni < NFRAG; ++ni)
-        for (int i = 0; i < 8; ++i)
-          bf[ni][i] =
-              Bs[cur][kseg * 8 + i][wc * (16 * NFRAG) + ni * 16 + row_a];
+    for (int ks = 0; ks < BK / 32; ++ks) {  // sequential k order: bitwise
+      bf16x8 af[2], bf[NFRAG];              // identical for any BKT
+      for (int mi = 0; mi < 2; ++mi)
+        af[mi] = *reinterpret_cast<const bf16x8*>(
+            &As[cur][wr * 32 + mi * 16 + row_a][ks * 32 + kseg * 8]);
+      if (TRANS_B) {
+        for (int ni = 0; ni < NFRAG; ++ni)
+          bf[ni] = *reinterpret_cast<const bf16x8*>(
+              &Bs[cur][wc * (16 * NFRAG) + ni * 16 + row_a][ks * 32 + kseg * 8]);
+      } else {
+        // transpose at the LDS read: Bs holds [k][n]
+        for (int ni = 0; ni < NFRAG; ++ni)
+          for (int i = 0; i < 8; ++i)
+            bf[ni][i] = Bs[cur][ks * 32 + kseg * 8 + i]
+                          [wc * (16 * NFRAG) + ni * 16 + row_a];
+      }
+      for (int mi = 0; mi < 2; ++mi)
+        for (int ni = 0; ni < NFRAG; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     }
-    for (int mi = 0; mi < 2; ++mi)
-      for (int ni = 0; ni < NFRAG; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
 
     if (kt + 1 < ktiles && !nxt_glds) store_tile(1 - cur);
     __syncthreads();
@@ -1871,8 +1885,15 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
     const char* e = getenv("GYMFX_GEMM_WIDE");
     return e ? atoi(e) : -1;
   }();
+  // BK=64 staging pipeline (half the barriers) for long-K trans_b shapes —
+  // ROADMAP lever 2; GYMFX_GEMM_BK64=1 forces on, =0 forces off
+  static const int bk64_env = [] {
+    const char* e = getenv("GYMFX_GEMM_BK64");
+    return e ? atoi(e) : -1;
+  }();
   const bool mid = (N >= 96) && wide_env == 4;
   const bool wide = (N >= 192) && !mid && wide_env == 1;
+  const bool bk64 = trans_b && !wide && !mid && K >= 192 && bk64_env == 1;
   dim3 grid(ceil_div(M, 64), ceil_div(N, wide ? 256 : (mid ? 128 : 64)));
   dim3 block(256);
 
@@ -1884,6 +1905,9 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
     else if (mid)                                                             \
       hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 4>), grid, block, 0,   \
                          stream, a, b, bias, C, y, M, N, K);                  \
+    else if (bk64)                                                            \
+      hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 2, false, 64>), grid,  \
+                         block, 0, stream, a, b, bias, C, y, M, N, K);        \
     else                                                                      \
       hipLaunchKernelGGL((gemm_kernel<TB, ACT, DT, AB, 2>), grid, block, 0,   \
                          stream, a, b, bias, C, y, M, N, K);                  \
