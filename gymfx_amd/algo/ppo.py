@@ -55,6 +55,14 @@ class PPOConfig:
     fuse_sample: bool = True  # sample the action inside the env-step kernel
                               # (native engine): one fewer launch per rollout
                               # step; bitwise == the sample_head kernel.
+    fuse_head: bool = True    # also compute head = h @ W3 + b3 INSIDE the
+                              # env-step kernel (per-env dot against the
+                              # L2-resident 2 KB head weights): removes the
+                              # tiny head GEMM launch from the rollout chain.
+                              # Requires fuse_sample; scalar-dot accumulation
+                              # order differs from the MFMA head GEMM, so
+                              # trajectories are deterministic but not
+                              # bitwise-comparable with fuse_head=False.
     overlap_gather: bool = False  # run minibatch gather on a side HIP
                                   # stream overlapped with the previous
                                   # minibatch's fwd/bwd (ping-pong slots).
@@ -88,6 +96,7 @@ class PPOConfig:
             "rollout_streams": "rollout_streams",
             "fused_rollout": "fused_rollout",
             "fuse_sample": "fuse_sample",
+            "fuse_head": "fuse_head",
             "overlap_gather": "overlap_gather",
         }
         from ..config.merger import convert_type
@@ -236,6 +245,9 @@ class PPOTrainer:
         self._fuse_sample = (self.device.type == "cuda" and cfg.fuse_sample
                              and getattr(env, "_native", None) is not None
                              and env.params.action_space_mode != "continuous")
+        self._fuse_head = (self._fuse_sample and cfg.fuse_head
+                           and cfg.hidden % 8 == 0 and n_actions + 1 <= 8
+                           and not cfg.fused_rollout)
         if self._split:
             self._s2 = torch.cuda.Stream()
             self.acts_half = [self.model.alloc_acts(N // 2),
@@ -266,13 +278,19 @@ class PPOTrainer:
                      "c": self.rnn_state["c"][lo:hi]}
         for t in range(T):
             obs_t = self.obs_buf[t][lo:hi]
+            h2 = None
             if rec:
                 if t % L == 0:
                     # chunk-boundary recurrent state for BPTT (f32 snapshot)
                     ch = t // L
                     self.h0_buf[ch][lo:hi].copy_(state["h"])
                     self.c0_buf[ch][lo:hi].copy_(state["c"])
-                head = model.step_forward(obs_t, state, acts)
+                if self._fuse_head:
+                    # head computed inside the env kernel from the new h
+                    model.step_forward(obs_t, state, acts, skip_head=True)
+                    head, h2 = None, state["h"]
+                else:
+                    head = model.step_forward(obs_t, state, acts)
             elif self._fused:
                 model.fused_step(
                     obs_t, self.act_buf[t][lo:hi], self.logp_buf[t][lo:hi],
@@ -280,9 +298,12 @@ class PPOTrainer:
                     step_base=self.step_base, row_offset=lo,
                 )
                 head = None
+            elif self._fuse_head:
+                h2 = model.forward_hidden(obs_t, acts)
+                head = None
             else:
                 head = model.forward(obs_t, acts)
-            fuse = self._fuse_sample and head is not None
+            fuse = self._fuse_sample and (head is not None or h2 is not None)
             if head is not None and not fuse:
                 api.sample_head(
                     head, self.sample_seed, t,
@@ -294,11 +315,16 @@ class PPOTrainer:
             # the NEXT observation (bf16) into obs_buf[t+1] directly; with
             # fuse_sample it also samples the action from `head` itself.
             nxt = self.obs_buf[t + 1] if t + 1 < T else self.obs_bf16_step
+            wname = "Wy" if rec else "W3"
+            bname = "by" if rec else "b3"
             env.step(
                 self.act_buf[t], reward_out=self.rew_buf[t],
                 terminated_out=self.done_buf[t], obs_bf16_out=nxt,
                 env_lo=lo, env_hi=hi,
                 head=head if fuse else None,
+                h2=h2 if fuse else None,
+                w3t=model.wt(wname) if (fuse and h2 is not None) else None,
+                b3=model.f32(bname) if (fuse and h2 is not None) else None,
                 logp_out=self.logp_buf[t] if fuse else None,
                 value_out=self.val_buf[t] if fuse else None,
                 step_base=self.step_base if fuse else None,

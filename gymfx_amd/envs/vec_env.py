@@ -175,7 +175,10 @@ class VecFxEnv:
              step_base: Optional[torch.Tensor] = None,
              sample_seed: int = 0,
              sample_step: int = 0,
-             fuse_obs: bool = False) -> Dict[str, torch.Tensor]:
+             fuse_obs: bool = False,
+             h2: Optional[torch.Tensor] = None,
+             w3t: Optional[torch.Tensor] = None,
+             b3: Optional[torch.Tensor] = None) -> Dict[str, torch.Tensor]:
         """Advance all envs. Returns dict with obs/reward/terminated tensors.
 
         reward_out / terminated_out / obs_bf16_out: optional preallocated
@@ -195,10 +198,10 @@ class VecFxEnv:
                                      terminated_out, obs_bf16_out,
                                      env_lo, env_hi, head, logp_out,
                                      value_out, step_base, sample_seed,
-                                     sample_step, fuse_obs)
+                                     sample_step, fuse_obs, h2, w3t, b3)
             info["obs"] = self._obs
             return info
-        if head is not None:
+        if head is not None or h2 is not None:
             raise ValueError("fused sampling requires the native engine")
         if env_lo != 0 or env_hi not in (0, self.params.n_envs):
             raise ValueError("env range stepping requires the native engine")

@@ -164,7 +164,8 @@ class ActorCriticLSTM:
     # -- single step (rollout) -------------------------------------------
     def step_forward(self, obs_bf16: torch.Tensor, state: Dict[str, torch.Tensor],
                      acts: Dict[str, torch.Tensor], *,
-                     fused: bool = True) -> torch.Tensor:
+                     fused: bool = True,
+                     skip_head: bool = False) -> "Optional[torch.Tensor]":
         """One policy step: updates state['h']/state['c'] IN PLACE (the h
         TENSOR IDENTITY rotates with acts['h_tmp'] on the fused path —
         callers must read state['h'] through the dict), returns head
@@ -183,6 +184,9 @@ class ActorCriticLSTM:
                      trans_b=True)
             api.lstm_cell_fwd(gates, acts["gates_h"], state["c"], state["c"],
                               state["h"])
+        if skip_head:
+            # head-in-step fusion: the env kernel computes h @ Wy + by
+            return None
         api.gemm(state["h"], self.wt("Wy"), self.f32("by"), acts["head"], act=0, trans_b=True)
         return acts["head"]
 
@@ -276,14 +280,14 @@ class ActorCriticLSTM:
     def adam(self, lr: float, *, beta1=0.9, beta2=0.999, eps=1e-8,
              max_grad_norm: float = 0.0) -> None:
         self.adam_step += 1
-        gscale = None
+        clip = None
         if max_grad_norm and max_grad_norm > 0:
-            api.grad_clip_scale(self.grads, max_grad_norm, self._clip_part,
-                                self._clip_scale)
-            gscale = self._clip_scale
+            # fused clipping: sumsq partials + per-block scale derivation
+            # inside the adam launch (no separate clip_scale kernel)
+            clip = (self._clip_part, max_grad_norm)
         api.adam(self.params, self.grads, self.m, self.v, self.params_bf16,
                  lr=lr, beta1=beta1, beta2=beta2, eps=eps, step=self.adam_step,
-                 gscale=gscale, step_ctr=self.adam_ctr)
+                 step_ctr=self.adam_ctr, clip=clip)
         api.increment_counter(self.adam_ctr, 1)
         self._refresh_wt()
 

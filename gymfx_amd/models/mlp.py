@@ -166,6 +166,15 @@ class ActorCriticMLP:
         api.gemm(acts["h2"], self.wt("W3"), self.f32("b3"), acts["head"], act=0, trans_b=True)
         return acts["head"]
 
+    def forward_hidden(self, obs_bf16: torch.Tensor,
+                       acts: Dict[str, torch.Tensor]) -> torch.Tensor:
+        """Layers 1-2 only: the rollout's head-in-step fusion computes the
+        tiny head GEMM inside the env kernel (h2 @ W3 + b3 per env row),
+        removing one launch from the latency-bound per-step chain."""
+        api.gemm(obs_bf16, self.wt("W1"), self.f32("b1"), acts["h1"], act=2, trans_b=True)
+        api.gemm(acts["h1"], self.wt("W2"), self.f32("b2"), acts["h2"], act=2, trans_b=True)
+        return acts["h2"]
+
     # -- backward (dhead [M, A+1] bf16 -> accumulate grads) --------------
     def backward(
         self,
@@ -205,16 +214,16 @@ class ActorCriticMLP:
     def adam(self, lr: float, *, beta1=0.9, beta2=0.999, eps=1e-8,
              max_grad_norm: float = 0.0) -> None:
         self.adam_step += 1
-        gscale = None
+        clip = None
         if max_grad_norm and max_grad_norm > 0:
-            api.grad_clip_scale(self.grads, max_grad_norm, self._clip_part,
-                                self._clip_scale)
-            gscale = self._clip_scale
+            # fused clipping: sumsq partials + per-block scale derivation
+            # inside the adam launch (no separate clip_scale kernel)
+            clip = (self._clip_part, max_grad_norm)
         # bias correction reads the DEVICE counter (graph-replayable); the
         # host step is a fallback for paths that pass step_ctr=None.
         api.adam(self.params, self.grads, self.m, self.v, self.params_bf16,
                  lr=lr, beta1=beta1, beta2=beta2, eps=eps, step=self.adam_step,
-                 gscale=gscale, step_ctr=self.adam_ctr)
+                 step_ctr=self.adam_ctr, clip=clip)
         api.increment_counter(self.adam_ctr, 1)
         self._refresh_wt()
 

@@ -311,17 +311,31 @@ def adam(
     step: int = 1,
     gscale: Optional[torch.Tensor] = None,
     step_ctr: Optional[torch.Tensor] = None,
+    clip: "Optional[Tuple[torch.Tensor, float]]" = None,
 ) -> None:
     """step_ctr (device int32 scalar): when given, bias correction uses
     ``*step_ctr + 1`` instead of the host ``step`` (hipGraph-replayable; pair
-    with ``increment_counter(step_ctr, 1)`` after the call)."""
+    with ``increment_counter(step_ctr, 1)`` after the call).
+    clip=(partials, max_norm): fused grad clipping — the sumsq partials are
+    computed and the scale derived inside the adam launch (one fewer kernel
+    in the 32x-per-update optimizer chain)."""
     if _use_native(p):
-        native.require().adam(p, g, m, v, p_bf16, lr, beta1, beta2, eps, step,
-                              gscale, step_ctr)
+        if clip is not None:
+            part, max_norm = clip
+            native.require().adam(p, g, m, v, p_bf16, lr, beta1, beta2, eps,
+                                  step, None, step_ctr, part, float(max_norm))
+        else:
+            native.require().adam(p, g, m, v, p_bf16, lr, beta1, beta2, eps,
+                                  step, gscale, step_ctr)
         return
     if step_ctr is not None:
         step = int(step_ctr.item()) + 1
-    s = float(gscale.item()) if gscale is not None else 1.0
+    if clip is not None:
+        _, max_norm = clip
+        norm = g.norm()
+        s = float(max_norm / norm) if (max_norm > 0 and norm > max_norm) else 1.0
+    else:
+        s = float(gscale.item()) if gscale is not None else 1.0
     geff = g * s
     m.mul_(beta1).add_(geff, alpha=1 - beta1)
     v.mul_(beta2).addcmul_(geff, geff, value=1 - beta2)

@@ -75,6 +75,13 @@ void launch_adam(float* p, const float* g, float* m, float* v, void* p_bf16,
                  hipStream_t stream);
 void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
                       float* scale, int nparts, hipStream_t stream);
+void launch_grad_sumsq(const float* g, int64_t n, float* part, int nparts,
+                       hipStream_t stream);
+void launch_adam_clip(float* p, const float* g, float* m, float* v,
+                      void* p_bf16, int64_t n, float lr, float beta1,
+                      float beta2, float eps, float bc1, float bc2,
+                      const int* step_ctr, const float* clip_part, int nparts,
+                      float max_norm, hipStream_t stream);
 void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,
                         uint64_t step, int64_t* actions, float* logp,
                         float* value, float* entropy, int greedy,
@@ -300,7 +307,10 @@ struct GymFxEngine {
                 c10::optional<torch::Tensor> logp_out,
                 c10::optional<torch::Tensor> value_out,
                 c10::optional<torch::Tensor> step_base,
-                int64_t sample_seed, int64_t sample_step, bool fuse_obs) {
+                int64_t sample_seed, int64_t sample_step, bool fuse_obs,
+                c10::optional<torch::Tensor> h2,
+                c10::optional<torch::Tensor> w3t,
+                c10::optional<torch::Tensor> b3) {
     TORCH_CHECK(actions.is_cuda() == state["cursor"].is_cuda(),
                 "actions must live on the env device");
     TORCH_CHECK(actions.numel() == K.n_envs, "actions numel != n_envs");
@@ -319,10 +329,47 @@ struct GymFxEngine {
     // optional fused sampling: the step kernel samples from `head` and
     // writes actions/logp/value itself (one fewer launch per step).
     P.head = nullptr;
+    P.h2 = nullptr;
+    P.w3t = nullptr;
+    P.b3 = nullptr;
     P.actions_out = nullptr;
     P.logp_out = nullptr;
     P.value_out = nullptr;
     P.step_base = nullptr;
+    if (h2.has_value()) {
+      // head-in-step fusion: the kernel computes head = h2 @ W3 + b3
+      TORCH_CHECK(!continuous, "fused sampling requires discrete actions");
+      TORCH_CHECK(!head.has_value(), "pass either head or h2, not both");
+      TORCH_CHECK(w3t.has_value() && b3.has_value(), "h2 needs w3t and b3");
+      TORCH_CHECK(h2->is_contiguous() &&
+                      h2->scalar_type() == torch::kBFloat16 && h2->dim() == 2,
+                  "h2 must be contiguous bf16 [env_cnt, H]");
+      TORCH_CHECK(w3t->is_contiguous() &&
+                      w3t->scalar_type() == torch::kBFloat16 &&
+                      w3t->dim() == 2 && w3t->size(1) == h2->size(1),
+                  "w3t must be contiguous bf16 [A+1, H]");
+      TORCH_CHECK(h2->size(1) % 8 == 0, "H must be a multiple of 8");
+      TORCH_CHECK(w3t->size(0) <= 8, "head width must be <= 8");
+      TORCH_CHECK(b3->scalar_type() == torch::kFloat32 &&
+                      b3->numel() == w3t->size(0),
+                  "b3 must be f32 [A+1]");
+      TORCH_CHECK(logp_out.has_value() && value_out.has_value(),
+                  "fused sampling needs logp_out and value_out");
+      P.h2 = h2->data_ptr();
+      P.w3t = w3t->data_ptr();
+      P.b3 = b3->data_ptr<float>();
+      P.actions_out = actions.data_ptr<int64_t>();
+      P.logp_out = logp_out->data_ptr<float>();
+      P.value_out = value_out->data_ptr<float>();
+      K.head_hidden = (int)h2->size(1);
+      K.sample_nact = (int)w3t->size(0) - 1;
+      K.sample_seed = (unsigned long long)sample_seed;
+      K.sample_step = (long long)sample_step;
+      if (step_base.has_value()) {
+        P.step_base = reinterpret_cast<const unsigned long long*>(
+            step_base->data_ptr());
+      }
+    }
     if (head.has_value()) {
       TORCH_CHECK(!continuous, "fused sampling requires discrete actions");
       TORCH_CHECK(head->is_contiguous() &&
@@ -722,7 +769,8 @@ void adam_op(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
              c10::optional<torch::Tensor> p_bf16, double lr, double beta1,
              double beta2, double eps, int64_t step,
              c10::optional<torch::Tensor> gscale,
-             c10::optional<torch::Tensor> step_ctr) {
+             c10::optional<torch::Tensor> step_ctr,
+             c10::optional<torch::Tensor> clip_part, double clip_max_norm) {
   check_f32(p, "p");
   check_f32(g, "g");
   check_f32(m, "m");
@@ -743,6 +791,20 @@ void adam_op(torch::Tensor p, torch::Tensor g, torch::Tensor m, torch::Tensor v,
   }
   const float bc1 = 1.f - powf((float)beta1, (float)step);
   const float bc2 = 1.f - powf((float)beta2, (float)step);
+  if (clip_part.has_value() && clip_max_norm > 0) {
+    // fused clip: sumsq partials, then adam derives the scale per block
+    check_f32(*clip_part, "clip_part");
+    float* part = clip_part->data_ptr<float>();
+    const int nparts = (int)clip_part->numel();
+    gymfx::launch_grad_sumsq(g.data_ptr<float>(), n, part, nparts,
+                             cur_stream());
+    gymfx::launch_adam_clip(p.data_ptr<float>(), g.data_ptr<float>(),
+                            m.data_ptr<float>(), v.data_ptr<float>(), pb, n,
+                            (float)lr, (float)beta1, (float)beta2, (float)eps,
+                            bc1, bc2, sc, part, nparts, (float)clip_max_norm,
+                            cur_stream());
+    return;
+  }
   gymfx::launch_adam(p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(), pb, n, (float)lr,
                      (float)beta1, (float)beta2, (float)eps, bc1, bc2, gs, sc,
@@ -957,7 +1019,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam", &adam_op, py::arg("p"), py::arg("g"), py::arg("m"),
         py::arg("v"), py::arg("p_bf16"), py::arg("lr"), py::arg("beta1"),
         py::arg("beta2"), py::arg("eps"), py::arg("step"),
-        py::arg("gscale") = py::none(), py::arg("step_ctr") = py::none());
+        py::arg("gscale") = py::none(), py::arg("step_ctr") = py::none(),
+        py::arg("clip_part") = py::none(), py::arg("clip_max_norm") = 0.0);
   m.def("grad_clip", &grad_clip_op);
   m.def("sample_head", &sample_head_op, py::arg("head"), py::arg("seed"),
         py::arg("step"), py::arg("actions"), py::arg("logp"),
@@ -993,7 +1056,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("env_hi") = 0, py::arg("head") = py::none(),
            py::arg("logp_out") = py::none(), py::arg("value_out") = py::none(),
            py::arg("step_base") = py::none(), py::arg("sample_seed") = 0,
-           py::arg("sample_step") = 0, py::arg("fuse_obs") = false)
+           py::arg("sample_step") = 0, py::arg("fuse_obs") = false,
+           py::arg("h2") = py::none(), py::arg("w3t") = py::none(),
+           py::arg("b3") = py::none())
       .def("build_obs", &gymfx::GymFxEngine::build_obs, py::arg("obs_out"),
            py::arg("obs_bf16_out") = py::none(), py::arg("env_lo") = 0,
            py::arg("env_hi") = 0);

@@ -140,6 +140,7 @@ struct EnvParamsK {
   double feature_clip, overlay_threshold;
   double margin_init_rate;
   // fused policy sampling (set per step() call when head != null)
+  int head_hidden;           // h2 width for the head-in-step fusion
   unsigned long long sample_seed;
   long long sample_step;
   int sample_nact;
@@ -194,6 +195,12 @@ struct EnvPtrs {
   // itself — identical math to sample_head_kernel — and writes it to
   // actions_out before decoding it.
   const float *head;         // [env_cnt, sample_nact+1] f32 (rows local to env_lo)
+  // head-in-step fusion: when h2/w3t/b3 are set (and head is null) the
+  // step kernel computes head = h2 @ W3 + b3 per env itself — removes the
+  // tiny head GEMM launch from the latency-bound rollout chain
+  const void *h2;            // [env_cnt, head_hidden] bf16
+  const void *w3t;           // [sample_nact+1, head_hidden] bf16
+  const float *b3;           // [sample_nact+1]
   int64_t *actions_out;      // sampled action (same buffer `actions` reads)
   float *logp_out;           // [N] f32
   float *value_out;          // [N] f32

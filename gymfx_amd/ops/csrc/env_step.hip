@@ -109,9 +109,32 @@ GFX_DEV void env_step_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
   // One thread already owns env row n; sampling here (identical math to
   // sample_head_kernel, shared helpers in env_common.h) removes one kernel
   // launch per rollout step from the latency-bound per-step chain.
-  if (P.head) {
+  if (P.head || P.h2) {
     const int A = K.sample_nact;
-    const float* row = P.head + (int64_t)(n - env_lo) * (A + 1);
+    float row_l[8];
+    const float* row;
+    if (P.h2) {
+      // head = h2 @ W3 + b3 computed inline (W3t is 2 KB, L2-resident;
+      // the h2 row streams as bf16x8 chunks) — one fewer rollout launch
+      const __bf16* h = reinterpret_cast<const __bf16*>(P.h2) +
+                        (int64_t)(n - env_lo) * K.head_hidden;
+      const __bf16* w = reinterpret_cast<const __bf16*>(P.w3t);
+      float s[8];
+      for (int j = 0; j <= A; ++j) s[j] = 0.f;
+      typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
+      for (int k8 = 0; k8 < K.head_hidden; k8 += 8) {
+        const bf16x8v hv = *reinterpret_cast<const bf16x8v*>(&h[k8]);
+        for (int j = 0; j <= A; ++j) {
+          const bf16x8v wv = *reinterpret_cast<const bf16x8v*>(
+              &w[(int64_t)j * K.head_hidden + k8]);
+          for (int i = 0; i < 8; ++i) s[j] += (float)hv[i] * (float)wv[i];
+        }
+      }
+      for (int j = 0; j <= A; ++j) row_l[j] = s[j] + P.b3[j];
+      row = row_l;
+    } else {
+      row = P.head + (int64_t)(n - env_lo) * (A + 1);
+    }
     const float logz = head_logz(row, A);
     uint64_t stp = (uint64_t)K.sample_step;
     if (P.step_base) stp += *P.step_base;

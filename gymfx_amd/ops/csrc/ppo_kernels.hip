@@ -673,12 +673,18 @@ __global__ void gae_kernel(const float* __restrict__ rewards,
 // ---------------------------------------------------------------------------
 // Fused Adam (f32 master params; bf16 mirror refreshed in the same pass)
 // ---------------------------------------------------------------------------
+// clip_part/nparts/max_norm: when given, every block tree-reduces the
+// sumsq partials itself and derives the clip scale inline (the partials
+// are ~1 KB, L2-resident) — this removed the separate one-block
+// clip_scale launch from the 32x-per-update optimizer chain.
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             __bf16* __restrict__ p_bf16, int64_t n, float lr,
                             float beta1, float beta2, float eps, float bc1,
                             float bc2, const float* __restrict__ gscale,
-                            const int* __restrict__ step_ctr) {
+                            const int* __restrict__ step_ctr,
+                            const float* __restrict__ clip_part, int nparts,
+                            float max_norm) {
   if (step_ctr) {
     // device step counter (hipGraph-replayable): effective step = *ctr + 1;
     // the companion increment kernel advances it after this launch.
@@ -686,7 +692,22 @@ __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
     bc1 = 1.f - powf(beta1, st);
     bc2 = 1.f - powf(beta2, st);
   }
-  const float s = gscale ? *gscale : 1.f;
+  float s = 1.f;
+  if (clip_part) {
+    __shared__ float red[256];
+    float acc = 0.f;
+    for (int i = threadIdx.x; i < nparts; i += 256) acc += clip_part[i];
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int w = 128; w > 0; w >>= 1) {
+      if (threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+      __syncthreads();
+    }
+    const float norm = sqrtf(red[0]);
+    s = (max_norm > 0.f && norm > max_norm) ? max_norm / norm : 1.f;
+  } else if (gscale) {
+    s = *gscale;
+  }
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     const float gi = g[i] * s;
@@ -1893,7 +1914,11 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
   }();
   const bool mid = (N >= 96) && wide_env == 4;
   const bool wide = (N >= 192) && !mid && wide_env == 1;
-  const bool bk64 = trans_b && !wide && !mid && K >= 192 && bk64_env == 1;
+  // measured: BK=64 wins for long-K shapes (LSTM dgates dgrad K=1024:
+  // update 41.5 -> 39.9 ms) and LOSES at K<=260 (MLP headline 28.4 ->
+  // 27.4M) — auto-enable only at K >= 512
+  const bool bk64 = trans_b && !wide && !mid &&
+                    (bk64_env == 1 || (bk64_env != 0 && K >= 512));
   dim3 grid(ceil_div(M, 64), ceil_div(N, wide ? 256 : (mid ? 128 : 64)));
   dim3 block(256);
 
@@ -2043,7 +2068,19 @@ void launch_adam(float* p, const float* g, float* m, float* v, void* p_bf16,
   int blocks = (int)std::min<int64_t>((n + 255) / 256, 2048);
   hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, stream, p, g, m,
                      v, reinterpret_cast<__bf16*>(p_bf16), n, lr, beta1, beta2,
-                     eps, bc1, bc2, gscale, step_ctr);
+                     eps, bc1, bc2, gscale, step_ctr, nullptr, 0, 0.f);
+}
+
+void launch_adam_clip(float* p, const float* g, float* m, float* v,
+                      void* p_bf16, int64_t n, float lr, float beta1,
+                      float beta2, float eps, float bc1, float bc2,
+                      const int* step_ctr, const float* clip_part, int nparts,
+                      float max_norm, hipStream_t stream) {
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, 2048);
+  hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, stream, p, g, m,
+                     v, reinterpret_cast<__bf16*>(p_bf16), n, lr, beta1, beta2,
+                     eps, bc1, bc2, nullptr, step_ctr, clip_part, nparts,
+                     max_norm);
 }
 
 void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
@@ -2052,6 +2089,12 @@ void launch_grad_clip(const float* g, int64_t n, float max_norm, float* part,
                      g, n, part);
   hipLaunchKernelGGL(clip_scale_kernel, dim3(1), dim3(256), 0, stream, part,
                      nparts, max_norm, scale);
+}
+
+void launch_grad_sumsq(const float* g, int64_t n, float* part, int nparts,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(sumsq_partial_kernel, dim3(nparts), dim3(256), 0, stream,
+                     g, n, part);
 }
 
 void launch_sample_head(const float* head, int M, int n_actions, uint64_t seed,

@@ -181,7 +181,9 @@ class NativeEngine:
              env_hi: int = 0, head: torch.Tensor = None,
              logp_out: torch.Tensor = None, value_out: torch.Tensor = None,
              step_base: torch.Tensor = None, sample_seed: int = 0,
-             sample_step: int = 0, fuse_obs: bool = False) -> Dict[str, torch.Tensor]:
+             sample_step: int = 0, fuse_obs: bool = False,
+             h2: torch.Tensor = None, w3t: torch.Tensor = None,
+             b3: torch.Tensor = None) -> Dict[str, torch.Tensor]:
         if self._params.action_space_mode == "continuous":
             actions = actions.to(torch.float32).contiguous()
         else:
@@ -190,7 +192,7 @@ class NativeEngine:
                                       terminated_out, obs_bf16_out,
                                       env_lo, env_hi, head, logp_out,
                                       value_out, step_base, sample_seed,
-                                      sample_step, fuse_obs))
+                                      sample_step, fuse_obs, h2, w3t, b3))
 
     def build_obs(self, obs_out: torch.Tensor,
                   obs_bf16_out: torch.Tensor = None) -> None:
