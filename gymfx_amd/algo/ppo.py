@@ -55,14 +55,14 @@ class PPOConfig:
     fuse_sample: bool = True  # sample the action inside the env-step kernel
                               # (native engine): one fewer launch per rollout
                               # step; bitwise == the sample_head kernel.
-    fuse_head: bool = True    # also compute head = h @ W3 + b3 INSIDE the
-                              # env-step kernel (per-env dot against the
-                              # L2-resident 2 KB head weights): removes the
-                              # tiny head GEMM launch from the rollout chain.
-                              # Requires fuse_sample; scalar-dot accumulation
-                              # order differs from the MFMA head GEMM, so
-                              # trajectories are deterministic but not
-                              # bitwise-comparable with fuse_head=False.
+    fuse_head: bool = False   # compute head = h @ W3 + b3 INSIDE the
+                              # env-step kernel.  Measured SLOWER (rollout
+                              # 6.4 -> 10.1 ms at N=4096): the env kernel
+                              # runs only N/256 workgroups, so the per-lane
+                              # 1024-FMA scalar dot serializes instead of
+                              # the 6.6 us MFMA head GEMM it replaced —
+                              # kept as a knob for much wider fleets
+                              # (profiles/PERF_NOTES.md).
     overlap_gather: bool = False  # run minibatch gather on a side HIP
                                   # stream overlapped with the previous
                                   # minibatch's fwd/bwd (ping-pong slots).

@@ -44,6 +44,62 @@ GFX_DEV float fast_tanh(float x) {
   return 1.f - 2.f * __builtin_amdgcn_rcpf(e + 1.f);
 }
 
+GFX_DEV uint32_t feistel_encrypt_once(uint32_t x, int half, uint32_t mask,
+                                      uint64_t key) {
+  uint32_t a = x & mask;        // low half
+  uint32_t b = x >> half;       // high half
+  for (int r = 0; r < 4; ++r) {
+    const uint32_t f =
+        (uint32_t)(splitmix64(key ^ ((uint64_t)r << 48) ^ (uint64_t)b)) & mask;
+    const uint32_t t = b;
+    b = a ^ f;
+    a = t;
+  }
+  return (b << half) | a;
+}
+
+GFX_DEV uint32_t feistel_perm_idx(uint32_t i, uint32_t n, int half,
+                                  uint64_t key) {
+  const uint32_t mask = (1u << half) - 1u;
+  uint32_t x = i;
+  do {
+    x = feistel_encrypt_once(x, half, mask, key);
+  } while (x >= n);  // cycle-walk: terminates (bijection over 2^(2*half))
+  return x;
+}
+
+// Row indirection for the gather+first-GEMM fusion: the L1 forward GEMM
+// (and the W1 wgrad) read the rollout slab obs_flat THROUGH the epoch
+// permutation instead of a materialized obs_mb copy (ROADMAP lever 3 —
+// saves one full write+read pass over the minibatch observations).
+// ctr_off: the gather that produced the OTHER minibatch fields has already
+// advanced mb_ctr when these kernels run (-1 in the trainer's sequence).
+struct FeistelMap {
+  uint32_t n;  // permutation domain (total rollout rows)
+  int half;
+  uint64_t seed;
+  int minibatches;
+  int M_mb;    // rows per minibatch
+  long long ctr_off;
+  const unsigned long long* step_base;
+  const unsigned long long* mb_ctr;
+};
+
+GFX_DEV uint64_t feistel_key(const FeistelMap& fm, uint32_t* mb) {
+  const unsigned long long ctr =
+      (unsigned long long)((long long)*fm.mb_ctr + fm.ctr_off);
+  const uint32_t epoch = (uint32_t)(ctr / (unsigned)fm.minibatches);
+  *mb = (uint32_t)(ctr % (unsigned)fm.minibatches);
+  return splitmix64(fm.seed ^ (*fm.step_base * 0x9E3779B97F4A7C15ull) ^
+                    ((uint64_t)epoch << 32));
+}
+
+GFX_DEV int64_t feistel_src_row(const FeistelMap& fm, uint64_t key,
+                                uint32_t mb, int row) {
+  return (int64_t)feistel_perm_idx(mb * (uint32_t)fm.M_mb + (uint32_t)row,
+                                   fm.n, fm.half, key);
+}
+
 // ---------------------------------------------------------------------------
 // GEMM: C[M,N] = act(A[M,K] @ B + bias)
 //   TRANS_B = false: B stored [K, N] (ldb = N)
@@ -70,12 +126,12 @@ GFX_DEV float fast_tanh(float x) {
 // via GYMFX_GEMM_BK64.  The MFMA k order is unchanged (two sequential
 // 16x16x32 steps per 64-chunk), so results stay bitwise identical.
 template <bool TRANS_B, int ACT, bool DACT_TANH, bool ADD_BIAS, int NFRAG,
-          bool ACCUM = false, int BKT = 32>
+          bool ACCUM = false, int BKT = 32, bool A_PERM = false>
 __global__ __launch_bounds__(256, 2) void gemm_kernel(
     const __bf16* __restrict__ A, const __bf16* __restrict__ B,
     const float* __restrict__ bias, void* __restrict__ C,
     const __bf16* __restrict__ Yact,  // activation output (DACT_TANH)
-    int M, int N, int K) {
+    int M, int N, int K, FeistelMap fm = FeistelMap{}) {
   constexpr int BM = 64, BK = BKT;
   constexpr int BN = 32 * NFRAG;
   // TRANS_B images are UNPADDED linear (required by the glds staging; the
@@ -101,6 +157,14 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
 
   f32x4 acc[2][NFRAG] = {};
 
+  // gather+first-GEMM fusion: A rows go through the epoch permutation
+  uint64_t fkey = 0;
+  uint32_t fmb = 0;
+  if (A_PERM) fkey = feistel_key(fm, &fmb);
+  auto a_row = [&](int gr) -> int64_t {
+    return A_PERM ? feistel_src_row(fm, fkey, fmb, gr) : (int64_t)gr;
+  };
+
   const int row_a = lane & 15;
   const int kseg = lane >> 4;  // 0..3 -> k-base = kseg*8
 
@@ -118,12 +182,13 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
       const int e = (tid + j * 256) * 8;
       const int gr = bm + e / BK;
       if (gr < M) {
+        const int64_t sr = a_row(gr);
         const int gk = k0 + e % BK;
         if (gk + 8 <= K) {
-          ra[j] = *reinterpret_cast<const bf16x8*>(&A[(int64_t)gr * K + gk]);
+          ra[j] = *reinterpret_cast<const bf16x8*>(&A[sr * K + gk]);
         } else {
           for (int i = 0; i < 8; ++i)
-            ra[j][i] = (gk + i < K) ? A[(int64_t)gr * K + gk + i] : (__bf16)0.f;
+            ra[j][i] = (gk + i < K) ? A[sr * K + gk + i] : (__bf16)0.f;
         }
       }
     }
@@ -194,7 +259,7 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
       const int col = (e0 + lane * 8) % BK;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)(uintptr_t)(
-              &A[(int64_t)(bm + row) * K + k0 + col]),
+              &A[a_row(bm + row) * K + k0 + col]),
           (__attribute__((address_space(3))) void*)(uintptr_t)(
               &As[buf][0][0] + e0),
           16, 0, 0);
@@ -1574,29 +1639,8 @@ __global__ void increment_i32_kernel(int* __restrict__ ctr, int delta) {
 // replaces host randperm + five giant index-gathers (and is exactly
 // reproducible on the CPU oracle: ops/api.feistel_perm).
 // ---------------------------------------------------------------------------
-GFX_DEV uint32_t feistel_encrypt_once(uint32_t x, int half, uint32_t mask,
-                                      uint64_t key) {
-  uint32_t a = x & mask;        // low half
-  uint32_t b = x >> half;       // high half
-  for (int r = 0; r < 4; ++r) {
-    const uint32_t f =
-        (uint32_t)(splitmix64(key ^ ((uint64_t)r << 48) ^ (uint64_t)b)) & mask;
-    const uint32_t t = b;
-    b = a ^ f;
-    a = t;
-  }
-  return (b << half) | a;
-}
-
-GFX_DEV uint32_t feistel_perm_idx(uint32_t i, uint32_t n, int half,
-                                  uint64_t key) {
-  const uint32_t mask = (1u << half) - 1u;
-  uint32_t x = i;
-  do {
-    x = feistel_encrypt_once(x, half, mask, key);
-  } while (x >= n);  // cycle-walk: terminates (bijection over 2^(2*half))
-  return x;
-}
+// (feistel helpers moved above gemm_kernel: the gather+first-GEMM
+// fusion indexes A rows through the permutation)
 
 // One wavefront copies one minibatch row. grid.x = ceil(M/4), block = 256.
 // mb_ctr (device) = epoch * minibatches + mb, advanced by increment after
