@@ -232,6 +232,18 @@ class PPOTrainer:
         self.sample_seed = cfg.seed * 1_000_003 + rank
         self.shuffle_seed = cfg.seed * 9973 + rank
 
+        # gather+first-GEMM fusion (ROADMAP lever 3): L1 fwd and the W1
+        # wgrad read obs rows through the epoch permutation, so the gather
+        # kernel skips its biggest copy.  A/B: GYMFX_FUSE_GATHER=0
+        import os as _os
+        self._fuse_gather = (
+            not self.recurrent and self.device.type == "cuda"
+            and cfg.shuffle_rows and not self._overlap
+            and _os.environ.get("GYMFX_FUSE_GATHER", "1") != "0")
+        self._fm = {"seed": self.shuffle_seed, "minibatches": cfg.minibatches,
+                    "ctr_off": -1, "step_base": self.step_base,
+                    "mb_ctr": self.mb_ctr}
+
         self.global_step = 0   # host mirror of step_base (logging/ckpt)
         self.update_count = 0
         self.use_graphs = bool(cfg.use_graphs) and self.device.type == "cuda"
@@ -397,6 +409,7 @@ class PPOTrainer:
                 self.logp_mb_s[slot], self.adv_mb_s[slot], self.ret_mb_s[slot],
                 seed=self.shuffle_seed, minibatches=cfg.minibatches,
                 step_base=self.step_base, mb_ctr=self.mb_ctr,
+                skip_obs=self._fuse_gather,
             )
         api.increment_counter(self.mb_ctr, 1)
 
@@ -418,7 +431,14 @@ class PPOTrainer:
             model.bptt_backward(self.obs_mb_seq, self.done_mb, self.dhead,
                                 self.bptt)
             return
-        head = model.forward(self.obs_mb_s[slot], self.acts_train)
+        if self._fuse_gather:
+            # obs rows come straight from the rollout slab through the
+            # permutation (mb_ctr was already advanced by the gather of the
+            # small fields: ctr_off=-1 in self._fm)
+            head = model.forward(self.obs_flat, self.acts_train,
+                                 a_feistel=self._fm)
+        else:
+            head = model.forward(self.obs_mb_s[slot], self.acts_train)
         api.ppo_loss_bwd(
             head, self.act_mb_s[slot], self.logp_mb_s[slot],
             self.adv_mb_s[slot], self.ret_mb_s[slot],
@@ -426,8 +446,12 @@ class PPOTrainer:
             vf_coef=cfg.vf_coef, inv_count=1.0 / self.mb_rows,
             losses=self.losses,
         )
-        model.backward(self.obs_mb_s[slot], self.acts_train, self.dhead,
-                       self.scratch)
+        if self._fuse_gather:
+            model.backward(self.obs_flat, self.acts_train, self.dhead,
+                           self.scratch, a_feistel=self._fm)
+        else:
+            model.backward(self.obs_mb_s[slot], self.acts_train, self.dhead,
+                           self.scratch)
 
     def _mb_body(self, slot: int = 0) -> None:
         self._gather_body(slot)

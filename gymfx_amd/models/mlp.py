@@ -156,12 +156,16 @@ class ActorCriticMLP:
             self.f32("b2"), self.wt("W3"), self.f32("b3"), actions, logp,
             value, seed, step, step_base, row_offset, greedy)
 
-    def forward(self, obs_bf16: torch.Tensor, acts: Dict[str, torch.Tensor]) -> torch.Tensor:
+    def forward(self, obs_bf16: torch.Tensor, acts: Dict[str, torch.Tensor],
+                a_feistel: "Optional[dict]" = None) -> torch.Tensor:
         """obs_bf16 [M, obs_dim] -> head f32 [M, A+1]; saves h1/h2 for bwd.
 
         B operands are the transposed mirrors (TRANS_B path): contiguous
-        vector staging on gfx950 regardless of tile width."""
-        api.gemm(obs_bf16, self.wt("W1"), self.f32("b1"), acts["h1"], act=2, trans_b=True)
+        vector staging on gfx950 regardless of tile width.
+        a_feistel: gather+first-GEMM fusion — obs_bf16 is the FULL rollout
+        slab and L1 reads its rows through the epoch permutation."""
+        api.gemm(obs_bf16, self.wt("W1"), self.f32("b1"), acts["h1"], act=2,
+                 trans_b=True, a_perm=a_feistel)
         api.gemm(acts["h1"], self.wt("W2"), self.f32("b2"), acts["h2"], act=2, trans_b=True)
         api.gemm(acts["h2"], self.wt("W3"), self.f32("b3"), acts["head"], act=0, trans_b=True)
         return acts["head"]
@@ -182,8 +186,9 @@ class ActorCriticMLP:
         acts: Dict[str, torch.Tensor],
         dhead: torch.Tensor,
         scratch: Dict[str, torch.Tensor],
+        a_feistel: "Optional[dict]" = None,
     ) -> None:
-        M = obs_bf16.shape[0]
+        M = dhead.shape[0]
         dh2 = scratch["dh2"]
         dh1 = scratch["dh1"]
         # head layer
@@ -201,7 +206,7 @@ class ActorCriticMLP:
         # layer 1
         dw_p, db_p, slabs = self._wg_workspace("W1", self.obs_dim, self.hidden, True)
         api.wgrad(obs_bf16, dh1, self.grad("W1"), self.grad("b1"),
-                  workspace=(dw_p, db_p), slabs=slabs)
+                  workspace=(dw_p, db_p), slabs=slabs, x_perm=a_feistel)
 
     def alloc_scratch(self, M: int) -> Dict[str, torch.Tensor]:
         dev = self.device

@@ -48,11 +48,26 @@ def gemm(
     act: int = 1,
     dact_tanh: bool = False,
     accum: bool = False,
+    a_perm: "Optional[dict]" = None,
 ) -> torch.Tensor:
+    """a_perm (gather+first-GEMM fusion): dict(seed, minibatches, ctr_off,
+    step_base, mb_ctr) — A rows are read through the epoch Feistel
+    permutation; A is the FULL rollout slab and C has minibatch rows."""
     if _use_native(A):
+        if a_perm is not None:
+            native.require().gemm(A, B, bias, C, Yact, trans_b, act,
+                                  dact_tanh, accum,
+                                  ap_seed=a_perm["seed"],
+                                  ap_minibatches=a_perm["minibatches"],
+                                  ap_ctr_off=a_perm.get("ctr_off", 0),
+                                  ap_step_base=a_perm["step_base"],
+                                  ap_mb_ctr=a_perm["mb_ctr"])
+            return C
         native.require().gemm(A, B, bias, C, Yact, trans_b, act, dact_tanh,
                               accum)
         return C
+    if a_perm is not None:
+        A = A[_perm_rows(A.shape[0], C.shape[0], a_perm)]
     a = A.to(torch.float32)
     b = B.to(torch.float32)
     if trans_b:
@@ -253,8 +268,10 @@ def wgrad(
     *,
     workspace: Optional[Tuple[torch.Tensor, Optional[torch.Tensor]]] = None,
     slabs: int = 64,
+    x_perm: "Optional[dict]" = None,
 ) -> None:
-    """dW[K,N] = X^T @ dY ; db[N] = colsum(dY). Deterministic split-M."""
+    """dW[K,N] = X^T @ dY ; db[N] = colsum(dY). Deterministic split-M.
+    x_perm: same contract as gemm's a_perm (X rows via the permutation)."""
     if _use_native(X):
         K, N = dW.shape
         if workspace is None:
@@ -266,8 +283,18 @@ def wgrad(
             )
         else:
             dW_part, db_part = workspace
+        if x_perm is not None:
+            native.require().wgrad(X, dY, dW_part, db_part, dW, db, slabs,
+                                   ap_seed=x_perm["seed"],
+                                   ap_minibatches=x_perm["minibatches"],
+                                   ap_ctr_off=x_perm.get("ctr_off", 0),
+                                   ap_step_base=x_perm["step_base"],
+                                   ap_mb_ctr=x_perm["mb_ctr"])
+            return
         native.require().wgrad(X, dY, dW_part, db_part, dW, db, slabs)
         return
+    if x_perm is not None:
+        X = X[_perm_rows(X.shape[0], dY.shape[0], x_perm)]
     x = X.to(torch.float32)
     dy = dY.to(torch.float32)
     dW.copy_(x.t() @ dy)
@@ -511,6 +538,16 @@ def feistel_key(seed: int, step_base: int, epoch: int) -> int:
     )
 
 
+def _perm_rows(n_rows: int, m_rows: int, perm: dict) -> torch.Tensor:
+    """CPU fallback of the a_perm row selection (same indices the fused
+    kernels compute on device)."""
+    ctr = int(perm["mb_ctr"].item()) + int(perm.get("ctr_off", 0))
+    minibatches = int(perm["minibatches"])
+    epoch, mb = ctr // minibatches, ctr % minibatches
+    key = feistel_key(int(perm["seed"]), int(perm["step_base"].item()), epoch)
+    return feistel_perm(n_rows, key)[mb * m_rows:(mb + 1) * m_rows]
+
+
 def feistel_perm(n: int, key: int) -> torch.Tensor:
     """dst -> src index map: 4-round balanced Feistel over the smallest
     even-bit power-of-two domain >= n, cycle-walked into [0, n).  CPU oracle
@@ -560,14 +597,18 @@ def mb_gather(
     minibatches: int,
     step_base: torch.Tensor,
     mb_ctr: torch.Tensor,
+    skip_obs: bool = False,
 ) -> None:
     """Gather minibatch ``mb_ctr % minibatches`` of epoch
     ``mb_ctr // minibatches`` under the Feistel permutation keyed by
-    (seed, *step_base, epoch).  One fused kernel on GPU."""
+    (seed, *step_base, epoch).  One fused kernel on GPU.  skip_obs: the
+    gather+first-GEMM fusion reads obs through the permutation itself, so
+    only the small per-row fields are copied."""
     if _use_native(obs_src):
         native.require().mb_gather(
             obs_src, act_src, logp_src, adv_src, ret_src, obs_mb, act_mb,
             logp_mb, adv_mb, ret_mb, seed, minibatches, step_base, mb_ctr,
+            skip_obs,
         )
         return
     n = obs_src.shape[0]
@@ -576,7 +617,8 @@ def mb_gather(
     epoch, mb = ctr // minibatches, ctr % minibatches
     key = feistel_key(seed, int(step_base.item()), epoch)
     src = feistel_perm(n, key)[mb * M:(mb + 1) * M]
-    obs_mb.copy_(obs_src[src])
+    if not skip_obs:
+        obs_mb.copy_(obs_src[src])
     act_mb.copy_(act_src[src])
     logp_mb.copy_(logp_src[src])
     adv_mb.copy_(adv_src[src])

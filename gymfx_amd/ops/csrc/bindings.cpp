@@ -22,7 +22,8 @@ void launch_env_step_obs(const EnvPtrs& P, const EnvParamsK& K, int env_lo,
                          int env_cnt, hipStream_t stream);
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
-                 bool dact_tanh, bool add_bias, bool accum, hipStream_t stream);
+                 bool dact_tanh, bool add_bias, bool accum, hipStream_t stream,
+                 const FeistelMap* fmp = nullptr);
 void launch_lstm_cell_fwd(const float* gates_pre, const void* gates_h,
                           const float* c_prev, float* c_new, void* h_new,
                           const bool* done, void* h_masked, float* c_masked,
@@ -65,7 +66,7 @@ void launch_mb_gather_seq(const void* obs_src, const int64_t* act_src,
                           hipStream_t stream);
 void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                   float* dW, float* db, int M, int N, int K, int slabs,
-                  hipStream_t stream);
+                  hipStream_t stream, const FeistelMap* fmp = nullptr);
 void launch_gae(const float* rewards, const float* values, const bool* dones,
                 float* adv, float* ret, int T, int N, float gamma, float lam,
                 hipStream_t stream);
@@ -104,7 +105,8 @@ void launch_mb_gather(const void* obs_src, const int64_t* act_src,
                       float* logp_mb, float* adv_mb, float* ret_mb, int M,
                       int D, uint32_t n, int half, uint64_t seed,
                       int minibatches, const unsigned long long* step_base,
-                      const unsigned long long* mb_ctr, hipStream_t stream);
+                      const unsigned long long* mb_ctr, hipStream_t stream,
+                      int skip_obs = 0);
 void launch_ppo_loss_bwd(const float* head, const int64_t* actions,
                          const float* old_logp, const float* adv,
                          const float* ret, void* dhead, int M, int n_actions,
@@ -485,9 +487,32 @@ void check_f32(const torch::Tensor& t, const char* name) {
               name, " must be contiguous f32");
 }
 
+// build a FeistelMap from the python-side dict-args of the gather fusion
+static gymfx::FeistelMap make_fmap(int64_t n_rows, int M_mb, int64_t seed,
+                                   int64_t minibatches, int64_t ctr_off,
+                                   torch::Tensor step_base,
+                                   torch::Tensor mb_ctr) {
+  gymfx::FeistelMap fm;
+  fm.n = (uint32_t)n_rows;
+  int bits = 2;
+  while ((1ll << bits) < n_rows) bits += 2;
+  fm.half = bits / 2;
+  fm.seed = (uint64_t)seed;
+  fm.minibatches = (int)minibatches;
+  fm.M_mb = M_mb;
+  fm.ctr_off = (long long)ctr_off;
+  fm.step_base = reinterpret_cast<const unsigned long long*>(
+      step_base.data_ptr());
+  fm.mb_ctr = reinterpret_cast<const unsigned long long*>(mb_ctr.data_ptr());
+  return fm;
+}
+
 void gemm_op(torch::Tensor A, torch::Tensor B, c10::optional<torch::Tensor> bias,
              torch::Tensor C, c10::optional<torch::Tensor> Yact, bool trans_b,
-             int64_t act, bool dact_tanh, bool accum) {
+             int64_t act, bool dact_tanh, bool accum,
+             int64_t ap_seed, int64_t ap_minibatches, int64_t ap_ctr_off,
+             c10::optional<torch::Tensor> ap_step_base,
+             c10::optional<torch::Tensor> ap_mb_ctr) {
   TORCH_CHECK(!accum || (act == 0 && !dact_tanh && !bias.has_value()),
               "accum only supported for plain f32-out gemm");
   check_bf16(A, "A");
@@ -495,7 +520,8 @@ void gemm_op(torch::Tensor A, torch::Tensor B, c10::optional<torch::Tensor> bias
   const int M = (int)A.size(0), K = (int)A.size(1);
   const int N = trans_b ? (int)B.size(0) : (int)B.size(1);
   TORCH_CHECK((trans_b ? B.size(1) : B.size(0)) == K, "gemm K mismatch");
-  TORCH_CHECK(C.size(0) == M && C.size(1) == N, "gemm C shape");
+  TORCH_CHECK(C.size(1) == N && (C.size(0) == M || ap_step_base.has_value()),
+              "gemm C shape");
   if (act == 0) check_f32(C, "C"); else check_bf16(C, "C");
   const float* bias_p = nullptr;
   if (bias.has_value()) {
@@ -508,6 +534,17 @@ void gemm_op(torch::Tensor A, torch::Tensor B, c10::optional<torch::Tensor> bias
     TORCH_CHECK(Yact.has_value(), "dact_tanh needs Yact");
     check_bf16(*Yact, "Yact");
     y_p = Yact->data_ptr();
+  }
+  if (ap_step_base.has_value()) {
+    TORCH_CHECK(ap_mb_ctr.has_value(), "a_perm needs mb_ctr");
+    const int M_mb = (int)C.size(0);
+    gymfx::FeistelMap fm = make_fmap(A.size(0), M_mb, ap_seed,
+                                     ap_minibatches, ap_ctr_off,
+                                     *ap_step_base, *ap_mb_ctr);
+    gymfx::launch_gemm(A.data_ptr(), B.data_ptr(), bias_p, C.data_ptr(), y_p,
+                       M_mb, N, K, trans_b, (int)act, dact_tanh,
+                       bias_p != nullptr, accum, cur_stream(), &fm);
+    return;
   }
   gymfx::launch_gemm(A.data_ptr(), B.data_ptr(), bias_p, C.data_ptr(), y_p, M,
                      N, K, trans_b, (int)act, dact_tanh, bias_p != nullptr,
@@ -726,14 +763,23 @@ void mb_gather_seq_op(torch::Tensor obs_src, torch::Tensor act_src,
       cur_stream());
 }
 
+
+
 void wgrad_op(torch::Tensor X, torch::Tensor dY, torch::Tensor dW_part,
               c10::optional<torch::Tensor> db_part, torch::Tensor dW,
-              c10::optional<torch::Tensor> db, int64_t slabs) {
+              c10::optional<torch::Tensor> db, int64_t slabs,
+              int64_t ap_seed, int64_t ap_minibatches, int64_t ap_ctr_off,
+              c10::optional<torch::Tensor> ap_step_base,
+              c10::optional<torch::Tensor> ap_mb_ctr) {
   check_bf16(X, "X");
   check_bf16(dY, "dY");
   check_f32(dW, "dW");
   check_f32(dW_part, "dW_part");
-  const int M = (int)X.size(0), K = (int)X.size(1), N = (int)dY.size(1);
+  const bool aperm = ap_step_base.has_value();
+  // gather fusion: X is the FULL rollout slab, rows selected through the
+  // permutation; the logical reduction length is the minibatch (dY rows)
+  const int M = aperm ? (int)dY.size(0) : (int)X.size(0);
+  const int K = (int)X.size(1), N = (int)dY.size(1);
   TORCH_CHECK(dY.size(0) == M, "wgrad M mismatch");
   TORCH_CHECK(dW.size(0) == K && dW.size(1) == N, "dW shape");
   TORCH_CHECK(dW_part.numel() >= slabs * (int64_t)K * N, "dW_part too small");
@@ -744,6 +790,15 @@ void wgrad_op(torch::Tensor X, torch::Tensor dY, torch::Tensor dW_part,
     check_f32(*db, "db");
     dbp = db_part->data_ptr<float>();
     dbo = db->data_ptr<float>();
+  }
+  if (aperm) {
+    TORCH_CHECK(ap_mb_ctr.has_value(), "a_perm needs mb_ctr");
+    gymfx::FeistelMap fm = make_fmap(X.size(0), M, ap_seed, ap_minibatches,
+                                     ap_ctr_off, *ap_step_base, *ap_mb_ctr);
+    gymfx::launch_wgrad(X.data_ptr(), dY.data_ptr(),
+                        dW_part.data_ptr<float>(), dbp, dW.data_ptr<float>(),
+                        dbo, M, N, K, (int)slabs, cur_stream(), &fm);
+    return;
   }
   gymfx::launch_wgrad(X.data_ptr(), dY.data_ptr(), dW_part.data_ptr<float>(),
                       dbp, dW.data_ptr<float>(), dbo, M, N, K, (int)slabs,
@@ -897,7 +952,7 @@ void mb_gather_op(torch::Tensor obs_src, torch::Tensor act_src,
                   torch::Tensor act_mb, torch::Tensor logp_mb,
                   torch::Tensor adv_mb, torch::Tensor ret_mb, int64_t seed,
                   int64_t minibatches, torch::Tensor step_base,
-                  torch::Tensor mb_ctr) {
+                  torch::Tensor mb_ctr, bool skip_obs) {
   check_bf16(obs_src, "obs_src");
   check_bf16(obs_mb, "obs_mb");
   const int64_t n = obs_src.size(0);
@@ -923,7 +978,7 @@ void mb_gather_op(torch::Tensor obs_src, torch::Tensor act_src,
       bits / 2, (uint64_t)seed, (int)minibatches,
       reinterpret_cast<const unsigned long long*>(step_base.data_ptr()),
       reinterpret_cast<const unsigned long long*>(mb_ctr.data_ptr()),
-      cur_stream());
+      cur_stream(), skip_obs ? 1 : 0);
 }
 
 void ppo_loss_bwd_op(torch::Tensor head, torch::Tensor actions,
@@ -976,7 +1031,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("bias"), py::arg("C"),
         py::arg("Yact") = py::none(), py::arg("trans_b") = false,
         py::arg("act") = 1, py::arg("dact_tanh") = false,
-        py::arg("accum") = false);
+        py::arg("accum") = false, py::arg("ap_seed") = 0,
+        py::arg("ap_minibatches") = 1, py::arg("ap_ctr_off") = 0,
+        py::arg("ap_step_base") = py::none(),
+        py::arg("ap_mb_ctr") = py::none());
   m.def("lstm_cell_fwd", &lstm_cell_fwd_op, py::arg("gates_pre"),
         py::arg("gates_h"), py::arg("c_prev"), py::arg("c_new"),
         py::arg("h_new"), py::arg("done") = py::none(),
@@ -1014,7 +1072,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("c0_mb"), py::arg("L"), py::arg("N"), py::arg("seed"),
         py::arg("minibatches"), py::arg("step_base"), py::arg("mb_ctr"));
   m.def("wgrad", &wgrad_op, py::arg("X"), py::arg("dY"), py::arg("dW_part"),
-        py::arg("db_part"), py::arg("dW"), py::arg("db"), py::arg("slabs"));
+        py::arg("db_part"), py::arg("dW"), py::arg("db"), py::arg("slabs"),
+        py::arg("ap_seed") = 0, py::arg("ap_minibatches") = 1,
+        py::arg("ap_ctr_off") = 0, py::arg("ap_step_base") = py::none(),
+        py::arg("ap_mb_ctr") = py::none());
   m.def("gae", &gae_op);
   m.def("adam", &adam_op, py::arg("p"), py::arg("g"), py::arg("m"),
         py::arg("v"), py::arg("p_bf16"), py::arg("lr"), py::arg("beta1"),
@@ -1039,7 +1100,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("logp_src"), py::arg("adv_src"), py::arg("ret_src"),
         py::arg("obs_mb"), py::arg("act_mb"), py::arg("logp_mb"),
         py::arg("adv_mb"), py::arg("ret_mb"), py::arg("seed"),
-        py::arg("minibatches"), py::arg("step_base"), py::arg("mb_ctr"));
+        py::arg("minibatches"), py::arg("step_base"), py::arg("mb_ctr"), py::arg("skip_obs") = false);
   m.def("ppo_loss_bwd", &ppo_loss_bwd_op, py::arg("head"), py::arg("actions"),
         py::arg("old_logp"), py::arg("adv"), py::arg("ret"), py::arg("dhead"),
         py::arg("clip_eps"), py::arg("ent_coef"), py::arg("vf_coef"),

@@ -119,6 +119,20 @@ GFX_DEV int sample_categorical(const float* row, int n_actions, float logz,
 }
 #endif  // __HIPCC__
 
+// Row-permutation descriptor for the gather+first-GEMM fusion (the L1
+// forward GEMM / W1 wgrad read rollout rows through the epoch Feistel
+// permutation instead of a materialized minibatch copy).
+struct FeistelMap {
+  uint32_t n = 0;   // permutation domain (total rollout rows)
+  int half = 0;
+  uint64_t seed = 0;
+  int minibatches = 1;
+  int M_mb = 0;     // rows per minibatch
+  long long ctr_off = 0;
+  const unsigned long long* step_base = nullptr;
+  const unsigned long long* mb_ctr = nullptr;
+};
+
 struct EnvParamsK {
   int n_envs, T, window, n_features;
   int reward_id, strategy_id, prep_id;

@@ -74,17 +74,7 @@ GFX_DEV uint32_t feistel_perm_idx(uint32_t i, uint32_t n, int half,
 // saves one full write+read pass over the minibatch observations).
 // ctr_off: the gather that produced the OTHER minibatch fields has already
 // advanced mb_ctr when these kernels run (-1 in the trainer's sequence).
-struct FeistelMap {
-  uint32_t n;  // permutation domain (total rollout rows)
-  int half;
-  uint64_t seed;
-  int minibatches;
-  int M_mb;    // rows per minibatch
-  long long ctr_off;
-  const unsigned long long* step_base;
-  const unsigned long long* mb_ctr;
-};
-
+// (FeistelMap struct lives in env_common.h — host code builds it.)
 GFX_DEV uint64_t feistel_key(const FeistelMap& fm, uint32_t* mb) {
   const unsigned long long ctr =
       (unsigned long long)((long long)*fm.mb_ctr + fm.ctr_off);
@@ -370,11 +360,11 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
 // m-chunk; the launcher falls back to the register-staged kernel for edge
 // tiles (K%64 tails, the head layer's N=4, partial slabs).
 // ---------------------------------------------------------------------------
-template <bool WANT_DB, int FK, int FN>
+template <bool WANT_DB, int FK, int FN, bool A_PERM = false>
 __global__ __launch_bounds__(256) void wgrad_glds_kernel(
     const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
     float* __restrict__ dW_part, float* __restrict__ db_part,
-    int M, int N, int K, int slabs) {
+    int M, int N, int K, int slabs, FeistelMap fm = FeistelMap{}) {
   constexpr int BKm = 64;
   constexpr int TK = 32 * FK, TN = 32 * FN;
   // UNPADDED linear images (glds cannot scatter past a row pad)
@@ -396,6 +386,14 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
   f32x4 acc[FK][FN] = {};
   float db_acc = 0.f;
 
+  // gather+wgrad fusion: X rows go through the epoch permutation
+  uint64_t fkey = 0;
+  uint32_t fmb = 0;
+  if (A_PERM) fkey = feistel_key(fm, &fmb);
+  auto x_row = [&](int gm) -> int64_t {
+    return A_PERM ? feistel_src_row(fm, fkey, fmb, gm) : (int64_t)gm;
+  };
+
   // one glds instruction stages 64 lanes x 16B = 1KB = 512 bf16; a 64xTW
   // chunk is 64*TW*2 B = TW/8 instructions = TW/32 per wave.
   auto glds_chunk = [&](int buf, int m0) {
@@ -405,7 +403,7 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
       const int col = (e0 + lane * 8) % TK;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)(uintptr_t)(
-              &X[(int64_t)(m0 + row) * K + bk + col]),
+              &X[x_row(m0 + row) * K + bk + col]),
           (__attribute__((address_space(3))) void*)(uintptr_t)(
               &Xs[buf][0][0] + e0),
           16, 0, 0);
@@ -472,11 +470,11 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
     db_part[(int64_t)slab * N + bn + tid] = db_acc;
 }
 
-template <bool WANT_DB, int FK, int FN>
+template <bool WANT_DB, int FK, int FN, bool A_PERM = false>
 __global__ __launch_bounds__(256) void wgrad_partial_kernel(
     const __bf16* __restrict__ X, const __bf16* __restrict__ dY,
     float* __restrict__ dW_part, float* __restrict__ db_part,
-    int M, int N, int K, int slabs, int kt0) {
+    int M, int N, int K, int slabs, int kt0, FeistelMap fm = FeistelMap{}) {
   // FK/FN = MFMA fragments per wave along K/N: block tile (32*FK) x (32*FN)
   // over the dW output; reduction dim is M, chunked BKm=32 rows at a time
   // with double-buffered natural-layout LDS staging (coalesced bf16x8
@@ -501,6 +499,10 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
   f32x4 acc[FK][FN] = {};
   float db_acc = 0.f;
 
+  uint64_t fkey = 0;
+  uint32_t fmb = 0;
+  if (A_PERM) fkey = feistel_key(fm, &fmb);
+
   bf16x8 rx[FK / 2], ry[FN / 2];
   auto load_chunk = [&](int m0) {
     // X chunk [32 m][TK k]: 32*TK elems = 256 threads * (FK/2) vec8
@@ -511,12 +513,14 @@ __global__ __launch_bounds__(256) void wgrad_partial_kernel(
       rx[j] = bf16x8{};
       const int gm = m0 + m;
       if (gm < m_end) {
+        const int64_t sm = A_PERM ? feistel_src_row(fm, fkey, fmb, gm)
+                                  : (int64_t)gm;
         const int gk = bk + c8;
         if (gk + 8 <= K) {
-          rx[j] = *reinterpret_cast<const bf16x8*>(&X[(int64_t)gm * K + gk]);
+          rx[j] = *reinterpret_cast<const bf16x8*>(&X[sm * K + gk]);
         } else {
           for (int i = 0; i < 8; ++i)
-            rx[j][i] = (gk + i < K) ? X[(int64_t)gm * K + gk + i] : (__bf16)0.f;
+            rx[j][i] = (gk + i < K) ? X[sm * K + gk + i] : (__bf16)0.f;
         }
       }
     }
@@ -1654,7 +1658,7 @@ __global__ __launch_bounds__(256) void mb_gather_kernel(
     float* __restrict__ adv_mb, float* __restrict__ ret_mb, int M, int D,
     uint32_t n, int half, uint64_t seed, int minibatches,
     const unsigned long long* __restrict__ step_base,
-    const unsigned long long* __restrict__ mb_ctr) {
+    const unsigned long long* __restrict__ mb_ctr, int skip_obs) {
   const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   if (row >= M) return;
@@ -1668,7 +1672,10 @@ __global__ __launch_bounds__(256) void mb_gather_kernel(
 
   const __bf16* src_row = obs_src + (int64_t)src * D;
   __bf16* dst_row = obs_mb + (int64_t)row * D;
-  if ((D & 3) == 0) {
+  if (skip_obs) {
+    // gather+first-GEMM fusion: the L1 GEMM / W1 wgrad read obs rows
+    // through the permutation themselves — only the small fields copy
+  } else if ((D & 3) == 0) {
     // 8-byte chunks (bf16 x4)
     const int chunks = D >> 2;
     const uint64_t* s64 = reinterpret_cast<const uint64_t*>(src_row);
@@ -1924,19 +1931,20 @@ void launch_mb_gather(const void* obs_src, const int64_t* act_src,
                       float* logp_mb, float* adv_mb, float* ret_mb, int M,
                       int D, uint32_t n, int half, uint64_t seed,
                       int minibatches, const unsigned long long* step_base,
-                      const unsigned long long* mb_ctr, hipStream_t stream) {
+                      const unsigned long long* mb_ctr, hipStream_t stream,
+                      int skip_obs) {
   hipLaunchKernelGGL(mb_gather_kernel, dim3(ceil_div(M, 4)), dim3(256), 0,
                      stream, reinterpret_cast<const __bf16*>(obs_src), act_src,
                      logp_src, adv_src, ret_src,
                      reinterpret_cast<__bf16*>(obs_mb), act_mb, logp_mb, adv_mb,
                      ret_mb, M, D, n, half, seed, minibatches, step_base,
-                     mb_ctr);
+                     mb_ctr, skip_obs);
 }
 
 void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
                  bool dact_tanh, bool add_bias, bool accum,
-                 hipStream_t stream) {
+                 hipStream_t stream, const FeistelMap* fmp = nullptr) {
   const __bf16* a = reinterpret_cast<const __bf16*>(A);
   const __bf16* b = reinterpret_cast<const __bf16*>(B);
   const __bf16* y = reinterpret_cast<const __bf16*>(Yact);
@@ -1965,6 +1973,19 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                     (bk64_env == 1 || (bk64_env != 0 && K >= 512));
   dim3 grid(ceil_div(M, 64), ceil_div(N, wide ? 256 : (mid ? 128 : 64)));
   dim3 block(256);
+  if (fmp) {
+    // gather+first-GEMM fusion: only the L1 forward combo is instantiated
+    // (trans_b, tanh epilogue, bias, 64x64 tile)
+    if (!(trans_b && act == 2 && !dact_tanh && add_bias && !accum)) {
+      // unreachable by construction (models/mlp.py passes a_feistel only
+      // on the first layer); abort loudly if it ever is
+      abort();
+    }
+    hipLaunchKernelGGL((gemm_kernel<true, 2, false, true, 2, false, 32, true>),
+                       grid, block, 0, stream, a, b, bias, C, y, M, N, K,
+                       *fmp);
+    return;
+  }
 
 #define GEMM_LAUNCH(TB, ACT, DT, AB)                                          \
   do {                                                                        \
@@ -2011,7 +2032,7 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
 
 void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                   float* dW, float* db, int M, int N, int K, int slabs,
-                  hipStream_t stream) {
+                  hipStream_t stream, const FeistelMap* fmp = nullptr) {
   const __bf16* x = reinterpret_cast<const __bf16*>(X);
   const __bf16* dy = reinterpret_cast<const __bf16*>(dY);
   // Tile choice: 128x128 halves the HBM re-streaming of X/dY — wgrad runs
@@ -2038,21 +2059,24 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
   if (glds_ok) {
     const int kfull = K / TW;
     dim3 g0(kfull, N / TW, slabs);
+#define WG_GLDS(WDB, FK, FN)                                                  \
+  do {                                                                        \
+    if (fmp)                                                                  \
+      hipLaunchKernelGGL((wgrad_glds_kernel<WDB, FK, FN, true>), g0,          \
+                         dim3(256), 0, stream, x, dy, dW_part, db_part, M, N, \
+                         K, slabs, *fmp);                                     \
+    else                                                                      \
+      hipLaunchKernelGGL((wgrad_glds_kernel<WDB, FK, FN>), g0, dim3(256), 0,  \
+                         stream, x, dy, dW_part, db_part, M, N, K, slabs);    \
+  } while (0)
     if (big) {
-      if (db_part)
-        hipLaunchKernelGGL((wgrad_glds_kernel<true, 4, 4>), g0, dim3(256), 0,
-                           stream, x, dy, dW_part, db_part, M, N, K, slabs);
-      else
-        hipLaunchKernelGGL((wgrad_glds_kernel<false, 4, 4>), g0, dim3(256), 0,
-                           stream, x, dy, dW_part, db_part, M, N, K, slabs);
+      if (db_part) WG_GLDS(true, 4, 4);
+      else WG_GLDS(false, 4, 4);
     } else {
-      if (db_part)
-        hipLaunchKernelGGL((wgrad_glds_kernel<true, 2, 2>), g0, dim3(256), 0,
-                           stream, x, dy, dW_part, db_part, M, N, K, slabs);
-      else
-        hipLaunchKernelGGL((wgrad_glds_kernel<false, 2, 2>), g0, dim3(256), 0,
-                           stream, x, dy, dW_part, db_part, M, N, K, slabs);
+      if (db_part) WG_GLDS(true, 2, 2);
+      else WG_GLDS(false, 2, 2);
     }
+#undef WG_GLDS
     if (K % TW) {
       // K tail (obs_dim=260 -> 4 leftover dW rows): one MFMA tile-column
       // of the register-staged kernel at tile offset kfull*TW/64.  The
@@ -2062,27 +2086,36 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
       // dead K columns.
       // tail width < TW but can span up to two 64-tiles when TW=128
       dim3 g1(ceil_div(K - kfull * TW, 64), ceil_div(N, 64), slabs);
-      hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), g1, dim3(256),
-                         0, stream, x, dy, dW_part, nullptr, M, N, K, slabs,
-                         kfull * TW / 64);
+      if (fmp)
+        hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2, true>), g1,
+                           dim3(256), 0, stream, x, dy, dW_part, nullptr, M,
+                           N, K, slabs, kfull * TW / 64, *fmp);
+      else
+        hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), g1, dim3(256),
+                           0, stream, x, dy, dW_part, nullptr, M, N, K, slabs,
+                           kfull * TW / 64);
     }
   } else {
     dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);
+#define WG_PART(WDB, FK, FN)                                                  \
+  do {                                                                        \
+    if (fmp)                                                                  \
+      hipLaunchKernelGGL((wgrad_partial_kernel<WDB, FK, FN, true>), grid,     \
+                         dim3(256), 0, stream, x, dy, dW_part, db_part, M, N, \
+                         K, slabs, 0, *fmp);                                  \
+    else                                                                      \
+      hipLaunchKernelGGL((wgrad_partial_kernel<WDB, FK, FN>), grid,           \
+                         dim3(256), 0, stream, x, dy, dW_part, db_part, M, N, \
+                         K, slabs, 0);                                        \
+  } while (0)
     if (db_part) {
-      if (big)
-        hipLaunchKernelGGL((wgrad_partial_kernel<true, 4, 4>), grid, dim3(256),
-                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs, 0);
-      else
-        hipLaunchKernelGGL((wgrad_partial_kernel<true, 2, 2>), grid, dim3(256),
-                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs, 0);
+      if (big) WG_PART(true, 4, 4);
+      else WG_PART(true, 2, 2);
     } else {
-      if (big)
-        hipLaunchKernelGGL((wgrad_partial_kernel<false, 4, 4>), grid, dim3(256),
-                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs, 0);
-      else
-        hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), grid, dim3(256),
-                           0, stream, x, dy, dW_part, db_part, M, N, K, slabs, 0);
+      if (big) WG_PART(false, 4, 4);
+      else WG_PART(false, 2, 2);
     }
+#undef WG_PART
   }
   int64_t elems = (int64_t)K * N;
   if (elems <= 4096) {

@@ -276,3 +276,36 @@ def test_trace_file_on_gpu(tmp_path):
     for rec in lines:
         assert rec["phases_ms"]["rollout"] > 0
         assert rec["phases_ms"]["update"] > 0
+
+
+def test_fuse_gather_equals_materialized():
+    """Gather+first-GEMM fusion (L1 fwd and W1 wgrad read obs rows through
+    the epoch permutation) must be BITWISE identical to the materialized
+    obs_mb path: identical bf16 rows, identical MFMA K order."""
+    from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    def make(fuse):
+        md = synthetic_ohlcv(2000, seed=9, vol=4e-4)
+        cfg = {"n_envs": 256, "device": "cuda", "window_size": 16,
+               "env_start_mode": "spread", "autoreset": True,
+               "position_size": 1000.0, "seed": 31}
+        env = build_vec_environment(cfg, md)
+        env.reset(seed=31)
+        pc = PPOConfig(rollout_steps=16, minibatches=4, ppo_epochs=2, seed=31,
+                       hidden=64)
+        t = PPOTrainer(env, pc)
+        t._fuse_gather = fuse
+        return t
+
+    tf = make(True)
+    tu = make(False)
+    for _ in range(3):
+        tf.train_update(with_stats=False)
+        tu.train_update(with_stats=False)
+    torch.cuda.synchronize()
+    assert torch.equal(tf.model.params, tu.model.params)
+    assert torch.equal(tf.model.grads, tu.model.grads)
+    assert torch.equal(tf.act_buf, tu.act_buf)
+    assert torch.equal(tf.env.st.equity, tu.env.st.equity)
