@@ -232,14 +232,17 @@ class PPOTrainer:
         self.sample_seed = cfg.seed * 1_000_003 + rank
         self.shuffle_seed = cfg.seed * 9973 + rank
 
-        # gather+first-GEMM fusion (ROADMAP lever 3): L1 fwd and the W1
-        # wgrad read obs rows through the epoch permutation, so the gather
-        # kernel skips its biggest copy.  A/B: GYMFX_FUSE_GATHER=0
+        # gather+first-GEMM fusion (ROADMAP lever 3): MEASURED NEGATIVE at
+        # the flagship shape — update 12.0 -> 18.0 ms: the per-lane Feistel
+        # recompute in every staging loop of L1-fwd and W1-wgrad costs far
+        # more than the one materialized 36 us gather pass it removes
+        # (profiles/PERF_NOTES.md).  Opt-in via GYMFX_FUSE_GATHER=1; the
+        # bitwise-equality GPU test keeps the path honest.
         import os as _os
         self._fuse_gather = (
             not self.recurrent and self.device.type == "cuda"
             and cfg.shuffle_rows and not self._overlap
-            and _os.environ.get("GYMFX_FUSE_GATHER", "1") != "0")
+            and _os.environ.get("GYMFX_FUSE_GATHER", "0") == "1")
         self._fm = {"seed": self.shuffle_seed, "minibatches": cfg.minibatches,
                     "ctr_off": -1, "step_base": self.step_base,
                     "mb_ctr": self.mb_ctr}
