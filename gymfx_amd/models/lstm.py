@@ -153,7 +153,10 @@ class ActorCriticLSTM:
     def alloc_acts(self, M: int) -> Dict[str, torch.Tensor]:
         dev, H = self.device, self.hidden
         return {
-            "gates": torch.empty(M, 4 * H, dtype=torch.float32, device=dev),
+            # gates_pre stored bf16 (like gates_h and h): the f32 original
+            # cost 268 MB of write traffic per minibatch on the Wx GEMM and
+            # doubled every cell read; the cell still sums gx+gh in f32
+            "gates": torch.empty(M, 4 * H, dtype=torch.bfloat16, device=dev),
             "gates_h": torch.empty(M, 4 * H, dtype=torch.bfloat16, device=dev),
             "head": torch.empty(M, self.head_dim, dtype=torch.float32, device=dev),
             # fused-step h output (the fused GEMM+cell kernel must not write
@@ -171,7 +174,7 @@ class ActorCriticLSTM:
         callers must read state['h'] through the dict), returns head
         [M, A+1] f32."""
         gates = acts["gates"]
-        api.gemm(obs_bf16, self.wt("Wx"), self.f32("b"), gates, act=0, trans_b=True)
+        api.gemm(obs_bf16, self.wt("Wx"), self.f32("b"), gates, act=1, trans_b=True)
         if fused and "h_tmp" in acts:
             # recurrent GEMM + cell in one kernel; rollout never reads
             # gates_h back, so its global write is skipped on the GPU path
@@ -201,7 +204,7 @@ class ActorCriticLSTM:
             # raw outputs of step l (head input; cell-bwd c_new)
             "h_raw": torch.empty(L, M, H, dtype=torch.bfloat16, device=dev),
             "c_raw": torch.empty(L, M, H, dtype=torch.float32, device=dev),
-            "gates": torch.empty(L, M, 4 * H, dtype=torch.float32, device=dev),
+            "gates": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
             "gates_h": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
             "head": torch.empty(L * M, self.head_dim, dtype=torch.float32, device=dev),
             "dgates": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
@@ -224,7 +227,7 @@ class ActorCriticLSTM:
         # x-projection for ALL timesteps in one GEMM
         gates_flat = buf["gates"].view(L * M, 4 * H)
         api.gemm(obs_seq.view(L * M, D), self.wt("Wx"), self.f32("b"),
-                 gates_flat, act=0, trans_b=True)
+                 gates_flat, act=1, trans_b=True)
         buf["h_in"][0].copy_(h0)
         buf["c_in"][0].copy_(c0)
         for l in range(L):

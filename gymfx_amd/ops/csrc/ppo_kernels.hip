@@ -394,6 +394,13 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
     return A_PERM ? feistel_src_row(fm, fkey, fmb, gm) : (int64_t)gm;
   };
 
+  // partial-K final tile (e.g. obs_dim=260 with TK=160: second tile covers
+  // 100 live columns): X staged with guarded register loads instead of
+  // glds — the K TAIL no longer needs a second full pass over dY
+  // (profiles/lstm_kernel_stats_r2b.txt: the tail kernel re-read all of
+  // dgates for 4 columns, ~1.5 ms/update).
+  const bool k_tail = bk + TK > K;
+
   // one glds instruction stages 64 lanes x 16B = 1KB = 512 bf16; a 64xTW
   // chunk is 64*TW*2 B = TW/8 instructions = TW/32 per wave.
   auto glds_chunk = [&](int buf, int m0) {
@@ -401,6 +408,19 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
       const int e0 = (wave * FK + j) * 512;
       const int row = (e0 + lane * 8) / TK;
       const int col = (e0 + lane * 8) % TK;
+      if (k_tail) {
+        bf16x8 v = bf16x8{};
+        const int gk = bk + col;
+        const int64_t xr = x_row(m0 + row) * K;
+        if (gk + 8 <= K) {
+          v = *reinterpret_cast<const bf16x8*>(&X[xr + gk]);
+        } else {
+          for (int i = 0; i < 8; ++i)
+            v[i] = (gk + i < K) ? X[xr + gk + i] : (__bf16)0.f;
+        }
+        *reinterpret_cast<bf16x8*>(&Xs[buf][0][0] + e0 + lane * 8) = v;
+        continue;
+      }
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)(uintptr_t)(
               &X[x_row(m0 + row) * K + bk + col]),
@@ -463,7 +483,7 @@ __global__ __launch_bounds__(256) void wgrad_glds_kernel(
       const int gcol = bn + wc * (16 * FN) + ni * 16 + ccol;
       for (int r = 0; r < 4; ++r) {
         const int grow = bk + wr * (16 * FK) + fi * 16 + crow_base + r;
-        out[(int64_t)grow * N + gcol] = acc[fi][ni][r];
+        if (grow < K) out[(int64_t)grow * N + gcol] = acc[fi][ni][r];
       }
     }
   if (WANT_DB && tid < TN && blockIdx.x == 0)
@@ -1010,9 +1030,12 @@ GFX_DEV float fast_sigmoid(float x) {
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
 // 4-wide interleaved gate pre-activations of hidden unit k for row m
-GFX_DEV f32x4 gate_pre4(const float* __restrict__ gates_pre,
+// (gates_pre is bf16 like gates_h; the sum runs in f32)
+GFX_DEV f32x4 gate_pre4(const __bf16* __restrict__ gates_pre,
                         const __bf16* __restrict__ gates_h, int64_t idx4) {
-  f32x4 v = *reinterpret_cast<const f32x4*>(&gates_pre[idx4]);
+  const bf16x4 gx = *reinterpret_cast<const bf16x4*>(&gates_pre[idx4]);
+  f32x4 v;
+  for (int j = 0; j < 4; ++j) v[j] = bf2f(gx[j]);
   if (gates_h) {
     const bf16x4 gh = *reinterpret_cast<const bf16x4*>(&gates_h[idx4]);
     for (int j = 0; j < 4; ++j) v[j] += bf2f(gh[j]);
@@ -1021,7 +1044,7 @@ GFX_DEV f32x4 gate_pre4(const float* __restrict__ gates_pre,
 }
 
 __global__ void lstm_cell_fwd_kernel(
-    const float* __restrict__ gates_pre,  // [M, 4H] (x-projection + bias)
+    const __bf16* __restrict__ gates_pre,  // [M, 4H] (x-projection + bias)
     const __bf16* __restrict__ gates_h,   // [M, 4H] or null (h-projection)
     const float* __restrict__ c_prev,     // [M, H]
     float* __restrict__ c_new,            // [M, H]
@@ -1057,7 +1080,7 @@ __global__ void lstm_cell_fwd_kernel(
 // boundary AFTER this step) along with dc_next, so no gradient crosses a
 // reset.
 __global__ void lstm_cell_bwd_kernel(
-    const float* __restrict__ gates_pre,  // [M, 4H] (saved fwd, x part)
+    const __bf16* __restrict__ gates_pre,  // [M, 4H] (saved fwd, x part)
     const __bf16* __restrict__ gates_h,   // [M, 4H] or null (saved h part)
     const float* __restrict__ c_prev,     // [M, H] (masked input c)
     const float* __restrict__ c_new,      // [M, H] (raw output c)
@@ -1112,7 +1135,7 @@ template <bool WRITE_GH>
 __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
     const __bf16* __restrict__ A,         // h_in [M, K=H]
     const __bf16* __restrict__ B,         // Wh^T [N=4H, K=H]
-    const float* __restrict__ gates_pre,  // [M, 4H] (x-proj + bias)
+    const __bf16* __restrict__ gates_pre,  // [M, 4H] (x-proj + bias)
     __bf16* __restrict__ gates_h,         // [M, 4H] out (if WRITE_GH)
     const float* __restrict__ c_prev,     // [M, H]
     float* __restrict__ c_new,            // [M, H]
@@ -1251,8 +1274,10 @@ __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
       for (int j = 0; j < 4; ++j) ghb[j] = f2bf(gh[j]);
       *reinterpret_cast<bf16x4*>(&gates_h[(int64_t)grow * N + (int64_t)(bn + ul * 4)]) = ghb;
     }
-    const f32x4 gx = *reinterpret_cast<const f32x4*>(
+    const bf16x4 gxb = *reinterpret_cast<const bf16x4*>(
         &gates_pre[(int64_t)grow * N + (int64_t)(bn + ul * 4)]);
+    f32x4 gx;
+    for (int j = 0; j < 4; ++j) gx[j] = bf2f(gxb[j]);
     const float i = fast_sigmoid(gx[0] + gh[0]);
     const float f = fast_sigmoid(gx[1] + gh[1]);
     const float g = fast_tanh(gx[2] + gh[2]);
@@ -1283,7 +1308,7 @@ __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
 // ---------------------------------------------------------------------------
 template <int NT>
 __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
-    const float* __restrict__ gates_pre,  // [M, 4H]
+    const __bf16* __restrict__ gates_pre,  // [M, 4H]
     const __bf16* __restrict__ gates_h,   // [M, 4H] or null
     const float* __restrict__ c_prev,     // [M, H]
     const float* __restrict__ c_new,      // [M, H]
@@ -1770,7 +1795,7 @@ __global__ __launch_bounds__(256) void mb_gather_seq_kernel(
 // ---------------------------------------------------------------------------
 static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
-void launch_lstm_cell_fwd(const float* gates_pre, const void* gates_h,
+void launch_lstm_cell_fwd(const __bf16* gates_pre, const void* gates_h,
                           const float* c_prev, float* c_new, void* h_new,
                           const bool* done, void* h_masked, float* c_masked,
                           int64_t M, int H, hipStream_t stream) {
@@ -1782,7 +1807,7 @@ void launch_lstm_cell_fwd(const float* gates_pre, const void* gates_h,
                      reinterpret_cast<__bf16*>(h_masked), c_masked, M, H);
 }
 
-void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
+void launch_lstm_cell_bwd(const __bf16* gates_pre, const void* gates_h,
                           const float* c_prev,
                           const float* c_new, const float* dh_head,
                           const float* dh_next, const float* dc_next,
@@ -1797,7 +1822,7 @@ void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
 }
 
 bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
-                               const float* gates_pre, void* gates_h,
+                               const __bf16* gates_pre, void* gates_h,
                                const float* c_prev, float* c_new, void* h_new,
                                const bool* done, void* h_masked,
                                float* c_masked, int M, int N, int K,
@@ -1821,7 +1846,7 @@ bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
   return true;
 }
 
-bool launch_lstm_bwd_fused(const float* gates_pre, const void* gates_h,
+bool launch_lstm_bwd_fused(const __bf16* gates_pre, const void* gates_h,
                            const float* c_prev, const float* c_new,
                            const float* dh_head, const float* dh_next,
                            const float* dc_next, const bool* done,
@@ -2056,45 +2081,40 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
   const int TW = big ? 128 : 64;
   const bool glds_ok = N % TW == 0 && K >= TW && m_per_slab % 64 == 0 &&
                        M % slabs == 0;
+  (void)0;
   if (glds_ok) {
-    const int kfull = K / TW;
-    dim3 g0(kfull, N / TW, slabs);
-#define WG_GLDS(WDB, FK, FN)                                                  \
+    // the glds kernel masks a partial final K tile in-kernel (guarded
+    // register staging + grow<K epilogue guard), so NO separate tail pass
+    // over dY is needed.  For K=260 the 5x4 tile (TK=160) covers K in TWO
+    // passes where 128-wide tiles need three — one fewer full dY read
+    // (~1.5 ms/update on the LSTM wgrads).
+#define WG_GLDS(WDB, FK, FN, GRID)                                            \
   do {                                                                        \
     if (fmp)                                                                  \
-      hipLaunchKernelGGL((wgrad_glds_kernel<WDB, FK, FN, true>), g0,          \
+      hipLaunchKernelGGL((wgrad_glds_kernel<WDB, FK, FN, true>), GRID,        \
                          dim3(256), 0, stream, x, dy, dW_part, db_part, M, N, \
                          K, slabs, *fmp);                                     \
     else                                                                      \
-      hipLaunchKernelGGL((wgrad_glds_kernel<WDB, FK, FN>), g0, dim3(256), 0,  \
-                         stream, x, dy, dW_part, db_part, M, N, K, slabs);    \
+      hipLaunchKernelGGL((wgrad_glds_kernel<WDB, FK, FN>), GRID, dim3(256),   \
+                         0, stream, x, dy, dW_part, db_part, M, N, K, slabs); \
   } while (0)
     if (big) {
-      if (db_part) WG_GLDS(true, 4, 4);
-      else WG_GLDS(false, 4, 4);
+      const int t44 = ceil_div(K, 128), t54 = ceil_div(K, 160);
+      if (t54 < t44) {
+        dim3 g0(t54, N / TW, slabs);
+        if (db_part) WG_GLDS(true, 5, 4, g0);
+        else WG_GLDS(false, 5, 4, g0);
+      } else {
+        dim3 g0(t44, N / TW, slabs);
+        if (db_part) WG_GLDS(true, 4, 4, g0);
+        else WG_GLDS(false, 4, 4, g0);
+      }
     } else {
-      if (db_part) WG_GLDS(true, 2, 2);
-      else WG_GLDS(false, 2, 2);
+      dim3 g0(ceil_div(K, 64), N / TW, slabs);
+      if (db_part) WG_GLDS(true, 2, 2, g0);
+      else WG_GLDS(false, 2, 2, g0);
     }
 #undef WG_GLDS
-    if (K % TW) {
-      // K tail (obs_dim=260 -> 4 leftover dW rows): one MFMA tile-column
-      // of the register-staged kernel at tile offset kfull*TW/64.  The
-      // scalar wgrad_ktail_kernel (one (k,n) pair per thread, dependent
-      // scalar loads over m) measured 48.7 us — as much as the whole main
-      // wgrad; the LDS-staged MFMA column streams dY once and masks the
-      // dead K columns.
-      // tail width < TW but can span up to two 64-tiles when TW=128
-      dim3 g1(ceil_div(K - kfull * TW, 64), ceil_div(N, 64), slabs);
-      if (fmp)
-        hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2, true>), g1,
-                           dim3(256), 0, stream, x, dy, dW_part, nullptr, M,
-                           N, K, slabs, kfull * TW / 64, *fmp);
-      else
-        hipLaunchKernelGGL((wgrad_partial_kernel<false, 2, 2>), g1, dim3(256),
-                           0, stream, x, dy, dW_part, nullptr, M, N, K, slabs,
-                           kfull * TW / 64);
-    }
   } else {
     dim3 grid(ceil_div(K, big ? 128 : 64), ceil_div(N, big ? 128 : 64), slabs);
 #define WG_PART(WDB, FK, FN)                                                  \
