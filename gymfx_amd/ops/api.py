@@ -109,7 +109,7 @@ def lstm_cell_fwd(
         native.require().lstm_cell_fwd(gates_pre, gates_h, c_prev, c_new,
                                        h_new, done, h_masked, c_masked)
         return
-    gp = gates_pre
+    gp = gates_pre.to(torch.float32)
     if gates_h is not None:
         gp = gp + gates_h.to(torch.float32)
     M, H4 = gp.shape
@@ -148,7 +148,7 @@ def lstm_cell_bwd(
                                        dh_head, dh_next, dc_next, done,
                                        dgates, dc_prev)
         return
-    gp = gates_pre
+    gp = gates_pre.to(torch.float32)
     if gates_h is not None:
         gp = gp + gates_h.to(torch.float32)
     M, H4 = gp.shape

@@ -24,23 +24,23 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                  const void* Yact, int M, int N, int K, bool trans_b, int act,
                  bool dact_tanh, bool add_bias, bool accum, hipStream_t stream,
                  const FeistelMap* fmp = nullptr);
-void launch_lstm_cell_fwd(const float* gates_pre, const void* gates_h,
+void launch_lstm_cell_fwd(const void* gates_pre, const void* gates_h,
                           const float* c_prev, float* c_new, void* h_new,
                           const bool* done, void* h_masked, float* c_masked,
                           int64_t M, int H, hipStream_t stream);
-void launch_lstm_cell_bwd(const float* gates_pre, const void* gates_h,
+void launch_lstm_cell_bwd(const void* gates_pre, const void* gates_h,
                           const float* c_prev,
                           const float* c_new, const float* dh_head,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
                           int64_t M, int H, hipStream_t stream);
 bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
-                               const float* gates_pre, void* gates_h,
+                               const void* gates_pre, void* gates_h,
                                const float* c_prev, float* c_new, void* h_new,
                                const bool* done, void* h_masked,
                                float* c_masked, int M, int N, int K,
                                hipStream_t stream);
-bool launch_lstm_bwd_fused(const float* gates_pre, const void* gates_h,
+bool launch_lstm_bwd_fused(const void* gates_pre, const void* gates_h,
                            const float* c_prev, const float* c_new,
                            const float* dh_head, const float* dh_next,
                            const float* dc_next, const bool* done,
@@ -558,7 +558,7 @@ void lstm_cell_fwd_op(torch::Tensor gates_pre,
                       c10::optional<torch::Tensor> done,
                       c10::optional<torch::Tensor> h_masked,
                       c10::optional<torch::Tensor> c_masked) {
-  check_f32(gates_pre, "gates_pre");
+  check_bf16(gates_pre, "gates_pre");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
   check_bf16(h_new, "h_new");
@@ -585,7 +585,7 @@ void lstm_cell_fwd_op(torch::Tensor gates_pre,
     hm_p = h_masked->data_ptr();
     cm_p = c_masked->data_ptr<float>();
   }
-  gymfx::launch_lstm_cell_fwd(gates_pre.data_ptr<float>(), gh,
+  gymfx::launch_lstm_cell_fwd(gates_pre.data_ptr(), gh,
                               c_prev.data_ptr<float>(),
                               c_new.data_ptr<float>(), h_new.data_ptr(),
                               done_p, hm_p, cm_p, M, H, cur_stream());
@@ -599,7 +599,7 @@ void lstm_cell_bwd_op(torch::Tensor gates_pre,
                       c10::optional<torch::Tensor> dc_next,
                       c10::optional<torch::Tensor> done,
                       torch::Tensor dgates, torch::Tensor dc_prev) {
-  check_f32(gates_pre, "gates_pre");
+  check_bf16(gates_pre, "gates_pre");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
   check_f32(dh_head, "dh_head");
@@ -616,7 +616,7 @@ void lstm_cell_bwd_op(torch::Tensor gates_pre,
     gh = gates_h->data_ptr();
   }
   gymfx::launch_lstm_cell_bwd(
-      gates_pre.data_ptr<float>(), gh, c_prev.data_ptr<float>(),
+      gates_pre.data_ptr(), gh, c_prev.data_ptr<float>(),
       c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
       dgates.data_ptr(), dc_prev.data_ptr<float>(), M, H, cur_stream());
 }
@@ -631,7 +631,7 @@ bool lstm_gemm_cell_fwd_op(torch::Tensor A, torch::Tensor B,
                            c10::optional<torch::Tensor> c_masked) {
   check_bf16(A, "A");
   check_bf16(B, "B");
-  check_f32(gates_pre, "gates_pre");
+  check_bf16(gates_pre, "gates_pre");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
   check_bf16(h_new, "h_new");
@@ -658,7 +658,7 @@ bool lstm_gemm_cell_fwd_op(torch::Tensor A, torch::Tensor B,
     cm_p = c_masked->data_ptr<float>();
   }
   return gymfx::launch_lstm_gemm_cell_fwd(
-      A.data_ptr(), B.data_ptr(), gates_pre.data_ptr<float>(), gh_p,
+      A.data_ptr(), B.data_ptr(), gates_pre.data_ptr(), gh_p,
       c_prev.data_ptr<float>(), c_new.data_ptr<float>(), h_new.data_ptr(),
       done_p, hm_p, cm_p, M, N, K, cur_stream());
 }
@@ -672,7 +672,7 @@ bool lstm_bwd_fused_op(torch::Tensor gates_pre,
                        c10::optional<torch::Tensor> done, torch::Tensor B,
                        torch::Tensor dgates, torch::Tensor dc_prev,
                        c10::optional<torch::Tensor> dh_prev) {
-  check_f32(gates_pre, "gates_pre");
+  check_bf16(gates_pre, "gates_pre");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
   check_f32(dh_head, "dh_head");
@@ -692,7 +692,7 @@ bool lstm_bwd_fused_op(torch::Tensor gates_pre,
   const bool* dn = done.has_value() ? done->data_ptr<bool>() : nullptr;
   float* dhp = dh_prev.has_value() ? dh_prev->data_ptr<float>() : nullptr;
   return gymfx::launch_lstm_bwd_fused(
-      gates_pre.data_ptr<float>(), gh, c_prev.data_ptr<float>(),
+      gates_pre.data_ptr(), gh, c_prev.data_ptr<float>(),
       c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
       B.data_ptr(), dgates.data_ptr(), dc_prev.data_ptr<float>(), dhp, M, H,
       cur_stream());

@@ -1795,19 +1795,20 @@ __global__ __launch_bounds__(256) void mb_gather_seq_kernel(
 // ---------------------------------------------------------------------------
 static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 
-void launch_lstm_cell_fwd(const __bf16* gates_pre, const void* gates_h,
+void launch_lstm_cell_fwd(const void* gates_pre, const void* gates_h,
                           const float* c_prev, float* c_new, void* h_new,
                           const bool* done, void* h_masked, float* c_masked,
                           int64_t M, int H, hipStream_t stream) {
   int64_t total = M * H;
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_fwd_kernel, dim3(blocks), dim3(256), 0, stream,
-                     gates_pre, reinterpret_cast<const __bf16*>(gates_h),
+                     reinterpret_cast<const __bf16*>(gates_pre),
+                     reinterpret_cast<const __bf16*>(gates_h),
                      c_prev, c_new, reinterpret_cast<__bf16*>(h_new), done,
                      reinterpret_cast<__bf16*>(h_masked), c_masked, M, H);
 }
 
-void launch_lstm_cell_bwd(const __bf16* gates_pre, const void* gates_h,
+void launch_lstm_cell_bwd(const void* gates_pre, const void* gates_h,
                           const float* c_prev,
                           const float* c_new, const float* dh_head,
                           const float* dh_next, const float* dc_next,
@@ -1816,13 +1817,14 @@ void launch_lstm_cell_bwd(const __bf16* gates_pre, const void* gates_h,
   int64_t total = M * H;
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_bwd_kernel, dim3(blocks), dim3(256), 0, stream,
-                     gates_pre, reinterpret_cast<const __bf16*>(gates_h),
+                     reinterpret_cast<const __bf16*>(gates_pre),
+                     reinterpret_cast<const __bf16*>(gates_h),
                      c_prev, c_new, dh_head, dh_next, dc_next, done,
                      reinterpret_cast<__bf16*>(dgates), dc_prev, M, H);
 }
 
 bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
-                               const __bf16* gates_pre, void* gates_h,
+                               const void* gates_pre, void* gates_h,
                                const float* c_prev, float* c_new, void* h_new,
                                const bool* done, void* h_masked,
                                float* c_masked, int M, int N, int K,
@@ -1832,21 +1834,23 @@ bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
   if (gates_h) {
     hipLaunchKernelGGL((lstm_gemm_cell_fwd_kernel<true>), grid, dim3(256), 0,
                        stream, reinterpret_cast<const __bf16*>(A),
-                       reinterpret_cast<const __bf16*>(B), gates_pre,
+                       reinterpret_cast<const __bf16*>(B),
+                       reinterpret_cast<const __bf16*>(gates_pre),
                        reinterpret_cast<__bf16*>(gates_h), c_prev, c_new,
                        reinterpret_cast<__bf16*>(h_new), done,
                        reinterpret_cast<__bf16*>(h_masked), c_masked, M, N, K);
   } else {
     hipLaunchKernelGGL((lstm_gemm_cell_fwd_kernel<false>), grid, dim3(256), 0,
                        stream, reinterpret_cast<const __bf16*>(A),
-                       reinterpret_cast<const __bf16*>(B), gates_pre, nullptr,
+                       reinterpret_cast<const __bf16*>(B),
+                       reinterpret_cast<const __bf16*>(gates_pre), nullptr,
                        c_prev, c_new, reinterpret_cast<__bf16*>(h_new), done,
                        reinterpret_cast<__bf16*>(h_masked), c_masked, M, N, K);
   }
   return true;
 }
 
-bool launch_lstm_bwd_fused(const __bf16* gates_pre, const void* gates_h,
+bool launch_lstm_bwd_fused(const void* gates_pre, const void* gates_h,
                            const float* c_prev, const float* c_new,
                            const float* dh_head, const float* dh_next,
                            const float* dc_next, const bool* done,
@@ -1864,7 +1868,8 @@ bool launch_lstm_bwd_fused(const __bf16* gates_pre, const void* gates_h,
 #define GYMFX_LSTM_BWD_CASE(NT)                                               \
   case NT:                                                                    \
     hipLaunchKernelGGL((lstm_bwd_fused_kernel<NT>), grid, dim3(256), lds,     \
-                       stream, gates_pre, gh, c_prev, c_new, dh_head,         \
+                       stream, reinterpret_cast<const __bf16*>(gates_pre),    \
+                       gh, c_prev, c_new, dh_head,                            \
                        dh_next, dc_next, done, Bw, dg, dc_prev, dh_prev, M,   \
                        H);                                                    \
     break;

@@ -28,7 +28,7 @@ def test_fused_fwd_step_bitwise(H, with_done):
     g = torch.Generator(device="cuda").manual_seed(0)
     h = torch.randn(M, H, generator=g, device="cuda").to(torch.bfloat16)
     c = torch.randn(M, H, generator=g, device="cuda")
-    gates = torch.randn(M, 4 * H, generator=g, device="cuda")
+    gates = torch.randn(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
     done = None
     hm_f = cm_f = hm_u = cm_u = None
     if with_done:
@@ -67,7 +67,7 @@ def test_fused_bwd_step_bitwise(H, last_step):
     M, D = 512 + 16, 32
     m = _model(H=H, D=D)
     g = torch.Generator(device="cuda").manual_seed(1)
-    gates = torch.randn(M, 4 * H, generator=g, device="cuda")
+    gates = torch.randn(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
     gh = torch.randn(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g, device="cuda")
     c_new = torch.randn(M, H, generator=g, device="cuda")
@@ -102,7 +102,7 @@ def test_fused_bwd_skips_dh_for_step0():
     H, M = 64, 128
     m = _model(H=H, D=16)
     g = torch.Generator(device="cuda").manual_seed(2)
-    gates = torch.randn(M, 4 * H, generator=g, device="cuda")
+    gates = torch.randn(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g, device="cuda")
     c_new = torch.randn(M, H, generator=g, device="cuda")
     dh_head = torch.randn(M, H, generator=g, device="cuda")

@@ -109,8 +109,10 @@ def test_bptt_matches_autograd_replica():
     heads = []
     for l in range(L):
         gh = (h @ Wh_c)
-        gh = gh.to(torch.bfloat16).float() + (gh - gh.detach())  # bf16 fwd
-        gates = obs[l].float() @ Wx_c + gh + b
+        gh = gh.to(torch.bfloat16).float().detach() + (gh - gh.detach())  # bf16 fwd
+        gx = obs[l].float() @ Wx_c + b
+        gx = gx.to(torch.bfloat16).float().detach() + (gx - gx.detach())  # bf16 gates
+        gates = gx + gh
         # interleaved gate layout: column 4*k + {0=i,1=f,2=g,3=o}
         i, f, gg, o = gates.view(gates.shape[0], -1, 4).unbind(2)
         c = torch.sigmoid(f) * c + torch.sigmoid(i) * torch.tanh(gg)
