@@ -352,7 +352,7 @@ def test_lstm_cell_kernels_match_cpu_oracle():
     ext = _ext()
     M, H = 4096, 256
     g = torch.Generator().manual_seed(5)
-    gates = torch.randn(M, 4 * H, generator=g)
+    gates = torch.randn(M, 4 * H, generator=g).to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g)
     dh_head = torch.randn(M, H, generator=g)
     dh_next = torch.randn(M, H, generator=g)
