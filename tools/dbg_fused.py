@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from gymfx_amd.models.lstm import ActorCriticLSTM
 from gymfx_amd.ops import api, native
