@@ -1043,6 +1043,43 @@ GFX_DEV f32x4 gate_pre4(const __bf16* __restrict__ gates_pre,
   return v;
 }
 
+// Cell forward math shared by the standalone cell kernel AND the fused
+// GEMM epilogue: explicit fmaf pins the contraction so both compile to
+// the same instruction sequence (a 1-ulp fma-vs-mul+add divergence
+// between the two kernels broke bitwise fused==unfused once).
+GFX_DEV void lstm_cell_math(const f32x4 gp, float cp, float* c_out,
+                            float* h_out) {
+  const float i = fast_sigmoid(gp[0]);
+  const float f = fast_sigmoid(gp[1]);
+  const float g = fast_tanh(gp[2]);
+  const float o = fast_sigmoid(gp[3]);
+  const float c = __builtin_fmaf(f, cp, i * g);
+  *c_out = c;
+  *h_out = o * fast_tanh(c);
+}
+
+
+// Cell backward math shared by the standalone bwd kernel and the fused
+// bwd kernel (same contraction-pinning rationale as lstm_cell_math).
+// dh_next/dc_next arrive PRE-masked (0 when absent or across a reset).
+GFX_DEV void lstm_cell_bwd_math(const f32x4 gp, float cp, float c,
+                                float dh_head_v, float dh_next_v,
+                                float dc_next_v, bf16x4* dg, float* dcp) {
+  const float i = fast_sigmoid(gp[0]);
+  const float f = fast_sigmoid(gp[1]);
+  const float g = fast_tanh(gp[2]);
+  const float o = fast_sigmoid(gp[3]);
+  const float tc = fast_tanh(c);
+  const float dhv = dh_head_v + dh_next_v;
+  float dc = dhv * o * (1.f - tc * tc);
+  dc += dc_next_v;
+  (*dg)[0] = f2bf(dc * g * i * (1.f - i));
+  (*dg)[1] = f2bf(dc * cp * f * (1.f - f));
+  (*dg)[2] = f2bf(dc * i * (1.f - g * g));
+  (*dg)[3] = f2bf(dhv * tc * o * (1.f - o));
+  *dcp = dc * f;
+}
+
 __global__ void lstm_cell_fwd_kernel(
     const __bf16* __restrict__ gates_pre,  // [M, 4H] (x-projection + bias)
     const __bf16* __restrict__ gates_h,   // [M, 4H] or null (h-projection)
@@ -1058,13 +1095,10 @@ __global__ void lstm_cell_fwd_kernel(
        idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
     const int64_t m = idx / H;
     const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
-    const float i = fast_sigmoid(gp[0]);
-    const float f = fast_sigmoid(gp[1]);
-    const float g = fast_tanh(gp[2]);
-    const float o = fast_sigmoid(gp[3]);
-    const float c = f * c_prev[idx] + i * g;
+    float c, hval;
+    lstm_cell_math(gp, c_prev[idx], &c, &hval);
     c_new[idx] = c;
-    const __bf16 hb = f2bf(o * fast_tanh(c));
+    const __bf16 hb = f2bf(hval);
     h_new[idx] = hb;
     if (done) {
       const bool d = done[m];
@@ -1097,23 +1131,13 @@ __global__ void lstm_cell_bwd_kernel(
     const int64_t m = idx / H;
     const float mask = (done && done[m]) ? 0.f : 1.f;
     const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
-    const float i = fast_sigmoid(gp[0]);
-    const float f = fast_sigmoid(gp[1]);
-    const float g = fast_tanh(gp[2]);
-    const float o = fast_sigmoid(gp[3]);
-    const float c = c_new[idx];
-    const float tc = fast_tanh(c);
-    float dhv = dh_head[idx];
-    if (dh_next) dhv += mask * dh_next[idx];
-    float dc = dhv * o * (1.f - tc * tc);
-    if (dc_next) dc += mask * dc_next[idx];
     bf16x4 dg;
-    dg[0] = f2bf(dc * g * i * (1.f - i));
-    dg[1] = f2bf(dc * c_prev[idx] * f * (1.f - f));
-    dg[2] = f2bf(dc * i * (1.f - g * g));
-    dg[3] = f2bf(dhv * tc * o * (1.f - o));
+    float dcp;
+    lstm_cell_bwd_math(gp, c_prev[idx], c_new[idx], dh_head[idx],
+                       dh_next ? mask * dh_next[idx] : 0.f,
+                       dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
     *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
-    dc_prev[idx] = dc * f;
+    dc_prev[idx] = dcp;
   }
 }
 
@@ -1276,15 +1300,12 @@ __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
     }
     const bf16x4 gxb = *reinterpret_cast<const bf16x4*>(
         &gates_pre[(int64_t)grow * N + (int64_t)(bn + ul * 4)]);
-    f32x4 gx;
-    for (int j = 0; j < 4; ++j) gx[j] = bf2f(gxb[j]);
-    const float i = fast_sigmoid(gx[0] + gh[0]);
-    const float f = fast_sigmoid(gx[1] + gh[1]);
-    const float g = fast_tanh(gx[2] + gh[2]);
-    const float o = fast_sigmoid(gx[3] + gh[3]);
-    const float c = f * c_prev[idx] + i * g;
+    f32x4 gp;
+    for (int j = 0; j < 4; ++j) gp[j] = bf2f(gxb[j]) + gh[j];
+    float c, hval;
+    lstm_cell_math(gp, c_prev[idx], &c, &hval);
     c_new[idx] = c;
-    const __bf16 hb = f2bf(o * fast_tanh(c));
+    const __bf16 hb = f2bf(hval);
     h_new[idx] = hb;
     if (done) {
       const bool d = done[grow];
@@ -1345,21 +1366,9 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
       const int64_t idx = (int64_t)grow * H + u;
       const float mask = (done && done[grow]) ? 0.f : 1.f;
       const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
-      const float i = fast_sigmoid(gp[0]);
-      const float f = fast_sigmoid(gp[1]);
-      const float g = fast_tanh(gp[2]);
-      const float o = fast_sigmoid(gp[3]);
-      const float c = c_new[idx];
-      const float tc = fast_tanh(c);
-      float dhv = dh_head[idx];
-      if (dh_next) dhv += mask * dh_next[idx];
-      float dc = dhv * o * (1.f - tc * tc);
-      if (dc_next) dc += mask * dc_next[idx];
-      dg[0] = f2bf(dc * g * i * (1.f - i));
-      dg[1] = f2bf(dc * c_prev[idx] * f * (1.f - f));
-      dg[2] = f2bf(dc * i * (1.f - g * g));
-      dg[3] = f2bf(dhv * tc * o * (1.f - o));
-      dcp = dc * f;
+      lstm_cell_bwd_math(gp, c_prev[idx], c_new[idx], dh_head[idx],
+                         dh_next ? mask * dh_next[idx] : 0.f,
+                         dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
       *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
       dc_prev[idx] = dcp;
     }
