@@ -17,11 +17,11 @@ from . import native
 # A/B knob for the fused LSTM step kernels (GYMFX_LSTM_FUSED=0 forces the
 # unfused gemm+cell pairs — bitwise-identical, for measurement)
 _LSTM_FUSED = os.environ.get("GYMFX_LSTM_FUSED", "1") != "0"
-# The fused BACKWARD step measured SLOWER than the unfused pair at the
-# flagship shape (row-slab reads 4x the Wh traffic of the 64x64-tiled
-# dgrad GEMM and runs 1 workgroup/CU): 54.3 vs 42.9 ms/update — kept
-# behind an opt-in knob with its bitwise tests (profiles/PERF_NOTES.md).
-_LSTM_BWD_FUSED = os.environ.get("GYMFX_LSTM_BWD_FUSED", "0") == "1"
+# Fused BACKWARD step v2 (tile-parallel: each 64x64 dgrad tile computes
+# its dgates chunk on the fly; the bn==0 block owns the global writes).
+# v1 (16-row slab across 4H) measured slower than the unfused pair and
+# was replaced (profiles/PERF_NOTES.md).  GYMFX_LSTM_BWD_FUSED=0 disables.
+_LSTM_BWD_FUSED = os.environ.get("GYMFX_LSTM_BWD_FUSED", "1") != "0"
 
 
 def _use_native(t: torch.Tensor) -> bool:

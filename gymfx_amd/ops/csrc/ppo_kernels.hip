@@ -1316,120 +1316,127 @@ __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// FUSED BPTT backward step: the cell backward (dgates from saved gates /
-// cell state / incoming grads) and the recurrent dgrad GEMM
-// dh_prev[M,H] = dgates[M,4H] @ Wh^T merged into ONE kernel — the two were
-// 15.3 us + 14.2 us per sequential step (profiles/lstm_kernel_stats_r2.txt).
-// One workgroup owns a 16-row slab across the FULL 4H gate width: phase 1
-// computes dgates elementwise into LDS (and to global, for the wgrads),
-// phase 2 runs the MFMA K-loop with A already resident in LDS.
-// NT = H/64 = 16-column output tiles per wave (H in {64,128,192,256}).
-// Bitwise parity: dgates are bf16-rounded before the MFMA (matching the
-// unfused global tensor) and the K-chunk order equals gemm_kernel's.
+// FUSED BPTT backward step, tile-parallel: the cell backward (dgates) and
+// the recurrent dgrad GEMM dh_prev = dgates @ Wh^T in ONE kernel with the
+// standard 64x64 output tiling — each (bm, bn) block COMPUTES its A tile
+// (dgates chunk) on the fly instead of reading a materialized tensor.
+// The cell math is recomputed by each of the H/64 bn-blocks (the gate
+// slabs are LLC-resident, the transcendentals are cheap), and only the
+// bn==0 block writes dgates/dc_prev to global.  v1 of this fusion was a
+// 16-row slab across the full 4H width: 4x the Wh traffic of this tiling
+// and 1 workgroup/CU — measured SLOWER than the unfused pair
+// (profiles/PERF_NOTES.md).  Bitwise parity: dgates are bf16-rounded
+// before the MFMA, shared lstm_cell_bwd_math, gemm-identical K order.
 // ---------------------------------------------------------------------------
-template <int NT>
 __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
     const __bf16* __restrict__ gates_pre,  // [M, 4H]
-    const __bf16* __restrict__ gates_h,   // [M, 4H] or null
-    const float* __restrict__ c_prev,     // [M, H]
-    const float* __restrict__ c_new,      // [M, H]
-    const float* __restrict__ dh_head,    // [M, H]
-    const float* __restrict__ dh_next,    // [M, H] or null
-    const float* __restrict__ dc_next,    // [M, H] or null
-    const bool* __restrict__ done,        // [M] or null
-    const __bf16* __restrict__ B,         // Wh [N=H, K=4H] (trans_b layout)
-    __bf16* __restrict__ dgates,          // [M, 4H] out
-    float* __restrict__ dc_prev,          // [M, H] out
-    float* __restrict__ dh_prev,          // [M, H] out, or null (step 0)
+    const __bf16* __restrict__ gates_h,    // [M, 4H] or null
+    const float* __restrict__ c_prev,      // [M, H]
+    const float* __restrict__ c_new,       // [M, H]
+    const float* __restrict__ dh_head,     // [M, H]
+    const float* __restrict__ dh_next,     // [M, H] or null
+    const float* __restrict__ dc_next,     // [M, H] or null
+    const bool* __restrict__ done,         // [M] or null
+    const __bf16* __restrict__ B,          // Wh [N=H, K=4H] (trans_b layout)
+    __bf16* __restrict__ dgates,           // [M, 4H] out
+    float* __restrict__ dc_prev,           // [M, H] out
+    float* __restrict__ dh_prev,           // [M, H] out, or null (step 0)
     int M, int H) {
-  constexpr int BM = 16, BK = 32;
-  const int N = H;
+  constexpr int BM = 64, BK = 32, BN = 64;
   const int K = 4 * H;
-  // LDS: As [BM][4H] bf16 (dgates slab), Bs [2][N][BK] bf16
-  extern __shared__ __align__(16) __bf16 lds[];
-  __bf16* As = lds;                       // BM * K
-  __bf16* Bs = lds + BM * K;              // 2 * N * BK
+  const int N = H;
+  __shared__ __align__(16) __bf16 As[2][BM][BK];
+  __shared__ __align__(16) __bf16 Bs[2][BN][BK];
 
   const int bm = blockIdx.x * BM;
+  const int bn = blockIdx.y * BN;
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+  const bool write_dg = blockIdx.y == 0;
 
-  // ---- phase 1: cell backward into LDS + global ----------------------
-  for (int p = tid; p < BM * H; p += 256) {
-    const int rl = p / H;
-    const int u = p % H;
-    const int grow = bm + rl;
-    bf16x4 dg = {};
-    float dcp = 0.f;
-    if (grow < M) {
-      const int64_t idx = (int64_t)grow * H + u;
-      const float mask = (done && done[grow]) ? 0.f : 1.f;
-      const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
-      lstm_cell_bwd_math(gp, c_prev[idx], c_new[idx], dh_head[idx],
-                         dh_next ? mask * dh_next[idx] : 0.f,
-                         dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
-      *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
-      dc_prev[idx] = dcp;
-    }
-    *reinterpret_cast<bf16x4*>(&As[rl * K + u * 4]) = dg;
-  }
-  if (dh_prev == nullptr) return;  // step 0: no recurrent grad consumer
-
-  // ---- phase 2: dh_prev = dgates @ Wh^T (A in LDS) -------------------
-  // B staged via global_load_lds double-buffering, exactly the gemm_kernel
-  // recipe: the glds for chunk k+1 is in flight across chunk k's MFMAs and
-  // the barrier's vmcnt(0) drains it — ONE barrier per chunk.  (The first
-  // version staged through registers with two barriers per chunk and
-  // serialized ~32 L2 round-trips; measured SLOWER than the unfused pair.)
-  auto glds_b = [&](int buf, int k0) {
-    // B chunk [N][BK] = N*BK elements, 8 bf16 (16 B) per lane-instr
-    const int total8 = N * BK / 8;          // bf16x8 transfers
-    const int per_wave = total8 / 4;        // lanes*instrs per wave
-    for (int j = 0; j < per_wave / 64; ++j) {
-      const int e0 = (wave * (per_wave / 64) + j) * 512;
-      const int row = (e0 + lane * 8) / BK;
-      const int col = (e0 + lane * 8) % BK;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)(uintptr_t)(
-              &B[(int64_t)row * K + k0 + col]),
-          (__attribute__((address_space(3))) void*)(uintptr_t)(
-              &Bs[(int64_t)buf * N * BK] + e0),
-          16, 0, 0);
+  // compute the dgates chunk for units [kt*8, kt*8+8) of rows [bm, bm+64)
+  auto compute_a = [&](int buf, int kt) {
+    const int u0 = kt * 8;
+    for (int j = 0; j < 2; ++j) {
+      const int p = tid + j * 256;       // (row, unit) pair
+      const int rl = p >> 3;
+      const int ul = p & 7;
+      const int grow = bm + rl;
+      bf16x4 dg = {};
+      float dcp = 0.f;
+      if (grow < M) {
+        const int64_t idx = (int64_t)grow * H + (u0 + ul);
+        const float mask = (done && done[grow]) ? 0.f : 1.f;
+        const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
+        lstm_cell_bwd_math(gp, c_prev[idx], c_new[idx], dh_head[idx],
+                           dh_next ? mask * dh_next[idx] : 0.f,
+                           dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
+        if (write_dg) {
+          *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
+          dc_prev[idx] = dcp;
+        }
+      }
+      *reinterpret_cast<bf16x4*>(&As[buf][rl][ul * 4]) = dg;
     }
   };
-  f32x4 acc[NT] = {};
+  if (dh_prev == nullptr) {
+    // step 0: no recurrent-grad consumer — cell backward only (bn==0 grid)
+    const int ktiles = K / BK;
+    for (int kt = 0; kt < ktiles; ++kt) compute_a(0, kt);
+    return;
+  }
+  auto glds_b = [&](int buf, int k0) {
+    const int e0 = wave * 512;  // BN*BK = 2048 elems = 4 waves x 512
+    const int row = (e0 + lane * 8) / BK;
+    const int col = (e0 + lane * 8) % BK;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(uintptr_t)(
+            &B[(int64_t)(bn + row) * K + k0 + col]),
+        (__attribute__((address_space(3))) void*)(uintptr_t)(
+            &Bs[buf][0][0] + e0),
+        16, 0, 0);
+  };
+
+  f32x4 acc[2][2] = {};
   const int row_a = lane & 15;
   const int kseg = lane >> 4;
-  const int ktiles = K / BK;  // 4H % 32 == 0 for H % 8 == 0
+  const int ktiles = K / BK;
   glds_b(0, 0);
+  compute_a(0, 0);
   __syncthreads();
   for (int kt = 0; kt < ktiles; ++kt) {
     const int cur = kt & 1;
-    if (kt + 1 < ktiles) glds_b(1 - cur, (kt + 1) * BK);
-    bf16x8 af = *reinterpret_cast<const bf16x8*>(
-        &As[row_a * K + kt * BK + kseg * 8]);
-    bf16x8 bf[NT];
-    for (int ni = 0; ni < NT; ++ni) {
-      const int col = wave * (NT * 16) + ni * 16 + row_a;
-      bf[ni] = *reinterpret_cast<const bf16x8*>(
-          &Bs[(cur * N + col) * BK + kseg * 8]);
+    if (kt + 1 < ktiles) {
+      glds_b(1 - cur, (kt + 1) * BK);
+      compute_a(1 - cur, kt + 1);
     }
-    for (int ni = 0; ni < NT; ++ni)
-      acc[ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf[ni], acc[ni],
-                                                        0, 0, 0);
+    bf16x8 af[2], bf[2];
+    for (int mi = 0; mi < 2; ++mi)
+      af[mi] = *reinterpret_cast<const bf16x8*>(
+          &As[cur][wr * 32 + mi * 16 + row_a][kseg * 8]);
+    for (int ni = 0; ni < 2; ++ni)
+      bf[ni] = *reinterpret_cast<const bf16x8*>(
+          &Bs[cur][wc * 32 + ni * 16 + row_a][kseg * 8]);
+    for (int mi = 0; mi < 2; ++mi)
+      for (int ni = 0; ni < 2; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
     __syncthreads();
   }
   const int crow = (lane >> 4) * 4;
   const int ccol = lane & 15;
-  for (int ni = 0; ni < NT; ++ni) {
-    const int gcol = wave * (NT * 16) + ni * 16 + ccol;
-    for (int r = 0; r < 4; ++r) {
-      const int grow = bm + crow + r;
-      if (grow < M && gcol < N)
-        dh_prev[(int64_t)grow * N + gcol] = acc[ni][r];
+  for (int mi = 0; mi < 2; ++mi)
+    for (int ni = 0; ni < 2; ++ni) {
+      const int gcol = bn + wc * 32 + ni * 16 + ccol;
+      for (int r = 0; r < 4; ++r) {
+        const int grow = bm + wr * 32 + mi * 16 + crow + r;
+        if (grow < M && gcol < N)
+          dh_prev[(int64_t)grow * N + gcol] = acc[mi][ni][r];
+      }
     }
-  }
 }
 
 // zero the recurrent state of terminated envs (rollout path: applied right
@@ -1866,30 +1873,15 @@ bool launch_lstm_bwd_fused(const void* gates_pre, const void* gates_h,
                            const void* B, void* dgates, float* dc_prev,
                            float* dh_prev, int M, int H,
                            hipStream_t stream) {
-  if (H % 64 != 0 || H > 256) return false;
-  dim3 grid(ceil_div(M, 16));
-  const size_t lds = (size_t)16 * 4 * H * sizeof(__bf16) +
-                     (size_t)2 * H * 32 * sizeof(__bf16);
-  const __bf16* Bw = reinterpret_cast<const __bf16*>(B);
-  __bf16* dg = reinterpret_cast<__bf16*>(dgates);
-  const __bf16* gh = reinterpret_cast<const __bf16*>(gates_h);
-  switch (H / 64) {
-#define GYMFX_LSTM_BWD_CASE(NT)                                               \
-  case NT:                                                                    \
-    hipLaunchKernelGGL((lstm_bwd_fused_kernel<NT>), grid, dim3(256), lds,     \
-                       stream, reinterpret_cast<const __bf16*>(gates_pre),    \
-                       gh, c_prev, c_new, dh_head,                            \
-                       dh_next, dc_next, done, Bw, dg, dc_prev, dh_prev, M,   \
-                       H);                                                    \
-    break;
-    GYMFX_LSTM_BWD_CASE(1)
-    GYMFX_LSTM_BWD_CASE(2)
-    GYMFX_LSTM_BWD_CASE(3)
-    GYMFX_LSTM_BWD_CASE(4)
-#undef GYMFX_LSTM_BWD_CASE
-    default:
-      return false;
-  }
+  if (H % 64 != 0) return false;
+  dim3 grid(ceil_div(M, 64), dh_prev ? H / 64 : 1);
+  hipLaunchKernelGGL(lstm_bwd_fused_kernel, grid, dim3(256), 0, stream,
+                     reinterpret_cast<const __bf16*>(gates_pre),
+                     reinterpret_cast<const __bf16*>(gates_h), c_prev, c_new,
+                     dh_head, dh_next, dc_next, done,
+                     reinterpret_cast<const __bf16*>(B),
+                     reinterpret_cast<__bf16*>(dgates), dc_prev, dh_prev, M,
+                     H);
   return true;
 }
 
