@@ -1357,53 +1357,26 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
   const int wc = wave & 1;
   const bool write_dg = blockIdx.y == 0;
 
-  // A-tile production is split so the cell-input GLOBAL loads issue early
-  // (overlapped with the MFMA window) and only the cheap math + LDS store
-  // run between barriers — v2a with loads inside the chunk serialized a
-  // full memory round-trip per chunk (73.8 ms update vs 39.4 unfused).
-  struct APair {
-    f32x4 gp;
-    float cp, c, dhh, dhn, dcn, mask;
-    int64_t idx;
-    bool live;
-  };
-  APair ap[2];
-  auto load_a = [&](int kt) {
+  // compute the dgates chunk for units [kt*8, kt*8+8) of rows [bm, bm+64)
+  auto compute_a = [&](int buf, int kt) {
     const int u0 = kt * 8;
     for (int j = 0; j < 2; ++j) {
       const int p = tid + j * 256;       // (row, unit) pair
       const int rl = p >> 3;
       const int ul = p & 7;
       const int grow = bm + rl;
-      ap[j].live = grow < M;
-      if (!ap[j].live) continue;
-      const int64_t idx = (int64_t)grow * H + (u0 + ul);
-      ap[j].idx = idx;
-      ap[j].mask = (done && done[grow]) ? 0.f : 1.f;
-      ap[j].gp = gate_pre4(gates_pre, gates_h, idx * 4);
-      ap[j].cp = c_prev[idx];
-      ap[j].c = c_new[idx];
-      ap[j].dhh = dh_head[idx];
-      ap[j].dhn = dh_next ? dh_next[idx] : 0.f;
-      ap[j].dcn = dc_next ? dc_next[idx] : 0.f;
-    }
-  };
-  auto finish_a = [&](int buf, int kt) {
-    const int u0 = kt * 8;
-    (void)u0;
-    for (int j = 0; j < 2; ++j) {
-      const int p = tid + j * 256;
-      const int rl = p >> 3;
-      const int ul = p & 7;
       bf16x4 dg = {};
       float dcp = 0.f;
-      if (ap[j].live) {
-        lstm_cell_bwd_math(ap[j].gp, ap[j].cp, ap[j].c, ap[j].dhh,
-                           ap[j].mask * ap[j].dhn, ap[j].mask * ap[j].dcn,
-                           &dg, &dcp);
+      if (grow < M) {
+        const int64_t idx = (int64_t)grow * H + (u0 + ul);
+        const float mask = (done && done[grow]) ? 0.f : 1.f;
+        const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
+        lstm_cell_bwd_math(gp, c_prev[idx], c_new[idx], dh_head[idx],
+                           dh_next ? mask * dh_next[idx] : 0.f,
+                           dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
         if (write_dg) {
-          *reinterpret_cast<bf16x4*>(&dgates[ap[j].idx * 4]) = dg;
-          dc_prev[ap[j].idx] = dcp;
+          *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
+          dc_prev[idx] = dcp;
         }
       }
       *reinterpret_cast<bf16x4*>(&As[buf][rl][ul * 4]) = dg;
@@ -1412,10 +1385,7 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
   if (dh_prev == nullptr) {
     // step 0: no recurrent-grad consumer — cell backward only (bn==0 grid)
     const int ktiles = K / BK;
-    for (int kt = 0; kt < ktiles; ++kt) {
-      load_a(kt);
-      finish_a(0, kt);
-    }
+    for (int kt = 0; kt < ktiles; ++kt) compute_a(0, kt);
     return;
   }
   auto glds_b = [&](int buf, int k0) {
@@ -1434,17 +1404,14 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
   const int row_a = lane & 15;
   const int kseg = lane >> 4;
   const int ktiles = K / BK;
-  load_a(0);
   glds_b(0, 0);
-  finish_a(0, 0);
+  compute_a(0, 0);
   __syncthreads();
   for (int kt = 0; kt < ktiles; ++kt) {
     const int cur = kt & 1;
     if (kt + 1 < ktiles) {
-      // next chunk's cell inputs + Wh tile go in flight NOW; the math and
-      // LDS store run after the MFMAs so the loads hide under them
-      load_a(kt + 1);
       glds_b(1 - cur, (kt + 1) * BK);
+      compute_a(1 - cur, kt + 1);
     }
     bf16x8 af[2], bf[2];
     for (int mi = 0; mi < 2; ++mi)
@@ -1457,7 +1424,6 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
       for (int ni = 0; ni < 2; ++ni)
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
-    if (kt + 1 < ktiles) finish_a(1 - cur, kt + 1);
     __syncthreads();
   }
   const int crow = (lane >> 4) * 4;
