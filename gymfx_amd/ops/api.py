@@ -17,11 +17,13 @@ from . import native
 # A/B knob for the fused LSTM step kernels (GYMFX_LSTM_FUSED=0 forces the
 # unfused gemm+cell pairs — bitwise-identical, for measurement)
 _LSTM_FUSED = os.environ.get("GYMFX_LSTM_FUSED", "1") != "0"
-# Fused BACKWARD step v2 (tile-parallel: each 64x64 dgrad tile computes
-# its dgates chunk on the fly; the bn==0 block owns the global writes).
-# v1 (16-row slab across 4H) measured slower than the unfused pair and
-# was replaced (profiles/PERF_NOTES.md).  GYMFX_LSTM_BWD_FUSED=0 disables.
-_LSTM_BWD_FUSED = os.environ.get("GYMFX_LSTM_BWD_FUSED", "1") != "0"
+# Fused BACKWARD step: measured SLOWER than the unfused pair in every
+# structure tried (v1 row-slab 54.3 ms, v2 tile-parallel-recompute 73.8 ms,
+# v2b register-prefetch 66.8 ms vs 39.4 ms unfused) — the cell's scattered
+# input reads need high occupancy, the MFMA dgrad wants fat registers, and
+# one kernel cannot have both.  Kept behind an opt-in knob with bitwise
+# tests (profiles/PERF_NOTES.md).
+_LSTM_BWD_FUSED = os.environ.get("GYMFX_LSTM_BWD_FUSED", "0") == "1"
 
 
 def _use_native(t: torch.Tensor) -> bool:
