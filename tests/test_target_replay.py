@@ -365,3 +365,31 @@ def test_execution_report_export():
                 "filled_price", "slippage_cost", "financing",
                 "conversion_cost", "broker_ids"):
         assert key in rep
+
+
+def test_replay_cli_runs_flash_crash_example(tmp_path):
+    """tools/run_target_replay.py end-to-end over the shipped example:
+    the flash-crash path fills the STOP (dip printed first) and exports
+    canonical execution reports."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parents[1]
+    out = tmp_path / "result.json"
+    reports = tmp_path / "reports.json"
+    r = subprocess.run(
+        [sys.executable, str(repo / "tools" / "run_target_replay.py"),
+         str(repo / "examples" / "replay" / "flash_crash.json"),
+         "--out", str(out), "--reports", str(reports)],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-2000:]
+    result = json.loads(out.read_text())
+    kinds = [e.get("kind") for e in result["events"]
+             if e["event_type"] == "order_filled"]
+    assert kinds == ["order_filled", "bracket_sl_fill"]
+    assert result["positions"]["EUR/USD.SIM"]["units"] == "0"
+    reps = json.loads(reports.read_text())
+    assert len(reps) == 2 and all(
+        x["schema"] == "gymfx.execution_report.v1" for x in reps)
