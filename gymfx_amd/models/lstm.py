@@ -180,7 +180,7 @@ class ActorCriticLSTM:
             # gates_h back, so its global write is skipped on the GPU path
             api.lstm_step_fused(state["h"], self.wt("Wh"), gates,
                                 acts["gates_h"], state["c"], state["c"],
-                                acts["h_tmp"], keep_gates_h=False)
+                                acts["h_tmp"])  # rollout: no saved acts
             state["h"], acts["h_tmp"] = acts["h_tmp"], state["h"]
         else:
             api.gemm(state["h"], self.wt("Wh"), None, acts["gates_h"], act=1,
@@ -205,7 +205,11 @@ class ActorCriticLSTM:
             "h_raw": torch.empty(L, M, H, dtype=torch.bfloat16, device=dev),
             "c_raw": torch.empty(L, M, H, dtype=torch.float32, device=dev),
             "gates": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
-            "gates_h": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
+            # saved gate ACTIVATIONS (i,f,g,o interleaved): what the
+            # backward consumes — gates/gates_h are never re-read by bwd
+            "acts": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
+            # scratch for the unfused fwd fallback's recurrent projection
+            "gh_scratch": torch.empty(M, 4 * H, dtype=torch.bfloat16, device=dev),
             "head": torch.empty(L * M, self.head_dim, dtype=torch.float32, device=dev),
             "dgates": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
             "dh_all": torch.empty(L, M, H, dtype=torch.float32, device=dev),
@@ -232,12 +236,14 @@ class ActorCriticLSTM:
         buf["c_in"][0].copy_(c0)
         for l in range(L):
             # one fused kernel per sequential step: recurrent GEMM + cell +
-            # reset-masked next-step state; gates_h is kept (bwd recompute)
+            # reset-masked next-step state; the four gate activations are
+            # saved (acts) for the backward
             api.lstm_step_fused(buf["h_in"][l], self.wt("Wh"),
-                                buf["gates"][l], buf["gates_h"][l],
+                                buf["gates"][l], buf["gh_scratch"],
                                 buf["c_in"][l], buf["c_raw"][l],
                                 buf["h_raw"][l], done_seq[l],
-                                buf["h_in"][l + 1], buf["c_in"][l + 1])
+                                buf["h_in"][l + 1], buf["c_in"][l + 1],
+                                acts_out=buf["acts"][l])
         api.gemm(buf["h_raw"].view(L * M, H), self.wt("Wy"), self.f32("by"),
                  buf["head"], act=0, trans_b=True)
         return buf["head"]
@@ -263,7 +269,7 @@ class ActorCriticLSTM:
         for l in range(L - 1, -1, -1):
             dc_out = dc_bufs[l & 1]
             dh_out = dh_bufs[l & 1] if l > 0 else None
-            api.lstm_bwd_step(buf["gates"][l], buf["gates_h"][l],
+            api.lstm_bwd_step(buf["acts"][l],
                               buf["c_in"][l], buf["c_raw"][l],
                               buf["dh_all"][l], dh_next, dc_next,
                               done_seq[l], self.w("Wh"), buf["dgates"][l],

@@ -101,15 +101,20 @@ def lstm_cell_fwd(
     done: "Optional[torch.Tensor]" = None,
     h_masked: "Optional[torch.Tensor]" = None,
     c_masked: "Optional[torch.Tensor]" = None,
+    acts_out: "Optional[torch.Tensor]" = None,
 ) -> None:
-    """gate pre-activations = gates_pre (f32, x@Wx + b) [+ gates_h (bf16,
+    """gate pre-activations = gates_pre (bf16, x@Wx + b) [+ gates_h (bf16,
     h@Wh kept separate so the recurrent GEMM writes 2-byte outputs instead
-    of read-modify-writing the f32 buffer)] -> c_new f32, h_new bf16.
+    of read-modify-writing a fat buffer)] -> c_new f32, h_new bf16.
     With done/h_masked/c_masked the kernel also emits the reset-masked
-    state that feeds the next BPTT step (fuses the masked_state launch)."""
+    state that feeds the next BPTT step (fuses the masked_state launch).
+    acts_out [M,4H] bf16: saves the four gate ACTIVATIONS (i,f,g,o
+    interleaved) — the backward consumes these instead of re-reading and
+    re-activating the pre-activation tensors."""
     if _use_native(gates_pre):
         native.require().lstm_cell_fwd(gates_pre, gates_h, c_prev, c_new,
-                                       h_new, done, h_masked, c_masked)
+                                       h_new, done, h_masked, c_masked,
+                                       acts_out)
         return
     gp = gates_pre.to(torch.float32)
     if gates_h is not None:
@@ -122,6 +127,9 @@ def lstm_cell_fwd(
     f = torch.sigmoid(gv[..., 1])
     g = torch.tanh(gv[..., 2])
     o = torch.sigmoid(gv[..., 3])
+    if acts_out is not None:
+        acts_out.copy_(torch.stack([i, f, g, o], dim=2).view(M, 4 * H)
+                       .to(torch.bfloat16))
     c = f * c_prev + i * g
     c_new.copy_(c)
     h_new.copy_((o * torch.tanh(c)).to(torch.bfloat16))
@@ -132,8 +140,7 @@ def lstm_cell_fwd(
 
 
 def lstm_cell_bwd(
-    gates_pre: torch.Tensor,
-    gates_h: "Optional[torch.Tensor]",
+    acts: torch.Tensor,
     c_prev: torch.Tensor,
     c_new: torch.Tensor,
     dh_head: torch.Tensor,
@@ -143,27 +150,26 @@ def lstm_cell_bwd(
     dgates: torch.Tensor,
     dc_prev: torch.Tensor,
 ) -> None:
-    """BPTT cell backward; dh_next/dc_next (grads arriving from step l+1)
+    """BPTT cell backward from the SAVED bf16 activations (i,f,g,o
+    interleaved, written by the forward cell) — no gate recompute, no
+    pre-activation reads.  dh_next/dc_next (grads arriving from step l+1)
     are masked by `done` so nothing propagates across an episode reset."""
-    if _use_native(gates_pre):
-        native.require().lstm_cell_bwd(gates_pre, gates_h, c_prev, c_new,
+    if _use_native(acts):
+        native.require().lstm_cell_bwd(acts, c_prev, c_new,
                                        dh_head, dh_next, dc_next, done,
                                        dgates, dc_prev)
         return
-    gp = gates_pre.to(torch.float32)
-    if gates_h is not None:
-        gp = gp + gates_h.to(torch.float32)
-    M, H4 = gp.shape
+    M, H4 = acts.shape
     H = H4 // 4
     mask = None
     if done is not None:
         mask = (~done).to(torch.float32).unsqueeze(1)
     # interleaved gate layout: column 4*k + {0=i,1=f,2=g,3=o} (models/lstm.py)
-    gv = gp.view(M, H, 4)
-    i = torch.sigmoid(gv[..., 0])
-    f = torch.sigmoid(gv[..., 1])
-    g = torch.tanh(gv[..., 2])
-    o = torch.sigmoid(gv[..., 3])
+    av = acts.to(torch.float32).view(M, H, 4)
+    i = av[..., 0]
+    f = av[..., 1]
+    g = av[..., 2]
+    o = av[..., 3]
     tc = torch.tanh(c_new)
     dh = dh_head.clone()
     if dh_next is not None:
@@ -192,28 +198,27 @@ def lstm_step_fused(
     done: "Optional[torch.Tensor]" = None,
     h_masked: "Optional[torch.Tensor]" = None,
     c_masked: "Optional[torch.Tensor]" = None,
-    keep_gates_h: bool = True,
+    acts_out: "Optional[torch.Tensor]" = None,
 ) -> None:
     """One recurrent step: gates_h = h_in @ Wh^T then the LSTM cell, fused
     into a single kernel on GPU (cell in the GEMM epilogue — removes one
-    launch from the sequential BPTT chain).  ``keep_gates_h=False`` (rollout)
-    skips the gates_h global write entirely.  Falls back to the unfused
-    gemm + lstm_cell_fwd pair (bitwise-identical result) on CPU or for
-    shapes without a fused kernel."""
+    launch from the sequential BPTT chain).  acts_out (BPTT) saves the
+    four gate activations for the backward; the rollout passes None and
+    skips that write entirely.  gates_h_buf is scratch for the unfused
+    fallback (CPU / unsupported shapes), which is numerically identical."""
     if _LSTM_FUSED and _use_native(gates_pre):
         ok = native.require().lstm_gemm_cell_fwd(
-            h_in, wh_t, gates_pre, gates_h_buf if keep_gates_h else None,
+            h_in, wh_t, gates_pre, acts_out,
             c_prev, c_new, h_new, done, h_masked, c_masked)
         if ok:
             return
     gemm(h_in, wh_t, None, gates_h_buf, act=1, trans_b=True)
     lstm_cell_fwd(gates_pre, gates_h_buf, c_prev, c_new, h_new, done,
-                  h_masked, c_masked)
+                  h_masked, c_masked, acts_out)
 
 
 def lstm_bwd_step(
-    gates_pre: torch.Tensor,
-    gates_h: "Optional[torch.Tensor]",
+    acts: torch.Tensor,
     c_prev: torch.Tensor,
     c_new: torch.Tensor,
     dh_head: torch.Tensor,
@@ -225,17 +230,18 @@ def lstm_bwd_step(
     dc_prev: torch.Tensor,
     dh_prev: "Optional[torch.Tensor]" = None,
 ) -> None:
-    """One BPTT backward step: cell backward (dgates, dc_prev) plus the
-    recurrent dgrad ``dh_prev = dgates @ Wh^T`` for step l-1, fused into a
-    single kernel on GPU.  ``wh`` is the [H, 4H] weight (trans_b layout).
-    Falls back to lstm_cell_bwd + gemm (bitwise-identical)."""
-    if _LSTM_BWD_FUSED and _use_native(gates_pre):
+    """One BPTT backward step from saved activations: cell backward
+    (dgates, dc_prev) plus the recurrent dgrad ``dh_prev = dgates @ Wh^T``
+    for step l-1, fused into a single kernel when GYMFX_LSTM_BWD_FUSED=1.
+    ``wh`` is the [H, 4H] weight (trans_b layout).  Default path is the
+    unfused pair (measured faster, profiles/PERF_NOTES.md)."""
+    if _LSTM_BWD_FUSED and _use_native(acts):
         ok = native.require().lstm_bwd_fused(
-            gates_pre, gates_h, c_prev, c_new, dh_head, dh_next, dc_next,
+            acts, c_prev, c_new, dh_head, dh_next, dc_next,
             done, wh, dgates, dc_prev, dh_prev)
         if ok:
             return
-    lstm_cell_bwd(gates_pre, gates_h, c_prev, c_new, dh_head, dh_next,
+    lstm_cell_bwd(acts, c_prev, c_new, dh_head, dh_next,
                   dc_next, done, dgates, dc_prev)
     if dh_prev is not None:
         gemm(dgates, wh, None, dh_prev, act=0, trans_b=True)

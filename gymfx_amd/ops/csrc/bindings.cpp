@@ -27,25 +27,25 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
 void launch_lstm_cell_fwd(const void* gates_pre, const void* gates_h,
                           const float* c_prev, float* c_new, void* h_new,
                           const bool* done, void* h_masked, float* c_masked,
-                          int64_t M, int H, hipStream_t stream);
-void launch_lstm_cell_bwd(const void* gates_pre, const void* gates_h,
-                          const float* c_prev,
+                          void* acts_out, int64_t M, int H,
+                          hipStream_t stream);
+void launch_lstm_cell_bwd(const void* acts, const float* c_prev,
                           const float* c_new, const float* dh_head,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
                           int64_t M, int H, hipStream_t stream);
 bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
-                               const void* gates_pre, void* gates_h,
+                               const void* gates_pre, void* acts_out,
                                const float* c_prev, float* c_new, void* h_new,
                                const bool* done, void* h_masked,
                                float* c_masked, int M, int N, int K,
                                hipStream_t stream);
-bool launch_lstm_bwd_fused(const void* gates_pre, const void* gates_h,
-                           const float* c_prev, const float* c_new,
-                           const float* dh_head, const float* dh_next,
-                           const float* dc_next, const bool* done,
-                           const void* B, void* dgates, float* dc_prev,
-                           float* dh_prev, int M, int H, hipStream_t stream);
+bool launch_lstm_bwd_fused(const void* acts, const float* c_prev,
+                           const float* c_new, const float* dh_head,
+                           const float* dh_next, const float* dc_next,
+                           const bool* done, const void* B, void* dgates,
+                           float* dc_prev, float* dh_prev, int M, int H,
+                           hipStream_t stream);
 void launch_mask_reset(void* h, float* c, const bool* done, int64_t M, int H,
                        hipStream_t stream);
 void launch_masked_state(const void* h_raw, const float* c_raw,
@@ -557,7 +557,8 @@ void lstm_cell_fwd_op(torch::Tensor gates_pre,
                       torch::Tensor c_new, torch::Tensor h_new,
                       c10::optional<torch::Tensor> done,
                       c10::optional<torch::Tensor> h_masked,
-                      c10::optional<torch::Tensor> c_masked) {
+                      c10::optional<torch::Tensor> c_masked,
+                      c10::optional<torch::Tensor> acts_out) {
   check_bf16(gates_pre, "gates_pre");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
@@ -585,21 +586,25 @@ void lstm_cell_fwd_op(torch::Tensor gates_pre,
     hm_p = h_masked->data_ptr();
     cm_p = c_masked->data_ptr<float>();
   }
+  void* acts_p = nullptr;
+  if (acts_out.has_value()) {
+    check_bf16(*acts_out, "acts_out");
+    acts_p = acts_out->data_ptr();
+  }
   gymfx::launch_lstm_cell_fwd(gates_pre.data_ptr(), gh,
                               c_prev.data_ptr<float>(),
                               c_new.data_ptr<float>(), h_new.data_ptr(),
-                              done_p, hm_p, cm_p, M, H, cur_stream());
+                              done_p, hm_p, cm_p, acts_p, M, H,
+                              cur_stream());
 }
 
-void lstm_cell_bwd_op(torch::Tensor gates_pre,
-                      c10::optional<torch::Tensor> gates_h,
-                      torch::Tensor c_prev,
+void lstm_cell_bwd_op(torch::Tensor acts, torch::Tensor c_prev,
                       torch::Tensor c_new, torch::Tensor dh_head,
                       c10::optional<torch::Tensor> dh_next,
                       c10::optional<torch::Tensor> dc_next,
                       c10::optional<torch::Tensor> done,
                       torch::Tensor dgates, torch::Tensor dc_prev) {
-  check_bf16(gates_pre, "gates_pre");
+  check_bf16(acts, "acts");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
   check_f32(dh_head, "dh_head");
@@ -607,23 +612,19 @@ void lstm_cell_bwd_op(torch::Tensor gates_pre,
   check_f32(dc_prev, "dc_prev");
   const int64_t M = c_prev.size(0);
   const int H = (int)c_prev.size(1);
+  TORCH_CHECK(acts.size(0) == M && acts.size(1) == 4 * H, "acts shape");
   const float* dhn = dh_next.has_value() ? dh_next->data_ptr<float>() : nullptr;
   const float* dcn = dc_next.has_value() ? dc_next->data_ptr<float>() : nullptr;
   const bool* dn = done.has_value() ? done->data_ptr<bool>() : nullptr;
-  const void* gh = nullptr;
-  if (gates_h.has_value()) {
-    check_bf16(*gates_h, "gates_h");
-    gh = gates_h->data_ptr();
-  }
   gymfx::launch_lstm_cell_bwd(
-      gates_pre.data_ptr(), gh, c_prev.data_ptr<float>(),
+      acts.data_ptr(), c_prev.data_ptr<float>(),
       c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
       dgates.data_ptr(), dc_prev.data_ptr<float>(), M, H, cur_stream());
 }
 
 bool lstm_gemm_cell_fwd_op(torch::Tensor A, torch::Tensor B,
                            torch::Tensor gates_pre,
-                           c10::optional<torch::Tensor> gates_h,
+                           c10::optional<torch::Tensor> acts_out,
                            torch::Tensor c_prev, torch::Tensor c_new,
                            torch::Tensor h_new,
                            c10::optional<torch::Tensor> done,
@@ -642,10 +643,10 @@ bool lstm_gemm_cell_fwd_op(torch::Tensor A, torch::Tensor B,
   TORCH_CHECK(N == 4 * K, "N must equal 4*H");
   TORCH_CHECK(gates_pre.size(0) == M && gates_pre.size(1) == N,
               "gates_pre shape");
-  void* gh_p = nullptr;
-  if (gates_h.has_value()) {
-    check_bf16(*gates_h, "gates_h");
-    gh_p = gates_h->data_ptr();
+  void* acts_p = nullptr;
+  if (acts_out.has_value()) {
+    check_bf16(*acts_out, "acts_out");
+    acts_p = acts_out->data_ptr();
   }
   const bool* done_p = nullptr;
   void* hm_p = nullptr;
@@ -658,21 +659,19 @@ bool lstm_gemm_cell_fwd_op(torch::Tensor A, torch::Tensor B,
     cm_p = c_masked->data_ptr<float>();
   }
   return gymfx::launch_lstm_gemm_cell_fwd(
-      A.data_ptr(), B.data_ptr(), gates_pre.data_ptr(), gh_p,
+      A.data_ptr(), B.data_ptr(), gates_pre.data_ptr(), acts_p,
       c_prev.data_ptr<float>(), c_new.data_ptr<float>(), h_new.data_ptr(),
       done_p, hm_p, cm_p, M, N, K, cur_stream());
 }
 
-bool lstm_bwd_fused_op(torch::Tensor gates_pre,
-                       c10::optional<torch::Tensor> gates_h,
-                       torch::Tensor c_prev, torch::Tensor c_new,
-                       torch::Tensor dh_head,
+bool lstm_bwd_fused_op(torch::Tensor acts, torch::Tensor c_prev,
+                       torch::Tensor c_new, torch::Tensor dh_head,
                        c10::optional<torch::Tensor> dh_next,
                        c10::optional<torch::Tensor> dc_next,
                        c10::optional<torch::Tensor> done, torch::Tensor B,
                        torch::Tensor dgates, torch::Tensor dc_prev,
                        c10::optional<torch::Tensor> dh_prev) {
-  check_bf16(gates_pre, "gates_pre");
+  check_bf16(acts, "acts");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
   check_f32(dh_head, "dh_head");
@@ -682,17 +681,12 @@ bool lstm_bwd_fused_op(torch::Tensor gates_pre,
   const int M = (int)c_prev.size(0);
   const int H = (int)c_prev.size(1);
   TORCH_CHECK(B.size(0) == H && B.size(1) == 4 * H, "B must be Wh [H, 4H]");
-  const void* gh = nullptr;
-  if (gates_h.has_value()) {
-    check_bf16(*gates_h, "gates_h");
-    gh = gates_h->data_ptr();
-  }
   const float* dhn = dh_next.has_value() ? dh_next->data_ptr<float>() : nullptr;
   const float* dcn = dc_next.has_value() ? dc_next->data_ptr<float>() : nullptr;
   const bool* dn = done.has_value() ? done->data_ptr<bool>() : nullptr;
   float* dhp = dh_prev.has_value() ? dh_prev->data_ptr<float>() : nullptr;
   return gymfx::launch_lstm_bwd_fused(
-      gates_pre.data_ptr(), gh, c_prev.data_ptr<float>(),
+      acts.data_ptr(), c_prev.data_ptr<float>(),
       c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
       B.data_ptr(), dgates.data_ptr(), dc_prev.data_ptr<float>(), dhp, M, H,
       cur_stream());
@@ -1038,23 +1032,23 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_cell_fwd", &lstm_cell_fwd_op, py::arg("gates_pre"),
         py::arg("gates_h"), py::arg("c_prev"), py::arg("c_new"),
         py::arg("h_new"), py::arg("done") = py::none(),
-        py::arg("h_masked") = py::none(), py::arg("c_masked") = py::none());
+        py::arg("h_masked") = py::none(), py::arg("c_masked") = py::none(),
+        py::arg("acts_out") = py::none());
   m.def("lstm_gemm_cell_fwd", &lstm_gemm_cell_fwd_op,
         "fused h@Wh^T recurrent GEMM + LSTM cell epilogue; returns False "
         "when the shape has no fused kernel (caller falls back)",
-        py::arg("A"), py::arg("B"), py::arg("gates_pre"), py::arg("gates_h"),
-        py::arg("c_prev"), py::arg("c_new"), py::arg("h_new"),
-        py::arg("done") = py::none(), py::arg("h_masked") = py::none(),
-        py::arg("c_masked") = py::none());
+        py::arg("A"), py::arg("B"), py::arg("gates_pre"),
+        py::arg("acts_out"), py::arg("c_prev"), py::arg("c_new"),
+        py::arg("h_new"), py::arg("done") = py::none(),
+        py::arg("h_masked") = py::none(), py::arg("c_masked") = py::none());
   m.def("lstm_bwd_fused", &lstm_bwd_fused_op,
         "fused LSTM cell backward + dgates @ Wh recurrent dgrad; returns "
         "False when the shape has no fused kernel (caller falls back)",
-        py::arg("gates_pre"), py::arg("gates_h"), py::arg("c_prev"),
+        py::arg("acts"), py::arg("c_prev"),
         py::arg("c_new"), py::arg("dh_head"), py::arg("dh_next"),
         py::arg("dc_next"), py::arg("done"), py::arg("B"), py::arg("dgates"),
         py::arg("dc_prev"), py::arg("dh_prev") = py::none());
-  m.def("lstm_cell_bwd", &lstm_cell_bwd_op, py::arg("gates_pre"),
-        py::arg("gates_h"),
+  m.def("lstm_cell_bwd", &lstm_cell_bwd_op, py::arg("acts"),
         py::arg("c_prev"), py::arg("c_new"), py::arg("dh_head"),
         py::arg("dh_next"), py::arg("dc_next"), py::arg("done"),
         py::arg("dgates"), py::arg("dc_prev"));

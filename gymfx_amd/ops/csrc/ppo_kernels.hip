@@ -1048,7 +1048,7 @@ GFX_DEV f32x4 gate_pre4(const __bf16* __restrict__ gates_pre,
 // the same instruction sequence (a 1-ulp fma-vs-mul+add divergence
 // between the two kernels broke bitwise fused==unfused once).
 GFX_DEV void lstm_cell_math(const f32x4 gp, float cp, float* c_out,
-                            float* h_out) {
+                            float* h_out, bf16x4* acts_out) {
   const float i = fast_sigmoid(gp[0]);
   const float f = fast_sigmoid(gp[1]);
   const float g = fast_tanh(gp[2]);
@@ -1056,19 +1056,28 @@ GFX_DEV void lstm_cell_math(const f32x4 gp, float cp, float* c_out,
   const float c = __builtin_fmaf(f, cp, i * g);
   *c_out = c;
   *h_out = o * fast_tanh(c);
+  if (acts_out) {
+    (*acts_out)[0] = f2bf(i);
+    (*acts_out)[1] = f2bf(f);
+    (*acts_out)[2] = f2bf(g);
+    (*acts_out)[3] = f2bf(o);
+  }
 }
 
 
 // Cell backward math shared by the standalone bwd kernel and the fused
 // bwd kernel (same contraction-pinning rationale as lstm_cell_math).
+// The four gate ACTIVATIONS arrive as saved bf16 (written by the forward
+// cell) — the backward no longer touches gates_pre/gates_h at all, which
+// removes 268 MB/minibatch of reads and all transcendental recompute.
 // dh_next/dc_next arrive PRE-masked (0 when absent or across a reset).
-GFX_DEV void lstm_cell_bwd_math(const f32x4 gp, float cp, float c,
+GFX_DEV void lstm_cell_bwd_math(const bf16x4 acts, float cp, float c,
                                 float dh_head_v, float dh_next_v,
                                 float dc_next_v, bf16x4* dg, float* dcp) {
-  const float i = fast_sigmoid(gp[0]);
-  const float f = fast_sigmoid(gp[1]);
-  const float g = fast_tanh(gp[2]);
-  const float o = fast_sigmoid(gp[3]);
+  const float i = bf2f(acts[0]);
+  const float f = bf2f(acts[1]);
+  const float g = bf2f(acts[2]);
+  const float o = bf2f(acts[3]);
   const float tc = fast_tanh(c);
   const float dhv = dh_head_v + dh_next_v;
   float dc = dhv * o * (1.f - tc * tc);
@@ -1089,6 +1098,7 @@ __global__ void lstm_cell_fwd_kernel(
     const bool* __restrict__ done,        // [M] or null: also emit the
     __bf16* __restrict__ h_masked,        //   reset-masked state that feeds
     float* __restrict__ c_masked,         //   the NEXT step (one fewer
+    __bf16* __restrict__ acts_out,        // [M, 4H] saved activations or null
     int64_t M, int H) {                   //   launch than masked_state)
   const int64_t total = M * H;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1096,7 +1106,9 @@ __global__ void lstm_cell_fwd_kernel(
     const int64_t m = idx / H;
     const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
     float c, hval;
-    lstm_cell_math(gp, c_prev[idx], &c, &hval);
+    bf16x4 av;
+    lstm_cell_math(gp, c_prev[idx], &c, &hval, acts_out ? &av : nullptr);
+    if (acts_out) *reinterpret_cast<bf16x4*>(&acts_out[idx * 4]) = av;
     c_new[idx] = c;
     const __bf16 hb = f2bf(hval);
     h_new[idx] = hb;
@@ -1114,8 +1126,7 @@ __global__ void lstm_cell_fwd_kernel(
 // boundary AFTER this step) along with dc_next, so no gradient crosses a
 // reset.
 __global__ void lstm_cell_bwd_kernel(
-    const __bf16* __restrict__ gates_pre,  // [M, 4H] (saved fwd, x part)
-    const __bf16* __restrict__ gates_h,   // [M, 4H] or null (saved h part)
+    const __bf16* __restrict__ acts,      // [M, 4H] saved activations
     const float* __restrict__ c_prev,     // [M, H] (masked input c)
     const float* __restrict__ c_new,      // [M, H] (raw output c)
     const float* __restrict__ dh_head,    // [M, H]
@@ -1130,10 +1141,10 @@ __global__ void lstm_cell_bwd_kernel(
        idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
     const int64_t m = idx / H;
     const float mask = (done && done[m]) ? 0.f : 1.f;
-    const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
+    const bf16x4 av = *reinterpret_cast<const bf16x4*>(&acts[idx * 4]);
     bf16x4 dg;
     float dcp;
-    lstm_cell_bwd_math(gp, c_prev[idx], c_new[idx], dh_head[idx],
+    lstm_cell_bwd_math(av, c_prev[idx], c_new[idx], dh_head[idx],
                        dh_next ? mask * dh_next[idx] : 0.f,
                        dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
     *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
@@ -1160,7 +1171,7 @@ __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
     const __bf16* __restrict__ A,         // h_in [M, K=H]
     const __bf16* __restrict__ B,         // Wh^T [N=4H, K=H]
     const __bf16* __restrict__ gates_pre,  // [M, 4H] (x-proj + bias)
-    __bf16* __restrict__ gates_h,         // [M, 4H] out (if WRITE_GH)
+    __bf16* __restrict__ acts_out,        // [M, 4H] saved activations (if WRITE_GH)
     const float* __restrict__ c_prev,     // [M, H]
     float* __restrict__ c_new,            // [M, H]
     __bf16* __restrict__ h_new,           // [M, H]
@@ -1293,17 +1304,16 @@ __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
     const int64_t idx = (int64_t)grow * H + ug;
     f32x4 gh;
     for (int j = 0; j < 4; ++j) gh[j] = Cs[r0 + r][ul * 4 + j];
-    if (WRITE_GH) {
-      bf16x4 ghb;
-      for (int j = 0; j < 4; ++j) ghb[j] = f2bf(gh[j]);
-      *reinterpret_cast<bf16x4*>(&gates_h[(int64_t)grow * N + (int64_t)(bn + ul * 4)]) = ghb;
-    }
     const bf16x4 gxb = *reinterpret_cast<const bf16x4*>(
         &gates_pre[(int64_t)grow * N + (int64_t)(bn + ul * 4)]);
     f32x4 gp;
     for (int j = 0; j < 4; ++j) gp[j] = bf2f(gxb[j]) + gh[j];
     float c, hval;
-    lstm_cell_math(gp, c_prev[idx], &c, &hval);
+    bf16x4 av;
+    lstm_cell_math(gp, c_prev[idx], &c, &hval, WRITE_GH ? &av : nullptr);
+    if (WRITE_GH)
+      *reinterpret_cast<bf16x4*>(
+          &acts_out[(int64_t)grow * N + (int64_t)(bn + ul * 4)]) = av;
     c_new[idx] = c;
     const __bf16 hb = f2bf(hval);
     h_new[idx] = hb;
@@ -1329,8 +1339,7 @@ __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
 // before the MFMA, shared lstm_cell_bwd_math, gemm-identical K order.
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
-    const __bf16* __restrict__ gates_pre,  // [M, 4H]
-    const __bf16* __restrict__ gates_h,    // [M, 4H] or null
+    const __bf16* __restrict__ acts,       // [M, 4H] saved activations
     const float* __restrict__ c_prev,      // [M, H]
     const float* __restrict__ c_new,       // [M, H]
     const float* __restrict__ dh_head,     // [M, H]
@@ -1370,8 +1379,8 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
       if (grow < M) {
         const int64_t idx = (int64_t)grow * H + (u0 + ul);
         const float mask = (done && done[grow]) ? 0.f : 1.f;
-        const f32x4 gp = gate_pre4(gates_pre, gates_h, idx * 4);
-        lstm_cell_bwd_math(gp, c_prev[idx], c_new[idx], dh_head[idx],
+        const bf16x4 av = *reinterpret_cast<const bf16x4*>(&acts[idx * 4]);
+        lstm_cell_bwd_math(av, c_prev[idx], c_new[idx], dh_head[idx],
                            dh_next ? mask * dh_next[idx] : 0.f,
                            dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
         if (write_dg) {
@@ -1814,18 +1823,19 @@ static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
 void launch_lstm_cell_fwd(const void* gates_pre, const void* gates_h,
                           const float* c_prev, float* c_new, void* h_new,
                           const bool* done, void* h_masked, float* c_masked,
-                          int64_t M, int H, hipStream_t stream) {
+                          void* acts_out, int64_t M, int H,
+                          hipStream_t stream) {
   int64_t total = M * H;
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_fwd_kernel, dim3(blocks), dim3(256), 0, stream,
                      reinterpret_cast<const __bf16*>(gates_pre),
                      reinterpret_cast<const __bf16*>(gates_h),
                      c_prev, c_new, reinterpret_cast<__bf16*>(h_new), done,
-                     reinterpret_cast<__bf16*>(h_masked), c_masked, M, H);
+                     reinterpret_cast<__bf16*>(h_masked), c_masked,
+                     reinterpret_cast<__bf16*>(acts_out), M, H);
 }
 
-void launch_lstm_cell_bwd(const void* gates_pre, const void* gates_h,
-                          const float* c_prev,
+void launch_lstm_cell_bwd(const void* acts, const float* c_prev,
                           const float* c_new, const float* dh_head,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
@@ -1833,26 +1843,25 @@ void launch_lstm_cell_bwd(const void* gates_pre, const void* gates_h,
   int64_t total = M * H;
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_bwd_kernel, dim3(blocks), dim3(256), 0, stream,
-                     reinterpret_cast<const __bf16*>(gates_pre),
-                     reinterpret_cast<const __bf16*>(gates_h),
+                     reinterpret_cast<const __bf16*>(acts),
                      c_prev, c_new, dh_head, dh_next, dc_next, done,
                      reinterpret_cast<__bf16*>(dgates), dc_prev, M, H);
 }
 
 bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
-                               const void* gates_pre, void* gates_h,
+                               const void* gates_pre, void* acts_out,
                                const float* c_prev, float* c_new, void* h_new,
                                const bool* done, void* h_masked,
                                float* c_masked, int M, int N, int K,
                                hipStream_t stream) {
   if (N % 64 != 0) return false;  // 64-col tiles must hold whole units
   dim3 grid(ceil_div(M, 64), N / 64);
-  if (gates_h) {
+  if (acts_out) {
     hipLaunchKernelGGL((lstm_gemm_cell_fwd_kernel<true>), grid, dim3(256), 0,
                        stream, reinterpret_cast<const __bf16*>(A),
                        reinterpret_cast<const __bf16*>(B),
                        reinterpret_cast<const __bf16*>(gates_pre),
-                       reinterpret_cast<__bf16*>(gates_h), c_prev, c_new,
+                       reinterpret_cast<__bf16*>(acts_out), c_prev, c_new,
                        reinterpret_cast<__bf16*>(h_new), done,
                        reinterpret_cast<__bf16*>(h_masked), c_masked, M, N, K);
   } else {
@@ -1866,18 +1875,16 @@ bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
   return true;
 }
 
-bool launch_lstm_bwd_fused(const void* gates_pre, const void* gates_h,
-                           const float* c_prev, const float* c_new,
-                           const float* dh_head, const float* dh_next,
-                           const float* dc_next, const bool* done,
-                           const void* B, void* dgates, float* dc_prev,
-                           float* dh_prev, int M, int H,
+bool launch_lstm_bwd_fused(const void* acts, const float* c_prev,
+                           const float* c_new, const float* dh_head,
+                           const float* dh_next, const float* dc_next,
+                           const bool* done, const void* B, void* dgates,
+                           float* dc_prev, float* dh_prev, int M, int H,
                            hipStream_t stream) {
   if (H % 64 != 0) return false;
   dim3 grid(ceil_div(M, 64), dh_prev ? H / 64 : 1);
   hipLaunchKernelGGL(lstm_bwd_fused_kernel, grid, dim3(256), 0, stream,
-                     reinterpret_cast<const __bf16*>(gates_pre),
-                     reinterpret_cast<const __bf16*>(gates_h), c_prev, c_new,
+                     reinterpret_cast<const __bf16*>(acts), c_prev, c_new,
                      dh_head, dh_next, dc_next, done,
                      reinterpret_cast<const __bf16*>(B),
                      reinterpret_cast<__bf16*>(dgates), dc_prev, dh_prev, M,

@@ -37,21 +37,22 @@ def test_fused_fwd_step_bitwise(H, with_done):
         cm_f = torch.empty(M, H, device="cuda")
         hm_u = torch.empty_like(hm_f)
         cm_u = torch.empty_like(cm_f)
-    # unfused reference
+    # unfused reference (saves the activations like the fused kernel)
     gh_u = torch.empty(M, 4 * H, dtype=torch.bfloat16, device="cuda")
+    acts_u = torch.empty(M, 4 * H, dtype=torch.bfloat16, device="cuda")
     c_u = torch.empty(M, H, device="cuda")
     h_u = torch.empty(M, H, dtype=torch.bfloat16, device="cuda")
     api.gemm(h, m.wt("Wh"), None, gh_u, act=1, trans_b=True)
-    api.lstm_cell_fwd(gates, gh_u, c, c_u, h_u, done, hm_u, cm_u)
+    api.lstm_cell_fwd(gates, gh_u, c, c_u, h_u, done, hm_u, cm_u, acts_u)
     # fused
-    gh_f = torch.empty_like(gh_u)
+    acts_f = torch.empty_like(acts_u)
     c_f = torch.empty_like(c_u)
     h_f = torch.empty_like(h_u)
-    ok = native.require().lstm_gemm_cell_fwd(h, m.wt("Wh"), gates, gh_f, c,
+    ok = native.require().lstm_gemm_cell_fwd(h, m.wt("Wh"), gates, acts_f, c,
                                              c_f, h_f, done, hm_f, cm_f)
     assert ok, "fused fwd kernel refused a supported shape"
     torch.cuda.synchronize()
-    assert torch.equal(gh_f, gh_u)
+    assert torch.equal(acts_f, acts_u)
     assert torch.equal(c_f, c_u)
     assert torch.equal(h_f, h_u)
     if with_done:
@@ -67,8 +68,7 @@ def test_fused_bwd_step_bitwise(H, last_step):
     M, D = 512 + 16, 32
     m = _model(H=H, D=D)
     g = torch.Generator(device="cuda").manual_seed(1)
-    gates = torch.randn(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
-    gh = torch.randn(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
+    acts = torch.rand(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g, device="cuda")
     c_new = torch.randn(M, H, generator=g, device="cuda")
     dh_head = torch.randn(M, H, generator=g, device="cuda")
@@ -79,14 +79,14 @@ def test_fused_bwd_step_bitwise(H, last_step):
     dg_u = torch.empty(M, 4 * H, dtype=torch.bfloat16, device="cuda")
     dcp_u = torch.empty(M, H, device="cuda")
     dhp_u = torch.empty(M, H, device="cuda")
-    api.lstm_cell_bwd(gates, gh, c_prev, c_new, dh_head, dh_next, dc_next,
+    api.lstm_cell_bwd(acts, c_prev, c_new, dh_head, dh_next, dc_next,
                       done, dg_u, dcp_u)
     api.gemm(dg_u, m.w("Wh"), None, dhp_u, act=0, trans_b=True)
     # fused
     dg_f = torch.empty_like(dg_u)
     dcp_f = torch.empty_like(dcp_u)
     dhp_f = torch.empty_like(dhp_u)
-    ok = native.require().lstm_bwd_fused(gates, gh, c_prev, c_new, dh_head,
+    ok = native.require().lstm_bwd_fused(acts, c_prev, c_new, dh_head,
                                          dh_next, dc_next, done, m.w("Wh"),
                                          dg_f, dcp_f, dhp_f)
     assert ok, "fused bwd kernel refused a supported shape"
@@ -102,20 +102,20 @@ def test_fused_bwd_skips_dh_for_step0():
     H, M = 64, 128
     m = _model(H=H, D=16)
     g = torch.Generator(device="cuda").manual_seed(2)
-    gates = torch.randn(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
+    acts = torch.rand(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g, device="cuda")
     c_new = torch.randn(M, H, generator=g, device="cuda")
     dh_head = torch.randn(M, H, generator=g, device="cuda")
     dg = torch.empty(M, 4 * H, dtype=torch.bfloat16, device="cuda")
     dcp = torch.empty(M, H, device="cuda")
-    ok = native.require().lstm_bwd_fused(gates, None, c_prev, c_new, dh_head,
+    ok = native.require().lstm_bwd_fused(acts, c_prev, c_new, dh_head,
                                          None, None, None, m.w("Wh"), dg,
                                          dcp, None)
     assert ok
     torch.cuda.synchronize()
     dg_u = torch.empty_like(dg)
     dcp_u = torch.empty_like(dcp)
-    api.lstm_cell_bwd(gates, None, c_prev, c_new, dh_head, None, None, None,
+    api.lstm_cell_bwd(acts, c_prev, c_new, dh_head, None, None, None,
                       dg_u, dcp_u)
     torch.cuda.synchronize()
     assert torch.equal(dg, dg_u) and torch.equal(dcp, dcp_u)

@@ -372,14 +372,15 @@ def test_lstm_cell_kernels_match_cpu_oracle():
                                rtol=1e-4, atol=1e-5)
     np.testing.assert_allclose(h_new_g.float().cpu().numpy(),
                                h_new_c.float().numpy(), rtol=1e-2, atol=1e-2)
-    # bwd (with masking)
+    # bwd (with masking) — consumes the saved bf16 activations
+    acts = torch.rand(M, 4 * H, generator=g).to(torch.bfloat16)
     dg_c = torch.empty(M, 4 * H, dtype=torch.bfloat16)
     dcp_c = torch.empty(M, H)
-    api.lstm_cell_bwd(gates, gates_h, c_prev, c_new_c, dh_head, dh_next,
+    api.lstm_cell_bwd(acts, c_prev, c_new_c, dh_head, dh_next,
                       dc_next, done, dg_c, dcp_c)
     dg_g = torch.empty(M, 4 * H, dtype=torch.bfloat16).cuda()
     dcp_g = torch.empty(M, H).cuda()
-    ext.lstm_cell_bwd(gates.cuda(), gates_h.cuda(), c_prev.cuda(), c_new_g,
+    ext.lstm_cell_bwd(acts.cuda(), c_prev.cuda(), c_new_g,
                       dh_head.cuda(),
                       dh_next.cuda(), dc_next.cuda(), done.cuda(), dg_g, dcp_g)
     torch.cuda.synchronize()

@@ -47,16 +47,20 @@ def test_lstm_cell_bwd_matches_autograd():
 
     dgates = torch.empty(M, 4 * H, dtype=torch.bfloat16)
     dc_prev = torch.empty(M, H)
-    api.lstm_cell_bwd(gates.detach(), None, c_prev.detach(), c_new.detach(),
+    # backward consumes the SAVED bf16 activations (fwd-cell output)
+    acts = torch.stack([torch.sigmoid(i), torch.sigmoid(f), torch.tanh(gg),
+                        torch.sigmoid(o)], dim=2).reshape(M, 4 * H) \
+        .detach().to(torch.bfloat16)
+    api.lstm_cell_bwd(acts, c_prev.detach(), c_new.detach(),
                       dh, None, dc_next, None, dgates, dc_prev)
-    assert torch.allclose(dgates.float(), dgates_ref, atol=2e-2, rtol=2e-2)
-    assert torch.allclose(dc_prev, dc_prev_ref, atol=1e-5, rtol=1e-4)
+    assert torch.allclose(dgates.float(), dgates_ref, atol=3e-2, rtol=3e-2)
+    assert torch.allclose(dc_prev, dc_prev_ref, atol=2e-2, rtol=2e-2)
 
 
 def test_lstm_cell_bwd_done_masks_recurrent_grads():
     M, H = 8, 4
     g = torch.Generator().manual_seed(2)
-    gates = torch.randn(M, 4 * H, generator=g)
+    acts = torch.rand(M, 4 * H, generator=g).to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g)
     c_new = torch.randn(M, H, generator=g)
     dh_head = torch.randn(M, H, generator=g)
@@ -66,12 +70,12 @@ def test_lstm_cell_bwd_done_masks_recurrent_grads():
     done[::2] = True
     dgates_m = torch.empty(M, 4 * H, dtype=torch.bfloat16)
     dc_prev_m = torch.empty(M, H)
-    api.lstm_cell_bwd(gates, None, c_prev, c_new, dh_head, dh_next, dc_next,
+    api.lstm_cell_bwd(acts, c_prev, c_new, dh_head, dh_next, dc_next,
                       done, dgates_m, dc_prev_m)
     # for done rows the result must equal the no-next-grad case
     dgates_0 = torch.empty(M, 4 * H, dtype=torch.bfloat16)
     dc_prev_0 = torch.empty(M, H)
-    api.lstm_cell_bwd(gates, None, c_prev, c_new, dh_head, None, None, None,
+    api.lstm_cell_bwd(acts, c_prev, c_new, dh_head, None, None, None,
                       dgates_0, dc_prev_0)
     assert torch.equal(dgates_m[::2], dgates_0[::2])
     assert torch.equal(dc_prev_m[::2], dc_prev_0[::2])
