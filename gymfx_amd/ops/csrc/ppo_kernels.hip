@@ -2002,8 +2002,14 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
     const char* e = getenv("GYMFX_GEMM_BK64");
     return e ? atoi(e) : -1;
   }();
+  // Tile-width heuristic: at N >= 1024 the 64-wide tile re-streams the A
+  // operand N/64 times (the LSTM Wx gates GEMM read its 34 MB A slab 16x
+  // = 544 MB/call); the 64x256 tile cuts that 4x and wins big there.
+  // The round-1 "wide loses" measurement holds only for N <= 256 outputs.
   const bool mid = (N >= 96) && wide_env == 4;
-  const bool wide = (N >= 192) && !mid && wide_env == 1;
+  const bool wide = (N >= 192) && !mid &&
+                    (wide_env == 1 ||
+                     (wide_env == -1 && N >= 1024 && M >= 16384));
   // measured: BK=64 wins for long-K shapes (LSTM dgates dgrad K=1024:
   // update 41.5 -> 39.9 ms) and LOSES at K<=260 (MLP headline 28.4 ->
   // 27.4M) — auto-enable only at K >= 512
@@ -2112,8 +2118,10 @@ void launch_wgrad(const void* X, const void* dY, float* dW_part, float* db_part,
                          0, stream, x, dy, dW_part, db_part, M, N, K, slabs); \
   } while (0)
     if (big) {
+      // 5x4 (TK=160) saves a dY pass for K=260 but its 80-VGPR accs spill
+      // at N >= 1024 (LSTM Wx wgrad measured 187 us vs 144 for 4x4+tail)
       const int t44 = ceil_div(K, 128), t54 = ceil_div(K, 160);
-      if (t54 < t44) {
+      if (t54 < t44 && N <= 512) {
         dim3 g0(t54, N / TW, slabs);
         if (db_part) WG_GLDS(true, 5, 4, g0);
         else WG_GLDS(false, 5, 4, g0);
