@@ -2006,10 +2006,10 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
   // operand N/64 times (the LSTM Wx gates GEMM read its 34 MB A slab 16x
   // = 544 MB/call); the 64x256 tile cuts that 4x and wins big there.
   // The round-1 "wide loses" measurement holds only for N <= 256 outputs.
-  const bool mid = (N >= 96) && wide_env == 4;
-  const bool wide = (N >= 192) && !mid &&
-                    (wide_env == 1 ||
-                     (wide_env == -1 && N >= 1024 && M >= 16384));
+  const bool mid = (N >= 96) &&
+                   (wide_env == 4 ||
+                    (wide_env == -1 && N >= 1024 && M >= 16384));
+  const bool wide = (N >= 192) && !mid && wide_env == 1;
   // measured: BK=64 wins for long-K shapes (LSTM dgates dgrad K=1024:
   // update 41.5 -> 39.9 ms) and LOSES at K<=260 (MLP headline 28.4 ->
   // 27.4M) — auto-enable only at K >= 512
