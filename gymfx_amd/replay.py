@@ -250,6 +250,7 @@ class ReplayAdapter:
             n_steps += 1
 
         bar = int(torch.clamp(env.st.cursor[0] - 1, min=0).item())
+        last_bar = bar
         engine = {
             "equity": float(env.st.equity[0].item()),
             "cash": float(env.st.cash[0].item()),
@@ -273,6 +274,7 @@ class ReplayAdapter:
         result = {
             "schema": "gymfx.replay.v1",
             "steps": n_steps,
+            "last_bar": last_bar,
             "engine": engine,
             "oracle": oracle,
             "reconciled": all(recon.values()),

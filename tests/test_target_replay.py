@@ -393,3 +393,52 @@ def test_replay_cli_runs_flash_crash_example(tmp_path):
     reps = json.loads(reports.read_text())
     assert len(reps) == 2 and all(
         x["schema"] == "gymfx.execution_report.v1" for x in reps)
+
+
+def _parallel_replay_worker(q):
+    """Isolated-process evaluator (ref tools/nautilus_parallel_smoke.py):
+    same inputs must hash identically across spawned processes."""
+    import json
+    from pathlib import Path
+
+    from gymfx_amd.target_replay import TargetReplay
+    import tools.run_target_replay as cli
+
+    repo = Path(__file__).resolve().parents[1]
+    raw = json.loads((repo / "examples" / "replay" / "flash_crash.json")
+                     .read_text())
+    profile, specs, frames, actions = cli.load_inputs(raw)
+    result = TargetReplay(profile).run(instrument_specs=specs, frames=frames,
+                                       actions=actions)
+    q.put(result["result_hash"])
+
+
+def test_parallel_evaluators_hash_identically():
+    import multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_parallel_replay_worker, args=(q,))
+             for _ in range(2)]
+    for p in procs:
+        p.start()
+    hashes = [q.get(timeout=120) for _ in procs]
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert hashes[0] == hashes[1]
+
+
+def test_cross_engine_bakeoff_reconciles():
+    """tools/engine_bakeoff.py: the vectorized engine, the scalar ledger
+    and the Decimal target replay agree on the same action script
+    (the reference's cross-engine bakeoff, natively)."""
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+    from tools.engine_bakeoff import run_bakeoff
+
+    out = run_bakeoff(rows=250, steps=80, seed=13)
+    assert out["vec_vs_ledger_ok"]
+    assert out["vec_vs_target_replay_ok"], out
