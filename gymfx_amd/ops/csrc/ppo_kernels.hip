@@ -1292,35 +1292,52 @@ __global__ __launch_bounds__(256, 2) void lstm_gemm_cell_fwd_kernel(
             bf2f(f2bf(acc[mi][ni][r]));
   __syncthreads();
 
-  // ---- cell phase: 16 units x 64 rows; thread t -> unit t%16, rows
-  // (t/16)*4 .. +3 ------------------------------------------------------
-  const int ul = tid & 15;            // unit within the tile
-  const int r0 = (tid >> 4) * 4;      // first local row
+  // ---- cell phase: one thread owns FOUR CONSECUTIVE units of one row
+  // (4 threads/row x 64 rows), so every global access is a vector op:
+  // gates/acts 32 B, c 16 B f32x4, h 8 B bf16x4.  (The first mapping —
+  // 4 rows x 1 unit per thread — emitted 20 scattered scalar stores.)
+  const int rl = tid >> 2;            // local row 0..63
+  const int u0l = (tid & 3) * 4;      // first local unit (of 16)
   const int H = N / 4;
-  const int ug = bn / 4 + ul;         // global hidden unit
-  for (int r = 0; r < 4; ++r) {
-    const int grow = bm + r0 + r;
-    if (grow >= M) continue;
-    const int64_t idx = (int64_t)grow * H + ug;
-    f32x4 gh;
-    for (int j = 0; j < 4; ++j) gh[j] = Cs[r0 + r][ul * 4 + j];
-    const bf16x4 gxb = *reinterpret_cast<const bf16x4*>(
-        &gates_pre[(int64_t)grow * N + (int64_t)(bn + ul * 4)]);
-    f32x4 gp;
-    for (int j = 0; j < 4; ++j) gp[j] = bf2f(gxb[j]) + gh[j];
-    float c, hval;
-    bf16x4 av;
-    lstm_cell_math(gp, c_prev[idx], &c, &hval, WRITE_GH ? &av : nullptr);
-    if (WRITE_GH)
-      *reinterpret_cast<bf16x4*>(
-          &acts_out[(int64_t)grow * N + (int64_t)(bn + ul * 4)]) = av;
-    c_new[idx] = c;
-    const __bf16 hb = f2bf(hval);
-    h_new[idx] = hb;
+  const int grow = bm + rl;
+  if (grow < M) {
+    const int64_t idx0 = (int64_t)grow * H + (bn / 4 + u0l);
+    const int64_t g0 = (int64_t)grow * N + (int64_t)(bn + u0l * 4);
+    float c4[4], h4[4];
+    bf16x4 av4[4];
+    for (int j = 0; j < 4; ++j) {
+      f32x4 gp;
+      const bf16x4 gxb = *reinterpret_cast<const bf16x4*>(
+          &gates_pre[g0 + j * 4]);
+      for (int q = 0; q < 4; ++q)
+        gp[q] = bf2f(gxb[q]) + Cs[rl][(u0l + j) * 4 + q];
+      lstm_cell_math(gp, c_prev[idx0 + j], &c4[j], &h4[j],
+                     WRITE_GH ? &av4[j] : nullptr);
+    }
+    if (WRITE_GH) {
+      *reinterpret_cast<bf16x8*>(&acts_out[g0]) =
+          *reinterpret_cast<const bf16x8*>(&av4[0]);
+      *reinterpret_cast<bf16x8*>(&acts_out[g0 + 8]) =
+          *reinterpret_cast<const bf16x8*>(&av4[2]);
+    }
+    f32x4 cv;
+    bf16x4 hv;
+    for (int j = 0; j < 4; ++j) {
+      cv[j] = c4[j];
+      hv[j] = f2bf(h4[j]);
+    }
+    *reinterpret_cast<f32x4*>(&c_new[idx0]) = cv;
+    *reinterpret_cast<bf16x4*>(&h_new[idx0]) = hv;
     if (done) {
       const bool d = done[grow];
-      h_masked[idx] = d ? (__bf16)0.f : hb;
-      c_masked[idx] = d ? 0.f : c;
+      f32x4 cm;
+      bf16x4 hm;
+      for (int j = 0; j < 4; ++j) {
+        cm[j] = d ? 0.f : cv[j];
+        hm[j] = d ? (__bf16)0.f : hv[j];
+      }
+      *reinterpret_cast<f32x4*>(&c_masked[idx0]) = cm;
+      *reinterpret_cast<bf16x4*>(&h_masked[idx0]) = hm;
     }
   }
 }
