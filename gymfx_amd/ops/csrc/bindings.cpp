@@ -613,6 +613,7 @@ void lstm_cell_bwd_op(torch::Tensor acts, torch::Tensor c_prev,
   const int64_t M = c_prev.size(0);
   const int H = (int)c_prev.size(1);
   TORCH_CHECK(acts.size(0) == M && acts.size(1) == 4 * H, "acts shape");
+  TORCH_CHECK(H % 4 == 0, "H must be a multiple of 4");
   const float* dhn = dh_next.has_value() ? dh_next->data_ptr<float>() : nullptr;
   const float* dcn = dc_next.has_value() ? dc_next->data_ptr<float>() : nullptr;
   const bool* dn = done.has_value() ? done->data_ptr<bool>() : nullptr;

@@ -1136,19 +1136,37 @@ __global__ void lstm_cell_bwd_kernel(
     __bf16* __restrict__ dgates,          // [M, 4H] out (bf16: feeds wgrad)
     float* __restrict__ dc_prev,          // [M, H] out
     int64_t M, int H) {
-  const int64_t total = M * H;
-  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
-    const int64_t m = idx / H;
+  // one thread owns FOUR consecutive hidden units: acts/dgates move as
+  // bf16x8 pairs, the f32 streams as f32x4 (H % 4 == 0 enforced by the
+  // binding) — the scalar-per-unit version ran at half its traffic floor
+  const int64_t groups = M * (H / 4);
+  for (int64_t gidx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       gidx < groups; gidx += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t m = gidx / (H / 4);
+    const int64_t idx0 = gidx * 4;  // == m * H + u0
     const float mask = (done && done[m]) ? 0.f : 1.f;
-    const bf16x4 av = *reinterpret_cast<const bf16x4*>(&acts[idx * 4]);
-    bf16x4 dg;
-    float dcp;
-    lstm_cell_bwd_math(av, c_prev[idx], c_new[idx], dh_head[idx],
-                       dh_next ? mask * dh_next[idx] : 0.f,
-                       dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
-    *reinterpret_cast<bf16x4*>(&dgates[idx * 4]) = dg;
-    dc_prev[idx] = dcp;
+    const f32x4 cp = *reinterpret_cast<const f32x4*>(&c_prev[idx0]);
+    const f32x4 cn = *reinterpret_cast<const f32x4*>(&c_new[idx0]);
+    const f32x4 dhh = *reinterpret_cast<const f32x4*>(&dh_head[idx0]);
+    f32x4 dhn = {}, dcn = {};
+    if (dh_next) dhn = *reinterpret_cast<const f32x4*>(&dh_next[idx0]);
+    if (dc_next) dcn = *reinterpret_cast<const f32x4*>(&dc_next[idx0]);
+    bf16x4 dg4[4];
+    f32x4 dcp4;
+    for (int j = 0; j < 4; ++j) {
+      const bf16x4 av = *reinterpret_cast<const bf16x4*>(
+          &acts[(idx0 + j) * 4]);
+      float dcp;
+      lstm_cell_bwd_math(av, cp[j], cn[j], dhh[j],
+                         dh_next ? mask * dhn[j] : 0.f,
+                         dc_next ? mask * dcn[j] : 0.f, &dg4[j], &dcp);
+      dcp4[j] = dcp;
+    }
+    *reinterpret_cast<bf16x8*>(&dgates[idx0 * 4]) =
+        *reinterpret_cast<const bf16x8*>(&dg4[0]);
+    *reinterpret_cast<bf16x8*>(&dgates[idx0 * 4 + 8]) =
+        *reinterpret_cast<const bf16x8*>(&dg4[2]);
+    *reinterpret_cast<f32x4*>(&dc_prev[idx0]) = dcp4;
   }
 }
 
@@ -1857,7 +1875,7 @@ void launch_lstm_cell_bwd(const void* acts, const float* c_prev,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
                           int64_t M, int H, hipStream_t stream) {
-  int64_t total = M * H;
+  int64_t total = M * (H / 4);
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_bwd_kernel, dim3(blocks), dim3(256), 0, stream,
                      reinterpret_cast<const __bf16*>(acts),
