@@ -1049,6 +1049,10 @@ GFX_DEV f32x4 gate_pre4(const __bf16* __restrict__ gates_pre,
 // between the two kernels broke bitwise fused==unfused once).
 GFX_DEV void lstm_cell_math(const f32x4 gp, float cp, float* c_out,
                             float* h_out, bf16x4* acts_out) {
+// contraction off: the compiler must not fuse mul+add differently in the
+// different kernels that inline this helper (bitwise fused==unfused);
+// the one fma we DO want is written explicitly below
+#pragma clang fp contract(off)
   const float i = fast_sigmoid(gp[0]);
   const float f = fast_sigmoid(gp[1]);
   const float g = fast_tanh(gp[2]);
@@ -1074,6 +1078,7 @@ GFX_DEV void lstm_cell_math(const f32x4 gp, float cp, float* c_out,
 GFX_DEV void lstm_cell_bwd_math(const bf16x4 acts, float cp, float c,
                                 float dh_head_v, float dh_next_v,
                                 float dc_next_v, bf16x4* dg, float* dcp) {
+#pragma clang fp contract(off)
   const float i = bf2f(acts[0]);
   const float f = bf2f(acts[1]);
   const float g = bf2f(acts[2]);
