@@ -17,8 +17,7 @@ from ..ops import api
 
 def evaluate_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
     from .. import build_vec_environment
-    from ..utils.checkpoint import load_checkpoint
-    from .ppo import PPOConfig, PPOTrainer
+    from .ppo import PPOConfig
 
     cfg = PPOConfig.from_config(config)
     vec_cfg = dict(config)
@@ -26,13 +25,33 @@ def evaluate_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
     vec_cfg.setdefault("env_start_mode", "spread")
     env = build_vec_environment(vec_cfg)
     env.reset(seed=cfg.seed)
-    trainer = PPOTrainer(env, cfg)
     ckpt = config.get("checkpoint_file")
     if ckpt:
-        load_checkpoint(trainer, ckpt)
-        # evaluation starts from a fresh fleet, not the training cursor
-        env.reset(seed=cfg.seed)
-    model = trainer.model
+        # inference-only load: evaluation needs the model weights, not the
+        # training-time n_envs or env cursor (a 4096-env checkpoint must
+        # evaluate on any fleet size)
+        from ..utils.checkpoint import load_model_for_inference
+
+        model, meta = load_model_for_inference(ckpt, env.device)
+        if meta["obs_dim"] != env.obs_dim:
+            raise ValueError(
+                f"checkpoint obs_dim {meta['obs_dim']} != eval env obs_dim "
+                f"{env.obs_dim} (feature/window config mismatch)")
+        recurrent = meta["policy"] == "lstm"
+        policy_name = meta["policy"]
+    else:
+        if cfg.policy == "lstm":
+            from ..models.lstm import ActorCriticLSTM
+
+            model = ActorCriticLSTM(env.obs_dim, 3, cfg.hidden,
+                                    device=env.device, seed=cfg.seed)
+        else:
+            from ..models.mlp import ActorCriticMLP
+
+            model = ActorCriticMLP(env.obs_dim, 3, cfg.hidden,
+                                   device=env.device, seed=cfg.seed)
+        recurrent = cfg.policy == "lstm"
+        policy_name = cfg.policy
     steps = int(config.get("eval_steps", config.get("steps", 500)))
     N, D = env.n_envs, env.obs_dim
     dev = env.device
@@ -40,20 +59,20 @@ def evaluate_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
     actions = torch.empty(N, dtype=torch.int64, device=dev)
     logp = torch.empty(N, dtype=torch.float32, device=dev)
     acts = model.alloc_acts(N)
-    state = model.alloc_state(N) if trainer.recurrent else None
+    state = model.alloc_state(N) if recurrent else None
     total_reward = torch.zeros(N, dtype=torch.float64, device=dev)
     t0 = time.perf_counter()
     n_steps = 0
     for t in range(steps):
         api.f32_to_bf16(env._obs, obs_bf16)
-        if trainer.recurrent:
+        if recurrent:
             head = model.step_forward(obs_bf16, state, acts)
         else:
             head = model.forward(obs_bf16, acts)
         api.sample_head(head, 0, t, actions, logp, greedy=True)
         out = env.step(actions)
         total_reward += out["reward"].to(torch.float64)
-        if trainer.recurrent:
+        if recurrent:
             api.mask_reset(state["h"], state["c"], out["terminated"])
         n_steps += 1
         if bool(env.st.terminated.all()):
@@ -65,7 +84,7 @@ def evaluate_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
     eq = env.st.equity
     return {
         "mode": "inference",
-        "policy_model": cfg.policy,
+        "policy_model": policy_name,
         "checkpoint_file": ckpt,
         "eval_steps": n_steps,
         "n_envs": N,

@@ -210,3 +210,24 @@ def test_periodic_checkpoint_interval(tmp_path):
         ck.save_checkpoint = orig
     # interval saves after updates 2 and 4, final save after 5
     assert saves == [2, 4, 5]
+
+
+def test_evaluate_loads_checkpoint_from_different_n_envs(tmp_path):
+    """mode=inference evaluation only needs model weights: a 16-env
+    checkpoint must evaluate on a different fleet size (same ADVICE class
+    as serving)."""
+    from gymfx_amd.algo.evaluate import evaluate_from_config
+    from gymfx_amd.config import DEFAULT_VALUES
+
+    tr = _make("lstm")
+    tr.train_update()
+    path = str(tmp_path / "e.pt")
+    save_checkpoint(tr, path)
+    cfg = {**DEFAULT_VALUES,
+           "data_feed_plugin": "synthetic_data_feed", "synthetic_rows": 500,
+           "n_envs": 4, "window_size": 8, "device": "cpu", "seed": 9,
+           "checkpoint_file": path, "eval_steps": 12, "hidden_size": 16,
+           "position_size": 1000.0}
+    out = evaluate_from_config(cfg)
+    assert out["policy_model"] == "lstm" and out["n_envs"] == 4
+    assert out["eval_steps"] > 0
