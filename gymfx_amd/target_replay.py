@@ -187,10 +187,15 @@ class TargetReplay:
         maker = taker = profile.commission_rate_per_side
 
         frames_sorted = sorted(frames, key=lambda f: (f.ts_event_ns, f.instrument_id))
+        frame_keys = {(f.ts_event_ns, f.instrument_id) for f in frames_sorted}
         by_ts: Dict[int, List[TargetAction]] = {}
+        unmatched = 0
         for a in actions:
             if a.instrument_id not in specs:
                 raise ValueError(f"action for unknown instrument {a.instrument_id}")
+            if (a.ts_event_ns, a.instrument_id) not in frame_keys:
+                unmatched += 1  # no frame at that timestamp: never applied
+                continue
             by_ts.setdefault(a.ts_event_ns, []).append(a)
 
         cash = Decimal(initial_cash)
@@ -434,6 +439,7 @@ class TargetReplay:
             "final_balance": str(cash + margin_held),
             "final_equity": str(cash + margin_held + unrealized),
             "positions": final_positions,
+            "unmatched_actions": unmatched,
             "events": events,
         }
         result["event_hash"] = hashlib.sha256(
