@@ -212,7 +212,9 @@ class ActorCriticLSTM:
             "gh_scratch": torch.empty(M, 4 * H, dtype=torch.bfloat16, device=dev),
             "head": torch.empty(L * M, self.head_dim, dtype=torch.float32, device=dev),
             "dgates": torch.empty(L, M, 4 * H, dtype=torch.bfloat16, device=dev),
-            "dh_all": torch.empty(L, M, H, dtype=torch.float32, device=dev),
+            # head-dgrad slab bf16: halves the 2x67 MB/minibatch of
+            # write+read traffic on the dh path (cell math still f32)
+            "dh_all": torch.empty(L, M, H, dtype=torch.bfloat16, device=dev),
             # ping-pong recurrent-grad buffers (the fused bwd kernel writes
             # dh for step l-1 while reading this step's dh_next)
             "dh_na": torch.empty(M, H, dtype=torch.float32, device=dev),
@@ -259,7 +261,7 @@ class ActorCriticLSTM:
         api.wgrad(h_raw_flat, dhead, self.grad("Wy"), self.grad("by"),
                   workspace=(dw_p, db_p), slabs=slabs)
         dh_flat = buf["dh_all"].view(L * M, H)
-        api.gemm(dhead, self.w("Wy"), None, dh_flat, act=0, trans_b=True)
+        api.gemm(dhead, self.w("Wy"), None, dh_flat, act=1, trans_b=True)
         # backward through time: ONE fused kernel per step computes dgates
         # (cell bwd) AND the recurrent dgrad dh for step l-1
         dc_next: Optional[torch.Tensor] = None

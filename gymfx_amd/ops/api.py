@@ -164,6 +164,7 @@ def lstm_cell_bwd(
     mask = None
     if done is not None:
         mask = (~done).to(torch.float32).unsqueeze(1)
+    dh_head = dh_head.to(torch.float32)
     # interleaved gate layout: column 4*k + {0=i,1=f,2=g,3=o} (models/lstm.py)
     av = acts.to(torch.float32).view(M, H, 4)
     i = av[..., 0]

@@ -30,7 +30,7 @@ void launch_lstm_cell_fwd(const void* gates_pre, const void* gates_h,
                           void* acts_out, int64_t M, int H,
                           hipStream_t stream);
 void launch_lstm_cell_bwd(const void* acts, const float* c_prev,
-                          const float* c_new, const float* dh_head,
+                          const float* c_new, const void* dh_head,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
                           int64_t M, int H, hipStream_t stream);
@@ -41,7 +41,7 @@ bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
                                float* c_masked, int M, int N, int K,
                                hipStream_t stream);
 bool launch_lstm_bwd_fused(const void* acts, const float* c_prev,
-                           const float* c_new, const float* dh_head,
+                           const float* c_new, const void* dh_head,
                            const float* dh_next, const float* dc_next,
                            const bool* done, const void* B, void* dgates,
                            float* dc_prev, float* dh_prev, int M, int H,
@@ -607,7 +607,7 @@ void lstm_cell_bwd_op(torch::Tensor acts, torch::Tensor c_prev,
   check_bf16(acts, "acts");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
-  check_f32(dh_head, "dh_head");
+  check_bf16(dh_head, "dh_head");
   check_bf16(dgates, "dgates");
   check_f32(dc_prev, "dc_prev");
   const int64_t M = c_prev.size(0);
@@ -619,7 +619,7 @@ void lstm_cell_bwd_op(torch::Tensor acts, torch::Tensor c_prev,
   const bool* dn = done.has_value() ? done->data_ptr<bool>() : nullptr;
   gymfx::launch_lstm_cell_bwd(
       acts.data_ptr(), c_prev.data_ptr<float>(),
-      c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
+      c_new.data_ptr<float>(), dh_head.data_ptr(), dhn, dcn, dn,
       dgates.data_ptr(), dc_prev.data_ptr<float>(), M, H, cur_stream());
 }
 
@@ -675,7 +675,7 @@ bool lstm_bwd_fused_op(torch::Tensor acts, torch::Tensor c_prev,
   check_bf16(acts, "acts");
   check_f32(c_prev, "c_prev");
   check_f32(c_new, "c_new");
-  check_f32(dh_head, "dh_head");
+  check_bf16(dh_head, "dh_head");
   check_bf16(B, "B");
   check_bf16(dgates, "dgates");
   check_f32(dc_prev, "dc_prev");
@@ -688,7 +688,7 @@ bool lstm_bwd_fused_op(torch::Tensor acts, torch::Tensor c_prev,
   float* dhp = dh_prev.has_value() ? dh_prev->data_ptr<float>() : nullptr;
   return gymfx::launch_lstm_bwd_fused(
       acts.data_ptr(), c_prev.data_ptr<float>(),
-      c_new.data_ptr<float>(), dh_head.data_ptr<float>(), dhn, dcn, dn,
+      c_new.data_ptr<float>(), dh_head.data_ptr(), dhn, dcn, dn,
       B.data_ptr(), dgates.data_ptr(), dc_prev.data_ptr<float>(), dhp, M, H,
       cur_stream());
 }

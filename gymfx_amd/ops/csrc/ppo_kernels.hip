@@ -1134,7 +1134,7 @@ __global__ void lstm_cell_bwd_kernel(
     const __bf16* __restrict__ acts,      // [M, 4H] saved activations
     const float* __restrict__ c_prev,     // [M, H] (masked input c)
     const float* __restrict__ c_new,      // [M, H] (raw output c)
-    const float* __restrict__ dh_head,    // [M, H]
+    const __bf16* __restrict__ dh_head,   // [M, H] (head-dgrad slab, bf16)
     const float* __restrict__ dh_next,    // [M, H] or null
     const float* __restrict__ dc_next,    // [M, H] or null (last step)
     const bool* __restrict__ done,        // [M] or null
@@ -1152,7 +1152,9 @@ __global__ void lstm_cell_bwd_kernel(
     const float mask = (done && done[m]) ? 0.f : 1.f;
     const f32x4 cp = *reinterpret_cast<const f32x4*>(&c_prev[idx0]);
     const f32x4 cn = *reinterpret_cast<const f32x4*>(&c_new[idx0]);
-    const f32x4 dhh = *reinterpret_cast<const f32x4*>(&dh_head[idx0]);
+    const bf16x4 dhb = *reinterpret_cast<const bf16x4*>(&dh_head[idx0]);
+    f32x4 dhh;
+    for (int j = 0; j < 4; ++j) dhh[j] = bf2f(dhb[j]);
     f32x4 dhn = {}, dcn = {};
     if (dh_next) dhn = *reinterpret_cast<const f32x4*>(&dh_next[idx0]);
     if (dc_next) dcn = *reinterpret_cast<const f32x4*>(&dc_next[idx0]);
@@ -1382,7 +1384,7 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
     const __bf16* __restrict__ acts,       // [M, 4H] saved activations
     const float* __restrict__ c_prev,      // [M, H]
     const float* __restrict__ c_new,       // [M, H]
-    const float* __restrict__ dh_head,     // [M, H]
+    const __bf16* __restrict__ dh_head,    // [M, H] (bf16 head-dgrad slab)
     const float* __restrict__ dh_next,     // [M, H] or null
     const float* __restrict__ dc_next,     // [M, H] or null
     const bool* __restrict__ done,         // [M] or null
@@ -1420,7 +1422,7 @@ __global__ __launch_bounds__(256, 2) void lstm_bwd_fused_kernel(
         const int64_t idx = (int64_t)grow * H + (u0 + ul);
         const float mask = (done && done[grow]) ? 0.f : 1.f;
         const bf16x4 av = *reinterpret_cast<const bf16x4*>(&acts[idx * 4]);
-        lstm_cell_bwd_math(av, c_prev[idx], c_new[idx], dh_head[idx],
+        lstm_cell_bwd_math(av, c_prev[idx], c_new[idx], bf2f(dh_head[idx]),
                            dh_next ? mask * dh_next[idx] : 0.f,
                            dc_next ? mask * dc_next[idx] : 0.f, &dg, &dcp);
         if (write_dg) {
@@ -1876,7 +1878,7 @@ void launch_lstm_cell_fwd(const void* gates_pre, const void* gates_h,
 }
 
 void launch_lstm_cell_bwd(const void* acts, const float* c_prev,
-                          const float* c_new, const float* dh_head,
+                          const float* c_new, const void* dh_head,
                           const float* dh_next, const float* dc_next,
                           const bool* done, void* dgates, float* dc_prev,
                           int64_t M, int H, hipStream_t stream) {
@@ -1884,7 +1886,8 @@ void launch_lstm_cell_bwd(const void* acts, const float* c_prev,
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
   hipLaunchKernelGGL(lstm_cell_bwd_kernel, dim3(blocks), dim3(256), 0, stream,
                      reinterpret_cast<const __bf16*>(acts),
-                     c_prev, c_new, dh_head, dh_next, dc_next, done,
+                     c_prev, c_new, reinterpret_cast<const __bf16*>(dh_head),
+                     dh_next, dc_next, done,
                      reinterpret_cast<__bf16*>(dgates), dc_prev, M, H);
 }
 
@@ -1916,7 +1919,7 @@ bool launch_lstm_gemm_cell_fwd(const void* A, const void* B,
 }
 
 bool launch_lstm_bwd_fused(const void* acts, const float* c_prev,
-                           const float* c_new, const float* dh_head,
+                           const float* c_new, const void* dh_head,
                            const float* dh_next, const float* dc_next,
                            const bool* done, const void* B, void* dgates,
                            float* dc_prev, float* dh_prev, int M, int H,
@@ -1925,7 +1928,8 @@ bool launch_lstm_bwd_fused(const void* acts, const float* c_prev,
   dim3 grid(ceil_div(M, 64), dh_prev ? H / 64 : 1);
   hipLaunchKernelGGL(lstm_bwd_fused_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<const __bf16*>(acts), c_prev, c_new,
-                     dh_head, dh_next, dc_next, done,
+                     reinterpret_cast<const __bf16*>(dh_head), dh_next,
+                     dc_next, done,
                      reinterpret_cast<const __bf16*>(B),
                      reinterpret_cast<__bf16*>(dgates), dc_prev, dh_prev, M,
                      H);

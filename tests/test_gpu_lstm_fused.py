@@ -71,7 +71,7 @@ def test_fused_bwd_step_bitwise(H, last_step):
     acts = torch.rand(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g, device="cuda")
     c_new = torch.randn(M, H, generator=g, device="cuda")
-    dh_head = torch.randn(M, H, generator=g, device="cuda")
+    dh_head = torch.randn(M, H, generator=g, device="cuda").to(torch.bfloat16)
     dh_next = None if last_step else torch.randn(M, H, generator=g, device="cuda")
     dc_next = None if last_step else torch.randn(M, H, generator=g, device="cuda")
     done = torch.rand(M, generator=g, device="cuda") < 0.25
@@ -105,7 +105,7 @@ def test_fused_bwd_skips_dh_for_step0():
     acts = torch.rand(M, 4 * H, generator=g, device="cuda").to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g, device="cuda")
     c_new = torch.randn(M, H, generator=g, device="cuda")
-    dh_head = torch.randn(M, H, generator=g, device="cuda")
+    dh_head = torch.randn(M, H, generator=g, device="cuda").to(torch.bfloat16)
     dg = torch.empty(M, 4 * H, dtype=torch.bfloat16, device="cuda")
     dcp = torch.empty(M, H, device="cuda")
     ok = native.require().lstm_bwd_fused(acts, c_prev, c_new, dh_head,

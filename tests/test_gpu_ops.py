@@ -354,7 +354,7 @@ def test_lstm_cell_kernels_match_cpu_oracle():
     g = torch.Generator().manual_seed(5)
     gates = torch.randn(M, 4 * H, generator=g).to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g)
-    dh_head = torch.randn(M, H, generator=g)
+    dh_head = torch.randn(M, H, generator=g).to(torch.bfloat16)
     dh_next = torch.randn(M, H, generator=g)
     dc_next = torch.randn(M, H, generator=g)
     done = torch.rand(M, generator=g) < 0.1

@@ -63,7 +63,7 @@ def test_lstm_cell_bwd_done_masks_recurrent_grads():
     acts = torch.rand(M, 4 * H, generator=g).to(torch.bfloat16)
     c_prev = torch.randn(M, H, generator=g)
     c_new = torch.randn(M, H, generator=g)
-    dh_head = torch.randn(M, H, generator=g)
+    dh_head = torch.randn(M, H, generator=g).to(torch.bfloat16)
     dh_next = torch.randn(M, H, generator=g)
     dc_next = torch.randn(M, H, generator=g)
     done = torch.zeros(M, dtype=torch.bool)
