@@ -344,11 +344,14 @@ class PPOTrainer:
                 value_out=self.val_buf[t] if fuse else None,
                 step_base=self.step_base if fuse else None,
                 sample_seed=self.sample_seed, sample_step=t,
+                # fresh episode -> zero recurrent state, fused into the
+                # step kernel (saves a mask_reset launch per rollout step;
+                # state["h"] ping-pongs per step and the recorded pointer
+                # is the buffer the NEXT step reads — same tensor the old
+                # mask_reset call saw at this point in the loop)
+                rnn_h=state["h"] if rec else None,
+                rnn_c=state["c"] if rec else None,
             )
-            if rec:
-                # fresh episode -> zero recurrent state
-                api.mask_reset(state["h"], state["c"],
-                               self.done_buf[t][lo:hi])
 
     def _rollout_body(self) -> None:
         env, model = self.env, self.model

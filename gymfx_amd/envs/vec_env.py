@@ -178,7 +178,9 @@ class VecFxEnv:
              fuse_obs: bool = False,
              h2: Optional[torch.Tensor] = None,
              w3t: Optional[torch.Tensor] = None,
-             b3: Optional[torch.Tensor] = None) -> Dict[str, torch.Tensor]:
+             b3: Optional[torch.Tensor] = None,
+             rnn_h: Optional[torch.Tensor] = None,
+             rnn_c: Optional[torch.Tensor] = None) -> Dict[str, torch.Tensor]:
         """Advance all envs. Returns dict with obs/reward/terminated tensors.
 
         reward_out / terminated_out / obs_bf16_out: optional preallocated
@@ -188,7 +190,11 @@ class VecFxEnv:
         head / logp_out / value_out / step_base / sample_seed / sample_step:
         fused policy sampling (native engine only) — the step kernel samples
         the action from `head` itself, writing into `actions`, saving one
-        kernel launch per rollout step."""
+        kernel launch per rollout step.
+
+        rnn_h / rnn_c: recurrent-state autoreset fused into the step — the
+        kernel zeros the terminated env's LSTM state rows [env_lo:env_hi)
+        itself (replaces the trainer's per-step mask_reset launch)."""
         if not isinstance(actions, torch.Tensor):
             actions = torch.as_tensor(actions, device=self.device)
         actions = actions.to(self.device)
@@ -198,7 +204,8 @@ class VecFxEnv:
                                      terminated_out, obs_bf16_out,
                                      env_lo, env_hi, head, logp_out,
                                      value_out, step_base, sample_seed,
-                                     sample_step, fuse_obs, h2, w3t, b3)
+                                     sample_step, fuse_obs, h2, w3t, b3,
+                                     rnn_h, rnn_c)
             info["obs"] = self._obs
             return info
         if head is not None or h2 is not None:
@@ -212,6 +219,11 @@ class VecFxEnv:
             if bool(done.any()):
                 reset_state_(self.st, self.params, done)
             info["terminated"] = done
+        if rnn_h is not None:
+            # same semantics as the native kernel's fused state reset
+            done_rows = info["terminated"]
+            rnn_h[done_rows] = 0
+            rnn_c[done_rows] = 0
         self._build_obs()
         info["obs"] = self._obs
         if reward_out is not None:

@@ -312,7 +312,9 @@ struct GymFxEngine {
                 int64_t sample_seed, int64_t sample_step, bool fuse_obs,
                 c10::optional<torch::Tensor> h2,
                 c10::optional<torch::Tensor> w3t,
-                c10::optional<torch::Tensor> b3) {
+                c10::optional<torch::Tensor> b3,
+                c10::optional<torch::Tensor> rnn_h,
+                c10::optional<torch::Tensor> rnn_c) {
     TORCH_CHECK(actions.is_cuda() == state["cursor"].is_cuda(),
                 "actions must live on the env device");
     TORCH_CHECK(actions.numel() == K.n_envs, "actions numel != n_envs");
@@ -338,6 +340,27 @@ struct GymFxEngine {
     P.logp_out = nullptr;
     P.value_out = nullptr;
     P.step_base = nullptr;
+    P.rnn_h = nullptr;
+    P.rnn_c = nullptr;
+    K.rnn_hidden = 0;
+    if (rnn_h.has_value()) {
+      // fused recurrent-state autoreset: the step kernel zeros the
+      // terminated env's h/c rows itself (replaces the per-step
+      // mask_reset launch in the recurrent rollout)
+      TORCH_CHECK(rnn_c.has_value(), "rnn_h needs rnn_c");
+      const int64_t cnt = env_hi - env_lo;
+      TORCH_CHECK(rnn_h->is_contiguous() &&
+                      rnn_h->scalar_type() == torch::kBFloat16 &&
+                      rnn_h->dim() == 2 && rnn_h->size(0) == cnt,
+                  "rnn_h must be contiguous bf16 [env_cnt, H]");
+      TORCH_CHECK(rnn_c->is_contiguous() &&
+                      rnn_c->scalar_type() == torch::kFloat32 &&
+                      rnn_c->sizes() == rnn_h->sizes(),
+                  "rnn_c must be contiguous f32 [env_cnt, H]");
+      P.rnn_h = rnn_h->data_ptr();
+      P.rnn_c = rnn_c->data_ptr<float>();
+      K.rnn_hidden = (int)rnn_h->size(1);
+    }
     if (h2.has_value()) {
       // head-in-step fusion: the kernel computes head = h2 @ W3 + b3
       TORCH_CHECK(!continuous, "fused sampling requires discrete actions");
@@ -1114,7 +1137,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("step_base") = py::none(), py::arg("sample_seed") = 0,
            py::arg("sample_step") = 0, py::arg("fuse_obs") = false,
            py::arg("h2") = py::none(), py::arg("w3t") = py::none(),
-           py::arg("b3") = py::none())
+           py::arg("b3") = py::none(), py::arg("rnn_h") = py::none(),
+           py::arg("rnn_c") = py::none())
       .def("build_obs", &gymfx::GymFxEngine::build_obs, py::arg("obs_out"),
            py::arg("obs_bf16_out") = py::none(), py::arg("env_lo") = 0,
            py::arg("env_hi") = 0);

@@ -155,6 +155,7 @@ struct EnvParamsK {
   double margin_init_rate;
   // fused policy sampling (set per step() call when head != null)
   int head_hidden;           // h2 width for the head-in-step fusion
+  int rnn_hidden;            // LSTM state width for the fused state reset
   unsigned long long sample_seed;
   long long sample_step;
   int sample_nact;
@@ -215,6 +216,12 @@ struct EnvPtrs {
   const void *h2;            // [env_cnt, head_hidden] bf16
   const void *w3t;           // [sample_nact+1, head_hidden] bf16
   const float *b3;           // [sample_nact+1]
+  // recurrent-state autoreset: when set, the step kernel zeros the
+  // terminated env's LSTM state rows itself — one fewer launch per
+  // rollout step than the separate mask_reset kernel (done rows are
+  // rare, so the per-lane zero loop is almost always skipped)
+  void *rnn_h;               // [env_cnt, rnn_hidden] bf16 (rows local to env_lo)
+  float *rnn_c;              // [env_cnt, rnn_hidden] f32
   int64_t *actions_out;      // sampled action (same buffer `actions` reads)
   float *logp_out;           // [N] f32
   float *value_out;          // [N] f32

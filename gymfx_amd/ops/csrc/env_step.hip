@@ -573,6 +573,18 @@ GFX_DEV void env_step_one(const EnvPtrs& P, const EnvParamsK& K, const int n,
   P.terminated_out[n] = P.terminated[n];
   P.coerced_out[n] = a;
 
+  // ---- 7b. recurrent-state autoreset (mask_reset fused in) -----------
+  // zero the terminated env's LSTM h/c rows here instead of launching a
+  // separate mask_reset kernel after every rollout step (the separate
+  // launch is pure launch latency: done rows are rare).
+  if (P.rnn_h != nullptr && P.terminated[n]) {
+    __bf16 *hrow = reinterpret_cast<__bf16 *>(P.rnn_h) +
+                   (int64_t)(n - env_lo) * K.rnn_hidden;
+    float *crow = P.rnn_c + (int64_t)(n - env_lo) * K.rnn_hidden;
+    for (int j = 0; j < K.rnn_hidden; ++j) hrow[j] = (__bf16)0.f;
+    for (int j = 0; j < K.rnn_hidden; ++j) crow[j] = 0.f;
+  }
+
   // ---- 8. autoreset (training path; no reference counterpart) --------
   if ((K.flags & F_AUTORESET) && P.terminated[n]) reset_env(P, K, n);
 }

@@ -186,3 +186,41 @@ def test_equity_accounting_identity_property():
             prev_comm = comm
 
     check()
+
+
+def test_fused_rnn_state_reset_matches_mask_reset(data_files):
+    """env.step(rnn_h=, rnn_c=) zeros exactly the terminated rows — the
+    same semantics as api.mask_reset after the step (the fused form the
+    recurrent rollout uses; ppo.py:_rollout_half)."""
+    import torch
+
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.config import DEFAULT_VALUES
+    from gymfx_amd.ops import api
+
+    from gymfx_amd.data.feed import synthetic_ohlcv, write_csv
+    tiny = data_files / "tiny.csv"
+    write_csv(synthetic_ohlcv(20, seed=9, vol=2e-4), str(tiny))
+    cfg = {**DEFAULT_VALUES, "n_envs": 8, "device": "cpu",
+           "autoreset": True, "quiet_mode": True, "window_size": 8,
+           "input_data_file": str(tiny)}
+    env_a = build_vec_environment(cfg)
+    env_b = build_vec_environment(cfg)
+    env_a.reset(seed=3)
+    env_b.reset(seed=3)
+    N, H = 8, 16
+    g = torch.Generator().manual_seed(5)
+    h_a = torch.randn(N, H, generator=g).to(torch.bfloat16)
+    c_a = torch.randn(N, H, generator=g)
+    h_b, c_b = h_a.clone(), c_a.clone()
+    acts = torch.ones(N, dtype=torch.int64)
+    hit = False
+    for _ in range(25):
+        out_a = env_a.step(acts, rnn_h=h_a, rnn_c=c_a)
+        out_b = env_b.step(acts)
+        api.mask_reset(h_b, c_b, out_b["terminated"])
+        assert torch.equal(out_a["terminated"], out_b["terminated"])
+        assert torch.equal(h_a, h_b) and torch.equal(c_a, c_b)
+        hit = hit or bool(out_a["terminated"].any())
+    assert hit, "20-bar data must terminate at least once in 25 steps"
+    assert (h_a[out_a["terminated"]] == 0).all()
