@@ -183,7 +183,8 @@ class NativeEngine:
              step_base: torch.Tensor = None, sample_seed: int = 0,
              sample_step: int = 0, fuse_obs: bool = False,
              h2: torch.Tensor = None, w3t: torch.Tensor = None,
-             b3: torch.Tensor = None) -> Dict[str, torch.Tensor]:
+             b3: torch.Tensor = None, rnn_h: torch.Tensor = None,
+             rnn_c: torch.Tensor = None) -> Dict[str, torch.Tensor]:
         if self._params.action_space_mode == "continuous":
             actions = actions.to(torch.float32).contiguous()
         else:
@@ -192,7 +193,8 @@ class NativeEngine:
                                       terminated_out, obs_bf16_out,
                                       env_lo, env_hi, head, logp_out,
                                       value_out, step_base, sample_seed,
-                                      sample_step, fuse_obs, h2, w3t, b3))
+                                      sample_step, fuse_obs, h2, w3t, b3,
+                                      rnn_h, rnn_c))
 
     def build_obs(self, obs_out: torch.Tensor,
                   obs_bf16_out: torch.Tensor = None) -> None:

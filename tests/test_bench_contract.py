@@ -44,3 +44,28 @@ def test_bench_lstm_variant_contract():
     assert "LSTM" in d["config"]["model"]
     assert d["config"]["reward"] == "dd_penalized_reward"
     assert d["config"]["strategy"] == "direct_atr_sltp"
+
+
+def test_bench_torchrun_ws2_contract():
+    """The driver's exact SCALE invocation shape (torch.distributed.run,
+    nnodes=1, nproc-per-node N, 127.0.0.1 rendezvous): rank 0 prints ONE
+    JSON line, parallelism reflects the world size, and the run exits 0
+    on the gloo fallback here (nccl=RCCL path on a GPU box)."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), str(REPO / "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--n-envs", "32", "--rollout", "8"],
+        capture_output=True, text=True, timeout=420, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout
+    d = json.loads(lines[0])
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0
