@@ -57,12 +57,25 @@ def run_env(config: Dict[str, Any]) -> Dict[str, Any]:
         done = False
         steps = int(config.get("steps", 500))
         step_count = 0
+        plugin_apply_errors = 0
         while not done and step_count < steps:
-            action = strategy.decide_action(obs=obs, info=info, step=step_count)
+            try:
+                action = strategy.decide_action(obs=obs, info=info,
+                                                step=step_count)
+            except Exception:
+                # graceful degradation, never crash mid-rollout: a failing
+                # (third-party) strategy plugin falls back to hold and is
+                # counted, matching the reference's plugin_apply_errors
+                # semantics (bt_bridge.py:191-201)
+                plugin_apply_errors += 1
+                action = 0
             obs, _, terminated, truncated, info = env.step(action)
             done = bool(terminated or truncated)
             step_count += 1
-        return env.summary()
+        summary = env.summary()
+        if plugin_apply_errors:
+            summary["plugin_apply_errors"] = plugin_apply_errors
+        return summary
     finally:
         env.close()
 

@@ -81,13 +81,20 @@ class DirectFixedSLTP(PluginBase):
         "position_size": 1.0,
     }
 
+    def __init__(self, config: Dict[str, Any] | None = None):
+        super().__init__(config)
+        # the runner's driver_mode drives actions even when a bracket
+        # strategy converts them to orders (reference app/main.py:57-66:
+        # the driver loop is independent of the strategy plugin)
+        self._driver = DefaultStrategy(config)
+
     def set_params(self, **kwargs: Any) -> None:
         for k, v in kwargs.items():
             if k in self.plugin_params:
                 self.params[k] = v
 
     def decide_action(self, obs, info, step: int) -> int:
-        return 0  # bracket manager, not a driver
+        return self._driver.decide_action(obs, info, step)
 
 
 class DirectAtrSLTP(PluginBase):
@@ -122,13 +129,18 @@ class DirectAtrSLTP(PluginBase):
         "force_close_hour": 20,
     }
 
+    def __init__(self, config: Optional[Dict[str, Any]] = None):
+        super().__init__(config)
+        # same driver delegation as DirectFixedSLTP (app/main.py:57-66)
+        self._driver = DefaultStrategy(config)
+
     def set_params(self, **kwargs: Any) -> None:
         for k, v in kwargs.items():
             if k in self.plugin_params:
                 self.params[k] = v
 
     def decide_action(self, obs, info, step: int) -> int:
-        return 0
+        return self._driver.decide_action(obs, info, step)
 
     def effective_sltp_multiples(self, config: Optional[Dict[str, Any]] = None):
         """Scalar oracle for the risk-mode shrink math

@@ -224,3 +224,48 @@ def test_fused_rnn_state_reset_matches_mask_reset(data_files):
         hit = hit or bool(out_a["terminated"].any())
     assert hit, "20-bar data must terminate at least once in 25 steps"
     assert (h_a[out_a["terminated"]] == 0).all()
+
+
+def test_plugin_apply_error_graceful_fallback(data_files):
+    """A strategy plugin that raises mid-run must not crash the driver:
+    the step falls back to hold and the summary reports
+    plugin_apply_errors (reference bt_bridge.py:191-201 semantics)."""
+    from gymfx_amd.main import run_env
+    from gymfx_amd.plugins.strategies import DefaultStrategy
+
+    calls = {"n": 0}
+    orig = DefaultStrategy.decide_action
+
+    def flaky(self, obs, info, step):
+        calls["n"] += 1
+        if step in (2, 5):
+            raise RuntimeError("plugin exploded")
+        return orig(self, obs, info, step)
+
+    DefaultStrategy.decide_action = flaky
+    try:
+        s = run_env(_cfg(data_files, "flat", steps=10))
+    finally:
+        DefaultStrategy.decide_action = orig
+    assert calls["n"] == 10
+    assert s["plugin_apply_errors"] == 2
+    assert s["trades_total"] == 0  # fallback action is hold
+
+
+def test_bracket_audit_jsonl(data_files, tmp_path):
+    """GYMFX_BRACKET_AUDIT parity (direct_atr_sltp.py:40-50): bracket
+    arms and exits are appended as JSONL records."""
+    import json
+
+    audit = tmp_path / "audit.jsonl"
+    s = run_env(_cfg(
+        data_files, "random",
+        strategy_plugin="direct_fixed_sltp",
+        sl_pips=2.0, tp_pips=2.0, seed=5, steps=200,
+        bracket_audit_file=str(audit)))
+    assert audit.exists(), "audit file must be written"
+    recs = [json.loads(l) for l in audit.read_text().splitlines()]
+    kinds = {r["event"] for r in recs}
+    assert "bracket_armed" in kinds, kinds
+    armed = [r for r in recs if r["event"] == "bracket_armed"]
+    assert all("sl" in r and "tp" in r and "bar_index" in r for r in armed)
