@@ -162,6 +162,9 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(
   // (row, col) decomposition reduces to the classic tid>>2 / (tid&3)*8
   constexpr int ALPT = (BM * BK) / 2048;  // bf16x8 loads per thread, A tile
   constexpr int BLPT = (BN * BK) / 2048;  // bf16x8 loads per thread, B tile
+  // NFRAG=1 (BN=32) tiles need BKT=64 so every thread still stages >=1
+  // bf16x8 of B (32*32/2048 would truncate to 0 loads)
+  static_assert(ALPT >= 1 && BLPT >= 1, "tile too small for 256-thread staging");
 
   bf16x8 ra[ALPT], rb[BLPT];
 
@@ -2061,6 +2064,23 @@ void launch_gemm(const void* A, const void* B, const float* bias, void* C,
                     (bk64_env == 1 || (bk64_env != 0 && K >= 512));
   dim3 grid(ceil_div(M, 64), ceil_div(N, wide ? 256 : (mid ? 128 : 64)));
   dim3 block(256);
+  // Narrow 64x32 tile (NFRAG=1, BK=64 only — see the BLPT static_assert):
+  // at N=256 the 64x64 tile fills exactly 1 block/CU (the LSTM bwd-chain
+  // dgrad: 64x4 = 256 workgroups) while occupancy allows more; BN=32
+  // doubles the grid for latency hiding at 2x B re-streaming.
+  // GYMFX_GEMM_N1=1 forces on, =0 off, unset = auto (M*N small, long K).
+  static const int n1_env = [] {
+    const char* e = getenv("GYMFX_GEMM_N1");
+    return e ? atoi(e) : -1;
+  }();
+  if (bk64 && act == 0 && !dact_tanh && !add_bias && !accum && !fmp &&
+      N <= 256 &&
+      (n1_env == 1 || (n1_env == -1 && (int64_t)M * N <= 64 * 64 * 256))) {
+    dim3 grid_n1(ceil_div(M, 64), ceil_div(N, 32));
+    hipLaunchKernelGGL((gemm_kernel<true, 0, false, false, 1, false, 64>),
+                       grid_n1, block, 0, stream, a, b, bias, C, y, M, N, K);
+    return;
+  }
   if (fmp) {
     // gather+first-GEMM fusion: only the L1 forward combo is instantiated
     // (trans_b, tanh epilogue, bias, 64x64 tile)
