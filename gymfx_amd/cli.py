@@ -9,7 +9,9 @@ def parse_args(argv=None):
     parser = argparse.ArgumentParser(
         description="gymfx-amd env runtime (MI355X-native, env + PPO training)."
     )
-    parser.add_argument("--mode", choices=["training", "optimization", "inference"])
+    parser.add_argument("--mode",
+                        choices=["training", "optimization", "inference",
+                                 "serve"])
     parser.add_argument("--driver_mode", choices=["random", "buy_hold", "flat", "replay"])
     parser.add_argument("--steps", type=int)
 

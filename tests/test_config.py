@@ -271,3 +271,11 @@ def test_optimization_refinement_narrows_around_best():
         for k, v in r["params"].items():
             lo, hi = schema[k]
             assert lo - 1e-9 <= float(v) <= hi + 1e-9
+
+
+def test_cli_mode_serve_accepted():
+    from gymfx_amd.cli import parse_args
+
+    args, unknown = parse_args(["--mode", "serve", "--serve_port", "9999"])
+    assert args.mode == "serve"
+    assert unknown == ["--serve_port", "9999"]
