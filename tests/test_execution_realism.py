@@ -194,3 +194,46 @@ def test_profile_policy_fields_reach_the_engine(tmp_path):
     assert p.margin_model == 1
     assert p.latency_bars == 3          # 180 s of latency on 1-minute bars
     assert p.enforce_margin_preflight
+
+
+def test_policy_cross_product_reconciles_property():
+    """Hypothesis sweep over the FULL policy cross-product (collision x
+    limit x latency x margin-model x costs) with random walks and random
+    action strings: the engine must reconcile against the independent
+    ScalarLedger for every combination, not just the hand-picked configs
+    above."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=40, deadline=None)
+    @given(
+        st.sampled_from(["worst_case", "ohlc", "adaptive"]),
+        st.sampled_from(["touch", "cross", "conservative"]),
+        st.sampled_from([0, 60_000, 180_000]),
+        st.sampled_from(["leveraged", "standard"]),
+        st.integers(min_value=0, max_value=2 ** 31 - 1),
+        st.lists(st.integers(min_value=0, max_value=2),
+                 min_size=4, max_size=24),
+    )
+    def check(collision, limit, latency_ms, margin, seed, actions):
+        rng = np.random.default_rng(seed)
+        n = max(len(actions) + 6, 16)
+        o = 1.0 + np.cumsum(rng.normal(0, 8e-4, size=n))
+        h = o + np.abs(rng.normal(0, 6e-4, size=n))
+        lo = o - np.abs(rng.normal(0, 6e-4, size=n))
+        c = np.clip(o + rng.normal(0, 4e-4, size=n), lo, h)
+        md = _md(list(zip(o, h, lo, c)))
+        cfg = {**BASE,
+               "intrabar_collision_policy": collision,
+               "limit_fill_policy": limit,
+               "latency_ms": latency_ms,
+               "margin_model": margin,
+               "margin_init_rate": 0.05,
+               "commission": 2e-5, "slippage": 1e-5,
+               "leverage": 20.0}
+        result = ReplayAdapter().run(cfg, md, actions)
+        assert result["reconciled"], (
+            collision, limit, latency_ms, margin, seed,
+            result["reconciliation"])
+
+    check()
