@@ -442,3 +442,75 @@ def test_cross_engine_bakeoff_reconciles():
     out = run_bakeoff(rows=250, steps=80, seed=13)
     assert out["vec_vs_ledger_ok"]
     assert out["vec_vs_target_replay_ok"], out
+
+
+def test_event_stream_reconstructs_final_balance_property():
+    """Independent reconciliation from IMMUTABLE fill facts (the
+    reference's reconcile_fills idiom, bakeoff.py:228-303): rebuild the
+    account from the event log alone — average-price netting over
+    order_filled quantities/prices/commissions plus financing amounts —
+    and it must land on the engine's final_balance for random walks and
+    random target scripts."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=30, deadline=None)
+    @given(st.integers(min_value=0, max_value=2 ** 31 - 1),
+           st.lists(st.sampled_from([-2000, -1000, 0, 1000, 2000]),
+                    min_size=2, max_size=12))
+    def check(seed, targets):
+        import numpy as np
+
+        rng = np.random.default_rng(seed)
+        n = len(targets) + 4
+        mids = 1.1 + np.cumsum(rng.normal(0, 5e-4, size=n))
+        frames = []
+        for i, m in enumerate(mids):
+            o = m
+            h = m + abs(rng.normal(0, 4e-4))
+            l = m - abs(rng.normal(0, 4e-4))
+            c = min(max(m + rng.normal(0, 2e-4), l), h)
+            frames.append(_frame(i, f"{o:.5f}", f"{h:.5f}", f"{l:.5f}",
+                                 f"{c:.5f}"))
+        actions = [TargetAction("EUR/USD.SIM", _ts(i + 1),
+                                Decimal(t), f"a{i}")
+                   for i, t in enumerate(targets)]
+        prof = _profile(commission_rate_per_side="0.00002",
+                        slippage_bps_per_side="0.1",
+                        random_seed=seed % 1000)
+        res = TargetReplay(prof).run(
+            instrument_specs=[EURUSD], frames=frames, actions=actions,
+            initial_cash=Decimal("100000"))
+
+        # -- rebuild from events only (conv = 1: USD quote) --------------
+        bal = Decimal("100000")
+        units = Decimal(0)
+        avg = Decimal(0)
+        for ev in res["events"]:
+            if ev["event_type"] == "financing":
+                bal += Decimal(ev["amount"])
+            if ev["event_type"] != "order_filled":
+                continue
+            qty = Decimal(ev["quantity"])
+            px = Decimal(ev["price"])
+            signed = qty if ev["side"] == "BUY" else -qty
+            bal -= Decimal(ev["commission"])
+            if units != 0 and units * signed < 0:
+                closing = min(abs(units), qty)
+                bal += closing * (px - avg) * (1 if units > 0 else -1)
+                new_units = units + signed
+                if new_units == 0:
+                    avg = Decimal(0)
+                elif units * new_units < 0:
+                    avg = px
+            else:
+                new_units = units + signed
+                if units == 0:
+                    avg = px
+                elif new_units != 0:
+                    avg = (abs(units) * avg + qty * px) / abs(new_units)
+            units = new_units
+        assert abs(bal - Decimal(res["final_balance"])) <= Decimal("1e-6"), (
+            seed, targets, str(bal), res["final_balance"])
+
+    check()
