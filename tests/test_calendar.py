@@ -92,3 +92,48 @@ def test_calendar_feature_ranges_property():
             assert f["is_no_new_position_window"] == 1.0
 
     check()
+
+
+def test_window_start_boundaries_exact_ny_times():
+    """The reference pins its Friday windows to exact NY wall-clock starts
+    (oanda_calendar.py:30-48; tests test_friday_*_window_starts_at_*):
+    no-new-position at 14:00, risk-reduction at 15:00, force-flat at
+    15:45 — one minute before each start the flag must be off, at the
+    start it must be on.  Checked in BOTH EST (Jan, UTC-5) and EDT
+    (Jul, UTC-4) so a fixed-UTC-offset implementation fails."""
+    from gymfx_amd.calendar import (is_broker_daily_break_near,
+                                    is_friday_risk_reduction_window,
+                                    is_no_trade_window)
+
+    for friday, off in ((dt.datetime(2024, 1, 5), 5),   # EST
+                        (dt.datetime(2024, 7, 5), 4)):  # EDT
+        def ny(hh, mm):
+            return _to_ny(_epoch(friday.year, friday.month, friday.day,
+                                 hh + off, mm))
+
+        for pred, (hh, mm) in ((is_no_new_position_window, (14, 0)),
+                               (is_friday_risk_reduction_window, (15, 0)),
+                               (is_force_flat_window, (15, 45))):
+            before = ny(hh, mm - 1) if mm else ny(hh - 1, 59)
+            assert not pred(before), (pred.__name__, "one minute early", off)
+            assert pred(ny(hh, mm)), (pred.__name__, "at start", off)
+        # daily break near 16:59 NY and the 16:50-17:10 no-trade window
+        assert is_broker_daily_break_near(ny(16, 59))
+        assert is_no_trade_window(ny(16, 55))
+        assert is_no_trade_window(ny(17, 5))
+        assert not is_no_trade_window(ny(16, 45))
+        assert not is_no_trade_window(ny(17, 15))
+
+
+def test_calendar_bars_scale_with_timeframe():
+    """bars_to_* features are hour-countdowns divided by the timeframe:
+    halving timeframe_hours doubles the bar counts (reference
+    test_feature_dict_keys_complete_and_bars_scale_with_timeframe)."""
+    ts = _epoch(2024, 1, 3, 12, 0)  # Wednesday, mid-session
+    f1 = compute_fx_calendar_features(ts, timeframe_hours=1.0)
+    f2 = compute_fx_calendar_features(ts, timeframe_hours=0.5)
+    assert set(f1) == set(CALENDAR_FEATURE_KEYS)
+    for k in CALENDAR_FEATURE_KEYS:
+        if k.startswith("bars_"):
+            assert f2[k] == pytest.approx(2.0 * f1[k], rel=1e-9), k
+            assert f1[k] > 0.0
