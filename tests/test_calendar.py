@@ -137,3 +137,60 @@ def test_calendar_bars_scale_with_timeframe():
         if k.startswith("bars_"):
             assert f2[k] == pytest.approx(2.0 * f1[k], rel=1e-9), k
             assert f1[k] > 0.0
+
+
+def test_broker_profile_auto_enables_calendar_obs():
+    """broker_profile='oanda_us_fx' flips the calendar-obs flag on without
+    setting oanda_fx_calendar_obs explicitly (reference
+    test_oanda_fx_broker_profile_auto_enables_calendar_obs), and the info
+    dict carries the calendar block + broker metadata; without either
+    flag the calendar keys must be absent."""
+    import torch
+
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    md = synthetic_ohlcv(60, seed=3, vol=1e-4)
+    base = {"n_envs": 1, "device": "cpu", "window_size": 4,
+            "timeframe_hours": 1.0}
+    on = build_vec_environment({**base, "broker_profile": "oanda_us_fx"}, md)
+    assert on.params.oanda_fx_calendar_obs
+    off = build_vec_environment(dict(base), md)
+    assert not off.params.oanda_fx_calendar_obs
+
+    import tempfile
+
+    from gymfx_amd import build_environment
+    from gymfx_amd.config import DEFAULT_VALUES
+    from gymfx_amd.data.feed import write_csv
+    from gymfx_amd.plugins import load_plugin
+
+    d = tempfile.mkdtemp()
+    write_csv(md, d + "/cal.csv")
+
+    def gym_env(extra):
+        cfg = {**DEFAULT_VALUES, **base, "input_data_file": d + "/cal.csv",
+               "quiet_mode": True, **extra}
+        plugins = {}
+        for group, key in [("data_feed.plugins", "data_feed_plugin"),
+                           ("broker.plugins", "broker_plugin"),
+                           ("strategy.plugins", "strategy_plugin"),
+                           ("preprocessor.plugins", "preprocessor_plugin"),
+                           ("reward.plugins", "reward_plugin"),
+                           ("metrics.plugins", "metrics_plugin")]:
+            klass, _ = load_plugin(group, cfg[key])
+            plugins[key] = klass(cfg)
+        return build_environment(config=cfg, **plugins)
+
+    env = gym_env({"broker_profile": "oanda_us_fx"})
+    env.reset()
+    _, _, _, _, info = env.step(0)
+    assert "hours_to_friday_close" in info
+    assert info["broker_profile"] == "oanda_us_fx"
+    assert "margin_available_norm" in info
+    env.close()
+    env2 = gym_env({})
+    env2.reset()
+    _, _, _, _, info2 = env2.step(0)
+    assert "hours_to_friday_close" not in info2
+    env2.close()
