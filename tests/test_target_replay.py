@@ -514,3 +514,38 @@ def test_event_stream_reconstructs_final_balance_property():
             seed, targets, str(bal), res["final_balance"])
 
     check()
+
+
+def test_future_market_mutation_cannot_change_earlier_fill_facts():
+    """No-future-leakage at the engine level (reference
+    test_nautilus_bakeoff.py:124-156): scaling the LAST bar's prices x5
+    must leave every event with an earlier timestamp byte-identical."""
+    import numpy as np
+    from dataclasses import replace
+
+    rng = np.random.default_rng(17)
+    n = 20
+    mids = 1.1 + np.cumsum(rng.normal(0, 5e-4, size=n))
+    frames = [_frame(i, f"{m:.5f}", f"{m + 3e-4:.5f}", f"{m - 3e-4:.5f}",
+                     f"{m:.5f}") for i, m in enumerate(mids)]
+    actions = [TargetAction("EUR/USD.SIM", _ts(i + 1),
+                            Decimal([1000, -1000, 0, 2000][i % 4]), f"a{i}")
+               for i in range(12)]
+    prof = _profile(commission_rate_per_side="0.00002",
+                    slippage_bps_per_side="0.1")
+
+    def run(fr):
+        return TargetReplay(prof).run(
+            instrument_specs=[EURUSD], frames=fr, actions=actions,
+            initial_cash=Decimal("100000"))
+
+    cutoff = max(f.ts_event_ns for f in frames)
+    base = run(frames)
+    mutated_frames = [
+        replace(f, open=f.open * 5, high=f.high * 5, low=f.low * 5,
+                close=f.close * 5) if f.ts_event_ns == cutoff else f
+        for f in frames]
+    mut = run(mutated_frames)
+    base_prefix = [e for e in base["events"] if e["ts_event_ns"] < cutoff]
+    mut_prefix = [e for e in mut["events"] if e["ts_event_ns"] < cutoff]
+    assert base_prefix == mut_prefix and len(base_prefix) > 4
