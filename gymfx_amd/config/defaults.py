@@ -69,6 +69,7 @@ DEFAULT_VALUES = {
     "min_equity": None,          # default: initial_cash * 0.01
     "financing_enabled": False,  # FX rollover interest at 22:00 UTC
     "rollover_rate_data": None,  # monthly central-bank rates (LOCATION/TIME/Value)
+    "financing_rate_data_file": None,  # or a CSV path with the same columns
     "rollover_hour_utc": 22,
     "enforce_margin_preflight": False,  # deny fills lacking free margin
 

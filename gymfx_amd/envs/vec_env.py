@@ -102,7 +102,18 @@ class VecFxEnv:
             raise ValueError("financing_enabled requires bar timestamps")
         rate_data = self.config.get("rollover_rate_data")
         if not rate_data:
-            raise ValueError("financing_enabled requires rollover_rate_data")
+            # reference-style CSV path (LOCATION,TIME,Value monthly rows —
+            # the config key of tests/test_nautilus_gym_bridge.py:24-26)
+            path = self.config.get("financing_rate_data_file")
+            if path:
+                import csv
+
+                with open(path, newline="", encoding="utf-8") as fh:
+                    rate_data = list(csv.DictReader(fh))
+        if not rate_data:
+            raise ValueError(
+                "financing_enabled requires rollover_rate_data (inline) or "
+                "financing_rate_data_file (CSV path)")
         sched = np.zeros(md.n_rows, dtype=np.float32)
         for b in self.instrument_blocks:
             sched[b["lo"]:b["end"]] = compute_rollover_schedule(

@@ -58,3 +58,52 @@ def test_financing_requires_rate_data():
            "financing_enabled": True}
     with pytest.raises(ValueError, match="rollover_rate_data"):
         build_vec_environment(cfg, md)
+
+
+def test_engine_alias_gym_contract_with_profile_and_rate_file(tmp_path):
+    """The reference's nautilus-bridge contract test
+    (tests/test_nautilus_gym_bridge.py): simulation_engine='nautilus'
+    (an alias of the native engine here) + an execution-cost profile +
+    financing_rate_data_file CSV must still give a conforming Gymnasium
+    step tuple with the usual info keys."""
+    from pathlib import Path
+
+    from gymfx_amd import build_environment
+    from gymfx_amd.config import DEFAULT_VALUES
+    from gymfx_amd.data.feed import write_csv
+    from gymfx_amd.plugins import load_plugin
+
+    root = Path(__file__).resolve().parents[1]
+    md = synthetic_ohlcv(200, seed=9, vol=2e-4, start="2024-01-02 21:00:00")
+    data = tmp_path / "px.csv"
+    write_csv(md, str(data))
+    cfg = {
+        **DEFAULT_VALUES,
+        "simulation_engine": "nautilus",
+        "execution_cost_profile": str(
+            root / "examples/config/execution_cost_profiles/default.json"),
+        "financing_rate_data_file": str(
+            root / "examples/data/fx_rollover_rates_smoke.csv"),
+        "financing_enabled": True,
+        "input_data_file": str(data),
+        "window_size": 4, "initial_cash": 10000.0,
+        "position_size": 1000.0, "quiet_mode": True,
+    }
+    plugins = {}
+    for group, key in [("data_feed.plugins", "data_feed_plugin"),
+                       ("broker.plugins", "broker_plugin"),
+                       ("strategy.plugins", "strategy_plugin"),
+                       ("preprocessor.plugins", "preprocessor_plugin"),
+                       ("reward.plugins", "reward_plugin"),
+                       ("metrics.plugins", "metrics_plugin")]:
+        klass, _ = load_plugin(group, cfg[key])
+        plugins[key] = klass(cfg)
+    env = build_environment(config=cfg, **plugins)
+    obs, info = env.reset(seed=1)
+    assert env.observation_space.contains(obs)
+    out = env.step(1)
+    assert len(out) == 5
+    obs2, reward, terminated, truncated, info2 = out
+    assert isinstance(reward, float) and truncated is False
+    assert "equity" in info2 and "execution_diagnostics" in info2
+    env.close()
