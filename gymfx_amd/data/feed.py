@@ -122,24 +122,35 @@ def load_csv(
 
     timestamps = None
     date_strings = None
-    if date_column in df.columns:
-        raw_dates = df[date_column].astype(str).to_numpy()
+    # case-insensitive date-column detection; duplicates dropped (the
+    # reference matches 'DATE_TIME' with strip().lower() and removes the
+    # extras, app/data_handler.py:38-51)
+    want = date_column.strip().lower()
+    dt_cols = [c for c in df.columns if str(c).strip().lower() == want]
+    if dt_cols:
+        raw_dates = df[dt_cols[0]].astype(str).to_numpy()
         ts = _parse_timestamps(raw_dates)
         keep = ts != np.iinfo(np.int64).min
         df = df.loc[keep].reset_index(drop=True)
         timestamps = ts[keep]
         date_strings = raw_dates[keep]
-        df = df.drop(columns=[date_column])
+        df = df.drop(columns=dt_cols)
 
     if price_column not in df.columns:
-        raise ValueError(f"price_column '{price_column}' not found in data")
+        m = [c for c in df.columns
+             if str(c).strip().lower() == price_column.strip().lower()]
+        if not m:
+            raise ValueError(f"price_column '{price_column}' not found in data")
+        price_column = m[0]
 
     columns: Dict[str, np.ndarray] = {}
     for name in df.columns:
-        try:
-            columns[name] = df[name].to_numpy(dtype=np.float64)
-        except (TypeError, ValueError):
-            continue  # non-numeric auxiliary column
+        # numeric coercion with NaN->0 (data_handler.py:63-64); columns
+        # with no parseable value at all are treated as auxiliary and
+        # skipped rather than becoming all-zero features
+        vals = pd.to_numeric(df[name], errors="coerce")
+        if vals.notna().any() or len(vals) == 0:
+            columns[name] = vals.fillna(0.0).to_numpy(dtype=np.float64)
 
     for col in OHLC_COLUMNS:
         if col not in columns:

@@ -166,3 +166,24 @@ def test_warmup_zero_history(sample_market):
     }
     obs, _ = _vec_obs_at_steps(md, config, [1])
     assert np.all(obs.numpy() == 0.0)
+
+
+def test_load_csv_case_insensitive_and_nan_fill(tmp_path):
+    """Reference data_handler semantics (app/data_handler.py:38-64):
+    'date_time' matches case-insensitively (duplicates dropped), the
+    price column matches case-insensitively, and NaN cells in numeric
+    columns fill with 0."""
+    from gymfx_amd.data.feed import load_csv
+
+    p = tmp_path / "mixed.csv"
+    p.write_text(
+        "date_time,close,Volume,NOTE\n"
+        "2024-01-02 00:00:00,1.10,100,alpha\n"
+        "2024-01-02 00:01:00,,200,beta\n"
+        "2024-01-02 00:02:00,1.12,,gamma\n")
+    md = load_csv(str(p), date_column="DATE_TIME", price_column="CLOSE")
+    assert md.timestamps is not None and len(md.timestamps) == 3
+    assert "close" in md.columns
+    assert md.columns["close"][1] == 0.0       # NaN -> 0 (reference fill)
+    assert md.columns["Volume"][2] == 0.0
+    assert "NOTE" not in md.columns            # fully non-numeric: auxiliary
