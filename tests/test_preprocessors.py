@@ -187,3 +187,38 @@ def test_load_csv_case_insensitive_and_nan_fill(tmp_path):
     assert md.columns["close"][1] == 0.0       # NaN -> 0 (reference fill)
     assert md.columns["Volume"][2] == 0.0
     assert "NOTE" not in md.columns            # fully non-numeric: auxiliary
+
+
+def test_feature_aware_obs_space_excludes_disabled_raw_prices():
+    """With feature_columns configured, the observation space drops the
+    raw prices/returns blocks unless include_price_window is explicitly
+    on (reference test_feature_aware_observation_space_excludes_disabled_
+    raw_prices / test_legacy_observation_space_keeps_prices_returns)."""
+    from gymfx_amd.envs.gym_env import build_base_observation_space
+
+    feat = build_base_observation_space(
+        {"feature_columns": FEATURES}, window_size=8)
+    assert "features" in feat.spaces
+    assert "prices" not in feat.spaces and "returns" not in feat.spaces
+    both = build_base_observation_space(
+        {"feature_columns": FEATURES, "include_price_window": True},
+        window_size=8)
+    assert {"features", "prices", "returns"} <= set(both.spaces)
+    legacy = build_base_observation_space({}, window_size=8)
+    assert "prices" in legacy.spaces and "features" not in legacy.spaces
+
+
+def test_missing_and_empty_feature_columns_raise(sample_market):
+    """Configured-but-absent feature columns and an empty feature list
+    both fail loudly (reference test_missing_columns_raises /
+    test_empty_feature_list_raises)."""
+    pre = FeatureWindowPreprocessor({})
+    with pytest.raises(ValueError, match="missing from data"):
+        pre.make_observation(
+            data=sample_market, step=4, bridge_state={},
+            config={"feature_columns": ["CLOSE", "NO_SUCH_COL"],
+                    "window_size": 4})
+    with pytest.raises(ValueError, match="non-empty"):
+        pre.make_observation(
+            data=sample_market, step=4, bridge_state={},
+            config={"feature_columns": [], "window_size": 4})
