@@ -686,8 +686,17 @@ def train_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
     vec_cfg.setdefault("env_start_mode", "spread")
     env = build_vec_environment(vec_cfg)
     env.reset(seed=cfg.seed)
-    trainer = PPOTrainer(env, cfg)
+    # torchrun-launched config training joins the data-parallel group the
+    # same way bench.py does (one rank per GPU over RCCL; gloo on CPU) —
+    # without this, WORLD_SIZE>1 would train unsynchronized replicas.
+    from ..parallel.ddp import init_from_env
+
+    rank, world_size, pg = init_from_env(env.device)
+    trainer = PPOTrainer(env, cfg, rank=rank, world_size=world_size,
+                         process_group=pg)
     ckpt_path = config.get("checkpoint_file")
+    if rank != 0:
+        ckpt_path = None  # only rank 0 writes checkpoints/results
     resumed = False
     if ckpt_path and config.get("resume"):
         from ..utils.checkpoint import load_checkpoint
@@ -736,13 +745,15 @@ def train_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
         save_checkpoint(trainer, ckpt_path,
                         extra={"config": {k: v for k, v in config.items()
                                           if isinstance(v, (int, float, str, bool, type(None)))}})
-    steps = trainer.global_step * env.n_envs
+    steps = trainer.global_step * env.n_envs * world_size
     summary = {
         "mode": "training",
         "updates": updates,
         "resumed": resumed,
         "checkpoint_file": ckpt_path,
         "policy_model": cfg.policy,
+        "rank": rank,
+        "world_size": world_size,
         "env_steps": steps,
         "wall_seconds": wall,
         "env_steps_per_sec": steps / wall if wall > 0 else 0.0,

@@ -5,6 +5,7 @@ env, drive it with the configured driver, write results + config."""
 from __future__ import annotations
 
 import json
+import os
 from pathlib import Path
 from typing import Any, Dict
 
@@ -131,6 +132,11 @@ def main(argv=None) -> None:
         summary = evaluate_from_config(config)
     else:
         summary = run_env(config)
+
+    # under torchrun only rank 0 writes results/logs (mode=training joins
+    # the data-parallel group inside train_from_config)
+    if int(os.environ.get("RANK", "0")) != 0:
+        return
 
     results_file = Path(config.get("results_file", "results.json"))
     results_file.parent.mkdir(parents=True, exist_ok=True)

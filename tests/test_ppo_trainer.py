@@ -316,3 +316,48 @@ def test_optimize_train_driver_trains_per_trial():
     }
     out = optimize_from_config(cfg)
     assert out["trials"] == 2 and "best_params" in out
+
+
+def test_mode_training_under_torchrun_ws2_synchronizes(tmp_path):
+    """python -m gymfx_amd.main --mode training under torchrun ws=2 must
+    join the data-parallel group (gloo here, RCCL on GPUs): both ranks'
+    models see all-reduced gradients, only rank 0 writes the checkpoint
+    and results, and the checkpoint stays loadable."""
+    import json
+    import socket
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    import torch
+
+    repo = Path(__file__).resolve().parents[1]
+    from gymfx_amd.config import DEFAULT_VALUES
+    from gymfx_amd.data.feed import synthetic_ohlcv, write_csv
+
+    write_csv(synthetic_ohlcv(600, seed=5, vol=2e-4), str(tmp_path / "px.csv"))
+    cfg = {**DEFAULT_VALUES, "mode": "training",
+           "input_data_file": str(tmp_path / "px.csv"),
+           "n_envs": 8, "window_size": 6, "device": "cpu",
+           "train_updates": 2, "rollout_steps": 8, "minibatches": 2,
+           "ppo_epochs": 1, "hidden_size": 16, "seed": 3,
+           "quiet_mode": True,
+           "checkpoint_file": str(tmp_path / "ck.pt"),
+           "results_file": str(tmp_path / "results.json")}
+    cfg_path = tmp_path / "train.json"
+    cfg_path.write_text(json.dumps(cfg))
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), "-m", "gymfx_amd.main",
+         "--load_config", str(cfg_path)],
+        capture_output=True, text=True, timeout=420, cwd=repo)
+    assert r.returncode == 0, r.stderr[-2500:]
+    res = json.loads((tmp_path / "results.json").read_text())
+    assert res["world_size"] == 2 and res["rank"] == 0
+    sd = torch.load(tmp_path / "ck.pt", map_location="cpu",
+                    weights_only=True)
+    assert sd["policy"] == "mlp"
