@@ -11,6 +11,8 @@ training.
 Endpoints
 ---------
 GET  /health            -> {status, policy, obs_dim, n_actions, device}
+GET  /metrics           -> Prometheus exposition (request/action counters,
+                           batch-latency histogram, live session gauge)
 POST /act               -> {"obs": [[f32 x obs_dim] x B], "greedy": bool,
                             "session": str|null}
                         -> {actions, logp, value}
@@ -158,6 +160,29 @@ def create_app(config: Dict[str, Any]):
     app = FastAPI(title="gymfx-amd policy server")
     app.state.server = server
 
+    # Prometheus observability (production serving): a private registry so
+    # repeated create_app calls (tests, multi-app processes) don't collide
+    # in the global default registry.
+    from prometheus_client import (CollectorRegistry, Counter, Gauge,
+                                   Histogram, generate_latest,
+                                   CONTENT_TYPE_LATEST)
+
+    registry = CollectorRegistry()
+    m_requests = Counter("gymfx_serve_requests_total",
+                         "act requests served", ["outcome"],
+                         registry=registry)
+    m_actions = Counter("gymfx_serve_actions_total",
+                        "actions produced", registry=registry)
+    m_latency = Histogram(
+        "gymfx_serve_batch_latency_seconds",
+        "wall time of one batched act() call",
+        buckets=(.0005, .001, .0025, .005, .01, .025, .05, .1, .25, 1.0),
+        registry=registry)
+    m_sessions = Gauge("gymfx_serve_live_sessions",
+                       "recurrent sessions held server-side",
+                       registry=registry)
+    m_sessions.set_function(lambda: float(len(server._sessions)))
+
     class ActRequest(BaseModel):
         obs: List[List[float]]
         greedy: bool = True
@@ -179,13 +204,29 @@ def create_app(config: Dict[str, Any]):
 
     @app.post("/act")
     def act(req: ActRequest):
+        import time as _time
+
+        t0 = _time.perf_counter()
         try:
             arr = np.asarray(req.obs, dtype=np.float32)
-            return server.act(arr, greedy=req.greedy, session=req.session)
+            out = server.act(arr, greedy=req.greedy, session=req.session)
         except SessionBatchMismatch as exc:
+            m_requests.labels("conflict").inc()
             raise HTTPException(status_code=409, detail=str(exc))
         except ValueError as exc:
+            m_requests.labels("invalid").inc()
             raise HTTPException(status_code=422, detail=str(exc))
+        m_latency.observe(_time.perf_counter() - t0)
+        m_requests.labels("ok").inc()
+        m_actions.inc(len(out["actions"]))
+        return out
+
+    @app.get("/metrics")
+    def metrics():
+        from fastapi import Response
+
+        return Response(generate_latest(registry),
+                        media_type=CONTENT_TYPE_LATEST)
 
     @app.post("/session/reset")
     def session_reset(req: SessionRequest):

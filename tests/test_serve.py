@@ -164,3 +164,27 @@ def test_serve_session_map_bounded_lru(tmp_path):
     srv.act(obs, session="k6")
     srv.act(obs, session="new")
     assert "k6" in srv._sessions and "k7" not in srv._sessions
+
+
+def test_serve_prometheus_metrics(tmp_path):
+    """GET /metrics exposes request/action counters, the batch-latency
+    histogram and the live-session gauge (production observability —
+    SURVEY §5.5's 'structured logging/metrics' requirement on the
+    serving side)."""
+    from fastapi.testclient import TestClient
+
+    from gymfx_amd.serve import create_app
+
+    cfg = _serve_cfg(tmp_path, "lstm")
+    client = TestClient(create_app(cfg))
+    D = client.get("/health").json()["obs_dim"]
+    obs = np.zeros((3, D), dtype=np.float32).tolist()
+    assert client.post("/act", json={"obs": obs, "session": "m1"}).status_code == 200
+    # a 409 must count as a conflict, not an ok
+    assert client.post("/act", json={"obs": obs[:2], "session": "m1"}).status_code == 409
+    text = client.get("/metrics").text
+    assert 'gymfx_serve_requests_total{outcome="ok"} 1.0' in text
+    assert 'gymfx_serve_requests_total{outcome="conflict"} 1.0' in text
+    assert "gymfx_serve_actions_total 3.0" in text
+    assert "gymfx_serve_batch_latency_seconds_bucket" in text
+    assert "gymfx_serve_live_sessions 1.0" in text
