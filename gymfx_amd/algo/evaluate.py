@@ -82,6 +82,16 @@ def evaluate_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
     wall = time.perf_counter() - t0
     ic = env.params.initial_cash
     eq = env.st.equity
+    # fleet-level trading-metrics digest through the CONFIGURED metrics
+    # plugin (the reference's inference mode reports the metrics plugin's
+    # summary, app/main.py:66-69 -> env.summary())
+    from ..plugins import load_plugin
+
+    mname = str(config.get("metrics_plugin", "trading_metrics"))
+    klass, _ = load_plugin("metrics.plugins", mname)
+    metrics = klass(config).summarize(
+        initial_cash=ic, final_equity=float(eq.mean()),
+        analyzers=env.fleet_analyzers(), config=config)
     return {
         "mode": "inference",
         "policy_model": policy_name,
@@ -94,5 +104,7 @@ def evaluate_from_config(config: Dict[str, Any]) -> Dict[str, Any]:
         "mean_step_reward": float((total_reward / max(n_steps, 1)).mean()),
         "terminated_envs": int(env.st.terminated.sum()),
         "total_trades": int(env.st.trade_count.sum()),
+        "metrics_plugin": mname,
+        "metrics": metrics,
         "vec_summary": env.vec_summary(),
     }

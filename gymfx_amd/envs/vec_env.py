@@ -332,6 +332,50 @@ class VecFxEnv:
             "time_return": {},
         }
 
+    def fleet_analyzers(self) -> Dict[str, Any]:
+        """Fleet-pooled analyzer digest (same shape as :meth:`analyzers`)
+        so the metrics plugins can summarize a vectorized evaluation run:
+        trade counters and pnl moments sum over envs, the Sharpe pools
+        every env's step returns, drawdown reports the fleet mean (pct)
+        and the worst env (money)."""
+        st = self.st
+        n_tr = int(st.trade_count.sum().item())
+        pnl_sum = float(st.trade_pnl_sum.sum().item())
+        pnl_sumsq = float(st.trade_pnl_sumsq.sum().item())
+        sqn = None
+        if n_tr >= 2:
+            mean = pnl_sum / n_tr
+            var = max(pnl_sumsq / n_tr - mean * mean, 0.0)
+            if var > 0:
+                sqn = (n_tr ** 0.5) * mean / (var ** 0.5)
+        n_ret = int(st.ret_count.sum().item())
+        sharpe = None
+        if n_ret >= 2:
+            rs = float(st.ret_sum.sum().item())
+            rss = float(st.ret_sumsq.sum().item())
+            mean = rs / n_ret
+            var = max((rss - n_ret * mean * mean) / (n_ret - 1), 0.0)
+            if var > 0:
+                ann = float(self.config.get("annualization_factor", 252.0))
+                sharpe = mean / (var ** 0.5) * (ann ** 0.5)
+        return {
+            "trades": {
+                "total": {"total": n_tr},
+                "won": {"total": int(st.trade_won.sum().item())},
+                "lost": {"total": int(st.trade_lost.sum().item())},
+                "pnl": {"net": {"average": (pnl_sum / n_tr) if n_tr else None}},
+            },
+            "sharpe": {"sharperatio": sharpe},
+            "drawdown": {
+                "max": {
+                    "drawdown": float(st.max_dd_pct.mean().item()),
+                    "moneydown": float(st.max_dd_money.max().item()),
+                }
+            },
+            "sqn": {"sqn": sqn},
+            "time_return": {},
+        }
+
     def vec_summary(self) -> Dict[str, Any]:
         """Whole-fleet aggregates (no reference counterpart)."""
         st = self.st

@@ -227,7 +227,11 @@ def test_evaluate_loads_checkpoint_from_different_n_envs(tmp_path):
            "data_feed_plugin": "synthetic_data_feed", "synthetic_rows": 500,
            "n_envs": 4, "window_size": 8, "device": "cpu", "seed": 9,
            "checkpoint_file": path, "eval_steps": 12, "hidden_size": 16,
-           "position_size": 1000.0}
+           "position_size": 1000.0, "metrics_plugin": "trading_metrics"}
     out = evaluate_from_config(cfg)
     assert out["policy_model"] == "lstm" and out["n_envs"] == 4
     assert out["eval_steps"] > 0
+    # fleet-level digest through the configured metrics plugin
+    assert out["metrics"]["metric_schema"] == "trading.metrics.v1"
+    assert out["metrics"]["initial_cash"] == cfg["initial_cash"]
+    assert "risk_adjusted_return" in out["metrics"] or "rap" in out["metrics"]
