@@ -343,7 +343,9 @@ def test_mode_training_under_torchrun_ws2_synchronizes(tmp_path):
            "ppo_epochs": 1, "hidden_size": 16, "seed": 3,
            "quiet_mode": True,
            "checkpoint_file": str(tmp_path / "ck.pt"),
-           "results_file": str(tmp_path / "results.json")}
+           "results_file": str(tmp_path / "results.json"),
+           "save_config": str(tmp_path / "config_out.json"),
+           "save_log": None}
     cfg_path = tmp_path / "train.json"
     cfg_path.write_text(json.dumps(cfg))
     with socket.socket() as s:
