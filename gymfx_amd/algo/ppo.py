@@ -122,6 +122,12 @@ class PPOTrainer:
         self.world_size = world_size
         self.pg = process_group
         self.device = env.device
+        if getattr(env.params, "action_space_mode", "discrete") != "discrete":
+            raise ValueError(
+                "PPO training uses the 3-way discrete head; set "
+                "action_space_mode='discrete' (continuous mode is for "
+                "external agents driving the env, reference "
+                "env.py:343-360 coercion semantics)")
 
         N = env.n_envs
         T = cfg.rollout_steps

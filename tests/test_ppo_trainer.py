@@ -363,3 +363,23 @@ def test_mode_training_under_torchrun_ws2_synchronizes(tmp_path):
     sd = torch.load(tmp_path / "ck.pt", map_location="cpu",
                     weights_only=True)
     assert sd["policy"] == "mlp"
+
+
+def test_trainer_rejects_continuous_action_mode(tmp_path):
+    """PPO's head is 3-way discrete; a continuous-mode env must be
+    refused at construction with a clear message, not a kernel dtype
+    error mid-rollout."""
+    import pytest as _pytest
+    import torch
+
+    from gymfx_amd import build_vec_environment
+    from gymfx_amd.algo.ppo import PPOConfig, PPOTrainer
+    from gymfx_amd.data.feed import synthetic_ohlcv
+
+    md = synthetic_ohlcv(200, seed=2, vol=1e-4)
+    env = build_vec_environment(
+        {"n_envs": 4, "device": "cpu", "window_size": 4,
+         "action_space_mode": "continuous"}, md)
+    with _pytest.raises(ValueError, match="discrete"):
+        PPOTrainer(env, PPOConfig(rollout_steps=4, minibatches=2,
+                                  ppo_epochs=1, hidden=16))
